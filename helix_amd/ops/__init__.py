@@ -1,0 +1,173 @@
+"""helix_amd.ops — CDNA4 (gfx950) kernel dispatch.
+
+GPU tensors run the hand-written HIP kernels in helix_amd._C and FAIL
+LOUDLY if the extension is missing (no silent eager fallback on a GPU
+box). CPU tensors use the pure-torch reference implementations so engine
+logic is testable without a GPU.
+"""
+from __future__ import annotations
+
+import torch
+
+from . import reference as ref
+from .reference import make_cos_sin_cache  # re-export
+
+_C = None
+_C_ERR: str | None = None
+try:
+    from helix_amd import _C  # type: ignore  # noqa: F401
+    from helix_amd import _C as _C_mod
+    _C = _C_mod
+except ImportError as e:  # pragma: no cover
+    _C_ERR = str(e)
+
+
+def _native():
+    if _C is None:
+        raise RuntimeError(
+            "helix_amd._C native extension is not built but a GPU tensor was "
+            "passed. Build it with `python setup.py build_ext --inplace` "
+            f"(PYTORCH_ROCM_ARCH=gfx950). Import error: {_C_ERR}")
+    return _C
+
+
+def have_native() -> bool:
+    return _C is not None
+
+
+def rms_norm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native().rms_norm(out, x, w, eps)
+        return out
+    return ref.rms_norm(x, w, eps)
+
+
+def fused_add_rms_norm(x: torch.Tensor, residual: torch.Tensor,
+                       w: torch.Tensor, eps: float):
+    """In-place on GPU: x <- norm(x+res), residual <- x+res. Returns both."""
+    if x.is_cuda:
+        _native().fused_add_rms_norm(x, residual, w, eps)
+        return x, residual
+    out, new_res = ref.fused_add_rms_norm(x, residual, w, eps)
+    x.copy_(out)
+    residual.copy_(new_res)
+    return x, residual
+
+
+def layer_norm(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
+               eps: float) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native().layer_norm(out, x, w, b, eps)
+        return out
+    return ref.layer_norm(x, w, b, eps)
+
+
+def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
+                     k: torch.Tensor, cos_sin: torch.Tensor, head_dim: int):
+    """In-place on GPU. Returns (q, k)."""
+    if q.is_cuda:
+        _native().rotary_embedding(positions, q, k, cos_sin, head_dim)
+        return q, k
+    qo, ko = ref.rotary_embedding(positions, q, k, cos_sin, head_dim)
+    q.copy_(qo)
+    k.copy_(ko)
+    return q, k
+
+
+def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        i = x.shape[-1] // 2
+        out = torch.empty(*x.shape[:-1], i, dtype=x.dtype, device=x.device)
+        _native().silu_and_mul(out, x)
+        return out
+    return ref.silu_and_mul(x)
+
+
+def gelu_tanh(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty_like(x)
+        _native().gelu_tanh(out, x)
+        return out
+    return ref.gelu_tanh(x)
+
+
+def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
+                      k_cache: torch.Tensor, v_cache: torch.Tensor,
+                      slot_mapping: torch.Tensor):
+    if k.is_cuda:
+        _native().reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+        return
+    ref.reshape_and_cache(k, v, k_cache, v_cache, slot_mapping)
+
+
+def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 cu_seqlens: torch.Tensor, max_seqlen: int,
+                 scale: float) -> torch.Tensor:
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        _native().attn_prefill(out, q, k, v, cu_seqlens, max_seqlen, scale)
+        return out
+    return ref.attn_prefill(q, k, v, cu_seqlens, max_seqlen, scale)
+
+
+# Flash-decoding split-K sizing (must match paged_attn_decode.hip).
+DECODE_PARTITION = 512
+
+
+def decode_workspace(max_batch: int, num_q_heads: int, head_dim: int,
+                     max_seq_len: int, device) -> tuple[torch.Tensor, torch.Tensor]:
+    """Persistent fp32 scratch for partitioned decode attention."""
+    max_parts = max(1, (max_seq_len + DECODE_PARTITION - 1) // DECODE_PARTITION)
+    tmp_out = torch.empty(max_batch, num_q_heads, max_parts, head_dim,
+                          dtype=torch.float32, device=device)
+    tmp_ml = torch.empty(max_batch, num_q_heads, max_parts, 2,
+                         dtype=torch.float32, device=device)
+    return tmp_out, tmp_ml
+
+
+def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                      v_cache: torch.Tensor, block_tables: torch.Tensor,
+                      seq_lens: torch.Tensor, scale: float,
+                      workspace=None) -> torch.Tensor:
+    if q.is_cuda:
+        out = torch.empty_like(q)
+        if workspace is None:
+            max_len = int(seq_lens.max())
+            workspace = decode_workspace(q.shape[0], q.shape[1], q.shape[2],
+                                         max_len, q.device)
+        tmp_out, tmp_ml = workspace
+        _native().paged_attn_decode(out, q, k_cache, v_cache, block_tables,
+                                    seq_lens, scale,
+                                    tmp_out[:q.shape[0]], tmp_ml[:q.shape[0]],
+                                    DECODE_PARTITION)
+        return out
+    return ref.paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens,
+                                 scale)
+
+
+def sample_tokens(logits: torch.Tensor, temperatures: torch.Tensor,
+                  seeds: torch.Tensor) -> torch.Tensor:
+    if logits.is_cuda:
+        out = torch.empty(logits.shape[0], dtype=torch.int64,
+                          device=logits.device)
+        _native().sample_tokens(out, logits, temperatures, seeds)
+        return out
+    return ref.sample_tokens(logits, temperatures, seeds)
+
+
+def gemm_bf16(x: torch.Tensor, w: torch.Tensor, bias=None,
+              act: int = 0) -> torch.Tensor:
+    if x.is_cuda:
+        out = torch.empty(x.shape[0], w.shape[0], dtype=x.dtype,
+                          device=x.device)
+        _native().gemm_bf16(out, x, w, bias, act)
+        return out
+    return ref.gemm_bf16(x, w, bias, act)
+
+
+def mfma_probe(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    d = torch.empty(16, 16, dtype=torch.float32, device=a.device)
+    _native().mfma_probe(d, a, b)
+    return d

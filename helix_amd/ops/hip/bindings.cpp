@@ -1,0 +1,49 @@
+// Python bindings for the helix_amd CDNA4 kernel library.
+#include <torch/extension.h>
+
+void rms_norm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps);
+void fused_add_rms_norm(torch::Tensor x, torch::Tensor residual,
+                        torch::Tensor w, double eps);
+void layer_norm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                torch::Tensor b, double eps);
+void rotary_embedding(torch::Tensor positions, torch::Tensor q,
+                      torch::Tensor k, torch::Tensor cos_sin,
+                      int64_t head_dim);
+void silu_and_mul(torch::Tensor out, torch::Tensor x);
+void gelu_tanh(torch::Tensor out, torch::Tensor x);
+void reshape_and_cache(torch::Tensor k, torch::Tensor v,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor slot_mapping);
+void paged_attn_decode(torch::Tensor out, torch::Tensor q,
+                       torch::Tensor k_cache, torch::Tensor v_cache,
+                       torch::Tensor block_tables, torch::Tensor seq_lens,
+                       double scale, torch::Tensor tmp_out,
+                       torch::Tensor tmp_ml, int64_t partition_size);
+void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
+                  torch::Tensor v, torch::Tensor cu_seqlens,
+                  int64_t max_seqlen, double scale);
+void sample_tokens(torch::Tensor out, torch::Tensor logits,
+                   torch::Tensor temperatures, torch::Tensor seeds);
+void gemm_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+               c10::optional<torch::Tensor> bias, int64_t act);
+void mfma_probe(torch::Tensor d, torch::Tensor a, torch::Tensor b);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "RMSNorm (bf16, CDNA4)");
+  m.def("fused_add_rms_norm", &fused_add_rms_norm,
+        "Fused residual-add + RMSNorm (in-place)");
+  m.def("layer_norm", &layer_norm, "LayerNorm (bf16)");
+  m.def("rotary_embedding", &rotary_embedding,
+        "Apply rotary embedding to q,k in-place");
+  m.def("silu_and_mul", &silu_and_mul, "Fused SiLU-gated multiply");
+  m.def("gelu_tanh", &gelu_tanh, "GELU (tanh approx)");
+  m.def("reshape_and_cache", &reshape_and_cache,
+        "Scatter K/V rows into paged cache");
+  m.def("paged_attn_decode", &paged_attn_decode,
+        "Paged decode attention (GQA, flash-decoding partitions)");
+  m.def("attn_prefill", &attn_prefill,
+        "Varlen causal flash prefill attention (MFMA)");
+  m.def("sample_tokens", &sample_tokens, "Greedy/Gumbel token sampling");
+  m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM: x @ w^T (+bias, act)");
+  m.def("mfma_probe", &mfma_probe, "16x16x32 MFMA layout probe");
+}

@@ -1,0 +1,109 @@
+// Token sampling kernels (CDNA4).
+//
+// Replaces the "logits -> next token" sampling the reference delegates to
+// vLLM (SURVEY.md §2.8 "Logits GEMM + sampling").
+//
+// Exact temperature sampling without a vocab sort via the Gumbel-max trick:
+//   next = argmax(logits / T + G_i),  G_i = -log(-log(U_i))
+// Greedy (T == 0) is a plain argmax. One 256-thread block per row;
+// vectorized bf16x8 logit reads (memory-bound over V ~ 128k).
+// top-k / top-p restriction is applied upstream (engine masks logits).
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+using namespace helix;
+
+namespace {
+
+template <typename T>
+__device__ __forceinline__ float as_f32(T v);
+template <>
+__device__ __forceinline__ float as_f32<uint16_t>(uint16_t v) {
+  return bf16_to_f32(v);
+}
+template <>
+__device__ __forceinline__ float as_f32<float>(float v) {
+  return v;
+}
+
+template <typename T>
+__global__ void sample_kernel(int64_t* __restrict__ out,
+                              const T* __restrict__ logits,
+                              const float* __restrict__ temperatures,
+                              const uint64_t* __restrict__ seeds, int V) {
+  const int row = blockIdx.x;
+  const T* lrow = logits + (int64_t)row * V;
+  const float temp = temperatures[row];
+  const uint64_t seed = seeds[row];
+  const bool greedy = temp <= 0.f;
+  const float inv_t = greedy ? 1.f : 1.f / temp;
+
+  float best = -INFINITY;
+  int best_i = 0;
+  for (int i = threadIdx.x; i < V; i += blockDim.x) {
+    float val = as_f32(lrow[i]) * inv_t;
+    if (!greedy) {
+      const float u = u64_to_uniform(splitmix64(seed ^ (uint64_t)i));
+      val += -__logf(-__logf(u));
+    }
+    if (val > best || (val == best && i < best_i)) {
+      best = val;
+      best_i = i;
+    }
+  }
+  // block argmax reduce
+  __shared__ float smax[4];
+  __shared__ int sidx[4];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ov = __shfl_xor(best, off, WAVE);
+    const int oi = __shfl_xor(best_i, off, WAVE);
+    if (ov > best || (ov == best && oi < best_i)) {
+      best = ov;
+      best_i = oi;
+    }
+  }
+  if (lane == 0) {
+    smax[wid] = best;
+    sidx[wid] = best_i;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < (int)(blockDim.x / WAVE); ++w) {
+      if (smax[w] > best || (smax[w] == best && sidx[w] < best_i)) {
+        best = smax[w];
+        best_i = sidx[w];
+      }
+    }
+    out[row] = best_i;
+  }
+}
+
+}  // namespace
+
+void sample_tokens(torch::Tensor out, torch::Tensor logits,
+                   torch::Tensor temperatures, torch::Tensor seeds) {
+  const int B = logits.size(0);
+  const int V = logits.size(1);
+  TORCH_CHECK(out.scalar_type() == torch::kInt64);
+  TORCH_CHECK(temperatures.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(seeds.scalar_type() == torch::kUInt64 ||
+              seeds.scalar_type() == torch::kInt64);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (logits.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((sample_kernel<uint16_t>), dim3(B), dim3(256), 0,
+                       stream, out.data_ptr<int64_t>(),
+                       (const uint16_t*)logits.data_ptr(),
+                       temperatures.data_ptr<float>(),
+                       (const uint64_t*)seeds.data_ptr(), V);
+  } else {
+    TORCH_CHECK(logits.scalar_type() == torch::kFloat32);
+    hipLaunchKernelGGL((sample_kernel<float>), dim3(B), dim3(256), 0, stream,
+                       out.data_ptr<int64_t>(), logits.data_ptr<float>(),
+                       temperatures.data_ptr<float>(),
+                       (const uint64_t*)seeds.data_ptr(), V);
+  }
+}

@@ -1,0 +1,210 @@
+"""GPU numerics tests: every CDNA4 HIP kernel vs the fp32 torch reference.
+
+All marked `gpu`; run on an MI355X via gpurun. Tolerances are bf16-scale.
+"""
+import pytest
+import torch
+
+import helix_amd.ops as ops
+import helix_amd.ops.reference as ref
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def assert_close_bf16(a, b, atol=2e-2, rtol=2e-2, msg=""):
+    torch.testing.assert_close(a.float().cpu(), b.float().cpu(), atol=atol,
+                               rtol=rtol, msg=msg)
+
+
+def test_native_loaded():
+    assert ops.have_native(), "HIP extension must be loaded on a GPU box"
+
+
+def test_mfma_probe_layout():
+    """Verify the assumed 16x16x32 bf16 MFMA fragment layouts with
+    asymmetric random operands (guide §3)."""
+    torch.manual_seed(0)
+    a = torch.randn(16, 32, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn(32, 16, dtype=torch.bfloat16, device=DEV)
+    d = ops.mfma_probe(a, b)
+    want = a.float() @ b.float()
+    assert_close_bf16(d, want, atol=5e-2, rtol=5e-2,
+                      msg="MFMA fragment layout mismatch")
+
+
+@pytest.mark.parametrize("H", [768, 4096, 8192])
+def test_rms_norm(H):
+    torch.manual_seed(0)
+    x = torch.randn(33, H, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(H, dtype=torch.bfloat16, device=DEV)
+    got = ops.rms_norm(x, w, 1e-5)
+    want = ref.rms_norm(x.cpu(), w.cpu(), 1e-5)
+    assert_close_bf16(got, want)
+
+
+def test_fused_add_rms_norm():
+    torch.manual_seed(1)
+    x = torch.randn(17, 4096, dtype=torch.bfloat16, device=DEV)
+    r = torch.randn(17, 4096, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(4096, dtype=torch.bfloat16, device=DEV)
+    want_out, want_res = ref.fused_add_rms_norm(x.cpu(), r.cpu(), w.cpu(), 1e-5)
+    got_out, got_res = ops.fused_add_rms_norm(x, r, w, 1e-5)
+    assert_close_bf16(got_res, want_res)
+    assert_close_bf16(got_out, want_out)
+
+
+def test_layer_norm():
+    torch.manual_seed(2)
+    x = torch.randn(9, 768, dtype=torch.bfloat16, device=DEV)
+    w = torch.randn(768, dtype=torch.bfloat16, device=DEV)
+    b = torch.randn(768, dtype=torch.bfloat16, device=DEV)
+    got = ops.layer_norm(x, w, b, 1e-5)
+    want = ref.layer_norm(x.cpu(), w.cpu(), b.cpu(), 1e-5)
+    assert_close_bf16(got, want)
+
+
+@pytest.mark.parametrize("hq,hk,d", [(32, 8, 128), (12, 12, 64)])
+def test_rope(hq, hk, d):
+    torch.manual_seed(3)
+    T = 21
+    q = torch.randn(T, hq * d, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, hk * d, dtype=torch.bfloat16, device=DEV)
+    pos = torch.randint(0, 500, (T,), dtype=torch.int64, device=DEV)
+    cs = ref.make_cos_sin_cache(d, 512).to(DEV)
+    want_q, want_k = ref.rotary_embedding(pos.cpu(), q.cpu(), k.cpu(),
+                                          cs.cpu(), d)
+    got_q, got_k = ops.rotary_embedding(pos, q, k, cs, d)
+    assert_close_bf16(got_q, want_q)
+    assert_close_bf16(got_k, want_k)
+
+
+def test_silu_and_mul():
+    torch.manual_seed(4)
+    x = torch.randn(13, 2 * 14336, dtype=torch.bfloat16, device=DEV)
+    got = ops.silu_and_mul(x)
+    want = ref.silu_and_mul(x.cpu())
+    assert_close_bf16(got, want)
+
+
+def test_gelu_tanh():
+    torch.manual_seed(5)
+    x = torch.randn(13, 3072, dtype=torch.bfloat16, device=DEV)
+    got = ops.gelu_tanh(x)
+    want = ref.gelu_tanh(x.cpu())
+    assert_close_bf16(got, want)
+
+
+def test_reshape_and_cache():
+    torch.manual_seed(6)
+    T, Hkv, D, bs, nb = 10, 8, 128, 16, 4
+    k = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, Hkv, D, dtype=torch.bfloat16, device=DEV)
+    kc = torch.zeros(nb, Hkv, bs, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    slots = torch.tensor([0, 1, 17, 18, 19, 35, 36, 63, 5, 40],
+                         dtype=torch.int64, device=DEV)
+    kc_ref, vc_ref = kc.cpu().clone(), vc.cpu().clone()
+    ref.reshape_and_cache(k.cpu(), v.cpu(), kc_ref, vc_ref, slots.cpu())
+    ops.reshape_and_cache(k, v, kc, vc, slots)
+    assert_close_bf16(kc, kc_ref, atol=0, rtol=0)
+    assert_close_bf16(vc, vc_ref, atol=0, rtol=0)
+
+
+@pytest.mark.parametrize("hq,hkv,d,lens", [
+    (32, 8, 128, [1, 17, 128, 63]),
+    (8, 8, 128, [200]),
+    (12, 12, 64, [33, 64]),
+    (32, 8, 128, [512, 300]),
+])
+def test_attn_prefill(hq, hkv, d, lens):
+    torch.manual_seed(7)
+    T = sum(lens)
+    q = torch.randn(T, hq, d, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, hkv, d, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, hkv, d, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    scale = d ** -0.5
+    got = ops.attn_prefill(q, k, v, cu, max(lens), scale)
+    want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), max(lens),
+                            scale)
+    assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("hq,hkv,d,lens", [
+    (32, 8, 128, [1, 16, 100, 333]),
+    (8, 8, 128, [257]),
+    (64, 8, 128, [90, 1024]),
+    (12, 12, 64, [50]),
+    (32, 8, 128, [2048]),   # exercises split-K partitions
+])
+def test_paged_attn_decode(hq, hkv, d, lens):
+    torch.manual_seed(8)
+    B = len(lens)
+    bs = 16
+    max_blocks = (max(lens) + bs - 1) // bs
+    total_blocks = sum((l + bs - 1) // bs for l in lens) + 1
+    q = torch.randn(B, hq, d, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(total_blocks, hkv, bs, d, dtype=torch.bfloat16,
+                     device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(B, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for b, l in enumerate(lens):
+        n = (l + bs - 1) // bs
+        bt[b, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    scale = d ** -0.5
+    got = ops.paged_attn_decode(q, kc, vc, bt, seq_lens, scale)
+    want = ref.paged_attn_decode(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                                 seq_lens.cpu(), scale)
+    assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
+
+
+def test_sample_greedy():
+    torch.manual_seed(9)
+    logits = torch.randn(64, 32000, dtype=torch.bfloat16, device=DEV)
+    temps = torch.zeros(64, dtype=torch.float32, device=DEV)
+    seeds = torch.arange(64, dtype=torch.int64, device=DEV)
+    got = ops.sample_tokens(logits, temps, seeds)
+    want = logits.float().argmax(-1)
+    assert torch.equal(got.cpu(), want.cpu())
+
+
+def test_sample_gumbel_distribution():
+    """Gumbel-max sampling should land near the softmax distribution."""
+    torch.manual_seed(10)
+    V = 8
+    logits_row = torch.tensor([0.0, 1.0, 2.0, 0.5, -1.0, 0.0, 3.0, -2.0])
+    N = 4096
+    logits = logits_row.expand(N, V).contiguous().bfloat16().to(DEV)
+    temps = torch.ones(N, dtype=torch.float32, device=DEV)
+    seeds = torch.arange(N, dtype=torch.int64, device=DEV) * 7919
+    got = ops.sample_tokens(logits, temps, seeds)
+    counts = torch.bincount(got.cpu(), minlength=V).float() / N
+    want = torch.softmax(logits_row, -1)
+    assert (counts - want).abs().max() < 0.05, (counts, want)
+
+
+@pytest.mark.parametrize("m,n,k", [(128, 128, 64), (200, 768, 768),
+                                   (512, 3072, 768), (1000, 768, 3072)])
+def test_gemm_bf16(m, n, k):
+    torch.manual_seed(11)
+    x = torch.randn(m, k, dtype=torch.bfloat16, device=DEV) / (k ** 0.25)
+    w = torch.randn(n, k, dtype=torch.bfloat16, device=DEV) / (k ** 0.25)
+    bias = torch.randn(n, dtype=torch.bfloat16, device=DEV)
+    got = ops.gemm_bf16(x, w, bias)
+    want = ref.gemm_bf16(x.cpu(), w.cpu(), bias.cpu())
+    assert_close_bf16(got, want, atol=5e-2, rtol=5e-2)
+
+
+def test_gemm_bf16_gelu():
+    torch.manual_seed(12)
+    x = torch.randn(100, 768, dtype=torch.bfloat16, device=DEV) / 5
+    w = torch.randn(3072, 768, dtype=torch.bfloat16, device=DEV) / 5
+    got = ops.gemm_bf16(x, w, None, act=1)
+    want = ref.gemm_bf16(x.cpu(), w.cpu(), None, act=1)
+    assert_close_bf16(got, want, atol=5e-2, rtol=5e-2)
