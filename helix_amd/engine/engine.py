@@ -1,0 +1,339 @@
+"""LLMEngine — continuous-batching inference engine for MI355X.
+
+Replaces the vLLM container the reference dispatches to
+(SURVEY.md §3.3 hot loop: "continuous batching, paged attention,
+sampling — all delegated"). Prefill and decode interleave: each step()
+either admits waiting sequences with one varlen prefill or advances all
+running sequences one decode token over the paged KV cache.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from enum import Enum
+from typing import Callable, Dict, List, Optional
+
+import torch
+
+from helix_amd import ops
+from helix_amd.engine.kv_cache import KVCache
+from helix_amd.engine.sampling_params import SamplingParams
+from helix_amd.models.llama import (DecodeMeta, LlamaConfig, LlamaForCausalLM,
+                                    PrefillMeta, PRESETS)
+
+_SEED_MIX = 0x9E3779B97F4A7C15
+
+
+class SeqStatus(Enum):
+    WAITING = "waiting"
+    RUNNING = "running"
+    FINISHED = "finished"
+    CANCELLED = "cancelled"
+
+
+@dataclass
+class Sequence:
+    seq_id: str
+    prompt_ids: List[int]
+    params: SamplingParams
+    output_ids: List[int] = field(default_factory=list)
+    status: SeqStatus = SeqStatus.WAITING
+    block_table: List[int] = field(default_factory=list)
+    arrival_time: float = field(default_factory=time.monotonic)
+    first_token_time: Optional[float] = None
+    finish_reason: Optional[str] = None
+    # streaming callback: fn(seq, new_token_id, finished)
+    on_token: Optional[Callable] = None
+
+    @property
+    def total_len(self) -> int:
+        return len(self.prompt_ids) + len(self.output_ids)
+
+
+@dataclass
+class EngineConfig:
+    model: str = "llama3-8b"
+    block_size: int = 16
+    max_num_seqs: int = 64
+    max_model_len: int = 8192
+    max_prefill_tokens: int = 8192
+    kv_cache_blocks: Optional[int] = None     # None => size from free HBM
+    gpu_memory_utilization: float = 0.90
+    eos_token_id: int = 2
+    seed: int = 0
+    enforce_eager: bool = False               # disable hipGraph capture
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, device: str = "cuda",
+                 model: Optional[LlamaForCausalLM] = None,
+                 tp_size: int = 1, tp_rank: int = 0,
+                 tp_group=None):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.model_cfg: LlamaConfig = PRESETS[cfg.model]
+        self.tp_size = tp_size
+        self.tp_rank = tp_rank
+        self.tp_group = tp_group
+        if model is None:
+            dtype = torch.bfloat16
+            model = LlamaForCausalLM(self.model_cfg, tp_size, tp_rank)
+            model = model.to(dtype).to(self.device)
+            model.init_random(cfg.seed)
+        self.model = model
+
+        nkv = self.model_cfg.num_kv_heads // tp_size
+        num_blocks = cfg.kv_cache_blocks
+        if num_blocks is None:
+            num_blocks = self._auto_kv_blocks(nkv)
+        self.kv = KVCache(self.model_cfg.num_layers, nkv,
+                          self.model_cfg.head_dim, cfg.block_size, num_blocks,
+                          self.device)
+        self.waiting: List[Sequence] = []
+        self.running: List[Sequence] = []
+        self.seqs: Dict[str, Sequence] = {}
+        self.steps = 0
+        if self.device.type == "cuda":
+            self.decode_ws = ops.decode_workspace(
+                cfg.max_num_seqs, self.model_cfg.num_heads // tp_size,
+                self.model_cfg.head_dim, cfg.max_model_len, self.device)
+        else:
+            self.decode_ws = None
+
+    # ------------------------------------------------------------------
+    def _auto_kv_blocks(self, nkv: int) -> int:
+        if self.device.type != "cuda":
+            return 512
+        free, total = torch.cuda.mem_get_info(self.device)
+        budget = int(total * self.cfg.gpu_memory_utilization
+                     - (total - free))
+        budget = max(budget, 1 << 28)
+        n = KVCache.blocks_for_bytes(budget, self.model_cfg.num_layers, nkv,
+                                     self.model_cfg.head_dim,
+                                     self.cfg.block_size)
+        # cap: max_num_seqs sequences of max_model_len
+        cap = (self.cfg.max_num_seqs *
+               (self.cfg.max_model_len // self.cfg.block_size + 1))
+        return max(1, min(n, cap))
+
+    # ------------------------------------------------------------------
+    def add_request(self, seq_id: str, prompt_ids: List[int],
+                    params: SamplingParams,
+                    on_token: Optional[Callable] = None) -> Sequence:
+        if len(prompt_ids) >= self.cfg.max_model_len:
+            raise ValueError(
+                f"prompt length {len(prompt_ids)} >= max_model_len "
+                f"{self.cfg.max_model_len}")
+        seq = Sequence(seq_id=seq_id, prompt_ids=list(prompt_ids),
+                       params=params, on_token=on_token)
+        self.seqs[seq_id] = seq
+        self.waiting.append(seq)
+        return seq
+
+    def cancel(self, seq_id: str):
+        seq = self.seqs.get(seq_id)
+        if seq is None:
+            return
+        if seq.status == SeqStatus.WAITING:
+            self.waiting.remove(seq)
+        elif seq.status == SeqStatus.RUNNING:
+            self.running.remove(seq)
+            self.kv.allocator.free(seq.block_table)
+            seq.block_table = []
+        seq.status = SeqStatus.CANCELLED
+        seq.finish_reason = "cancelled"
+
+    @property
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def num_free_blocks(self) -> int:
+        return self.kv.allocator.free_count
+
+    # ------------------------------------------------------------------
+    def step(self) -> List[tuple]:
+        """Run one engine iteration. Returns [(seq, token_id, finished)]."""
+        prefill_batch = self._schedule_prefill()
+        if prefill_batch:
+            out = self._run_prefill(prefill_batch)
+        elif self.running:
+            out = self._run_decode()
+        else:
+            return []
+        self.steps += 1
+        events = []
+        for seq, tok in out:
+            finished = self._append_token(seq, tok)
+            events.append((seq, tok, finished))
+            if seq.on_token is not None:
+                seq.on_token(seq, tok, finished)
+        return events
+
+    # ------------------------------------------------------------------
+    def _schedule_prefill(self) -> List[Sequence]:
+        batch: List[Sequence] = []
+        tokens = 0
+        while self.waiting and len(self.running) + len(batch) < self.cfg.max_num_seqs:
+            seq = self.waiting[0]
+            need = len(seq.prompt_ids)
+            if batch and tokens + need > self.cfg.max_prefill_tokens:
+                break
+            nblocks = (need + self.cfg.block_size - 1) // self.cfg.block_size
+            # +1 headroom block so the first decode step can't OOM-deadlock
+            if not self.kv.allocator.can_allocate(nblocks + 1):
+                break
+            seq.block_table = self.kv.allocator.allocate(nblocks)
+            batch.append(self.waiting.pop(0))
+            tokens += need
+        return batch
+
+    def _slot(self, seq: Sequence, pos: int) -> int:
+        bs = self.cfg.block_size
+        return seq.block_table[pos // bs] * bs + pos % bs
+
+    def _run_prefill(self, batch: List[Sequence]):
+        bs = self.cfg.block_size
+        input_ids, positions, slots, cu = [], [], [], [0]
+        for seq in batch:
+            L = len(seq.prompt_ids)
+            input_ids.extend(seq.prompt_ids)
+            positions.extend(range(L))
+            slots.extend(self._slot(seq, p) for p in range(L))
+            cu.append(cu[-1] + L)
+        dev = self.device
+        meta = PrefillMeta(
+            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
+            max_seqlen=max(len(s.prompt_ids) for s in batch),
+            slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
+            positions=torch.tensor(positions, dtype=torch.int64, device=dev))
+        ids = torch.tensor(input_ids, dtype=torch.int64, device=dev)
+        hidden = self.model(ids, self.kv.caches, meta)
+        last_rows = torch.tensor([c - 1 for c in cu[1:]], dtype=torch.int64,
+                                 device=dev)
+        logits = self.model.compute_logits(hidden[last_rows])
+        tokens = self._sample(batch, logits)
+        now = time.monotonic()
+        for seq in batch:
+            seq.status = SeqStatus.RUNNING
+            seq.first_token_time = now
+            self.running.append(seq)
+        return list(zip(batch, tokens))
+
+    def _run_decode(self):
+        bs = self.cfg.block_size
+        batch = self.running
+        dev = self.device
+        input_ids, positions, slots, seq_lens = [], [], [], []
+        for seq in batch:
+            pos = seq.total_len - 1          # position of the new input token
+            nblk = pos // bs + 1
+            while len(seq.block_table) < nblk:
+                seq.block_table.extend(self.kv.allocator.allocate(1))
+            input_ids.append(seq.output_ids[-1] if seq.output_ids
+                             else seq.prompt_ids[-1])
+            positions.append(pos)
+            slots.append(self._slot(seq, pos))
+            seq_lens.append(pos + 1)
+        max_blocks = max(len(s.block_table) for s in batch)
+        bt = torch.zeros(len(batch), max_blocks, dtype=torch.int32, device=dev)
+        for i, seq in enumerate(batch):
+            bt[i, :len(seq.block_table)] = torch.tensor(
+                seq.block_table, dtype=torch.int32)
+        meta = DecodeMeta(
+            block_tables=bt,
+            seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
+            positions=torch.tensor(positions, dtype=torch.int64, device=dev),
+            workspace=self.decode_ws)
+        ids = torch.tensor(input_ids, dtype=torch.int64, device=dev)
+        hidden = self.model(ids, self.kv.caches, meta)
+        logits = self.model.compute_logits(hidden)
+        tokens = self._sample(batch, logits)
+        return list(zip(batch, tokens))
+
+    # ------------------------------------------------------------------
+    def _sample(self, batch: List[Sequence], logits: torch.Tensor) -> List[int]:
+        # wait: decode path's new input token is appended by _append_token;
+        # here logits are [B, V].
+        needs_proc = any(s.params.needs_logit_processing for s in batch)
+        if needs_proc:
+            logits = self._process_logits(batch, logits.float())
+        temps = torch.tensor([s.params.temperature for s in batch],
+                             dtype=torch.float32, device=logits.device)
+        seeds = []
+        for s in batch:
+            base = s.params.seed if s.params.seed is not None else (
+                hash(s.seq_id) & 0x7FFFFFFF)
+            seeds.append((base + len(s.output_ids) * _SEED_MIX)
+                         & 0x7FFFFFFFFFFFFFFF)
+        seeds_t = torch.tensor(seeds, dtype=torch.int64, device=logits.device)
+        toks = ops.sample_tokens(logits.contiguous(), temps, seeds_t)
+        return [int(t) for t in toks.cpu()]
+
+    def _process_logits(self, batch, logits: torch.Tensor) -> torch.Tensor:
+        for i, seq in enumerate(batch):
+            p = seq.params
+            if p.repetition_penalty != 1.0 or p.presence_penalty != 0.0 \
+               or p.frequency_penalty != 0.0:
+                seen = torch.tensor(
+                    list(set(seq.prompt_ids + seq.output_ids)),
+                    dtype=torch.int64, device=logits.device)
+                row = logits[i]
+                if p.repetition_penalty != 1.0:
+                    vals = row[seen]
+                    row[seen] = torch.where(vals > 0,
+                                            vals / p.repetition_penalty,
+                                            vals * p.repetition_penalty)
+                if p.presence_penalty != 0.0:
+                    row[seen] -= p.presence_penalty
+                if p.frequency_penalty != 0.0:
+                    from collections import Counter
+                    cnt = Counter(seq.output_ids)
+                    idx = torch.tensor(list(cnt.keys()), dtype=torch.int64,
+                                       device=logits.device)
+                    freq = torch.tensor(list(cnt.values()),
+                                        dtype=torch.float32,
+                                        device=logits.device)
+                    row[idx] -= p.frequency_penalty * freq
+            if p.top_k > 0:
+                kth = torch.topk(logits[i], p.top_k).values[-1]
+                logits[i][logits[i] < kth] = -float("inf")
+            if p.top_p < 1.0:
+                sorted_logits, idx = torch.sort(logits[i], descending=True)
+                probs = torch.softmax(sorted_logits, -1)
+                cum = probs.cumsum(-1)
+                cut = (cum - probs) >= p.top_p   # keep first token past p
+                sorted_logits[cut] = -float("inf")
+                logits[i].scatter_(0, idx, sorted_logits)
+        return logits
+
+    def _append_token(self, seq: Sequence, tok: int) -> bool:
+        seq.output_ids.append(tok)
+        p = seq.params
+        finished = None
+        if not p.ignore_eos and (tok == self.cfg.eos_token_id
+                                 or tok in p.stop_token_ids):
+            finished = "stop"
+        elif len(seq.output_ids) >= p.max_tokens:
+            finished = "length"
+        elif seq.total_len >= self.cfg.max_model_len:
+            finished = "length"
+        if finished:
+            seq.status = SeqStatus.FINISHED
+            seq.finish_reason = finished
+            self.running.remove(seq)
+            self.kv.allocator.free(seq.block_table)
+            seq.block_table = []
+            return True
+        return False
+
+    # ------------------------------------------------------------------
+    def generate(self, prompts: List[List[int]],
+                 params: SamplingParams) -> List[List[int]]:
+        """Synchronous batch generate (used by tests and bench)."""
+        ids = [f"gen-{id(prompts)}-{i}" for i in range(len(prompts))]
+        for sid, p in zip(ids, prompts):
+            self.add_request(sid, p, params)
+        while self.has_work:
+            self.step()
+        return [self.seqs[sid].output_ids for sid in ids]

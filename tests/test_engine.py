@@ -1,0 +1,127 @@
+"""Engine logic tests on CPU (reference ops, tiny model)."""
+import pytest
+import torch
+
+from helix_amd.engine.engine import EngineConfig, LLMEngine, SeqStatus
+from helix_amd.engine.kv_cache import BlockAllocator, KVCache
+from helix_amd.engine.sampling_params import SamplingParams
+
+
+def make_engine(**kw):
+    cfg = EngineConfig(model="tiny", max_model_len=256, max_num_seqs=8,
+                       kv_cache_blocks=128, eos_token_id=0, **kw)
+    return LLMEngine(cfg, device="cpu")
+
+
+def test_block_allocator():
+    a = BlockAllocator(10)
+    b1 = a.allocate(4)
+    assert a.free_count == 6
+    a.free(b1)
+    assert a.free_count == 10
+    with pytest.raises(RuntimeError):
+        a.allocate(11)
+
+
+def test_kv_cache_sizing():
+    bb = KVCache.block_bytes(2, 2, 64, 16)
+    assert bb == 2 * 2 * 2 * 16 * 64 * 2
+    assert KVCache.blocks_for_bytes(bb * 7 + 1, 2, 2, 64, 16) == 7
+
+
+def test_generate_greedy_deterministic():
+    torch.manual_seed(0)
+    eng = make_engine()
+    prompts = [[1, 2, 3, 4, 5], [7, 8, 9]]
+    sp = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+    out1 = eng.generate(prompts, sp)
+    out2 = eng.generate(prompts, sp)
+    assert out1 == out2
+    assert all(len(o) == 8 for o in out1)
+
+
+def test_decode_matches_full_prefill():
+    """Incremental decode over paged KV must agree with a fresh prefill of
+    the full sequence (greedy tokens equal)."""
+    torch.manual_seed(0)
+    eng = make_engine()
+    prompt = list(range(1, 11))
+    sp = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    out = eng.generate([prompt], sp)[0]
+    # now feed prompt+out[:-1] as a fresh prompt; next greedy token must be out[-1]
+    eng2 = make_engine()
+    out2 = eng2.generate([prompt + out[:-1]],
+                         SamplingParams(temperature=0.0, max_tokens=1,
+                                        ignore_eos=True))[0]
+    assert out2[0] == out[-1]
+
+
+def test_streaming_callback_and_ttft():
+    eng = make_engine()
+    got = []
+    sp = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
+    seq = eng.add_request("s1", [1, 2, 3], sp,
+                          on_token=lambda s, t, f: got.append((t, f)))
+    while eng.has_work:
+        eng.step()
+    assert len(got) == 4
+    assert got[-1][1] is True
+    assert seq.first_token_time is not None
+
+
+def test_cancel():
+    eng = make_engine()
+    sp = SamplingParams(temperature=0.0, max_tokens=100, ignore_eos=True)
+    eng.add_request("a", [1, 2, 3], sp)
+    eng.add_request("b", [4, 5, 6], sp)
+    eng.step()
+    free_before = eng.num_free_blocks()
+    eng.cancel("a")
+    assert eng.seqs["a"].status == SeqStatus.CANCELLED
+    assert eng.num_free_blocks() > free_before
+    # engine continues with remaining seq
+    eng.step()
+    assert len(eng.seqs["b"].output_ids) >= 2
+
+
+def test_max_tokens_and_continuous_batching():
+    eng = make_engine()
+    sp = SamplingParams(temperature=0.0, max_tokens=3, ignore_eos=True)
+    eng.add_request("a", [1, 2], sp)
+    eng.step()  # prefill a
+    eng.add_request("b", [3, 4], sp)
+    # next step admits b (prefill) while a keeps decoding after
+    while eng.has_work:
+        eng.step()
+    assert len(eng.seqs["a"].output_ids) == 3
+    assert len(eng.seqs["b"].output_ids) == 3
+    assert eng.seqs["a"].finish_reason == "length"
+
+
+def test_eos_stops():
+    eng = make_engine()
+    # find whichever token greedy emits first, then declare it EOS
+    sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+    tok = eng.generate([[1, 2, 3]], sp)[0][0]
+    eng2 = make_engine()
+    eng2.cfg.eos_token_id = tok
+    sp2 = SamplingParams(temperature=0.0, max_tokens=5)
+    out = eng2.generate([[1, 2, 3]], sp2)[0]
+    assert out == [tok]
+    assert eng2.seqs[list(eng2.seqs)[0]].finish_reason == "stop"
+
+
+def test_top_k_top_p_and_penalties_run():
+    eng = make_engine()
+    sp = SamplingParams(temperature=0.8, top_k=5, top_p=0.9,
+                        repetition_penalty=1.2, presence_penalty=0.1,
+                        frequency_penalty=0.1, max_tokens=4, ignore_eos=True,
+                        seed=42)
+    out = eng.generate([[1, 2, 3]], sp)[0]
+    assert len(out) == 4
+
+
+def test_prompt_too_long_rejected():
+    eng = make_engine()
+    with pytest.raises(ValueError):
+        eng.add_request("x", list(range(300)), SamplingParams())
