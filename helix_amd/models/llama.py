@@ -95,6 +95,7 @@ class Attention(nn.Module):
         self.nkv = cfg.num_kv_heads // tp_size
         self.hd = cfg.head_dim
         self.scale = self.hd ** -0.5
+        self.tp_size = tp_size
         q, kv, h = self.nh * self.hd, self.nkv * self.hd, cfg.hidden_size
         self.qkv_proj = nn.Linear(h, q + 2 * kv, bias=False)
         self.o_proj = nn.Linear(q, h, bias=False)
@@ -121,19 +122,28 @@ class Attention(nn.Module):
             o = ops.paged_attn_decode(q, k_cache, v_cache, meta.block_tables,
                                       meta.seq_lens, self.scale,
                                       meta.workspace)
-        return self.o_proj(o.view(T, -1))
+        out = self.o_proj(o.view(T, -1))
+        if self.tp_size > 1:
+            from helix_amd import parallel
+            out = parallel.tp_all_reduce(out)
+        return out
 
 
 class MLP(nn.Module):
     def __init__(self, cfg: LlamaConfig, tp_size: int = 1):
         super().__init__()
         assert cfg.intermediate_size % tp_size == 0
+        self.tp_size = tp_size
         i = cfg.intermediate_size // tp_size
         self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * i, bias=False)
         self.down_proj = nn.Linear(i, cfg.hidden_size, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
+        out = self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
+        if self.tp_size > 1:
+            from helix_amd import parallel
+            out = parallel.tp_all_reduce(out)
+        return out
 
 
 class DecoderLayer(nn.Module):
@@ -176,6 +186,14 @@ class LlamaForCausalLM(nn.Module):
                                      bias=False)
         cs = make_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_base)
         self.register_buffer("cos_sin", cs, persistent=False)
+
+    def _apply(self, fn, recurse=True):
+        # keep the RoPE table fp32 through .to(bfloat16) casts — the HIP
+        # kernel consumes fp32 cos/sin.
+        super()._apply(fn, recurse)
+        if self.cos_sin.dtype != torch.float32:
+            self.cos_sin = self.cos_sin.float()
+        return self
 
     @torch.inference_mode()
     def forward(self, input_ids: torch.Tensor, kv_caches, meta) -> torch.Tensor:
