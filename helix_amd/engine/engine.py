@@ -99,6 +99,13 @@ class LLMEngine:
                 self.model_cfg.head_dim, cfg.max_model_len, self.device)
         else:
             self.decode_ws = None
+        self.graph_runner = None
+        if self.device.type == "cuda" and not cfg.enforce_eager:
+            from helix_amd.engine.graph_runner import CUDAGraphRunner
+            self.graph_runner = CUDAGraphRunner(
+                self.model, self.kv.caches, self.decode_ws,
+                cfg.max_model_len, cfg.block_size, cfg.max_num_seqs,
+                self.device)
 
     # ------------------------------------------------------------------
     def _auto_kv_blocks(self, nkv: int) -> int:
@@ -235,19 +242,27 @@ class LLMEngine:
             slots.append(self._slot(seq, pos))
             seq_lens.append(pos + 1)
         max_blocks = max(len(s.block_table) for s in batch)
-        bt = torch.zeros(len(batch), max_blocks, dtype=torch.int32, device=dev)
+        bt_cpu = torch.zeros(len(batch), max_blocks, dtype=torch.int32)
         for i, seq in enumerate(batch):
-            bt[i, :len(seq.block_table)] = torch.tensor(
+            bt_cpu[i, :len(seq.block_table)] = torch.tensor(
                 seq.block_table, dtype=torch.int32)
-        meta = DecodeMeta(
-            block_tables=bt,
-            seq_lens=torch.tensor(seq_lens, dtype=torch.int32, device=dev),
-            slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
-            positions=torch.tensor(positions, dtype=torch.int64, device=dev),
-            workspace=self.decode_ws)
-        ids = torch.tensor(input_ids, dtype=torch.int64, device=dev)
-        hidden = self.model(ids, self.kv.caches, meta)
-        logits = self.model.compute_logits(hidden)
+        if self.graph_runner is not None:
+            logits = self.graph_runner.run(input_ids, positions, slots,
+                                           bt_cpu, seq_lens, max(seq_lens))
+        else:
+            meta = DecodeMeta(
+                block_tables=bt_cpu.to(dev),
+                seq_lens=torch.tensor(seq_lens, dtype=torch.int32,
+                                      device=dev),
+                slot_mapping=torch.tensor(slots, dtype=torch.int64,
+                                          device=dev),
+                positions=torch.tensor(positions, dtype=torch.int64,
+                                       device=dev),
+                max_len=max(seq_lens),
+                workspace=self.decode_ws)
+            ids = torch.tensor(input_ids, dtype=torch.int64, device=dev)
+            hidden = self.model(ids, self.kv.caches, meta)
+            logits = self.model.compute_logits(hidden)
         tokens = self._sample(batch, logits)
         return list(zip(batch, tokens))
 

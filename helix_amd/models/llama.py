@@ -82,6 +82,7 @@ class DecodeMeta:
     seq_lens: torch.Tensor         # [B] int32 (context incl. current token)
     slot_mapping: torch.Tensor     # [B] int64
     positions: torch.Tensor        # [B] int64
+    max_len: int = 0               # host-side max(seq_lens) (avoids sync)
     workspace: Optional[tuple] = None
     is_prefill: bool = False
 
@@ -121,7 +122,8 @@ class Attention(nn.Module):
             k_cache, v_cache = kv_cache
             o = ops.paged_attn_decode(q, k_cache, v_cache, meta.block_tables,
                                       meta.seq_lens, self.scale,
-                                      meta.workspace)
+                                      meta.workspace,
+                                      meta.max_len or None)
         out = self.o_proj(o.view(T, -1))
         if self.tp_size > 1:
             from helix_amd import parallel

@@ -104,16 +104,18 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
 
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  cu_seqlens: torch.Tensor, max_seqlen: int,
-                 scale: float) -> torch.Tensor:
+                 scale: float, causal: bool = True) -> torch.Tensor:
     if q.is_cuda:
         out = torch.empty_like(q)
-        _native().attn_prefill(out, q, k, v, cu_seqlens, max_seqlen, scale)
+        _native().attn_prefill(out, q, k, v, cu_seqlens, max_seqlen, scale,
+                               causal)
         return out
-    return ref.attn_prefill(q, k, v, cu_seqlens, max_seqlen, scale)
+    return ref.attn_prefill(q, k, v, cu_seqlens, max_seqlen, scale, causal)
 
 
-# Flash-decoding split-K sizing (must match paged_attn_decode.hip).
-DECODE_PARTITION = 512
+# Flash-decoding split-K quantum (must match PART_QUANT in
+# paged_attn_decode.hip).
+DECODE_PARTITION = 128
 
 
 def decode_workspace(max_batch: int, num_q_heads: int, head_dim: int,
@@ -130,18 +132,20 @@ def decode_workspace(max_batch: int, num_q_heads: int, head_dim: int,
 def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, block_tables: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float,
-                      workspace=None) -> torch.Tensor:
+                      workspace=None, max_len: int | None = None
+                      ) -> torch.Tensor:
     if q.is_cuda:
         out = torch.empty_like(q)
+        if max_len is None:
+            max_len = int(seq_lens.max())   # host sync — pass max_len to avoid
         if workspace is None:
-            max_len = int(seq_lens.max())
             workspace = decode_workspace(q.shape[0], q.shape[1], q.shape[2],
                                          max_len, q.device)
         tmp_out, tmp_ml = workspace
         _native().paged_attn_decode(out, q, k_cache, v_cache, block_tables,
                                     seq_lens, scale,
                                     tmp_out[:q.shape[0]], tmp_ml[:q.shape[0]],
-                                    DECODE_PARTITION)
+                                    max_len)
         return out
     return ref.paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens,
                                  scale)

@@ -39,7 +39,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
     const uint16_t* __restrict__ k,    // [T, Hkv, D]
     const uint16_t* __restrict__ v,    // [T, Hkv, D]
     const int* __restrict__ cu_seqlens,  // [B+1]
-    float scale, int Hq, int Hkv, int D) {
+    float scale, int Hq, int Hkv, int D, int causal) {
   const int qtile = blockIdx.x;
   const int seq = blockIdx.y;
   const int hq = blockIdx.z;
@@ -88,7 +88,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
   for (int c = 0; c < DMAX / 16; ++c) o_acc[c] = floatx4{0, 0, 0, 0};
   const int nc = D / 16;
 
-  const int kv_max = min(len, qbase + QTILE);  // causal bound
+  const int kv_max = causal ? min(len, qbase + QTILE) : len;
   const int ntiles = (kv_max + KTILE - 1) / KTILE;
 
   for (int t = 0; t < ntiles; ++t) {
@@ -135,8 +135,8 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
       float s0 = s_frag[0][r] * scale;
       float s1 = s_frag[1][r] * scale;
       const int t0 = kv_base + lane_lo, t1 = kv_base + 16 + lane_lo;
-      if (t0 > qpos || t0 >= len) s0 = -1e30f;
-      if (t1 > qpos || t1 >= len) s1 = -1e30f;
+      if ((causal && t0 > qpos) || t0 >= len) s0 = -1e30f;
+      if ((causal && t1 > qpos) || t1 >= len) s1 = -1e30f;
       float m_tile = fmaxf(s0, s1);
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
@@ -205,7 +205,7 @@ __global__ __launch_bounds__(256) void attn_prefill_kernel(
 
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   torch::Tensor v, torch::Tensor cu_seqlens,
-                  int64_t max_seqlen, double scale) {
+                  int64_t max_seqlen, double scale, bool causal) {
   const int Hq = q.size(1);
   const int D = q.size(2);
   const int Hkv = k.size(1);
@@ -219,5 +219,6 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                      (const uint16_t*)q.data_ptr(),
                      (const uint16_t*)k.data_ptr(),
                      (const uint16_t*)v.data_ptr(),
-                     cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv, D);
+                     cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv, D,
+                     causal ? 1 : 0);
 }

@@ -21,7 +21,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor tmp_ml, int64_t partition_size);
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   torch::Tensor v, torch::Tensor cu_seqlens,
-                  int64_t max_seqlen, double scale);
+                  int64_t max_seqlen, double scale, bool causal);
 void sample_tokens(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperatures, torch::Tensor seeds);
 void gemm_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,

@@ -4,18 +4,19 @@
 // Replaces the decode attention the reference delegates to vLLM
 // (SURVEY.md §2.8 "Decode attention (paged KV)").
 //
-// Design (MI355X-first):
+// Design (MI355X-first, v2 — occupancy-driven):
 //  - Decode attention is KV-bandwidth-bound: the kernel reads each KV byte
 //    exactly once per kv-head and amortizes it across the G = Hq/Hkv query
-//    heads of the GQA group (G-way traffic saving vs per-q-head kernels).
-//  - grid = (num_seqs, Hkv, num_partitions): flash-decoding style split-K
-//    over the sequence so small batches still fill 256 CUs.
-//  - Each 256-thread block walks its partition in 256-token chunks:
-//      A: thread t loads K row of token t (vectorized 16B), dots against the
-//         G query vectors held in LDS (broadcast reads), scores -> LDS.
-//      B: per-head online-softmax update (running m, l) by one wave per head.
-//      C: V accumulation with dim-owned accumulators: thread owns dim d for
-//         all G heads; V reads are lane-contiguous (coalesced 2B*64 = 128B).
+//    heads of the GQA group.
+//  - grid = (num_seqs, Hkv, num_partitions): flash-decoding split-K, with
+//    the partition count chosen on the host to put >= ~1024 workgroups on
+//    the 256 CUs (v1 used 1 block/CU and was 10x off the bandwidth bound —
+//    profiles/r01_decode_profile_v1.md).
+//  - 128-thread blocks (2 waves) walk their partition in 128-token chunks:
+//      A: thread t loads the K row of token t (vectorized 16B), dots against
+//         the G query vectors held in LDS (broadcast reads), scores -> LDS.
+//      B: per-head online-softmax update (running m, l), one wave per head.
+//      C: V accumulation, thread owns a dim PAIR (b32 loads) for all G heads.
 //  - Single-partition grids write normalized bf16 straight to `out`;
 //    multi-partition grids write fp32 partials + (m, l) for a reduce kernel.
 #include "common.h"
@@ -26,11 +27,12 @@ using namespace helix;
 
 namespace {
 
-constexpr int NTHREADS = 256;
-constexpr int CHUNK = 256;
+constexpr int NTHREADS = 128;
+constexpr int CHUNK = 128;
 constexpr int MAX_G = 8;
+constexpr int PART_QUANT = 128;   // partition sizes are multiples of this
 
-__global__ __launch_bounds__(NTHREADS) void paged_attn_decode_kernel(
+__global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
     float* __restrict__ tmp_ml,          // [B, Hq, maxP, 2]
@@ -46,11 +48,11 @@ __global__ __launch_bounds__(NTHREADS) void paged_attn_decode_kernel(
   const int part = blockIdx.z;
   const int nparts = gridDim.z;
   const int len = seq_lens[seq];
+  const int G = Hq / Hkv;
   const int p_start = part * partition_size;
   if (p_start >= len) {
     // Dead partition: mark so the reduce kernel skips it.
     if (nparts > 1 && threadIdx.x == 0) {
-      const int G = Hq / Hkv;
       for (int g = 0; g < G; ++g) {
         const int hq = hkv * G + g;
         float* ml = tmp_ml + (((int64_t)seq * Hq + hq) * max_parts + part) * 2;
@@ -61,12 +63,10 @@ __global__ __launch_bounds__(NTHREADS) void paged_attn_decode_kernel(
     return;
   }
   const int p_end = min(len, p_start + partition_size);
-  const int G = Hq / Hkv;
 
   __shared__ float q_lds[MAX_G][128];
   __shared__ float s_lds[MAX_G][CHUNK];
   __shared__ float head_m[MAX_G], head_l[MAX_G], head_corr[MAX_G];
-  __shared__ float comb[MAX_G * NTHREADS];  // [par][g][d] flattened
 
   // Load the G query vectors (pre-scaled) into LDS.
   for (int idx = threadIdx.x; idx < G * D; idx += NTHREADS) {
@@ -80,13 +80,15 @@ __global__ __launch_bounds__(NTHREADS) void paged_attn_decode_kernel(
   }
   __syncthreads();
 
-  // Phase-C ownership: thread owns dim d for tokens of parity `par`.
-  const int n_par = NTHREADS / D;          // D in {64, 128}
-  const int d_own = threadIdx.x % D;
-  const int par = threadIdx.x / D;
-  float acc[MAX_G];
+  // Phase-C ownership: thread owns dim pair (2*t, 2*t+1) within D, and
+  // loops token parities if D < 2*NTHREADS is not exact.
+  const int pairs = D / 2;                  // D in {64, 128}
+  const int d_own = (threadIdx.x % pairs) * 2;
+  const int par = threadIdx.x / pairs;      // 0..n_par-1
+  const int n_par = NTHREADS / pairs;
+  float acc[MAX_G][2];
 #pragma unroll
-  for (int g = 0; g < MAX_G; ++g) acc[g] = 0.f;
+  for (int g = 0; g < MAX_G; ++g) acc[g][0] = acc[g][1] = 0.f;
 
   const int* btable = block_tables + (int64_t)seq * max_blocks;
   const int nwaves = NTHREADS / WAVE;
@@ -118,7 +120,7 @@ __global__ __launch_bounds__(NTHREADS) void paged_attn_decode_kernel(
     }
     __syncthreads();
 
-    // --- Phase B: online softmax per head (one wave per head slot) --------
+    // --- Phase B: online softmax per head (waves rotate over heads) -------
     for (int g = wid; g < G; g += nwaves) {
       float m_chunk = -INFINITY;
       for (int i = lane; i < chunk_n; i += WAVE)
@@ -142,35 +144,61 @@ __global__ __launch_bounds__(NTHREADS) void paged_attn_decode_kernel(
     }
     __syncthreads();
 
-    // --- Phase C: V accumulation -------------------------------------------
+    // --- Phase C: V accumulation (b32 dim-pair loads) ----------------------
     {
-      for (int g = 0; g < G; ++g) acc[g] *= head_corr[g];
+      for (int g = 0; g < G; ++g) {
+        acc[g][0] *= head_corr[g];
+        acc[g][1] *= head_corr[g];
+      }
       for (int tok_i = par; tok_i < chunk_n; tok_i += n_par) {
         const int tok = base + tok_i;
         const int64_t blk = btable[tok / block_size];
-        const float v = bf16_to_f32(
-            v_cache[(((blk * Hkv + hkv) * (int64_t)block_size +
-                      tok % block_size)) * D + d_own]);
-        for (int g = 0; g < G; ++g) acc[g] += s_lds[g][tok_i] * v;
+        const uint32_t vv = *reinterpret_cast<const uint32_t*>(
+            v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
+                        tok % block_size)) * D + d_own);
+        const float v0 = bf16_to_f32((uint16_t)(vv & 0xffff));
+        const float v1 = bf16_to_f32((uint16_t)(vv >> 16));
+        for (int g = 0; g < G; ++g) {
+          const float p = s_lds[g][tok_i];
+          acc[g][0] += p * v0;
+          acc[g][1] += p * v1;
+        }
       }
     }
     __syncthreads();  // s_lds reused next chunk
   }
 
-  // Combine parities via LDS: comb[par*G*D + g*D + d]
-  for (int g = 0; g < G; ++g) comb[(par * G + g) * D + d_own] = acc[g];
-  __syncthreads();
+  // Combine token parities (n_par == 1 for D=128; 2 for D=64) and write.
+  __shared__ float comb[MAX_G * NTHREADS * 2];
+  if (n_par > 1) {
+    for (int g = 0; g < G; ++g) {
+      comb[((par * MAX_G + g) * pairs + d_own / 2) * 2 + 0] = acc[g][0];
+      comb[((par * MAX_G + g) * pairs + d_own / 2) * 2 + 1] = acc[g][1];
+    }
+    __syncthreads();
+    if (par == 0) {
+      for (int g = 0; g < G; ++g)
+        for (int p = 1; p < n_par; ++p) {
+          acc[g][0] += comb[((p * MAX_G + g) * pairs + d_own / 2) * 2 + 0];
+          acc[g][1] += comb[((p * MAX_G + g) * pairs + d_own / 2) * 2 + 1];
+        }
+    }
+  }
   if (par == 0) {
     for (int g = 0; g < G; ++g) {
-      float o = comb[g * D + d_own];
-      for (int p = 1; p < n_par; ++p) o += comb[(p * G + g) * D + d_own];
       const int hq = hkv * G + g;
       if (nparts == 1) {
-        const float l = head_l[g];
-        out[((int64_t)seq * Hq + hq) * D + d_own] =
-            f32_to_bf16(o / fmaxf(l, 1e-20f));
+        const float inv_l = 1.f / fmaxf(head_l[g], 1e-20f);
+        uint32_t packed =
+            (uint32_t)f32_to_bf16(acc[g][0] * inv_l) |
+            ((uint32_t)f32_to_bf16(acc[g][1] * inv_l) << 16);
+        *reinterpret_cast<uint32_t*>(
+            out + ((int64_t)seq * Hq + hq) * D + d_own) = packed;
       } else {
-        tmp_out[(((int64_t)seq * Hq + hq) * max_parts + part) * D + d_own] = o;
+        float* tp =
+            tmp_out + (((int64_t)seq * Hq + hq) * max_parts + part) * D;
+        tp[d_own] = acc[g][0];
+        tp[d_own + 1] = acc[g][1];
         if (d_own == 0) {
           float* ml =
               tmp_ml + (((int64_t)seq * Hq + hq) * max_parts + part) * 2;
@@ -183,7 +211,7 @@ __global__ __launch_bounds__(NTHREADS) void paged_attn_decode_kernel(
 }
 
 // Reduce partials: out[seq, hq, :] = sum_p w_p * tmp_out[p] / L
-__global__ void paged_attn_reduce_kernel(
+__global__ __launch_bounds__(128) void paged_attn_reduce_kernel(
     uint16_t* __restrict__ out, const float* __restrict__ tmp_out,
     const float* __restrict__ tmp_ml, const int* __restrict__ seq_lens,
     int Hq, int D, int partition_size, int max_parts) {
@@ -193,8 +221,8 @@ __global__ void paged_attn_reduce_kernel(
       min(max_parts, (seq_lens[seq] + partition_size - 1) / partition_size);
   const float* ml = tmp_ml + ((int64_t)seq * Hq + hq) * max_parts * 2;
 
-  __shared__ float w[256];
-  __shared__ float m_sh, l_sh;
+  __shared__ float w[512];
+  __shared__ float l_sh;
   if (threadIdx.x == 0) {
     float M = -INFINITY;
     for (int p = 0; p < nparts; ++p) M = fmaxf(M, ml[p * 2]);
@@ -221,7 +249,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor block_tables, torch::Tensor seq_lens,
                        double scale, torch::Tensor tmp_out,
-                       torch::Tensor tmp_ml, int64_t partition_size) {
+                       torch::Tensor tmp_ml, int64_t max_len_hint) {
   const int B = q.size(0);
   const int Hq = q.size(1);
   const int D = q.size(2);
@@ -234,13 +262,14 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
 
-  const int max_len = seq_lens.max().item<int>();
-  int nparts = cdiv(max_len, (int)partition_size);
-  nparts = std::min(nparts, max_parts);
-  // If a single partition covers everything, write out directly.
-  if ((int64_t)B * Hkv >= 512 || max_len <= partition_size) nparts = 1;
-  int eff_part = (int)partition_size;
-  if (nparts == 1) eff_part = max_len;  // single pass over full length
+  const int max_len = (int)max_len_hint;
+  // Pick a partition count that fills the chip (~>=1024 workgroups) without
+  // exceeding the workspace.
+  const int max_useful = cdiv(max_len, PART_QUANT);
+  int nparts = std::max(1, 1024 / std::max(1, B * Hkv));
+  nparts = std::min({nparts, max_useful, max_parts});
+  int eff_part = cdiv(cdiv(max_len, nparts), PART_QUANT) * PART_QUANT;
+  nparts = cdiv(max_len, eff_part);
 
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(paged_attn_decode_kernel, dim3(B, Hkv, nparts),
@@ -253,7 +282,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                      (float)scale, Hq, Hkv, D, block_size, max_blocks,
                      eff_part, max_parts);
   if (nparts > 1) {
-    hipLaunchKernelGGL(paged_attn_reduce_kernel, dim3(B, Hq), dim3(256), 0,
+    hipLaunchKernelGGL(paged_attn_reduce_kernel, dim3(B, Hq), dim3(128), 0,
                        stream, (uint16_t*)out.data_ptr(),
                        tmp_out.data_ptr<float>(), tmp_ml.data_ptr<float>(),
                        seq_lens.data_ptr<int>(), Hq, D, eff_part, max_parts);

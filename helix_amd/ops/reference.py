@@ -86,7 +86,7 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
 
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  cu_seqlens: torch.Tensor, max_seqlen: int,
-                 scale: float) -> torch.Tensor:
+                 scale: float, causal: bool = True) -> torch.Tensor:
     """Varlen causal GQA attention. q: [T, Hq, D], k/v: [T, Hkv, D]."""
     Hq, Hkv = q.shape[1], k.shape[1]
     G = Hq // Hkv
@@ -98,7 +98,7 @@ def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         vs = v[s:e].float().repeat_interleave(G, dim=1)
         o = torch.nn.functional.scaled_dot_product_attention(
             qs.transpose(0, 1), ks.transpose(0, 1), vs.transpose(0, 1),
-            is_causal=True, scale=scale)
+            is_causal=causal, scale=scale)
         out[s:e] = o.transpose(0, 1).to(q.dtype)
     return out
 

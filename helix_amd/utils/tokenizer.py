@@ -1,0 +1,66 @@
+"""Byte-level tokenizer + chat template.
+
+No network => no real tokenizer files; this deterministic byte-level
+tokenizer (256 byte tokens + specials) makes the text->tokens->text path
+fully functional end-to-end with random-init weights ("data: synthetic"
+per BASELINE.json). Vocabulary ids stay < 300 so any model preset works.
+"""
+from __future__ import annotations
+
+from typing import List
+
+PAD = 0
+BOS = 1
+EOS = 2
+ROLE_SYSTEM = 3
+ROLE_USER = 4
+ROLE_ASSISTANT = 5
+ROLE_TOOL = 6
+N_SPECIAL = 8
+
+_ROLE_TOKENS = {
+    "system": ROLE_SYSTEM,
+    "user": ROLE_USER,
+    "assistant": ROLE_ASSISTANT,
+    "tool": ROLE_TOOL,
+}
+
+
+class ByteTokenizer:
+    vocab_size = N_SPECIAL + 256
+    bos_token_id = BOS
+    eos_token_id = EOS
+
+    def encode(self, text: str, add_bos: bool = False) -> List[int]:
+        ids = [BOS] if add_bos else []
+        ids.extend(b + N_SPECIAL for b in text.encode("utf-8"))
+        return ids
+
+    def decode(self, ids: List[int]) -> str:
+        bs = bytes(i - N_SPECIAL for i in ids
+                   if N_SPECIAL <= i < N_SPECIAL + 256)
+        return bs.decode("utf-8", errors="replace")
+
+    def apply_chat_template(self, messages: List[dict],
+                            add_generation_prompt: bool = True) -> List[int]:
+        ids = [BOS]
+        for m in messages:
+            role = m.get("role", "user")
+            content = m.get("content") or ""
+            if not isinstance(content, str):
+                # OpenAI content-parts form
+                content = " ".join(p.get("text", "") for p in content
+                                   if isinstance(p, dict))
+            ids.append(_ROLE_TOKENS.get(role, ROLE_USER))
+            ids.extend(self.encode(content))
+            ids.append(EOS)
+        if add_generation_prompt:
+            ids.append(ROLE_ASSISTANT)
+        return ids
+
+
+_DEFAULT = ByteTokenizer()
+
+
+def get_tokenizer(model: str = "") -> ByteTokenizer:
+    return _DEFAULT
