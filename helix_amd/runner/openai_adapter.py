@@ -1,0 +1,219 @@
+"""OpenAI-compatible request handling over RunnerService.
+
+Used by the runner's HTTP app and by the control plane's in-process
+"helix" provider (the seam the reference keeps between
+helix_openai_server.go and the sandbox vLLM containers).
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+import uuid
+from typing import AsyncIterator, Dict, List, Optional
+
+from helix_amd.engine.sampling_params import SamplingParams
+from helix_amd.runner.service import LLMInstance, RunnerService
+from helix_amd.utils.tokenizer import get_tokenizer
+
+
+class StreamDetokenizer:
+    """Incremental byte-level detokenizer (UTF-8 partials buffered)."""
+
+    def __init__(self, tokenizer):
+        self.tok = tokenizer
+        self.ids: List[int] = []
+        self.emitted = ""
+
+    def push(self, token_id: int) -> str:
+        self.ids.append(token_id)
+        full = self.tok.decode(self.ids)
+        # strip a trailing replacement char from an incomplete sequence
+        while full.endswith("�"):
+            full = full[:-1]
+        delta = full[len(self.emitted):]
+        self.emitted = full
+        return delta
+
+
+def _params_from_request(req: dict, max_model_len: int) -> SamplingParams:
+    temp = req.get("temperature")
+    if temp is None:
+        temp = 1.0
+    stop = req.get("stop")
+    return SamplingParams(
+        temperature=float(temp),
+        top_p=float(req.get("top_p") or 1.0),
+        top_k=int(req.get("top_k") or 0),
+        max_tokens=int(req.get("max_tokens") or 256),
+        presence_penalty=float(req.get("presence_penalty") or 0.0),
+        frequency_penalty=float(req.get("frequency_penalty") or 0.0),
+        seed=req.get("seed"),
+    ), ([stop] if isinstance(stop, str) else list(stop or []))
+
+
+class TokenStream:
+    """Bridges the engine's callback thread into an asyncio queue."""
+
+    def __init__(self, loop: asyncio.AbstractEventLoop):
+        self.loop = loop
+        self.q: asyncio.Queue = asyncio.Queue()
+
+    def on_token(self, seq, token_id: int, finished: bool):
+        self.loop.call_soon_threadsafe(
+            self.q.put_nowait, (token_id, finished, seq.finish_reason))
+
+    async def __aiter__(self):
+        while True:
+            tok, fin, reason = await self.q.get()
+            yield tok, fin, reason
+            if fin:
+                return
+
+
+async def chat_completion(service: RunnerService, req: dict,
+                          request_id: Optional[str] = None):
+    """Returns a response dict, or an async iterator of chunk dicts when
+    req['stream'] is true."""
+    model = req.get("model", "")
+    loop = asyncio.get_event_loop()
+    inst = await loop.run_in_executor(None, service.ensure_loaded, model)
+    assert isinstance(inst, LLMInstance), f"{model} is not an LLM"
+    tok = get_tokenizer(model)
+    if "messages" in req and req["messages"] is not None:
+        prompt_ids = tok.apply_chat_template(req["messages"])
+    else:
+        prompt_ids = tok.encode(str(req.get("prompt", "")), add_bos=True)
+    max_ctx = inst.spec.max_model_len
+    if len(prompt_ids) >= max_ctx - 8:
+        prompt_ids = prompt_ids[-(max_ctx - 8):]
+    params, stop_strs = _params_from_request(req, max_ctx)
+    params.stop_token_ids = [tok.eos_token_id]
+    rid = request_id or f"chatcmpl-{uuid.uuid4().hex[:24]}"
+    created = int(time.time())
+
+    if req.get("stream"):
+        return _stream(service, inst, model, rid, created, prompt_ids,
+                       params, stop_strs, tok)
+
+    n = int(req.get("n") or 1)
+    choices = []
+    total_completion = 0
+    for i in range(n):
+        text, finish_reason, ntok = await _generate_one(
+            inst, f"{rid}-{i}", prompt_ids, params, stop_strs, tok, loop)
+        total_completion += ntok
+        choices.append({
+            "index": i,
+            "message": {"role": "assistant", "content": text},
+            "finish_reason": finish_reason,
+        })
+    return {
+        "id": rid,
+        "object": "chat.completion",
+        "created": created,
+        "model": model,
+        "choices": choices,
+        "usage": {
+            "prompt_tokens": len(prompt_ids),
+            "completion_tokens": total_completion,
+            "total_tokens": len(prompt_ids) + total_completion,
+        },
+    }
+
+
+async def _generate_one(inst, seq_id, prompt_ids, params, stop_strs, tok,
+                        loop):
+    ts = TokenStream(loop)
+    inst.submit(seq_id, prompt_ids, params, ts.on_token)
+    detok = StreamDetokenizer(tok)
+    text = ""
+    finish_reason = "stop"
+    ntok = 0
+    async for token_id, fin, reason in ts.__aiter__():
+        ntok += 1
+        text += detok.push(token_id)
+        if fin:
+            finish_reason = reason or "stop"
+            break
+        hit = _find_stop(text, stop_strs)
+        if hit is not None:
+            inst.cancel(seq_id)
+            text = text[:hit]
+            finish_reason = "stop"
+            break
+    return text, finish_reason, ntok
+
+
+def _find_stop(text: str, stop_strs: List[str]) -> Optional[int]:
+    best = None
+    for s in stop_strs:
+        if s and s in text:
+            i = text.index(s)
+            best = i if best is None else min(best, i)
+    return best
+
+
+async def _stream(service, inst, model, rid, created, prompt_ids, params,
+                  stop_strs, tok) -> AsyncIterator[dict]:
+    loop = asyncio.get_event_loop()
+    ts = TokenStream(loop)
+    inst.submit(rid, prompt_ids, params, ts.on_token)
+    detok = StreamDetokenizer(tok)
+    ntok = 0
+
+    def chunk(delta: dict, finish: Optional[str] = None, usage=None):
+        c = {
+            "id": rid,
+            "object": "chat.completion.chunk",
+            "created": created,
+            "model": model,
+            "choices": [{"index": 0, "delta": delta,
+                         "finish_reason": finish}],
+        }
+        if usage:
+            c["usage"] = usage
+        return c
+
+    yield chunk({"role": "assistant", "content": ""})
+    emitted = 0
+    finish_reason = "stop"
+    async for token_id, fin, reason in ts.__aiter__():
+        ntok += 1
+        delta = detok.push(token_id)
+        hit = _find_stop(detok.emitted, stop_strs)
+        if hit is not None:
+            inst.cancel(rid)
+            keep = max(0, hit - emitted)
+            if keep:
+                yield chunk({"content": delta[:keep]})
+            finish_reason = "stop"
+            break
+        if delta:
+            emitted += len(delta)
+            yield chunk({"content": delta})
+        if fin:
+            finish_reason = reason or "stop"
+            break
+    yield chunk({}, finish=finish_reason, usage={
+        "prompt_tokens": len(prompt_ids),
+        "completion_tokens": ntok,
+        "total_tokens": len(prompt_ids) + ntok,
+    })
+
+
+async def embeddings(service: RunnerService, req: dict) -> dict:
+    model = req.get("model", "")
+    loop = asyncio.get_event_loop()
+    inst = await loop.run_in_executor(None, service.ensure_loaded, model)
+    inputs = req.get("input", "")
+    if isinstance(inputs, str):
+        inputs = [inputs]
+    vecs = await loop.run_in_executor(None, inst.embed, inputs)
+    return {
+        "object": "list",
+        "model": model,
+        "data": [{"object": "embedding", "index": i, "embedding": v}
+                 for i, v in enumerate(vecs)],
+        "usage": {"prompt_tokens": sum(len(t) for t in inputs),
+                  "total_tokens": sum(len(t) for t in inputs)},
+    }

@@ -1,0 +1,291 @@
+"""RunnerService — the native MI355X model runner.
+
+Replaces the reference's entire delegated GPU plane (compose-manager +
+inference-proxy + vLLM/Ollama containers) with in-process
+engines on hand-written CDNA4 kernels, plus the HBM-aware multi-model
+scheduler the reference deleted (SURVEY.md §2.8 "Model pack/evict":
+estimate-based admission driven by live free HBM, LRU eviction that
+coordinates with in-flight batches).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from helix_amd.engine.engine import EngineConfig, LLMEngine, Sequence
+from helix_amd.engine.sampling_params import SamplingParams
+from helix_amd.models.bert import BERT_PRESETS, BertEmbeddingModel
+from helix_amd.models.llama import PRESETS as LLAMA_PRESETS
+from helix_amd.utils.tokenizer import get_tokenizer
+
+log = logging.getLogger("helix_amd.runner")
+
+
+@dataclass
+class ModelSpec:
+    """Manifest entry (replaces compose-profile YAML model entries)."""
+    name: str
+    kind: str = "llm"                  # llm | embedding
+    preset: str = "llama3-8b"
+    max_model_len: int = 8192
+    max_num_seqs: int = 64
+    kv_cache_blocks: Optional[int] = None
+    kv_memory_fraction: float = 0.30   # of total HBM, for auto-sizing
+
+
+DEFAULT_SPECS = {
+    "llama3-8b": ModelSpec("llama3-8b", "llm", "llama3-8b"),
+    "llama3-70b": ModelSpec("llama3-70b", "llm", "llama3-70b"),
+    "mistral-7b": ModelSpec("mistral-7b", "llm", "mistral-7b"),
+    "bge-base": ModelSpec("bge-base", "embedding", "bge-base"),
+    "bge-large": ModelSpec("bge-large", "embedding", "bge-large"),
+    # tiny models for CPU tests
+    "tiny": ModelSpec("tiny", "llm", "tiny", max_model_len=256,
+                      kv_cache_blocks=256),
+    "tiny-gqa": ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=512,
+                          kv_cache_blocks=256),
+    "tiny-bert": ModelSpec("tiny-bert", "embedding", "tiny-bert"),
+}
+
+
+def estimate_model_bytes(spec: ModelSpec, block_size: int = 16) -> int:
+    """Admission estimate: weights + KV budget + workspace headroom.
+    Descendant of the reference's GGUF estimator (api/pkg/memory/estimate.go)
+    re-based on bf16 dense checkpoints."""
+    if spec.kind == "embedding":
+        cfg = BERT_PRESETS[spec.preset]
+        n = cfg.vocab_size * cfg.hidden_size + cfg.max_position * cfg.hidden_size
+        per_layer = 4 * cfg.hidden_size ** 2 + \
+            2 * cfg.hidden_size * cfg.intermediate_size
+        return 2 * (n + cfg.num_layers * per_layer) + (64 << 20)
+    cfg = LLAMA_PRESETS[spec.preset]
+    emb = cfg.vocab_size * cfg.hidden_size * (1 if cfg.tie_embeddings else 2)
+    per_layer = (cfg.hidden_size * (cfg.q_size + 2 * cfg.kv_size)  # qkv
+                 + cfg.q_size * cfg.hidden_size                     # o
+                 + 3 * cfg.hidden_size * cfg.intermediate_size)     # mlp
+    weights = 2 * (emb + cfg.num_layers * per_layer)
+    kv_block = 2 * cfg.num_layers * cfg.num_kv_heads * block_size * \
+        cfg.head_dim * 2
+    if spec.kv_cache_blocks:
+        kv = kv_block * spec.kv_cache_blocks
+    else:
+        kv = kv_block * spec.max_num_seqs * \
+            (spec.max_model_len // block_size + 1)
+    return weights + kv + (1 << 30)    # +1 GiB activations/graphs headroom
+
+
+class LLMInstance:
+    """A loaded LLM: engine + dedicated step-loop thread."""
+
+    def __init__(self, spec: ModelSpec, device: str):
+        self.spec = spec
+        self.device = device
+        kv_blocks = spec.kv_cache_blocks
+        if kv_blocks is None and torch.cuda.is_available():
+            free, total = torch.cuda.mem_get_info(torch.device(device))
+            from helix_amd.engine.kv_cache import KVCache
+            cfg = LLAMA_PRESETS[spec.preset]
+            budget = int(total * spec.kv_memory_fraction)
+            kv_blocks = KVCache.blocks_for_bytes(
+                min(budget, max(free - (2 << 30), 1 << 28)),
+                cfg.num_layers, cfg.num_kv_heads, cfg.head_dim, 16)
+        self.engine = LLMEngine(
+            EngineConfig(model=spec.preset, max_model_len=spec.max_model_len,
+                         max_num_seqs=spec.max_num_seqs,
+                         kv_cache_blocks=kv_blocks),
+            device=device)
+        self.lock = threading.Lock()
+        self.wake = threading.Event()
+        self.stop = False
+        self.last_used = time.time()
+        self.thread = threading.Thread(target=self._loop, daemon=True,
+                                       name=f"engine-{spec.name}")
+        self.thread.start()
+
+    @property
+    def in_flight(self) -> int:
+        with self.lock:
+            return len(self.engine.waiting) + len(self.engine.running)
+
+    def _loop(self):
+        while not self.stop:
+            with self.lock:
+                has_work = self.engine.has_work
+            if not has_work:
+                self.wake.wait(timeout=0.05)
+                self.wake.clear()
+                continue
+            with self.lock:
+                self.engine.step()
+
+    def submit(self, seq_id: str, prompt_ids: List[int],
+               params: SamplingParams, on_token) -> Sequence:
+        self.last_used = time.time()
+        with self.lock:
+            seq = self.engine.add_request(seq_id, prompt_ids, params,
+                                          on_token=on_token)
+        self.wake.set()
+        return seq
+
+    def cancel(self, seq_id: str):
+        with self.lock:
+            self.engine.cancel(seq_id)
+
+    def shutdown(self):
+        self.stop = True
+        self.wake.set()
+        self.thread.join(timeout=10)
+        del self.engine
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
+
+class EmbeddingInstance:
+    def __init__(self, spec: ModelSpec, device: str):
+        self.spec = spec
+        self.device = torch.device(device)
+        cfg = BERT_PRESETS[spec.preset]
+        dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        self.model = BertEmbeddingModel(cfg).to(dtype).to(self.device)
+        self.model.init_random(0)
+        self.tokenizer = get_tokenizer(spec.name)
+        self.lock = threading.Lock()
+        self.last_used = time.time()
+
+    @property
+    def in_flight(self) -> int:
+        return 0
+
+    def embed(self, texts: List[str]) -> List[List[float]]:
+        self.last_used = time.time()
+        cfg = self.model.cfg
+        ids_list = [self.tokenizer.encode(t)[: cfg.max_position - 1] or [1]
+                    for t in texts]
+        flat, cu = [], [0]
+        for ids in ids_list:
+            flat.extend(ids)
+            cu.append(cu[-1] + len(ids))
+        with self.lock:
+            ids_t = torch.tensor(flat, dtype=torch.int64, device=self.device)
+            cu_t = torch.tensor(cu, dtype=torch.int32, device=self.device)
+            out = self.model(ids_t, cu_t, max(len(i) for i in ids_list))
+        return out.cpu().tolist()
+
+    def shutdown(self):
+        del self.model
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
+
+class RunnerService:
+    """Owns model instances on one GPU (or CPU for tests); packs/evicts
+    under the HBM budget."""
+
+    def __init__(self, device: str = "cuda:0",
+                 specs: Optional[Dict[str, ModelSpec]] = None,
+                 memory_budget: Optional[int] = None):
+        self.device = device
+        self.specs = dict(DEFAULT_SPECS)
+        if specs:
+            self.specs.update(specs)
+        self.instances: Dict[str, object] = {}
+        self._lock = threading.Lock()
+        if memory_budget is None:
+            if torch.cuda.is_available() and device.startswith("cuda"):
+                _, total = torch.cuda.mem_get_info(torch.device(device))
+                memory_budget = int(total * 0.92)
+            else:
+                memory_budget = 64 << 30
+        self.memory_budget = memory_budget
+
+    # ---------------- scheduler -------------------------------------------
+    def free_hbm(self) -> int:
+        if torch.cuda.is_available() and self.device.startswith("cuda"):
+            free, _ = torch.cuda.mem_get_info(torch.device(self.device))
+            return free
+        used = sum(estimate_model_bytes(i.spec)
+                   for i in self.instances.values())
+        return self.memory_budget - used
+
+    def loaded_models(self) -> List[str]:
+        return list(self.instances.keys())
+
+    def ensure_loaded(self, model: str):
+        """Admission control: load `model`, LRU-evicting idle models if the
+        estimate does not fit in free HBM. Raises NoCapacityError if it
+        cannot fit even after eviction (control plane surfaces 503,
+        matching the reference's NoRunnerError behavior, router.go:62)."""
+        with self._lock:
+            if model in self.instances:
+                return self.instances[model]
+            spec = self.specs.get(model)
+            if spec is None:
+                raise ModelNotFoundError(model)
+            need = estimate_model_bytes(spec)
+            # Evict LRU idle models until it fits.
+            while self.free_hbm() < need:
+                victim = self._pick_victim()
+                if victim is None:
+                    raise NoCapacityError(
+                        f"model {model} needs {need >> 30} GiB; "
+                        f"free {self.free_hbm() >> 30} GiB and no evictable "
+                        f"model")
+                log.info("evicting %s to fit %s", victim, model)
+                self._unload(victim)
+            log.info("loading %s (%d GiB est.)", model, need >> 30)
+            t0 = time.time()
+            if spec.kind == "embedding":
+                inst = EmbeddingInstance(spec, self.device)
+            else:
+                inst = LLMInstance(spec, self.device)
+            log.info("loaded %s in %.1fs", model, time.time() - t0)
+            self.instances[model] = inst
+            return inst
+
+    def _pick_victim(self) -> Optional[str]:
+        idle = [(inst.last_used, name)
+                for name, inst in self.instances.items()
+                if inst.in_flight == 0]
+        if not idle:
+            return None
+        return min(idle)[1]
+
+    def _unload(self, model: str):
+        inst = self.instances.pop(model)
+        inst.shutdown()
+
+    def unload(self, model: str):
+        with self._lock:
+            if model in self.instances:
+                self._unload(model)
+
+    def status(self) -> List[dict]:
+        out = []
+        for name, inst in self.instances.items():
+            out.append({
+                "model_id": name,
+                "state": "ready",
+                "memory_bytes": estimate_model_bytes(inst.spec),
+                "in_flight": inst.in_flight,
+                "last_used": inst.last_used,
+            })
+        return out
+
+    def shutdown(self):
+        with self._lock:
+            for name in list(self.instances):
+                self._unload(name)
+
+
+class ModelNotFoundError(Exception):
+    pass
+
+
+class NoCapacityError(Exception):
+    pass

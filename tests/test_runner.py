@@ -1,0 +1,121 @@
+"""Runner service + OpenAI adapter tests on CPU (tiny models).
+
+Mirrors the reference's inferenceproxy/proxy_test.go + gpucloud
+inference_roundtrip scenario at unit scale.
+"""
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from helix_amd.runner.http import create_runner_app
+from helix_amd.runner.service import (DEFAULT_SPECS, ModelSpec,
+                                      NoCapacityError, RunnerService,
+                                      estimate_model_bytes)
+
+
+@pytest.fixture()
+def service():
+    svc = RunnerService(device="cpu", memory_budget=64 << 30)
+    yield svc
+    svc.shutdown()
+
+
+@pytest.fixture()
+def client(service):
+    app = create_runner_app(service)
+    with TestClient(app) as c:
+        yield c
+
+
+def test_estimate_llama8b_size():
+    est = estimate_model_bytes(DEFAULT_SPECS["llama3-8b"])
+    # 8B params bf16 ~16 GB + KV + headroom
+    assert 16 << 30 < est < 120 << 30
+
+
+def test_chat_completion_roundtrip(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "tiny",
+        "messages": [{"role": "user", "content": "hello"}],
+        "max_tokens": 8, "temperature": 0,
+    })
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body["object"] == "chat.completion"
+    assert body["choices"][0]["message"]["role"] == "assistant"
+    assert body["usage"]["completion_tokens"] >= 1
+
+
+def test_chat_completion_stream(client):
+    with client.stream("POST", "/v1/chat/completions", json={
+        "model": "tiny",
+        "messages": [{"role": "user", "content": "hi"}],
+        "max_tokens": 6, "temperature": 0, "stream": True,
+    }) as r:
+        assert r.status_code == 200
+        lines = [l for l in r.iter_lines() if l.startswith("data: ")]
+    assert lines[-1] == "data: [DONE]"
+    import json
+    chunks = [json.loads(l[6:]) for l in lines[:-1]]
+    assert chunks[0]["choices"][0]["delta"].get("role") == "assistant"
+    assert chunks[-1]["choices"][0]["finish_reason"] is not None
+    assert chunks[-1]["usage"]["completion_tokens"] >= 1
+
+
+def test_model_not_found(client):
+    r = client.post("/v1/chat/completions", json={
+        "model": "nope", "messages": [{"role": "user", "content": "x"}]})
+    assert r.status_code == 404
+
+
+def test_embeddings(client):
+    r = client.post("/v1/embeddings", json={
+        "model": "tiny-bert", "input": ["hello world", "second text"]})
+    assert r.status_code == 200
+    data = r.json()["data"]
+    assert len(data) == 2
+    v = torch.tensor(data[0]["embedding"])
+    assert abs(float(v.norm()) - 1.0) < 1e-3  # L2 normalized
+
+
+def test_models_list(client):
+    r = client.get("/v1/models")
+    ids = [m["id"] for m in r.json()["data"]]
+    assert "tiny" in ids and "llama3-8b" in ids
+
+
+def test_scheduler_evicts_lru(service):
+    # budget fits ~1 tiny model at a time (estimate ~1.1 GiB incl. the
+    # 1 GiB activation headroom)
+    svc = RunnerService(device="cpu", memory_budget=int(1.5e9))
+    specs = {
+        "a": ModelSpec("a", "llm", "tiny", max_model_len=256,
+                       kv_cache_blocks=64),
+        "b": ModelSpec("b", "llm", "tiny", max_model_len=256,
+                       kv_cache_blocks=64),
+    }
+    svc.specs.update(specs)
+    try:
+        svc.ensure_loaded("a")
+        assert "a" in svc.loaded_models()
+        svc.ensure_loaded("b")   # must evict a
+        assert "b" in svc.loaded_models()
+        assert "a" not in svc.loaded_models()
+    finally:
+        svc.shutdown()
+
+
+def test_scheduler_no_capacity():
+    svc = RunnerService(device="cpu", memory_budget=1000)
+    try:
+        with pytest.raises(NoCapacityError):
+            svc.ensure_loaded("tiny")
+    finally:
+        svc.shutdown()
+
+
+def test_status_endpoint(client):
+    client.post("/api/v1/models/tiny/load")
+    r = client.get("/api/v1/status")
+    body = r.json()
+    assert any(m["model_id"] == "tiny" for m in body["models"])
