@@ -1,0 +1,339 @@
+"""HelixAPIServer — the control-plane HTTP app (parity with
+api/pkg/server registerRoutes, server.go:863): OpenAI-compatible surface,
+sessions, apps/agents, runner plane, admin.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+from typing import Optional
+
+from fastapi import Depends, FastAPI, HTTPException, Request, WebSocket
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from helix_amd.server import pubsub as ps
+from helix_amd.server.auth import (Authenticator, AuthUser, make_admin_dep,
+                                   make_auth_dep, make_runner_dep)
+from helix_amd.server.config import ServerConfig, load_config
+from helix_amd.server.controller import Controller
+from helix_amd.server.providers import (LocalRunnerClient, MockClient,
+                                        OpenAIHTTPClient, ProviderError,
+                                        ProviderManager, RetryableClient,
+                                        RouterClient)
+from helix_amd.server.router import InferenceRouter, NoRunnerError
+from helix_amd.server.types import (App, AppHelixConfig, RunnerHeartbeat,
+                                    new_id)
+from helix_amd.store import Store
+
+log = logging.getLogger("helix_amd.server")
+
+auth_dep = make_auth_dep()
+admin_dep = make_admin_dep()
+runner_dep = make_runner_dep()
+
+
+def create_app(cfg: Optional[ServerConfig] = None,
+               store: Optional[Store] = None,
+               providers: Optional[ProviderManager] = None,
+               runner_service=None) -> FastAPI:
+    cfg = cfg or load_config()
+    store = store if store is not None else Store(cfg.store.path)
+    app = FastAPI(title="helix_amd control plane", docs_url=None)
+    auth = Authenticator(store, cfg.web.admin_api_key,
+                         cfg.runner_plane.runner_token)
+    router = InferenceRouter(cfg.runner_plane.dispatch_stale_s,
+                             cfg.runner_plane.offline_after_s)
+    pubsub = ps.PubSub()
+
+    if providers is None:
+        providers = ProviderManager(store)
+        if cfg.providers.openai_api_key:
+            providers.register("openai", RetryableClient(OpenAIHTTPClient(
+                "openai", cfg.providers.openai_base_url,
+                cfg.providers.openai_api_key)))
+        if cfg.providers.together_api_key:
+            providers.register("togetherai", RetryableClient(OpenAIHTTPClient(
+                "togetherai", cfg.providers.together_base_url,
+                cfg.providers.together_api_key)))
+        if runner_service is not None:
+            providers.register("helix", LocalRunnerClient(runner_service))
+        elif cfg.runner_plane.local_runner:
+            from helix_amd.runner.service import RunnerService
+            runner_service = RunnerService(
+                device=cfg.runner_plane.local_runner_device)
+            providers.register("helix", LocalRunnerClient(runner_service))
+        else:
+            providers.register("helix", RouterClient(router))
+
+    # RAG + agents wired in by their modules (optional deps).
+    rag = None
+    try:
+        from helix_amd.rag.service import RAGService
+        rag = RAGService(cfg, store, providers)
+    except Exception as e:  # pragma: no cover
+        log.warning("RAG unavailable: %s", e)
+    agent_runner = None
+    try:
+        from helix_amd.agent.runner import AgentRunner
+        agent_runner = AgentRunner(cfg, store, providers, pubsub)
+    except Exception as e:  # pragma: no cover
+        log.warning("agent runtime unavailable: %s", e)
+
+    controller = Controller(cfg, store, providers, pubsub, rag=rag,
+                            agent_runner=agent_runner)
+
+    app.state.cfg = cfg
+    app.state.store = store
+    app.state.auth = auth
+    app.state.router = router
+    app.state.pubsub = pubsub
+    app.state.providers = providers
+    app.state.controller = controller
+    app.state.runner_service = runner_service
+    app.state.rag = rag
+
+    @app.exception_handler(ProviderError)
+    async def _pe(request, exc: ProviderError):
+        return JSONResponse({"error": {"message": str(exc)}},
+                            status_code=exc.status)
+
+    @app.exception_handler(NoRunnerError)
+    async def _nr(request, exc: NoRunnerError):
+        # reference router.go:62: NoRunnerError -> 503 with available list
+        return JSONResponse({"error": {"message": str(exc),
+                                       "available": exc.available}},
+                            status_code=503)
+
+    # ------------------------------------------------------------------
+    # OpenAI-compatible surface
+    # ------------------------------------------------------------------
+    async def _chat(request: Request, user: AuthUser, body: dict):
+        app_id = body.pop("app_id", "") or \
+            request.headers.get("X-Helix-App-Id", "")
+        assistant_id = body.pop("assistant_id", "")
+        ctx = {"owner": user.id}
+        if body.get("stream"):
+            async def sse():
+                try:
+                    async for chunk in controller.chat_completion_stream(
+                            body, user.id, app_id, assistant_id, ctx):
+                        yield f"data: {json.dumps(chunk)}\n\n"
+                except (ProviderError, NoRunnerError) as e:
+                    yield ("data: " + json.dumps(
+                        {"error": {"message": str(e)}}) + "\n\n")
+                yield "data: [DONE]\n\n"
+            return StreamingResponse(sse(), media_type="text/event-stream")
+        return await controller.chat_completion(body, user.id, app_id,
+                                                assistant_id, ctx)
+
+    @app.post("/v1/chat/completions")
+    async def chat_completions(request: Request,
+                               user: AuthUser = Depends(auth_dep)):
+        return await _chat(request, user, await request.json())
+
+    @app.post("/openai/deployments/{model}/chat/completions")
+    async def azure_chat(model: str, request: Request,
+                         user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        body.setdefault("model", model)
+        return await _chat(request, user, body)
+
+    @app.post("/v1/embeddings")
+    async def embeddings(request: Request,
+                         user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        provider, model = providers.resolve(
+            body.get("model", ""), cfg.rag.embeddings_provider, user.id)
+        client = providers.get_client(provider, user.id)
+        return await client.embeddings({**body, "model": model})
+
+    @app.get("/v1/models")
+    async def models(user: AuthUser = Depends(auth_dep)):
+        data = await providers.aggregate_models(user.id)
+        return {"object": "list", "data": data}
+
+    # ------------------------------------------------------------------
+    # Sessions API (reference server.go:1024-1064)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/sessions/chat")
+    async def session_chat(request: Request,
+                           user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        session_id = body.get("session_id", "")
+        if session_id:
+            session = controller.get_session(session_id)
+            if session is None or session.owner != user.id:
+                raise HTTPException(404, "session not found")
+        else:
+            session = controller.create_session(
+                user.id, model=body.get("model", ""),
+                provider=body.get("provider", ""),
+                app_id=body.get("app_id", ""),
+                name=(body.get("messages") or [{}])[-1].get(
+                    "content", "")[:40] or "New Session")
+        msgs = body.get("messages") or []
+        prompt = msgs[-1].get("content", "") if msgs else body.get("text", "")
+        interaction = controller.add_interaction(session, prompt)
+
+        async def sse():
+            yield ("data: " + json.dumps(
+                {"type": "session", "session_id": session.id,
+                 "interaction_id": interaction.id}) + "\n\n")
+            try:
+                async for chunk in controller.run_session_turn(session,
+                                                               interaction):
+                    yield f"data: {json.dumps(chunk)}\n\n"
+            except (ProviderError, NoRunnerError) as e:
+                yield ("data: " + json.dumps(
+                    {"error": {"message": str(e)}}) + "\n\n")
+            yield "data: [DONE]\n\n"
+        return StreamingResponse(sse(), media_type="text/event-stream")
+
+    @app.get("/api/v1/sessions")
+    async def list_sessions(user: AuthUser = Depends(auth_dep)):
+        return [s.model_dump() for s in controller.list_sessions(user.id)]
+
+    @app.get("/api/v1/sessions/{session_id}")
+    async def get_session(session_id: str,
+                          user: AuthUser = Depends(auth_dep)):
+        s = controller.get_session(session_id)
+        if s is None or (s.owner != user.id and not user.admin):
+            raise HTTPException(404, "session not found")
+        doc = s.model_dump()
+        doc["interactions"] = store.list("interactions", parent=session_id,
+                                         desc=False)
+        return doc
+
+    @app.delete("/api/v1/sessions/{session_id}")
+    async def delete_session(session_id: str,
+                             user: AuthUser = Depends(auth_dep)):
+        s = controller.get_session(session_id)
+        if s is None or (s.owner != user.id and not user.admin):
+            raise HTTPException(404, "session not found")
+        controller.delete_session(session_id)
+        return {"ok": True}
+
+    @app.websocket("/api/v1/ws/user")
+    async def ws_user(ws: WebSocket):
+        token = ws.query_params.get("access_token", "")
+        user = auth.resolve(token)
+        if user is None:
+            await ws.close(code=4401)
+            return
+        await ws.accept()
+        sub = await pubsub.subscribe(ps.session_queue(user.id, "*"))
+        try:
+            while True:
+                topic, msg = await sub.get()
+                await ws.send_json({"topic": topic, "payload": msg})
+        except Exception:
+            pass
+        finally:
+            await sub.close()
+
+    # ------------------------------------------------------------------
+    # Apps / agents (registered under BOTH aliases, agent_routes.go:12-50)
+    # ------------------------------------------------------------------
+    def _get_app(app_id: str, user: AuthUser) -> App:
+        doc = store.get("apps", app_id)
+        if doc is None:
+            raise HTTPException(404, "app not found")
+        a = App.model_validate(doc)
+        if a.owner != user.id and not a.global_ and not user.admin:
+            raise HTTPException(403, "forbidden")
+        return a
+
+    for prefix in ("/api/v1/apps", "/api/v1/agents"):
+        def _bind(prefix=prefix):
+            @app.post(prefix, name=f"create_app_{prefix}")
+            async def create_app_h(request: Request,
+                                   user: AuthUser = Depends(auth_dep)):
+                body = await request.json()
+                config = AppHelixConfig.model_validate(
+                    body.get("config", body))
+                a = App(owner=user.id, config=config,
+                        global_=bool(body.get("global", False)))
+                store.put("apps", a.id, a.model_dump(by_alias=True),
+                          owner=user.id)
+                return a.model_dump(by_alias=True)
+
+            @app.get(prefix, name=f"list_apps_{prefix}")
+            async def list_apps_h(user: AuthUser = Depends(auth_dep)):
+                docs = store.list("apps", owner=user.id)
+                return docs
+
+            @app.get(prefix + "/{app_id}", name=f"get_app_{prefix}")
+            async def get_app_h(app_id: str,
+                                user: AuthUser = Depends(auth_dep)):
+                return _get_app(app_id, user).model_dump(by_alias=True)
+
+            @app.put(prefix + "/{app_id}", name=f"update_app_{prefix}")
+            async def update_app_h(app_id: str, request: Request,
+                                   user: AuthUser = Depends(auth_dep)):
+                a = _get_app(app_id, user)
+                body = await request.json()
+                a.config = AppHelixConfig.model_validate(
+                    body.get("config", body))
+                from helix_amd.server.types import now_ms
+                a.updated = now_ms()
+                store.put("apps", a.id, a.model_dump(by_alias=True),
+                          owner=a.owner)
+                return a.model_dump(by_alias=True)
+
+            @app.delete(prefix + "/{app_id}", name=f"delete_app_{prefix}")
+            async def delete_app_h(app_id: str,
+                                   user: AuthUser = Depends(auth_dep)):
+                _get_app(app_id, user)
+                store.delete("apps", app_id)
+                return {"ok": True}
+        _bind()
+
+    # ------------------------------------------------------------------
+    # Runner plane (heartbeats -> router, reference sandbox_handlers.go:84)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/runner/heartbeat")
+    async def runner_heartbeat(request: Request,
+                               _=Depends(runner_dep)):
+        hb = RunnerHeartbeat.model_validate(await request.json())
+        router.on_heartbeat(hb)
+        return {"ok": True}
+
+    @app.get("/api/v1/admin/runners")
+    async def list_runners(user: AuthUser = Depends(admin_dep)):
+        router.reap_offline()
+        return [r.model_dump() for r in router.runners()]
+
+    # ------------------------------------------------------------------
+    # Users / keys / admin
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/users")
+    async def create_user(request: Request,
+                          user: AuthUser = Depends(admin_dep)):
+        body = await request.json()
+        u = auth.create_user(body["username"], bool(body.get("admin")))
+        key = auth.create_api_key(u["id"])
+        return {**u, "api_key": key}
+
+    @app.post("/api/v1/api_keys")
+    async def create_key(user: AuthUser = Depends(auth_dep)):
+        return {"api_key": auth.create_api_key(user.id)}
+
+    @app.get("/api/v1/llm_calls")
+    async def llm_calls(user: AuthUser = Depends(admin_dep),
+                        session_id: str = ""):
+        if session_id:
+            return store.list("llm_calls", parent=session_id)
+        return store.list("llm_calls", limit=200)
+
+    @app.get("/api/v1/config")
+    async def get_config(user: AuthUser = Depends(auth_dep)):
+        return {"version": "helix_amd-0.1.0",
+                "default_model": cfg.inference.default_model,
+                "default_provider": cfg.inference.default_provider}
+
+    @app.get("/healthz")
+    async def healthz():
+        return {"ok": True}
+
+    return app

@@ -1,0 +1,210 @@
+"""Control-plane API tests (memorystore + mock provider pattern,
+mirroring the reference's server.NewTestServer + gomock approach)."""
+import json
+
+import pytest
+from fastapi.testclient import TestClient
+
+from helix_amd.server.app import create_app
+from helix_amd.server.config import ServerConfig
+from helix_amd.server.providers import MockClient, ProviderManager
+from helix_amd.store import Store
+
+
+@pytest.fixture()
+def stack():
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    mock = MockClient()
+    pm.register("mock", mock)
+    app = create_app(cfg, store=store, providers=pm)
+    client = TestClient(app)
+    # bootstrap a user via admin key
+    r = client.post("/api/v1/users", json={"username": "alice"},
+                    headers={"Authorization": "Bearer admin-key"})
+    key = r.json()["api_key"]
+    return app, client, mock, key, store
+
+
+def H(key):
+    return {"Authorization": f"Bearer {key}"}
+
+
+def test_auth_required(stack):
+    _, client, _, _, _ = stack
+    assert client.post("/v1/chat/completions", json={}).status_code == 401
+    assert client.post("/v1/chat/completions", json={}, headers=H(
+        "bogus")).status_code == 401
+
+
+def test_chat_completion_mock(stack):
+    _, client, mock, key, store = stack
+    r = client.post("/v1/chat/completions", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "hi"}]}, headers=H(key))
+    assert r.status_code == 200, r.text
+    assert r.json()["choices"][0]["message"]["content"] == "mock response"
+    # llm_calls logged with usage
+    calls = store.list("llm_calls")
+    assert len(calls) == 1
+    assert calls[0]["prompt_tokens"] == 7
+
+
+def test_chat_completion_stream(stack):
+    _, client, _, key, _ = stack
+    with client.stream("POST", "/v1/chat/completions", json={
+        "model": "mock-model", "stream": True,
+        "messages": [{"role": "user", "content": "hi"}]},
+            headers=H(key)) as r:
+        lines = [l for l in r.iter_lines() if l.startswith("data: ")]
+    assert lines[-1] == "data: [DONE]"
+    chunks = [json.loads(l[6:]) for l in lines[:-1]]
+    text = "".join(c["choices"][0]["delta"].get("content", "")
+                   for c in chunks)
+    assert text == "mock response"
+
+
+def test_models_aggregate(stack):
+    _, client, _, key, _ = stack
+    r = client.get("/v1/models", headers=H(key))
+    ids = [m["id"] for m in r.json()["data"]]
+    assert "mock/mock-model" in ids
+
+
+def test_azure_alias(stack):
+    _, client, _, key, _ = stack
+    r = client.post("/openai/deployments/mock-model/chat/completions",
+                    json={"messages": [{"role": "user", "content": "x"}]},
+                    headers=H(key))
+    assert r.status_code == 200
+
+
+def test_session_chat_and_history(stack):
+    _, client, _, key, store = stack
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "first turn"}]},
+            headers=H(key)) as r:
+        lines = [l for l in r.iter_lines() if l.startswith("data: ")]
+    head = json.loads(lines[0][6:])
+    sid = head["session_id"]
+    assert head["type"] == "session"
+
+    # history persisted
+    r = client.get(f"/api/v1/sessions/{sid}", headers=H(key))
+    body = r.json()
+    assert len(body["interactions"]) == 1
+    assert body["interactions"][0]["response_message"] == "mock response"
+    assert body["interactions"][0]["state"] == "complete"
+    assert body["interactions"][0]["ttft_ms"] >= 0
+
+    # second turn carries history
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "session_id": sid,
+        "messages": [{"role": "user", "content": "second turn"}]},
+            headers=H(key)) as r:
+        [l for l in r.iter_lines()]
+    r = client.get(f"/api/v1/sessions/{sid}", headers=H(key))
+    assert len(r.json()["interactions"]) == 2
+
+
+def test_session_isolation(stack):
+    app, client, _, key, _ = stack
+    r = client.post("/api/v1/users", json={"username": "bob"},
+                    headers=H("admin-key"))
+    bob_key = r.json()["api_key"]
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "secret"}]},
+            headers=H(key)) as r:
+        sid = json.loads(next(l for l in r.iter_lines()
+                              if l.startswith("data: "))[6:])["session_id"]
+    assert client.get(f"/api/v1/sessions/{sid}",
+                      headers=H(bob_key)).status_code == 404
+
+
+def test_app_crud_and_assistant_config(stack):
+    _, client, mock, key, _ = stack
+    helix_yaml = {
+        "name": "My Agent",
+        "assistants": [{
+            "name": "default",
+            "model": "mock-model",
+            "provider": "mock",
+            "system_prompt": "You are terse.",
+            "temperature": 0.3,
+        }],
+    }
+    r = client.post("/api/v1/apps", json={"config": helix_yaml},
+                    headers=H(key))
+    assert r.status_code == 200, r.text
+    app_id = r.json()["id"]
+
+    # both aliases list it
+    assert any(a["id"] == app_id for a in client.get(
+        "/api/v1/agents", headers=H(key)).json())
+
+    # chat through the app applies system prompt + sampling params
+    r = client.post("/v1/chat/completions", json={
+        "app_id": app_id,
+        "messages": [{"role": "user", "content": "hello"}]}, headers=H(key))
+    assert r.status_code == 200
+    sent = mock.calls[-1]
+    assert sent["messages"][0]["role"] == "system"
+    assert sent["messages"][0]["content"] == "You are terse."
+    assert sent["temperature"] == 0.3
+    assert sent["model"] == "mock-model"
+
+    # update + delete
+    helix_yaml["assistants"][0]["system_prompt"] = "Changed."
+    r = client.put(f"/api/v1/apps/{app_id}", json={"config": helix_yaml},
+                   headers=H(key))
+    assert r.json()["config"]["assistants"][0]["system_prompt"] == "Changed."
+    assert client.delete(f"/api/v1/apps/{app_id}",
+                         headers=H(key)).json()["ok"]
+
+
+def test_runner_heartbeat_and_router(stack):
+    _, client, _, key, _ = stack
+    hb = {"runner_id": "r1", "address": "http://10.0.0.5:8090",
+          "gpus": [{"index": 0, "arch": "cdna4", "total_memory": 309237645312,
+                    "free_memory": 200000000000}],
+          "models": [{"model_id": "llama3-8b", "state": "ready"}]}
+    r = client.post("/api/v1/runner/heartbeat", json=hb,
+                    headers=H("runner-token"))
+    assert r.status_code == 200
+    # wrong token rejected
+    assert client.post("/api/v1/runner/heartbeat", json=hb,
+                       headers=H(key)).status_code == 401
+    r = client.get("/api/v1/admin/runners", headers=H("admin-key"))
+    assert r.json()[0]["runner_id"] == "r1"
+
+
+def test_no_runner_503(stack):
+    app, client, _, key, _ = stack
+    # helix provider with empty router -> 503 NoRunnerError
+    cfg = app.state.cfg
+    r = client.post("/v1/chat/completions", json={
+        "model": "helix/whatever",
+        "messages": [{"role": "user", "content": "x"}]}, headers=H(key))
+    # 'helix' not registered as prefix in this stack (mock only) -> falls to
+    # default provider. Register router-backed helix instead:
+    from helix_amd.server.providers import RouterClient
+    app.state.providers.register("helix", RouterClient(app.state.router))
+    app.state.providers._model_cache.clear()
+    r = client.post("/v1/chat/completions", json={
+        "model": "helix/llama3-8b",
+        "messages": [{"role": "user", "content": "x"}]}, headers=H(key))
+    assert r.status_code == 503
+    assert "available" in r.json()["error"]
+
+
+def test_llm_calls_admin_only(stack):
+    _, client, _, key, _ = stack
+    assert client.get("/api/v1/llm_calls",
+                      headers=H(key)).status_code == 403
+    assert client.get("/api/v1/llm_calls",
+                      headers=H("admin-key")).status_code == 200
