@@ -4,21 +4,21 @@
 // Replaces the decode attention the reference delegates to vLLM
 // (SURVEY.md §2.8 "Decode attention (paged KV)").
 //
-// Design (MI355X-first, v2 — occupancy-driven):
-//  - Decode attention is KV-bandwidth-bound: the kernel reads each KV byte
-//    exactly once per kv-head and amortizes it across the G = Hq/Hkv query
-//    heads of the GQA group.
-//  - grid = (num_seqs, Hkv, num_partitions): flash-decoding split-K, with
-//    the partition count chosen on the host to put >= ~1024 workgroups on
-//    the 256 CUs (v1 used 1 block/CU and was 10x off the bandwidth bound —
+// Design (MI355X-first, v3 — templated + load-batched):
+//  - KV-bandwidth-bound: each KV byte is read once per kv-head and
+//    amortized across the G = Hq/Hkv GQA query heads.
+//  - grid = (num_seqs, Hkv, num_partitions): flash-decoding split-K, host
+//    sizes partitions to fill the 256 CUs (>= ~1024 workgroups).
+//  - DHEAD and G are template parameters so every inner loop fully
+//    unrolls; K rows and V tiles are loaded into registers in batches
+//    BEFORE any math so the wave keeps many VMEM ops in flight (v2's
+//    runtime-D loops serialized load->use and reached only 0.5 TB/s —
 //    profiles/r01_decode_profile_v1.md).
-//  - 128-thread blocks (2 waves) walk their partition in 128-token chunks:
-//      A: thread t loads the K row of token t (vectorized 16B), dots against
-//         the G query vectors held in LDS (broadcast reads), scores -> LDS.
-//      B: per-head online-softmax update (running m, l), one wave per head.
-//      C: V accumulation, thread owns a dim PAIR (b32 loads) for all G heads.
-//  - Single-partition grids write normalized bf16 straight to `out`;
-//    multi-partition grids write fp32 partials + (m, l) for a reduce kernel.
+//  - 128-thread blocks (2 waves) walk the partition in 128-token chunks:
+//      A: thread t loads K row of token t (DHEAD/8 x 16B batched), dots
+//         against the G query vectors in LDS (broadcast reads).
+//      B: per-head online softmax (running m, l), one wave per head.
+//      C: V accumulation, thread owns a dim pair, 8-token load batches.
 #include "common.h"
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -30,8 +30,9 @@ namespace {
 constexpr int NTHREADS = 128;
 constexpr int CHUNK = 128;
 constexpr int MAX_G = 8;
-constexpr int PART_QUANT = 128;   // partition sizes are multiples of this
+constexpr int PART_QUANT = 128;
 
+template <int DHEAD, int G>
 __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
@@ -41,18 +42,18 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
     const uint16_t* __restrict__ v_cache,
     const int* __restrict__ block_tables,// [B, max_blocks]
     const int* __restrict__ seq_lens,    // [B]
-    float scale, int Hq, int Hkv, int D, int block_size, int max_blocks,
+    float scale, int Hkv, int block_size, int max_blocks,
     int partition_size, int max_parts) {
   const int seq = blockIdx.x;
   const int hkv = blockIdx.y;
   const int part = blockIdx.z;
   const int nparts = gridDim.z;
+  const int Hq = Hkv * G;
   const int len = seq_lens[seq];
-  const int G = Hq / Hkv;
   const int p_start = part * partition_size;
   if (p_start >= len) {
-    // Dead partition: mark so the reduce kernel skips it.
     if (nparts > 1 && threadIdx.x == 0) {
+#pragma unroll
       for (int g = 0; g < G; ++g) {
         const int hq = hkv * G + g;
         float* ml = tmp_ml + (((int64_t)seq * Hq + hq) * max_parts + part) * 2;
@@ -64,15 +65,14 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
   }
   const int p_end = min(len, p_start + partition_size);
 
-  __shared__ float q_lds[MAX_G][128];
-  __shared__ float s_lds[MAX_G][CHUNK];
+  __shared__ float q_lds[G][DHEAD];
+  __shared__ float s_lds[G][CHUNK];
   __shared__ float head_m[MAX_G], head_l[MAX_G], head_corr[MAX_G];
 
-  // Load the G query vectors (pre-scaled) into LDS.
-  for (int idx = threadIdx.x; idx < G * D; idx += NTHREADS) {
-    const int g = idx / D, d = idx % D;
+  for (int idx = threadIdx.x; idx < G * DHEAD; idx += NTHREADS) {
+    const int g = idx / DHEAD, d = idx % DHEAD;
     q_lds[g][d] =
-        bf16_to_f32(q[((int64_t)seq * Hq + hkv * G + g) * D + d]) * scale;
+        bf16_to_f32(q[((int64_t)seq * Hq + hkv * G + g) * DHEAD + d]) * scale;
   }
   if (threadIdx.x < MAX_G) {
     head_m[threadIdx.x] = -INFINITY;
@@ -80,15 +80,13 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
   }
   __syncthreads();
 
-  // Phase-C ownership: thread owns dim pair (2*t, 2*t+1) within D, and
-  // loops token parities if D < 2*NTHREADS is not exact.
-  const int pairs = D / 2;                  // D in {64, 128}
-  const int d_own = (threadIdx.x % pairs) * 2;
-  const int par = threadIdx.x / pairs;      // 0..n_par-1
-  const int n_par = NTHREADS / pairs;
-  float acc[MAX_G][2];
+  constexpr int PAIRS = DHEAD / 2;
+  constexpr int N_PAR = NTHREADS / PAIRS;   // 1 for D=128, 2 for D=64
+  const int d_own = (threadIdx.x % PAIRS) * 2;
+  const int par = threadIdx.x / PAIRS;
+  float acc[G][2];
 #pragma unroll
-  for (int g = 0; g < MAX_G; ++g) acc[g][0] = acc[g][1] = 0.f;
+  for (int g = 0; g < G; ++g) acc[g][0] = acc[g][1] = 0.f;
 
   const int* btable = block_tables + (int64_t)seq * max_blocks;
   const int nwaves = NTHREADS / WAVE;
@@ -98,29 +96,37 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
   for (int base = p_start; base < p_end; base += CHUNK) {
     const int chunk_n = min(CHUNK, p_end - base);
 
-    // --- Phase A: scores ---------------------------------------------------
+    // --- Phase A: scores (batched K-row loads, then unrolled math) -------
     if ((int)threadIdx.x < chunk_n) {
       const int tok = base + threadIdx.x;
       const int64_t blk = btable[tok / block_size];
       const uint16_t* krow =
           k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                      tok % block_size)) * D;
-      float s[MAX_G];
+                      tok % block_size)) * DHEAD;
+      u16x8 kraw[DHEAD / 8];
 #pragma unroll
-      for (int g = 0; g < MAX_G; ++g) s[g] = 0.f;
-      for (int j = 0; j < D; j += 8) {
+      for (int j = 0; j < DHEAD / 8; ++j)
+        kraw[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+      float s[G];
+#pragma unroll
+      for (int g = 0; g < G; ++g) s[g] = 0.f;
+#pragma unroll
+      for (int j = 0; j < DHEAD / 8; ++j) {
         float kv[8];
-        load_bf16x8(krow + j, kv);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) kv[i] = bf16_to_f32(kraw[j][i]);
+#pragma unroll
         for (int g = 0; g < G; ++g) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i) s[g] += q_lds[g][j + i] * kv[i];
+          for (int i = 0; i < 8; ++i) s[g] += q_lds[g][j * 8 + i] * kv[i];
         }
       }
+#pragma unroll
       for (int g = 0; g < G; ++g) s_lds[g][threadIdx.x] = s[g];
     }
     __syncthreads();
 
-    // --- Phase B: online softmax per head (waves rotate over heads) -------
+    // --- Phase B: online softmax per head --------------------------------
     for (int g = wid; g < G; g += nwaves) {
       float m_chunk = -INFINITY;
       for (int i = lane; i < chunk_n; i += WAVE)
@@ -144,20 +150,48 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
     }
     __syncthreads();
 
-    // --- Phase C: V accumulation (b32 dim-pair loads) ----------------------
+    // --- Phase C: V accumulation (8-token load batches) -------------------
     {
+#pragma unroll
       for (int g = 0; g < G; ++g) {
         acc[g][0] *= head_corr[g];
         acc[g][1] *= head_corr[g];
       }
-      for (int tok_i = par; tok_i < chunk_n; tok_i += n_par) {
+      const int ntok = (chunk_n - par + N_PAR - 1) / N_PAR;  // my tokens
+      int t8 = 0;
+      for (; t8 + 8 <= ntok; t8 += 8) {
+        uint32_t vv[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int tok = base + (t8 + u) * N_PAR + par;
+          const int64_t blk = btable[tok / block_size];
+          vv[u] = *reinterpret_cast<const uint32_t*>(
+              v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
+                          tok % block_size)) * DHEAD + d_own);
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int tok_i = (t8 + u) * N_PAR + par;
+          const float v0 = bf16_to_f32((uint16_t)(vv[u] & 0xffff));
+          const float v1 = bf16_to_f32((uint16_t)(vv[u] >> 16));
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = s_lds[g][tok_i];
+            acc[g][0] += p * v0;
+            acc[g][1] += p * v1;
+          }
+        }
+      }
+      for (; t8 < ntok; ++t8) {
+        const int tok_i = t8 * N_PAR + par;
         const int tok = base + tok_i;
         const int64_t blk = btable[tok / block_size];
         const uint32_t vv = *reinterpret_cast<const uint32_t*>(
             v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                        tok % block_size)) * D + d_own);
+                        tok % block_size)) * DHEAD + d_own);
         const float v0 = bf16_to_f32((uint16_t)(vv & 0xffff));
         const float v1 = bf16_to_f32((uint16_t)(vv >> 16));
+#pragma unroll
         for (int g = 0; g < G; ++g) {
           const float p = s_lds[g][tok_i];
           acc[g][0] += p * v0;
@@ -165,26 +199,29 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
         }
       }
     }
-    __syncthreads();  // s_lds reused next chunk
+    __syncthreads();
   }
 
-  // Combine token parities (n_par == 1 for D=128; 2 for D=64) and write.
-  __shared__ float comb[MAX_G * NTHREADS * 2];
-  if (n_par > 1) {
+  // Combine token parities (N_PAR==2 only for D=64) and write out.
+  if (N_PAR > 1) {
+    __shared__ float comb[N_PAR > 1 ? G * NTHREADS * 2 : 1];
+    #pragma unroll
     for (int g = 0; g < G; ++g) {
-      comb[((par * MAX_G + g) * pairs + d_own / 2) * 2 + 0] = acc[g][0];
-      comb[((par * MAX_G + g) * pairs + d_own / 2) * 2 + 1] = acc[g][1];
+      comb[((par * G + g) * PAIRS + d_own / 2) * 2 + 0] = acc[g][0];
+      comb[((par * G + g) * PAIRS + d_own / 2) * 2 + 1] = acc[g][1];
     }
     __syncthreads();
     if (par == 0) {
+#pragma unroll
       for (int g = 0; g < G; ++g)
-        for (int p = 1; p < n_par; ++p) {
-          acc[g][0] += comb[((p * MAX_G + g) * pairs + d_own / 2) * 2 + 0];
-          acc[g][1] += comb[((p * MAX_G + g) * pairs + d_own / 2) * 2 + 1];
+        for (int p = 1; p < N_PAR; ++p) {
+          acc[g][0] += comb[((p * G + g) * PAIRS + d_own / 2) * 2 + 0];
+          acc[g][1] += comb[((p * G + g) * PAIRS + d_own / 2) * 2 + 1];
         }
     }
   }
   if (par == 0) {
+#pragma unroll
     for (int g = 0; g < G; ++g) {
       const int hq = hkv * G + g;
       if (nparts == 1) {
@@ -193,10 +230,10 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
             (uint32_t)f32_to_bf16(acc[g][0] * inv_l) |
             ((uint32_t)f32_to_bf16(acc[g][1] * inv_l) << 16);
         *reinterpret_cast<uint32_t*>(
-            out + ((int64_t)seq * Hq + hq) * D + d_own) = packed;
+            out + ((int64_t)seq * Hq + hq) * DHEAD + d_own) = packed;
       } else {
         float* tp =
-            tmp_out + (((int64_t)seq * Hq + hq) * max_parts + part) * D;
+            tmp_out + (((int64_t)seq * Hq + hq) * max_parts + part) * DHEAD;
         tp[d_own] = acc[g][0];
         tp[d_own + 1] = acc[g][1];
         if (d_own == 0) {
@@ -243,6 +280,18 @@ __global__ __launch_bounds__(128) void paged_attn_reduce_kernel(
   }
 }
 
+template <int DHEAD, int G>
+void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
+                   const uint16_t* q, const uint16_t* kc, const uint16_t* vc,
+                   const int* bt, const int* lens, float scale, int B,
+                   int Hkv, int block_size, int max_blocks, int eff_part,
+                   int nparts, int max_parts, hipStream_t stream) {
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G>),
+                     dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
+                     tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
+                     block_size, max_blocks, eff_part, max_parts);
+}
+
 }  // namespace
 
 void paged_attn_decode(torch::Tensor out, torch::Tensor q,
@@ -257,14 +306,14 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   const int block_size = k_cache.size(2);
   const int max_blocks = block_tables.size(1);
   const int max_parts = tmp_out.size(2);
+  const int G = Hq / Hkv;
   TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
-  TORCH_CHECK(Hq / Hkv <= MAX_G && Hq % Hkv == 0);
+  TORCH_CHECK(G <= MAX_G && Hq % Hkv == 0);
+  TORCH_CHECK(G == 1 || G == 2 || G == 4 || G == 8, "G must be 1/2/4/8");
   TORCH_CHECK(block_tables.scalar_type() == torch::kInt32);
   TORCH_CHECK(seq_lens.scalar_type() == torch::kInt32);
 
   const int max_len = (int)max_len_hint;
-  // Pick a partition count that fills the chip (~>=1024 workgroups) without
-  // exceeding the workspace.
   const int max_useful = cdiv(max_len, PART_QUANT);
   int nparts = std::max(1, 1024 / std::max(1, B * Hkv));
   nparts = std::min({nparts, max_useful, max_parts});
@@ -272,19 +321,34 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   nparts = cdiv(max_len, eff_part);
 
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(paged_attn_decode_kernel, dim3(B, Hkv, nparts),
-                     dim3(NTHREADS), 0, stream, (uint16_t*)out.data_ptr(),
-                     tmp_out.data_ptr<float>(), tmp_ml.data_ptr<float>(),
-                     (const uint16_t*)q.data_ptr(),
-                     (const uint16_t*)k_cache.data_ptr(),
-                     (const uint16_t*)v_cache.data_ptr(),
-                     block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                     (float)scale, Hq, Hkv, D, block_size, max_blocks,
-                     eff_part, max_parts);
+  auto* o = (uint16_t*)out.data_ptr();
+  auto* to = tmp_out.data_ptr<float>();
+  auto* tm = tmp_ml.data_ptr<float>();
+  auto* qp = (const uint16_t*)q.data_ptr();
+  auto* kp = (const uint16_t*)k_cache.data_ptr();
+  auto* vp = (const uint16_t*)v_cache.data_ptr();
+  auto* bp = block_tables.data_ptr<int>();
+  auto* lp = seq_lens.data_ptr<int>();
+
+#define DISPATCH(DH, GG)                                                     \
+  launch_decode<DH, GG>(o, to, tm, qp, kp, vp, bp, lp, (float)scale, B,     \
+                        Hkv, block_size, max_blocks, eff_part, nparts,      \
+                        max_parts, stream)
+  if (D == 128) {
+    if (G == 1) DISPATCH(128, 1);
+    else if (G == 2) DISPATCH(128, 2);
+    else if (G == 4) DISPATCH(128, 4);
+    else DISPATCH(128, 8);
+  } else {
+    if (G == 1) DISPATCH(64, 1);
+    else if (G == 2) DISPATCH(64, 2);
+    else if (G == 4) DISPATCH(64, 4);
+    else DISPATCH(64, 8);
+  }
+#undef DISPATCH
+
   if (nparts > 1) {
     hipLaunchKernelGGL(paged_attn_reduce_kernel, dim3(B, Hq), dim3(128), 0,
-                       stream, (uint16_t*)out.data_ptr(),
-                       tmp_out.data_ptr<float>(), tmp_ml.data_ptr<float>(),
-                       seq_lens.data_ptr<int>(), Hq, D, eff_part, max_parts);
+                       stream, o, to, tm, lp, Hq, D, eff_part, max_parts);
   }
 }
