@@ -1,0 +1,217 @@
+"""Agent runtime — the skill-based agentic loop (parity with the
+reference's api/pkg/agent Agent.Run, agent.go:374): iterate up to
+max_iterations, decideNextAction via LLM with skills-as-tools, execute
+tool calls, stream the final answer; step-info emitted per step
+(observability.go:20 StepInfoEmitter -> pubsub + step_info table).
+"""
+from __future__ import annotations
+
+import json
+import logging
+import time
+from typing import Any, AsyncIterator, Dict, List, Optional
+
+from helix_amd.agent.skills import (APISkill, CalculatorSkill,
+                                    KnowledgeSkill, MemorySkill, Skill,
+                                    WebSearchSkill)
+from helix_amd.server import pubsub as ps
+from helix_amd.server.types import AssistantConfig, new_id
+
+log = logging.getLogger("helix_amd.agent")
+
+
+class AgentRunner:
+    def __init__(self, cfg, store, providers, pubsub, rag=None):
+        self.cfg = cfg
+        self.store = store
+        self.providers = providers
+        self.pubsub = pubsub
+        self.rag = rag
+
+    # ------------------------------------------------------------------
+    def build_skills(self, assistant: AssistantConfig, owner: str,
+                     app_id: str = "") -> List[Skill]:
+        skills: List[Skill] = []
+        if assistant.calculator.get("enabled", bool(assistant.calculator)):
+            skills.append(CalculatorSkill())
+        if assistant.knowledge:
+            skills.append(KnowledgeSkill(
+                self.rag, self.store,
+                [k.name for k in assistant.knowledge], owner))
+        for api in assistant.apis:
+            try:
+                skills.append(APISkill(api))
+            except Exception as e:
+                log.warning("failed to build API skill %s: %s", api.name, e)
+        if assistant.web_search.get("enabled",
+                                    bool(assistant.web_search)):
+            import os
+            skills.append(WebSearchSkill(os.environ.get("SEARXNG_URL", "")))
+        if assistant.memory.get("enabled", bool(assistant.memory)):
+            skills.append(MemorySkill(self.store, owner, app_id))
+        return skills
+
+    def _model_for(self, assistant: AssistantConfig, slot: str) -> tuple:
+        """4-slot model selection (reference llm_client.go:14-19)."""
+        mc = getattr(assistant, slot, None)
+        if mc is not None and mc.model:
+            return (mc.provider or assistant.provider or
+                    self.cfg.inference.default_provider, mc.model)
+        return (assistant.provider or self.cfg.inference.default_provider,
+                assistant.model or self.cfg.inference.default_model)
+
+    def _emit_step(self, ctx: dict, step: dict):
+        sid = new_id("step")
+        doc = {"id": sid, "session_id": ctx.get("session_id", ""),
+               "interaction_id": ctx.get("interaction_id", ""),
+               "created": time.time(), **step}
+        self.store.put("step_info", sid, doc, owner=ctx.get("owner", ""),
+                       parent=ctx.get("session_id", ""))
+        return doc
+
+    async def _publish_step(self, ctx: dict, doc: dict):
+        owner = ctx.get("owner", "")
+        session_id = ctx.get("session_id", "")
+        if owner and session_id:
+            await self.pubsub.publish(ps.session_queue(owner, session_id),
+                                      {"type": "step_info", "step": doc})
+
+    # ------------------------------------------------------------------
+    async def _loop(self, assistant: AssistantConfig, req: dict, owner: str,
+                    ctx: dict) -> tuple:
+        """Run the agentic loop; returns (messages, final_req) where
+        final_req is what produces the user-facing answer."""
+        app_id = ctx.get("app_id", "")
+        skills = self.build_skills(assistant, owner, app_id)
+        tools = [s.to_tool() for s in skills]
+        by_name = {s.name: s for s in skills}
+
+        messages = [dict(m) for m in req.get("messages", [])]
+        if assistant.system_prompt and not any(
+                m.get("role") == "system" for m in messages):
+            messages.insert(0, {"role": "system",
+                                "content": assistant.system_prompt})
+        provider, model = self._model_for(assistant, "generation_model")
+        r_provider, r_model = self._model_for(assistant, "reasoning_model")
+        client = self.providers.get_client(r_provider, owner)
+        from helix_amd.server.providers import LoggingClient
+        client = LoggingClient(client, self.store)
+
+        max_iter = self.cfg.agent_max_iterations
+        if not skills:
+            # no-skill direct path (reference agent.go:290)
+            return messages, {"model": model, "provider": provider}
+
+        for it in range(max_iter):
+            decide_req = {
+                "model": r_model,
+                "messages": messages,
+                "tools": tools,
+                "tool_choice": "auto",
+                "temperature": (assistant.temperature
+                                if assistant.temperature is not None else 0.1),
+                "_ctx": {**ctx, "step": f"agent_decide_{it}"},
+            }
+            resp = await client.chat(decide_req)
+            choice = resp["choices"][0]
+            msg = choice.get("message", {})
+            tool_calls = msg.get("tool_calls") or []
+            if not tool_calls:
+                # model answered directly — done
+                messages.append({"role": "assistant",
+                                 "content": msg.get("content", "")})
+                return messages, None
+            messages.append({"role": "assistant",
+                             "content": msg.get("content") or "",
+                             "tool_calls": tool_calls})
+            for tc in tool_calls:
+                fn = tc.get("function", {})
+                name = fn.get("name", "")
+                try:
+                    args = json.loads(fn.get("arguments") or "{}")
+                except json.JSONDecodeError:
+                    args = {}
+                skill = by_name.get(name)
+                t0 = time.time()
+                if skill is None:
+                    result = f"unknown tool: {name}"
+                else:
+                    result = await skill.execute(args, ctx)
+                doc = self._emit_step(ctx, {
+                    "step": name, "arguments": args,
+                    "result": result[:2000],
+                    "duration_ms": int((time.time() - t0) * 1000),
+                    "iteration": it})
+                await self._publish_step(ctx, doc)
+                messages.append({"role": "tool",
+                                 "tool_call_id": tc.get("id", ""),
+                                 "name": name, "content": result})
+        # iteration cap reached: ask for a final summary answer
+        messages.append({
+            "role": "user",
+            "content": "Summarize your findings and answer the original "
+                       "question now, without calling more tools."})
+        return messages, {"model": model, "provider": provider}
+
+    # ------------------------------------------------------------------
+    async def run_blocking(self, assistant: AssistantConfig, req: dict,
+                           owner: str, ctx: dict) -> dict:
+        messages, final = await self._loop(assistant, req, owner, ctx)
+        if final is None and messages and messages[-1]["role"] == "assistant":
+            content = messages[-1].get("content", "")
+            return _completion_dict(req.get("model", ""), content)
+        provider, model = (final or {}).get("provider"), \
+            (final or {}).get("model")
+        client = self.providers.get_client(
+            provider or self.cfg.inference.default_provider, owner)
+        from helix_amd.server.providers import LoggingClient
+        resp = await LoggingClient(client, self.store).chat({
+            "model": model, "messages": messages,
+            "temperature": assistant.temperature,
+            "max_tokens": assistant.max_tokens,
+            "_ctx": {**ctx, "step": "agent_final"}})
+        return resp
+
+    async def run_stream(self, assistant: AssistantConfig, req: dict,
+                         owner: str, ctx: dict) -> AsyncIterator[dict]:
+        messages, final = await self._loop(assistant, req, owner, ctx)
+        if final is None and messages and messages[-1]["role"] == "assistant":
+            content = messages[-1].get("content", "")
+            base = _completion_dict(req.get("model", ""), content)
+            yield {
+                "id": base["id"], "object": "chat.completion.chunk",
+                "created": base["created"], "model": base["model"],
+                "choices": [{"index": 0, "delta": {
+                    "role": "assistant", "content": content},
+                    "finish_reason": None}]}
+            yield {
+                "id": base["id"], "object": "chat.completion.chunk",
+                "created": base["created"], "model": base["model"],
+                "choices": [{"index": 0, "delta": {},
+                             "finish_reason": "stop"}]}
+            return
+        provider, model = (final or {}).get("provider"), \
+            (final or {}).get("model")
+        client = self.providers.get_client(
+            provider or self.cfg.inference.default_provider, owner)
+        from helix_amd.server.providers import LoggingClient
+        async for chunk in LoggingClient(client, self.store).chat_stream({
+                "model": model, "messages": messages, "stream": True,
+                "temperature": assistant.temperature,
+                "max_tokens": assistant.max_tokens,
+                "_ctx": {**ctx, "step": "agent_final"}}):
+            yield chunk
+
+
+def _completion_dict(model: str, content: str) -> dict:
+    return {
+        "id": new_id("chatcmpl"),
+        "object": "chat.completion",
+        "created": int(time.time()),
+        "model": model,
+        "choices": [{"index": 0, "message": {
+            "role": "assistant", "content": content},
+            "finish_reason": "stop"}],
+        "usage": {"prompt_tokens": 0, "completion_tokens": 0,
+                  "total_tokens": 0},
+    }

@@ -66,19 +66,12 @@ def create_app(cfg: Optional[ServerConfig] = None,
         else:
             providers.register("helix", RouterClient(router))
 
-    # RAG + agents wired in by their modules (optional deps).
-    rag = None
-    try:
-        from helix_amd.rag.service import RAGService
-        rag = RAGService(cfg, store, providers)
-    except Exception as e:  # pragma: no cover
-        log.warning("RAG unavailable: %s", e)
-    agent_runner = None
-    try:
-        from helix_amd.agent.runner import AgentRunner
-        agent_runner = AgentRunner(cfg, store, providers, pubsub)
-    except Exception as e:  # pragma: no cover
-        log.warning("agent runtime unavailable: %s", e)
+    from helix_amd.agent.runner import AgentRunner
+    from helix_amd.rag.service import RAGService
+    from helix_amd.server.knowledge import KnowledgeReconciler
+    rag = RAGService(cfg, store, providers)
+    agent_runner = AgentRunner(cfg, store, providers, pubsub, rag=rag)
+    knowledge = KnowledgeReconciler(cfg, store, rag)
 
     controller = Controller(cfg, store, providers, pubsub, rag=rag,
                             agent_runner=agent_runner)
@@ -92,6 +85,17 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.controller = controller
     app.state.runner_service = runner_service
     app.state.rag = rag
+    app.state.knowledge = knowledge
+
+    @app.on_event("startup")
+    async def _start_reconciler():
+        app.state._reconciler_task = asyncio.create_task(knowledge.run())
+
+    @app.on_event("shutdown")
+    async def _stop_reconciler():
+        t = getattr(app.state, "_reconciler_task", None)
+        if t:
+            t.cancel()
 
     @app.exception_handler(ProviderError)
     async def _pe(request, exc: ProviderError):
@@ -288,6 +292,57 @@ def create_app(cfg: Optional[ServerConfig] = None,
                 store.delete("apps", app_id)
                 return {"ok": True}
         _bind()
+
+    # ------------------------------------------------------------------
+    # Knowledge API (reference server.go knowledge routes + reconciler)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/knowledge")
+    async def create_knowledge(request: Request,
+                               user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        doc = knowledge.create(user.id, body.get("name", ""),
+                               body.get("source", {}),
+                               body.get("app_id", ""))
+        return doc
+
+    @app.get("/api/v1/knowledge")
+    async def list_knowledge(user: AuthUser = Depends(auth_dep)):
+        return knowledge.list(user.id)
+
+    @app.get("/api/v1/knowledge/{kid}")
+    async def get_knowledge(kid: str, user: AuthUser = Depends(auth_dep)):
+        doc = knowledge.get(kid)
+        if doc is None or (doc["owner"] != user.id and not user.admin):
+            raise HTTPException(404, "knowledge not found")
+        return doc
+
+    @app.delete("/api/v1/knowledge/{kid}")
+    async def delete_knowledge(kid: str,
+                               user: AuthUser = Depends(auth_dep)):
+        doc = knowledge.get(kid)
+        if doc is None or (doc["owner"] != user.id and not user.admin):
+            raise HTTPException(404, "knowledge not found")
+        knowledge.delete(kid)
+        return {"ok": True}
+
+    @app.post("/api/v1/knowledge/{kid}/refresh")
+    async def refresh_knowledge(kid: str,
+                                user: AuthUser = Depends(auth_dep)):
+        doc = knowledge.get(kid)
+        if doc is None or (doc["owner"] != user.id and not user.admin):
+            raise HTTPException(404, "knowledge not found")
+        knowledge.request_refresh(kid)
+        return {"ok": True}
+
+    @app.post("/api/v1/knowledge/{kid}/query")
+    async def query_knowledge(kid: str, request: Request,
+                              user: AuthUser = Depends(auth_dep)):
+        doc = knowledge.get(kid)
+        if doc is None or (doc["owner"] != user.id and not user.admin):
+            raise HTTPException(404, "knowledge not found")
+        body = await request.json()
+        return await rag.query(kid, body.get("query", ""),
+                               body.get("k"))
 
     # ------------------------------------------------------------------
     # Runner plane (heartbeats -> router, reference sandbox_handlers.go:84)

@@ -5,8 +5,11 @@ Times each hot kernel at serving shapes with hip events and prints
 bandwidth/TFLOPS vs the roofline. Usage: python scripts/bench_kernels.py
 [decode|prefill|gemm|norm|sample|all]
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
