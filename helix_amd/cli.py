@@ -1,0 +1,252 @@
+"""helix-amd CLI (parity with the reference's `helix` cobra CLI,
+api/cmd/helix root.go:45-72: serve, apply, chat, knowledge, secret,
+session, model, runner, version, test).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import os
+import sys
+
+import typer
+
+app = typer.Typer(name="helix-amd", help="MI355X-native private GenAI stack")
+
+
+def _api(ctx_url: str = "") -> tuple[str, dict]:
+    url = ctx_url or os.environ.get("HELIX_URL", "http://localhost:8080")
+    key = os.environ.get("HELIX_API_KEY", "admin-key")
+    return url, {"Authorization": f"Bearer {key}"}
+
+
+@app.command()
+def serve(host: str = typer.Option(None), port: int = typer.Option(None),
+          local_runner: bool = typer.Option(False, "--local-runner"),
+          store_path: str = typer.Option(None)):
+    """Start the control plane (optionally with an in-process GPU runner)."""
+    import uvicorn
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import load_config
+    cfg = load_config()
+    if host:
+        cfg.web.host = host
+    if port:
+        cfg.web.port = port
+    if local_runner:
+        cfg.runner_plane.local_runner = True
+    if store_path:
+        cfg.store.path = store_path
+    api = create_app(cfg)
+    uvicorn.run(api, host=cfg.web.host, port=cfg.web.port, log_level="info")
+
+
+@app.command()
+def runner(api_url: str = typer.Option("http://localhost:8080"),
+           runner_id: str = typer.Option("runner-0"),
+           port: int = typer.Option(8090),
+           advertise: str = typer.Option(""),
+           device: str = typer.Option("cuda:0"),
+           preload: str = typer.Option("", help="comma-separated models")):
+    """Start a GPU runner: OpenAI-compatible server + heartbeat."""
+    import threading
+
+    import uvicorn
+    from helix_amd.runner.heartbeat import heartbeat_loop
+    from helix_amd.runner.http import create_runner_app
+    from helix_amd.runner.service import RunnerService
+    from helix_amd.server.config import load_config
+    cfg = load_config()
+    svc = RunnerService(device=device)
+    for m in filter(None, preload.split(",")):
+        svc.ensure_loaded(m.strip())
+    api = create_runner_app(svc, runner_id)
+    addr = advertise or f"http://{_local_ip()}:{port}"
+
+    def beat():
+        asyncio.run(heartbeat_loop(
+            api_url, cfg.runner_plane.runner_token, runner_id, addr, svc,
+            interval=cfg.runner_plane.heartbeat_interval_s))
+    threading.Thread(target=beat, daemon=True).start()
+    uvicorn.run(api, host="0.0.0.0", port=port, log_level="info")
+
+
+def _local_ip() -> str:
+    import socket
+    s = socket.socket(socket.AF_INET, socket.SOCK_DGRAM)
+    try:
+        s.connect(("10.255.255.255", 1))
+        return s.getsockname()[0]
+    except Exception:
+        return "127.0.0.1"
+    finally:
+        s.close()
+
+
+@app.command()
+def apply(file: str = typer.Option(..., "-f", "--file"),
+          url: str = typer.Option("", "--url"),
+          global_: bool = typer.Option(False, "--global")):
+    """Create/update an app from helix.yaml (reference `helix apply`)."""
+    import httpx
+    from helix_amd.server.apps import load_app_file
+    cfg = load_app_file(file)
+    api, headers = _api(url)
+    # update-if-exists by name
+    with httpx.Client(timeout=30) as http:
+        existing = http.get(f"{api}/api/v1/apps", headers=headers).json()
+        match = next((a for a in existing
+                      if a.get("config", {}).get("name") == cfg.name), None)
+        if match:
+            r = http.put(f"{api}/api/v1/apps/{match['id']}",
+                         json={"config": cfg.model_dump(by_alias=True),
+                               "global": global_}, headers=headers)
+        else:
+            r = http.post(f"{api}/api/v1/apps",
+                          json={"config": cfg.model_dump(by_alias=True),
+                                "global": global_}, headers=headers)
+        r.raise_for_status()
+        doc = r.json()
+        typer.echo(f"applied app {doc['id']} ({cfg.name})")
+
+
+@app.command()
+def chat(message: str = typer.Argument(...),
+         model: str = typer.Option("", "--model"),
+         app_id: str = typer.Option("", "--app"),
+         url: str = typer.Option("", "--url"),
+         stream: bool = typer.Option(True)):
+    """One-shot chat against /v1/chat/completions."""
+    import httpx
+    api, headers = _api(url)
+    body = {"messages": [{"role": "user", "content": message}],
+            "stream": stream}
+    if model:
+        body["model"] = model
+    if app_id:
+        body["app_id"] = app_id
+    with httpx.Client(timeout=300) as http:
+        if stream:
+            with http.stream("POST", f"{api}/v1/chat/completions",
+                             json=body, headers=headers) as r:
+                r.raise_for_status()
+                for line in r.iter_lines():
+                    if not line.startswith("data: ") or \
+                            line[6:].strip() == "[DONE]":
+                        continue
+                    chunk = json.loads(line[6:])
+                    if chunk.get("choices"):
+                        sys.stdout.write(
+                            chunk["choices"][0].get("delta", {}).get(
+                                "content") or "")
+                        sys.stdout.flush()
+                sys.stdout.write("\n")
+        else:
+            r = http.post(f"{api}/v1/chat/completions", json=body,
+                          headers=headers)
+            r.raise_for_status()
+            typer.echo(r.json()["choices"][0]["message"]["content"])
+
+
+knowledge_app = typer.Typer(help="Manage knowledge sources")
+app.add_typer(knowledge_app, name="knowledge")
+
+
+@knowledge_app.command("list")
+def knowledge_list(url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    rows = httpx.get(f"{api}/api/v1/knowledge", headers=headers).json()
+    for k in rows:
+        typer.echo(f"{k['id']}  {k['name']:<24} {k['state']:<10} "
+                   f"v{k.get('version', 0)} chunks={k.get('chunks', 0)}")
+
+
+@knowledge_app.command("refresh")
+def knowledge_refresh(kid: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.post(f"{api}/api/v1/knowledge/{kid}/refresh", headers=headers)
+    typer.echo(r.json())
+
+
+secret_app = typer.Typer(help="Manage secrets")
+app.add_typer(secret_app, name="secret")
+
+
+@secret_app.command("set")
+def secret_set(name: str, value: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.post(f"{api}/api/v1/secrets", json={"name": name,
+                                                  "value": value},
+                   headers=headers)
+    typer.echo(r.json())
+
+
+@secret_app.command("list")
+def secret_list(url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    for s in httpx.get(f"{api}/api/v1/secrets", headers=headers).json():
+        typer.echo(f"{s['name']}")
+
+
+session_app = typer.Typer(help="Sessions")
+app.add_typer(session_app, name="session")
+
+
+@session_app.command("list")
+def session_list(url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    for s in httpx.get(f"{api}/api/v1/sessions", headers=headers).json():
+        typer.echo(f"{s['id']}  {s['name'][:40]}")
+
+
+@app.command()
+def version():
+    from helix_amd import __version__
+    typer.echo(f"helix_amd {__version__}")
+
+
+@app.command()
+def test(file: str = typer.Option("helix.yaml", "-f", "--file"),
+         url: str = typer.Option("", "--url")):
+    """Run helix.yaml assistant tests (reference `helix test`): each test
+    step sends the prompt and an LLM judge checks the expected response."""
+    import httpx
+    from helix_amd.server.apps import load_app_file
+    cfg = load_app_file(file)
+    api, headers = _api(url)
+    failures = 0
+    with httpx.Client(timeout=300) as http:
+        for asst in cfg.assistants:
+            for t in asst.tests:
+                for step in t.steps:
+                    prompt = step.get("prompt", "")
+                    expected = step.get("expected_output", "")
+                    r = http.post(f"{api}/v1/chat/completions", json={
+                        "model": asst.model,
+                        "messages": [{"role": "user", "content": prompt}]},
+                        headers=headers)
+                    answer = r.json()["choices"][0]["message"]["content"]
+                    judge = http.post(f"{api}/v1/chat/completions", json={
+                        "model": asst.model,
+                        "messages": [{"role": "user", "content":
+                                      f"Does this answer satisfy the "
+                                      f"expectation?\nExpectation: "
+                                      f"{expected}\nAnswer: {answer}\n"
+                                      f"Reply YES or NO."}]},
+                        headers=headers)
+                    verdict = judge.json()["choices"][0]["message"][
+                        "content"].strip().upper()
+                    ok = verdict.startswith("YES")
+                    failures += 0 if ok else 1
+                    typer.echo(f"[{'PASS' if ok else 'FAIL'}] {t.name}: "
+                               f"{prompt[:40]}")
+    raise typer.Exit(1 if failures else 0)
+
+
+if __name__ == "__main__":
+    app()

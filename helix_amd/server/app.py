@@ -360,6 +360,32 @@ def create_app(cfg: Optional[ServerConfig] = None,
         return [r.model_dump() for r in router.runners()]
 
     # ------------------------------------------------------------------
+    # Secrets (reference api/pkg/server secrets routes)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/secrets")
+    async def set_secret(request: Request,
+                         user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        name = body.get("name", "")
+        if not name:
+            raise HTTPException(400, "name required")
+        sid = f"{user.id}:{name}"
+        store.put("secrets", sid, {"id": sid, "name": name,
+                                   "value": body.get("value", "")},
+                  owner=user.id)
+        return {"name": name, "ok": True}
+
+    @app.get("/api/v1/secrets")
+    async def list_secrets(user: AuthUser = Depends(auth_dep)):
+        return [{"name": s["name"]}
+                for s in store.list("secrets", owner=user.id)]
+
+    @app.delete("/api/v1/secrets/{name}")
+    async def delete_secret(name: str, user: AuthUser = Depends(auth_dep)):
+        ok = store.delete("secrets", f"{user.id}:{name}")
+        return {"ok": ok}
+
+    # ------------------------------------------------------------------
     # Users / keys / admin
     # ------------------------------------------------------------------
     @app.post("/api/v1/users")

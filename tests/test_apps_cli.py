@@ -1,0 +1,78 @@
+"""helix.yaml parsing + CLI-level behaviors."""
+import textwrap
+
+import pytest
+
+from helix_amd.server.apps import parse_helix_yaml
+
+
+def test_parse_plain_form(tmp_path):
+    y = textwrap.dedent("""
+        name: My App
+        description: demo
+        assistants:
+          - name: default
+            model: llama3-8b
+            provider: helix
+            system_prompt: Be helpful.
+            temperature: 0.5
+            knowledge:
+              - name: docs
+                source:
+                  text: "inline knowledge"
+            apis:
+              - name: petstore
+                url: https://petstore.example.com
+                schema: '{"paths": {}}'
+            tests:
+              - name: smoke
+                steps:
+                  - prompt: hello
+                    expected_output: a greeting
+    """)
+    cfg = parse_helix_yaml(y)
+    assert cfg.name == "My App"
+    a = cfg.assistants[0]
+    assert a.model == "llama3-8b"
+    assert a.temperature == 0.5
+    assert a.knowledge[0].name == "docs"
+    assert a.apis[0].name == "petstore"
+    assert a.tests[0].steps[0]["prompt"] == "hello"
+
+
+def test_parse_crd_form():
+    y = textwrap.dedent("""
+        apiVersion: app.aispec.org/v1alpha1
+        kind: AIApp
+        metadata:
+          name: crd-app
+        spec:
+          description: from crd
+          assistants:
+            - name: a
+              model: m
+    """)
+    cfg = parse_helix_yaml(y)
+    assert cfg.name == "crd-app"
+    assert cfg.assistants[0].model == "m"
+
+
+def test_file_ref_inlining(tmp_path):
+    (tmp_path / "prompt.txt").write_text("You are from a file.")
+    y = textwrap.dedent("""
+        name: f
+        assistants:
+          - name: a
+            model: m
+            system_prompt: file://prompt.txt
+    """)
+    cfg = parse_helix_yaml(y, base_dir=str(tmp_path))
+    assert cfg.assistants[0].system_prompt == "You are from a file."
+
+
+def test_cli_version():
+    from typer.testing import CliRunner
+    from helix_amd.cli import app
+    res = CliRunner().invoke(app, ["version"])
+    assert res.exit_code == 0
+    assert "helix_amd" in res.output
