@@ -42,14 +42,14 @@ __device__ __forceinline__ int v_swz(int dim, int byte_in_row) {
   return dim * (KTILE * 2) + (byte_in_row ^ ((dim & 7) << 4));
 }
 
-template <int CAUSAL>
-__global__ __launch_bounds__(256, 3) void attn_prefill_kernel(
+template <int CAUSAL, int D>
+__global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
     uint16_t* __restrict__ out,        // [T, Hq, D]
     const uint16_t* __restrict__ q,    // [T, Hq, D]
     const uint16_t* __restrict__ k,    // [T, Hkv, D]
     const uint16_t* __restrict__ v,    // [T, Hkv, D]
     const int* __restrict__ cu_seqlens,  // [B+1]
-    float scale, int Hq, int Hkv, int D) {
+    float scale, int Hq, int Hkv) {
   const int qtile = blockIdx.x;
   const int seq = blockIdx.y;
   const int hq = blockIdx.z;
@@ -71,11 +71,11 @@ __global__ __launch_bounds__(256, 3) void attn_prefill_kernel(
 
   // ---- Q fragments (A-operand layout), 16 rows per wave ----
   const int my_qrow = qbase + wid * 16 + lane_lo;
-  const int nkt = D / 32;
-  bf16x8 q_frag[DMAX / 32];
+  constexpr int nkt = D / 32;
+  bf16x8 q_frag[nkt];
 #pragma unroll
-  for (int kt = 0; kt < DMAX / 32; ++kt) {
-    if (kt < nkt && my_qrow < len) {
+  for (int kt = 0; kt < nkt; ++kt) {
+    if (my_qrow < len) {
       const uint16_t* src =
           q + ((int64_t)(seq_start + my_qrow) * Hq + hq) * D + kt * 32 +
           lane_hi * 8;
@@ -92,22 +92,21 @@ __global__ __launch_bounds__(256, 3) void attn_prefill_kernel(
     m_run[r] = -INFINITY;
     l_run[r] = 0.f;
   }
-  floatx4 o_acc[DMAX / 16];
+  constexpr int nc = D / 16;
+  floatx4 o_acc[nc];
 #pragma unroll
-  for (int c = 0; c < DMAX / 16; ++c) o_acc[c] = floatx4{0, 0, 0, 0};
-  const int nc = D / 16;
+  for (int c = 0; c < nc; ++c) o_acc[c] = floatx4{0, 0, 0, 0};
 
   const int kv_max = CAUSAL ? min(len, qbase + QTILE) : len;
   const int ntiles = (kv_max + KTILE - 1) / KTILE;
 
   // Per-thread staging assignment: KTILE*D/8 16B-chunks over 256 threads.
-  const int nchunk = KTILE * D / 8 / 256;     // 4 for D=128, 2 for D=64
-  u16x8 k_reg[4], v_reg[4];
+  constexpr int nchunk = KTILE * D / 8 / 256;  // 4 for D=128, 2 for D=64
+  u16x8 k_reg[nchunk], v_reg[nchunk];
 
   auto load_tile = [&](int t) {
 #pragma unroll
-    for (int cch = 0; cch < 4; ++cch) {
-      if (cch >= nchunk) break;
+    for (int cch = 0; cch < nchunk; ++cch) {
       const int e = (cch * 256 + tid) * 8;
       const int tok = e / D;
       const int d0 = e % D;
@@ -125,8 +124,7 @@ __global__ __launch_bounds__(256, 3) void attn_prefill_kernel(
   };
   auto store_tile = [&]() {
 #pragma unroll
-    for (int cch = 0; cch < 4; ++cch) {
-      if (cch >= nchunk) break;
+    for (int cch = 0; cch < nchunk; ++cch) {
       const int e = (cch * 256 + tid) * 8;
       const int tok = e / D;
       const int d0 = e % D;
@@ -155,6 +153,7 @@ __global__ __launch_bounds__(256, 3) void attn_prefill_kernel(
     for (int h = 0; h < KTILE / 16; ++h) {
       s_frag[h] = floatx4{0, 0, 0, 0};
       const int tok = h * 16 + lane_lo;
+#pragma unroll
       for (int kt = 0; kt < nkt; ++kt) {
         const int byte = (kt * 32 + lane_hi * 8) * 2;
         u16x8 raw =
@@ -214,6 +213,7 @@ __global__ __launch_bounds__(256, 3) void attn_prefill_kernel(
           &p_lds[wid][lane_lo * P_PITCH + kk * 32 + lane_hi * 8]);
       p_frag[kk] = *reinterpret_cast<bf16x8*>(&raw);
     }
+#pragma unroll
     for (int c = 0; c < nc; ++c) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) o_acc[c][r] *= corr[r];
@@ -231,6 +231,7 @@ __global__ __launch_bounds__(256, 3) void attn_prefill_kernel(
   }
 
   // ---- Epilogue ----
+#pragma unroll
   for (int c = 0; c < nc; ++c) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -254,21 +255,20 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
   const int B = cu_seqlens.size(0) - 1;
   TORCH_CHECK(D % 32 == 0 && D <= DMAX, "head_dim must be mult of 32, <=128");
   TORCH_CHECK(cu_seqlens.scalar_type() == torch::kInt32);
+  TORCH_CHECK(D == 64 || D == 128, "head_dim must be 64 or 128");
   auto stream = at::hip::getCurrentHIPStream();
   const int n_qtiles = cdiv((int)max_seqlen, QTILE);
+#define LAUNCH(C, DH)                                                        \
+  hipLaunchKernelGGL((attn_prefill_kernel<C, DH>), dim3(n_qtiles, B, Hq),    \
+                     dim3(256), 0, stream, (uint16_t*)out.data_ptr(),        \
+                     (const uint16_t*)q.data_ptr(),                          \
+                     (const uint16_t*)k.data_ptr(),                          \
+                     (const uint16_t*)v.data_ptr(),                          \
+                     cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv)
   if (causal) {
-    hipLaunchKernelGGL((attn_prefill_kernel<1>), dim3(n_qtiles, B, Hq),
-                       dim3(256), 0, stream, (uint16_t*)out.data_ptr(),
-                       (const uint16_t*)q.data_ptr(),
-                       (const uint16_t*)k.data_ptr(),
-                       (const uint16_t*)v.data_ptr(),
-                       cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv, D);
+    if (D == 128) LAUNCH(1, 128); else LAUNCH(1, 64);
   } else {
-    hipLaunchKernelGGL((attn_prefill_kernel<0>), dim3(n_qtiles, B, Hq),
-                       dim3(256), 0, stream, (uint16_t*)out.data_ptr(),
-                       (const uint16_t*)q.data_ptr(),
-                       (const uint16_t*)k.data_ptr(),
-                       (const uint16_t*)v.data_ptr(),
-                       cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv, D);
+    if (D == 128) LAUNCH(0, 128); else LAUNCH(0, 64);
   }
+#undef LAUNCH
 }
