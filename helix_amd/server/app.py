@@ -68,13 +68,21 @@ def create_app(cfg: Optional[ServerConfig] = None,
 
     from helix_amd.agent.runner import AgentRunner
     from helix_amd.rag.service import RAGService
+    from helix_amd.server.filestore import FileStore
     from helix_amd.server.knowledge import KnowledgeReconciler
+    from helix_amd.server.models_catalog import ModelCatalog
+    from helix_amd.server.triggers import TriggerManager
+    from helix_amd.server.usage import QuotaExceededError, UsageService
     rag = RAGService(cfg, store, providers)
     agent_runner = AgentRunner(cfg, store, providers, pubsub, rag=rag)
     knowledge = KnowledgeReconciler(cfg, store, rag)
+    catalog = ModelCatalog(store)
+    usage = UsageService(store, catalog)
+    filestore = FileStore(cfg.filestore.path)
 
     controller = Controller(cfg, store, providers, pubsub, rag=rag,
-                            agent_runner=agent_runner)
+                            agent_runner=agent_runner, usage=usage)
+    triggers = TriggerManager(store, controller)
 
     app.state.cfg = cfg
     app.state.store = store
@@ -86,21 +94,32 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.runner_service = runner_service
     app.state.rag = rag
     app.state.knowledge = knowledge
+    app.state.catalog = catalog
+    app.state.usage = usage
+    app.state.filestore = filestore
+    app.state.triggers = triggers
 
     @app.on_event("startup")
     async def _start_reconciler():
         app.state._reconciler_task = asyncio.create_task(knowledge.run())
+        app.state._trigger_task = asyncio.create_task(triggers.run())
 
     @app.on_event("shutdown")
     async def _stop_reconciler():
-        t = getattr(app.state, "_reconciler_task", None)
-        if t:
-            t.cancel()
+        for attr in ("_reconciler_task", "_trigger_task"):
+            t = getattr(app.state, attr, None)
+            if t:
+                t.cancel()
 
     @app.exception_handler(ProviderError)
     async def _pe(request, exc: ProviderError):
         return JSONResponse({"error": {"message": str(exc)}},
                             status_code=exc.status)
+
+    @app.exception_handler(PermissionError)
+    async def _perm(request, exc):
+        return JSONResponse({"error": {"message": str(exc)}},
+                            status_code=400)
 
     @app.exception_handler(NoRunnerError)
     async def _nr(request, exc: NoRunnerError):
@@ -358,6 +377,134 @@ def create_app(cfg: Optional[ServerConfig] = None,
     async def list_runners(user: AuthUser = Depends(admin_dep)):
         router.reap_offline()
         return [r.model_dump() for r in router.runners()]
+
+    # ------------------------------------------------------------------
+    # Anthropic-compatible surface (reference api/pkg/anthropic proxy)
+    # ------------------------------------------------------------------
+    @app.post("/v1/messages")
+    async def anthropic_messages(request: Request,
+                                 user: AuthUser = Depends(auth_dep)):
+        from helix_amd.server.anthropic_api import (anthropic_to_openai,
+                                                    openai_to_anthropic,
+                                                    stream_anthropic_events)
+        areq = await request.json()
+        oreq = anthropic_to_openai(areq)
+        ctx = {"owner": user.id}
+        if oreq.get("stream"):
+            chunks = controller.chat_completion_stream(oreq, user.id,
+                                                       ctx=ctx)
+
+            async def sse():
+                async for ev in stream_anthropic_events(
+                        chunks, oreq.get("model", "")):
+                    yield ev
+            return StreamingResponse(sse(), media_type="text/event-stream")
+        resp = await controller.chat_completion(oreq, user.id, ctx=ctx)
+        return openai_to_anthropic(resp)
+
+    # ------------------------------------------------------------------
+    # Model catalog (reference helix-models + model-info routes)
+    # ------------------------------------------------------------------
+    @app.get("/api/v1/helix-models")
+    async def helix_models(user: AuthUser = Depends(auth_dep)):
+        return catalog.list()
+
+    @app.get("/api/v1/model-info/{model}")
+    async def model_info(model: str, user: AuthUser = Depends(auth_dep)):
+        info = catalog.get(model)
+        if info is None:
+            raise HTTPException(404, "unknown model")
+        return info
+
+    @app.put("/api/v1/model-info/{model}")
+    async def set_model_info(model: str, request: Request,
+                             user: AuthUser = Depends(admin_dep)):
+        catalog.set_override(model, await request.json())
+        return catalog.get(model)
+
+    # ------------------------------------------------------------------
+    # Usage / wallet (reference usage_metrics + wallets)
+    # ------------------------------------------------------------------
+    @app.get("/api/v1/usage")
+    async def my_usage(user: AuthUser = Depends(auth_dep), day: str = ""):
+        return usage.usage_for(user.id, day or None)
+
+    @app.get("/api/v1/wallet")
+    async def my_wallet(user: AuthUser = Depends(auth_dep)):
+        return usage.wallet(user.id)
+
+    @app.post("/api/v1/wallet/topup")
+    async def wallet_topup(request: Request,
+                           user: AuthUser = Depends(admin_dep)):
+        body = await request.json()
+        return usage.topup(body.get("owner", user.id),
+                           float(body.get("amount_usd", 0)))
+
+    # ------------------------------------------------------------------
+    # Triggers (reference api/pkg/trigger: cron + webhook)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/triggers")
+    async def create_trigger(request: Request,
+                             user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        try:
+            return triggers.create(user.id, body.get("app_id", ""),
+                                   body.get("kind", "cron"),
+                                   body.get("config", {}))
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+
+    @app.get("/api/v1/triggers")
+    async def list_triggers(user: AuthUser = Depends(auth_dep)):
+        return triggers.list(user.id)
+
+    @app.delete("/api/v1/triggers/{tid}")
+    async def delete_trigger(tid: str, user: AuthUser = Depends(auth_dep)):
+        doc = store.get("triggers", tid)
+        if doc is None or (doc["owner"] != user.id and not user.admin):
+            raise HTTPException(404, "trigger not found")
+        triggers.delete(tid)
+        return {"ok": True}
+
+    @app.post("/api/v1/webhooks/{tid}")
+    async def fire_webhook(tid: str, request: Request):
+        doc = store.get("triggers", tid)
+        if doc is None or doc.get("kind") != "webhook":
+            raise HTTPException(404, "webhook not found")
+        try:
+            payload = await request.json()
+        except Exception:
+            payload = {}
+        return await triggers.fire(doc, payload)
+
+    # ------------------------------------------------------------------
+    # Filestore (reference api/pkg/filestore routes)
+    # ------------------------------------------------------------------
+    @app.get("/api/v1/filestore/list")
+    async def fs_list(user: AuthUser = Depends(auth_dep), path: str = ""):
+        return filestore.list(user.id, path)
+
+    @app.put("/api/v1/filestore/upload")
+    async def fs_upload(request: Request,
+                        user: AuthUser = Depends(auth_dep), path: str = ""):
+        if not path:
+            raise HTTPException(400, "path query param required")
+        data = await request.body()
+        return filestore.write(user.id, path, data)
+
+    @app.get("/api/v1/filestore/download")
+    async def fs_download(user: AuthUser = Depends(auth_dep),
+                          path: str = ""):
+        from fastapi.responses import Response
+        try:
+            return Response(filestore.read(user.id, path),
+                            media_type="application/octet-stream")
+        except FileNotFoundError:
+            raise HTTPException(404, "not found")
+
+    @app.delete("/api/v1/filestore")
+    async def fs_delete(user: AuthUser = Depends(auth_dep), path: str = ""):
+        return {"ok": filestore.delete(user.id, path)}
 
     # ------------------------------------------------------------------
     # Secrets (reference api/pkg/server secrets routes)

@@ -20,13 +20,19 @@ log = logging.getLogger("helix_amd.controller")
 
 class Controller:
     def __init__(self, cfg, store, providers: ProviderManager,
-                 pubsub: ps.PubSub, rag=None, agent_runner=None):
+                 pubsub: ps.PubSub, rag=None, agent_runner=None, usage=None):
         self.cfg = cfg
         self.store = store
         self.providers = providers
         self.pubsub = pubsub
         self.rag = rag
         self.agent_runner = agent_runner
+        self.usage = usage
+
+    def _logging(self, client):
+        from helix_amd.server.providers import LoggingClient
+        logger = self.usage.log_call if self.usage is not None else None
+        return LoggingClient(client, self.store, usage_logger=logger)
 
     # ------------------------------------------------------------------
     # Assistant / app resolution (reference inference.go:81 loadAssistant)
@@ -145,8 +151,7 @@ class Controller:
         provider, model = self._resolve(req, owner, assistant)
         client = self.providers.get_client(provider, owner)
         call_req = {**req, "model": model, "_ctx": ctx or {"owner": owner}}
-        from helix_amd.server.providers import LoggingClient
-        return await LoggingClient(client, self.store).chat(call_req)
+        return await self._logging(client).chat(call_req)
 
     async def chat_completion_stream(self, req: dict, owner: str,
                                      app_id: str = "", assistant_id: str = "",
@@ -169,9 +174,7 @@ class Controller:
         client = self.providers.get_client(provider, owner)
         call_req = {**req, "model": model, "stream": True,
                     "_ctx": ctx or {"owner": owner}}
-        from helix_amd.server.providers import LoggingClient
-        async for chunk in LoggingClient(client, self.store).chat_stream(
-                call_req):
+        async for chunk in self._logging(client).chat_stream(call_req):
             yield chunk
 
     # ------------------------------------------------------------------

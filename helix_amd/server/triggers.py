@@ -1,0 +1,124 @@
+"""Trigger manager (parity with api/pkg/trigger: cron triggers via a
+5-field cron parser, webhook triggers; each fires an agent session turn).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+from typing import List, Optional
+
+from helix_amd.server.types import new_id
+
+log = logging.getLogger("helix_amd.triggers")
+
+
+def _parse_field(field: str, lo: int, hi: int) -> set:
+    vals = set()
+    for part in field.split(","):
+        step = 1
+        if "/" in part:
+            part, step_s = part.split("/")
+            step = int(step_s)
+        if part in ("*", ""):
+            rng = range(lo, hi + 1)
+        elif "-" in part:
+            a, b = part.split("-")
+            rng = range(int(a), int(b) + 1)
+        else:
+            rng = range(int(part), int(part) + 1)
+        vals.update(v for v in rng if (v - lo) % step == 0)
+    return vals
+
+
+class CronSchedule:
+    """Standard 5-field cron: minute hour dom month dow."""
+
+    def __init__(self, expr: str):
+        f = expr.split()
+        if len(f) != 5:
+            raise ValueError(f"bad cron expression: {expr!r}")
+        self.minute = _parse_field(f[0], 0, 59)
+        self.hour = _parse_field(f[1], 0, 23)
+        self.dom = _parse_field(f[2], 1, 31)
+        self.month = _parse_field(f[3], 1, 12)
+        self.dow = _parse_field(f[4], 0, 6)
+
+    def matches(self, t: time.struct_time) -> bool:
+        # python tm_wday: Mon=0..Sun=6; cron dow: Sun=0..Sat=6
+        return (t.tm_min in self.minute and t.tm_hour in self.hour and
+                t.tm_mday in self.dom and t.tm_mon in self.month and
+                ((t.tm_wday + 1) % 7) in self.dow)
+
+
+class TriggerManager:
+    def __init__(self, store, controller):
+        self.store = store
+        self.controller = controller
+        self._last_minute = -1
+
+    def create(self, owner: str, app_id: str, kind: str, config: dict) -> dict:
+        tid = new_id("trg")
+        if kind == "cron":
+            CronSchedule(config.get("schedule", ""))  # validate
+        doc = {"id": tid, "owner": owner, "app_id": app_id, "kind": kind,
+               "config": config, "enabled": True, "last_fired": 0,
+               "fire_count": 0}
+        self.store.put("triggers", tid, doc, owner=owner, parent=app_id)
+        return doc
+
+    def list(self, owner: str) -> List[dict]:
+        return self.store.list("triggers", owner=owner)
+
+    def delete(self, tid: str) -> bool:
+        return self.store.delete("triggers", tid)
+
+    async def fire(self, doc: dict, payload: Optional[dict] = None) -> dict:
+        """Run the trigger's prompt through a session turn on its app."""
+        prompt = doc.get("config", {}).get("prompt", "Run the scheduled task.")
+        if payload:
+            prompt = f"{prompt}\n\nWebhook payload: {payload}"
+        session = self.controller.create_session(
+            doc["owner"], app_id=doc.get("app_id", ""),
+            name=f"trigger {doc['id'][:12]}")
+        interaction = self.controller.add_interaction(session, prompt)
+        async for _ in self.controller.run_session_turn(session, interaction,
+                                                        stream_to_pubsub=True):
+            pass
+        doc["last_fired"] = time.time()
+        doc["fire_count"] = doc.get("fire_count", 0) + 1
+        self.store.put("triggers", doc["id"], doc, owner=doc["owner"],
+                       parent=doc.get("app_id", ""))
+        return {"session_id": session.id}
+
+    async def tick(self, now: Optional[float] = None) -> int:
+        """Fire cron triggers whose schedule matches the current minute."""
+        now = now or time.time()
+        t = time.localtime(now)
+        minute_key = t.tm_min + 60 * (t.tm_hour + 24 * t.tm_yday)
+        if minute_key == self._last_minute:
+            return 0
+        self._last_minute = minute_key
+        fired = 0
+        for doc in self.store.list("triggers", limit=10000):
+            if doc.get("kind") != "cron" or not doc.get("enabled"):
+                continue
+            try:
+                sched = CronSchedule(doc["config"].get("schedule", ""))
+            except ValueError:
+                continue
+            if sched.matches(t):
+                try:
+                    await self.fire(doc)
+                    fired += 1
+                except Exception:
+                    log.exception("trigger %s failed", doc["id"])
+        return fired
+
+    async def run(self, interval: float = 20.0):
+        while True:
+            try:
+                await self.tick()
+            except Exception:
+                log.exception("trigger loop error")
+            await asyncio.sleep(interval)

@@ -1,0 +1,194 @@
+"""Platform services: catalog, usage/wallet, triggers, filestore,
+Anthropic surface, weight IO."""
+import asyncio
+import time
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from helix_amd.server.app import create_app
+from helix_amd.server.config import ServerConfig
+from helix_amd.server.providers import MockClient, ProviderManager
+from helix_amd.server.triggers import CronSchedule
+from helix_amd.store import Store
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    cfg.filestore.path = str(tmp_path / "fs")
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    mock = MockClient()
+    pm.register("mock", mock)
+    app = create_app(cfg, store=store, providers=pm)
+    client = TestClient(app)
+    r = client.post("/api/v1/users", json={"username": "alice"},
+                    headers={"Authorization": "Bearer admin-key"})
+    key = r.json()["api_key"]
+    return app, client, mock, key, store
+
+
+def H(key):
+    return {"Authorization": f"Bearer {key}"}
+
+
+def test_model_catalog(stack):
+    _, client, _, key, _ = stack
+    r = client.get("/api/v1/helix-models", headers=H(key))
+    ids = [m["id"] for m in r.json()]
+    assert "llama3-8b" in ids and "bge-base" in ids
+    r = client.get("/api/v1/model-info/llama3-8b", headers=H(key))
+    assert r.json()["context_length"] == 8192
+    # admin override
+    r = client.put("/api/v1/model-info/llama3-8b",
+                   json={"context_length": 16384},
+                   headers=H("admin-key"))
+    assert r.json()["context_length"] == 16384
+
+
+def test_usage_metering_and_wallet(stack):
+    app, client, _, key, store = stack
+    uid = store.list("users")[0]["id"]
+    client.post("/api/v1/wallet/topup", json={"owner": uid,
+                                              "amount_usd": 10.0},
+                headers=H("admin-key"))
+    # mock-model has no price -> usage rows but no debit
+    client.post("/v1/chat/completions", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "hi"}]}, headers=H(key))
+    r = client.get("/api/v1/usage", headers=H(key))
+    rows = r.json()
+    assert rows and rows[0]["prompt_tokens"] == 7
+    assert rows[0]["calls"] == 1
+    w = client.get("/api/v1/wallet", headers=H(key)).json()
+    assert w["balance_usd"] == 10.0
+
+
+def test_cron_schedule():
+    s = CronSchedule("*/15 9-17 * * 1-5")
+    t = time.struct_time((2026, 9, 7, 10, 30, 0, 0, 250, -1))  # Monday
+    assert s.matches(t)
+    t2 = time.struct_time((2026, 9, 6, 10, 30, 0, 6, 249, -1))  # Sunday
+    assert not s.matches(t2)
+    assert not s.matches(time.struct_time((2026, 9, 7, 10, 7, 0, 0, 250, -1)))
+    with pytest.raises(ValueError):
+        CronSchedule("* * *")
+
+
+def test_trigger_fire_and_webhook(stack):
+    app, client, _, key, store = stack
+    r = client.post("/api/v1/triggers", json={
+        "kind": "webhook",
+        "config": {"prompt": "handle the webhook"}}, headers=H(key))
+    tid = r.json()["id"]
+    r = client.post(f"/api/v1/webhooks/{tid}", json={"event": "push"})
+    assert r.status_code == 200
+    sid = r.json()["session_id"]
+    its = store.list("interactions", parent=sid)
+    assert its and its[0]["response_message"] == "mock response"
+    assert "push" in its[0]["prompt_message"]
+
+
+def test_trigger_cron_validation(stack):
+    _, client, _, key, _ = stack
+    r = client.post("/api/v1/triggers", json={
+        "kind": "cron", "config": {"schedule": "bogus"}}, headers=H(key))
+    assert r.status_code == 400
+    r = client.post("/api/v1/triggers", json={
+        "kind": "cron", "config": {"schedule": "0 9 * * *",
+                                   "prompt": "daily"}}, headers=H(key))
+    assert r.status_code == 200
+
+
+def test_filestore_roundtrip(stack):
+    _, client, _, key, _ = stack
+    r = client.put("/api/v1/filestore/upload?path=docs/a.txt",
+                   content=b"hello file", headers=H(key))
+    assert r.status_code == 200
+    assert r.json()["size"] == 10
+    r = client.get("/api/v1/filestore/list?path=docs", headers=H(key))
+    assert r.json()[0]["path"].endswith("a.txt")
+    r = client.get("/api/v1/filestore/download?path=docs/a.txt",
+                   headers=H(key))
+    assert r.content == b"hello file"
+    r = client.request("DELETE", "/api/v1/filestore?path=docs/a.txt",
+                       headers=H(key))
+    assert r.json()["ok"]
+
+
+def test_filestore_path_escape(stack):
+    _, client, _, key, _ = stack
+    r = client.put("/api/v1/filestore/upload?path=../../etc/passwd",
+                   content=b"x", headers=H(key))
+    assert r.status_code >= 400
+
+
+def test_anthropic_messages(stack):
+    _, client, _, key, _ = stack
+    r = client.post("/v1/messages", json={
+        "model": "mock-model", "max_tokens": 32,
+        "system": "be terse",
+        "messages": [{"role": "user", "content": "hello"}]}, headers=H(key))
+    assert r.status_code == 200
+    body = r.json()
+    assert body["type"] == "message"
+    assert body["content"][0]["text"] == "mock response"
+    assert body["stop_reason"] == "end_turn"
+    assert body["usage"]["input_tokens"] == 7
+
+
+def test_anthropic_streaming(stack):
+    _, client, _, key, _ = stack
+    with client.stream("POST", "/v1/messages", json={
+        "model": "mock-model", "max_tokens": 32, "stream": True,
+        "messages": [{"role": "user", "content": "hello"}]},
+            headers=H(key)) as r:
+        text = "".join(r.iter_text())
+    assert "message_start" in text
+    assert "content_block_delta" in text
+    assert "message_stop" in text
+    assert "mock response".split()[0] in text
+
+
+def test_weight_io_roundtrip(tmp_path):
+    from helix_amd.engine.weights import load_llama_weights, save_sharded
+    from helix_amd.models.llama import PRESETS, LlamaForCausalLM
+    torch.manual_seed(0)
+    m1 = LlamaForCausalLM(PRESETS["tiny"]).float()
+    m1.init_random(1)
+    save_sharded(m1, str(tmp_path / "ckpt"), shard_bytes=1 << 20)
+    m2 = LlamaForCausalLM(PRESETS["tiny"]).float()
+    m2.init_random(2)
+    load_llama_weights(m2, str(tmp_path / "ckpt"))
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(),
+                                  m2.named_parameters()):
+        assert torch.equal(p1, p2), n1
+
+
+def test_weight_io_hf_names(tmp_path):
+    """HF-style checkpoint names map onto fused projections."""
+    from safetensors.torch import save_file
+    from helix_amd.engine.weights import load_llama_weights
+    from helix_amd.models.llama import PRESETS, LlamaForCausalLM
+    torch.manual_seed(3)
+    cfg = PRESETS["tiny"]
+    m = LlamaForCausalLM(cfg).float()
+    m.init_random(3)
+    q, kv, h = cfg.q_size, cfg.kv_size, cfg.hidden_size
+    sd = {}
+    qw = torch.randn(q, h)
+    kw = torch.randn(kv, h)
+    vw = torch.randn(kv, h)
+    sd["model.layers.0.self_attn.q_proj.weight"] = qw
+    sd["model.layers.0.self_attn.k_proj.weight"] = kw
+    sd["model.layers.0.self_attn.v_proj.weight"] = vw
+    save_file(sd, str(tmp_path / "hf.safetensors"))
+    load_llama_weights(m, str(tmp_path))
+    fused = m.layers[0].attn.qkv_proj.weight.data
+    assert torch.equal(fused[:q], qw)
+    assert torch.equal(fused[q:q + kv], kw)
+    assert torch.equal(fused[q + kv:], vw)
