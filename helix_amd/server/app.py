@@ -80,9 +80,16 @@ def create_app(cfg: Optional[ServerConfig] = None,
     usage = UsageService(store, catalog)
     filestore = FileStore(cfg.filestore.path)
 
+    from helix_amd.server.oauth import OAuthManager
+    from helix_amd.server.rbac import RBACService
+    from helix_amd.server.runner_profiles import (ProfileService,
+                                                  RunnerProfile)
     controller = Controller(cfg, store, providers, pubsub, rag=rag,
                             agent_runner=agent_runner, usage=usage)
     triggers = TriggerManager(store, controller)
+    rbac = RBACService(store)
+    oauth = OAuthManager(store)
+    profiles = ProfileService(store)
 
     app.state.cfg = cfg
     app.state.store = store
@@ -98,6 +105,9 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.usage = usage
     app.state.filestore = filestore
     app.state.triggers = triggers
+    app.state.rbac = rbac
+    app.state.oauth = oauth
+    app.state.profiles = profiles
 
     @app.on_event("startup")
     async def _start_reconciler():
@@ -377,6 +387,150 @@ def create_app(cfg: Optional[ServerConfig] = None,
     async def list_runners(user: AuthUser = Depends(admin_dep)):
         router.reap_offline()
         return [r.model_dump() for r in router.runners()]
+
+    # ------------------------------------------------------------------
+    # Runner profiles (reference server.go:1295-1320 + §3.5 flow)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/runner-profiles")
+    async def create_profile(request: Request,
+                             user: AuthUser = Depends(admin_dep)):
+        prof = RunnerProfile.model_validate(await request.json())
+        return profiles.create(prof).model_dump()
+
+    @app.get("/api/v1/runner-profiles")
+    async def list_profiles(user: AuthUser = Depends(admin_dep)):
+        return [p.model_dump() for p in profiles.list()]
+
+    @app.delete("/api/v1/runner-profiles/{pid}")
+    async def delete_profile(pid: str,
+                             user: AuthUser = Depends(admin_dep)):
+        return {"ok": profiles.delete(pid)}
+
+    @app.post("/api/v1/runners/{runner_id}/assign-profile")
+    async def assign_profile(runner_id: str, request: Request,
+                             user: AuthUser = Depends(admin_dep)):
+        body = await request.json()
+        state = next((r for r in router.runners()
+                      if r.runner_id == runner_id), None)
+        gpus = state.gpus if state else []
+        ok, why = profiles.assign(runner_id, body.get("profile_id", ""),
+                                  gpus)
+        if not ok:
+            raise HTTPException(409, why)
+        return {"ok": True}
+
+    @app.get("/api/v1/runner/{runner_id}/assignment")
+    async def get_assignment(runner_id: str, _=Depends(runner_dep)):
+        # runner polls its assigned profile (reference compose-manager
+        # 15 s poll, §3.5)
+        return profiles.assignment(runner_id) or {}
+
+    @app.delete("/api/v1/runners/{runner_id}/assignment")
+    async def clear_assignment(runner_id: str,
+                               user: AuthUser = Depends(admin_dep)):
+        return {"ok": profiles.clear_assignment(runner_id)}
+
+    # ------------------------------------------------------------------
+    # Orgs / teams / access grants (reference org routes + authz)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/organizations")
+    async def create_org(request: Request,
+                         user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return rbac.create_org(user.id, body.get("name", ""))
+
+    @app.get("/api/v1/organizations")
+    async def list_orgs(user: AuthUser = Depends(auth_dep)):
+        return rbac.list_orgs_for(user.id)
+
+    @app.post("/api/v1/organizations/{oid}/members")
+    async def add_org_member(oid: str, request: Request,
+                             user: AuthUser = Depends(auth_dep)):
+        if rbac.member_role(oid, user.id) not in ("owner", "admin") \
+                and not user.admin:
+            raise HTTPException(403, "org admin only")
+        body = await request.json()
+        return rbac.add_member(oid, body["user_id"],
+                               body.get("role", "member"))
+
+    @app.post("/api/v1/organizations/{oid}/teams")
+    async def create_team(oid: str, request: Request,
+                          user: AuthUser = Depends(auth_dep)):
+        if rbac.member_role(oid, user.id) is None and not user.admin:
+            raise HTTPException(403, "not a member")
+        body = await request.json()
+        return rbac.create_team(oid, body.get("name", ""))
+
+    @app.post("/api/v1/teams/{tid}/members")
+    async def add_team_member(tid: str, request: Request,
+                              user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return rbac.add_team_member(tid, body["user_id"])
+
+    @app.post("/api/v1/access-grants")
+    async def create_grant(request: Request,
+                           user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return rbac.grant(body.get("resource_type", ""),
+                          body.get("resource_id", ""),
+                          body.get("role", "viewer"),
+                          user_id=body.get("user_id", ""),
+                          team_id=body.get("team_id", ""),
+                          org_id=body.get("org_id", ""))
+
+    @app.get("/api/v1/access-grants/{resource_id}")
+    async def list_grants(resource_id: str,
+                          user: AuthUser = Depends(auth_dep)):
+        return rbac.grants_for(resource_id)
+
+    # ------------------------------------------------------------------
+    # OAuth (reference api/pkg/oauth routes)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/oauth/providers")
+    async def configure_oauth(request: Request,
+                              user: AuthUser = Depends(admin_dep)):
+        body = await request.json()
+        return oauth.configure_provider(
+            body["name"], body.get("client_id", ""),
+            body.get("client_secret", ""), body.get("auth_url", ""),
+            body.get("token_url", ""), body.get("scopes"))
+
+    @app.get("/api/v1/oauth/{provider}/authorize-url")
+    async def oauth_authorize(provider: str, redirect_uri: str,
+                              user: AuthUser = Depends(auth_dep)):
+        try:
+            return {"url": oauth.authorize_url(provider, redirect_uri,
+                                               state=user.id)}
+        except KeyError as e:
+            raise HTTPException(404, str(e))
+
+    @app.post("/api/v1/oauth/{provider}/token")
+    async def oauth_save_token(provider: str, request: Request,
+                               user: AuthUser = Depends(auth_dep)):
+        tok = await request.json()
+        doc = oauth.save_token(user.id, provider, tok)
+        return {"provider": provider,
+                "expires_at": doc["expires_at"]}
+
+    # ------------------------------------------------------------------
+    # Provider endpoints (user/org-defined OpenAI-compatible providers)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/provider-endpoints")
+    async def create_provider_endpoint(request: Request,
+                                       user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        name = body.get("name", "")
+        pid = f"{user.id}:{name}"
+        store.put("provider_endpoints", pid,
+                  {"id": pid, "name": name,
+                   "base_url": body.get("base_url", ""),
+                   "api_key": body.get("api_key", "")}, owner=user.id)
+        return {"name": name, "ok": True}
+
+    @app.get("/api/v1/provider-endpoints")
+    async def list_provider_endpoints(user: AuthUser = Depends(auth_dep)):
+        return [{"name": e["name"], "base_url": e["base_url"]}
+                for e in store.list("provider_endpoints", owner=user.id)]
 
     # ------------------------------------------------------------------
     # Anthropic-compatible surface (reference api/pkg/anthropic proxy)

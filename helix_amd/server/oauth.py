@@ -1,0 +1,133 @@
+"""OAuth manager (parity with api/pkg/oauth: provider registry, authorize
+URL + code exchange + token refresh over standard OAuth2 endpoints,
+per-app token injection into API tools)."""
+from __future__ import annotations
+
+import time
+from typing import Dict, List, Optional
+
+import httpx
+
+from helix_amd.server.types import new_id
+
+# Built-in provider endpoint templates (reference oauth providers incl.
+# github/google/hubspot)
+BUILTIN_PROVIDERS = {
+    "github": {
+        "auth_url": "https://github.com/login/oauth/authorize",
+        "token_url": "https://github.com/login/oauth/access_token",
+        "scopes": ["repo", "user"],
+    },
+    "google": {
+        "auth_url": "https://accounts.google.com/o/oauth2/v2/auth",
+        "token_url": "https://oauth2.googleapis.com/token",
+        "scopes": ["openid", "email"],
+    },
+    "hubspot": {
+        "auth_url": "https://app.hubspot.com/oauth/authorize",
+        "token_url": "https://api.hubapi.com/oauth/v1/token",
+        "scopes": ["crm.objects.contacts.read"],
+    },
+}
+
+
+class OAuthManager:
+    def __init__(self, store, http_client: Optional[httpx.AsyncClient] = None):
+        self.store = store
+        self._http = http_client
+
+    def _client(self) -> httpx.AsyncClient:
+        if self._http is None:
+            self._http = httpx.AsyncClient(timeout=30)
+        return self._http
+
+    # -- provider configs (client_id/secret stored per deployment) ---------
+    def configure_provider(self, name: str, client_id: str,
+                           client_secret: str, auth_url: str = "",
+                           token_url: str = "", scopes: List[str] = None):
+        base = BUILTIN_PROVIDERS.get(name, {})
+        doc = {
+            "id": f"oauth-provider:{name}", "name": name,
+            "client_id": client_id, "client_secret": client_secret,
+            "auth_url": auth_url or base.get("auth_url", ""),
+            "token_url": token_url or base.get("token_url", ""),
+            "scopes": scopes or base.get("scopes", []),
+        }
+        self.store.put("system_settings", doc["id"], doc)
+        return doc
+
+    def provider(self, name: str) -> Optional[dict]:
+        return self.store.get("system_settings", f"oauth-provider:{name}")
+
+    def authorize_url(self, name: str, redirect_uri: str,
+                      state: str) -> str:
+        p = self.provider(name)
+        if p is None:
+            raise KeyError(f"oauth provider not configured: {name}")
+        from urllib.parse import urlencode
+        return p["auth_url"] + "?" + urlencode({
+            "client_id": p["client_id"], "redirect_uri": redirect_uri,
+            "scope": " ".join(p["scopes"]), "state": state,
+            "response_type": "code"})
+
+    # -- token lifecycle ----------------------------------------------------
+    async def exchange_code(self, name: str, code: str, redirect_uri: str,
+                            owner: str) -> dict:
+        p = self.provider(name)
+        r = await self._client().post(p["token_url"], data={
+            "client_id": p["client_id"],
+            "client_secret": p["client_secret"],
+            "code": code, "redirect_uri": redirect_uri,
+            "grant_type": "authorization_code"},
+            headers={"Accept": "application/json"})
+        tok = r.json()
+        return self.save_token(owner, name, tok)
+
+    def save_token(self, owner: str, provider: str, tok: dict) -> dict:
+        doc = {
+            "id": f"{owner}:{provider}", "provider": provider,
+            "access_token": tok.get("access_token", ""),
+            "refresh_token": tok.get("refresh_token", ""),
+            "expires_at": time.time() + float(tok.get("expires_in", 3600)),
+        }
+        self.store.put("oauth_tokens", doc["id"], doc, owner=owner)
+        return doc
+
+    def token_for(self, owner: str, provider: str) -> Optional[dict]:
+        return self.store.get("oauth_tokens", f"{owner}:{provider}")
+
+    async def get_valid_token(self, owner: str, provider: str
+                              ) -> Optional[str]:
+        """Returns an access token, refreshing if expired (reference
+        token refresh loop)."""
+        doc = self.token_for(owner, provider)
+        if doc is None:
+            return None
+        if doc["expires_at"] > time.time() + 60:
+            return doc["access_token"]
+        p = self.provider(provider)
+        if not p or not doc.get("refresh_token"):
+            return doc["access_token"] or None
+        r = await self._client().post(p["token_url"], data={
+            "client_id": p["client_id"],
+            "client_secret": p["client_secret"],
+            "refresh_token": doc["refresh_token"],
+            "grant_type": "refresh_token"},
+            headers={"Accept": "application/json"})
+        tok = r.json()
+        if tok.get("access_token"):
+            tok.setdefault("refresh_token", doc["refresh_token"])
+            doc = self.save_token(owner, provider, tok)
+        return doc["access_token"] or None
+
+    async def inject_auth(self, owner: str, oauth_provider: str,
+                          headers: Dict[str, str]) -> Dict[str, str]:
+        """OAuth token injection for API tools (reference
+        inference.go:137 -> :2041)."""
+        if not oauth_provider:
+            return headers
+        token = await self.get_valid_token(owner, oauth_provider)
+        if token:
+            headers = dict(headers)
+            headers["Authorization"] = f"Bearer {token}"
+        return headers

@@ -20,7 +20,7 @@ _TABLES = [
     "triggers", "organizations", "teams", "memberships", "access_grants",
     "runner_profiles", "projects", "spec_tasks", "git_repositories",
     "evaluation_runs", "system_settings", "wallets", "transactions",
-    "rag_chunks", "usage_rollups",
+    "rag_chunks", "usage_rollups", "runner_assignments",
 ]
 
 
