@@ -32,6 +32,10 @@ def main():
     ap.add_argument("--batch", type=int, default=256)
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--model", type=str, default="llama3-8b")
+    ap.add_argument("--parallelism", type=str, default="dp",
+                    choices=["dp", "tp"],
+                    help="dp: one engine per GPU (weak scaling); "
+                         "tp: one tensor-parallel engine over all GPUs")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -48,6 +52,10 @@ def main():
         world = dist.get_world_size()
         torch.cuda.set_device(rank % torch.cuda.device_count())
     device = f"cuda:{rank % max(1, torch.cuda.device_count())}"
+    tp = args.parallelism == "tp" and world > 1
+    if tp:
+        from helix_amd import parallel
+        parallel.init_tp(world)
 
     max_len = args.prompt_len + args.warmup + args.steps + 64
     cfg = EngineConfig(
@@ -58,9 +66,18 @@ def main():
         eos_token_id=-1,
         seed=rank,
     )
-    eng = LLMEngine(cfg, device=device)
+    if tp:
+        # SPMD tensor parallelism: every rank runs the identical engine
+        # loop; the model's row-parallel all-reduces (RCCL over xGMI) are
+        # the only cross-rank communication. Identical seeds keep
+        # scheduling and sampling converged across ranks.
+        cfg.seed = 0
+        cfg.enforce_eager = True
+        eng = LLMEngine(cfg, device=device, tp_size=world, tp_rank=rank)
+    else:
+        eng = LLMEngine(cfg, device=device)
 
-    torch.manual_seed(1234 + rank)
+    torch.manual_seed(1234 + (0 if tp else rank))
     vocab = eng.model_cfg.vocab_size
     prompts = torch.randint(3, vocab - 1,
                             (args.batch, args.prompt_len)).tolist()
@@ -97,7 +114,7 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    tokens = args.batch * args.steps * world
+    tokens = args.batch * args.steps * (1 if tp else world)
     value = tokens / elapsed
     if rank == 0:
         print(json.dumps({
@@ -109,7 +126,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1000, 3),
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "strong" if tp else "weak",
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
@@ -117,7 +134,7 @@ def main():
                 "model": args.model,
                 "global_batch": args.batch * world,
                 "seq_len": args.prompt_len,
-                "parallelism": f"dp{world}",
+                "parallelism": (f"tp{world}" if tp else f"dp{world}"),
                 "ttft_p50_ms": round(ttft_p50, 2),
                 "prefill_s": round(prefill_s, 3),
             },
