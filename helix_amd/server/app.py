@@ -80,16 +80,24 @@ def create_app(cfg: Optional[ServerConfig] = None,
     usage = UsageService(store, catalog)
     filestore = FileStore(cfg.filestore.path)
 
+    from helix_amd.server.code_intel import CodeIntelService
+    from helix_amd.server.git_service import GitService
+    from helix_amd.server.notifications import NotificationService
     from helix_amd.server.oauth import OAuthManager
     from helix_amd.server.rbac import RBACService
     from helix_amd.server.runner_profiles import (ProfileService,
                                                   RunnerProfile)
+    from helix_amd.server.spec_tasks import SpecTaskService
     controller = Controller(cfg, store, providers, pubsub, rag=rag,
                             agent_runner=agent_runner, usage=usage)
     triggers = TriggerManager(store, controller)
     rbac = RBACService(store)
     oauth = OAuthManager(store)
     profiles = ProfileService(store)
+    git_svc = GitService(store, cfg.filestore.path)
+    spec_tasks = SpecTaskService(store, controller, git_svc)
+    code_intel = CodeIntelService(rag, git_svc)
+    notifications = NotificationService(store=store)
 
     app.state.cfg = cfg
     app.state.store = store
@@ -108,6 +116,10 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.rbac = rbac
     app.state.oauth = oauth
     app.state.profiles = profiles
+    app.state.git = git_svc
+    app.state.spec_tasks = spec_tasks
+    app.state.code_intel = code_intel
+    app.state.notifications = notifications
 
     @app.on_event("startup")
     async def _start_reconciler():
@@ -707,6 +719,78 @@ def create_app(cfg: Optional[ServerConfig] = None,
         if session_id:
             return store.list("llm_calls", parent=session_id)
         return store.list("llm_calls", limit=200)
+
+    # ------------------------------------------------------------------
+    # Projects / spec-tasks / git / code-intel (reference services)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/projects")
+    async def create_project(request: Request,
+                             user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return spec_tasks.create_project(user.id, body.get("name", ""))
+
+    @app.get("/api/v1/projects")
+    async def list_projects(user: AuthUser = Depends(auth_dep)):
+        return spec_tasks.list_projects(user.id)
+
+    @app.post("/api/v1/projects/{pid}/tasks")
+    async def create_task(pid: str, request: Request,
+                          user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return spec_tasks.create_task(user.id, pid, body.get("title", ""),
+                                      body.get("description", ""))
+
+    @app.get("/api/v1/projects/{pid}/tasks")
+    async def list_tasks(pid: str, user: AuthUser = Depends(auth_dep)):
+        return spec_tasks.list_tasks(pid)
+
+    @app.post("/api/v1/spec-tasks/{tid}/transition")
+    async def transition_task(tid: str, request: Request,
+                              user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        try:
+            return spec_tasks.transition(tid, body.get("state", ""))
+        except (KeyError, ValueError) as e:
+            raise HTTPException(400, str(e))
+
+    @app.post("/api/v1/spec-tasks/{tid}/plan")
+    async def plan_task(tid: str, user: AuthUser = Depends(auth_dep)):
+        try:
+            return await spec_tasks.plan(tid)
+        except KeyError:
+            raise HTTPException(404, "task not found")
+
+    @app.post("/api/v1/spec-tasks/{tid}/comments")
+    async def comment_task(tid: str, request: Request,
+                           user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return spec_tasks.add_comment(tid, user.id, body.get("text", ""))
+
+    @app.get("/api/v1/git/repos")
+    async def list_repos(user: AuthUser = Depends(auth_dep)):
+        return git_svc.list(user.id)
+
+    @app.get("/api/v1/git/repos/{rid}/log")
+    async def repo_log(rid: str, user: AuthUser = Depends(auth_dep),
+                       ref: str = "HEAD"):
+        return git_svc.log(rid, ref)
+
+    @app.get("/api/v1/git/repos/{rid}/files")
+    async def repo_files(rid: str, user: AuthUser = Depends(auth_dep),
+                         ref: str = "HEAD"):
+        return git_svc.ls_tree(rid, ref)
+
+    @app.post("/api/v1/git/repos/{rid}/index")
+    async def index_repo(rid: str, user: AuthUser = Depends(auth_dep)):
+        n = await code_intel.index_repo(rid)
+        return {"chunks": n}
+
+    @app.post("/api/v1/git/repos/{rid}/search")
+    async def search_repo(rid: str, request: Request,
+                          user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return await code_intel.query(rid, body.get("query", ""),
+                                      body.get("k", 6))
 
     @app.get("/api/v1/config")
     async def get_config(user: AuthUser = Depends(auth_dep)):

@@ -1,0 +1,109 @@
+"""Git repository service (parity with api/pkg/services git_*: server-side
+bare repos under the filestore, branch listing, commit log, file access;
+PR flows against external forges are config-gated and offline-safe)."""
+from __future__ import annotations
+
+import os
+import subprocess
+import time
+from typing import List, Optional
+
+from helix_amd.server.types import new_id
+
+
+def _git(repo: str, *args: str, check: bool = True) -> str:
+    res = subprocess.run(["git", *args], cwd=repo, capture_output=True,
+                         text=True, timeout=60)
+    if check and res.returncode != 0:
+        raise RuntimeError(f"git {' '.join(args)} failed: {res.stderr}")
+    return res.stdout
+
+
+class GitService:
+    def __init__(self, store, root: str):
+        self.store = store
+        self.root = os.path.join(os.path.abspath(root), "git-repositories")
+        os.makedirs(self.root, exist_ok=True)
+
+    def _path(self, repo_id: str) -> str:
+        return os.path.join(self.root, f"{repo_id}.git")
+
+    def create(self, owner: str, name: str, project_id: str = "") -> dict:
+        rid = new_id("repo")
+        path = self._path(rid)
+        subprocess.run(["git", "init", "--bare", path], check=True,
+                       capture_output=True)
+        subprocess.run(["git", "symbolic-ref", "HEAD", "refs/heads/main"],
+                       cwd=path, check=True, capture_output=True)
+        doc = {"id": rid, "name": name, "owner": owner,
+               "project_id": project_id, "path": path,
+               "created": time.time()}
+        self.store.put("git_repositories", rid, doc, owner=owner,
+                       parent=project_id)
+        return doc
+
+    def get(self, rid: str) -> Optional[dict]:
+        return self.store.get("git_repositories", rid)
+
+    def list(self, owner: str) -> List[dict]:
+        return self.store.list("git_repositories", owner=owner)
+
+    def delete(self, rid: str) -> bool:
+        import shutil
+        path = self._path(rid)
+        if os.path.isdir(path):
+            shutil.rmtree(path)
+        return self.store.delete("git_repositories", rid)
+
+    # -- repo ops ----------------------------------------------------------
+    def branches(self, rid: str) -> List[str]:
+        out = _git(self._path(rid), "branch", "--list",
+                   "--format=%(refname:short)")
+        return [b for b in out.splitlines() if b]
+
+    def log(self, rid: str, ref: str = "HEAD", n: int = 20) -> List[dict]:
+        try:
+            out = _git(self._path(rid), "log", ref, f"-{n}",
+                       "--format=%H%x1f%an%x1f%at%x1f%s")
+        except RuntimeError:
+            return []
+        commits = []
+        for line in out.splitlines():
+            h, an, at, s = line.split("\x1f")
+            commits.append({"hash": h, "author": an, "ts": int(at),
+                            "subject": s})
+        return commits
+
+    def read_file(self, rid: str, path: str, ref: str = "HEAD") -> str:
+        return _git(self._path(rid), "show", f"{ref}:{path}")
+
+    def ls_tree(self, rid: str, ref: str = "HEAD") -> List[str]:
+        try:
+            out = _git(self._path(rid), "ls-tree", "-r", "--name-only", ref)
+        except RuntimeError:
+            return []
+        return [p for p in out.splitlines() if p]
+
+    def commit_files(self, rid: str, files: dict, message: str,
+                     branch: str = "main") -> str:
+        """Author a commit of {path: content} onto `branch` (used by the
+        spec-task planner to write helix-specs, reference
+        git_repository_service behavior)."""
+        import tempfile
+        bare = self._path(rid)
+        with tempfile.TemporaryDirectory() as td:
+            wt = os.path.join(td, "wt")
+            subprocess.run(["git", "clone", "-q", bare, wt], check=True,
+                           capture_output=True)
+            _git(wt, "checkout", "-B", branch)
+            for rel, content in files.items():
+                full = os.path.join(wt, rel)
+                os.makedirs(os.path.dirname(full) or wt, exist_ok=True)
+                with open(full, "w") as f:
+                    f.write(content)
+            _git(wt, "add", "-A")
+            _git(wt, "-c", "user.email=agent@helix", "-c",
+                 "user.name=helix-agent", "commit", "-m", message,
+                 check=False)
+            _git(wt, "push", "-q", "origin", branch)
+            return _git(wt, "rev-parse", "HEAD").strip()

@@ -1,0 +1,125 @@
+"""Spec-task system (parity with api/pkg/services spec_*: the kanban
+pipeline Backlog -> Planning -> SpecReview -> InProgress -> PR -> Merged,
+driven by planning/implementation agents; design docs written into the
+project's git repo)."""
+from __future__ import annotations
+
+import logging
+import time
+from typing import List, Optional
+
+from helix_amd.server.types import new_id
+
+log = logging.getLogger("helix_amd.spec_tasks")
+
+STATES = ["backlog", "planning", "spec_review", "in_progress", "pr",
+          "merged", "failed"]
+# allowed transitions
+_NEXT = {
+    "backlog": {"planning"},
+    "planning": {"spec_review", "failed"},
+    "spec_review": {"in_progress", "planning", "failed"},
+    "in_progress": {"pr", "failed"},
+    "pr": {"merged", "in_progress", "failed"},
+    "merged": set(),
+    "failed": {"backlog"},
+}
+
+
+class SpecTaskService:
+    def __init__(self, store, controller=None, git=None):
+        self.store = store
+        self.controller = controller
+        self.git = git
+
+    # -- projects ----------------------------------------------------------
+    def create_project(self, owner: str, name: str) -> dict:
+        pid = new_id("proj")
+        doc = {"id": pid, "name": name, "owner": owner,
+               "created": time.time()}
+        self.store.put("projects", pid, doc, owner=owner)
+        if self.git is not None:
+            repo = self.git.create(owner, name, project_id=pid)
+            doc["repo_id"] = repo["id"]
+            self.store.put("projects", pid, doc, owner=owner)
+        return doc
+
+    def list_projects(self, owner: str) -> List[dict]:
+        return self.store.list("projects", owner=owner)
+
+    # -- tasks -------------------------------------------------------------
+    def create_task(self, owner: str, project_id: str, title: str,
+                    description: str = "") -> dict:
+        tid = new_id("task")
+        doc = {"id": tid, "project_id": project_id, "owner": owner,
+               "title": title, "description": description,
+               "state": "backlog", "spec": "", "comments": [],
+               "created": time.time(), "updated": time.time()}
+        self.store.put("spec_tasks", tid, doc, owner=owner,
+                       parent=project_id)
+        return doc
+
+    def get_task(self, tid: str) -> Optional[dict]:
+        return self.store.get("spec_tasks", tid)
+
+    def list_tasks(self, project_id: str) -> List[dict]:
+        return self.store.list("spec_tasks", parent=project_id, desc=False)
+
+    def transition(self, tid: str, new_state: str) -> dict:
+        doc = self.get_task(tid)
+        if doc is None:
+            raise KeyError(tid)
+        if new_state not in STATES:
+            raise ValueError(f"unknown state: {new_state}")
+        if new_state not in _NEXT[doc["state"]]:
+            raise ValueError(
+                f"illegal transition {doc['state']} -> {new_state}")
+        doc["state"] = new_state
+        doc["updated"] = time.time()
+        self.store.put("spec_tasks", tid, doc, owner=doc["owner"],
+                       parent=doc["project_id"])
+        return doc
+
+    def add_comment(self, tid: str, author: str, text: str) -> dict:
+        doc = self.get_task(tid)
+        doc["comments"].append({"author": author, "text": text,
+                                "ts": time.time()})
+        self.store.put("spec_tasks", tid, doc, owner=doc["owner"],
+                       parent=doc["project_id"])
+        return doc
+
+    # -- planning agent (reference spec_task_orchestrator planning phase) --
+    async def plan(self, tid: str, model: str = "") -> dict:
+        doc = self.transition(tid, "planning")
+        if self.controller is None:
+            raise RuntimeError("controller unavailable")
+        req = {
+            "model": model or self.controller.cfg.inference.default_model,
+            "messages": [
+                {"role": "system",
+                 "content": "You are a software planning agent. Write a "
+                            "concise implementation spec in markdown."},
+                {"role": "user",
+                 "content": f"Task: {doc['title']}\n\n{doc['description']}"},
+            ],
+        }
+        try:
+            resp = await self.controller.chat_completion(
+                req, doc["owner"], ctx={"owner": doc["owner"],
+                                        "step": "spec_plan"})
+            spec = resp["choices"][0]["message"]["content"]
+            doc = self.get_task(tid)
+            doc["spec"] = spec
+            self.store.put("spec_tasks", tid, doc, owner=doc["owner"],
+                           parent=doc["project_id"])
+            # write the spec into the project repo (helix-specs authoring)
+            project = self.store.get("projects", doc["project_id"])
+            if self.git is not None and project and project.get("repo_id"):
+                self.git.commit_files(
+                    project["repo_id"],
+                    {f"specs/{tid}.md": spec},
+                    f"spec: {doc['title']}", branch="helix-specs")
+            return self.transition(tid, "spec_review")
+        except Exception:
+            log.exception("planning failed for %s", tid)
+            return self.transition(tid, "failed")

@@ -1,0 +1,123 @@
+"""Git service, spec-tasks, code intel, notifications."""
+import asyncio
+
+import pytest
+from fastapi.testclient import TestClient
+
+from helix_amd.server.app import create_app
+from helix_amd.server.config import ServerConfig
+from helix_amd.server.providers import MockClient, ProviderManager
+from helix_amd.store import Store
+from tests.test_rag_agent import EmbedMock
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    cfg.rag.embeddings_provider = "mock"
+    cfg.filestore.path = str(tmp_path / "fs")
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    pm.register("mock", EmbedMock(responses=["# Spec\nBuild it well."]))
+    app = create_app(cfg, store=store, providers=pm)
+    client = TestClient(app)
+    r = client.post("/api/v1/users", json={"username": "alice"},
+                    headers={"Authorization": "Bearer admin-key"})
+    return app, client, r.json()["api_key"], store
+
+
+def H(key):
+    return {"Authorization": f"Bearer {key}"}
+
+
+def test_project_and_git_repo(stack):
+    app, client, key, _ = stack
+    r = client.post("/api/v1/projects", json={"name": "proj1"},
+                    headers=H(key))
+    proj = r.json()
+    assert proj["repo_id"]
+    repos = client.get("/api/v1/git/repos", headers=H(key)).json()
+    assert repos[0]["id"] == proj["repo_id"]
+
+
+def test_spec_task_pipeline(stack):
+    app, client, key, store = stack
+    pid = client.post("/api/v1/projects", json={"name": "p"},
+                      headers=H(key)).json()["id"]
+    t = client.post(f"/api/v1/projects/{pid}/tasks", json={
+        "title": "Add login", "description": "users must log in"},
+        headers=H(key)).json()
+    assert t["state"] == "backlog"
+    # illegal transition rejected
+    r = client.post(f"/api/v1/spec-tasks/{t['id']}/transition",
+                    json={"state": "merged"}, headers=H(key))
+    assert r.status_code == 400
+    # planning agent writes spec + moves to spec_review
+    r = client.post(f"/api/v1/spec-tasks/{t['id']}/plan", headers=H(key))
+    doc = r.json()
+    assert doc["state"] == "spec_review"
+    task = store.get("spec_tasks", t["id"])
+    assert "Build it well" in task["spec"]
+    # spec landed in the project repo on helix-specs branch
+    proj = store.get("projects", pid)
+    git = app.state.git
+    assert "helix-specs" in git.branches(proj["repo_id"])
+    files = git.ls_tree(proj["repo_id"], "helix-specs")
+    assert any(f.startswith("specs/") for f in files)
+    # continue the pipeline
+    for state in ("in_progress", "pr", "merged"):
+        doc = client.post(f"/api/v1/spec-tasks/{t['id']}/transition",
+                          json={"state": state}, headers=H(key)).json()
+    assert doc["state"] == "merged"
+
+
+def test_code_intel_index_and_search(stack):
+    app, client, key, store = stack
+    pid = client.post("/api/v1/projects", json={"name": "code"},
+                      headers=H(key)).json()["id"]
+    proj = store.get("projects", pid)
+    rid = proj["repo_id"]
+    app.state.git.commit_files(rid, {
+        "main.py": "def hello_world():\n    return 'greetings'\n",
+        "util.py": "def compute_checksum(data):\n    return sum(data)\n",
+        "README.md": "# demo repo\n",
+    }, "initial", branch="main")
+    r = client.post(f"/api/v1/git/repos/{rid}/index", headers=H(key))
+    assert r.json()["chunks"] >= 3
+    r = client.post(f"/api/v1/git/repos/{rid}/search",
+                    json={"query": "compute checksum data"}, headers=H(key))
+    results = r.json()
+    assert results
+    assert any("checksum" in x["text"] for x in results[:2])
+
+
+def test_git_log_and_read(stack):
+    app, client, key, store = stack
+    pid = client.post("/api/v1/projects", json={"name": "g"},
+                      headers=H(key)).json()["id"]
+    rid = store.get("projects", pid)["repo_id"]
+    app.state.git.commit_files(rid, {"a.txt": "v1"}, "first", "main")
+    app.state.git.commit_files(rid, {"a.txt": "v2"}, "second", "main")
+    log = client.get(f"/api/v1/git/repos/{rid}/log?ref=main",
+                     headers=H(key)).json()
+    assert [c["subject"] for c in log[:2]] == ["second", "first"]
+    assert app.state.git.read_file(rid, "a.txt", "main") == "v2"
+
+
+def test_notifications_offline_safe():
+    from helix_amd.server.notifications import NotificationService
+    n = NotificationService()
+    res = n.notify("u1", "test", "body", email_to="x@y.z")
+    assert res["email"] is False and res["webhook"] is False
+    assert len(n.sent) == 1
+
+
+def test_code_chunker():
+    from helix_amd.server.code_intel import chunk_code
+    text = "\n".join(f"line {i}" for i in range(150))
+    chunks = chunk_code(text, "x.py", max_lines=60, overlap=10)
+    assert len(chunks) == 3
+    assert chunks[0]["metadata"]["start_line"] == 1
+    assert chunks[1]["metadata"]["start_line"] == 51
