@@ -33,7 +33,7 @@ constexpr int MAX_G = 8;
 constexpr int PART_QUANT = 128;
 
 template <int DHEAD, int G>
-__global__ __launch_bounds__(NTHREADS, (G <= 4 ? 3 : 2)) void paged_attn_decode_kernel(
+__global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
     float* __restrict__ tmp_ml,          // [B, Hq, maxP, 2]
@@ -66,47 +66,39 @@ __global__ __launch_bounds__(NTHREADS, (G <= 4 ? 3 : 2)) void paged_attn_decode_
   const int p_end = min(len, p_start + partition_size);
 
   __shared__ float q_lds[G][DHEAD];
-  __shared__ float p_lds[2][G][WAVE];      // per-wave P handoff (A->C)
-  __shared__ float merge[2][G][DHEAD + 2]; // cross-wave merge: acc + (m,l)
+  __shared__ float s_lds[G][CHUNK];
+  __shared__ float head_m[MAX_G], head_l[MAX_G], head_corr[MAX_G];
 
   for (int idx = threadIdx.x; idx < G * DHEAD; idx += NTHREADS) {
     const int g = idx / DHEAD, d = idx % DHEAD;
     q_lds[g][d] =
         bf16_to_f32(q[((int64_t)seq * Hq + hkv * G + g) * DHEAD + d]) * scale;
   }
+  if (threadIdx.x < MAX_G) {
+    head_m[threadIdx.x] = -INFINITY;
+    head_l[threadIdx.x] = 0.f;
+  }
   __syncthreads();
 
-  // Wave-private processing: each wave walks its own 64-token chunks of
-  // the partition with NO block barriers in the main loop (the v3
-  // 3-barrier structure convoyed both waves: SQ_WAIT_ANY was 14x busy
-  // cycles). Scores live one-per-lane, softmax reduces via shuffles,
-  // P crosses lanes through a wave-private LDS row.
   constexpr int PAIRS = DHEAD / 2;
-  constexpr int N_PAR = WAVE / PAIRS;       // 1 for D=128, 2 for D=64
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const int d_own = (lane % PAIRS) * 2;
-  const int par = lane / PAIRS;
-
-  float m_run[G], l_run[G], acc[G][2];
+  constexpr int N_PAR = NTHREADS / PAIRS;   // 1 for D=128, 2 for D=64
+  const int d_own = (threadIdx.x % PAIRS) * 2;
+  const int par = threadIdx.x / PAIRS;
+  float acc[G][2];
 #pragma unroll
-  for (int g = 0; g < G; ++g) {
-    m_run[g] = -INFINITY;
-    l_run[g] = 0.f;
-    acc[g][0] = acc[g][1] = 0.f;
-  }
+  for (int g = 0; g < G; ++g) acc[g][0] = acc[g][1] = 0.f;
 
   const int* btable = block_tables + (int64_t)seq * max_blocks;
+  const int nwaves = NTHREADS / WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
 
-  for (int base = p_start + wid * WAVE; base < p_end; base += 2 * WAVE) {
-    const int chunk_n = min(WAVE, p_end - base);
+  for (int base = p_start; base < p_end; base += CHUNK) {
+    const int chunk_n = min(CHUNK, p_end - base);
 
-    // --- A: my token's scores (batched K-row load) ---
-    float sg[G];
-#pragma unroll
-    for (int g = 0; g < G; ++g) sg[g] = -INFINITY;
-    if (lane < chunk_n) {
-      const int tok = base + lane;
+    // --- Phase A: scores (batched K-row loads, then unrolled math) -------
+    if ((int)threadIdx.x < chunk_n) {
+      const int tok = base + threadIdx.x;
       const int64_t blk = btable[tok / block_size];
       const uint16_t* krow =
           k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
@@ -115,11 +107,10 @@ __global__ __launch_bounds__(NTHREADS, (G <= 4 ? 3 : 2)) void paged_attn_decode_
 #pragma unroll
       for (int j = 0; j < DHEAD / 8; ++j)
         kraw[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+      float s[G];
 #pragma unroll
-      for (int g = 0; g < G; ++g) sg[g] = 0.f;
-      // partial unroll: full unrolling hoists G*DHEAD q_lds loads live
-      // at once and spills hundreds of VGPRs
-#pragma unroll 2
+      for (int g = 0; g < G; ++g) s[g] = 0.f;
+#pragma unroll
       for (int j = 0; j < DHEAD / 8; ++j) {
         float kv[8];
 #pragma unroll
@@ -127,123 +118,129 @@ __global__ __launch_bounds__(NTHREADS, (G <= 4 ? 3 : 2)) void paged_attn_decode_
 #pragma unroll
         for (int g = 0; g < G; ++g) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i) sg[g] += q_lds[g][j * 8 + i] * kv[i];
+          for (int i = 0; i < 8; ++i) s[g] += q_lds[g][j * 8 + i] * kv[i];
         }
       }
+#pragma unroll
+      for (int g = 0; g < G; ++g) s_lds[g][threadIdx.x] = s[g];
     }
+    __syncthreads();
 
-    // --- B: online softmax via shuffles (no LDS, no barrier) ---
-    float corr[G];
-#pragma unroll
-    for (int g = 0; g < G; ++g) {
-      const float m_chunk = wave_reduce_max(sg[g]);
-      const float m_new = fmaxf(m_run[g], m_chunk);
-      corr[g] = (m_run[g] == -INFINITY) ? 0.f : __expf(m_run[g] - m_new);
-      const float pv = (lane < chunk_n) ? __expf(sg[g] - m_new) : 0.f;
-      l_run[g] = l_run[g] * corr[g] + wave_reduce_sum(pv);
-      m_run[g] = m_new;
-      p_lds[wid][g][lane] = pv;
-    }
-
-    // --- C: V accumulation (lane owns a dim pair; 8-token batches) ---
-#pragma unroll
-    for (int g = 0; g < G; ++g) {
-      acc[g][0] *= corr[g];
-      acc[g][1] *= corr[g];
-    }
-    const int ntok = (chunk_n - par + N_PAR - 1) / N_PAR;
-    int t8 = 0;
-    for (; t8 + 8 <= ntok; t8 += 8) {
-      uint32_t vv[8];
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const int tok = base + (t8 + u) * N_PAR + par;
-        const int64_t blk = btable[tok / block_size];
-        vv[u] = *reinterpret_cast<const uint32_t*>(
-            v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                        tok % block_size)) * DHEAD + d_own);
+    // --- Phase B: online softmax per head --------------------------------
+    for (int g = wid; g < G; g += nwaves) {
+      float m_chunk = -INFINITY;
+      for (int i = lane; i < chunk_n; i += WAVE)
+        m_chunk = fmaxf(m_chunk, s_lds[g][i]);
+      m_chunk = wave_reduce_max(m_chunk);
+      const float m_old = head_m[g];
+      const float m_new = fmaxf(m_old, m_chunk);
+      const float corr = (m_old == -INFINITY) ? 0.f : __expf(m_old - m_new);
+      float l_add = 0.f;
+      for (int i = lane; i < chunk_n; i += WAVE) {
+        const float p = __expf(s_lds[g][i] - m_new);
+        s_lds[g][i] = p;
+        l_add += p;
       }
-#pragma unroll
-      for (int u = 0; u < 8; ++u) {
-        const int tok_i = (t8 + u) * N_PAR + par;
-        const float v0 = bf16_to_f32((uint16_t)(vv[u] & 0xffff));
-        const float v1 = bf16_to_f32((uint16_t)(vv[u] >> 16));
-#pragma unroll
-        for (int g = 0; g < G; ++g) {
-          const float pv = p_lds[wid][g][tok_i];
-          acc[g][0] += pv * v0;
-          acc[g][1] += pv * v1;
-        }
+      l_add = wave_reduce_sum(l_add);
+      if (lane == 0) {
+        head_l[g] = head_l[g] * corr + l_add;
+        head_m[g] = m_new;
+        head_corr[g] = corr;
       }
     }
-    for (; t8 < ntok; ++t8) {
-      const int tok_i = t8 * N_PAR + par;
-      const int tok = base + tok_i;
-      const int64_t blk = btable[tok / block_size];
-      const uint32_t vvs = *reinterpret_cast<const uint32_t*>(
-          v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                      tok % block_size)) * DHEAD + d_own);
-      const float v0 = bf16_to_f32((uint16_t)(vvs & 0xffff));
-      const float v1 = bf16_to_f32((uint16_t)(vvs >> 16));
+    __syncthreads();
+
+    // --- Phase C: V accumulation (8-token load batches) -------------------
+    {
 #pragma unroll
       for (int g = 0; g < G; ++g) {
-        const float pv = p_lds[wid][g][tok_i];
-        acc[g][0] += pv * v0;
-        acc[g][1] += pv * v1;
+        acc[g][0] *= head_corr[g];
+        acc[g][1] *= head_corr[g];
+      }
+      const int ntok = (chunk_n - par + N_PAR - 1) / N_PAR;  // my tokens
+      int t8 = 0;
+      for (; t8 + 8 <= ntok; t8 += 8) {
+        uint32_t vv[8];
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int tok = base + (t8 + u) * N_PAR + par;
+          const int64_t blk = btable[tok / block_size];
+          vv[u] = *reinterpret_cast<const uint32_t*>(
+              v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
+                          tok % block_size)) * DHEAD + d_own);
+        }
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int tok_i = (t8 + u) * N_PAR + par;
+          const float v0 = bf16_to_f32((uint16_t)(vv[u] & 0xffff));
+          const float v1 = bf16_to_f32((uint16_t)(vv[u] >> 16));
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = s_lds[g][tok_i];
+            acc[g][0] += p * v0;
+            acc[g][1] += p * v1;
+          }
+        }
+      }
+      for (; t8 < ntok; ++t8) {
+        const int tok_i = t8 * N_PAR + par;
+        const int tok = base + tok_i;
+        const int64_t blk = btable[tok / block_size];
+        const uint32_t vv = *reinterpret_cast<const uint32_t*>(
+            v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
+                        tok % block_size)) * DHEAD + d_own);
+        const float v0 = bf16_to_f32((uint16_t)(vv & 0xffff));
+        const float v1 = bf16_to_f32((uint16_t)(vv >> 16));
+#pragma unroll
+        for (int g = 0; g < G; ++g) {
+          const float p = s_lds[g][tok_i];
+          acc[g][0] += p * v0;
+          acc[g][1] += p * v1;
+        }
       }
     }
+    __syncthreads();
   }
 
-  // --- Merge the two waves' partial (m, l, acc) ---
-  // For D=64 also fold the two in-wave parities via shuffles first.
+  // Combine token parities (N_PAR==2 only for D=64) and write out.
   if (N_PAR > 1) {
-#pragma unroll
+    __shared__ float comb[N_PAR > 1 ? G * NTHREADS * 2 : 1];
+    #pragma unroll
     for (int g = 0; g < G; ++g) {
-      acc[g][0] += __shfl_xor(acc[g][0], PAIRS, WAVE);
-      acc[g][1] += __shfl_xor(acc[g][1], PAIRS, WAVE);
+      comb[((par * G + g) * PAIRS + d_own / 2) * 2 + 0] = acc[g][0];
+      comb[((par * G + g) * PAIRS + d_own / 2) * 2 + 1] = acc[g][1];
+    }
+    __syncthreads();
+    if (par == 0) {
+#pragma unroll
+      for (int g = 0; g < G; ++g)
+        for (int p = 1; p < N_PAR; ++p) {
+          acc[g][0] += comb[((p * G + g) * PAIRS + d_own / 2) * 2 + 0];
+          acc[g][1] += comb[((p * G + g) * PAIRS + d_own / 2) * 2 + 1];
+        }
     }
   }
-#pragma unroll
-  for (int g = 0; g < G; ++g) {
-    if (lane < PAIRS) {
-      merge[wid][g][d_own] = acc[g][0];
-      merge[wid][g][d_own + 1] = acc[g][1];
-    }
-    if (lane == 0) {
-      merge[wid][g][DHEAD] = m_run[g];
-      merge[wid][g][DHEAD + 1] = l_run[g];
-    }
-  }
-  __syncthreads();
-  if (wid == 0 && lane < PAIRS) {
+  if (par == 0) {
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-      const float m0 = merge[0][g][DHEAD], m1 = merge[1][g][DHEAD];
-      const float l0 = merge[0][g][DHEAD + 1], l1 = merge[1][g][DHEAD + 1];
-      const float M = fmaxf(m0, m1);
-      const float e0 = (m0 == -INFINITY) ? 0.f : __expf(m0 - M);
-      const float e1 = (m1 == -INFINITY) ? 0.f : __expf(m1 - M);
-      const float L = l0 * e0 + l1 * e1;
-      const float o0 = merge[0][g][d_own] * e0 + merge[1][g][d_own] * e1;
-      const float o1 =
-          merge[0][g][d_own + 1] * e0 + merge[1][g][d_own + 1] * e1;
       const int hq = hkv * G + g;
       if (nparts == 1) {
-        const float inv_l = 1.f / fmaxf(L, 1e-20f);
-        uint32_t packed = (uint32_t)f32_to_bf16(o0 * inv_l) |
-                          ((uint32_t)f32_to_bf16(o1 * inv_l) << 16);
+        const float inv_l = 1.f / fmaxf(head_l[g], 1e-20f);
+        uint32_t packed =
+            (uint32_t)f32_to_bf16(acc[g][0] * inv_l) |
+            ((uint32_t)f32_to_bf16(acc[g][1] * inv_l) << 16);
         *reinterpret_cast<uint32_t*>(
             out + ((int64_t)seq * Hq + hq) * DHEAD + d_own) = packed;
       } else {
         float* tp =
             tmp_out + (((int64_t)seq * Hq + hq) * max_parts + part) * DHEAD;
-        tp[d_own] = o0;
-        tp[d_own + 1] = o1;
+        tp[d_own] = acc[g][0];
+        tp[d_own + 1] = acc[g][1];
         if (d_own == 0) {
           float* ml =
               tmp_ml + (((int64_t)seq * Hq + hq) * max_parts + part) * 2;
-          ml[0] = M;
-          ml[1] = L;
+          ml[0] = head_m[g];
+          ml[1] = head_l[g];
         }
       }
     }
