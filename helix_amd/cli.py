@@ -68,6 +68,12 @@ def runner(api_url: str = typer.Option("http://localhost:8080"),
             api_url, cfg.runner_plane.runner_token, runner_id, addr, svc,
             interval=cfg.runner_plane.heartbeat_interval_s))
     threading.Thread(target=beat, daemon=True).start()
+
+    def poll_assignment():
+        from helix_amd.runner.assignment import assignment_loop
+        asyncio.run(assignment_loop(
+            api_url, cfg.runner_plane.runner_token, runner_id, svc))
+    threading.Thread(target=poll_assignment, daemon=True).start()
     uvicorn.run(api, host="0.0.0.0", port=port, log_level="info")
 
 

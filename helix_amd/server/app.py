@@ -123,12 +123,29 @@ def create_app(cfg: Optional[ServerConfig] = None,
 
     @app.on_event("startup")
     async def _start_reconciler():
+        # Startup recovery (reference serve.go:270-280
+        # ResetRunningInteractions): interactions stuck mid-stream from a
+        # previous process flip to error so clients never hang on them.
+        for it in store.list("interactions", limit=100000):
+            if it.get("state") in ("waiting", "editing"):
+                it["state"] = "error"
+                it["error"] = "server restarted mid-generation"
+                store.put("interactions", it["id"], it,
+                          owner=it.get("owner", ""),
+                          parent=it.get("session_id", ""))
         app.state._reconciler_task = asyncio.create_task(knowledge.run())
         app.state._trigger_task = asyncio.create_task(triggers.run())
 
+        async def _reaper():
+            # offline-runner reaper (reference 5 m threshold)
+            while True:
+                router.reap_offline()
+                await asyncio.sleep(30)
+        app.state._reaper_task = asyncio.create_task(_reaper())
+
     @app.on_event("shutdown")
     async def _stop_reconciler():
-        for attr in ("_reconciler_task", "_trigger_task"):
+        for attr in ("_reconciler_task", "_trigger_task", "_reaper_task"):
             t = getattr(app.state, attr, None)
             if t:
                 t.cancel()
@@ -801,5 +818,11 @@ def create_app(cfg: Optional[ServerConfig] = None,
     @app.get("/healthz")
     async def healthz():
         return {"ok": True}
+
+    @app.get("/")
+    async def index():
+        from fastapi.responses import HTMLResponse
+        from helix_amd.server.webui import INDEX_HTML
+        return HTMLResponse(INDEX_HTML)
 
     return app

@@ -119,3 +119,20 @@ def test_status_endpoint(client):
     r = client.get("/api/v1/status")
     body = r.json()
     assert any(m["model_id"] == "tiny" for m in body["models"])
+
+
+def test_apply_profile_reconciles(service):
+    from helix_amd.runner.assignment import apply_profile
+    res = apply_profile(service, {"id": "p1", "models": [
+        {"name": "tiny", "preset": "tiny", "max_model_len": 256,
+         "kv_cache_blocks": 64},
+        {"name": "tiny-bert", "preset": "tiny-bert", "kind": "embedding"},
+    ]})
+    assert set(res["loaded"]) == {"tiny", "tiny-bert"}
+    assert set(service.loaded_models()) == {"tiny", "tiny-bert"}
+    # switching profile unloads removed models (profile_switch scenario)
+    res = apply_profile(service, {"id": "p2", "models": [
+        {"name": "tiny-gqa", "preset": "tiny-gqa", "max_model_len": 256,
+         "kv_cache_blocks": 64}]})
+    assert set(res["unloaded"]) == {"tiny", "tiny-bert"}
+    assert service.loaded_models() == ["tiny-gqa"]

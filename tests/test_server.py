@@ -208,3 +208,23 @@ def test_llm_calls_admin_only(stack):
                       headers=H(key)).status_code == 403
     assert client.get("/api/v1/llm_calls",
                       headers=H("admin-key")).status_code == 200
+
+
+def test_webui_served(stack):
+    _, client, _, _, _ = stack
+    r = client.get("/")
+    assert r.status_code == 200
+    assert "helix_amd" in r.text
+
+
+def test_startup_recovery_resets_stuck_interactions(stack):
+    app, client, _, key, store = stack
+    store.put("interactions", "int_stuck", {
+        "id": "int_stuck", "session_id": "ses_x", "state": "waiting",
+        "prompt_message": "p", "response_message": ""}, parent="ses_x")
+    # re-enter startup handler
+    import asyncio
+    with TestClient(app):
+        pass
+    doc = store.get("interactions", "int_stuck")
+    assert doc["state"] == "error"
