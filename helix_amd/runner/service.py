@@ -41,7 +41,8 @@ class ModelSpec:
 
 DEFAULT_SPECS = {
     "llama3-8b": ModelSpec("llama3-8b", "llm", "llama3-8b"),
-    "llama3-70b": ModelSpec("llama3-70b", "llm", "llama3-70b"),
+    "llama3-70b": ModelSpec("llama3-70b", "llm", "llama3-70b",
+                            max_num_seqs=48, kv_cache_blocks=16384),
     "mistral-7b": ModelSpec("mistral-7b", "llm", "mistral-7b"),
     "bge-base": ModelSpec("bge-base", "embedding", "bge-base"),
     "bge-large": ModelSpec("bge-large", "embedding", "bge-large"),
