@@ -37,6 +37,13 @@ class CUDAGraphRunner:
         self.in_bt = torch.zeros(B, self.max_blocks, dtype=torch.int32,
                                  device=device)
         self.in_lens = torch.ones(B, dtype=torch.int32, device=device)
+        # pinned host staging (one DMA per field per step)
+        self.h_ids = torch.zeros(B, dtype=torch.int64, pin_memory=True)
+        self.h_pos = torch.zeros(B, dtype=torch.int64, pin_memory=True)
+        self.h_slots = torch.zeros(B, dtype=torch.int64, pin_memory=True)
+        self.h_lens = torch.ones(B, dtype=torch.int32, pin_memory=True)
+        self.h_bt = torch.zeros(B, self.max_blocks, dtype=torch.int32,
+                                pin_memory=True)
 
     def batch_bucket(self, n: int) -> int:
         for b in BATCH_BUCKETS:
@@ -87,23 +94,25 @@ class CUDAGraphRunner:
         nb = self.batch_bucket(B)
         lb = self.len_bucket(max_len)
         key = (nb, lb)
-        # Stage inputs into the static buffers (pad rows are inert:
-        # slot -1 => no KV write; seq_len 1 => one garbage token read).
-        self.in_ids[:nb] = 0
-        self.in_pos[:nb] = 0
-        self.in_slots[:nb] = -1
-        self.in_lens[:nb] = 1
-        self.in_bt[:nb].zero_()
-        self.in_ids[:B].copy_(torch.tensor(input_ids, dtype=torch.int64),
-                              non_blocking=True)
-        self.in_pos[:B].copy_(torch.tensor(positions, dtype=torch.int64),
-                              non_blocking=True)
-        self.in_slots[:B].copy_(torch.tensor(slots, dtype=torch.int64),
-                                non_blocking=True)
-        self.in_lens[:B].copy_(torch.tensor(seq_lens, dtype=torch.int32),
-                               non_blocking=True)
+        # Stage inputs into pinned host buffers then one async DMA per
+        # field (pad rows are inert: slot -1 => no KV write; seq_len 1 =>
+        # one garbage token read).
+        self.h_ids[:nb] = 0
+        self.h_pos[:nb] = 0
+        self.h_slots[:nb] = -1
+        self.h_lens[:nb] = 1
+        self.h_bt[:nb].zero_()
+        self.h_ids[:B] = torch.tensor(input_ids, dtype=torch.int64)
+        self.h_pos[:B] = torch.tensor(positions, dtype=torch.int64)
+        self.h_slots[:B] = torch.tensor(slots, dtype=torch.int64)
+        self.h_lens[:B] = torch.tensor(seq_lens, dtype=torch.int32)
         w = block_tables.shape[1]
-        self.in_bt[:B, :w].copy_(block_tables, non_blocking=True)
+        self.h_bt[:B, :w] = block_tables
+        self.in_ids[:nb].copy_(self.h_ids[:nb], non_blocking=True)
+        self.in_pos[:nb].copy_(self.h_pos[:nb], non_blocking=True)
+        self.in_slots[:nb].copy_(self.h_slots[:nb], non_blocking=True)
+        self.in_lens[:nb].copy_(self.h_lens[:nb], non_blocking=True)
+        self.in_bt[:nb].copy_(self.h_bt[:nb], non_blocking=True)
 
         if key not in self.graphs:
             self.graphs[key] = self._capture(nb, lb)
