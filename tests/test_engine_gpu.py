@@ -1,0 +1,50 @@
+"""GPU engine integration tests: hipGraph-vs-eager equivalence and
+paged-KV consistency on real hardware (tiny model)."""
+import pytest
+import torch
+
+from helix_amd.engine.engine import EngineConfig, LLMEngine
+from helix_amd.engine.sampling_params import SamplingParams
+
+pytestmark = pytest.mark.gpu
+
+
+def _mk(enforce_eager: bool, seed=7):
+    cfg = EngineConfig(model="tiny-gqa", max_model_len=512, max_num_seqs=8,
+                       kv_cache_blocks=256, eos_token_id=-1, seed=seed,
+                       enforce_eager=enforce_eager)
+    return LLMEngine(cfg, device="cuda:0")
+
+
+def test_graph_matches_eager_greedy():
+    prompts = [[1, 2, 3, 4, 5, 6, 7, 8], [9, 10, 11], [12] * 33]
+    sp = SamplingParams(temperature=0.0, max_tokens=24, ignore_eos=True)
+    out_eager = _mk(True).generate(prompts, sp)
+    out_graph = _mk(False).generate(prompts, sp)
+    assert out_eager == out_graph
+
+
+def test_continuous_batching_gpu():
+    eng = _mk(False)
+    sp = SamplingParams(temperature=0.0, max_tokens=12, ignore_eos=True)
+    eng.add_request("a", [1, 2, 3], sp)
+    eng.step()
+    eng.add_request("b", [4, 5, 6, 7], sp)
+    while eng.has_work:
+        eng.step()
+    assert len(eng.seqs["a"].output_ids) == 12
+    assert len(eng.seqs["b"].output_ids) == 12
+    # all KV blocks returned
+    assert eng.num_free_blocks() == 256
+
+
+def test_decode_matches_fresh_prefill_gpu():
+    eng = _mk(True)
+    sp = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    prompt = list(range(1, 20))
+    out = eng.generate([prompt], sp)[0]
+    eng2 = _mk(True)
+    out2 = eng2.generate([prompt + out[:-1]],
+                         SamplingParams(temperature=0.0, max_tokens=1,
+                                        ignore_eos=True))[0]
+    assert out2[0] == out[-1]
