@@ -267,6 +267,57 @@ def create_app(cfg: Optional[ServerConfig] = None,
                                          desc=False)
         return doc
 
+    @app.post("/api/v1/sessions/{session_id}/resume")
+    async def resume_session(session_id: str,
+                             user: AuthUser = Depends(auth_dep)):
+        """Re-run the last interaction if it errored mid-stream
+        (reference session_handlers.go:2110)."""
+        s = controller.get_session(session_id)
+        if s is None or s.owner != user.id:
+            raise HTTPException(404, "session not found")
+        its = store.list("interactions", parent=session_id, desc=True,
+                         limit=1)
+        if not its:
+            raise HTTPException(400, "nothing to resume")
+        from helix_amd.server.types import Interaction
+        last = Interaction.model_validate(its[0])
+        if last.state == "complete":
+            return {"ok": True, "note": "already complete"}
+
+        async def sse():
+            yield ("data: " + json.dumps(
+                {"type": "session", "session_id": s.id,
+                 "interaction_id": last.id}) + "\n\n")
+            try:
+                async for chunk in controller.run_session_turn(s, last):
+                    yield f"data: {json.dumps(chunk)}\n\n"
+            except (ProviderError, NoRunnerError) as e:
+                yield ("data: " + json.dumps(
+                    {"error": {"message": str(e)}}) + "\n\n")
+            yield "data: [DONE]\n\n"
+        return StreamingResponse(sse(), media_type="text/event-stream")
+
+    @app.post("/api/v1/sessions/{session_id}/fork")
+    async def fork_session(session_id: str,
+                           user: AuthUser = Depends(auth_dep)):
+        """Duplicate a session with its history (reference fork route)."""
+        s = controller.get_session(session_id)
+        if s is None or s.owner != user.id:
+            raise HTTPException(404, "session not found")
+        ns = controller.create_session(user.id, model=s.model_name,
+                                       provider=s.provider,
+                                       app_id=s.parent_app,
+                                       name=f"{s.name} (fork)")
+        from helix_amd.server.types import Interaction
+        for doc in store.list("interactions", parent=session_id,
+                              desc=False):
+            it = Interaction.model_validate(doc)
+            it.id = f"int_fork_{it.id[-12:]}_{ns.id[-6:]}"
+            it.session_id = ns.id
+            store.put("interactions", it.id, it.model_dump(),
+                      owner=user.id, parent=ns.id)
+        return ns.model_dump()
+
     @app.delete("/api/v1/sessions/{session_id}")
     async def delete_session(session_id: str,
                              user: AuthUser = Depends(auth_dep)):
@@ -350,6 +401,48 @@ def create_app(cfg: Optional[ServerConfig] = None,
                 store.delete("apps", app_id)
                 return {"ok": True}
         _bind()
+
+    # ------------------------------------------------------------------
+    # Local-model admin (reference local-models/load|unload,
+    # server.go:1018-1020): proxied to the runner serving the model.
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/local-models/{model}/load")
+    async def load_local_model(model: str,
+                               user: AuthUser = Depends(admin_dep)):
+        if runner_service is not None:
+            import asyncio as aio
+            await aio.get_event_loop().run_in_executor(
+                None, runner_service.ensure_loaded, model)
+            return {"ok": True, "where": "local"}
+        import httpx
+        errors = {}
+        for r in router.runners():
+            try:
+                async with httpx.AsyncClient(timeout=600) as http:
+                    resp = await http.post(
+                        f"{r.address}/api/v1/models/{model}/load")
+                    if resp.status_code == 200:
+                        return {"ok": True, "where": r.runner_id}
+                    errors[r.runner_id] = resp.text[:200]
+            except Exception as e:
+                errors[r.runner_id] = str(e)
+        raise HTTPException(503, f"no runner could load {model}: {errors}")
+
+    @app.post("/api/v1/local-models/{model}/unload")
+    async def unload_local_model(model: str,
+                                 user: AuthUser = Depends(admin_dep)):
+        if runner_service is not None:
+            runner_service.unload(model)
+            return {"ok": True}
+        import httpx
+        for r in router.runners():
+            try:
+                async with httpx.AsyncClient(timeout=60) as http:
+                    await http.post(
+                        f"{r.address}/api/v1/models/{model}/unload")
+            except Exception:
+                pass
+        return {"ok": True}
 
     # ------------------------------------------------------------------
     # Knowledge API (reference server.go knowledge routes + reconciler)

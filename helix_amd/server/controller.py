@@ -235,6 +235,7 @@ class Controller:
         t0 = time.monotonic()
         text = ""
         first_ms = 0
+        last_persist = t0
         try:
             async for chunk in self.chat_completion_stream(
                     req, session.owner, app_id=session.parent_app, ctx=ctx):
@@ -245,6 +246,16 @@ class Controller:
                 if delta and not first_ms:
                     first_ms = int((time.monotonic() - t0) * 1000)
                 text += delta
+                # 200 ms partial-response write throttle (reference
+                # wsprotocol accumulator, design/2026-02-25: O(N) deltas
+                # on the wire, throttled DB writes — not per-token)
+                now = time.monotonic()
+                if delta and now - last_persist >= 0.2:
+                    interaction.response_message = text
+                    self.store.put("interactions", interaction.id,
+                                   interaction.model_dump(),
+                                   owner=session.owner, parent=session.id)
+                    last_persist = now
                 if stream_to_pubsub:
                     await self.pubsub.publish(
                         ps.session_queue(session.owner, session.id),

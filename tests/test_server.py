@@ -228,3 +228,49 @@ def test_startup_recovery_resets_stuck_interactions(stack):
         pass
     doc = store.get("interactions", "int_stuck")
     assert doc["state"] == "error"
+
+
+def test_session_fork(stack):
+    _, client, _, key, store = stack
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "turn one"}]},
+            headers=H(key)) as r:
+        sid = json.loads(next(l for l in r.iter_lines()
+                              if l.startswith("data: "))[6:])["session_id"]
+    r = client.post(f"/api/v1/sessions/{sid}/fork", headers=H(key))
+    fid = r.json()["id"]
+    assert fid != sid
+    body = client.get(f"/api/v1/sessions/{fid}", headers=H(key)).json()
+    assert len(body["interactions"]) == 1
+    assert body["interactions"][0]["prompt_message"] == "turn one"
+
+
+def test_session_resume_after_error(stack):
+    app, client, _, key, store = stack
+    # create a session whose interaction errored mid-stream
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "hello"}]},
+            headers=H(key)) as r:
+        sid = json.loads(next(l for l in r.iter_lines()
+                              if l.startswith("data: "))[6:])["session_id"]
+    its = store.list("interactions", parent=sid)
+    doc = its[0]
+    doc["state"] = "error"
+    doc["response_message"] = ""
+    store.put("interactions", doc["id"], doc, parent=sid)
+    with client.stream("POST", f"/api/v1/sessions/{sid}/resume",
+                       headers=H(key)) as r:
+        assert r.status_code == 200
+        [l for l in r.iter_lines()]
+    doc = store.list("interactions", parent=sid)[0]
+    assert doc["state"] == "complete"
+    assert doc["response_message"] == "mock response"
+
+
+def test_local_model_admin_no_runner(stack):
+    _, client, _, _, _ = stack
+    r = client.post("/api/v1/local-models/llama3-8b/load",
+                    headers=H("admin-key"))
+    assert r.status_code == 503  # no runners registered
