@@ -912,6 +912,25 @@ def create_app(cfg: Optional[ServerConfig] = None,
     async def healthz():
         return {"ok": True}
 
+    @app.get("/debug/stats")
+    async def debug_stats(user: AuthUser = Depends(admin_dep)):
+        # pprof-equivalent quick stats (reference mounts net/http/pprof,
+        # server.go:1473)
+        import threading as _t
+
+        import psutil
+        p = psutil.Process()
+        return {
+            "rss_bytes": p.memory_info().rss,
+            "num_threads": p.num_threads(),
+            "cpu_percent": p.cpu_percent(interval=0.0),
+            "open_files": len(p.open_files()),
+            "threads": [t.name for t in _t.enumerate()][:50],
+            "store_counts": {t: store.count(t) for t in
+                             ("sessions", "interactions", "llm_calls",
+                              "apps", "knowledge")},
+        }
+
     @app.get("/")
     async def index():
         from fastapi.responses import HTMLResponse

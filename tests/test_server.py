@@ -274,3 +274,24 @@ def test_local_model_admin_no_runner(stack):
     r = client.post("/api/v1/local-models/llama3-8b/load",
                     headers=H("admin-key"))
     assert r.status_code == 503  # no runners registered
+
+
+def test_ws_user_stream(stack):
+    _, client, _, key, _ = stack
+    with client.websocket_connect(f"/api/v1/ws/user?access_token={key}") as ws:
+        # drive a session turn in another request; WS receives events
+        with client.stream("POST", "/api/v1/sessions/chat", json={
+            "model": "mock-model",
+            "messages": [{"role": "user", "content": "ws test"}]},
+                headers=H(key)) as r:
+            [l for l in r.iter_lines()]
+        msg = ws.receive_json()
+        assert msg["payload"]["type"] in ("chunk", "done", "step_info")
+
+
+def test_debug_stats(stack):
+    _, client, _, key, _ = stack
+    assert client.get("/debug/stats", headers=H(key)).status_code == 403
+    r = client.get("/debug/stats", headers=H("admin-key"))
+    assert r.status_code == 200
+    assert r.json()["rss_bytes"] > 0
