@@ -13,9 +13,10 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ROOT = os.path.dirname(os.path.abspath(__file__))
 HIP_DIR = os.path.join(ROOT, "helix_amd", "ops", "hip")
 
-sources = sorted(glob.glob(os.path.join(HIP_DIR, "*.hip"))) + [
-    os.path.join(HIP_DIR, "bindings.cpp")
-]
+sources = sorted(
+    f for f in glob.glob(os.path.join(HIP_DIR, "*.hip"))
+    if not f.endswith("_hip.hip")       # torch-hipify build artifacts
+) + [os.path.join(HIP_DIR, "bindings.cpp")]
 
 setup(
     name="helix_amd_C",
