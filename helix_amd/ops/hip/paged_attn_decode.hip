@@ -95,56 +95,33 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
   for (int base = p_start; base < p_end; base += CHUNK) {
     const int chunk_n = min(CHUNK, p_end - base);
 
-    // --- Phase A: scores ---------------------------------------------------
-    // LPT lanes cooperate on one K row so each 64B cache line is consumed
-    // by temporally-adjacent loads of one lane (the 1-lane-per-row layout
-    // thrashed L1 across 16 resident waves: TCC saw 2.3x logical bytes).
-    {
-      constexpr int LPT = (DHEAD * 2) / 64;       // lanes per token row
-      constexpr int TPP = NTHREADS / LPT;         // tokens per pass
-      const int sub = threadIdx.x % LPT;          // my 64B quarter
-      const int tok_off = threadIdx.x / LPT;
-#pragma unroll (G <= 2 ? 4 : (G <= 4 ? 2 : 1))
-      for (int pass = 0; pass < CHUNK / TPP; ++pass) {
-        const int ti = pass * TPP + tok_off;
-        float sg[G];
+    // --- Phase A: scores (batched K-row loads, then unrolled math) -------
+    if ((int)threadIdx.x < chunk_n) {
+      const int tok = base + threadIdx.x;
+      const int64_t blk = btable[tok / block_size];
+      const uint16_t* krow =
+          k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
+                      tok % block_size)) * DHEAD;
+      u16x8 kraw[DHEAD / 8];
 #pragma unroll
-        for (int g = 0; g < G; ++g) sg[g] = 0.f;
-        if (ti < chunk_n) {
-          const int tok = base + ti;
-          const int64_t blk = btable[tok / block_size];
-          const uint16_t* krow =
-              k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                          tok % block_size)) * DHEAD + sub * 32;
-          u16x8 kraw[4];
+      for (int j = 0; j < DHEAD / 8; ++j)
+        kraw[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+      float s[G];
 #pragma unroll
-          for (int q8 = 0; q8 < 4; ++q8)
-            kraw[q8] = *reinterpret_cast<const u16x8*>(krow + q8 * 8);
+      for (int g = 0; g < G; ++g) s[g] = 0.f;
 #pragma unroll
-          for (int q8 = 0; q8 < 4; ++q8) {
-            float kv[8];
+      for (int j = 0; j < DHEAD / 8; ++j) {
+        float kv[8];
 #pragma unroll
-            for (int i = 0; i < 8; ++i) kv[i] = bf16_to_f32(kraw[q8][i]);
+        for (int i = 0; i < 8; ++i) kv[i] = bf16_to_f32(kraw[j][i]);
 #pragma unroll
-            for (int g = 0; g < G; ++g) {
+        for (int g = 0; g < G; ++g) {
 #pragma unroll
-              for (int i = 0; i < 8; ++i)
-                sg[g] += q_lds[g][sub * 32 + q8 * 8 + i] * kv[i];
-            }
-          }
-        }
-        // fold the LPT partial dots
-#pragma unroll
-        for (int off = 1; off < LPT; off <<= 1) {
-#pragma unroll
-          for (int g = 0; g < G; ++g)
-            sg[g] += __shfl_xor(sg[g], off, WAVE);
-        }
-        if (sub == 0 && ti < chunk_n) {
-#pragma unroll
-          for (int g = 0; g < G; ++g) s_lds[g][ti] = sg[g];
+          for (int i = 0; i < 8; ++i) s[g] += q_lds[g][j * 8 + i] * kv[i];
         }
       }
+#pragma unroll
+      for (int g = 0; g < G; ++g) s_lds[g][threadIdx.x] = s[g];
     }
     __syncthreads();
 
