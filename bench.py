@@ -62,7 +62,9 @@ def main():
         model=args.model,
         max_num_seqs=args.batch,
         max_model_len=max_len,
-        max_prefill_tokens=args.batch * args.prompt_len,
+        # chunked prefill admission (16k-token waves): earlier requests
+        # start decoding sooner => TTFT p50 ~ halves vs one giant prefill
+        max_prefill_tokens=16384,
         eos_token_id=-1,
         seed=rank,
     )

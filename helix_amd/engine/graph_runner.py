@@ -12,7 +12,8 @@ from typing import Dict, Tuple
 
 import torch
 
-BATCH_BUCKETS = [1, 2, 4, 8, 16, 32, 48, 64, 96, 128, 192, 256]
+BATCH_BUCKETS = [1, 2, 4, 8, 16, 32, 48, 64, 96, 128, 192, 256,
+                 384, 512]
 MIN_LEN_BUCKET = 512
 
 
