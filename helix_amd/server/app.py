@@ -98,6 +98,9 @@ def create_app(cfg: Optional[ServerConfig] = None,
     spec_tasks = SpecTaskService(store, controller, git_svc)
     code_intel = CodeIntelService(rag, git_svc)
     notifications = NotificationService(store=store)
+    from helix_amd.server.evaluations import EvaluationService
+    evaluations = EvaluationService(store, controller, pubsub)
+    app.state.evaluations = evaluations
 
     app.state.cfg = cfg
     app.state.store = store
@@ -875,6 +878,35 @@ def create_app(cfg: Optional[ServerConfig] = None,
                            user: AuthUser = Depends(auth_dep)):
         body = await request.json()
         return spec_tasks.add_comment(tid, user.id, body.get("text", ""))
+
+    # ------------------------------------------------------------------
+    # Evaluation suites / runs (reference agent_routes evaluation API)
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/apps/{app_id}/evaluation-suites")
+    async def create_suite(app_id: str, request: Request,
+                           user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return evaluations.create_suite(user.id, app_id,
+                                        body.get("name", ""),
+                                        body.get("tests", []))
+
+    @app.get("/api/v1/evaluation-suites")
+    async def list_suites(user: AuthUser = Depends(auth_dep)):
+        return evaluations.list_suites(user.id)
+
+    @app.post("/api/v1/evaluation-suites/{sid}/runs")
+    async def run_suite(sid: str, user: AuthUser = Depends(auth_dep)):
+        try:
+            return await evaluations.run_suite(sid)
+        except KeyError:
+            raise HTTPException(404, "suite not found")
+
+    @app.get("/api/v1/evaluation-runs/{rid}")
+    async def get_run(rid: str, user: AuthUser = Depends(auth_dep)):
+        run = evaluations.get_run(rid)
+        if run is None:
+            raise HTTPException(404, "run not found")
+        return run
 
     @app.get("/api/v1/git/repos")
     async def list_repos(user: AuthUser = Depends(auth_dep)):

@@ -121,3 +121,30 @@ def test_code_chunker():
     assert len(chunks) == 3
     assert chunks[0]["metadata"]["start_line"] == 1
     assert chunks[1]["metadata"]["start_line"] == 51
+
+
+def test_evaluation_suite_run(stack):
+    app, client, key, store = stack
+    # app with a judged test; EmbedMock returns "# Spec..." -> judge says
+    # NO unless we script; use a mock that answers then judges YES
+    from helix_amd.server.providers import MockClient
+    scripted = MockClient(responses=["the answer is 42", "YES — correct"])
+    app.state.providers.register("mock", scripted)
+    r = client.post("/api/v1/apps", json={"config": {
+        "name": "eval app",
+        "assistants": [{"name": "a", "model": "mock-model",
+                        "provider": "mock"}]}}, headers=H(key))
+    app_id = r.json()["id"]
+    r = client.post(f"/api/v1/apps/{app_id}/evaluation-suites", json={
+        "name": "smoke", "tests": [{"name": "t1", "steps": [
+            {"prompt": "what is 6*7", "expected_output": "42"}]}]},
+        headers=H(key))
+    sid = r.json()["id"]
+    r = client.post(f"/api/v1/evaluation-suites/{sid}/runs", headers=H(key))
+    run = r.json()
+    assert run["state"] == "complete"
+    assert run["passed"] == 1 and run["total"] == 1
+    assert run["results"][0]["passed"] is True
+    # persisted and fetchable
+    r = client.get(f"/api/v1/evaluation-runs/{run['id']}", headers=H(key))
+    assert r.json()["passed"] == 1
