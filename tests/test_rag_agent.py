@@ -18,13 +18,15 @@ class EmbedMock(MockClient):
     similar vectors (enough for ranking tests)."""
 
     async def embeddings(self, req):
+        import zlib
         inputs = req.get("input")
         inputs = [inputs] if isinstance(inputs, str) else list(inputs)
         data = []
         for i, text in enumerate(inputs):
             v = [0.0] * 16
             for w in text.lower().split():
-                v[hash(w) % 16] += 1.0
+                # deterministic hash (python str hash is per-process seeded)
+                v[zlib.crc32(w.encode()) % 16] += 1.0
             data.append({"object": "embedding", "index": i, "embedding": v})
         return {"object": "list", "data": data, "model": req.get("model")}
 
