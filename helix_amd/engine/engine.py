@@ -226,21 +226,52 @@ class LLMEngine:
             self.running.append(seq)
         return list(zip(batch, tokens))
 
+    def _preempt_for_blocks(self, need: int) -> bool:
+        """Free KV by preempting the most recent running sequence back to
+        waiting (recompute-on-readmit, vLLM-style). Returns True if any
+        sequence was preempted."""
+        if len(self.running) <= 1:
+            return False
+        victim = self.running.pop()          # newest first
+        self.kv.allocator.free(victim.block_table)
+        victim.block_table = []
+        # re-admit with its generated tokens folded into the prompt so the
+        # next prefill recomputes the full context
+        victim.prompt_ids = victim.prompt_ids + victim.output_ids
+        victim.output_ids = []
+        victim.status = SeqStatus.WAITING
+        self.waiting.insert(0, victim)
+        return True
+
     def _run_decode(self):
         bs = self.cfg.block_size
-        batch = self.running
         dev = self.device
+        batch = []
         input_ids, positions, slots, seq_lens = [], [], [], []
-        for seq in batch:
+        for seq in list(self.running):
+            if seq not in self.running:
+                continue                      # preempted below
             pos = seq.total_len - 1          # position of the new input token
             nblk = pos // bs + 1
             while len(seq.block_table) < nblk:
+                if not self.kv.allocator.can_allocate(1):
+                    if not self._preempt_for_blocks(1):
+                        raise RuntimeError(
+                            "KV cache exhausted and nothing to preempt")
+                    if seq not in self.running:
+                        break                 # we preempted ourselves
+                    continue
                 seq.block_table.extend(self.kv.allocator.allocate(1))
+            if seq not in self.running:
+                continue
+            batch.append(seq)
             input_ids.append(seq.output_ids[-1] if seq.output_ids
                              else seq.prompt_ids[-1])
             positions.append(pos)
             slots.append(self._slot(seq, pos))
             seq_lens.append(pos + 1)
+        if not batch:
+            return []
         max_blocks = max(len(s.block_table) for s in batch)
         bt_cpu = torch.zeros(len(batch), max_blocks, dtype=torch.int32)
         for i, seq in enumerate(batch):
