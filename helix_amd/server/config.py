@@ -120,6 +120,10 @@ class ServerConfig:
     # agent loop cap (reference agent.go:26 maxIterations default 10)
     agent_max_iterations: int = field(
         default_factory=lambda: _env_int("HELIX_AGENT_MAX_ITERATIONS", 10))
+    # per-user daily token quota per provider (0 = unlimited; reference
+    # api/pkg/quota + controller/balance_check.go)
+    daily_token_limit: int = field(
+        default_factory=lambda: _env_int("HELIX_DAILY_TOKEN_LIMIT", 0))
 
 
 def load_config() -> ServerConfig:

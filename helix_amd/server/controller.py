@@ -29,6 +29,18 @@ class Controller:
         self.agent_runner = agent_runner
         self.usage = usage
 
+    def _check_quota(self, owner: str, provider: str):
+        if self.usage is None:
+            return
+        limit = getattr(self.cfg, "daily_token_limit", 0)
+        if limit <= 0:
+            return
+        from helix_amd.server.usage import QuotaExceededError
+        try:
+            self.usage.check_quota(owner, provider, limit)
+        except QuotaExceededError as e:
+            raise ProviderError(str(e), 429)
+
     def _logging(self, client):
         from helix_amd.server.providers import LoggingClient
         logger = self.usage.log_call if self.usage is not None else None
@@ -119,6 +131,10 @@ class Controller:
             v = getattr(assistant, field)
             if v is not None and out.get(field) is None:
                 out[field] = v
+        # reasoning-effort gating (reference llm_client.go:46: only pass
+        # the field when the assistant configures it)
+        if assistant.reasoning_effort and not out.get("reasoning_effort"):
+            out["reasoning_effort"] = assistant.reasoning_effort
         return out
 
     def _resolve(self, req: dict, owner: str,
@@ -149,6 +165,7 @@ class Controller:
             req["messages"] = await self.enrich_with_knowledge(
                 assistant, req["messages"], owner)
         provider, model = self._resolve(req, owner, assistant)
+        self._check_quota(owner, provider)
         client = self.providers.get_client(provider, owner)
         call_req = {**req, "model": model, "_ctx": ctx or {"owner": owner}}
         return await self._logging(client).chat(call_req)
@@ -171,6 +188,7 @@ class Controller:
             req["messages"] = await self.enrich_with_knowledge(
                 assistant, req["messages"], owner)
         provider, model = self._resolve(req, owner, assistant)
+        self._check_quota(owner, provider)
         client = self.providers.get_client(provider, owner)
         call_req = {**req, "model": model, "stream": True,
                     "_ctx": ctx or {"owner": owner}}

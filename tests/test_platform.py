@@ -192,3 +192,32 @@ def test_weight_io_hf_names(tmp_path):
     assert torch.equal(fused[:q], qw)
     assert torch.equal(fused[q:q + kv], kw)
     assert torch.equal(fused[q + kv:], vw)
+
+
+def test_quota_enforced(stack):
+    app, client, _, key, store = stack
+    app.state.cfg.daily_token_limit = 10
+    # first call consumes 12 tokens (mock usage), second must 429
+    r = client.post("/v1/chat/completions", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "a"}]}, headers=H(key))
+    assert r.status_code == 200
+    r = client.post("/v1/chat/completions", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "b"}]}, headers=H(key))
+    assert r.status_code == 429
+    assert "quota" in r.json()["error"]["message"]
+
+
+def test_reasoning_effort_passthrough(stack):
+    app, client, mock, key, _ = stack
+    r = client.post("/api/v1/apps", json={"config": {
+        "name": "re app",
+        "assistants": [{"name": "a", "model": "mock-model",
+                        "provider": "mock",
+                        "reasoning_effort": "high"}]}}, headers=H(key))
+    app_id = r.json()["id"]
+    client.post("/v1/chat/completions", json={
+        "app_id": app_id,
+        "messages": [{"role": "user", "content": "x"}]}, headers=H(key))
+    assert mock.calls[-1]["reasoning_effort"] == "high"
