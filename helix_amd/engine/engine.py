@@ -226,12 +226,12 @@ class LLMEngine:
             self.running.append(seq)
         return list(zip(batch, tokens))
 
-    def _preempt_for_blocks(self, need: int) -> bool:
+    def _preempt_for_blocks(self):
         """Free KV by preempting the most recent running sequence back to
-        waiting (recompute-on-readmit, vLLM-style). Returns True if any
-        sequence was preempted."""
+        waiting (recompute-on-readmit, vLLM-style). Returns the victim or
+        None."""
         if len(self.running) <= 1:
-            return False
+            return None
         victim = self.running.pop()          # newest first
         self.kv.allocator.free(victim.block_table)
         victim.block_table = []
@@ -241,28 +241,31 @@ class LLMEngine:
         victim.output_ids = []
         victim.status = SeqStatus.WAITING
         self.waiting.insert(0, victim)
-        return True
+        return victim
 
     def _run_decode(self):
         bs = self.cfg.block_size
         dev = self.device
         batch = []
+        preempted = set()                     # ids preempted this step
         input_ids, positions, slots, seq_lens = [], [], [], []
         for seq in list(self.running):
-            if seq not in self.running:
-                continue                      # preempted below
+            if seq.seq_id in preempted:
+                continue
             pos = seq.total_len - 1          # position of the new input token
             nblk = pos // bs + 1
             while len(seq.block_table) < nblk:
                 if not self.kv.allocator.can_allocate(1):
-                    if not self._preempt_for_blocks(1):
+                    victim = self._preempt_for_blocks()
+                    if victim is None:
                         raise RuntimeError(
                             "KV cache exhausted and nothing to preempt")
-                    if seq not in self.running:
+                    preempted.add(victim.seq_id)
+                    if victim is seq:
                         break                 # we preempted ourselves
                     continue
                 seq.block_table.extend(self.kv.allocator.allocate(1))
-            if seq not in self.running:
+            if seq.seq_id in preempted:
                 continue
             batch.append(seq)
             input_ids.append(seq.output_ids[-1] if seq.output_ids
