@@ -101,6 +101,9 @@ def create_app(cfg: Optional[ServerConfig] = None,
     from helix_amd.server.evaluations import EvaluationService
     evaluations = EvaluationService(store, controller, pubsub)
     app.state.evaluations = evaluations
+    from helix_amd.server.mcp_gateway import MCPGateway
+    mcp = MCPGateway(store, agent_runner, code_intel)
+    app.state.mcp = mcp
 
     app.state.cfg = cfg
     app.state.store = store
@@ -878,6 +881,20 @@ def create_app(cfg: Optional[ServerConfig] = None,
                            user: AuthUser = Depends(auth_dep)):
         body = await request.json()
         return spec_tasks.add_comment(tid, user.id, body.get("text", ""))
+
+    # ------------------------------------------------------------------
+    # MCP gateway (reference mcp gateway routes): JSON-RPC over POST
+    # ------------------------------------------------------------------
+    @app.post("/api/v1/mcp/{app_id}")
+    async def mcp_endpoint(app_id: str, request: Request,
+                           user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        if isinstance(body, list):  # JSON-RPC batch
+            out = [await mcp.handle(app_id, user.id, r) for r in body]
+            return [r for r in out if r is not None]
+        resp = await mcp.handle(app_id, user.id, body)
+        return resp if resp is not None else JSONResponse(status_code=202,
+                                                          content=None)
 
     # ------------------------------------------------------------------
     # Evaluation suites / runs (reference agent_routes evaluation API)

@@ -148,3 +148,33 @@ def test_evaluation_suite_run(stack):
     # persisted and fetchable
     r = client.get(f"/api/v1/evaluation-runs/{run['id']}", headers=H(key))
     assert r.json()["passed"] == 1
+
+
+def test_mcp_gateway(stack):
+    app, client, key, store = stack
+    r = client.post("/api/v1/apps", json={"config": {
+        "name": "mcp app",
+        "assistants": [{"name": "a", "model": "mock-model",
+                        "provider": "mock",
+                        "calculator": {"enabled": True}}]}}, headers=H(key))
+    app_id = r.json()["id"]
+    # initialize
+    r = client.post(f"/api/v1/mcp/{app_id}", json={
+        "jsonrpc": "2.0", "id": 1, "method": "initialize",
+        "params": {}}, headers=H(key))
+    assert r.json()["result"]["serverInfo"]["name"] == "helix_amd"
+    # tools/list
+    r = client.post(f"/api/v1/mcp/{app_id}", json={
+        "jsonrpc": "2.0", "id": 2, "method": "tools/list"}, headers=H(key))
+    tools = r.json()["result"]["tools"]
+    assert any(t["name"] == "calculator" for t in tools)
+    # tools/call
+    r = client.post(f"/api/v1/mcp/{app_id}", json={
+        "jsonrpc": "2.0", "id": 3, "method": "tools/call",
+        "params": {"name": "calculator",
+                   "arguments": {"expression": "6*7"}}}, headers=H(key))
+    assert r.json()["result"]["content"][0]["text"] == "42"
+    # unknown method
+    r = client.post(f"/api/v1/mcp/{app_id}", json={
+        "jsonrpc": "2.0", "id": 4, "method": "bogus"}, headers=H(key))
+    assert r.json()["error"]["code"] == -32601
