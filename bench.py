@@ -29,7 +29,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=32)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--batch", type=int, default=256)
+    ap.add_argument("--batch", type=int, default=512)
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--model", type=str, default="llama3-8b")
     ap.add_argument("--parallelism", type=str, default="dp",

@@ -160,6 +160,7 @@ async def _stream(service, inst, model, rid, created, prompt_ids, params,
     inst.submit(rid, prompt_ids, params, ts.on_token)
     detok = StreamDetokenizer(tok)
     ntok = 0
+    finished = False
 
     def chunk(delta: dict, finish: Optional[str] = None, usage=None):
         c = {
@@ -174,31 +175,40 @@ async def _stream(service, inst, model, rid, created, prompt_ids, params,
             c["usage"] = usage
         return c
 
-    yield chunk({"role": "assistant", "content": ""})
-    emitted = 0
-    finish_reason = "stop"
-    async for token_id, fin, reason in ts.__aiter__():
-        ntok += 1
-        delta = detok.push(token_id)
-        hit = _find_stop(detok.emitted, stop_strs)
-        if hit is not None:
+    try:
+        yield chunk({"role": "assistant", "content": ""})
+        emitted = 0
+        finish_reason = "stop"
+        async for token_id, fin, reason in ts.__aiter__():
+            ntok += 1
+            delta = detok.push(token_id)
+            hit = _find_stop(detok.emitted, stop_strs)
+            if hit is not None:
+                inst.cancel(rid)
+                keep = max(0, hit - emitted)
+                if keep:
+                    yield chunk({"content": delta[:keep]})
+                finish_reason = "stop"
+                finished = True
+                break
+            if delta:
+                emitted += len(delta)
+                yield chunk({"content": delta})
+            if fin:
+                finish_reason = reason or "stop"
+                finished = True
+                break
+        yield chunk({}, finish=finish_reason, usage={
+            "prompt_tokens": len(prompt_ids),
+            "completion_tokens": ntok,
+            "total_tokens": len(prompt_ids) + ntok,
+        })
+    finally:
+        # Client disconnect / generator close mid-stream: stop decoding
+        # (the reference's cancellation-correctness requirement —
+        # helix_openai_server.go:285-293 conn hard-close semantics).
+        if not finished:
             inst.cancel(rid)
-            keep = max(0, hit - emitted)
-            if keep:
-                yield chunk({"content": delta[:keep]})
-            finish_reason = "stop"
-            break
-        if delta:
-            emitted += len(delta)
-            yield chunk({"content": delta})
-        if fin:
-            finish_reason = reason or "stop"
-            break
-    yield chunk({}, finish=finish_reason, usage={
-        "prompt_tokens": len(prompt_ids),
-        "completion_tokens": ntok,
-        "total_tokens": len(prompt_ids) + ntok,
-    })
 
 
 async def embeddings(service: RunnerService, req: dict) -> dict:

@@ -136,3 +136,35 @@ def test_apply_profile_reconciles(service):
          "kv_cache_blocks": 64}]})
     assert set(res["unloaded"]) == {"tiny", "tiny-bert"}
     assert service.loaded_models() == ["tiny-gqa"]
+
+
+def test_stream_disconnect_cancels_sequence(service):
+    """Dropping the SSE stream mid-generation must cancel the engine
+    sequence (no orphan decode burning KV/compute)."""
+    import asyncio
+    from helix_amd.runner.openai_adapter import chat_completion
+
+    async def run():
+        it = await chat_completion(service, {
+            "model": "tiny", "stream": True, "temperature": 0,
+            "max_tokens": 200,
+            "messages": [{"role": "user", "content": "go"}]})
+        count = 0
+        async for _ in it:
+            count += 1
+            if count >= 3:
+                await it.aclose()      # simulate client disconnect
+                break
+        return count
+
+    asyncio.run(run())
+    inst = service.instances["tiny"]
+    # wait for the engine loop to drain the cancelled sequence
+    import time
+    for _ in range(100):
+        if inst.in_flight == 0:
+            break
+        time.sleep(0.05)
+    assert inst.in_flight == 0
+    seqs = inst.engine.seqs
+    assert any(s.status.value == "cancelled" for s in seqs.values())
