@@ -65,8 +65,10 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   const int lane_hi = lane >> 4;   // 0..3
   const int lane_lo = lane & 15;   // 0..15
 
-  __shared__ uint16_t k_lds[KTILE * DMAX];
-  __shared__ uint16_t v_lds[DMAX * KTILE];          // transposed
+  // double-buffered K/V so the main loop needs ONE barrier per tile
+  // instead of two (store for tile t+1 overlaps nothing that reads it)
+  __shared__ uint16_t k_lds[2][KTILE * DMAX];
+  __shared__ uint16_t v_lds[2][DMAX * KTILE];       // transposed
   __shared__ uint16_t p_lds[4][16 * P_PITCH];
 
   // ---- Q fragments (A-operand layout), 16 rows per wave ----
@@ -122,28 +124,28 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
       }
     }
   };
-  auto store_tile = [&]() {
+  auto store_tile = [&](int buf) {
 #pragma unroll
     for (int cch = 0; cch < nchunk; ++cch) {
       const int e = (cch * 256 + tid) * 8;
       const int tok = e / D;
       const int d0 = e % D;
-      *reinterpret_cast<u16x8*>((char*)k_lds + k_swz(tok, d0 * 2)) =
+      *reinterpret_cast<u16x8*>((char*)k_lds[buf] + k_swz(tok, d0 * 2)) =
           k_reg[cch];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        *reinterpret_cast<uint16_t*>((char*)v_lds + v_swz(d0 + i, tok * 2)) =
-            v_reg[cch][i];
+        *reinterpret_cast<uint16_t*>(
+            (char*)v_lds[buf] + v_swz(d0 + i, tok * 2)) = v_reg[cch][i];
       }
     }
   };
 
   load_tile(0);
+  store_tile(0);
+  if (ntiles > 1) load_tile(1);
   for (int t = 0; t < ntiles; ++t) {
-    __syncthreads();            // compute on previous LDS contents done
-    store_tile();
-    __syncthreads();
-    if (t + 1 < ntiles) load_tile(t + 1);   // prefetch under compute
+    const int buf = t & 1;
+    __syncthreads();            // buf's staging (and last compute) done
 
     const int kv_base = t * KTILE;
     // ---- QK^T: 4 halves of 16 tokens ----
@@ -156,8 +158,8 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 #pragma unroll
       for (int kt = 0; kt < nkt; ++kt) {
         const int byte = (kt * 32 + lane_hi * 8) * 2;
-        u16x8 raw =
-            *reinterpret_cast<const u16x8*>((char*)k_lds + k_swz(tok, byte));
+        u16x8 raw = *reinterpret_cast<const u16x8*>(
+            (char*)k_lds[buf] + k_swz(tok, byte));
         s_frag[h] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             q_frag[kt], *reinterpret_cast<bf16x8*>(&raw), s_frag[h], 0, 0, 0);
       }
@@ -222,11 +224,17 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
         // B-operand from transposed V: row dim = c*16+lane_lo, toks
         // kk*32 + lane_hi*8 .. +7 — one contiguous swizzled 16B read.
         u16x8 raw = *reinterpret_cast<const u16x8*>(
-            (char*)v_lds +
+            (char*)v_lds[buf] +
             v_swz(c * 16 + lane_lo, (kk * 32 + lane_hi * 8) * 2));
         o_acc[c] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             p_frag[kk], *reinterpret_cast<bf16x8*>(&raw), o_acc[c], 0, 0, 0);
       }
+    }
+    // stage tile t+1 into the other buffer (nobody reads it this iter);
+    // prefetch tile t+2 loads afterwards so they fly during the barrier
+    if (t + 1 < ntiles) {
+      store_tile(buf ^ 1);
+      if (t + 2 < ntiles) load_tile(t + 2);
     }
   }
 
