@@ -8,6 +8,7 @@ runner pushes, see heartbeat.py).
 """
 from __future__ import annotations
 
+import collections
 import json
 import logging
 from typing import Optional
@@ -23,11 +24,31 @@ from helix_amd.runner.service import (ModelNotFoundError, NoCapacityError,
 log = logging.getLogger("helix_amd.runner.http")
 
 
+class RingBufferHandler(logging.Handler):
+    """In-memory log ring (parity with hydra's logbuf + the admin
+    runner-logs surface, reference design/2026-05-29)."""
+
+    def __init__(self, capacity: int = 2000):
+        super().__init__()
+        self.buf = collections.deque(maxlen=capacity)
+        self.setFormatter(logging.Formatter(
+            "%(asctime)s %(levelname)s %(name)s %(message)s"))
+
+    def emit(self, record):
+        try:
+            self.buf.append(self.format(record))
+        except Exception:
+            pass
+
+
 def create_runner_app(service: RunnerService,
                       runner_id: str = "runner-0") -> FastAPI:
     app = FastAPI(title="helix_amd runner", docs_url=None)
     app.state.service = service
     app.state.runner_id = runner_id
+    logbuf = RingBufferHandler()
+    logging.getLogger("helix_amd").addHandler(logbuf)
+    app.state.logbuf = logbuf
 
     @app.exception_handler(ModelNotFoundError)
     async def _nf(request, exc):
@@ -77,6 +98,10 @@ def create_runner_app(service: RunnerService,
             "gpus": [g.model_dump() for g in gpudetect.detect()],
             "models": service.status(),
         }
+
+    @app.get("/api/v1/logs")
+    async def logs(n: int = 200):
+        return {"lines": list(logbuf.buf)[-n:]}
 
     @app.post("/api/v1/models/{model}/load")
     async def load(model: str):

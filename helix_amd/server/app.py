@@ -303,6 +303,34 @@ def create_app(cfg: Optional[ServerConfig] = None,
             yield "data: [DONE]\n\n"
         return StreamingResponse(sse(), media_type="text/event-stream")
 
+    @app.put("/api/v1/sessions/{session_id}/agent")
+    async def switch_agent(session_id: str, request: Request,
+                           user: AuthUser = Depends(auth_dep)):
+        """Switch the app/agent mid-session, history carried across
+        (reference server.go:1042 switch-agent)."""
+        s = controller.get_session(session_id)
+        if s is None or s.owner != user.id:
+            raise HTTPException(404, "session not found")
+        body = await request.json()
+        new_app = body.get("app_id", "")
+        if new_app and store.get("apps", new_app) is None:
+            raise HTTPException(404, "app not found")
+        s.parent_app = new_app
+        if body.get("model"):
+            s.model_name = body["model"]
+        store.put("sessions", s.id, s.model_dump(), owner=s.owner,
+                  parent=new_app)
+        return s.model_dump()
+
+    @app.get("/api/v1/sessions/{session_id}/step-info")
+    async def step_info(session_id: str,
+                        user: AuthUser = Depends(auth_dep)):
+        """Agent step trace for a session (reference server.go:1036)."""
+        s = controller.get_session(session_id)
+        if s is None or (s.owner != user.id and not user.admin):
+            raise HTTPException(404, "session not found")
+        return store.list("step_info", parent=session_id, desc=False)
+
     @app.post("/api/v1/sessions/{session_id}/fork")
     async def fork_session(session_id: str,
                            user: AuthUser = Depends(auth_dep)):
@@ -510,6 +538,19 @@ def create_app(cfg: Optional[ServerConfig] = None,
         hb = RunnerHeartbeat.model_validate(await request.json())
         router.on_heartbeat(hb)
         return {"ok": True}
+
+    @app.get("/api/v1/admin/runners/{runner_id}/logs")
+    async def runner_logs(runner_id: str, n: int = 200,
+                          user: AuthUser = Depends(admin_dep)):
+        import httpx
+        state = next((r for r in router.runners()
+                      if r.runner_id == runner_id), None)
+        if state is None:
+            raise HTTPException(404, "runner not found")
+        async with httpx.AsyncClient(timeout=20) as http:
+            r = await http.get(f"{state.address}/api/v1/logs",
+                               params={"n": n})
+            return r.json()
 
     @app.get("/api/v1/admin/runners")
     async def list_runners(user: AuthUser = Depends(admin_dep)):

@@ -295,3 +295,24 @@ def test_debug_stats(stack):
     r = client.get("/debug/stats", headers=H("admin-key"))
     assert r.status_code == 200
     assert r.json()["rss_bytes"] > 0
+
+
+def test_switch_agent_and_step_info(stack):
+    _, client, _, key, store = stack
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "x"}]},
+            headers=H(key)) as r:
+        sid = json.loads(next(l for l in r.iter_lines()
+                              if l.startswith("data: "))[6:])["session_id"]
+    r = client.post("/api/v1/apps", json={"config": {
+        "name": "other", "assistants": [{"name": "a",
+                                         "model": "mock-model",
+                                         "provider": "mock"}]}},
+        headers=H(key))
+    app_id = r.json()["id"]
+    r = client.put(f"/api/v1/sessions/{sid}/agent",
+                   json={"app_id": app_id}, headers=H(key))
+    assert r.json()["parent_app"] == app_id
+    assert client.get(f"/api/v1/sessions/{sid}/step-info",
+                      headers=H(key)).status_code == 200
