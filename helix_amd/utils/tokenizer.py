@@ -60,7 +60,51 @@ class ByteTokenizer:
 
 
 _DEFAULT = ByteTokenizer()
+_CACHE = {}
 
 
-def get_tokenizer(model: str = "") -> ByteTokenizer:
+class HFTokenizer:
+    """Real tokenizer files (tokenizer.json) when a deployment provides
+    them via HELIX_TOKENIZER_DIR/<model>/tokenizer.json — same interface
+    as ByteTokenizer."""
+
+    def __init__(self, path: str):
+        from tokenizers import Tokenizer
+        self._tok = Tokenizer.from_file(path)
+        self.vocab_size = self._tok.get_vocab_size()
+        self.bos_token_id = self._tok.token_to_id("<|begin_of_text|>") or             self._tok.token_to_id("<s>") or 1
+        self.eos_token_id = self._tok.token_to_id("<|end_of_text|>") or             self._tok.token_to_id("</s>") or 2
+
+    def encode(self, text: str, add_bos: bool = False):
+        ids = self._tok.encode(text, add_special_tokens=False).ids
+        return ([self.bos_token_id] + ids) if add_bos else ids
+
+    def decode(self, ids):
+        return self._tok.decode(list(ids), skip_special_tokens=True)
+
+    def apply_chat_template(self, messages, add_generation_prompt=True):
+        parts = []
+        for m in messages:
+            content = m.get("content") or ""
+            if not isinstance(content, str):
+                content = " ".join(p.get("text", "") for p in content
+                                   if isinstance(p, dict))
+            parts.append(f"<|{m.get('role', 'user')}|>\n{content}")
+        if add_generation_prompt:
+            parts.append("<|assistant|>\n")
+        return self.encode("\n".join(parts), add_bos=True)
+
+
+def get_tokenizer(model: str = ""):
+    import os
+    base = os.environ.get("HELIX_TOKENIZER_DIR", "")
+    if base and model:
+        path = os.path.join(base, model, "tokenizer.json")
+        if os.path.exists(path):
+            if path not in _CACHE:
+                try:
+                    _CACHE[path] = HFTokenizer(path)
+                except Exception:
+                    _CACHE[path] = _DEFAULT
+            return _CACHE[path]
     return _DEFAULT

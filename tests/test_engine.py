@@ -144,3 +144,42 @@ def test_kv_exhaustion_preempts_not_crashes():
     # both finish eventually (b preempted/recomputed at least once)
     assert len(eng.seqs["a"].output_ids) == 40
     assert len(eng.seqs["b"].output_ids) == 40
+
+
+def test_concurrent_submit_thread_safety():
+    """Stress the runner instance's engine lock: concurrent submits from
+    many threads while the step loop runs (reference relies on -race CI;
+    here we exercise the locking directly)."""
+    import threading
+    from helix_amd.runner.service import RunnerService
+    svc = RunnerService(device="cpu", memory_budget=64 << 30)
+    try:
+        inst = svc.ensure_loaded("tiny")
+        done = []
+        lock = threading.Lock()
+
+        def on_token(seq, tok, fin):
+            if fin:
+                with lock:
+                    done.append(seq.seq_id)
+
+        sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+
+        def submit(n):
+            for i in range(4):
+                inst.submit(f"t{n}-{i}", [1 + n, 2 + i, 3], sp, on_token)
+
+        threads = [threading.Thread(target=submit, args=(n,))
+                   for n in range(6)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        import time
+        for _ in range(200):
+            if len(done) == 24:
+                break
+            time.sleep(0.05)
+        assert len(done) == 24
+    finally:
+        svc.shutdown()
