@@ -32,7 +32,7 @@ constexpr int MAX_G = 8;
 constexpr int PART_QUANT = 128;
 
 template <int DHEAD, int G>
-__global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
+__global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
     float* __restrict__ tmp_ml,          // [B, Hq, maxP, 2]
@@ -79,13 +79,18 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
   }
   __syncthreads();
 
-  constexpr int PAIRS = DHEAD / 2;
-  constexpr int N_PAR = NTHREADS / PAIRS;   // 1 for D=128, 2 for D=64
-  const int d_own = (threadIdx.x % PAIRS) * 2;
-  const int par = threadIdx.x / PAIRS;
-  float acc[G][2];
+  // Phase-C ownership: a thread owns 8 consecutive dims (one 16B load
+  // per token) — the ablation probe showed b32 V loads were 64% of the
+  // kernel (8x more VMEM instructions per byte than phase A's b128s).
+  constexpr int NGRP = DHEAD / 8;            // dim-groups (16 for D=128)
+  constexpr int C_PAR = NTHREADS / NGRP;     // token parities (8 / 16)
+  const int d8 = (threadIdx.x % NGRP) * 8;   // my dims [d8, d8+8)
+  const int cpar = threadIdx.x / NGRP;
+  float acc[G][8];
 #pragma unroll
-  for (int g = 0; g < G; ++g) acc[g][0] = acc[g][1] = 0.f;
+  for (int g = 0; g < G; ++g)
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[g][i] = 0.f;
 
   const int* btable = block_tables + (int64_t)seq * max_blocks;
   const int nwaves = NTHREADS / WAVE;
@@ -149,93 +154,96 @@ __global__ __launch_bounds__(NTHREADS, 4) void paged_attn_decode_kernel(
     }
     __syncthreads();
 
-    // --- Phase C: V accumulation (8-token load batches) -------------------
+    // --- Phase C: V accumulation (16B lane loads, 4-pass batches) --------
     {
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        acc[g][0] *= head_corr[g];
-        acc[g][1] *= head_corr[g];
-      }
-      const int ntok = (chunk_n - par + N_PAR - 1) / N_PAR;  // my tokens
-      int t8 = 0;
-      for (; t8 + 8 <= ntok; t8 += 8) {
-        uint32_t vv[8];
+      for (int g = 0; g < G; ++g)
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const int tok = base + (t8 + u) * N_PAR + par;
+        for (int i = 0; i < 8; ++i) acc[g][i] *= head_corr[g];
+      const int npass = (chunk_n - cpar + C_PAR - 1) / C_PAR;  // my tokens
+      int ps = 0;
+      for (; ps + 4 <= npass; ps += 4) {
+        u16x8 vv[4];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          const int tok = base + (ps + u) * C_PAR + cpar;
           const int64_t blk = btable[tok / block_size];
-          vv[u] = *reinterpret_cast<const uint32_t*>(
+          vv[u] = *reinterpret_cast<const u16x8*>(
               v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                          tok % block_size)) * DHEAD + d_own);
+                          tok % block_size)) * DHEAD + d8);
         }
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const int tok_i = (t8 + u) * N_PAR + par;
-          const float v0 = bf16_to_f32((uint16_t)(vv[u] & 0xffff));
-          const float v1 = bf16_to_f32((uint16_t)(vv[u] >> 16));
+        for (int u = 0; u < 4; ++u) {
+          const int tok_i = (ps + u) * C_PAR + cpar;
+          float v[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) v[i] = bf16_to_f32(vv[u][i]);
 #pragma unroll
           for (int g = 0; g < G; ++g) {
-            const float p = s_lds[g][tok_i];
-            acc[g][0] += p * v0;
-            acc[g][1] += p * v1;
+            const float pv = s_lds[g][tok_i];
+#pragma unroll
+            for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
           }
         }
       }
-      for (; t8 < ntok; ++t8) {
-        const int tok_i = t8 * N_PAR + par;
+      for (; ps < npass; ++ps) {
+        const int tok_i = ps * C_PAR + cpar;
         const int tok = base + tok_i;
         const int64_t blk = btable[tok / block_size];
-        const uint32_t vv = *reinterpret_cast<const uint32_t*>(
+        const u16x8 vv = *reinterpret_cast<const u16x8*>(
             v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                        tok % block_size)) * DHEAD + d_own);
-        const float v0 = bf16_to_f32((uint16_t)(vv & 0xffff));
-        const float v1 = bf16_to_f32((uint16_t)(vv >> 16));
+                        tok % block_size)) * DHEAD + d8);
+        float v[8];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[i] = bf16_to_f32(vv[i]);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
-          const float p = s_lds[g][tok_i];
-          acc[g][0] += p * v0;
-          acc[g][1] += p * v1;
+          const float pv = s_lds[g][tok_i];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
         }
       }
     }
     __syncthreads();
   }
 
-  // Combine token parities (N_PAR==2 only for D=64) and write out.
-  if (N_PAR > 1) {
-    __shared__ float comb[N_PAR > 1 ? G * NTHREADS * 2 : 1];
-    #pragma unroll
-    for (int g = 0; g < G; ++g) {
-      comb[((par * G + g) * PAIRS + d_own / 2) * 2 + 0] = acc[g][0];
-      comb[((par * G + g) * PAIRS + d_own / 2) * 2 + 1] = acc[g][1];
-    }
-    __syncthreads();
-    if (par == 0) {
+  // Combine the C_PAR token parities via LDS, then write out.
+  __shared__ float comb[G * DHEAD];
 #pragma unroll
-      for (int g = 0; g < G; ++g)
-        for (int p = 1; p < N_PAR; ++p) {
-          acc[g][0] += comb[((p * G + g) * PAIRS + d_own / 2) * 2 + 0];
-          acc[g][1] += comb[((p * G + g) * PAIRS + d_own / 2) * 2 + 1];
-        }
+  for (int g = 0; g < G; ++g) {
+    if (cpar == 0) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) comb[g * DHEAD + d8 + i] = acc[g][i];
     }
   }
-  if (par == 0) {
+  __syncthreads();
+  for (int p = 1; p < C_PAR; ++p) {
+    if (cpar == p) {
+#pragma unroll
+      for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          comb[g * DHEAD + d8 + i] += acc[g][i];
+    }
+    __syncthreads();
+  }
+  if (cpar == 0) {
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       const int hq = hkv * G + g;
       if (nparts == 1) {
         const float inv_l = 1.f / fmaxf(head_l[g], 1e-20f);
-        uint32_t packed =
-            (uint32_t)f32_to_bf16(acc[g][0] * inv_l) |
-            ((uint32_t)f32_to_bf16(acc[g][1] * inv_l) << 16);
-        *reinterpret_cast<uint32_t*>(
-            out + ((int64_t)seq * Hq + hq) * DHEAD + d_own) = packed;
+        float o[8];
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          o[i] = comb[g * DHEAD + d8 + i] * inv_l;
+        store_bf16x8(out + ((int64_t)seq * Hq + hq) * DHEAD + d8, o);
       } else {
         float* tp =
             tmp_out + (((int64_t)seq * Hq + hq) * max_parts + part) * DHEAD;
-        tp[d_own] = acc[g][0];
-        tp[d_own + 1] = acc[g][1];
-        if (d_own == 0) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) tp[d8 + i] = comb[g * DHEAD + d8 + i];
+        if (d8 == 0) {
           float* ml =
               tmp_ml + (((int64_t)seq * Hq + hq) * max_parts + part) * 2;
           ml[0] = head_m[g];

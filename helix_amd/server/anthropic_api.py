@@ -6,7 +6,6 @@ local MI355X runner — so Anthropic SDK clients work against helix_amd).
 from __future__ import annotations
 
 import json
-import time
 import uuid
 from typing import AsyncIterator
 

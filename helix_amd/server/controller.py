@@ -5,7 +5,6 @@ sessions.go).
 """
 from __future__ import annotations
 
-import asyncio
 import logging
 import time
 from typing import Any, AsyncIterator, Dict, List, Optional

@@ -4,7 +4,6 @@ streaming): each suite holds helix.yaml-style multi-turn tests; a run
 executes them against the app's assistant and judges with an LLM."""
 from __future__ import annotations
 
-import asyncio
 import logging
 import time
 from typing import List, Optional
