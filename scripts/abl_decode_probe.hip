@@ -57,11 +57,15 @@ __global__ __launch_bounds__(NT, 3) void decode_abl(
   if (threadIdx.x < G) { hm[threadIdx.x] = -INFINITY; hl[threadIdx.x] = 0.f; }
   __syncthreads();
 
-  const int pairs = DH / 2;
-  const int d_own = (threadIdx.x % pairs) * 2;
-  float acc[G][2];
+  constexpr int NGRP = DH / 8;
+  constexpr int C_PAR = NT / NGRP;
+  const int d8 = (threadIdx.x % NGRP) * 8;
+  const int cpar = threadIdx.x / NGRP;
+  float acc[G][8];
 #pragma unroll
-  for (int g = 0; g < G; ++g) acc[g][0] = acc[g][1] = 0.f;
+  for (int g = 0; g < G; ++g)
+#pragma unroll
+    for (int i = 0; i < 8; ++i) acc[g][i] = 0.f;
   const int* bt = btab + (long)seq * max_blocks;
   const int wid = threadIdx.x / WAVE, lane = threadIdx.x & 63;
   const int nw = NT / WAVE;
@@ -131,52 +135,62 @@ __global__ __launch_bounds__(NT, 3) void decode_abl(
 
     {
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        acc[g][0] *= hc[g];
-        acc[g][1] *= hc[g];
-      }
-      int t8 = 0;
-      for (; t8 + 8 <= cn; t8 += 8) {
-        uint32_t vv[8];
+      for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int i = 0; i < 8; ++i) acc[g][i] *= hc[g];
+      const int npass = (cn - cpar + C_PAR - 1) / C_PAR;
+      int ps = 0;
+      for (; ps + 4 <= npass; ps += 4) {
+        u16x8 vv[4];
         if (ABL & 4) {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) vv[u] = 0x3f803f80u;  // ~1.0 bf16 x2
+          for (int u = 0; u < 4; ++u)
+#pragma unroll
+            for (int i = 0; i < 8; ++i) vv[u][i] = 0x3f80;
         } else {
 #pragma unroll
-          for (int u = 0; u < 8; ++u) {
-            const int tok = base + t8 + u;
+          for (int u = 0; u < 4; ++u) {
+            const int tok = base + (ps + u) * C_PAR + cpar;
             const long blk = bt[tok / bs];
-            vv[u] = *reinterpret_cast<const uint32_t*>(
+            vv[u] = *reinterpret_cast<const u16x8*>(
                 v_cache + ((blk * Hkv + hkv) * (long)bs + tok % bs) * DH +
-                d_own);
+                d8);
           }
         }
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-          const float v0 = bf2f((uint16_t)(vv[u] & 0xffff));
-          const float v1 = bf2f((uint16_t)(vv[u] >> 16));
+        for (int u = 0; u < 4; ++u) {
+          const int ti = (ps + u) * C_PAR + cpar;
+          float v[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) v[i] = bf2f(vv[u][i]);
 #pragma unroll
           for (int g = 0; g < G; ++g) {
-            const float p = s_lds[g][t8 + u];
-            acc[g][0] += p * v0;
-            acc[g][1] += p * v1;
+            const float pv = s_lds[g][ti];
+#pragma unroll
+            for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
           }
         }
       }
-      for (; t8 < cn; ++t8) {
-        const int tok = base + t8;
+      for (; ps < npass; ++ps) {
+        const int ti = ps * C_PAR + cpar;
+        const int tok = base + ti;
         const long blk = bt[tok / bs];
-        uint32_t vv = (ABL & 4) ? 0x3f803f80u
-            : *reinterpret_cast<const uint32_t*>(
-                  v_cache + ((blk * Hkv + hkv) * (long)bs + tok % bs) * DH +
-                  d_own);
-        const float v0 = bf2f((uint16_t)(vv & 0xffff));
-        const float v1 = bf2f((uint16_t)(vv >> 16));
+        u16x8 vv;
+        if (ABL & 4) {
+#pragma unroll
+          for (int i = 0; i < 8; ++i) vv[i] = 0x3f80;
+        } else {
+          vv = *reinterpret_cast<const u16x8*>(
+              v_cache + ((blk * Hkv + hkv) * (long)bs + tok % bs) * DH + d8);
+        }
+        float v[8];
+#pragma unroll
+        for (int i = 0; i < 8; ++i) v[i] = bf2f(vv[i]);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
-          const float p = s_lds[g][t8];
-          acc[g][0] += p * v0;
-          acc[g][1] += p * v1;
+          const float pv = s_lds[g][ti];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
         }
       }
     }
@@ -187,11 +201,11 @@ __global__ __launch_bounds__(NT, 3) void decode_abl(
   for (int g = 0; g < G; ++g) {
     const int hq = hkv * G + g;
     const float inv = 1.f / fmaxf(hl[g], 1e-20f);
-    // write raw fp32 bits truncated — numerics unused; keep acc live
-    out[((long)seq * Hq + hq) * DH + d_own] =
-        (uint16_t)(__float_as_uint(acc[g][0] * inv) >> 16);
-    out[((long)seq * Hq + hq) * DH + d_own + 1] =
-        (uint16_t)(__float_as_uint(acc[g][1] * inv) >> 16);
+    if (cpar == 0)
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        out[((long)seq * Hq + hq) * DH + d8 + i] =
+            (uint16_t)(__float_as_uint(acc[g][i] * inv) >> 16);
   }
 }
 
