@@ -748,6 +748,30 @@ def create_app(cfg: Optional[ServerConfig] = None,
     # ------------------------------------------------------------------
     # Usage / wallet (reference usage_metrics + wallets)
     # ------------------------------------------------------------------
+    @app.get("/api/v1/apps/{app_id}/usage")
+    async def app_usage(app_id: str, user: AuthUser = Depends(auth_dep),
+                        day: str = ""):
+        """Per-app daily usage (reference server.go:1021-1022): rolls up
+        llm_calls whose session belongs to the app."""
+        sessions = {s["id"] for s in store.list("sessions", parent=app_id,
+                                                limit=100000)}
+        out = {}
+        for m in store.list("usage_metrics", owner=user.id, limit=100000):
+            if m.get("session_id") not in sessions:
+                continue
+            if day and m.get("day") != day:
+                continue
+            k = (m["day"], m["model"])
+            agg = out.setdefault(k, {"day": m["day"], "model": m["model"],
+                                     "prompt_tokens": 0,
+                                     "completion_tokens": 0, "calls": 0,
+                                     "cost_usd": 0.0})
+            agg["prompt_tokens"] += m["prompt_tokens"]
+            agg["completion_tokens"] += m["completion_tokens"]
+            agg["calls"] += 1
+            agg["cost_usd"] += m.get("cost_usd", 0.0)
+        return list(out.values())
+
     @app.get("/api/v1/usage")
     async def my_usage(user: AuthUser = Depends(auth_dep), day: str = ""):
         return usage.usage_for(user.id, day or None)
