@@ -210,6 +210,92 @@ def session_list(url: str = typer.Option("", "--url")):
         typer.echo(f"{s['id']}  {s['name'][:40]}")
 
 
+project_app = typer.Typer(help="Projects & spec-tasks")
+app.add_typer(project_app, name="project")
+
+
+@project_app.command("create")
+def project_create(name: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.post(f"{api}/api/v1/projects", json={"name": name},
+                   headers=headers)
+    typer.echo(r.json())
+
+
+@project_app.command("list")
+def project_list(url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    for p in httpx.get(f"{api}/api/v1/projects", headers=headers).json():
+        typer.echo(f"{p['id']}  {p['name']}")
+
+
+spectask_app = typer.Typer(help="Spec-task kanban")
+app.add_typer(spectask_app, name="spectask")
+
+
+@spectask_app.command("create")
+def spectask_create(project_id: str, title: str,
+                    description: str = typer.Option("", "--description"),
+                    url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.post(f"{api}/api/v1/projects/{project_id}/tasks",
+                   json={"title": title, "description": description},
+                   headers=headers)
+    typer.echo(r.json())
+
+
+@spectask_app.command("list")
+def spectask_list(project_id: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    for t in httpx.get(f"{api}/api/v1/projects/{project_id}/tasks",
+                       headers=headers).json():
+        typer.echo(f"{t['id']}  [{t['state']:<12}] {t['title']}")
+
+
+@spectask_app.command("plan")
+def spectask_plan(task_id: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.post(f"{api}/api/v1/spec-tasks/{task_id}/plan",
+                   headers=headers, timeout=300)
+    typer.echo(r.json())
+
+
+model_app = typer.Typer(help="Model catalog & local models")
+app.add_typer(model_app, name="model")
+
+
+@model_app.command("list")
+def model_list(url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    for m in httpx.get(f"{api}/api/v1/helix-models", headers=headers).json():
+        typer.echo(f"{m['id']:<16} ctx={m.get('context_length', '?'):<7} "
+                   f"{m.get('runtime', '')}")
+
+
+@model_app.command("load")
+def model_load(model: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.post(f"{api}/api/v1/local-models/{model}/load",
+                   headers=headers, timeout=900)
+    typer.echo(r.json())
+
+
+@model_app.command("unload")
+def model_unload(model: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.post(f"{api}/api/v1/local-models/{model}/unload",
+                   headers=headers)
+    typer.echo(r.json())
+
+
 @app.command()
 def version():
     from helix_amd import __version__
