@@ -156,6 +156,8 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : (G <= 4 ? 3 : 2))) void pag
         for (int g = 0; g < G; ++g) s_lds[g][i1] = s1[g];
       }
     }
+    __syncthreads();
+
     // --- Phase B: online softmax per head --------------------------------
     for (int g = wid; g < G; g += nwaves) {
       float m_chunk = -INFINITY;
