@@ -27,12 +27,12 @@ using namespace helix;
 namespace {
 
 constexpr int NTHREADS = 128;
-constexpr int CHUNK = 256;   // 2 tokens per thread in phase A
+constexpr int CHUNK = 128;
 constexpr int MAX_G = 8;
 constexpr int PART_QUANT = 128;
 
 template <int DHEAD, int G>
-__global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : (G <= 4 ? 3 : 2))) void paged_attn_decode_kernel(
+__global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
     float* __restrict__ tmp_ml,          // [B, Hq, maxP, 2]
@@ -100,61 +100,33 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : (G <= 4 ? 3 : 2))) void pag
   for (int base = p_start; base < p_end; base += CHUNK) {
     const int chunk_n = min(CHUNK, p_end - base);
 
-    // --- Phase A: scores — 2 tokens per thread so each q_lds read is
-    // amortized over both (q re-reads were ~35% of the kernel, abl2) ----
-    {
-      const int i0 = threadIdx.x, i1 = threadIdx.x + NTHREADS;
-      const bool a0 = i0 < chunk_n, a1 = i1 < chunk_n;
-      u16x8 k0[DHEAD / 8], k1[DHEAD / 8];
-      if (a0) {
-        const int tok = base + i0;
-        const int64_t blk = btable[tok / block_size];
-        const uint16_t* kr =
-            k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                        tok % block_size)) * DHEAD;
+    // --- Phase A: scores (batched K-row loads, then unrolled math) -------
+    if ((int)threadIdx.x < chunk_n) {
+      const int tok = base + threadIdx.x;
+      const int64_t blk = btable[tok / block_size];
+      const uint16_t* krow =
+          k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
+                      tok % block_size)) * DHEAD;
+      u16x8 kraw[DHEAD / 8];
 #pragma unroll
-        for (int j = 0; j < DHEAD / 8; ++j)
-          k0[j] = *reinterpret_cast<const u16x8*>(kr + j * 8);
-      }
-      if (a1) {
-        const int tok = base + i1;
-        const int64_t blk = btable[tok / block_size];
-        const uint16_t* kr =
-            k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                        tok % block_size)) * DHEAD;
+      for (int j = 0; j < DHEAD / 8; ++j)
+        kraw[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+      float s[G];
 #pragma unroll
-        for (int j = 0; j < DHEAD / 8; ++j)
-          k1[j] = *reinterpret_cast<const u16x8*>(kr + j * 8);
-      }
-      float s0[G], s1[G];
+      for (int g = 0; g < G; ++g) s[g] = 0.f;
 #pragma unroll
-      for (int g = 0; g < G; ++g) s0[g] = s1[g] = 0.f;
-#pragma unroll 2
       for (int j = 0; j < DHEAD / 8; ++j) {
-        float kv0[8], kv1[8];
+        float kv[8];
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          kv0[i] = a0 ? bf16_to_f32(k0[j][i]) : 0.f;
-          kv1[i] = a1 ? bf16_to_f32(k1[j][i]) : 0.f;
-        }
+        for (int i = 0; i < 8; ++i) kv[i] = bf16_to_f32(kraw[j][i]);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
 #pragma unroll
-          for (int i = 0; i < 8; ++i) {
-            const float qv = q_lds[g][j * 8 + i];
-            s0[g] += qv * kv0[i];
-            s1[g] += qv * kv1[i];
-          }
+          for (int i = 0; i < 8; ++i) s[g] += q_lds[g][j * 8 + i] * kv[i];
         }
       }
-      if (a0) {
 #pragma unroll
-        for (int g = 0; g < G; ++g) s_lds[g][i0] = s0[g];
-      }
-      if (a1) {
-#pragma unroll
-        for (int g = 0; g < G; ++g) s_lds[g][i1] = s1[g];
-      }
+      for (int g = 0; g < G; ++g) s_lds[g][threadIdx.x] = s[g];
     }
     __syncthreads();
 
