@@ -217,9 +217,23 @@ def create_app(cfg: Optional[ServerConfig] = None,
         return await client.embeddings({**body, "model": model})
 
     @app.get("/v1/models")
-    async def models(user: AuthUser = Depends(auth_dep)):
+    async def models(request: Request, user: AuthUser = Depends(auth_dep)):
+        # anthropic-version header => Anthropic-style model list
+        # (reference openai_model_handlers.go:17-45)
+        if request.headers.get("anthropic-version"):
+            data = await providers.aggregate_models(user.id)
+            return {"data": [{"id": m["id"], "type": "model",
+                              "display_name": m["id"]} for m in data],
+                    "has_more": False}
         data = await providers.aggregate_models(user.id)
         return {"object": "list", "data": data}
+
+    @app.post("/v1/images/generations")
+    async def images(user: AuthUser = Depends(auth_dep)):
+        # kept returning 501 until a diffusion engine exists
+        # (SURVEY.md §2.8 "Image generation ... out of MVP")
+        raise HTTPException(501, "image generation is not implemented in "
+                                 "the MI355X runner yet")
 
     # ------------------------------------------------------------------
     # Sessions API (reference server.go:1024-1064)

@@ -316,3 +316,13 @@ def test_switch_agent_and_step_info(stack):
     assert r.json()["parent_app"] == app_id
     assert client.get(f"/api/v1/sessions/{sid}/step-info",
                       headers=H(key)).status_code == 200
+
+
+def test_images_501_and_anthropic_model_list(stack):
+    _, client, _, key, _ = stack
+    assert client.post("/v1/images/generations",
+                       headers=H(key)).status_code == 501
+    r = client.get("/v1/models", headers={**H(key),
+                                          "anthropic-version": "2023-06-01"})
+    assert r.json()["has_more"] is False
+    assert r.json()["data"][0]["type"] == "model"
