@@ -120,3 +120,21 @@ class CUDAGraphRunner:
         entry = self.graphs[key]
         entry["graph"].replay()
         return entry["logits"][:B]
+
+    def warmup(self, len_buckets=(MIN_LEN_BUCKET,)):
+        """Eagerly capture all batch buckets so serving never pays
+        multi-second capture latency mid-request (cold-start cost moves
+        to model load)."""
+        self.in_ids.zero_()
+        self.in_pos.zero_()
+        self.in_slots.fill_(-1)
+        self.in_lens.fill_(1)
+        self.in_bt.zero_()
+        for lb in len_buckets:
+            lb = min(lb, self.max_model_len)
+            for nb in BATCH_BUCKETS:
+                if nb > self.max_batch:
+                    break
+                key = (nb, lb)
+                if key not in self.graphs:
+                    self.graphs[key] = self._capture(nb, lb)

@@ -101,6 +101,12 @@ class LLMInstance:
                          max_num_seqs=spec.max_num_seqs,
                          kv_cache_blocks=kv_blocks),
             device=device)
+        if self.engine.graph_runner is not None:
+            # eager hipGraph capture for all batch buckets (serving never
+            # pays capture latency mid-request)
+            t0 = time.time()
+            self.engine.graph_runner.warmup()
+            log.info("captured decode graphs in %.1fs", time.time() - t0)
         self.lock = threading.Lock()
         self.wake = threading.Event()
         self.stop = False
