@@ -121,7 +121,7 @@ class CUDAGraphRunner:
         entry["graph"].replay()
         return entry["logits"][:B]
 
-    def warmup(self, len_buckets=(MIN_LEN_BUCKET,)):
+    def warmup(self, len_buckets=(MIN_LEN_BUCKET, 2 * MIN_LEN_BUCKET)):
         """Eagerly capture all batch buckets so serving never pays
         multi-second capture latency mid-request (cold-start cost moves
         to model load)."""
