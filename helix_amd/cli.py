@@ -47,6 +47,9 @@ def runner(api_url: str = typer.Option("http://localhost:8080"),
            port: int = typer.Option(8090),
            advertise: str = typer.Option(""),
            device: str = typer.Option("cuda:0"),
+           tunnel: bool = typer.Option(False, "--tunnel",
+                                       help="reverse-dial (NAT-friendly): "
+                                            "no inbound port needed"),
            preload: str = typer.Option("", help="comma-separated models")):
     """Start a GPU runner: OpenAI-compatible server + heartbeat."""
     import threading
@@ -61,7 +64,16 @@ def runner(api_url: str = typer.Option("http://localhost:8080"),
     for m in filter(None, preload.split(",")):
         svc.ensure_loaded(m.strip())
     api = create_runner_app(svc, runner_id)
-    addr = advertise or f"http://{_local_ip()}:{port}"
+    if tunnel:
+        addr = f"tunnel:{runner_id}"
+
+        def run_tunnel():
+            from helix_amd.server.tunnel import tunnel_loop
+            asyncio.run(tunnel_loop(api_url, cfg.runner_plane.runner_token,
+                                    runner_id, svc))
+        threading.Thread(target=run_tunnel, daemon=True).start()
+    else:
+        addr = advertise or f"http://{_local_ip()}:{port}"
 
     def beat():
         asyncio.run(heartbeat_loop(

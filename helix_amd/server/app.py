@@ -45,6 +45,8 @@ def create_app(cfg: Optional[ServerConfig] = None,
     router = InferenceRouter(cfg.runner_plane.dispatch_stale_s,
                              cfg.runner_plane.offline_after_s)
     pubsub = ps.PubSub()
+    from helix_amd.server.tunnel import TunnelRegistry
+    tunnels = TunnelRegistry()
 
     if providers is None:
         providers = ProviderManager(store)
@@ -64,7 +66,8 @@ def create_app(cfg: Optional[ServerConfig] = None,
                 device=cfg.runner_plane.local_runner_device)
             providers.register("helix", LocalRunnerClient(runner_service))
         else:
-            providers.register("helix", RouterClient(router))
+            providers.register("helix", RouterClient(router,
+                                                     tunnels=tunnels))
 
     from helix_amd.agent.runner import AgentRunner
     from helix_amd.rag.service import RAGService
@@ -113,6 +116,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.providers = providers
     app.state.controller = controller
     app.state.runner_service = runner_service
+    app.state.tunnels = tunnels
     app.state.rag = rag
     app.state.knowledge = knowledge
     app.state.catalog = catalog
@@ -552,6 +556,56 @@ def create_app(cfg: Optional[ServerConfig] = None,
         hb = RunnerHeartbeat.model_validate(await request.json())
         router.on_heartbeat(hb)
         return {"ok": True}
+
+    @app.get("/api/v1/runner/tunnel/{runner_id}")
+    async def runner_tunnel(runner_id: str, _=Depends(runner_dep)):
+        """Reverse-dial downlink: runner holds this SSE stream open and
+        receives dispatched requests (RevDial parity)."""
+        q = tunnels.connect(runner_id)
+
+        async def sse():
+            try:
+                while True:
+                    try:
+                        msg = await asyncio.wait_for(q.get(), 15.0)
+                    except asyncio.TimeoutError:
+                        yield 'data: {"type": "ping"}\n\n'
+                        continue
+                    yield f"data: {json.dumps(msg)}\n\n"
+            finally:
+                tunnels.disconnect(runner_id, q)
+        return StreamingResponse(sse(), media_type="text/event-stream")
+
+    @app.post("/api/v1/runner/tunnel/{runner_id}/reply")
+    async def runner_tunnel_reply(runner_id: str, request: Request,
+                                  _=Depends(runner_dep)):
+        body = await request.json()
+        ok = await tunnels.push_reply(body.get("id", ""),
+                                      body.get("event") or {})
+        return {"ok": ok}
+
+    @app.get("/api/v1/events/{session_id}")
+    async def session_events(session_id: str,
+                             user: AuthUser = Depends(auth_dep)):
+        """SSE session event stream (WS-free deployments)."""
+        s_obj = controller.get_session(session_id)
+        if s_obj is None or (s_obj.owner != user.id and not user.admin):
+            raise HTTPException(404, "session not found")
+        sub = await pubsub.subscribe(ps.session_queue(s_obj.owner,
+                                                      session_id))
+
+        async def sse():
+            try:
+                while True:
+                    try:
+                        topic, msg = await sub.get(timeout=30.0)
+                    except asyncio.TimeoutError:
+                        yield 'data: {"type": "ping"}\n\n'
+                        continue
+                    yield f"data: {json.dumps(msg)}\n\n"
+            finally:
+                await sub.close()
+        return StreamingResponse(sse(), media_type="text/event-stream")
 
     @app.get("/api/v1/admin/runners/{runner_id}/logs")
     async def runner_logs(runner_id: str, n: int = 200,

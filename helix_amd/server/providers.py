@@ -187,18 +187,31 @@ class LocalRunnerClient(Client):
 class RouterClient(Client):
     """The internal "helix" provider in control-plane mode: dispatches to
     remote runners through the inference router (reference
-    helix_openai_server.go: PickRunner -> dispatchToSandbox)."""
+    helix_openai_server.go: PickRunner -> dispatchToSandbox). Runners
+    reachable only via reverse tunnel advertise `tunnel:<id>` addresses
+    and are dispatched through the TunnelRegistry (RevDial parity)."""
 
-    def __init__(self, router, timeout: float = 300.0, transport=None):
+    def __init__(self, router, timeout: float = 300.0, transport=None,
+                 tunnels=None):
         self.provider = "helix"
         self.router = router
+        self.tunnels = tunnels
         self._http = httpx.AsyncClient(timeout=timeout, transport=transport)
 
-    def _pick(self, model: str) -> str:
-        return self.router.pick_runner(model)
+    def _pick(self, model: str):
+        addr = self.router.pick_runner(model)
+        from helix_amd.server.tunnel import TUNNEL_ADDR_PREFIX, TunnelClient
+        if addr.startswith(TUNNEL_ADDR_PREFIX):
+            if self.tunnels is None:
+                raise ProviderError("tunnel registry unavailable", 502)
+            return TunnelClient(self.tunnels,
+                                addr[len(TUNNEL_ADDR_PREFIX):])
+        return addr
 
     async def chat(self, req: dict) -> dict:
         addr = self._pick(req.get("model", ""))
+        if not isinstance(addr, str):
+            return await addr.chat(req)
         r = await self._http.post(f"{addr}/v1/chat/completions",
                                   json={**req, "stream": False})
         if r.status_code != 200:
@@ -207,6 +220,10 @@ class RouterClient(Client):
 
     async def chat_stream(self, req: dict) -> AsyncIterator[dict]:
         addr = self._pick(req.get("model", ""))
+        if not isinstance(addr, str):
+            async for chunk in addr.chat_stream(req):
+                yield chunk
+            return
         async with self._http.stream(
                 "POST", f"{addr}/v1/chat/completions",
                 json={**req, "stream": True}) as r:
@@ -222,6 +239,8 @@ class RouterClient(Client):
 
     async def embeddings(self, req: dict) -> dict:
         addr = self._pick(req.get("model", ""))
+        if not isinstance(addr, str):
+            return await addr.embeddings(req)
         r = await self._http.post(f"{addr}/v1/embeddings", json=req)
         if r.status_code != 200:
             raise ProviderError(r.text, r.status_code)
