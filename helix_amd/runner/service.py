@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import asyncio
 import logging
+import queue
 import threading
 import time
 from dataclasses import dataclass, field
@@ -107,7 +108,11 @@ class LLMInstance:
             t0 = time.time()
             self.engine.graph_runner.warmup()
             log.info("captured decode graphs in %.1fs", time.time() - t0)
-        self.lock = threading.Lock()
+        # Engine state is touched ONLY by the engine thread; submissions
+        # and cancellations flow through an intake queue. (Holding a lock
+        # across step() starved submitters: the engine thread re-acquired
+        # it instantly between steps, serializing all requests.)
+        self.intake: queue.Queue = queue.Queue()
         self.wake = threading.Event()
         self.stop = False
         self.last_used = time.time()
@@ -117,32 +122,53 @@ class LLMInstance:
 
     @property
     def in_flight(self) -> int:
-        with self.lock:
-            return len(self.engine.waiting) + len(self.engine.running)
+        # approximate (racy read is fine for scheduling/stats)
+        return (len(self.engine.waiting) + len(self.engine.running)
+                + self.intake.qsize())
+
+    def _drain_intake(self):
+        while True:
+            try:
+                kind, payload = self.intake.get_nowait()
+            except queue.Empty:
+                return
+            try:
+                if kind == "submit":
+                    seq_id, prompt_ids, params, on_token = payload
+                    self.engine.add_request(seq_id, prompt_ids, params,
+                                            on_token=on_token)
+                elif kind == "cancel":
+                    self.engine.cancel(payload)
+            except Exception as e:
+                log.warning("intake %s failed: %s", kind, e)
+                if kind == "submit" and payload[3] is not None:
+                    # surface the rejection to the waiting caller
+                    class _F:
+                        seq_id = payload[0]
+                        finish_reason = f"error: {e}"
+                    try:
+                        payload[3](_F(), 0, True)
+                    except Exception:
+                        pass
 
     def _loop(self):
         while not self.stop:
-            with self.lock:
-                has_work = self.engine.has_work
-            if not has_work:
+            self._drain_intake()
+            if not self.engine.has_work:
                 self.wake.wait(timeout=0.05)
                 self.wake.clear()
                 continue
-            with self.lock:
-                self.engine.step()
+            self.engine.step()
 
     def submit(self, seq_id: str, prompt_ids: List[int],
-               params: SamplingParams, on_token) -> Sequence:
+               params: SamplingParams, on_token) -> None:
         self.last_used = time.time()
-        with self.lock:
-            seq = self.engine.add_request(seq_id, prompt_ids, params,
-                                          on_token=on_token)
+        self.intake.put(("submit", (seq_id, prompt_ids, params, on_token)))
         self.wake.set()
-        return seq
 
     def cancel(self, seq_id: str):
-        with self.lock:
-            self.engine.cancel(seq_id)
+        self.intake.put(("cancel", seq_id))
+        self.wake.set()
 
     def shutdown(self):
         self.stop = True
