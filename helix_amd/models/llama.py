@@ -36,6 +36,7 @@ class LlamaConfig:
     rope_base: float = 500000.0
     max_position: int = 8192
     tie_embeddings: bool = False
+    sliding_window: int = 0        # 0 = full attention (Mistral v0.1: 4096)
 
     @property
     def q_size(self) -> int:
@@ -52,7 +53,8 @@ PRESETS = {
         name="llama3-70b", hidden_size=8192, intermediate_size=28672,
         num_layers=80, num_heads=64, num_kv_heads=8),
     "mistral-7b": LlamaConfig(
-        name="mistral-7b", vocab_size=32000, rope_base=10000.0),
+        name="mistral-7b", vocab_size=32000, rope_base=10000.0,
+        sliding_window=4096),
     # Tiny configs for CPU tests and smoke runs.
     "tiny": LlamaConfig(
         name="tiny", vocab_size=512, hidden_size=256, intermediate_size=512,
@@ -97,6 +99,7 @@ class Attention(nn.Module):
         self.hd = cfg.head_dim
         self.scale = self.hd ** -0.5
         self.tp_size = tp_size
+        self.window = cfg.sliding_window or 0
         q, kv, h = self.nh * self.hd, self.nkv * self.hd, cfg.hidden_size
         self.qkv_proj = nn.Linear(h, q + 2 * kv, bias=False)
         self.o_proj = nn.Linear(q, h, bias=False)
@@ -117,13 +120,14 @@ class Attention(nn.Module):
             ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
         if meta.is_prefill:
             o = ops.attn_prefill(q, k, v, meta.cu_seqlens, meta.max_seqlen,
-                                 self.scale)
+                                 self.scale, window=self.window)
         else:
             k_cache, v_cache = kv_cache
             o = ops.paged_attn_decode(q, k_cache, v_cache, meta.block_tables,
                                       meta.seq_lens, self.scale,
                                       meta.workspace,
-                                      meta.max_len or None)
+                                      meta.max_len or None,
+                                      window=self.window)
         out = self.o_proj(o.view(T, -1))
         if self.tp_size > 1:
             from helix_amd import parallel

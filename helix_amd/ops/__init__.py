@@ -104,13 +104,15 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
 
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  cu_seqlens: torch.Tensor, max_seqlen: int,
-                 scale: float, causal: bool = True) -> torch.Tensor:
+                 scale: float, causal: bool = True,
+                 window: int = 0) -> torch.Tensor:
     if q.is_cuda:
         out = torch.empty_like(q)
         _native().attn_prefill(out, q, k, v, cu_seqlens, max_seqlen, scale,
-                               causal)
+                               causal, window)
         return out
-    return ref.attn_prefill(q, k, v, cu_seqlens, max_seqlen, scale, causal)
+    return ref.attn_prefill(q, k, v, cu_seqlens, max_seqlen, scale, causal,
+                            window)
 
 
 # Flash-decoding split-K quantum (must match PART_QUANT in
@@ -132,8 +134,8 @@ def decode_workspace(max_batch: int, num_q_heads: int, head_dim: int,
 def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, block_tables: torch.Tensor,
                       seq_lens: torch.Tensor, scale: float,
-                      workspace=None, max_len: int | None = None
-                      ) -> torch.Tensor:
+                      workspace=None, max_len: int | None = None,
+                      window: int = 0) -> torch.Tensor:
     if q.is_cuda:
         out = torch.empty_like(q)
         if max_len is None:
@@ -145,10 +147,10 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
         _native().paged_attn_decode(out, q, k_cache, v_cache, block_tables,
                                     seq_lens, scale,
                                     tmp_out[:q.shape[0]], tmp_ml[:q.shape[0]],
-                                    max_len)
+                                    max_len, window)
         return out
     return ref.paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens,
-                                 scale)
+                                 scale, window)
 
 
 def sample_tokens(logits: torch.Tensor, temperatures: torch.Tensor,

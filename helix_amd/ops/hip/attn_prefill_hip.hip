@@ -50,7 +50,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
     const uint16_t* __restrict__ k,    // [T, Hkv, D]
     const uint16_t* __restrict__ v,    // [T, Hkv, D]
     const int* __restrict__ cu_seqlens,  // [B+1]
-    float scale, int Hq, int Hkv) {
+    float scale, int Hq, int Hkv, int window) {
   const int qtile = blockIdx.x;
   const int seq = blockIdx.y;
   const int hq = blockIdx.z;
@@ -101,6 +101,10 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   for (int c = 0; c < nc; ++c) o_acc[c] = floatx4{0, 0, 0, 0};
 
   const int kv_max = CAUSAL ? min(len, qbase + QTILE) : len;
+  // sliding window: rows in this q-tile never look below kv_lo
+  const int kv_lo = (CAUSAL && window > 0)
+                        ? max(0, qbase - window + 1) : 0;
+  const int t_first = kv_lo / KTILE;
   const int ntiles = (kv_max + KTILE - 1) / KTILE;
 
   // Per-thread staging assignment: KTILE*D/8 16B-chunks over 256 threads.
@@ -141,11 +145,11 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
     }
   };
 
-  load_tile(0);
+  load_tile(t_first);
   store_tile(0);
-  if (ntiles > 1) load_tile(1);
-  for (int t = 0; t < ntiles; ++t) {
-    const int buf = t & 1;
+  if (t_first + 1 < ntiles) load_tile(t_first + 1);
+  for (int t = t_first; t < ntiles; ++t) {
+    const int buf = (t - t_first) & 1;
     __syncthreads();            // buf's staging (and last compute) done
 
     const int kv_base = t * KTILE;
@@ -177,7 +181,9 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
       for (int h = 0; h < KTILE / 16; ++h) {
         s[h] = s_frag[h][r] * scale;
         const int tp = kv_base + h * 16 + lane_lo;
-        if ((CAUSAL && tp > qpos) || tp >= len) s[h] = -1e30f;
+        if ((CAUSAL && (tp > qpos ||
+                        (window > 0 && qpos - tp >= window))) ||
+            tp >= len) s[h] = -1e30f;
         m_tile = fmaxf(m_tile, s[h]);
       }
 #pragma unroll
@@ -257,7 +263,8 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   torch::Tensor v, torch::Tensor cu_seqlens,
-                  int64_t max_seqlen, double scale, bool causal) {
+                  int64_t max_seqlen, double scale, bool causal,
+                  int64_t window) {
   const int Hq = q.size(1);
   const int D = q.size(2);
   const int Hkv = k.size(1);
@@ -273,7 +280,8 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                      (const uint16_t*)q.data_ptr(),                          \
                      (const uint16_t*)k.data_ptr(),                          \
                      (const uint16_t*)v.data_ptr(),                          \
-                     cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv)
+                     cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv,      \
+                     (int)window)
   if (causal) {
     if (D == 128) LAUNCH(1, 128); else LAUNCH(1, 64);
   } else {

@@ -18,10 +18,12 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor block_tables, torch::Tensor seq_lens,
                        double scale, torch::Tensor tmp_out,
-                       torch::Tensor tmp_ml, int64_t partition_size);
+                       torch::Tensor tmp_ml, int64_t partition_size,
+                       int64_t window);
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   torch::Tensor v, torch::Tensor cu_seqlens,
-                  int64_t max_seqlen, double scale, bool causal);
+                  int64_t max_seqlen, double scale, bool causal,
+                  int64_t window);
 void sample_tokens(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperatures, torch::Tensor seeds);
 void gemm_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,

@@ -42,15 +42,18 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_
     const int* __restrict__ block_tables,// [B, max_blocks]
     const int* __restrict__ seq_lens,    // [B]
     float scale, int Hkv, int block_size, int max_blocks,
-    int partition_size, int max_parts) {
+    int partition_size, int max_parts, int window) {
   const int seq = blockIdx.x;
   const int hkv = blockIdx.y;
   const int part = blockIdx.z;
   const int nparts = gridDim.z;
   const int Hq = Hkv * G;
   const int len = seq_lens[seq];
-  const int p_start = part * partition_size;
-  if (p_start >= len) {
+  const int kv_lo = (window > 0) ? max(0, len - window) : 0;
+  int p_start = part * partition_size;
+  const int p_hi = min(len, p_start + partition_size);
+  if (p_start < kv_lo) p_start = kv_lo;
+  if (p_start >= len || p_hi <= kv_lo) {
     if (nparts > 1 && threadIdx.x == 0) {
 #pragma unroll
       for (int g = 0; g < G; ++g) {
@@ -62,7 +65,7 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_
     }
     return;
   }
-  const int p_end = min(len, p_start + partition_size);
+  const int p_end = p_hi;
 
   __shared__ float q_lds[G][DHEAD];
   __shared__ float s_lds[G][CHUNK];
@@ -292,11 +295,12 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
                    const uint16_t* q, const uint16_t* kc, const uint16_t* vc,
                    const int* bt, const int* lens, float scale, int B,
                    int Hkv, int block_size, int max_blocks, int eff_part,
-                   int nparts, int max_parts, hipStream_t stream) {
+                   int nparts, int max_parts, int window,
+                   hipStream_t stream) {
   hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G>),
                      dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
                      tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
-                     block_size, max_blocks, eff_part, max_parts);
+                     block_size, max_blocks, eff_part, max_parts, window);
 }
 
 }  // namespace
@@ -305,7 +309,8 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        torch::Tensor k_cache, torch::Tensor v_cache,
                        torch::Tensor block_tables, torch::Tensor seq_lens,
                        double scale, torch::Tensor tmp_out,
-                       torch::Tensor tmp_ml, int64_t max_len_hint) {
+                       torch::Tensor tmp_ml, int64_t max_len_hint,
+                       int64_t window) {
   const int B = q.size(0);
   const int Hq = q.size(1);
   const int D = q.size(2);
@@ -340,7 +345,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
 #define DISPATCH(DH, GG)                                                     \
   launch_decode<DH, GG>(o, to, tm, qp, kp, vp, bp, lp, (float)scale, B,     \
                         Hkv, block_size, max_blocks, eff_part, nparts,      \
-                        max_parts, stream)
+                        max_parts, (int)window, stream)
   if (D == 128) {
     if (G == 1) DISPATCH(128, 1);
     else if (G == 2) DISPATCH(128, 2);
@@ -357,5 +362,7 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   if (nparts > 1) {
     hipLaunchKernelGGL(paged_attn_reduce_kernel, dim3(B, Hq), dim3(128), 0,
                        stream, o, to, tm, lp, Hq, D, eff_part, max_parts);
+    // note: dead (windowed-out) partitions wrote ml = -inf and are
+    // skipped by the reduce weighting
   }
 }

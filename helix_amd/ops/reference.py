@@ -86,26 +86,37 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
 
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  cu_seqlens: torch.Tensor, max_seqlen: int,
-                 scale: float, causal: bool = True) -> torch.Tensor:
-    """Varlen causal GQA attention. q: [T, Hq, D], k/v: [T, Hkv, D]."""
+                 scale: float, causal: bool = True,
+                 window: int = 0) -> torch.Tensor:
+    """Varlen causal GQA attention (optional sliding window).
+    q: [T, Hq, D], k/v: [T, Hkv, D]."""
     Hq, Hkv = q.shape[1], k.shape[1]
     G = Hq // Hkv
     out = torch.empty_like(q)
     for i in range(cu_seqlens.shape[0] - 1):
         s, e = int(cu_seqlens[i]), int(cu_seqlens[i + 1])
+        L = e - s
         qs = q[s:e].float()
         ks = k[s:e].float().repeat_interleave(G, dim=1)
         vs = v[s:e].float().repeat_interleave(G, dim=1)
+        mask = None
+        is_causal = causal
+        if causal and window > 0:
+            pos = torch.arange(L)
+            m = (pos[None, :] <= pos[:, None]) &                 (pos[:, None] - pos[None, :] < window)
+            mask = torch.where(m, 0.0, float("-inf"))
+            is_causal = False
         o = torch.nn.functional.scaled_dot_product_attention(
             qs.transpose(0, 1), ks.transpose(0, 1), vs.transpose(0, 1),
-            is_causal=causal, scale=scale)
+            attn_mask=mask, is_causal=is_causal, scale=scale)
         out[s:e] = o.transpose(0, 1).to(q.dtype)
     return out
 
 
 def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                       v_cache: torch.Tensor, block_tables: torch.Tensor,
-                      seq_lens: torch.Tensor, scale: float) -> torch.Tensor:
+                      seq_lens: torch.Tensor, scale: float,
+                      window: int = 0) -> torch.Tensor:
     """q: [B, Hq, D] single token per seq. Gathers KV then full attention."""
     B, Hq, D = q.shape
     Hkv = k_cache.shape[1]
@@ -118,6 +129,9 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
         blocks = block_tables[b, :nb].long()
         k = k_cache[blocks].permute(0, 2, 1, 3).reshape(nb * bs, Hkv, D)[:L]
         v = v_cache[blocks].permute(0, 2, 1, 3).reshape(nb * bs, Hkv, D)[:L]
+        if window > 0:
+            k = k[max(0, L - window):]
+            v = v[max(0, L - window):]
         kf = k.float().repeat_interleave(G, dim=1)  # [L, Hq, D]
         vf = v.float().repeat_interleave(G, dim=1)
         qf = q[b].float()  # [Hq, D]

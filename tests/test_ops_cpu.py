@@ -105,3 +105,29 @@ def test_cos_sin_cache():
     assert torch.allclose(cs[0, :4], torch.ones(4))  # cos(0)
     assert torch.allclose(cs[0, 4:], torch.zeros(4))  # sin(0)
     assert math.isclose(float(cs[1, 0]), math.cos(1.0), rel_tol=1e-5)
+
+
+def test_sliding_window_reference_consistency():
+    """Windowed decode of the last position == windowed prefill last row."""
+    torch.manual_seed(5)
+    L, hq, hkv, d, bs, W = 50, 8, 4, 64, 16, 16
+    q = torch.randn(L, hq, d, dtype=torch.bfloat16)
+    k = torch.randn(L, hkv, d, dtype=torch.bfloat16)
+    v = torch.randn(L, hkv, d, dtype=torch.bfloat16)
+    cu = torch.tensor([0, L], dtype=torch.int32)
+    scale = d ** -0.5
+    pre = ops.attn_prefill(q, k, v, cu, L, scale, window=W)
+    nb = (L + bs - 1) // bs
+    kc = torch.zeros(nb + 1, hkv, bs, d, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    ops.reshape_and_cache(k, v, kc, vc,
+                          torch.arange(bs, bs + L, dtype=torch.int64))
+    bt = torch.arange(1, nb + 2, dtype=torch.int32).unsqueeze(0)
+    out = ops.paged_attn_decode(q[-1:].reshape(1, hq, d), kc, vc, bt,
+                                torch.tensor([L], dtype=torch.int32), scale,
+                                window=W)
+    assert torch.allclose(out[0].float(), pre[-1].float(), atol=3e-2,
+                          rtol=3e-2)
+    # and differs from full attention (the window actually applies)
+    full = ops.attn_prefill(q, k, v, cu, L, scale)
+    assert not torch.allclose(full[-1].float(), pre[-1].float(), atol=1e-3)

@@ -208,3 +208,47 @@ def test_gemm_bf16_gelu():
     got = ops.gemm_bf16(x, w, None, act=1)
     want = ref.gemm_bf16(x.cpu(), w.cpu(), None, act=1)
     assert_close_bf16(got, want, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("window,lens", [(64, [200]), (128, [100, 400])])
+def test_attn_prefill_sliding_window(window, lens):
+    torch.manual_seed(13)
+    hq, hkv, d = 32, 8, 128
+    T = sum(lens)
+    q = torch.randn(T, hq, d, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(T, hkv, d, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(T, hkv, d, dtype=torch.bfloat16, device=DEV)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)),
+                      dtype=torch.int32, device=DEV)
+    scale = d ** -0.5
+    got = ops.attn_prefill(q, k, v, cu, max(lens), scale, window=window)
+    want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), cu.cpu(), max(lens),
+                            scale, window=window)
+    assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("window,lens", [(64, [200]), (256, [100, 700]),
+                                         (512, [2048])])
+def test_paged_attn_decode_sliding_window(window, lens):
+    torch.manual_seed(14)
+    hq, hkv, d, bs = 32, 8, 128, 16
+    B = len(lens)
+    max_blocks = (max(lens) + bs - 1) // bs
+    total_blocks = sum((l + bs - 1) // bs for l in lens) + 1
+    q = torch.randn(B, hq, d, dtype=torch.bfloat16, device=DEV)
+    kc = torch.randn(total_blocks, hkv, bs, d, dtype=torch.bfloat16,
+                     device=DEV)
+    vc = torch.randn_like(kc)
+    bt = torch.zeros(B, max_blocks, dtype=torch.int32, device=DEV)
+    nxt = 1
+    for b, l in enumerate(lens):
+        n = (l + bs - 1) // bs
+        bt[b, :n] = torch.arange(nxt, nxt + n, dtype=torch.int32)
+        nxt += n
+    seq_lens = torch.tensor(lens, dtype=torch.int32, device=DEV)
+    scale = d ** -0.5
+    got = ops.paged_attn_decode(q, kc, vc, bt, seq_lens, scale,
+                                window=window)
+    want = ref.paged_attn_decode(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                                 seq_lens.cpu(), scale, window=window)
+    assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
