@@ -285,7 +285,11 @@ __global__ __launch_bounds__(128) void paged_attn_reduce_kernel(
   const float* tp = tmp_out + ((int64_t)seq * Hq + hq) * max_parts * D;
   for (int d = threadIdx.x; d < D; d += blockDim.x) {
     float o = 0.f;
-    for (int p = 0; p < nparts; ++p) o += w[p] * tp[p * D + d];
+    for (int p = 0; p < nparts; ++p) {
+      // skip dead partitions explicitly: their tmp_out is uninitialized
+      // and 0 * NaN would poison the sum (windowed-out partitions)
+      if (w[p] != 0.f) o += w[p] * tp[p * D + d];
+    }
     out[((int64_t)seq * Hq + hq) * D + d] = f32_to_bf16(o / l_sh);
   }
 }
