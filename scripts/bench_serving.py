@@ -23,7 +23,8 @@ def pct(v, p):
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--model", default="llama3-8b",
+                    help="comma-separated for co-resident multi-model")
     ap.add_argument("--rate", type=float, default=8.0, help="req/s")
     ap.add_argument("--requests", type=int, default=64)
     ap.add_argument("--prompt-len", type=int, default=512)
@@ -34,8 +35,8 @@ def main():
     from helix_amd.runner.service import RunnerService
     from helix_amd.engine.sampling_params import SamplingParams
     svc = RunnerService(device=args.device)
-    inst = svc.ensure_loaded(args.model)
-    vocab = inst.engine.model_cfg.vocab_size
+    models = [m.strip() for m in args.model.split(",")]
+    insts = [svc.ensure_loaded(m) for m in models]
 
     stats = {}
     lock = threading.Lock()
@@ -59,6 +60,8 @@ def main():
     torch.manual_seed(0)
     t0 = time.monotonic()
     for i in range(args.requests):
+        inst = insts[i % len(insts)]
+        vocab = inst.engine.model_cfg.vocab_size
         prompt = [random.randrange(3, vocab - 1)
                   for _ in range(args.prompt_len)]
         inst.submit(f"q{i}", prompt,
@@ -79,6 +82,7 @@ def main():
             tpots.append((b - a) * 1000)
     import json
     print(json.dumps({
+        "models": models,
         "requests": len(stats), "rate_rps": args.rate,
         "prompt_len": args.prompt_len, "max_tokens": args.max_tokens,
         "wall_s": round(wall, 2),
