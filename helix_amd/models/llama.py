@@ -64,6 +64,11 @@ PRESETS = {
         name="tiny-gqa", vocab_size=1024, hidden_size=512,
         intermediate_size=1024, num_layers=4, num_heads=8, num_kv_heads=2,
         head_dim=64, max_position=1024, rope_base=10000.0),
+    "tiny-sw": LlamaConfig(
+        name="tiny-sw", vocab_size=512, hidden_size=256,
+        intermediate_size=512, num_layers=2, num_heads=4, num_kv_heads=2,
+        head_dim=64, max_position=512, rope_base=10000.0,
+        sliding_window=24),
 }
 
 

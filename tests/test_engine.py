@@ -183,3 +183,20 @@ def test_concurrent_submit_thread_safety():
         assert len(done) == 24
     finally:
         svc.shutdown()
+
+
+def test_sliding_window_engine_consistency():
+    """A sliding-window model's incremental decode must match a fresh
+    windowed prefill of the same context (Mistral-style window)."""
+    torch.manual_seed(0)
+    cfg = EngineConfig(model="tiny-sw", max_model_len=128, max_num_seqs=4,
+                       kv_cache_blocks=128, eos_token_id=-1)
+    eng = LLMEngine(cfg, device="cpu")
+    prompt = list(range(1, 41))   # > window of 24
+    sp = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    out = eng.generate([prompt], sp)[0]
+    eng2 = LLMEngine(cfg, device="cpu")
+    out2 = eng2.generate([prompt + out[:-1]],
+                         SamplingParams(temperature=0.0, max_tokens=1,
+                                        ignore_eos=True))[0]
+    assert out2[0] == out[-1]
