@@ -326,3 +326,19 @@ def test_images_501_and_anthropic_model_list(stack):
                                           "anthropic-version": "2023-06-01"})
     assert r.json()["has_more"] is False
     assert r.json()["data"][0]["type"] == "model"
+
+
+def test_provider_resolution_order(stack):
+    """Reference openai_chat_handlers.go:148-192: cached model list wins,
+    then provider/ prefix, then default provider with the FULL name."""
+    app, _, _, _, _ = stack
+    pm = app.state.providers
+    pm._model_cache = {"mock": ["exact-model"]}
+    # 1. cached-list match
+    assert pm.resolve("exact-model", "default") == ("mock", "exact-model")
+    # 2. prefix parse for a registered provider
+    assert pm.resolve("mock/other", "default") == ("mock", "other")
+    # 3. unknown prefix falls through to default, name preserved intact
+    assert pm.resolve("unknown/model", "mock") == ("mock", "unknown/model")
+    # 4. plain unknown name -> default provider
+    assert pm.resolve("whatever", "mock") == ("mock", "whatever")
