@@ -16,7 +16,7 @@ from typing import Callable, Dict, List, Optional
 import torch
 
 from helix_amd import ops
-from helix_amd.engine.kv_cache import KVCache
+from helix_amd.engine.kv_cache import KVCache, chain_hash
 from helix_amd.engine.sampling_params import SamplingParams
 from helix_amd.models.llama import (DecodeMeta, LlamaConfig, LlamaForCausalLM,
                                     PrefillMeta, PRESETS)
@@ -39,6 +39,8 @@ class Sequence:
     output_ids: List[int] = field(default_factory=list)
     status: SeqStatus = SeqStatus.WAITING
     block_table: List[int] = field(default_factory=list)
+    cached_prefix: int = 0                 # tokens served from prefix cache
+    block_hashes: List[bytes] = field(default_factory=list)
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: Optional[float] = None
     finish_reason: Optional[str] = None
@@ -62,6 +64,7 @@ class EngineConfig:
     eos_token_id: int = 2
     seed: int = 0
     enforce_eager: bool = False               # disable hipGraph capture
+    enable_prefix_caching: bool = True
 
 
 class LLMEngine:
@@ -180,16 +183,39 @@ class LLMEngine:
     def _schedule_prefill(self) -> List[Sequence]:
         batch: List[Sequence] = []
         tokens = 0
+        bs = self.cfg.block_size
+        alloc = self.kv.allocator
         while self.waiting and len(self.running) + len(batch) < self.cfg.max_num_seqs:
             seq = self.waiting[0]
-            need = len(seq.prompt_ids)
+            L = len(seq.prompt_ids)
+            # prefix-cache match (chain hashes over full blocks)
+            matched: List[int] = []
+            if self.cfg.enable_prefix_caching:
+                if not seq.block_hashes:
+                    h = b""
+                    for i in range(L // bs):
+                        h = chain_hash(h, seq.prompt_ids[i * bs:(i + 1) * bs])
+                        seq.block_hashes.append(h)
+                for h in seq.block_hashes:
+                    b = alloc.lookup_hash(h)
+                    if b is None:
+                        break
+                    matched.append(b)
+                # always leave at least one token to prefill
+                while matched and len(matched) * bs > L - 1:
+                    matched.pop()
+            cached = len(matched) * bs
+            need = L - cached
             if batch and tokens + need > self.cfg.max_prefill_tokens:
                 break
-            nblocks = (need + self.cfg.block_size - 1) // self.cfg.block_size
+            nblocks_total = (L + bs - 1) // bs
+            new_blocks = nblocks_total - len(matched)
             # +1 headroom block so the first decode step can't OOM-deadlock
-            if not self.kv.allocator.can_allocate(nblocks + 1):
+            if not alloc.can_allocate(new_blocks + 1):
                 break
-            seq.block_table = self.kv.allocator.allocate(nblocks)
+            seq.block_table = [alloc.share(b) for b in matched] + \
+                alloc.allocate(new_blocks)
+            seq.cached_prefix = cached
             batch.append(self.waiting.pop(0))
             tokens += need
         return batch
@@ -200,22 +226,44 @@ class LLMEngine:
 
     def _run_prefill(self, batch: List[Sequence]):
         bs = self.cfg.block_size
-        input_ids, positions, slots, cu = [], [], [], [0]
+        input_ids, positions, slots, cu_q, cu_k = [], [], [], [0], [0]
+        gather_blk, gather_off = [], []
+        any_cached = any(s.cached_prefix for s in batch)
         for seq in batch:
             L = len(seq.prompt_ids)
-            input_ids.extend(seq.prompt_ids)
-            positions.extend(range(L))
-            slots.extend(self._slot(seq, p) for p in range(L))
-            cu.append(cu[-1] + L)
+            P = seq.cached_prefix
+            input_ids.extend(seq.prompt_ids[P:])
+            positions.extend(range(P, L))
+            slots.extend(self._slot(seq, p) for p in range(P, L))
+            cu_q.append(cu_q[-1] + L - P)
+            cu_k.append(cu_k[-1] + L)
+            if any_cached:
+                for p in range(L):
+                    gather_blk.append(seq.block_table[p // bs])
+                    gather_off.append(p % bs)
         dev = self.device
         meta = PrefillMeta(
-            cu_seqlens=torch.tensor(cu, dtype=torch.int32, device=dev),
-            max_seqlen=max(len(s.prompt_ids) for s in batch),
+            cu_seqlens=torch.tensor(cu_q, dtype=torch.int32, device=dev),
+            max_seqlen=max(len(s.prompt_ids) - s.cached_prefix
+                           for s in batch),
             slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
-            positions=torch.tensor(positions, dtype=torch.int64, device=dev))
+            positions=torch.tensor(positions, dtype=torch.int64, device=dev),
+            cu_seqlens_k=(torch.tensor(cu_k, dtype=torch.int32, device=dev)
+                          if any_cached else None),
+            gather_blk=(torch.tensor(gather_blk, dtype=torch.int64,
+                                     device=dev) if any_cached else None),
+            gather_off=(torch.tensor(gather_off, dtype=torch.int64,
+                                     device=dev) if any_cached else None))
         ids = torch.tensor(input_ids, dtype=torch.int64, device=dev)
         hidden = self.model(ids, self.kv.caches, meta)
-        last_rows = torch.tensor([c - 1 for c in cu[1:]], dtype=torch.int64,
+        # register this batch's new full blocks in the prefix cache
+        if self.cfg.enable_prefix_caching:
+            for seq in batch:
+                start = seq.cached_prefix // bs
+                for i in range(start, len(seq.block_hashes)):
+                    self.kv.allocator.register_hash(seq.block_table[i],
+                                                    seq.block_hashes[i])
+        last_rows = torch.tensor([c - 1 for c in cu_q[1:]], dtype=torch.int64,
                                  device=dev)
         logits = self.model.compute_logits(hidden[last_rows])
         tokens = self._sample(batch, logits)
@@ -239,6 +287,8 @@ class LLMEngine:
         # next prefill recomputes the full context
         victim.prompt_ids = victim.prompt_ids + victim.output_ids
         victim.output_ids = []
+        victim.cached_prefix = 0
+        victim.block_hashes = []
         victim.status = SeqStatus.WAITING
         self.waiting.insert(0, victim)
         return victim

@@ -74,11 +74,17 @@ PRESETS = {
 
 @dataclass
 class PrefillMeta:
-    """Varlen prefill: q/k/v are concatenated new tokens of all sequences."""
+    """Varlen prefill: q/k/v are concatenated new tokens of all sequences.
+    With prefix caching, q covers only the uncached suffix; attention
+    gathers the full context K/V from the paged cache (gather_blk/off)
+    and cu_seqlens_k carries the full-context boundaries."""
     cu_seqlens: torch.Tensor       # [B+1] int32 (new-token boundaries)
     max_seqlen: int
     slot_mapping: torch.Tensor     # [T] int64 cache slots for new tokens
     positions: torch.Tensor        # [T] int64
+    cu_seqlens_k: Optional[torch.Tensor] = None   # [B+1] int32 full context
+    gather_blk: Optional[torch.Tensor] = None     # [Tk] int64
+    gather_off: Optional[torch.Tensor] = None     # [Tk] int64
     is_prefill: bool = True
 
 
@@ -124,8 +130,20 @@ class Attention(nn.Module):
             k_cache, v_cache = kv_cache
             ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
         if meta.is_prefill:
-            o = ops.attn_prefill(q, k, v, meta.cu_seqlens, meta.max_seqlen,
-                                 self.scale, window=self.window)
+            if meta.cu_seqlens_k is not None:
+                # cached-prefix prefill: full-context K/V gathered from
+                # the paged cache (new tokens were just written above)
+                k_cache, v_cache = kv_cache
+                k_all = k_cache[meta.gather_blk, :, meta.gather_off]
+                v_all = v_cache[meta.gather_blk, :, meta.gather_off]
+                o = ops.attn_prefill(q, k_all, v_all, meta.cu_seqlens,
+                                     meta.max_seqlen, self.scale,
+                                     window=self.window,
+                                     cu_seqlens_k=meta.cu_seqlens_k)
+            else:
+                o = ops.attn_prefill(q, k, v, meta.cu_seqlens,
+                                     meta.max_seqlen, self.scale,
+                                     window=self.window)
         else:
             k_cache, v_cache = kv_cache
             o = ops.paged_attn_decode(q, k_cache, v_cache, meta.block_tables,

@@ -105,14 +105,20 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  cu_seqlens: torch.Tensor, max_seqlen: int,
                  scale: float, causal: bool = True,
-                 window: int = 0) -> torch.Tensor:
+                 window: int = 0,
+                 cu_seqlens_k: torch.Tensor | None = None) -> torch.Tensor:
+    """Varlen attention. When cu_seqlens_k is given, q rows are the
+    SUFFIX of each sequence (cached-prefix prefill): kv rows cover the
+    full context and causal positions are offset by klen - qlen."""
+    if cu_seqlens_k is None:
+        cu_seqlens_k = cu_seqlens
     if q.is_cuda:
         out = torch.empty_like(q)
-        _native().attn_prefill(out, q, k, v, cu_seqlens, max_seqlen, scale,
-                               causal, window)
+        _native().attn_prefill(out, q, k, v, cu_seqlens, cu_seqlens_k,
+                               max_seqlen, scale, causal, window)
         return out
     return ref.attn_prefill(q, k, v, cu_seqlens, max_seqlen, scale, causal,
-                            window)
+                            window, cu_seqlens_k)
 
 
 # Flash-decoding split-K quantum (must match PART_QUANT in

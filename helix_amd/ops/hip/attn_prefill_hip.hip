@@ -49,16 +49,20 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
     const uint16_t* __restrict__ q,    // [T, Hq, D]
     const uint16_t* __restrict__ k,    // [T, Hkv, D]
     const uint16_t* __restrict__ v,    // [T, Hkv, D]
-    const int* __restrict__ cu_seqlens,  // [B+1]
+    const int* __restrict__ cu_seqlens_q,  // [B+1] query rows
+    const int* __restrict__ cu_seqlens_k,  // [B+1] kv rows (>= q rows)
     float scale, int Hq, int Hkv, int window) {
   const int qtile = blockIdx.x;
   const int seq = blockIdx.y;
   const int hq = blockIdx.z;
   const int hkv = hq / (Hq / Hkv);
-  const int seq_start = cu_seqlens[seq];
-  const int len = cu_seqlens[seq + 1] - seq_start;
+  const int q_start = cu_seqlens_q[seq];
+  const int qlen = cu_seqlens_q[seq + 1] - q_start;
+  const int k_start = cu_seqlens_k[seq];
+  const int klen = cu_seqlens_k[seq + 1] - k_start;
+  const int ctx = klen - qlen;   // cached-prefix offset (0 = plain prefill)
   const int qbase = qtile * QTILE;
-  if (qbase >= len) return;
+  if (qbase >= qlen) return;
 
   const int tid = threadIdx.x;
   const int wid = tid / WAVE;
@@ -100,10 +104,10 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 #pragma unroll
   for (int c = 0; c < nc; ++c) o_acc[c] = floatx4{0, 0, 0, 0};
 
-  const int kv_max = CAUSAL ? min(len, qbase + QTILE) : len;
+  const int kv_max = CAUSAL ? min(klen, ctx + qbase + QTILE) : klen;
   // sliding window: rows in this q-tile never look below kv_lo
   const int kv_lo = (CAUSAL && window > 0)
-                        ? max(0, qbase - window + 1) : 0;
+                        ? max(0, ctx + qbase - window + 1) : 0;
   const int t_first = kv_lo / KTILE;
   const int ntiles = (kv_max + KTILE - 1) / KTILE;
 
@@ -118,9 +122,9 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
       const int tok = e / D;
       const int d0 = e % D;
       const int gtok = t * KTILE + tok;
-      if (gtok < len) {
+      if (gtok < klen) {
         const int64_t off =
-            ((int64_t)(seq_start + gtok) * Hkv + hkv) * D + d0;
+            ((int64_t)(k_start + gtok) * Hkv + hkv) * D + d0;
         k_reg[cch] = *reinterpret_cast<const u16x8*>(k + off);
         v_reg[cch] = *reinterpret_cast<const u16x8*>(v + off);
       } else {
@@ -174,7 +178,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
     float corr[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int qpos = qbase + wid * 16 + lane_hi * 4 + r;
+      const int qpos = ctx + qbase + wid * 16 + lane_hi * 4 + r;
       float s[KTILE / 16];
       float m_tile = -INFINITY;
 #pragma unroll
@@ -183,7 +187,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
         const int tp = kv_base + h * 16 + lane_lo;
         if ((CAUSAL && (tp > qpos ||
                         (window > 0 && qpos - tp >= window))) ||
-            tp >= len) s[h] = -1e30f;
+            tp >= klen) s[h] = -1e30f;
         m_tile = fmaxf(m_tile, s[h]);
       }
 #pragma unroll
@@ -251,8 +255,8 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = qbase + wid * 16 + lane_hi * 4 + r;
-      if (qrow < len) {
-        out[((int64_t)(seq_start + qrow) * Hq + hq) * D + c * 16 + lane_lo] =
+      if (qrow < qlen) {
+        out[((int64_t)(q_start + qrow) * Hq + hq) * D + c * 16 + lane_lo] =
             f32_to_bf16(o_acc[c][r] / fmaxf(l_run[r], 1e-20f));
       }
     }
@@ -280,7 +284,8 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                      (const uint16_t*)q.data_ptr(),                          \
                      (const uint16_t*)k.data_ptr(),                          \
                      (const uint16_t*)v.data_ptr(),                          \
-                     cu_seqlens.data_ptr<int>(), (float)scale, Hq, Hkv,      \
+                     cu_seqlens.data_ptr<int>(),                             \
+                     cu_seqlens_k.data_ptr<int>(), (float)scale, Hq, Hkv,    \
                      (int)window)
   if (causal) {
     if (D == 128) LAUNCH(1, 128); else LAUNCH(1, 64);

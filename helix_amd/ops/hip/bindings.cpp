@@ -22,8 +22,8 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
                        int64_t window);
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   torch::Tensor v, torch::Tensor cu_seqlens,
-                  int64_t max_seqlen, double scale, bool causal,
-                  int64_t window);
+                  torch::Tensor cu_seqlens_k, int64_t max_seqlen,
+                  double scale, bool causal, int64_t window);
 void sample_tokens(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperatures, torch::Tensor seeds);
 void gemm_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,

@@ -87,29 +87,35 @@ def reshape_and_cache(k: torch.Tensor, v: torch.Tensor,
 def attn_prefill(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                  cu_seqlens: torch.Tensor, max_seqlen: int,
                  scale: float, causal: bool = True,
-                 window: int = 0) -> torch.Tensor:
-    """Varlen causal GQA attention (optional sliding window).
-    q: [T, Hq, D], k/v: [T, Hkv, D]."""
+                 window: int = 0, cu_seqlens_k=None) -> torch.Tensor:
+    """Varlen causal GQA attention (optional sliding window and
+    cached-prefix query offset). q: [T, Hq, D], k/v: [Tk, Hkv, D]."""
     Hq, Hkv = q.shape[1], k.shape[1]
     G = Hq // Hkv
+    if cu_seqlens_k is None:
+        cu_seqlens_k = cu_seqlens
     out = torch.empty_like(q)
     for i in range(cu_seqlens.shape[0] - 1):
-        s, e = int(cu_seqlens[i]), int(cu_seqlens[i + 1])
-        L = e - s
-        qs = q[s:e].float()
-        ks = k[s:e].float().repeat_interleave(G, dim=1)
-        vs = v[s:e].float().repeat_interleave(G, dim=1)
-        mask = None
-        is_causal = causal
-        if causal and window > 0:
-            pos = torch.arange(L)
-            m = (pos[None, :] <= pos[:, None]) &                 (pos[:, None] - pos[None, :] < window)
-            mask = torch.where(m, 0.0, float("-inf"))
-            is_causal = False
+        qs0, qe = int(cu_seqlens[i]), int(cu_seqlens[i + 1])
+        ks0, ke = int(cu_seqlens_k[i]), int(cu_seqlens_k[i + 1])
+        qlen, klen = qe - qs0, ke - ks0
+        ctx = klen - qlen
+        qs = q[qs0:qe].float()
+        ks = k[ks0:ke].float().repeat_interleave(G, dim=1)
+        vs = v[ks0:ke].float().repeat_interleave(G, dim=1)
+        qpos = torch.arange(ctx, klen)
+        kpos = torch.arange(klen)
+        if causal:
+            m = kpos[None, :] <= qpos[:, None]
+            if window > 0:
+                m &= (qpos[:, None] - kpos[None, :]) < window
+        else:
+            m = torch.ones(qlen, klen, dtype=torch.bool)
+        mask = torch.where(m, 0.0, float("-inf"))
         o = torch.nn.functional.scaled_dot_product_attention(
             qs.transpose(0, 1), ks.transpose(0, 1), vs.transpose(0, 1),
-            attn_mask=mask, is_causal=is_causal, scale=scale)
-        out[s:e] = o.transpose(0, 1).to(q.dtype)
+            attn_mask=mask, scale=scale)
+        out[qs0:qe] = o.transpose(0, 1).to(q.dtype)
     return out
 
 

@@ -200,3 +200,68 @@ def test_sliding_window_engine_consistency():
                          SamplingParams(temperature=0.0, max_tokens=1,
                                         ignore_eos=True))[0]
     assert out2[0] == out[-1]
+
+
+def test_prefix_caching_reuses_blocks_and_matches():
+    """Second request with a shared prompt prefix reuses cached KV blocks
+    and produces IDENTICAL greedy output to the uncached run."""
+    torch.manual_seed(0)
+    cfg = EngineConfig(model="tiny", max_model_len=256, max_num_seqs=8,
+                       kv_cache_blocks=128, eos_token_id=-1,
+                       enable_prefix_caching=True)
+    eng = LLMEngine(cfg, device="cpu")
+    system = list(range(1, 49))           # 48 tokens = 3 full blocks
+    p1 = system + [100, 101, 102]
+    p2 = system + [200, 201]
+    sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+
+    out1 = eng.generate([p1], sp)[0]
+    free_after_1 = eng.num_free_blocks()
+    # second request: the 3 system blocks must come from cache
+    eng.add_request("r2", p2, sp)
+    eng.step()  # prefill
+    seq2 = eng.seqs["r2"]
+    assert seq2.cached_prefix == 48
+    while eng.has_work:
+        eng.step()
+    out2 = eng.seqs["r2"].output_ids
+
+    # uncached baseline: same model without prefix caching
+    cfg2 = EngineConfig(model="tiny", max_model_len=256, max_num_seqs=8,
+                        kv_cache_blocks=128, eos_token_id=-1,
+                        enable_prefix_caching=False)
+    eng2 = LLMEngine(cfg2, device="cpu")
+    base1 = eng2.generate([p1], sp)[0]
+    base2 = eng2.generate([p2], sp)[0]
+    assert out1 == base1
+    assert out2 == base2
+
+
+def test_prefix_cache_refcounts():
+    """Shared blocks survive one sequence's free and are reclaimed only
+    when the last reference drops."""
+    torch.manual_seed(0)
+    cfg = EngineConfig(model="tiny", max_model_len=256, max_num_seqs=8,
+                       kv_cache_blocks=64, eos_token_id=-1)
+    eng = LLMEngine(cfg, device="cpu")
+    system = list(range(1, 33))           # 2 full blocks
+    sp = SamplingParams(temperature=0.0, max_tokens=2, ignore_eos=True)
+    total = eng.num_free_blocks()
+    eng.generate([system + [7]], sp)
+    # finished seqs freed their refs; hashes stay until blocks reclaimed?
+    # blocks were freed to refcount 0 -> reclaimed, registry dropped
+    assert eng.num_free_blocks() == total
+    # two concurrent sequences sharing a live prefix
+    eng.add_request("a", system + [40], sp)
+    eng.step()   # prefill a (registers hashes)
+    eng.add_request("b", system + [50, 51], sp)
+    eng.step()   # prefill b -> shares a's blocks
+    assert eng.seqs["b"].cached_prefix == 32
+    shared = set(eng.seqs["a"].block_table[:2])
+    assert shared == set(eng.seqs["b"].block_table[:2])
+    eng.cancel("a")
+    # b still holds the shared blocks
+    assert all(eng.kv.allocator.ref.get(blk, 0) >= 1 for blk in shared)
+    while eng.has_work:
+        eng.step()
+    assert eng.num_free_blocks() == total

@@ -48,3 +48,33 @@ def test_decode_matches_fresh_prefill_gpu():
                          SamplingParams(temperature=0.0, max_tokens=1,
                                         ignore_eos=True))[0]
     assert out2[0] == out[-1]
+
+
+def test_prefix_caching_gpu_equivalence():
+    """Prefix-cached generation must match uncached greedy output on GPU
+    (exercises the gather + query-offset kernel path)."""
+    torch.manual_seed(3)
+    system = list(range(1, 49))
+    p1 = system + [100, 101, 102]
+    p2 = system + [200, 201]
+    sp = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+
+    def run(prefix_caching):
+        cfg = EngineConfig(model="tiny-gqa", max_model_len=512,
+                           max_num_seqs=8, kv_cache_blocks=256,
+                           eos_token_id=-1, seed=11,
+                           enable_prefix_caching=prefix_caching)
+        eng = LLMEngine(cfg, device="cuda:0")
+        o1 = eng.generate([p1], sp)[0]
+        eng.add_request("r2", p2, sp)
+        eng.step()
+        cached = eng.seqs["r2"].cached_prefix
+        while eng.has_work:
+            eng.step()
+        return o1, eng.seqs["r2"].output_ids, cached
+
+    o1a, o2a, cached_a = run(True)
+    o1b, o2b, cached_b = run(False)
+    assert cached_a == 48 and cached_b == 0
+    assert o1a == o1b
+    assert o2a == o2b

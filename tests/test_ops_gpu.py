@@ -252,3 +252,25 @@ def test_paged_attn_decode_sliding_window(window, lens):
     want = ref.paged_attn_decode(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
                                  seq_lens.cpu(), scale, window=window)
     assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("qlens,klens", [([32], [160]), ([7, 64], [7, 200]),
+                                         ([128], [128])])
+def test_attn_prefill_cached_prefix(qlens, klens):
+    """Query-offset prefill: q is the suffix, kv covers full context."""
+    torch.manual_seed(15)
+    hq, hkv, d = 32, 8, 128
+    Tq, Tk = sum(qlens), sum(klens)
+    q = torch.randn(Tq, hq, d, dtype=torch.bfloat16, device=DEV)
+    k = torch.randn(Tk, hkv, d, dtype=torch.bfloat16, device=DEV)
+    v = torch.randn(Tk, hkv, d, dtype=torch.bfloat16, device=DEV)
+    cu_q = torch.tensor([0] + list(torch.tensor(qlens).cumsum(0)),
+                        dtype=torch.int32, device=DEV)
+    cu_k = torch.tensor([0] + list(torch.tensor(klens).cumsum(0)),
+                        dtype=torch.int32, device=DEV)
+    scale = d ** -0.5
+    got = ops.attn_prefill(q, k, v, cu_q, max(qlens), scale,
+                           cu_seqlens_k=cu_k)
+    want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), cu_q.cpu(),
+                            max(qlens), scale, cu_seqlens_k=cu_k.cpu())
+    assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
