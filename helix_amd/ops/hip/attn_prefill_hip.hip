@@ -82,9 +82,9 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   bf16x8 q_frag[nkt];
 #pragma unroll
   for (int kt = 0; kt < nkt; ++kt) {
-    if (my_qrow < len) {
+    if (my_qrow < qlen) {
       const uint16_t* src =
-          q + ((int64_t)(seq_start + my_qrow) * Hq + hq) * D + kt * 32 +
+          q + ((int64_t)(q_start + my_qrow) * Hq + hq) * D + kt * 32 +
           lane_hi * 8;
       u16x8 raw = *reinterpret_cast<const u16x8*>(src);
       q_frag[kt] = *reinterpret_cast<bf16x8*>(&raw);
@@ -267,8 +267,8 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 
 void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   torch::Tensor v, torch::Tensor cu_seqlens,
-                  int64_t max_seqlen, double scale, bool causal,
-                  int64_t window) {
+                  torch::Tensor cu_seqlens_k, int64_t max_seqlen,
+                  double scale, bool causal, int64_t window) {
   const int Hq = q.size(1);
   const int D = q.size(2);
   const int Hkv = k.size(1);
