@@ -30,6 +30,9 @@ def main():
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--max-tokens", type=int, default=128)
     ap.add_argument("--device", default="cuda:0")
+    ap.add_argument("--shared-prefix", type=int, default=0,
+                    help="tokens of a common system prompt shared by all "
+                         "requests (exercises the prefix cache)")
     args = ap.parse_args()
 
     from helix_amd.runner.service import RunnerService
@@ -58,12 +61,18 @@ def main():
 
     random.seed(0)
     torch.manual_seed(0)
+    shared = {}
     t0 = time.monotonic()
     for i in range(args.requests):
         inst = insts[i % len(insts)]
         vocab = inst.engine.model_cfg.vocab_size
-        prompt = [random.randrange(3, vocab - 1)
-                  for _ in range(args.prompt_len)]
+        if args.shared_prefix and id(inst) not in shared:
+            shared[id(inst)] = [random.randrange(3, vocab - 1)
+                                for _ in range(args.shared_prefix)]
+        prefix = shared.get(id(inst), [])
+        prompt = prefix + [random.randrange(3, vocab - 1)
+                           for _ in range(args.prompt_len
+                                          - len(prefix))]
         inst.submit(f"q{i}", prompt,
                     SamplingParams(temperature=0.0,
                                    max_tokens=args.max_tokens,
@@ -82,7 +91,7 @@ def main():
             tpots.append((b - a) * 1000)
     import json
     print(json.dumps({
-        "models": models,
+        "models": models, "shared_prefix": args.shared_prefix,
         "requests": len(stats), "rate_rps": args.rate,
         "prompt_len": args.prompt_len, "max_tokens": args.max_tokens,
         "wall_s": round(wall, 2),
