@@ -342,3 +342,53 @@ def test_provider_resolution_order(stack):
     assert pm.resolve("unknown/model", "mock") == ("mock", "unknown/model")
     # 4. plain unknown name -> default provider
     assert pm.resolve("whatever", "mock") == ("mock", "whatever")
+
+
+def test_interrupt_before_streaming(stack):
+    """Cancel/new-turn on a session whose last interaction never streamed
+    (reference Zed e2e 'interrupt mid-stream' phase at unit scale)."""
+    _, client, _, key, store = stack
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "one"}]},
+            headers=H(key)) as r:
+        sid = json.loads(next(l for l in r.iter_lines()
+                              if l.startswith("data: "))[6:])["session_id"]
+    # mark the interaction as if it never completed
+    doc = store.list("interactions", parent=sid)[0]
+    doc["state"] = "waiting"
+    doc["response_message"] = ""
+    store.put("interactions", doc["id"], doc, parent=sid)
+    # a new turn must still work (history includes the unanswered prompt)
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+        "session_id": sid,
+        "messages": [{"role": "user", "content": "two"}]},
+            headers=H(key)) as r:
+        lines = [l for l in r.iter_lines() if l.startswith("data: ")]
+    assert lines[-1] == "data: [DONE]"
+    its = store.list("interactions", parent=sid, desc=False)
+    assert len(its) == 2
+    assert its[1]["state"] == "complete"
+
+
+def test_rapid_multi_turn(stack):
+    """Three rapid consecutive turns on one session stay consistent
+    (reference 'rapid 3-turn' e2e phase)."""
+    _, client, _, key, store = stack
+    sid = None
+    for i in range(3):
+        body = {"model": "mock-model",
+                "messages": [{"role": "user", "content": f"turn {i}"}]}
+        if sid:
+            body["session_id"] = sid
+        with client.stream("POST", "/api/v1/sessions/chat", json=body,
+                           headers=H(key)) as r:
+            head = json.loads(next(l for l in r.iter_lines()
+                                   if l.startswith("data: "))[6:])
+            sid = head["session_id"]
+            [l for l in r.iter_lines()]
+    its = store.list("interactions", parent=sid, desc=False)
+    assert len(its) == 3
+    assert all(i["state"] == "complete" for i in its)
+    assert [i["prompt_message"] for i in its] == \
+        ["turn 0", "turn 1", "turn 2"]
