@@ -383,10 +383,11 @@ def test_rapid_multi_turn(stack):
             body["session_id"] = sid
         with client.stream("POST", "/api/v1/sessions/chat", json=body,
                            headers=H(key)) as r:
-            head = json.loads(next(l for l in r.iter_lines()
-                                   if l.startswith("data: "))[6:])
+            head = None
+            for l in r.iter_lines():
+                if head is None and l.startswith("data: "):
+                    head = json.loads(l[6:])
             sid = head["session_id"]
-            [l for l in r.iter_lines()]
     its = store.list("interactions", parent=sid, desc=False)
     assert len(its) == 3
     assert all(i["state"] == "complete" for i in its)
