@@ -59,8 +59,12 @@ class TokenStream:
         self.q: asyncio.Queue = asyncio.Queue()
 
     def on_token(self, seq, token_id: int, finished: bool):
-        self.loop.call_soon_threadsafe(
-            self.q.put_nowait, (token_id, finished, seq.finish_reason))
+        try:
+            self.loop.call_soon_threadsafe(
+                self.q.put_nowait, (token_id, finished, seq.finish_reason))
+        except RuntimeError:
+            # client's event loop is gone (disconnect) — tokens go nowhere
+            pass
 
     async def __aiter__(self):
         while True:

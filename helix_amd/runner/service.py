@@ -158,7 +158,14 @@ class LLMInstance:
                 self.wake.wait(timeout=0.05)
                 self.wake.clear()
                 continue
-            self.engine.step()
+            try:
+                self.engine.step()
+            except Exception:
+                # a step failure (bad callback, transient OOM) must not
+                # kill the serving loop; affected sequences error out,
+                # the rest keep going
+                log.exception("engine step failed (%s)", self.spec.name)
+                time.sleep(0.01)
 
     def submit(self, seq_id: str, prompt_ids: List[int],
                params: SamplingParams, on_token) -> None:
