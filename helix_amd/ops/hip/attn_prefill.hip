@@ -38,8 +38,14 @@ __device__ __forceinline__ int k_swz(int tok, int byte_in_row) {
 }
 
 // V tile: TRANSPOSED [dim][KTILE], 128B rows; swizzle on slot bits.
+// The swizzle mixes dim>>3 so lanes of ONE scalar-write instruction
+// (which share dim&7) still land in different banks — with the plain
+// (dim&7) swizzle the XOR was constant per instruction and writes
+// collapsed to the 4 tok values per wave (16-way conflicts; PMC showed
+// 5.7e8 conflicts/dispatch, profiles/r01_decode_profile_v1.md).
 __device__ __forceinline__ int v_swz(int dim, int byte_in_row) {
-  return dim * (KTILE * 2) + (byte_in_row ^ ((dim & 7) << 4));
+  const int s = ((dim >> 3) ^ dim) & 7;
+  return dim * (KTILE * 2) + (byte_in_row ^ (s << 4));
 }
 
 template <int CAUSAL, int D>
