@@ -349,9 +349,9 @@ def load_gguf_weights(model, path: str) -> int:
             continue
         t = g.load_tensor(name)
         if role == "q":
-            t = unpermute_rope(t, cfg.num_attention_heads)
+            t = unpermute_rope(t, cfg.num_heads)
         elif role == "k":
-            t = unpermute_rope(t, cfg.num_key_value_heads)
+            t = unpermute_rope(t, cfg.num_kv_heads)
         p = params[our]
         t = t.to(p.dtype)
         if role == "q":

@@ -102,6 +102,21 @@ def load_llama_weights(model, ckpt_dir: str, use_async: bool = True):
     model. On GPU, tensors are staged through a pinned-host buffer and
     copied on a side stream (hipMemcpyAsync) so decode streams on the
     default stream are not serialized behind weight traffic."""
+    # GGUF checkpoints (llama.cpp ecosystem) route through the native
+    # GGUF loader (engine/gguf.py): file path or a dir holding one.
+    gguf_path = None
+    if ckpt_dir.endswith(".gguf") and os.path.isfile(ckpt_dir):
+        gguf_path = ckpt_dir
+    elif os.path.isdir(ckpt_dir):
+        ggufs = sorted(glob.glob(os.path.join(ckpt_dir, "*.gguf")))
+        if ggufs and not glob.glob(os.path.join(ckpt_dir, "*.safetensors")):
+            gguf_path = ggufs[0]
+    if gguf_path is not None:
+        from . import gguf as _gguf
+        n = _gguf.load_gguf_weights(model, gguf_path)
+        log.info("loaded %d tensors from GGUF %s", n, gguf_path)
+        return n
+
     params = dict(model.named_parameters())
     cfg = model.cfg
     q, kv = cfg.q_size, cfg.kv_size
