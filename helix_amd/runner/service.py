@@ -38,6 +38,7 @@ class ModelSpec:
     max_num_seqs: int = 64
     kv_cache_blocks: Optional[int] = None
     kv_memory_fraction: float = 0.30   # of total HBM, for auto-sizing
+    tp: int = 1                        # tensor-parallel degree (xGMI group)
 
 
 DEFAULT_SPECS = {
@@ -282,6 +283,9 @@ class RunnerService:
             t0 = time.time()
             if spec.kind == "embedding":
                 inst = EmbeddingInstance(spec, self.device)
+            elif spec.tp > 1:
+                from helix_amd.runner.tp_instance import TPLLMInstance
+                inst = TPLLMInstance(spec, spec.tp)
             else:
                 inst = LLMInstance(spec, self.device)
             log.info("loaded %s in %.1fs", model, time.time() - t0)

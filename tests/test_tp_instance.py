@@ -1,0 +1,111 @@
+"""TP serving instance over gloo (world_size=2, CPU): the spawned
+per-GPU worker daemon must stream the exact tokens of a single-process
+engine (runner-plane TP serving; SURVEY.md §2.8 "Model containers" /
+§2.6)."""
+import os
+import threading
+import time
+
+import pytest
+import torch
+
+from helix_amd.engine.engine import EngineConfig, LLMEngine
+from helix_amd.engine.sampling_params import SamplingParams
+from helix_amd.models.llama import LlamaForCausalLM, PRESETS
+from helix_amd.runner.service import ModelSpec
+
+
+def _collect_stream(inst, seq_id, prompt, params):
+    done = threading.Event()
+    toks = []
+
+    def cb(seq, tok, fin):
+        if not fin or seq.finish_reason not in ("abort",):
+            toks.append(tok)
+        if fin:
+            done.set()
+
+    inst.submit(seq_id, prompt, params, cb)
+    assert done.wait(timeout=120), "stream did not finish"
+    return toks
+
+
+@pytest.mark.timeout(300)
+def test_tp_instance_matches_single_process(tmp_path):
+    from helix_amd.runner.tp_instance import TPLLMInstance
+    cfg = PRESETS["tiny-gqa"]
+    torch.manual_seed(0)
+    full = LlamaForCausalLM(cfg).float()
+    full.init_random(0)
+    sd_path = str(tmp_path / "full_sd.pt")
+    torch.save(full.state_dict(), sd_path)
+
+    prompts = [[1, 2, 3, 4, 5], [9, 8, 7]]
+    sp = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+    eng = LLMEngine(EngineConfig(model="tiny-gqa", max_model_len=256,
+                                 max_num_seqs=4, kv_cache_blocks=128,
+                                 eos_token_id=-1),
+                    device="cpu", model=full)
+    want = eng.generate(prompts, sp)
+
+    spec = ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=256,
+                     max_num_seqs=4, kv_cache_blocks=128, tp=2)
+    inst = TPLLMInstance(spec, 2, device_type="cpu", backend="gloo",
+                         sd_path=sd_path)
+    try:
+        got0 = _collect_stream(inst, "s0", prompts[0], sp)
+        got1 = _collect_stream(inst, "s1", prompts[1], sp)
+        assert got0 == want[0]
+        assert got1 == want[1]
+        # concurrent submissions (continuous batching across ranks)
+        res = {}
+        done = {}
+        for i, p in enumerate(prompts):
+            done[i] = threading.Event()
+            res[i] = []
+
+            def mk(i):
+                def cb(seq, tok, fin):
+                    res[i].append(tok)
+                    if fin:
+                        done[i].set()
+                return cb
+            inst.submit(f"c{i}", p, sp, mk(i))
+        for i in done:
+            assert done[i].wait(timeout=120)
+        assert res[0] == want[0]
+        assert res[1] == want[1]
+        assert inst.in_flight == 0
+    finally:
+        inst.shutdown()
+    assert all(p.exitcode == 0 for p in inst.procs)
+
+
+@pytest.mark.timeout(300)
+def test_tp_instance_via_runner_service():
+    """RunnerService loads a tp=2 spec through the TP daemon path and
+    serves a request end-to-end (self-consistent random init)."""
+    from helix_amd.runner.service import RunnerService
+    spec = ModelSpec("tiny-tp2", "llm", "tiny-gqa", max_model_len=256,
+                     max_num_seqs=4, kv_cache_blocks=128, tp=2)
+    svc = RunnerService(device="cpu", specs={"tiny-tp2": spec})
+    try:
+        inst = svc.ensure_loaded("tiny-tp2")
+        sp = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+        toks = _collect_stream(inst, "r0", [1, 2, 3], sp)
+        assert len(toks) == 6
+        assert "tiny-tp2" in svc.loaded_models()
+        # cancel path: a long request aborts cleanly
+        ev = threading.Event()
+
+        def cb(seq, tok, fin):
+            if fin:
+                ev.set()
+        inst.submit("r1", [4, 5, 6],
+                    SamplingParams(temperature=0.0, max_tokens=4096,
+                                   ignore_eos=True), cb)
+        time.sleep(0.3)
+        inst.cancel("r1")
+        assert ev.wait(timeout=60)
+    finally:
+        svc.shutdown()
