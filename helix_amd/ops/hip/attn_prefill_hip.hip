@@ -219,13 +219,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
     for (int h = 0; h < KTILE / 16; ++h) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        // column-group XOR swizzle keyed on the q-row: varies across
-        // the lanes of one scalar-write instruction AND across the
-        // rows of one vector-read instruction (16B-aligned: swizzle
-        // bits are >= bit 3 of the half index)
-        const int prow = lane_hi * 4 + r;
-        p_lds[wid][prow * P_PITCH +
-                   ((h * 16 + lane_lo) ^ ((prow & 7) << 3))] =
+        p_lds[wid][(lane_hi * 4 + r) * P_PITCH + h * 16 + lane_lo] =
             f32_to_bf16(p_val[h][r]);
       }
     }
@@ -235,8 +229,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 #pragma unroll
     for (int kk = 0; kk < KTILE / 32; ++kk) {
       u16x8 raw = *reinterpret_cast<const u16x8*>(
-          &p_lds[wid][lane_lo * P_PITCH +
-                      ((kk * 32 + lane_hi * 8) ^ ((lane_lo & 7) << 3))]);
+          &p_lds[wid][lane_lo * P_PITCH + kk * 32 + lane_hi * 8]);
       p_frag[kk] = *reinterpret_cast<bf16x8*>(&raw);
     }
 #pragma unroll
