@@ -379,6 +379,19 @@ def create_app(cfg: Optional[ServerConfig] = None,
         controller.delete_session(session_id)
         return {"ok": True}
 
+    @app.post("/api/v1/auth/token")
+    async def issue_token(request: Request,
+                          user: AuthUser = Depends(auth_dep)):
+        """Exchange an API key for a short-lived HS256 JWT (the
+        reference's Keycloak-token flow, self-issued here)."""
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        ttl = min(int(body.get("ttl_s", 3600)), 86400)
+        return {"access_token": auth.issue_jwt(user, ttl),
+                "token_type": "Bearer", "expires_in": ttl}
+
     @app.websocket("/api/v1/ws/user")
     async def ws_user(ws: WebSocket):
         token = ws.query_params.get("access_token", "")
@@ -891,6 +904,35 @@ def create_app(cfg: Optional[ServerConfig] = None,
         except Exception:
             payload = {}
         return await triggers.fire(doc, payload)
+
+    @app.post("/api/v1/slack/events/{tid}")
+    async def slack_events(tid: str, request: Request):
+        """Slack Events API inbound (reference api/pkg/trigger slack):
+        url_verification challenge + signature-verified message events."""
+        doc = store.get("triggers", tid)
+        if doc is None or doc.get("kind") != "slack":
+            raise HTTPException(404, "slack trigger not found")
+        body = await request.body()
+        try:
+            return await triggers.handle_slack_event(
+                doc, body,
+                request.headers.get("X-Slack-Request-Timestamp", ""),
+                request.headers.get("X-Slack-Signature", ""))
+        except PermissionError:
+            raise HTTPException(401, "bad slack signature")
+
+    @app.post("/api/v1/teams/webhook/{tid}")
+    async def teams_webhook(tid: str, request: Request):
+        """Teams outgoing-webhook inbound (HMAC of raw body)."""
+        doc = store.get("triggers", tid)
+        if doc is None or doc.get("kind") != "teams":
+            raise HTTPException(404, "teams trigger not found")
+        body = await request.body()
+        try:
+            return await triggers.handle_teams_event(
+                doc, body, request.headers.get("Authorization", ""))
+        except PermissionError:
+            raise HTTPException(401, "bad teams hmac")
 
     # ------------------------------------------------------------------
     # Filestore (reference api/pkg/filestore routes)

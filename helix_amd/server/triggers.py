@@ -1,9 +1,14 @@
 """Trigger manager (parity with api/pkg/trigger: cron triggers via a
-5-field cron parser, webhook triggers; each fires an agent session turn).
+5-field cron parser, webhook triggers, and signature-verified Slack /
+Teams inbound events; each fires an agent session turn).
 """
 from __future__ import annotations
 
 import asyncio
+import base64
+import hashlib
+import hmac
+import json
 import logging
 import time
 from typing import List, Optional
@@ -122,3 +127,83 @@ class TriggerManager:
             except Exception:
                 log.exception("trigger loop error")
             await asyncio.sleep(interval)
+
+
+# -- inbound chat-platform events (reference api/pkg/trigger slack/teams)
+
+def verify_slack_signature(signing_secret: str, timestamp: str,
+                           body: bytes, signature: str,
+                           now: Optional[float] = None) -> bool:
+    """Slack Events API v0 signing: HMAC-SHA256 of "v0:{ts}:{body}",
+    replay-bounded to 5 minutes."""
+    try:
+        ts = float(timestamp)
+    except (TypeError, ValueError):
+        return False
+    if abs((now or time.time()) - ts) > 300:
+        return False
+    base = f"v0:{timestamp}:".encode() + body
+    want = "v0=" + hmac.new(signing_secret.encode(), base,
+                            hashlib.sha256).hexdigest()
+    return hmac.compare_digest(want, signature or "")
+
+
+def verify_teams_hmac(security_token_b64: str, body: bytes,
+                      auth_header: str) -> bool:
+    """Teams outgoing-webhook auth: base64 HMAC-SHA256 of the raw body
+    with the base64-decoded security token, sent as "HMAC <b64>"."""
+    try:
+        key = base64.b64decode(security_token_b64)
+    except Exception:
+        return False
+    want = "HMAC " + base64.b64encode(
+        hmac.new(key, body, hashlib.sha256).digest()).decode()
+    return hmac.compare_digest(want, auth_header or "")
+
+
+class SlackTeamsMixin:
+    """Inbound-event handling mixed into TriggerManager."""
+
+    async def handle_slack_event(self, doc: dict, body: bytes,
+                                 timestamp: str, signature: str) -> dict:
+        secret = doc.get("config", {}).get("signing_secret", "")
+        if secret and not verify_slack_signature(secret, timestamp, body,
+                                                 signature):
+            raise PermissionError("bad slack signature")
+        try:
+            payload = json.loads(body or b"{}")
+        except Exception:
+            payload = {}
+        if payload.get("type") == "url_verification":
+            return {"challenge": payload.get("challenge", "")}
+        if payload.get("type") == "event_callback":
+            ev = payload.get("event", {})
+            # ignore our own / other bots' messages (loop prevention)
+            if ev.get("type") in ("app_mention", "message") \
+                    and not ev.get("bot_id"):
+                res = await self.fire(doc, {
+                    "source": "slack", "channel": ev.get("channel"),
+                    "user": ev.get("user"), "text": ev.get("text", "")})
+                return {"ok": True, **res}
+        return {"ok": True, "ignored": True}
+
+    async def handle_teams_event(self, doc: dict, body: bytes,
+                                 auth_header: str) -> dict:
+        token = doc.get("config", {}).get("security_token", "")
+        if token and not verify_teams_hmac(token, body, auth_header):
+            raise PermissionError("bad teams hmac")
+        try:
+            payload = json.loads(body or b"{}")
+        except Exception:
+            payload = {}
+        text = payload.get("text", "")
+        res = await self.fire(doc, {
+            "source": "teams", "from": payload.get("from", {}).get("name"),
+            "text": text})
+        return {"type": "message",
+                "text": f"Started session {res['session_id']}"}
+
+
+# attach to TriggerManager (kept as a separate block for readability)
+TriggerManager.handle_slack_event = SlackTeamsMixin.handle_slack_event
+TriggerManager.handle_teams_event = SlackTeamsMixin.handle_teams_event
