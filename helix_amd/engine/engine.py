@@ -13,6 +13,7 @@ from dataclasses import dataclass, field
 from enum import Enum
 from typing import Callable, Dict, List, Optional
 
+import numpy as np
 import torch
 
 from helix_amd import ops
@@ -39,6 +40,7 @@ class Sequence:
     output_ids: List[int] = field(default_factory=list)
     status: SeqStatus = SeqStatus.WAITING
     block_table: List[int] = field(default_factory=list)
+    row: int = -1                          # persistent block-table row slot
     cached_prefix: int = 0                 # tokens served from prefix cache
     block_hashes: List[bytes] = field(default_factory=list)
     arrival_time: float = field(default_factory=time.monotonic)
@@ -102,6 +104,18 @@ class LLMEngine:
                 self.model_cfg.head_dim, cfg.max_model_len, self.device)
         else:
             self.decode_ws = None
+        # Persistent host-side batch state: per-seq block-table rows and
+        # scratch arrays. Rebuilding these as Python-list -> torch.tensor
+        # per step cost ~4 ms at B=512 (512 small tensor constructions);
+        # numpy row updates are O(changed entries) instead.
+        mb = cfg.max_model_len // cfg.block_size + 2
+        self._bt_np = np.zeros((cfg.max_num_seqs, mb), dtype=np.int32)
+        self._free_rows = list(range(cfg.max_num_seqs - 1, -1, -1))
+        self._ids_np = np.zeros(cfg.max_num_seqs, dtype=np.int64)
+        self._pos_np = np.zeros(cfg.max_num_seqs, dtype=np.int64)
+        self._slots_np = np.zeros(cfg.max_num_seqs, dtype=np.int64)
+        self._lens_np = np.zeros(cfg.max_num_seqs, dtype=np.int32)
+        self._rows_np = np.zeros(cfg.max_num_seqs, dtype=np.intp)
         self.graph_runner = None
         if self.device.type == "cuda" and not cfg.enforce_eager:
             from helix_amd.engine.graph_runner import CUDAGraphRunner
@@ -150,6 +164,7 @@ class LLMEngine:
             self.running.remove(seq)
             self.kv.allocator.free(seq.block_table)
             seq.block_table = []
+            self._release_row(seq)
         seq.status = SeqStatus.CANCELLED
         seq.finish_reason = "cancelled"
 
@@ -215,6 +230,7 @@ class LLMEngine:
                 break
             seq.block_table = [alloc.share(b) for b in matched] + \
                 alloc.allocate(new_blocks)
+            self._assign_row(seq)
             seq.cached_prefix = cached
             batch.append(self.waiting.pop(0))
             tokens += need
@@ -223,6 +239,18 @@ class LLMEngine:
     def _slot(self, seq: Sequence, pos: int) -> int:
         bs = self.cfg.block_size
         return seq.block_table[pos // bs] * bs + pos % bs
+
+    def _assign_row(self, seq: Sequence):
+        if seq.row < 0:
+            seq.row = self._free_rows.pop()
+        n = len(seq.block_table)
+        self._bt_np[seq.row, :n] = seq.block_table
+        self._bt_np[seq.row, n:] = 0
+
+    def _release_row(self, seq: Sequence):
+        if seq.row >= 0:
+            self._free_rows.append(seq.row)
+            seq.row = -1
 
     def _run_prefill(self, batch: List[Sequence]):
         bs = self.cfg.block_size
@@ -283,6 +311,7 @@ class LLMEngine:
         victim = self.running.pop()          # newest first
         self.kv.allocator.free(victim.block_table)
         victim.block_table = []
+        self._release_row(victim)
         # re-admit with its generated tokens folded into the prompt so the
         # next prefill recomputes the full context
         victim.prompt_ids = victim.prompt_ids + victim.output_ids
@@ -298,7 +327,9 @@ class LLMEngine:
         dev = self.device
         batch = []
         preempted = set()                     # ids preempted this step
-        input_ids, positions, slots, seq_lens = [], [], [], []
+        nb_ = 0
+        ids_np, pos_np = self._ids_np, self._pos_np
+        slots_np, lens_np, rows_np = self._slots_np, self._lens_np, self._rows_np
         for seq in list(self.running):
             if seq.seq_id in preempted:
                 continue
@@ -315,36 +346,38 @@ class LLMEngine:
                         break                 # we preempted ourselves
                     continue
                 seq.block_table.extend(self.kv.allocator.allocate(1))
+                self._bt_np[seq.row, len(seq.block_table) - 1] = \
+                    seq.block_table[-1]
             if seq.seq_id in preempted:
                 continue
             batch.append(seq)
-            input_ids.append(seq.output_ids[-1] if seq.output_ids
-                             else seq.prompt_ids[-1])
-            positions.append(pos)
-            slots.append(self._slot(seq, pos))
-            seq_lens.append(pos + 1)
+            ids_np[nb_] = (seq.output_ids[-1] if seq.output_ids
+                           else seq.prompt_ids[-1])
+            pos_np[nb_] = pos
+            slots_np[nb_] = self._slot(seq, pos)
+            lens_np[nb_] = pos + 1
+            rows_np[nb_] = seq.row
+            nb_ += 1
         if not batch:
             return []
+        B = nb_
         max_blocks = max(len(s.block_table) for s in batch)
-        bt_cpu = torch.zeros(len(batch), max_blocks, dtype=torch.int32)
-        for i, seq in enumerate(batch):
-            bt_cpu[i, :len(seq.block_table)] = torch.tensor(
-                seq.block_table, dtype=torch.int32)
+        # one C-speed gather of the persistent rows (no per-row tensors)
+        bt_np = self._bt_np[rows_np[:B], :max_blocks]
+        max_len = int(lens_np[:B].max())
         if self.graph_runner is not None:
-            logits = self.graph_runner.run(input_ids, positions, slots,
-                                           bt_cpu, seq_lens, max(seq_lens))
+            logits = self.graph_runner.run(ids_np[:B], pos_np[:B],
+                                           slots_np[:B], bt_np,
+                                           lens_np[:B], max_len)
         else:
             meta = DecodeMeta(
-                block_tables=bt_cpu.to(dev),
-                seq_lens=torch.tensor(seq_lens, dtype=torch.int32,
-                                      device=dev),
-                slot_mapping=torch.tensor(slots, dtype=torch.int64,
-                                          device=dev),
-                positions=torch.tensor(positions, dtype=torch.int64,
-                                       device=dev),
-                max_len=max(seq_lens),
+                block_tables=torch.from_numpy(bt_np).to(dev),
+                seq_lens=torch.from_numpy(lens_np[:B].copy()).to(dev),
+                slot_mapping=torch.from_numpy(slots_np[:B].copy()).to(dev),
+                positions=torch.from_numpy(pos_np[:B].copy()).to(dev),
+                max_len=max_len,
                 workspace=self.decode_ws)
-            ids = torch.tensor(input_ids, dtype=torch.int64, device=dev)
+            ids = torch.from_numpy(ids_np[:B].copy()).to(dev)
             hidden = self.model(ids, self.kv.caches, meta)
             logits = self.model.compute_logits(hidden)
         tokens = self._sample(batch, logits)
@@ -357,17 +390,19 @@ class LLMEngine:
         needs_proc = any(s.params.needs_logit_processing for s in batch)
         if needs_proc:
             logits = self._process_logits(batch, logits.float())
-        temps = torch.tensor([s.params.temperature for s in batch],
-                             dtype=torch.float32, device=logits.device)
-        seeds = []
-        for s in batch:
+        B = len(batch)
+        temps_np = np.empty(B, dtype=np.float32)
+        seeds_np = np.empty(B, dtype=np.int64)
+        for i, s in enumerate(batch):
+            temps_np[i] = s.params.temperature
             base = s.params.seed if s.params.seed is not None else (
                 hash(s.seq_id) & 0x7FFFFFFF)
-            seeds.append((base + len(s.output_ids) * _SEED_MIX)
-                         & 0x7FFFFFFFFFFFFFFF)
-        seeds_t = torch.tensor(seeds, dtype=torch.int64, device=logits.device)
+            seeds_np[i] = ((base + len(s.output_ids) * _SEED_MIX)
+                           & 0x7FFFFFFFFFFFFFFF)
+        temps = torch.from_numpy(temps_np).to(logits.device)
+        seeds_t = torch.from_numpy(seeds_np).to(logits.device)
         toks = ops.sample_tokens(logits.contiguous(), temps, seeds_t)
-        return [int(t) for t in toks.cpu()]
+        return toks.cpu().tolist()
 
     def _process_logits(self, batch, logits: torch.Tensor) -> torch.Tensor:
         for i, seq in enumerate(batch):
@@ -423,6 +458,7 @@ class LLMEngine:
             self.running.remove(seq)
             self.kv.allocator.free(seq.block_table)
             seq.block_table = []
+            self._release_row(seq)
             return True
         return False
 

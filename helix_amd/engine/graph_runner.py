@@ -90,7 +90,7 @@ class CUDAGraphRunner:
 
     def run(self, input_ids, positions, slots, block_tables, seq_lens,
             max_len: int) -> torch.Tensor:
-        """All args host lists / small tensors; returns logits[:B]."""
+        """All args numpy arrays (or lists); returns logits[:B]."""
         B = len(input_ids)
         nb = self.batch_bucket(B)
         lb = self.len_bucket(max_len)
@@ -103,12 +103,13 @@ class CUDAGraphRunner:
         self.h_slots[:nb] = -1
         self.h_lens[:nb] = 1
         self.h_bt[:nb].zero_()
-        self.h_ids[:B] = torch.tensor(input_ids, dtype=torch.int64)
-        self.h_pos[:B] = torch.tensor(positions, dtype=torch.int64)
-        self.h_slots[:B] = torch.tensor(slots, dtype=torch.int64)
-        self.h_lens[:B] = torch.tensor(seq_lens, dtype=torch.int32)
-        w = block_tables.shape[1]
-        self.h_bt[:B, :w] = block_tables
+        self.h_ids[:B] = torch.as_tensor(input_ids, dtype=torch.int64)
+        self.h_pos[:B] = torch.as_tensor(positions, dtype=torch.int64)
+        self.h_slots[:B] = torch.as_tensor(slots, dtype=torch.int64)
+        self.h_lens[:B] = torch.as_tensor(seq_lens, dtype=torch.int32)
+        bt_t = torch.as_tensor(block_tables)
+        w = bt_t.shape[1]
+        self.h_bt[:B, :w] = bt_t
         self.in_ids[:nb].copy_(self.h_ids[:nb], non_blocking=True)
         self.in_pos[:nb].copy_(self.h_pos[:nb], non_blocking=True)
         self.in_slots[:nb].copy_(self.h_slots[:nb], non_blocking=True)

@@ -95,6 +95,8 @@ def create_app(cfg: Optional[ServerConfig] = None,
                             agent_runner=agent_runner, usage=usage)
     triggers = TriggerManager(store, controller)
     rbac = RBACService(store)
+    from helix_amd.server.org_runtime import OrgRuntime
+    org_rt = OrgRuntime(store, controller)
     oauth = OAuthManager(store)
     profiles = ProfileService(store)
     git_svc = GitService(store, cfg.filestore.path)
@@ -124,6 +126,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.filestore = filestore
     app.state.triggers = triggers
     app.state.rbac = rbac
+    app.state.org_rt = org_rt
     app.state.oauth = oauth
     app.state.profiles = profiles
     app.state.git = git_svc
@@ -716,6 +719,84 @@ def create_app(cfg: Optional[ServerConfig] = None,
                               user: AuthUser = Depends(auth_dep)):
         body = await request.json()
         return rbac.add_team_member(tid, body["user_id"])
+
+    # -- org runtime: positions / bots / streams (reference api/pkg/org)
+    def _org_member(oid: str, user: AuthUser):
+        if rbac.member_role(oid, user.id) is None and not user.admin:
+            raise HTTPException(403, "not a member")
+
+    @app.post("/api/v1/organizations/{oid}/positions")
+    async def create_position(oid: str, request: Request,
+                              user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        b = await request.json()
+        return org_rt.create_position(
+            oid, b.get("name", ""), b.get("role", "worker"),
+            b.get("app_id", ""), b.get("system_prompt", ""),
+            b.get("model", ""))
+
+    @app.get("/api/v1/organizations/{oid}/positions")
+    async def list_positions(oid: str,
+                             user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        return org_rt.list_positions(oid)
+
+    @app.post("/api/v1/organizations/{oid}/bots")
+    async def create_bot(oid: str, request: Request,
+                         user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        b = await request.json()
+        try:
+            return org_rt.create_bot(oid, b.get("name", ""),
+                                     b.get("position_id", ""), user.id)
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+
+    @app.get("/api/v1/organizations/{oid}/bots")
+    async def list_bots(oid: str, user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        return org_rt.list_bots(oid)
+
+    @app.post("/api/v1/bots/{bid}/subscribe")
+    async def bot_subscribe(bid: str, request: Request,
+                            user: AuthUser = Depends(auth_dep)):
+        b = await request.json()
+        bot = org_rt.get_bot(bid)
+        if bot is None:
+            raise HTTPException(404, "bot not found")
+        _org_member(bot["org_id"], user)
+        try:
+            return org_rt.subscribe(bid, b.get("stream_id", ""))
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+
+    @app.post("/api/v1/organizations/{oid}/streams")
+    async def create_stream(oid: str, request: Request,
+                            user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        b = await request.json()
+        return org_rt.create_stream(oid, b.get("name", ""),
+                                    b.get("topic", ""))
+
+    @app.get("/api/v1/organizations/{oid}/streams")
+    async def list_streams(oid: str, user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        return org_rt.list_streams(oid)
+
+    @app.post("/api/v1/streams/{sid}/messages")
+    async def post_stream_message(sid: str, request: Request,
+                                  user: AuthUser = Depends(auth_dep)):
+        b = await request.json()
+        try:
+            return await org_rt.post_message(sid, user.username or user.id,
+                                             b.get("text", ""))
+        except ValueError as e:
+            raise HTTPException(404, str(e))
+
+    @app.get("/api/v1/streams/{sid}/messages")
+    async def get_stream_messages(sid: str,
+                                  user: AuthUser = Depends(auth_dep)):
+        return org_rt.stream_messages(sid)
 
     @app.post("/api/v1/access-grants")
     async def create_grant(request: Request,

@@ -21,6 +21,7 @@ _TABLES = [
     "runner_profiles", "projects", "spec_tasks", "git_repositories",
     "evaluation_runs", "system_settings", "wallets", "transactions",
     "rag_chunks", "usage_rollups", "runner_assignments",
+    "org_positions", "org_bots", "org_streams", "org_messages",
 ]
 
 

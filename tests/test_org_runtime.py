@@ -1,0 +1,105 @@
+"""Org runtime: positions/bots/streams with message fan-out to bots
+(reference "helix-org" api/pkg/org runtime)."""
+import pytest
+from fastapi.testclient import TestClient
+
+from helix_amd.server.app import create_app
+from helix_amd.server.config import ServerConfig
+from helix_amd.server.providers import MockClient, ProviderManager
+from helix_amd.store import Store
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    cfg.filestore.path = str(tmp_path / "fs")
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    pm.register("mock", MockClient())
+    app = create_app(cfg, store=store, providers=pm)
+    client = TestClient(app)
+    r = client.post("/api/v1/users", json={"username": "alice"},
+                    headers={"Authorization": "Bearer admin-key"})
+    key = r.json()["api_key"]
+    return client, key, store
+
+
+def H(key):
+    return {"Authorization": f"Bearer {key}"}
+
+
+def _mk_org(client, key):
+    return client.post("/api/v1/organizations", json={"name": "acme"},
+                       headers=H(key)).json()["id"]
+
+
+def test_position_bot_stream_crud(stack):
+    client, key, _ = stack
+    oid = _mk_org(client, key)
+    pos = client.post(f"/api/v1/organizations/{oid}/positions",
+                      json={"name": "support", "role": "worker",
+                            "system_prompt": "Be helpful."},
+                      headers=H(key)).json()
+    assert pos["org_id"] == oid
+    bots = client.post(f"/api/v1/organizations/{oid}/bots",
+                       json={"name": "helper", "position_id": pos["id"]},
+                       headers=H(key)).json()
+    assert bots["position_id"] == pos["id"]
+    strm = client.post(f"/api/v1/organizations/{oid}/streams",
+                       json={"name": "general"}, headers=H(key)).json()
+    assert client.get(f"/api/v1/organizations/{oid}/positions",
+                      headers=H(key)).json()[0]["name"] == "support"
+    assert client.get(f"/api/v1/organizations/{oid}/bots",
+                      headers=H(key)).json()[0]["name"] == "helper"
+    assert client.get(f"/api/v1/organizations/{oid}/streams",
+                      headers=H(key)).json()[0]["name"] == "general"
+    # bad position -> 400
+    r = client.post(f"/api/v1/organizations/{oid}/bots",
+                    json={"name": "x", "position_id": "nope"},
+                    headers=H(key))
+    assert r.status_code == 400
+    # non-member denied
+    r2 = client.post("/api/v1/users", json={"username": "mallory"},
+                     headers={"Authorization": "Bearer admin-key"})
+    mkey = r2.json()["api_key"]
+    r = client.get(f"/api/v1/organizations/{oid}/bots", headers=H(mkey))
+    assert r.status_code == 403
+    assert strm["org_id"] == oid
+
+
+def test_stream_message_fans_out_to_subscribed_bots(stack):
+    client, key, store = stack
+    oid = _mk_org(client, key)
+    pos = client.post(f"/api/v1/organizations/{oid}/positions",
+                      json={"name": "support",
+                            "system_prompt": "Answer tersely."},
+                      headers=H(key)).json()
+    bot = client.post(f"/api/v1/organizations/{oid}/bots",
+                      json={"name": "helper", "position_id": pos["id"]},
+                      headers=H(key)).json()
+    strm = client.post(f"/api/v1/organizations/{oid}/streams",
+                       json={"name": "general"}, headers=H(key)).json()
+    other = client.post(f"/api/v1/organizations/{oid}/streams",
+                        json={"name": "random"}, headers=H(key)).json()
+    client.post(f"/api/v1/bots/{bot['id']}/subscribe",
+                json={"stream_id": strm["id"]}, headers=H(key))
+
+    msg = client.post(f"/api/v1/streams/{strm['id']}/messages",
+                      json={"text": "What is our refund policy?"},
+                      headers=H(key)).json()
+    assert len(msg["replies"]) == 1
+    assert msg["replies"][0]["bot"] == "helper"
+    assert msg["replies"][0]["text"] == "mock response"
+    # bot session was created and turn counted
+    b = store.get("org_bots", bot["id"])
+    assert b["turns"] == 1
+    # unsubscribed stream does not fan out
+    msg2 = client.post(f"/api/v1/streams/{other['id']}/messages",
+                       json={"text": "hello?"}, headers=H(key)).json()
+    assert msg2["replies"] == []
+    # message history is ordered and complete
+    hist = client.get(f"/api/v1/streams/{strm['id']}/messages",
+                      headers=H(key)).json()
+    assert [m["text"] for m in hist] == ["What is our refund policy?"]
