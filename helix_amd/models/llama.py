@@ -119,16 +119,15 @@ class Attention(nn.Module):
                 meta) -> torch.Tensor:
         T = x.shape[0]
         qkv = self.qkv_proj(x)
-        q_sz, kv_sz = self.nh * self.hd, self.nkv * self.hd
-        q, k, v = qkv.split([q_sz, kv_sz, kv_sz], dim=-1)
-        q, k = q.contiguous(), k.contiguous()
-        q, k = ops.rotary_embedding(meta.positions, q, k, cos_sin, self.hd)
-        q = q.view(T, self.nh, self.hd)
-        k = k.view(T, self.nkv, self.hd)
-        v = v.contiguous().view(T, self.nkv, self.hd)
-        if kv_cache is not None:
-            k_cache, v_cache = kv_cache
-            ops.reshape_and_cache(k, v, k_cache, v_cache, meta.slot_mapping)
+        # fused strided split + rope + paged-cache write (one kernel, no
+        # .contiguous() copies); compact K/V emitted only when the
+        # fresh-prefill attention path consumes them directly
+        want_kv = meta.is_prefill and \
+            getattr(meta, "cu_seqlens_k", None) is None
+        q, k, v = ops.rope_qkv_cache(
+            meta.positions, qkv, cos_sin, self.nh, self.nkv, self.hd,
+            kv_cache=kv_cache, slot_mapping=meta.slot_mapping,
+            want_kv=want_kv)
         if meta.is_prefill:
             if meta.cu_seqlens_k is not None:
                 # cached-prefix prefill: full-context K/V gathered from

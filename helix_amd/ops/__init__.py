@@ -76,6 +76,46 @@ def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
     return q, k
 
 
+def rope_qkv_cache(positions: torch.Tensor, qkv: torch.Tensor,
+                   cos_sin: torch.Tensor, num_q: int, num_kv: int,
+                   head_dim: int, kv_cache=None, slot_mapping=None,
+                   want_kv: bool = False):
+    """Fused QKV epilogue: strided split + rope + paged-cache write.
+
+    Returns (q [T, num_q, head_dim], k, v) — k/v are None unless
+    want_kv (fresh-prefill path needs compact K/V for attn_prefill).
+    """
+    T = qkv.shape[0]
+    if qkv.is_cuda:
+        q_out = torch.empty(T, num_q * head_dim, dtype=qkv.dtype,
+                            device=qkv.device)
+        k_out = v_out = None
+        if want_kv:
+            k_out = torch.empty(T, num_kv * head_dim, dtype=qkv.dtype,
+                                device=qkv.device)
+            v_out = torch.empty_like(k_out)
+        kc = vc = None
+        if kv_cache is not None:
+            kc, vc = kv_cache
+        _native().rope_qkv_cache(positions, qkv, q_out, kc, vc,
+                                 slot_mapping, k_out, v_out, cos_sin,
+                                 num_q, num_kv, head_dim)
+        return (q_out.view(T, num_q, head_dim),
+                k_out.view(T, num_kv, head_dim) if want_kv else None,
+                v_out.view(T, num_kv, head_dim) if want_kv else None)
+    # CPU reference composition (same semantics)
+    q_sz, kv_sz = num_q * head_dim, num_kv * head_dim
+    q, k, v = qkv.split([q_sz, kv_sz, kv_sz], dim=-1)
+    q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+    q, k = rotary_embedding(positions, q, k, cos_sin, head_dim)
+    q = q.view(T, num_q, head_dim)
+    k = k.view(T, num_kv, head_dim)
+    v = v.view(T, num_kv, head_dim)
+    if kv_cache is not None:
+        reshape_and_cache(k, v, kv_cache[0], kv_cache[1], slot_mapping)
+    return q, (k if want_kv else None), (v if want_kv else None)
+
+
 def silu_and_mul(x: torch.Tensor) -> torch.Tensor:
     if x.is_cuda:
         i = x.shape[-1] // 2

@@ -10,6 +10,15 @@ void rotary_embedding(torch::Tensor positions, torch::Tensor q,
                       torch::Tensor k, torch::Tensor cos_sin,
                       int64_t head_dim);
 void silu_and_mul(torch::Tensor out, torch::Tensor x);
+void rope_qkv_cache(torch::Tensor positions, torch::Tensor qkv,
+                    torch::Tensor q_out,
+                    c10::optional<torch::Tensor> k_cache,
+                    c10::optional<torch::Tensor> v_cache,
+                    c10::optional<torch::Tensor> slot_mapping,
+                    c10::optional<torch::Tensor> k_out,
+                    c10::optional<torch::Tensor> v_out,
+                    torch::Tensor cos_sin, int64_t num_q_heads,
+                    int64_t num_kv_heads, int64_t head_dim);
 void gelu_tanh(torch::Tensor out, torch::Tensor x);
 void reshape_and_cache(torch::Tensor k, torch::Tensor v,
                        torch::Tensor k_cache, torch::Tensor v_cache,
@@ -38,6 +47,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rotary_embedding", &rotary_embedding,
         "Apply rotary embedding to q,k in-place");
   m.def("silu_and_mul", &silu_and_mul, "Fused SiLU-gated multiply");
+  m.def("rope_qkv_cache", &rope_qkv_cache,
+        "Fused strided-QKV rope + paged-cache write");
   m.def("gelu_tanh", &gelu_tanh, "GELU (tanh approx)");
   m.def("reshape_and_cache", &reshape_and_cache,
         "Scatter K/V rows into paged cache");
