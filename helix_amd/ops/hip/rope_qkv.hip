@@ -126,5 +126,5 @@ void rope_qkv_cache(torch::Tensor positions, torch::Tensor qkv,
       k_out.has_value() ? (uint16_t*)k_out->data_ptr() : nullptr,
       v_out.has_value() ? (uint16_t*)v_out->data_ptr() : nullptr,
       cos_sin.data_ptr<float>(), (int)num_q_heads, (int)num_kv_heads,
-      (int)head_dim);
+      (int)head_dim, block_size);
 }

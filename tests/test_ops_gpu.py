@@ -274,3 +274,43 @@ def test_attn_prefill_cached_prefix(qlens, klens):
     want = ref.attn_prefill(q.cpu(), k.cpu(), v.cpu(), cu_q.cpu(),
                             max(qlens), scale, cu_seqlens_k=cu_k.cpu())
     assert_close_bf16(got, want, atol=3e-2, rtol=3e-2)
+
+
+@pytest.mark.parametrize("want_kv,use_cache", [(True, True), (False, True),
+                                               (True, False)])
+def test_rope_qkv_cache_fused(want_kv, use_cache):
+    """Fused strided-QKV rope+cache kernel vs the unfused composition."""
+    torch.manual_seed(0)
+    T, hq, hkv, d, bs = 33, 8, 2, 64, 16
+    qkv = torch.randn(T, (hq + 2 * hkv) * d, dtype=torch.bfloat16,
+                      device="cuda")
+    pos = torch.randint(0, 100, (T,), dtype=torch.int64, device="cuda")
+    cs = ref.make_cos_sin_cache(d, 128).cuda()
+    nb = 8
+    slots = torch.arange(bs, bs + T, dtype=torch.int64, device="cuda") \
+        if use_cache else torch.full((T,), -1, dtype=torch.int64,
+                                     device="cuda")
+    kc = torch.zeros(nb, hkv, bs, d, dtype=torch.bfloat16, device="cuda")
+    vc = torch.zeros_like(kc)
+    cache = (kc, vc) if use_cache else None
+    q, k, v = ops.rope_qkv_cache(pos, qkv, cs, hq, hkv, d,
+                                 kv_cache=cache, slot_mapping=slots,
+                                 want_kv=want_kv)
+    # unfused reference path
+    q_sz, kv_sz = hq * d, hkv * d
+    q2, k2, v2 = qkv.split([q_sz, kv_sz, kv_sz], dim=-1)
+    q2, k2 = q2.contiguous(), k2.contiguous()
+    q2, k2 = ops.rotary_embedding(pos, q2.clone(), k2.clone(), cs, d)
+    v2 = v2.contiguous()
+    torch.testing.assert_close(q.reshape(T, -1), q2, atol=2e-2, rtol=2e-2)
+    if want_kv:
+        torch.testing.assert_close(k.reshape(T, -1), k2, atol=2e-2,
+                                   rtol=2e-2)
+        torch.testing.assert_close(v.reshape(T, -1), v2, atol=0, rtol=0)
+    if use_cache:
+        kc2 = torch.zeros_like(kc)
+        vc2 = torch.zeros_like(vc)
+        ops.reshape_and_cache(k2.view(T, hkv, d), v2.view(T, hkv, d),
+                              kc2, vc2, slots)
+        torch.testing.assert_close(kc, kc2, atol=2e-2, rtol=2e-2)
+        torch.testing.assert_close(vc, vc2, atol=0, rtol=0)
