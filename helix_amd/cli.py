@@ -352,5 +352,72 @@ def test(file: str = typer.Option("helix.yaml", "-f", "--file"),
     raise typer.Exit(1 if failures else 0)
 
 
+@app.command("export-gguf")
+def export_gguf(preset: str = typer.Option("llama3-8b"),
+                ckpt: str = typer.Option("", help="safetensors dir "
+                                         "(random-init if empty)"),
+                out: str = typer.Option(..., "-o", "--out"),
+                seed: int = typer.Option(0)):
+    """Export a model as a GGUF v3 checkpoint (llama.cpp tensor names,
+    Q/K rows permuted for GGML rope — interops with the llama.cpp
+    ecosystem; see engine/gguf.py)."""
+    import torch
+    from helix_amd.engine import gguf as g
+    from helix_amd.models.llama import LlamaForCausalLM, PRESETS
+
+    cfg = PRESETS[preset]
+    model = LlamaForCausalLM(cfg)
+    if ckpt:
+        from helix_amd.engine.weights import load_llama_weights
+        load_llama_weights(model, ckpt, use_async=False)
+    else:
+        model.init_random(seed)
+
+    def permute(w, n_head):
+        o, i = w.shape
+        return (w.reshape(n_head, 2, o // n_head // 2, i)
+                .swapaxes(1, 2).reshape(o, i))
+
+    sd = dict(model.named_parameters())
+    q, kv, inter = cfg.q_size, cfg.kv_size, cfg.intermediate_size
+    tensors = {
+        "token_embd.weight": sd["embed_tokens.weight"].data,
+        "output_norm.weight": sd["final_norm_w"].data,
+        "output.weight": sd["lm_head.weight"].data,
+    }
+    for i in range(cfg.num_layers):
+        qkv = sd[f"layers.{i}.attn.qkv_proj.weight"].data
+        gu = sd[f"layers.{i}.mlp.gate_up_proj.weight"].data
+        tensors[f"blk.{i}.attn_q.weight"] = permute(qkv[:q], cfg.num_heads)
+        tensors[f"blk.{i}.attn_k.weight"] = permute(qkv[q:q + kv],
+                                                    cfg.num_kv_heads)
+        tensors[f"blk.{i}.attn_v.weight"] = qkv[q + kv:]
+        tensors[f"blk.{i}.attn_output.weight"] = \
+            sd[f"layers.{i}.attn.o_proj.weight"].data
+        tensors[f"blk.{i}.ffn_gate.weight"] = gu[:inter]
+        tensors[f"blk.{i}.ffn_up.weight"] = gu[inter:]
+        tensors[f"blk.{i}.ffn_down.weight"] = \
+            sd[f"layers.{i}.mlp.down_proj.weight"].data
+        tensors[f"blk.{i}.attn_norm.weight"] = \
+            sd[f"layers.{i}.input_norm_w"].data
+        tensors[f"blk.{i}.ffn_norm.weight"] = \
+            sd[f"layers.{i}.post_norm_w"].data
+    meta = {
+        "general.architecture": "llama",
+        "general.name": cfg.name,
+        "llama.block_count": cfg.num_layers,
+        "llama.embedding_length": cfg.hidden_size,
+        "llama.feed_forward_length": cfg.intermediate_size,
+        "llama.attention.head_count": cfg.num_heads,
+        "llama.attention.head_count_kv": cfg.num_kv_heads,
+        "llama.context_length": cfg.max_position,
+        "llama.rope.freq_base": float(cfg.rope_base),
+    }
+    g.write_gguf(out, meta, tensors)
+    est = g.estimate_gguf_bytes(out)
+    typer.echo(f"wrote {out}: {len(tensors)} tensors, "
+               f"{est['weights'] >> 20} MiB")
+
+
 if __name__ == "__main__":
     app()

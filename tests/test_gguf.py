@@ -125,3 +125,24 @@ def test_bad_magic(tmp_path):
     p.write_bytes(b"NOPE" + struct.pack("<I", 3))
     with pytest.raises(ValueError):
         gguf.GGUFFile(str(p))
+
+
+def test_weights_loader_dispatches_gguf(tmp_path):
+    """load_llama_weights auto-detects a GGUF checkpoint dir."""
+    import subprocess
+    import sys
+    out = str(tmp_path / "tiny.gguf")
+    r = subprocess.run(
+        [sys.executable, "-m", "helix_amd.cli", "export-gguf",
+         "--preset", "tiny", "-o", out, "--seed", "3"],
+        capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr
+    from helix_amd.engine.weights import load_llama_weights
+    torch.manual_seed(0)
+    src = LlamaForCausalLM(PRESETS["tiny"])
+    src.init_random(3)
+    dst = LlamaForCausalLM(PRESETS["tiny"])
+    n = load_llama_weights(dst, out)
+    assert n > 0
+    for name, p in src.named_parameters():
+        assert torch.equal(p.data, dict(dst.named_parameters())[name].data), name
