@@ -314,3 +314,36 @@ def test_rope_qkv_cache_fused(want_kv, use_cache):
                               kc2, vc2, slots)
         torch.testing.assert_close(kc, kc2, atol=2e-2, rtol=2e-2)
         torch.testing.assert_close(vc, vc2, atol=0, rtol=0)
+
+
+def test_mfma_probe_fp8_layout():
+    """16x16x128 MX-fp8 MFMA fragment layout vs torch e4m3 reference
+    (asymmetric operands; unity e8m0 scales 0x7F = 2^0)."""
+    torch.manual_seed(1)
+    a = (torch.randn(16, 128) * 0.5).to(torch.float8_e4m3fn)
+    b = (torch.randn(128, 16) * 0.5).to(torch.float8_e4m3fn)
+    want = a.float() @ b.float()
+    d = torch.zeros(16, 16, dtype=torch.float32, device="cuda")
+    from helix_amd.ops import _native
+    _native().mfma_probe_fp8(d, a.view(torch.uint8).cuda(),
+                             b.view(torch.uint8).cuda(),
+                             0x7F7F7F7F, 0x7F7F7F7F)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(d.cpu(), want, atol=1e-2, rtol=1e-2)
+
+
+def test_mfma_probe_fp8_scale_doubles():
+    """e8m0 scale byte 0x80 (=2^1) on A must exactly double the result
+    relative to 0x7F (=2^0) — validates the unity-scale trick the fp8
+    GEMM epilogue-dequant design relies on."""
+    torch.manual_seed(2)
+    a = (torch.randn(16, 128) * 0.5).to(torch.float8_e4m3fn)
+    b = (torch.randn(128, 16) * 0.5).to(torch.float8_e4m3fn)
+    from helix_amd.ops import _native
+    d1 = torch.zeros(16, 16, dtype=torch.float32, device="cuda")
+    d2 = torch.zeros_like(d1)
+    au, bu = a.view(torch.uint8).cuda(), b.view(torch.uint8).cuda()
+    _native().mfma_probe_fp8(d1, au, bu, 0x7F7F7F7F, 0x7F7F7F7F)
+    _native().mfma_probe_fp8(d2, au, bu, 0x80808080, 0x7F7F7F7F)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(d2, d1 * 2.0)
