@@ -209,6 +209,37 @@ def sample_tokens(logits: torch.Tensor, temperatures: torch.Tensor,
     return ref.sample_tokens(logits, temperatures, seeds)
 
 
+def quantize_fp8(t: torch.Tensor, dim: int = -1):
+    """Per-row (dim=-1 reduces over the last axis) e4m3 quantization.
+    Returns (bytes uint8 view, float32 scales)."""
+    amax = t.float().abs().amax(dim=dim, keepdim=True).clamp(min=1e-8)
+    scale = amax / 448.0                      # e4m3fn max normal
+    q = (t.float() / scale).to(torch.float8_e4m3fn)
+    return q.view(torch.uint8), scale.squeeze(dim).float()
+
+
+def gemm_fp8(x8: torch.Tensor, w8: torch.Tensor, x_scale: torch.Tensor,
+             w_scale: torch.Tensor, bias: torch.Tensor | None = None,
+             act: int = 0) -> torch.Tensor:
+    """out[M,N] = dequant(x8 @ w8^T): e4m3 inputs as uint8 views with
+    per-row activation scales and per-output-channel weight scales
+    (epilogue dequant; MFMA runs at the 2x fp8 rate)."""
+    M, N = x8.shape[0], w8.shape[0]
+    if x8.is_cuda:
+        out = torch.empty(M, N, dtype=torch.bfloat16, device=x8.device)
+        _native().gemm_fp8(out, x8, w8, x_scale.contiguous(),
+                           w_scale.contiguous(), bias, act)
+        return out
+    xf = x8.view(torch.float8_e4m3fn).float() * x_scale[:, None]
+    wf = w8.view(torch.float8_e4m3fn).float() * w_scale[:, None]
+    out = xf @ wf.t()
+    if bias is not None:
+        out = out + bias.float()
+    if act == 1:
+        out = torch.nn.functional.gelu(out, approximate="tanh")
+    return out.to(torch.bfloat16)
+
+
 def gemm_bf16(x: torch.Tensor, w: torch.Tensor, bias=None,
               act: int = 0) -> torch.Tensor:
     if x.is_cuda:

@@ -37,6 +37,9 @@ void sample_tokens(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperatures, torch::Tensor seeds);
 void gemm_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias, int64_t act);
+void gemm_fp8(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+              torch::Tensor x_scale, torch::Tensor w_scale,
+              c10::optional<torch::Tensor> bias, int64_t act);
 void mfma_probe(torch::Tensor d, torch::Tensor a, torch::Tensor b);
 void mfma_probe_fp8(torch::Tensor d, torch::Tensor a, torch::Tensor b,
                     int64_t scale_a, int64_t scale_b);
@@ -60,6 +63,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Varlen causal flash prefill attention (MFMA)");
   m.def("sample_tokens", &sample_tokens, "Greedy/Gumbel token sampling");
   m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM: x @ w^T (+bias, act)");
+  m.def("gemm_fp8", &gemm_fp8,
+        "MX-fp8 e4m3 MFMA GEMM with epilogue per-row/col dequant");
   m.def("mfma_probe", &mfma_probe, "16x16x32 MFMA layout probe");
   m.def("mfma_probe_fp8", &mfma_probe_fp8,
         "16x16x128 MX-fp8 MFMA layout probe (unity e8m0 scales)");

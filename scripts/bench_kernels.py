@@ -83,6 +83,23 @@ def bench_gemm():
               f"({fl/t2/1e12:7.1f} TF)")
 
 
+def bench_fp8():
+    print("== gemm_fp8 (MX e4m3, epilogue dequant) vs bf16 paths ==")
+    for (M, N, K) in [(4096, 4096, 4096), (8192, 8192, 8192),
+                      (512, 28672, 4096), (512, 6144, 4096),
+                      (16384, 4096, 4096)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) * 0.05
+        x8, xs = ops.quantize_fp8(x)
+        w8, ws = ops.quantize_fp8(w)
+        t1 = timeit(lambda: ops.gemm_fp8(x8, w8, xs, ws), iters=30)
+        t2 = timeit(lambda: torch.nn.functional.linear(x, w), iters=30)
+        fl = 2 * M * N * K
+        print(f"M{M:5d} N{N:5d} K{K:5d}: fp8 {t1*1e6:9.1f} us "
+              f"({fl/t1/1e12:7.1f} TF) | hipblaslt-bf16 {t2*1e6:9.1f} us "
+              f"({fl/t2/1e12:7.1f} TF)")
+
+
 def bench_norm():
     print("== rms_norm / fused_add_rms_norm / silu / rope ==")
     for T in [64, 8192]:
@@ -119,7 +136,8 @@ if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     torch.manual_seed(0)
     fns = {"decode": bench_decode, "prefill": bench_prefill,
-           "gemm": bench_gemm, "norm": bench_norm, "sample": bench_sample}
+           "gemm": bench_gemm, "fp8": bench_fp8, "norm": bench_norm,
+           "sample": bench_sample}
     if which == "all":
         for f in fns.values():
             f()

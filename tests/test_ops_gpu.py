@@ -347,3 +347,37 @@ def test_mfma_probe_fp8_scale_doubles():
     _native().mfma_probe_fp8(d2, au, bu, 0x80808080, 0x7F7F7F7F)
     torch.cuda.synchronize()
     torch.testing.assert_close(d2, d1 * 2.0)
+
+
+def test_gemm_fp8_numerics():
+    """fp8 GEMM with epilogue dequant vs fp32 reference of the quantized
+    operands (exact modulo fp32 accumulation order)."""
+    torch.manual_seed(0)
+    M, N, K = 96, 200, 256
+    x = torch.randn(M, K) * 2.0
+    w = torch.randn(N, K) * 0.05
+    x8, xs = ops.quantize_fp8(x)
+    w8, ws = ops.quantize_fp8(w)
+    want = ops.gemm_fp8(x8, w8, xs, ws)            # CPU reference path
+    got = ops.gemm_fp8(x8.cuda(), w8.cuda(), xs.cuda(), ws.cuda()).cpu()
+    torch.testing.assert_close(got.float(), want.float(), atol=0.05,
+                               rtol=0.05)
+    # and it approximates the bf16 matmul (quantization error bounded)
+    ref = (x @ w.t()).float()
+    rel = (got.float() - ref).norm() / ref.norm()
+    assert rel < 0.05, f"fp8 rel error {rel}"
+
+
+def test_gemm_fp8_bias_gelu():
+    torch.manual_seed(1)
+    M, N, K = 64, 64, 128
+    x = torch.randn(M, K)
+    w = torch.randn(N, K) * 0.1
+    bias = torch.randn(N).bfloat16()
+    x8, xs = ops.quantize_fp8(x)
+    w8, ws = ops.quantize_fp8(w)
+    want = ops.gemm_fp8(x8, w8, xs, ws, bias=bias, act=1)
+    got = ops.gemm_fp8(x8.cuda(), w8.cuda(), xs.cuda(), ws.cuda(),
+                       bias=bias.cuda(), act=1).cpu()
+    torch.testing.assert_close(got.float(), want.float(), atol=0.08,
+                               rtol=0.08)
