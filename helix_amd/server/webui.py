@@ -1,24 +1,52 @@
-"""Minimal built-in web UI (the reference ships a 237k-LoC React app;
-this is a functional single-page chat + admin view served at / so the
-stack is usable from a browser without a separate frontend build)."""
+"""Built-in web UI (the reference ships a 237k-LoC React frontend; this
+is a dependency-free single-page app served at / covering the main
+panes: chat sessions with SSE streaming, apps/helix.yaml, knowledge,
+runner dashboard, and usage — so the stack is usable from a browser
+without a separate frontend build)."""
 
 INDEX_HTML = """<!doctype html>
 <html><head><meta charset="utf-8"><title>helix_amd</title>
 <style>
  body{font-family:system-ui,sans-serif;margin:0;display:flex;height:100vh}
- #side{width:260px;background:#111;color:#eee;padding:12px;overflow-y:auto}
- #side h1{font-size:16px} #side div.s{padding:6px;cursor:pointer;border-radius:6px;font-size:13px}
+ #side{width:260px;background:#111;color:#eee;padding:12px;overflow-y:auto;
+       display:flex;flex-direction:column}
+ #side h1{font-size:16px;margin:4px 0 10px}
+ #tabs{display:flex;flex-wrap:wrap;gap:4px;margin-bottom:10px}
+ #tabs button{background:#333;color:#eee;border:0;padding:6px 10px;
+   border-radius:6px;cursor:pointer;font-size:12px}
+ #tabs button.on{background:#4a7}
+ #side div.s{padding:6px;cursor:pointer;border-radius:6px;font-size:13px}
  #side div.s:hover{background:#333}
  #main{flex:1;display:flex;flex-direction:column}
+ .pane{flex:1;display:none;flex-direction:column;overflow:hidden}
+ .pane.on{display:flex}
  #log{flex:1;overflow-y:auto;padding:16px;background:#f7f7f8}
- .msg{max-width:760px;margin:8px auto;padding:10px 14px;border-radius:10px;white-space:pre-wrap}
+ .msg{max-width:760px;margin:8px auto;padding:10px 14px;border-radius:10px;
+   white-space:pre-wrap}
  .user{background:#d8e8ff} .assistant{background:#fff;border:1px solid #ddd}
  #bar{display:flex;padding:12px;gap:8px;border-top:1px solid #ddd}
  #inp{flex:1;padding:10px;font-size:15px}
- input,select,button{font-size:14px}
- #cfg{padding:8px 12px;display:flex;gap:8px;background:#eee;align-items:center}
+ input,select,button,textarea{font-size:14px}
+ #cfg{padding:8px 12px;display:flex;gap:8px;background:#eee;
+   align-items:center}
+ .lst{flex:1;overflow-y:auto;padding:16px}
+ table{border-collapse:collapse;width:100%;font-size:13px}
+ td,th{border:1px solid #ddd;padding:6px 8px;text-align:left}
+ textarea{width:100%;height:180px;font-family:monospace}
+ .card{background:#fff;border:1px solid #ddd;border-radius:8px;
+   padding:12px;margin-bottom:12px}
+ .pill{display:inline-block;padding:2px 8px;border-radius:10px;
+   font-size:11px;background:#dfd}
+ .pill.err{background:#fdd}
 </style></head><body>
 <div id="side"><h1>helix_amd</h1>
+ <div id="tabs">
+  <button data-p="chat" class="on">Chat</button>
+  <button data-p="apps">Apps</button>
+  <button data-p="knw">Knowledge</button>
+  <button data-p="run">Runners</button>
+  <button data-p="use">Usage</button>
+ </div>
  <button onclick="newSession()">+ new session</button>
  <div id="sessions"></div>
 </div>
@@ -28,38 +56,74 @@ INDEX_HTML = """<!doctype html>
   model <select id="model"></select>
   <span id="status"></span>
  </div>
- <div id="log"></div>
- <div id="bar">
-  <input id="inp" placeholder="Say something..." onkeydown="if(event.key==='Enter')send()">
-  <button onclick="send()">send</button>
+ <div class="pane on" id="p-chat">
+  <div id="log"></div>
+  <div id="bar">
+   <input id="inp" placeholder="Say something..."
+    onkeydown="if(event.key==='Enter')send()">
+   <button onclick="send()">send</button>
+  </div>
  </div>
+ <div class="pane" id="p-apps"><div class="lst">
+  <div class="card"><b>Apply helix.yaml / app config (JSON or CRD)</b>
+   <textarea id="appyaml">{"name":"my-agent","assistants":[{"name":"default",
+ "system_prompt":"You are helpful."}]}</textarea>
+   <button onclick="applyApp()">apply</button> <span id="appmsg"></span>
+  </div>
+  <div id="applist"></div>
+ </div></div>
+ <div class="pane" id="p-knw"><div class="lst">
+  <div class="card"><b>New knowledge source (text)</b><br>
+   name <input id="kname" value="notes">
+   <textarea id="ktext">Paste source text here.</textarea>
+   <button onclick="addKnowledge()">index</button>
+  </div>
+  <div id="knwlist"></div>
+ </div></div>
+ <div class="pane" id="p-run"><div class="lst" id="runlist"></div></div>
+ <div class="pane" id="p-use"><div class="lst" id="uselist"></div></div>
 </div>
 <script>
 let sessionId = null;
-const H = () => ({'Authorization':'Bearer '+document.getElementById('key').value,
+const $ = id => document.getElementById(id);
+const H = () => ({'Authorization':'Bearer '+$('key').value,
                   'Content-Type':'application/json'});
+for(const b of document.querySelectorAll('#tabs button')){
+  b.onclick = () => {
+    document.querySelectorAll('#tabs button').forEach(x=>x.classList
+      .remove('on'));
+    document.querySelectorAll('.pane').forEach(x=>x.classList.remove('on'));
+    b.classList.add('on');
+    $('p-'+b.dataset.p).classList.add('on');
+    ({apps:loadApps, knw:loadKnowledge, run:loadRunners,
+      use:loadUsage})[b.dataset.p]?.();
+  };
+}
 async function loadModels(){
   const r = await fetch('/v1/models',{headers:H()});
   if(!r.ok) return;
-  const sel = document.getElementById('model'); sel.innerHTML='';
+  const sel = $('model'); sel.innerHTML='';
   for(const m of (await r.json()).data){
-    const o=document.createElement('option');o.value=o.textContent=m.id;sel.appendChild(o);}
+    const o=document.createElement('option');
+    o.value=o.textContent=m.id;sel.appendChild(o);}
 }
 async function loadSessions(){
   const r = await fetch('/api/v1/sessions',{headers:H()});
   if(!r.ok) return;
-  const el = document.getElementById('sessions'); el.innerHTML='';
+  const el = $('sessions'); el.innerHTML='';
   for(const s of await r.json()){
-    const d=document.createElement('div');d.className='s';d.textContent=s.name;
+    const d=document.createElement('div');d.className='s';
+    d.textContent=s.name;
     d.onclick=()=>openSession(s.id);el.appendChild(d);}
 }
 function add(role, text){
-  const d=document.createElement('div');d.className='msg '+role;d.textContent=text;
-  document.getElementById('log').appendChild(d);
+  const d=document.createElement('div');d.className='msg '+role;
+  d.textContent=text;
+  $('log').appendChild(d);
   d.scrollIntoView();return d;}
-function newSession(){sessionId=null;document.getElementById('log').innerHTML='';}
+function newSession(){sessionId=null;$('log').innerHTML='';}
 async function openSession(id){
-  sessionId=id;document.getElementById('log').innerHTML='';
+  sessionId=id;$('log').innerHTML='';
   const r=await fetch('/api/v1/sessions/'+id,{headers:H()});
   const s=await r.json();
   for(const it of s.interactions||[]){
@@ -67,12 +131,12 @@ async function openSession(id){
     if(it.response_message)add('assistant',it.response_message);}
 }
 async function send(){
-  const inp=document.getElementById('inp');const text=inp.value.trim();
+  const inp=$('inp');const text=inp.value.trim();
   if(!text)return; inp.value='';
   add('user',text);
   const out=add('assistant','');
   const body={messages:[{role:'user',content:text}],
-              model:document.getElementById('model').value};
+              model:$('model').value};
   if(sessionId)body.session_id=sessionId;
   const r=await fetch('/api/v1/sessions/chat',{method:'POST',headers:H(),
                       body:JSON.stringify(body)});
@@ -91,6 +155,68 @@ async function send(){
       }catch(e){}
       out.scrollIntoView();
     }}
+}
+// ---- apps pane ----
+async function applyApp(){
+  let cfg;
+  try{cfg=JSON.parse($('appyaml').value);}catch(e){
+    $('appmsg').textContent='bad JSON: '+e;return;}
+  const r=await fetch('/api/v1/apps',{method:'POST',headers:H(),
+    body:JSON.stringify({config:cfg})});
+  $('appmsg').textContent=r.ok?'applied':'error '+r.status;
+  loadApps();
+}
+async function loadApps(){
+  const r=await fetch('/api/v1/apps',{headers:H()});if(!r.ok)return;
+  const el=$('applist');el.innerHTML='';
+  for(const a of await r.json()){
+    const d=document.createElement('div');d.className='card';
+    const cfg=a.config?.helix||a.config||{};
+    d.innerHTML='<b>'+(cfg.name||a.id)+'</b> <span class="pill">'+
+      (cfg.assistants?.length||0)+' assistants</span><br><small>'+a.id+
+      '</small>';
+    el.appendChild(d);}
+}
+// ---- knowledge pane ----
+async function addKnowledge(){
+  await fetch('/api/v1/knowledge',{method:'POST',headers:H(),
+    body:JSON.stringify({name:$('kname').value,
+      source:{text:$('ktext').value}})});
+  loadKnowledge();
+}
+async function loadKnowledge(){
+  const r=await fetch('/api/v1/knowledge',{headers:H()});if(!r.ok)return;
+  const el=$('knwlist');el.innerHTML='';
+  for(const k of await r.json()){
+    const d=document.createElement('div');d.className='card';
+    d.innerHTML='<b>'+k.name+'</b> <span class="pill'+
+      (k.state==='error'?' err':'')+'">'+k.state+'</span>'+
+      (k.version?' v'+k.version:'');
+    el.appendChild(d);}
+}
+// ---- runners pane ----
+async function loadRunners(){
+  const r=await fetch('/api/v1/admin/runners',{headers:H()});
+  const el=$('runlist');
+  if(!r.ok){el.textContent='admin key required';return;}
+  const rs=await r.json();el.innerHTML='';
+  if(!rs.length){el.textContent='no runners connected';return;}
+  for(const x of rs){
+    const d=document.createElement('div');d.className='card';
+    const models=(x.models||[]).map(m=>m.model_id||m).join(', ');
+    d.innerHTML='<b>'+x.id+'</b> <span class="pill">'+
+      (x.status||'ready')+'</span><br>GPU: '+(x.gpu||'?')+
+      '<br>models: '+models;
+    el.appendChild(d);}
+}
+// ---- usage pane ----
+async function loadUsage(){
+  const r=await fetch('/api/v1/usage',{headers:H()});
+  const el=$('uselist');
+  if(!r.ok){el.textContent='no usage yet';return;}
+  const u=await r.json();
+  el.innerHTML='<div class="card"><pre>'+
+    JSON.stringify(u,null,2).slice(0,4000)+'</pre></div>';
 }
 loadModels();loadSessions();
 </script></body></html>"""
