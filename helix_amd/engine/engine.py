@@ -67,6 +67,7 @@ class EngineConfig:
     seed: int = 0
     enforce_eager: bool = False               # disable hipGraph capture
     enable_prefix_caching: bool = True
+    quantization: Optional[str] = None        # None | "fp8" (e4m3 W8A8)
 
 
 class LLMEngine:
@@ -85,6 +86,9 @@ class LLMEngine:
             model = LlamaForCausalLM(self.model_cfg, tp_size, tp_rank)
             model = model.to(dtype).to(self.device)
             model.init_random(cfg.seed)
+        if cfg.quantization == "fp8":
+            from helix_amd.models.quant import quantize_model_fp8
+            quantize_model_fp8(model)
         self.model = model
 
         nkv = self.model_cfg.num_kv_heads // tp_size

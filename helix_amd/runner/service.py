@@ -39,6 +39,7 @@ class ModelSpec:
     kv_cache_blocks: Optional[int] = None
     kv_memory_fraction: float = 0.30   # of total HBM, for auto-sizing
     tp: int = 1                        # tensor-parallel degree (xGMI group)
+    quantization: Optional[str] = None  # None | "fp8"
 
 
 DEFAULT_SPECS = {
@@ -72,7 +73,9 @@ def estimate_model_bytes(spec: ModelSpec, block_size: int = 16) -> int:
     per_layer = (cfg.hidden_size * (cfg.q_size + 2 * cfg.kv_size)  # qkv
                  + cfg.q_size * cfg.hidden_size                     # o
                  + 3 * cfg.hidden_size * cfg.intermediate_size)     # mlp
-    weights = 2 * (emb + cfg.num_layers * per_layer)
+    wbytes = 1 if spec.quantization == "fp8" else 2
+    # fp8 halves the projection weights (embeddings/lm_head stay bf16)
+    weights = 2 * emb + wbytes * cfg.num_layers * per_layer
     kv_block = 2 * cfg.num_layers * cfg.num_kv_heads * block_size * \
         cfg.head_dim * 2
     if spec.kv_cache_blocks:
@@ -101,7 +104,8 @@ class LLMInstance:
         self.engine = LLMEngine(
             EngineConfig(model=spec.preset, max_model_len=spec.max_model_len,
                          max_num_seqs=spec.max_num_seqs,
-                         kv_cache_blocks=kv_blocks),
+                         kv_cache_blocks=kv_blocks,
+                         quantization=spec.quantization),
             device=device)
         if self.engine.graph_runner is not None:
             # eager hipGraph capture for all batch buckets (serving never
