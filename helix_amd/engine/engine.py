@@ -164,6 +164,10 @@ class LLMEngine:
             return
         if seq.status == SeqStatus.WAITING:
             self.waiting.remove(seq)
+            if seq.block_table:            # mid-chunked-prefill
+                self.kv.allocator.free(seq.block_table)
+                seq.block_table = []
+                self._release_row(seq)
         elif seq.status == SeqStatus.RUNNING:
             self.running.remove(seq)
             self.kv.allocator.free(seq.block_table)
@@ -200,44 +204,71 @@ class LLMEngine:
 
     # ------------------------------------------------------------------
     def _schedule_prefill(self) -> List[Sequence]:
+        """Admit waiting sequences for prefill. A prompt whose remaining
+        tokens exceed max_prefill_tokens is processed in CHUNKS across
+        steps: each chunk writes its KV and the next chunk re-enters as
+        a cached-prefix suffix (query-offset kernel + paged gather), so
+        one long prompt never monopolizes a step or blows activation
+        memory."""
         batch: List[Sequence] = []
         tokens = 0
         bs = self.cfg.block_size
         alloc = self.kv.allocator
+        self._chunk_target: Dict[str, int] = {}
         while self.waiting and len(self.running) + len(batch) < self.cfg.max_num_seqs:
             seq = self.waiting[0]
             L = len(seq.prompt_ids)
-            # prefix-cache match (chain hashes over full blocks)
+            first = not seq.block_table
             matched: List[int] = []
-            if self.cfg.enable_prefix_caching:
-                if not seq.block_hashes:
-                    h = b""
-                    for i in range(L // bs):
-                        h = chain_hash(h, seq.prompt_ids[i * bs:(i + 1) * bs])
-                        seq.block_hashes.append(h)
-                for h in seq.block_hashes:
-                    b = alloc.lookup_hash(h)
-                    if b is None:
-                        break
-                    matched.append(b)
-                # always leave at least one token to prefill
-                while matched and len(matched) * bs > L - 1:
-                    matched.pop()
-            cached = len(matched) * bs
-            need = L - cached
-            if batch and tokens + need > self.cfg.max_prefill_tokens:
+            if first:
+                # prefix-cache match (chain hashes over full blocks)
+                if self.cfg.enable_prefix_caching:
+                    if not seq.block_hashes:
+                        h = b""
+                        for i in range(L // bs):
+                            h = chain_hash(h,
+                                           seq.prompt_ids[i * bs:(i + 1) * bs])
+                            seq.block_hashes.append(h)
+                    for h in seq.block_hashes:
+                        b = alloc.lookup_hash(h)
+                        if b is None:
+                            break
+                        matched.append(b)
+                    # always leave at least one token to prefill
+                    while matched and len(matched) * bs > L - 1:
+                        matched.pop()
+                P = len(matched) * bs
+            else:
+                P = seq.cached_prefix          # continuation chunk
+            budget = self.cfg.max_prefill_tokens - tokens
+            if budget <= 0:
                 break
-            nblocks_total = (L + bs - 1) // bs
-            new_blocks = nblocks_total - len(matched)
+            T = min(L, P + budget)
+            if T <= P:
+                break
+            if batch and T < L:
+                break                           # partial chunks go alone-ish
+            nblocks = (T + bs - 1) // bs
+            new_blocks = nblocks - max(len(seq.block_table), len(matched))
             # +1 headroom block so the first decode step can't OOM-deadlock
-            if not alloc.can_allocate(new_blocks + 1):
+            head = 1 if T == L else 0
+            if not alloc.can_allocate(new_blocks + head):
                 break
-            seq.block_table = [alloc.share(b) for b in matched] + \
-                alloc.allocate(new_blocks)
+            if first:
+                seq.block_table = [alloc.share(b) for b in matched] + \
+                    alloc.allocate(new_blocks)
+            else:
+                seq.block_table.extend(alloc.allocate(new_blocks))
             self._assign_row(seq)
-            seq.cached_prefix = cached
-            batch.append(self.waiting.pop(0))
-            tokens += need
+            seq.cached_prefix = P
+            self._chunk_target[seq.seq_id] = T
+            if T == L:
+                batch.append(self.waiting.pop(0))
+            else:
+                batch.append(seq)               # stays queued for next chunk
+            tokens += T - P
+            if T < L:
+                break
         return batch
 
     def _slot(self, seq: Sequence, pos: int) -> int:
@@ -260,11 +291,12 @@ class LLMEngine:
         bs = self.cfg.block_size
         input_ids, positions, slots, cu_q, cu_k = [], [], [], [0], [0]
         gather_blk, gather_off = [], []
+        targets = getattr(self, "_chunk_target", {})
         any_cached = any(s.cached_prefix for s in batch)
         for seq in batch:
-            L = len(seq.prompt_ids)
+            L = targets.get(seq.seq_id, len(seq.prompt_ids))
             P = seq.cached_prefix
-            input_ids.extend(seq.prompt_ids[P:])
+            input_ids.extend(seq.prompt_ids[P:L])
             positions.extend(range(P, L))
             slots.extend(self._slot(seq, p) for p in range(P, L))
             cu_q.append(cu_q[-1] + L - P)
@@ -276,8 +308,8 @@ class LLMEngine:
         dev = self.device
         meta = PrefillMeta(
             cu_seqlens=torch.tensor(cu_q, dtype=torch.int32, device=dev),
-            max_seqlen=max(len(s.prompt_ids) - s.cached_prefix
-                           for s in batch),
+            max_seqlen=max(targets.get(s.seq_id, len(s.prompt_ids))
+                           - s.cached_prefix for s in batch),
             slot_mapping=torch.tensor(slots, dtype=torch.int64, device=dev),
             positions=torch.tensor(positions, dtype=torch.int64, device=dev),
             cu_seqlens_k=(torch.tensor(cu_k, dtype=torch.int32, device=dev)
@@ -288,11 +320,14 @@ class LLMEngine:
                                      device=dev) if any_cached else None))
         ids = torch.tensor(input_ids, dtype=torch.int64, device=dev)
         hidden = self.model(ids, self.kv.caches, meta)
-        # register this batch's new full blocks in the prefix cache
+        # register this batch's newly COMPLETED full blocks in the
+        # prefix cache (partial chunks register only what they wrote)
         if self.cfg.enable_prefix_caching:
             for seq in batch:
+                T = targets.get(seq.seq_id, len(seq.prompt_ids))
                 start = seq.cached_prefix // bs
-                for i in range(start, len(seq.block_hashes)):
+                end = min(len(seq.block_hashes), T // bs)
+                for i in range(start, end):
                     self.kv.allocator.register_hash(seq.block_table[i],
                                                     seq.block_hashes[i])
         last_rows = torch.tensor([c - 1 for c in cu_q[1:]], dtype=torch.int64,
@@ -300,16 +335,29 @@ class LLMEngine:
         logits = self.model.compute_logits(hidden[last_rows])
         tokens = self._sample(batch, logits)
         now = time.monotonic()
-        for seq in batch:
+        out = []
+        for seq, tok in zip(batch, tokens):
+            T = targets.get(seq.seq_id, len(seq.prompt_ids))
+            if T < len(seq.prompt_ids):
+                seq.cached_prefix = T      # partial chunk: KV written,
+                continue                   # stays WAITING for next chunk
             seq.status = SeqStatus.RUNNING
             seq.first_token_time = now
             self.running.append(seq)
-        return list(zip(batch, tokens))
+            out.append((seq, tok))
+        return out
 
     def _preempt_for_blocks(self):
         """Free KV by preempting the most recent running sequence back to
         waiting (recompute-on-readmit, vLLM-style). Returns the victim or
         None."""
+        for w in self.waiting:             # drop mid-chunk prefill first
+            if w.block_table:
+                self.kv.allocator.free(w.block_table)
+                w.block_table = []
+                w.cached_prefix = 0
+                self._release_row(w)
+                return w
         if len(self.running) <= 1:
             return None
         victim = self.running.pop()          # newest first

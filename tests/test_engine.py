@@ -265,3 +265,63 @@ def test_prefix_cache_refcounts():
     while eng.has_work:
         eng.step()
     assert eng.num_free_blocks() == total
+
+
+def test_chunked_prefill_long_prompt_equivalence():
+    """A prompt longer than max_prefill_tokens is prefilled in chunks
+    (KV-cache continuation via the query-offset path) and must produce
+    the exact greedy tokens of an unchunked engine."""
+    import torch
+    torch.manual_seed(0)
+    prompt = [((i * 37) % 500) + 1 for i in range(200)]
+    sp = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+
+    def run(budget):
+        eng = LLMEngine(EngineConfig(model="tiny", max_model_len=512,
+                                     max_num_seqs=4, kv_cache_blocks=128,
+                                     max_prefill_tokens=budget,
+                                     eos_token_id=-1, seed=3),
+                        device="cpu")
+        out = eng.generate([prompt], sp)
+        assert eng.num_free_blocks() == 128 or eng.running
+        return out[0], eng.steps
+
+    full, steps_full = run(8192)
+    chunked, steps_chunked = run(64)
+    assert chunked == full
+    assert steps_chunked > steps_full      # actually took multiple chunks
+
+
+def test_chunked_prefill_interleaves_with_decode():
+    """While a long prompt chunks through prefill, short running seqs
+    are unaffected and everything completes."""
+    eng = LLMEngine(EngineConfig(model="tiny", max_model_len=512,
+                                 max_num_seqs=4, kv_cache_blocks=128,
+                                 max_prefill_tokens=48, eos_token_id=-1,
+                                 seed=1),
+                    device="cpu")
+    sp = SamplingParams(temperature=0.0, max_tokens=10, ignore_eos=True)
+    eng.add_request("short", [1, 2, 3], sp)
+    eng.step()
+    long_prompt = [((i * 13) % 400) + 1 for i in range(150)]
+    eng.add_request("long", long_prompt, sp)
+    while eng.has_work:
+        eng.step()
+    assert len(eng.seqs["short"].output_ids) == 10
+    assert len(eng.seqs["long"].output_ids) == 10
+    assert eng.num_free_blocks() == 128
+
+
+def test_chunked_prefill_cancel_mid_chunk_frees_blocks():
+    eng = LLMEngine(EngineConfig(model="tiny", max_model_len=512,
+                                 max_num_seqs=4, kv_cache_blocks=128,
+                                 max_prefill_tokens=32, eos_token_id=-1),
+                    device="cpu")
+    sp = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
+    eng.add_request("x", [(i % 400) + 1 for i in range(100)], sp)
+    eng.step()                             # first chunk only
+    assert eng.seqs["x"].cached_prefix == 32
+    assert eng.num_free_blocks() < 128
+    eng.cancel("x")
+    assert eng.num_free_blocks() == 128
+    assert not eng.has_work
