@@ -99,3 +99,19 @@ def test_fp8_quantized_engine_gpu():
         assert len(g) == 12
         agree = sum(a == b for a, b in zip(g, w))
         assert agree >= 6, f"fp8 diverged early: {g} vs {w}"
+
+
+def test_chunked_prefill_gpu_equivalence():
+    """Within-prompt chunked prefill must match unchunked greedy output
+    on GPU (exercises query-offset kernel + gather continuation)."""
+    prompt = [((i * 37) % 900) + 1 for i in range(200)]
+    sp = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+
+    def run(budget):
+        cfg = EngineConfig(model="tiny-gqa", max_model_len=512,
+                           max_num_seqs=8, kv_cache_blocks=256,
+                           max_prefill_tokens=budget, eos_token_id=-1,
+                           seed=9, enforce_eager=True)
+        return LLMEngine(cfg, device="cuda:0").generate([prompt], sp)[0]
+
+    assert run(64) == run(8192)
