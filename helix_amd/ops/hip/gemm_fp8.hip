@@ -30,9 +30,8 @@ __global__ __launch_bounds__(256) void gemm_fp8_kernel(
     const uint8_t* __restrict__ w, const float* __restrict__ x_scale,
     const float* __restrict__ w_scale, const uint16_t* __restrict__ bias,
     int M, int N, int K, int act /*0=none,1=gelu_tanh*/) {
-  // double-buffered: load tile kt+1 while MFMAs consume tile kt
-  __shared__ uint8_t a_lds[2][BM * BK];
-  __shared__ uint8_t b_lds[2][BN * BK];
+  __shared__ uint8_t a_lds[BM * BK];
+  __shared__ uint8_t b_lds[BN * BK];
 
   const int tile_n = blockIdx.x;
   const int tile_m = blockIdx.y;
@@ -51,8 +50,9 @@ __global__ __launch_bounds__(256) void gemm_fp8_kernel(
 
   // 128x128 B per tile / 4 waves = 4 KiB/wave = 4 x 64 lanes x 16 B.
   const int nk = K / BK;
-  auto stage = [&](int kt, int buf) {
+  for (int kt = 0; kt < nk; ++kt) {
     const int k0 = kt * BK;
+    __syncthreads();
 #pragma unroll
     for (int it = 0; it < 4; ++it) {
       const int e = ((wid * 4 + it) * WAVE + lane) * 16;  // byte index
@@ -62,35 +62,31 @@ __global__ __launch_bounds__(256) void gemm_fp8_kernel(
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)(x +
               (int64_t)arow * K + k0 + col),
-          (__attribute__((address_space(3))) uint32_t*)(a_lds[buf] +
+          (__attribute__((address_space(3))) uint32_t*)(a_lds +
               (wid * 4 + it) * WAVE * 16),
           16, 0, 0);
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) uint32_t*)(w +
               (int64_t)brow * K + k0 + col),
-          (__attribute__((address_space(3))) uint32_t*)(b_lds[buf] +
+          (__attribute__((address_space(3))) uint32_t*)(b_lds +
               (wid * 4 + it) * WAVE * 16),
           16, 0, 0);
     }
-  };
-  stage(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-  for (int kt = 0; kt < nk; ++kt) {
-    const int buf = kt & 1;
-    if (kt + 1 < nk) stage(kt + 1, buf ^ 1);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+
     v8i a_frag[4], b_frag[4];
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const int row = wr * 64 + mi * 16 + lane_lo;
       a_frag[mi] = *reinterpret_cast<const v8i*>(
-          a_lds[buf] + row * BK + lane_hi * 32);
+          a_lds + row * BK + lane_hi * 32);
     }
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
       const int row = wc * 64 + ni * 16 + lane_lo;
       b_frag[ni] = *reinterpret_cast<const v8i*>(
-          b_lds[buf] + row * BK + lane_hi * 32);
+          b_lds + row * BK + lane_hi * 32);
     }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
@@ -98,10 +94,6 @@ __global__ __launch_bounds__(256) void gemm_fp8_kernel(
       for (int ni = 0; ni < 4; ++ni)
         acc[mi][ni] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
             a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0, UNITY, 0, UNITY);
-    if (kt + 1 < nk) {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      __syncthreads();
-    }
   }
 
   // Epilogue: per-row activation scale x per-column weight scale.
