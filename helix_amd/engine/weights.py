@@ -82,6 +82,12 @@ def _map_hf_name(name: str):
                 (f"layers.{i}.attn.qkv_proj.weight", "k"),
             "self_attn.v_proj.weight":
                 (f"layers.{i}.attn.qkv_proj.weight", "v"),
+            "self_attn.q_proj.bias":
+                (f"layers.{i}.attn.qkv_proj.bias", "q"),
+            "self_attn.k_proj.bias":
+                (f"layers.{i}.attn.qkv_proj.bias", "k"),
+            "self_attn.v_proj.bias":
+                (f"layers.{i}.attn.qkv_proj.bias", "v"),
             "self_attn.o_proj.weight":
                 (f"layers.{i}.attn.o_proj.weight", None),
             "mlp.gate_proj.weight":

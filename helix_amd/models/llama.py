@@ -37,6 +37,7 @@ class LlamaConfig:
     max_position: int = 8192
     tie_embeddings: bool = False
     sliding_window: int = 0        # 0 = full attention (Mistral v0.1: 4096)
+    attention_bias: bool = False   # Qwen2-style QKV bias
 
     @property
     def q_size(self) -> int:
@@ -64,6 +65,16 @@ PRESETS = {
         name="tiny-gqa", vocab_size=1024, hidden_size=512,
         intermediate_size=1024, num_layers=4, num_heads=8, num_kv_heads=2,
         head_dim=64, max_position=1024, rope_base=10000.0),
+    "qwen2-7b": LlamaConfig(
+        name="qwen2-7b", vocab_size=152064, hidden_size=3584,
+        intermediate_size=18944, num_layers=28, num_heads=28,
+        num_kv_heads=4, head_dim=128, rope_base=1000000.0,
+        max_position=32768, attention_bias=True),
+    "tiny-qwen": LlamaConfig(
+        name="tiny-qwen", vocab_size=512, hidden_size=256,
+        intermediate_size=512, num_layers=2, num_heads=4, num_kv_heads=2,
+        head_dim=64, max_position=512, rope_base=10000.0,
+        attention_bias=True),
     "tiny-sw": LlamaConfig(
         name="tiny-sw", vocab_size=512, hidden_size=256,
         intermediate_size=512, num_layers=2, num_heads=4, num_kv_heads=2,
@@ -112,7 +123,7 @@ class Attention(nn.Module):
         self.tp_size = tp_size
         self.window = cfg.sliding_window or 0
         q, kv, h = self.nh * self.hd, self.nkv * self.hd, cfg.hidden_size
-        self.qkv_proj = nn.Linear(h, q + 2 * kv, bias=False)
+        self.qkv_proj = nn.Linear(h, q + 2 * kv, bias=cfg.attention_bias)
         self.o_proj = nn.Linear(q, h, bias=False)
 
     def forward(self, x: torch.Tensor, cos_sin: torch.Tensor, kv_cache,
@@ -254,6 +265,9 @@ class LlamaForCausalLM(nn.Module):
         for name, p in self.named_parameters():
             if name.endswith("norm_w"):
                 p.data.fill_(1.0)
+            elif name.endswith(".bias"):
+                p.data.zero_() if on_gpu else p.data.copy_(
+                    torch.zeros(p.shape, dtype=torch.float32).to(p.dtype))
             elif on_gpu:
                 p.data.normal_(0.0, std)
             else:

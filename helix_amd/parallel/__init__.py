@@ -76,7 +76,12 @@ def shard_llama_state_dict(sd: dict, cfg, tp: int, rank: int) -> dict:
     ql, kvl = q // tp, kv // tp
     il = cfg.intermediate_size // tp
     for name, w in sd.items():
-        if name.endswith("qkv_proj.weight"):
+        if name.endswith("qkv_proj.bias"):
+            qs = w[rank * ql:(rank + 1) * ql]
+            ks = w[q + rank * kvl: q + (rank + 1) * kvl]
+            vs = w[q + kv + rank * kvl: q + kv + (rank + 1) * kvl]
+            out[name] = torch.cat([qs, ks, vs], dim=0).contiguous()
+        elif name.endswith("qkv_proj.weight"):
             qs = w[rank * ql:(rank + 1) * ql]
             ks = w[q + rank * kvl: q + (rank + 1) * kvl]
             vs = w[q + kv + rank * kvl: q + kv + (rank + 1) * kvl]

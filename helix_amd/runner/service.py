@@ -47,6 +47,8 @@ DEFAULT_SPECS = {
     "llama3-70b": ModelSpec("llama3-70b", "llm", "llama3-70b",
                             max_num_seqs=48, kv_cache_blocks=16384),
     "mistral-7b": ModelSpec("mistral-7b", "llm", "mistral-7b"),
+    "qwen2-7b": ModelSpec("qwen2-7b", "llm", "qwen2-7b",
+                          max_model_len=32768),
     "bge-base": ModelSpec("bge-base", "embedding", "bge-base"),
     "bge-large": ModelSpec("bge-large", "embedding", "bge-large"),
     # tiny models for CPU tests
