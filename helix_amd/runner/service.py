@@ -263,6 +263,15 @@ class RunnerService:
     def loaded_models(self) -> List[str]:
         return list(self.instances.keys())
 
+    def register_spec(self, spec: ModelSpec):
+        """Register/override a model manifest entry at runtime (the
+        reference's dynamic model config; pairs with local-models
+        load)."""
+        with self._lock:
+            if spec.name in self.instances:
+                raise ValueError(f"{spec.name} is loaded; unload first")
+            self.specs[spec.name] = spec
+
     def ensure_loaded(self, model: str):
         """Admission control: load `model`, LRU-evicting idle models if the
         estimate does not fit in free HBM. Raises NoCapacityError if it

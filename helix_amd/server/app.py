@@ -474,6 +474,31 @@ def create_app(cfg: Optional[ServerConfig] = None,
     # Local-model admin (reference local-models/load|unload,
     # server.go:1018-1020): proxied to the runner serving the model.
     # ------------------------------------------------------------------
+    @app.post("/api/v1/local-models")
+    async def register_local_model(request: Request,
+                                   user: AuthUser = Depends(admin_dep)):
+        """Register a model spec (preset + serving options: tp,
+        quantization, context length) on the local runner."""
+        if runner_service is None:
+            raise HTTPException(400, "no local runner on this server")
+        from helix_amd.runner.service import ModelSpec
+        b = await request.json()
+        try:
+            spec = ModelSpec(
+                name=b["name"], kind=b.get("kind", "llm"),
+                preset=b.get("preset", b["name"]),
+                max_model_len=int(b.get("max_model_len", 8192)),
+                max_num_seqs=int(b.get("max_num_seqs", 64)),
+                kv_cache_blocks=b.get("kv_cache_blocks"),
+                tp=int(b.get("tp", 1)),
+                quantization=b.get("quantization"))
+            runner_service.register_spec(spec)
+        except KeyError as e:
+            raise HTTPException(400, f"missing field {e}")
+        except ValueError as e:
+            raise HTTPException(409, str(e))
+        return {"ok": True, "model": spec.name}
+
     @app.post("/api/v1/local-models/{model}/load")
     async def load_local_model(model: str,
                                user: AuthUser = Depends(admin_dep)):

@@ -168,3 +168,33 @@ def test_stream_disconnect_cancels_sequence(service):
     assert inst.in_flight == 0
     seqs = inst.engine.seqs
     assert any(s.status.value == "cancelled" for s in seqs.values())
+
+
+def test_register_spec_and_load(tmp_path):
+    """Dynamic model registration: admin registers an fp8 tiny spec and
+    loads it through the admin API."""
+    from fastapi.testclient import TestClient
+    from helix_amd.runner.service import RunnerService
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import ServerConfig
+    from helix_amd.store import Store
+    cfg = ServerConfig()
+    cfg.filestore.path = str(tmp_path / "fs")
+    svc = RunnerService(device="cpu")
+    app = create_app(cfg, store=Store(":memory:"), runner_service=svc)
+    client = TestClient(app)
+    H = {"Authorization": "Bearer admin-key"}
+    r = client.post("/api/v1/local-models", json={
+        "name": "tiny-fp8", "preset": "tiny", "max_model_len": 256,
+        "kv_cache_blocks": 64, "quantization": "fp8"}, headers=H)
+    assert r.status_code == 200, r.text
+    r = client.post("/api/v1/local-models/tiny-fp8/load", headers=H)
+    assert r.status_code == 200
+    assert "tiny-fp8" in svc.loaded_models()
+    from helix_amd.models.quant import FP8Linear
+    inst = svc.instances["tiny-fp8"]
+    assert isinstance(inst.engine.model.layers[0].attn.qkv_proj, FP8Linear)
+    r = client.post("/api/v1/local-models", json={
+        "name": "tiny-fp8", "preset": "tiny"}, headers=H)
+    assert r.status_code == 409        # loaded: must unload first
+    svc.shutdown()
