@@ -1157,6 +1157,16 @@ def create_app(cfg: Optional[ServerConfig] = None,
         except KeyError:
             raise HTTPException(404, "task not found")
 
+    @app.post("/api/v1/spec-tasks/{tid}/implement")
+    async def implement_task(tid: str,
+                             user: AuthUser = Depends(auth_dep)):
+        try:
+            return await spec_tasks.implement(tid)
+        except KeyError:
+            raise HTTPException(404, "task not found")
+        except (ValueError, RuntimeError) as e:
+            raise HTTPException(400, str(e))
+
     @app.post("/api/v1/spec-tasks/{tid}/comments")
     async def comment_task(tid: str, request: Request,
                            user: AuthUser = Depends(auth_dep)):

@@ -123,3 +123,62 @@ class SpecTaskService:
         except Exception:
             log.exception("planning failed for %s", tid)
             return self.transition(tid, "failed")
+
+    # -- implementation agent (reference spec_task_orchestrator impl
+    # phase; runs in-process against the project's bare repo instead of
+    # a sandboxed dev container — no per-session dockerd on this stack) --
+    async def implement(self, tid: str, model: str = "") -> dict:
+        """Drive the approved spec to a commit on a task branch: the LLM
+        emits a JSON file-manifest which is committed to
+        `helix/task-{id}`; the task moves in_progress -> pr."""
+        import json as _json
+        doc = self.transition(tid, "in_progress")
+        if self.controller is None or self.git is None:
+            raise RuntimeError("controller/git unavailable")
+        project = self.store.get("projects", doc["project_id"])
+        rid = (project or {}).get("repo_id")
+        if not rid:
+            raise RuntimeError("project has no repo")
+        tree = []
+        try:
+            tree = self.git.ls_tree(rid)
+        except Exception:
+            pass
+        req = {
+            "model": model or self.controller.cfg.inference.default_model,
+            "messages": [
+                {"role": "system",
+                 "content": "You are an implementation agent. Respond "
+                            "with ONLY a JSON object: {\"message\": "
+                            "\"commit message\", \"files\": {\"path\": "
+                            "\"full file content\", ...}}."},
+                {"role": "user",
+                 "content": f"Spec:\n{doc['spec'] or doc['description']}"
+                            f"\n\nExisting repo files: {tree[:100]}"},
+            ],
+        }
+        try:
+            resp = await self.controller.chat_completion(
+                req, doc["owner"], ctx={"owner": doc["owner"],
+                                        "step": "spec_implement"})
+            text = resp["choices"][0]["message"]["content"]
+            # tolerate fenced or prefixed JSON
+            start = text.find("{")
+            end = text.rfind("}")
+            manifest = _json.loads(text[start:end + 1]) \
+                if start >= 0 else {}
+            files = manifest.get("files") or {}
+            if not isinstance(files, dict) or not files:
+                # the model produced no manifest: record the raw output
+                files = {f"tasks/{tid}/output.md": text}
+            msg = manifest.get("message") or f"task: {doc['title']}"
+            branch = f"helix/task-{tid[-8:]}"
+            self.git.commit_files(rid, files, msg, branch=branch)
+            doc = self.get_task(tid)
+            doc["branch"] = branch
+            self.store.put("spec_tasks", tid, doc, owner=doc["owner"],
+                           parent=doc["project_id"])
+            return self.transition(tid, "pr")
+        except Exception:
+            log.exception("implementation failed for %s", tid)
+            return self.transition(tid, "failed")

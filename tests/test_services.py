@@ -178,3 +178,42 @@ def test_mcp_gateway(stack):
     r = client.post(f"/api/v1/mcp/{app_id}", json={
         "jsonrpc": "2.0", "id": 4, "method": "bogus"}, headers=H(key))
     assert r.json()["error"]["code"] == -32601
+
+
+def test_spec_task_implementation_agent(tmp_path):
+    """implement(): the agent's JSON manifest becomes a commit on a
+    task branch and the task moves to `pr`."""
+    import json
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    cfg.filestore.path = str(tmp_path / "fs")
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    manifest = json.dumps({"message": "add login module",
+                           "files": {"src/login.py": "def login():\n"
+                                                     "    return True\n"}})
+    pm.register("mock", MockClient(
+        responses=["# Spec\nAdd a login module.", manifest]))
+    app = create_app(cfg, store=store, providers=pm)
+    client = TestClient(app)
+    key = client.post("/api/v1/users", json={"username": "bob"},
+                      headers={"Authorization": "Bearer admin-key"}
+                      ).json()["api_key"]
+    pid = client.post("/api/v1/projects", json={"name": "p"},
+                      headers=H(key)).json()["id"]
+    t = client.post(f"/api/v1/projects/{pid}/tasks",
+                    json={"title": "login"}, headers=H(key)).json()
+    client.post(f"/api/v1/spec-tasks/{t['id']}/plan", headers=H(key))
+    r = client.post(f"/api/v1/spec-tasks/{t['id']}/implement",
+                    headers=H(key))
+    doc = r.json()
+    assert doc["state"] == "pr", doc
+    assert doc["branch"].startswith("helix/task-")
+    git = app.state.git
+    rid = store.get("projects", pid)["repo_id"]
+    assert doc["branch"] in git.branches(rid)
+    assert git.read_file(rid, "src/login.py",
+                         doc["branch"]) == "def login():\n    return True\n"
+    log = git.log(rid, doc["branch"], 3)
+    assert log[0]["subject"].startswith("add login module")
