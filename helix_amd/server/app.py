@@ -80,7 +80,10 @@ def create_app(cfg: Optional[ServerConfig] = None,
     agent_runner = AgentRunner(cfg, store, providers, pubsub, rag=rag)
     knowledge = KnowledgeReconciler(cfg, store, rag)
     catalog = ModelCatalog(store)
-    usage = UsageService(store, catalog)
+    from helix_amd.server.metrics import Metrics
+    metrics = Metrics()
+    app.state.metrics = metrics
+    usage = UsageService(store, catalog, metrics=metrics)
     filestore = FileStore(cfg.filestore.path)
 
     from helix_amd.server.code_intel import CodeIntelService
@@ -162,6 +165,34 @@ def create_app(cfg: Optional[ServerConfig] = None,
             t = getattr(app.state, attr, None)
             if t:
                 t.cancel()
+
+    @app.middleware("http")
+    async def _prom_mw(request: Request, call_next):
+        import time as _t
+        t0 = _t.monotonic()
+        try:
+            resp = await call_next(request)
+            status = resp.status_code
+        except Exception:
+            status = 500
+            raise
+        finally:
+            route = request.scope.get("route")
+            rname = getattr(route, "path", request.url.path)
+            metrics.http_requests.labels(request.method, rname,
+                                         str(status)).inc()
+            metrics.http_latency.labels(rname).observe(_t.monotonic() - t0)
+        return resp
+
+    @app.get("/metrics")
+    async def prometheus_metrics():
+        from fastapi.responses import Response
+        metrics.runners_online.set(len(router.runners()))
+        metrics.models_loaded.set(
+            len(runner_service.loaded_models())
+            if runner_service is not None else 0)
+        body, ctype = metrics.render()
+        return Response(content=body, media_type=ctype)
 
     @app.exception_handler(ProviderError)
     async def _pe(request, exc: ProviderError):

@@ -14,13 +14,16 @@ class QuotaExceededError(Exception):
 
 
 class UsageService:
-    def __init__(self, store, catalog=None):
+    def __init__(self, store, catalog=None, metrics=None):
         self.store = store
         self.catalog = catalog
+        self.metrics = metrics
 
     # -- metering ----------------------------------------------------------
     def log_call(self, call: LLMCall):
         """Per-call metric row + daily rollup per (owner, provider, model)."""
+        if self.metrics is not None:
+            self.metrics.observe_call(call)
         day = time.strftime("%Y-%m-%d", time.gmtime())
         mid = new_id("use")
         cost = 0.0

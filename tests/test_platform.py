@@ -221,3 +221,19 @@ def test_reasoning_effort_passthrough(stack):
         "app_id": app_id,
         "messages": [{"role": "user", "content": "x"}]}, headers=H(key))
     assert mock.calls[-1]["reasoning_effort"] == "high"
+
+
+def test_prometheus_metrics(stack):
+    _, client, _, key, _ = stack
+    # generate one chat call so llm metrics have samples
+    r = client.post("/v1/chat/completions", json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "hi"}]}, headers=H(key))
+    assert r.status_code == 200
+    m = client.get("/metrics")
+    assert m.status_code == 200
+    text = m.text
+    assert "helix_http_requests_total" in text
+    assert "helix_llm_calls_total" in text
+    assert 'model="mock-model"' in text
+    assert "helix_runners_online" in text
