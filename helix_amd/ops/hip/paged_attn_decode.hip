@@ -31,7 +31,11 @@ constexpr int CHUNK = 128;
 constexpr int MAX_G = 8;
 constexpr int PART_QUANT = 128;
 
-template <int DHEAD, int G>
+// MFMA_A: phase A computes scores with f32_16x16x32_bf16 matrix ops
+// (16 tokens x G queries per instruction) instead of per-thread VALU
+// dots — motivated by the B=512 PMC: 37 VALU per vector load and
+// wait:busy 14:1 (profiles/r01_pmc_decode_b512.txt).
+template <int DHEAD, int G, bool MFMA_A>
 __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
@@ -82,6 +86,24 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_
   }
   __syncthreads();
 
+  // Q fragments for the MFMA path: A[row=query g][k=dim], rows >= G
+  // zero.  Lane holds A[lane&15][(lane>>4)*8+i] per 32-dim step.
+  constexpr int NKK = DHEAD / 32;
+  bf16x8 qfrag[MFMA_A ? NKK : 1];
+  if constexpr (MFMA_A) {
+    const int l_lo = threadIdx.x & 15;
+    const int l_hi = (threadIdx.x & (WAVE - 1)) >> 4;
+#pragma unroll
+    for (int kk = 0; kk < NKK; ++kk) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        const float qv = (l_lo < G)
+            ? q_lds[l_lo][kk * 32 + l_hi * 8 + i] : 0.f;
+        reinterpret_cast<uint16_t*>(&qfrag[kk])[i] = f32_to_bf16(qv);
+      }
+    }
+  }
+
   // Phase-C ownership: a thread owns 8 consecutive dims (one 16B load
   // per token) — the ablation probe showed b32 V loads were 64% of the
   // kernel (8x more VMEM instructions per byte than phase A's b128s).
@@ -103,8 +125,47 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_
   for (int base = p_start; base < p_end; base += CHUNK) {
     const int chunk_n = min(CHUNK, p_end - base);
 
-    // --- Phase A: scores (batched K-row loads, then unrolled math) -------
-    if ((int)threadIdx.x < chunk_n) {
+    // --- Phase A: scores -------------------------------------------------
+    if constexpr (MFMA_A) {
+      // 16-token MFMA tiles; each wave owns alternating groups of 4.
+      const int ntiles = (chunk_n + 15) / 16;
+      const int l = threadIdx.x & (WAVE - 1);
+      const int l_lo = l & 15, l_hi = l >> 4;
+      for (int t0 = wid * 4; t0 < ntiles; t0 += nwaves * 4) {
+        const int nt = min(4, ntiles - t0);
+        u16x8 bfr[4][NKK];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (u >= nt) continue;
+          const int tok = min(base + (t0 + u) * 16 + l_lo, p_end - 1);
+          const int64_t blk = btable[tok / block_size];
+          const uint16_t* krow =
+              k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
+                          tok % block_size)) * DHEAD;
+#pragma unroll
+          for (int kk = 0; kk < NKK; ++kk)
+            bfr[u][kk] = *reinterpret_cast<const u16x8*>(
+                krow + kk * 32 + l_hi * 8);
+        }
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+          if (u >= nt) continue;
+          floatx4 sacc = floatx4{0, 0, 0, 0};
+#pragma unroll
+          for (int kk = 0; kk < NKK; ++kk)
+            sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                qfrag[kk], *reinterpret_cast<bf16x8*>(&bfr[u][kk]), sacc,
+                0, 0, 0);
+          // C[row=l_hi*4+r][col=l_lo]: row = query g, col = token
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int g = l_hi * 4 + r;
+            const int ci = (t0 + u) * 16 + l_lo;
+            if (g < G && ci < chunk_n) s_lds[g][ci] = sacc[r];
+          }
+        }
+      }
+    } else if ((int)threadIdx.x < chunk_n) {
       const int tok = base + threadIdx.x;
       const int64_t blk = btable[tok / block_size];
       const uint16_t* krow =
@@ -294,6 +355,15 @@ __global__ __launch_bounds__(128) void paged_attn_reduce_kernel(
   }
 }
 
+static bool use_mfma_a() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("HELIX_DECODE_MFMA");
+    v = (e != nullptr && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+
 template <int DHEAD, int G>
 void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
                    const uint16_t* q, const uint16_t* kc, const uint16_t* vc,
@@ -301,10 +371,17 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
                    int Hkv, int block_size, int max_blocks, int eff_part,
                    int nparts, int max_parts, int window,
                    hipStream_t stream) {
-  hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G>),
-                     dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
-                     tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
-                     block_size, max_blocks, eff_part, max_parts, window);
+  if (use_mfma_a()) {
+    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, true>),
+                       dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
+                       tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
+                       block_size, max_blocks, eff_part, max_parts, window);
+  } else {
+    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, false>),
+                       dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
+                       tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
+                       block_size, max_blocks, eff_part, max_parts, window);
+  }
 }
 
 }  // namespace
