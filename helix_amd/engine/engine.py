@@ -68,6 +68,7 @@ class EngineConfig:
     enforce_eager: bool = False               # disable hipGraph capture
     enable_prefix_caching: bool = True
     quantization: Optional[str] = None        # None | "fp8" (e4m3 W8A8)
+    kv_cache_dtype: str = "bf16"              # "bf16" | "fp8" (e4m3 KV)
 
 
 class LLMEngine:
@@ -95,9 +96,11 @@ class LLMEngine:
         num_blocks = cfg.kv_cache_blocks
         if num_blocks is None:
             num_blocks = self._auto_kv_blocks(nkv)
+        kv_dtype = torch.uint8 if cfg.kv_cache_dtype == "fp8" \
+            else torch.bfloat16
         self.kv = KVCache(self.model_cfg.num_layers, nkv,
                           self.model_cfg.head_dim, cfg.block_size, num_blocks,
-                          self.device)
+                          self.device, dtype=kv_dtype)
         self.waiting: List[Sequence] = []
         self.running: List[Sequence] = []
         self.seqs: Dict[str, Sequence] = {}

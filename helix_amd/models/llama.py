@@ -146,6 +146,9 @@ class Attention(nn.Module):
                 k_cache, v_cache = kv_cache
                 k_all = k_cache[meta.gather_blk, :, meta.gather_off]
                 v_all = v_cache[meta.gather_blk, :, meta.gather_off]
+                if k_all.dtype == torch.uint8:      # fp8 KV cache
+                    k_all = ops.kv_fp8_dequant(k_all, q.dtype)
+                    v_all = ops.kv_fp8_dequant(v_all, q.dtype)
                 o = ops.attn_prefill(q, k_all, v_all, meta.cu_seqlens,
                                      meta.max_seqlen, self.scale,
                                      window=self.window,

@@ -76,6 +76,16 @@ def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
     return q, k
 
 
+def kv_fp8_quant(t: torch.Tensor) -> torch.Tensor:
+    """bf16 -> OCP e4m3 bytes (uint8 view), any shape."""
+    return t.to(torch.float8_e4m3fn).view(torch.uint8)
+
+
+def kv_fp8_dequant(t: torch.Tensor, dtype=torch.bfloat16) -> torch.Tensor:
+    """OCP e4m3 bytes (uint8) -> dtype."""
+    return t.view(torch.float8_e4m3fn).to(dtype)
+
+
 def rope_qkv_cache(positions: torch.Tensor, qkv: torch.Tensor,
                    cos_sin: torch.Tensor, num_q: int, num_kv: int,
                    head_dim: int, kv_cache=None, slot_mapping=None,
@@ -112,7 +122,11 @@ def rope_qkv_cache(positions: torch.Tensor, qkv: torch.Tensor,
     k = k.view(T, num_kv, head_dim)
     v = v.view(T, num_kv, head_dim)
     if kv_cache is not None:
-        reshape_and_cache(k, v, kv_cache[0], kv_cache[1], slot_mapping)
+        if kv_cache[0].dtype == torch.uint8:    # fp8 KV cache
+            reshape_and_cache(kv_fp8_quant(k), kv_fp8_quant(v),
+                              kv_cache[0], kv_cache[1], slot_mapping)
+        else:
+            reshape_and_cache(k, v, kv_cache[0], kv_cache[1], slot_mapping)
     return q, (k if want_kv else None), (v if want_kv else None)
 
 
@@ -195,6 +209,9 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                                     tmp_out[:q.shape[0]], tmp_ml[:q.shape[0]],
                                     max_len, window)
         return out
+    if k_cache.dtype == torch.uint8:            # fp8 KV cache
+        k_cache = kv_fp8_dequant(k_cache)
+        v_cache = kv_fp8_dequant(v_cache)
     return ref.paged_attn_decode(q, k_cache, v_cache, block_tables, seq_lens,
                                  scale, window)
 

@@ -32,6 +32,8 @@ typedef __attribute__((ext_vector_type(8))) uint16_t u16x8;   // 8 bf16 = 16B
 typedef __attribute__((ext_vector_type(2))) uint32_t u32x2;
 typedef __attribute__((ext_vector_type(4))) uint32_t u32x4;
 // MFMA fragment types for f32_16x16x32_bf16
+typedef __attribute__((ext_vector_type(8))) uint8_t u8x8;     // 8 fp8 = 8B
+typedef __attribute__((ext_vector_type(16))) uint8_t u8x16;   // 16 fp8 = 16B
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;    // A/B operand (4 VGPRs)
 typedef __attribute__((ext_vector_type(4))) float floatx4;    // C/D accumulator
 
@@ -131,6 +133,15 @@ __device__ __forceinline__ uint64_t splitmix64(uint64_t x) {
 // uniform in (0,1]
 __device__ __forceinline__ float u64_to_uniform(uint64_t r) {
   return (float)((r >> 40) + 1) * (1.0f / 16777216.0f);
+}
+
+// OCP e4m3 <-> f32 via the gfx950 cvt units (NOT the MI300X fnuz
+// encoding). Byte-select variant for scalar use; pack for stores.
+__device__ __forceinline__ float fp8_to_f32(uint8_t b) {
+  return __builtin_amdgcn_cvt_f32_fp8((int)b, 0);
+}
+__device__ __forceinline__ uint8_t f32_to_fp8(float f) {
+  return (uint8_t)(__builtin_amdgcn_cvt_pk_fp8_f32(f, f, 0, false) & 0xFF);
 }
 
 }  // namespace helix

@@ -35,7 +35,11 @@ constexpr int PART_QUANT = 128;
 // (16 tokens x G queries per instruction) instead of per-thread VALU
 // dots — motivated by the B=512 PMC: 37 VALU per vector load and
 // wait:busy 14:1 (profiles/r01_pmc_decode_b512.txt).
-template <int DHEAD, int G, bool MFMA_A>
+// KV8: the paged cache holds OCP e4m3 bytes (opt-in
+// kv_cache_dtype="fp8"): halves the KV traffic this kernel is
+// latency/BW-bound on; dequant is one v_cvt per element in the math
+// loops. MFMA_A and KV8 are mutually exclusive instantiations.
+template <int DHEAD, int G, bool MFMA_A, bool KV8>
 __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
@@ -168,25 +172,47 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_
     } else if ((int)threadIdx.x < chunk_n) {
       const int tok = base + threadIdx.x;
       const int64_t blk = btable[tok / block_size];
-      const uint16_t* krow =
-          k_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                      tok % block_size)) * DHEAD;
-      u16x8 kraw[DHEAD / 8];
-#pragma unroll
-      for (int j = 0; j < DHEAD / 8; ++j)
-        kraw[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+      const int64_t roff =
+          (((blk * Hkv + hkv) * (int64_t)block_size + tok % block_size)) *
+          DHEAD;
       float s[G];
 #pragma unroll
       for (int g = 0; g < G; ++g) s[g] = 0.f;
+      if constexpr (KV8) {
+        const uint8_t* krow = (const uint8_t*)k_cache + roff;
+        u8x16 kraw[DHEAD / 16];
 #pragma unroll
-      for (int j = 0; j < DHEAD / 8; ++j) {
-        float kv[8];
+        for (int j = 0; j < DHEAD / 16; ++j)
+          kraw[j] = *reinterpret_cast<const u8x16*>(krow + j * 16);
 #pragma unroll
-        for (int i = 0; i < 8; ++i) kv[i] = bf16_to_f32(kraw[j][i]);
+        for (int j = 0; j < DHEAD / 16; ++j) {
+          float kv[16];
 #pragma unroll
-        for (int g = 0; g < G; ++g) {
+          for (int i = 0; i < 16; ++i) kv[i] = fp8_to_f32(kraw[j][i]);
 #pragma unroll
-          for (int i = 0; i < 8; ++i) s[g] += q_lds[g][j * 8 + i] * kv[i];
+          for (int g = 0; g < G; ++g) {
+#pragma unroll
+            for (int i = 0; i < 16; ++i)
+              s[g] += q_lds[g][j * 16 + i] * kv[i];
+          }
+        }
+      } else {
+        const uint16_t* krow = k_cache + roff;
+        u16x8 kraw[DHEAD / 8];
+#pragma unroll
+        for (int j = 0; j < DHEAD / 8; ++j)
+          kraw[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+#pragma unroll
+        for (int j = 0; j < DHEAD / 8; ++j) {
+          float kv[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) kv[i] = bf16_to_f32(kraw[j][i]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+#pragma unroll
+            for (int i = 0; i < 8; ++i)
+              s[g] += q_lds[g][j * 8 + i] * kv[i];
+          }
         }
       }
 #pragma unroll
@@ -225,41 +251,47 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_
 #pragma unroll
         for (int i = 0; i < 8; ++i) acc[g][i] *= head_corr[g];
       const int npass = (chunk_n - cpar + C_PAR - 1) / C_PAR;  // my tokens
+      auto vrow_off = [&](int tok) {
+        const int64_t blk = btable[tok / block_size];
+        return (((blk * Hkv + hkv) * (int64_t)block_size +
+                 tok % block_size)) * DHEAD + d8;
+      };
+      auto load_v = [&](int64_t off, float v[8]) {
+        if constexpr (KV8) {
+          const u8x8 raw = *reinterpret_cast<const u8x8*>(
+              (const uint8_t*)v_cache + off);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) v[i] = fp8_to_f32(raw[i]);
+        } else {
+          const u16x8 raw = *reinterpret_cast<const u16x8*>(v_cache + off);
+#pragma unroll
+          for (int i = 0; i < 8; ++i) v[i] = bf16_to_f32(raw[i]);
+        }
+      };
       int ps = 0;
       for (; ps + 4 <= npass; ps += 4) {
-        u16x8 vv[4];
+        int64_t voff[4];
 #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-          const int tok = base + (ps + u) * C_PAR + cpar;
-          const int64_t blk = btable[tok / block_size];
-          vv[u] = *reinterpret_cast<const u16x8*>(
-              v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                          tok % block_size)) * DHEAD + d8);
-        }
+        for (int u = 0; u < 4; ++u)
+          voff[u] = vrow_off(base + (ps + u) * C_PAR + cpar);
+        float v4[4][8];
+#pragma unroll
+        for (int u = 0; u < 4; ++u) load_v(voff[u], v4[u]);
 #pragma unroll
         for (int u = 0; u < 4; ++u) {
           const int tok_i = (ps + u) * C_PAR + cpar;
-          float v[8];
-#pragma unroll
-          for (int i = 0; i < 8; ++i) v[i] = bf16_to_f32(vv[u][i]);
 #pragma unroll
           for (int g = 0; g < G; ++g) {
             const float pv = s_lds[g][tok_i];
 #pragma unroll
-            for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
+            for (int i = 0; i < 8; ++i) acc[g][i] += pv * v4[u][i];
           }
         }
       }
       for (; ps < npass; ++ps) {
         const int tok_i = ps * C_PAR + cpar;
-        const int tok = base + tok_i;
-        const int64_t blk = btable[tok / block_size];
-        const u16x8 vv = *reinterpret_cast<const u16x8*>(
-            v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                        tok % block_size)) * DHEAD + d8);
         float v[8];
-#pragma unroll
-        for (int i = 0; i < 8; ++i) v[i] = bf16_to_f32(vv[i]);
+        load_v(vrow_off(base + tok_i), v);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
           const float pv = s_lds[g][tok_i];
@@ -369,15 +401,20 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
                    const uint16_t* q, const uint16_t* kc, const uint16_t* vc,
                    const int* bt, const int* lens, float scale, int B,
                    int Hkv, int block_size, int max_blocks, int eff_part,
-                   int nparts, int max_parts, int window,
+                   int nparts, int max_parts, int window, bool kv8,
                    hipStream_t stream) {
-  if (use_mfma_a()) {
-    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, true>),
+  if (kv8) {
+    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, false, true>),
+                       dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
+                       tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
+                       block_size, max_blocks, eff_part, max_parts, window);
+  } else if (use_mfma_a()) {
+    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, true, false>),
                        dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
                        tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
                        block_size, max_blocks, eff_part, max_parts, window);
   } else {
-    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, false>),
+    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, false, false>),
                        dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
                        tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
                        block_size, max_blocks, eff_part, max_parts, window);
@@ -423,10 +460,11 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
   auto* bp = block_tables.data_ptr<int>();
   auto* lp = seq_lens.data_ptr<int>();
 
+  const bool kv8 = k_cache.scalar_type() == torch::kUInt8;
 #define DISPATCH(DH, GG)                                                     \
   launch_decode<DH, GG>(o, to, tm, qp, kp, vp, bp, lp, (float)scale, B,     \
                         Hkv, block_size, max_blocks, eff_part, nparts,      \
-                        max_parts, (int)window, stream)
+                        max_parts, (int)window, kv8, stream)
   if (D == 128) {
     if (G == 1) DISPATCH(128, 1);
     else if (G == 2) DISPATCH(128, 2);

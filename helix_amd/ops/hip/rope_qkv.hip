@@ -25,7 +25,7 @@ __global__ void rope_qkv_cache_kernel(
     const int64_t* __restrict__ slot_mapping,
     uint16_t* __restrict__ k_out, uint16_t* __restrict__ v_out,
     const float* __restrict__ cos_sin,
-    int Hq, int Hkv, int D, int block_size) {
+    int Hq, int Hkv, int D, int block_size, int kv8) {
   const int t = blockIdx.x;
   const int64_t pos = positions[t];
   const float* cs = cos_sin + pos * D;
@@ -63,10 +63,17 @@ __global__ void rope_qkv_cache_kernel(
         o[d + half] = y2;
       }
       if (slot >= 0) {
-        uint16_t* o = k_cache +
+        const int64_t ko =
             ((cblock * Hkv + hk) * (int64_t)block_size + coff) * D;
-        o[d] = y1;
-        o[d + half] = y2;
+        if (kv8) {           // OCP e4m3 cache (kv_cache_dtype="fp8")
+          uint8_t* o = (uint8_t*)k_cache + ko;
+          o[d] = f32_to_fp8(x1 * c - x2 * sn);
+          o[d + half] = f32_to_fp8(x2 * c + x1 * sn);
+        } else {
+          uint16_t* o = k_cache + ko;
+          o[d] = y1;
+          o[d + half] = y2;
+        }
       }
     }
   }
@@ -81,10 +88,19 @@ __global__ void rope_qkv_cache_kernel(
     if (v_out != nullptr)
       *reinterpret_cast<u16x8*>(v_out + (int64_t)t * Hkv * D + idx * 8) =
           val;
-    if (slot >= 0)
-      *reinterpret_cast<u16x8*>(
-          v_cache +
-          ((cblock * Hkv + h) * (int64_t)block_size + coff) * D + d) = val;
+    if (slot >= 0) {
+      const int64_t vo =
+          ((cblock * Hkv + h) * (int64_t)block_size + coff) * D + d;
+      if (kv8) {
+        u8x8 q8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          q8[e] = f32_to_fp8(bf16_to_f32(val[e]));
+        *reinterpret_cast<u8x8*>((uint8_t*)v_cache + vo) = q8;
+      } else {
+        *reinterpret_cast<u16x8*>(v_cache + vo) = val;
+      }
+    }
   }
 }
 
@@ -106,10 +122,12 @@ void rope_qkv_cache(torch::Tensor positions, torch::Tensor qkv,
   TORCH_CHECK(qkv.stride(-1) == 1, "qkv innermost must be contiguous");
   TORCH_CHECK(head_dim % 8 == 0);
   int block_size = 1;
+  int kv8 = 0;
   uint16_t* kc = nullptr;
   uint16_t* vc = nullptr;
   const int64_t* slots = nullptr;
   if (k_cache.has_value()) {
+    kv8 = (k_cache->scalar_type() == torch::kUInt8) ? 1 : 0;
     TORCH_CHECK(v_cache.has_value() && slot_mapping.has_value());
     TORCH_CHECK(slot_mapping->scalar_type() == torch::kInt64);
     block_size = k_cache->size(2);
@@ -126,5 +144,5 @@ void rope_qkv_cache(torch::Tensor positions, torch::Tensor qkv,
       k_out.has_value() ? (uint16_t*)k_out->data_ptr() : nullptr,
       v_out.has_value() ? (uint16_t*)v_out->data_ptr() : nullptr,
       cos_sin.data_ptr<float>(), (int)num_q_heads, (int)num_kv_heads,
-      (int)head_dim, block_size);
+      (int)head_dim, block_size, kv8);
 }

@@ -40,6 +40,7 @@ class ModelSpec:
     kv_memory_fraction: float = 0.30   # of total HBM, for auto-sizing
     tp: int = 1                        # tensor-parallel degree (xGMI group)
     quantization: Optional[str] = None  # None | "fp8"
+    kv_cache_dtype: str = "bf16"
 
 
 DEFAULT_SPECS = {
@@ -107,7 +108,8 @@ class LLMInstance:
             EngineConfig(model=spec.preset, max_model_len=spec.max_model_len,
                          max_num_seqs=spec.max_num_seqs,
                          kv_cache_blocks=kv_blocks,
-                         quantization=spec.quantization),
+                         quantization=spec.quantization,
+                         kv_cache_dtype=spec.kv_cache_dtype),
             device=device)
         if self.engine.graph_runner is not None:
             # eager hipGraph capture for all batch buckets (serving never

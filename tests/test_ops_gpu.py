@@ -381,3 +381,63 @@ def test_gemm_fp8_bias_gelu():
                        bias=bias.cuda(), act=1).cpu()
     torch.testing.assert_close(got.float(), want.float(), atol=0.08,
                                rtol=0.08)
+
+
+def test_fp8_kv_decode_gpu():
+    """Decode kernel over an fp8 (e4m3) cache matches the dequantized
+    reference, and the HW cvt encodings match torch float8_e4m3fn."""
+    torch.manual_seed(0)
+    B, hq, hkv, d, bs, L = 4, 8, 2, 128, 16, 77
+    nb = (L + bs - 1) // bs
+    k = torch.randn(L, hkv, d, dtype=torch.bfloat16, device="cuda")
+    v = torch.randn(L, hkv, d, dtype=torch.bfloat16, device="cuda")
+    q = torch.randn(B, hq, d, dtype=torch.bfloat16, device="cuda")
+    kc = torch.zeros(B * nb + 1, hkv, bs, d, dtype=torch.uint8,
+                     device="cuda")
+    vc = torch.zeros_like(kc)
+    bt = (torch.arange(B * nb, dtype=torch.int32, device="cuda")
+          .reshape(B, nb) + 1)
+    lens = torch.full((B,), L, dtype=torch.int32, device="cuda")
+    for b in range(B):
+        slots = (torch.arange(L, dtype=torch.int64, device="cuda")
+                 + (1 + b * nb) * bs)
+        ops.reshape_and_cache(ops.kv_fp8_quant(k), ops.kv_fp8_quant(v),
+                              kc, vc, slots)
+    scale = d ** -0.5
+    out = ops.paged_attn_decode(q, kc, vc, bt, lens, scale)
+    # fp32 torch reference on the dequantized cache (CPU)
+    want = ops.paged_attn_decode(q.cpu(), ops.kv_fp8_dequant(kc).cpu(),
+                                 ops.kv_fp8_dequant(vc).cpu(), bt.cpu(),
+                                 lens.cpu(), scale)
+    torch.testing.assert_close(out.cpu().float(), want.float(), atol=3e-2,
+                               rtol=3e-2)
+
+
+def test_fp8_kv_fused_rope_write_gpu():
+    """rope_qkv_cache writing an fp8 cache matches quantizing the bf16
+    path's cache (HW v_cvt_fp8 == torch e4m3fn encodings)."""
+    torch.manual_seed(1)
+    T, hq, hkv, d, bs = 21, 8, 2, 64, 16
+    qkv = torch.randn(T, (hq + 2 * hkv) * d, dtype=torch.bfloat16,
+                      device="cuda")
+    pos = torch.randint(0, 60, (T,), dtype=torch.int64, device="cuda")
+    cs = ref.make_cos_sin_cache(d, 128).cuda()
+    slots = torch.arange(bs, bs + T, dtype=torch.int64, device="cuda")
+    kc16 = torch.zeros(8, hkv, bs, d, dtype=torch.bfloat16, device="cuda")
+    vc16 = torch.zeros_like(kc16)
+    ops.rope_qkv_cache(pos, qkv, cs, hq, hkv, d, kv_cache=(kc16, vc16),
+                       slot_mapping=slots)
+    kc8 = torch.zeros(8, hkv, bs, d, dtype=torch.uint8, device="cuda")
+    vc8 = torch.zeros_like(kc8)
+    ops.rope_qkv_cache(pos, qkv, cs, hq, hkv, d, kv_cache=(kc8, vc8),
+                       slot_mapping=slots)
+    # quantization of the bf16-cache contents must equal the direct
+    # fp8 write (up to the float32 rounding point of rope outputs)
+    torch.testing.assert_close(ops.kv_fp8_dequant(vc8).float(),
+                               ops.kv_fp8_dequant(
+                                   ops.kv_fp8_quant(vc16)).float(),
+                               atol=0, rtol=0)
+    torch.testing.assert_close(ops.kv_fp8_dequant(kc8).float(),
+                               ops.kv_fp8_dequant(
+                                   ops.kv_fp8_quant(kc16)).float(),
+                               atol=0.06, rtol=0.06)
