@@ -11,27 +11,32 @@ using namespace helix;
 
 namespace {
 
+// EB = element bytes (2 = bf16 cache, 1 = fp8/e4m3 cache: inputs are
+// already-quantized byte views with identical [T, Hkv, D] geometry).
+template <int EB>
 __global__ void reshape_and_cache_kernel(
-    const uint16_t* __restrict__ k, const uint16_t* __restrict__ v,
-    uint16_t* __restrict__ k_cache, uint16_t* __restrict__ v_cache,
+    const uint8_t* __restrict__ k, const uint8_t* __restrict__ v,
+    uint8_t* __restrict__ k_cache, uint8_t* __restrict__ v_cache,
     const int64_t* __restrict__ slot_mapping, int Hkv, int D, int block_size) {
   const int t = blockIdx.x;
   const int64_t slot = slot_mapping[t];
   if (slot < 0) return;
   const int64_t block = slot / block_size;
   const int off = slot % block_size;
-  const int nvec = Hkv * D / 8;
+  const int row_bytes = Hkv * D * EB;
+  const int nvec = row_bytes / 16;
 
   for (int idx = threadIdx.x; idx < nvec; idx += blockDim.x) {
-    const int h = (idx * 8) / D;
-    const int d = (idx * 8) % D;
-    const int64_t src = (int64_t)t * Hkv * D + idx * 8;
+    const int b16 = idx * 16;               // byte offset within the row
+    const int h = b16 / (D * EB);
+    const int d = b16 % (D * EB);
+    const int64_t src = (int64_t)t * row_bytes + b16;
     const int64_t dst =
-        ((block * Hkv + h) * (int64_t)block_size + off) * D + d;
-    *reinterpret_cast<u16x8*>(k_cache + dst) =
-        *reinterpret_cast<const u16x8*>(k + src);
-    *reinterpret_cast<u16x8*>(v_cache + dst) =
-        *reinterpret_cast<const u16x8*>(v + src);
+        (((block * Hkv + h) * (int64_t)block_size + off) * D) * EB + d;
+    *reinterpret_cast<u8x16*>(k_cache + dst) =
+        *reinterpret_cast<const u8x16*>(k + src);
+    *reinterpret_cast<u8x16*>(v_cache + dst) =
+        *reinterpret_cast<const u8x16*>(v + src);
   }
 }
 
@@ -44,14 +49,27 @@ void reshape_and_cache(torch::Tensor k, torch::Tensor v,
   const int Hkv = k_cache.size(1);
   const int block_size = k_cache.size(2);
   const int D = k_cache.size(3);
-  TORCH_CHECK(D % 8 == 0);
+  TORCH_CHECK(D % 16 == 0);
   TORCH_CHECK(slot_mapping.scalar_type() == torch::kInt64);
+  const bool kv8 = k_cache.scalar_type() == torch::kUInt8;
+  TORCH_CHECK(!kv8 || k.scalar_type() == torch::kUInt8,
+              "fp8 cache takes pre-quantized uint8 K/V views");
   auto stream = at::hip::getCurrentHIPStream();
-  const int threads = std::min(256, Hkv * D / 8);
-  hipLaunchKernelGGL(reshape_and_cache_kernel, dim3(T), dim3(threads), 0,
-                     stream, (const uint16_t*)k.data_ptr(),
-                     (const uint16_t*)v.data_ptr(),
-                     (uint16_t*)k_cache.data_ptr(),
-                     (uint16_t*)v_cache.data_ptr(),
-                     slot_mapping.data_ptr<int64_t>(), Hkv, D, block_size);
+  const int eb = kv8 ? 1 : 2;
+  const int threads = std::min(256, Hkv * D * eb / 16);
+  if (kv8) {
+    hipLaunchKernelGGL(reshape_and_cache_kernel<1>, dim3(T), dim3(threads),
+                       0, stream, (const uint8_t*)k.data_ptr(),
+                       (const uint8_t*)v.data_ptr(),
+                       (uint8_t*)k_cache.data_ptr(),
+                       (uint8_t*)v_cache.data_ptr(),
+                       slot_mapping.data_ptr<int64_t>(), Hkv, D, block_size);
+  } else {
+    hipLaunchKernelGGL(reshape_and_cache_kernel<2>, dim3(T), dim3(threads),
+                       0, stream, (const uint8_t*)k.data_ptr(),
+                       (const uint8_t*)v.data_ptr(),
+                       (uint8_t*)k_cache.data_ptr(),
+                       (uint8_t*)v_cache.data_ptr(),
+                       slot_mapping.data_ptr<int64_t>(), Hkv, D, block_size);
+  }
 }
