@@ -522,7 +522,8 @@ def create_app(cfg: Optional[ServerConfig] = None,
                 max_num_seqs=int(b.get("max_num_seqs", 64)),
                 kv_cache_blocks=b.get("kv_cache_blocks"),
                 tp=int(b.get("tp", 1)),
-                quantization=b.get("quantization"))
+                quantization=b.get("quantization"),
+                kv_cache_dtype=b.get("kv_cache_dtype", "bf16"))
             runner_service.register_spec(spec)
         except KeyError as e:
             raise HTTPException(400, f"missing field {e}")
