@@ -237,3 +237,12 @@ def test_prometheus_metrics(stack):
     assert "helix_llm_calls_total" in text
     assert 'model="mock-model"' in text
     assert "helix_runners_online" in text
+
+
+def test_openapi_spec_served(stack):
+    _, client, _, _, _ = stack
+    r = client.get("/openapi.json")
+    assert r.status_code == 200
+    paths = r.json()["paths"]
+    assert "/v1/chat/completions" in paths
+    assert len(paths) > 80
