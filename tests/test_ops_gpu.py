@@ -446,6 +446,8 @@ def test_fp8_kv_fused_rope_write_gpu():
     kb = ops.kv_fp8_dequant(ops.kv_fp8_quant(kc16)).float()
     exact = (ka == kb).float().mean().item()
     assert exact > 0.99, f"only {exact:.3f} exact"
-    rel = ((ka - kb).abs() /
-           kb.abs().clamp(min=1e-3)).max().item()
-    assert rel <= 0.26, f"K off by more than one e4m3 ulp: {rel}"
+    # one e4m3 ulp is ~2^-3 of the magnitude (plus a small absolute
+    # floor for the denormal range)
+    bound = 0.135 * torch.maximum(ka.abs(), kb.abs()) + 0.02
+    bad = ((ka - kb).abs() > bound).sum().item()
+    assert bad == 0, f"{bad} K elements off by more than one e4m3 ulp"
