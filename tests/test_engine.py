@@ -374,3 +374,30 @@ def test_qwen2_hf_bias_weight_mapping(tmp_path):
     for i in range(cfg.num_layers):
         assert torch.equal(dst.layers[i].attn.qkv_proj.bias.data,
                            src.layers[i].attn.qkv_proj.bias.data)
+
+
+def test_row_pool_stress_chunked_preempt():
+    """Stress the persistent-row pool: many sequences, tiny KV, chunked
+    prefill and preemption interleaved — no row exhaustion, all
+    sequences finish, all blocks and rows return."""
+    import torch
+    torch.manual_seed(1)
+    eng = LLMEngine(EngineConfig(model="tiny", max_model_len=128,
+                                 max_num_seqs=6, kv_cache_blocks=48,
+                                 max_prefill_tokens=40, eos_token_id=-1,
+                                 seed=4),
+                    device="cpu")
+    sp = SamplingParams(temperature=0.0, max_tokens=24, ignore_eos=True)
+    for i in range(12):
+        L = 9 + (i * 17) % 70
+        eng.add_request(f"s{i}", [(j * 7 + i) % 500 + 1 for j in range(L)],
+                        sp)
+    steps = 0
+    while eng.has_work:
+        eng.step()
+        steps += 1
+        assert steps < 3000, "engine wedged"
+    for i in range(12):
+        assert len(eng.seqs[f"s{i}"].output_ids) == 24, i
+    assert eng.num_free_blocks() == 48
+    assert len(eng._free_rows) == 6
