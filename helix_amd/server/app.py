@@ -137,6 +137,42 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.code_intel = code_intel
     app.state.notifications = notifications
 
+    async def reap_wedged() -> int:
+        """Stuck-interaction auto-wake (reference
+        auto_wake_stuck_interactions.go): a turn whose partial-persist
+        pulse went stale while "waiting" is wedged — flip it to error
+        so clients and retries unblock. Returns #flipped."""
+        import time as _t
+        cutoff = int((_t.time() - cfg.web.wedge_timeout_s) * 1000)
+        n = 0
+        for it in store.list("interactions", limit=10000):
+            if it.get("state") in ("waiting", "editing") and \
+                    it.get("updated", 0) < cutoff:
+                it["state"] = "error"
+                it["error"] = "wedged: no generation progress; auto-errored"
+                store.put("interactions", it["id"], it,
+                          owner=it.get("owner", ""),
+                          parent=it.get("session_id", ""))
+                await pubsub.publish(
+                    ps.session_queue(it.get("owner", ""),
+                                     it.get("session_id", "")),
+                    {"type": "error", "interaction_id": it["id"],
+                     "error": it["error"]})
+                n += 1
+        return n
+
+    app.state.reap_wedged = reap_wedged
+
+    async def _reaper_loop():
+        # offline-runner reaper (reference 5 m threshold) + wedge sweep
+        while True:
+            router.reap_offline()
+            try:
+                await reap_wedged()
+            except Exception:
+                log.exception("wedge reaper failed")
+            await asyncio.sleep(30)
+
     @app.on_event("startup")
     async def _start_reconciler():
         # Startup recovery (reference serve.go:270-280
@@ -152,12 +188,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
         app.state._reconciler_task = asyncio.create_task(knowledge.run())
         app.state._trigger_task = asyncio.create_task(triggers.run())
 
-        async def _reaper():
-            # offline-runner reaper (reference 5 m threshold)
-            while True:
-                router.reap_offline()
-                await asyncio.sleep(30)
-        app.state._reaper_task = asyncio.create_task(_reaper())
+        app.state._reaper_task = asyncio.create_task(_reaper_loop())
 
     @app.on_event("shutdown")
     async def _stop_reconciler():

@@ -90,6 +90,10 @@ class RAGConfig:
 
 @dataclass
 class WebServerConfig:
+    # a streaming turn with no partial-persist pulse for this long is
+    # wedged (reference auto_wake_stuck_interactions)
+    wedge_timeout_s: float = float(os.environ.get(
+        "HELIX_WEDGE_TIMEOUT", "600"))
     host: str = field(default_factory=lambda: _env("SERVER_HOST", "0.0.0.0"))
     port: int = field(default_factory=lambda: _env_int("SERVER_PORT", 8080))
     admin_api_key: str = field(

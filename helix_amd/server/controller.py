@@ -269,6 +269,7 @@ class Controller:
                 now = time.monotonic()
                 if delta and now - last_persist >= 0.2:
                     interaction.response_message = text
+                    interaction.updated = now_ms()   # wedge-detector pulse
                     self.store.put("interactions", interaction.id,
                                    interaction.model_dump(),
                                    owner=session.owner, parent=session.id)

@@ -393,3 +393,25 @@ def test_rapid_multi_turn(stack):
     assert all(i["state"] == "complete" for i in its)
     assert [i["prompt_message"] for i in its] == \
         ["turn 0", "turn 1", "turn 2"]
+
+
+def test_wedged_interaction_auto_errors(stack):
+    """Reference auto_wake_stuck_interactions: a waiting interaction
+    with a stale partial-persist pulse flips to error; live ones are
+    untouched."""
+    import asyncio
+    app, store = stack[0], stack[-1]
+    from helix_amd.server.types import Interaction, InteractionState
+    stale = Interaction(session_id="s1", prompt_message="hi",
+                        state=InteractionState.WAITING, updated=1000)
+    store.put("interactions", stale.id, stale.model_dump(), owner="u",
+              parent="s1")
+    fresh = Interaction(session_id="s1", prompt_message="hi2",
+                        state=InteractionState.WAITING)
+    store.put("interactions", fresh.id, fresh.model_dump(), owner="u",
+              parent="s1")
+    n = asyncio.get_event_loop().run_until_complete(
+        app.state.reap_wedged())
+    assert n == 1
+    assert store.get("interactions", stale.id)["state"] == "error"
+    assert store.get("interactions", fresh.id)["state"] == "waiting"
