@@ -410,8 +410,7 @@ def test_wedged_interaction_auto_errors(stack):
                         state=InteractionState.WAITING)
     store.put("interactions", fresh.id, fresh.model_dump(), owner="u",
               parent="s1")
-    n = asyncio.get_event_loop().run_until_complete(
-        app.state.reap_wedged())
+    n = asyncio.run(app.state.reap_wedged())
     assert n == 1
     assert store.get("interactions", stale.id)["state"] == "error"
     assert store.get("interactions", fresh.id)["state"] == "waiting"
