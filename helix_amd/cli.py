@@ -352,6 +352,41 @@ def test(file: str = typer.Option("helix.yaml", "-f", "--file"),
     raise typer.Exit(1 if failures else 0)
 
 
+@app.command()
+def status(url: str = typer.Option("http://localhost:8080"),
+           key: str = typer.Option("admin-key", envvar="HELIX_API_KEY")):
+    """Cluster overview: runners, loaded models, store counts."""
+    import httpx
+    H = {"Authorization": f"Bearer {key}"}
+    try:
+        runners = httpx.get(f"{url}/api/v1/admin/runners", headers=H,
+                            timeout=10).json()
+    except Exception as e:
+        typer.echo(f"control plane unreachable at {url}: {e}")
+        raise typer.Exit(1)
+    typer.echo(f"runners: {len(runners)}")
+    for r in runners:
+        models = ", ".join(m.get("model_id", str(m))
+                           for m in r.get("models", []))
+        typer.echo(f"  {r.get('id')}: {r.get('status', 'ready')} "
+                   f"[{models}]")
+    try:
+        stats = httpx.get(f"{url}/debug/stats", headers=H,
+                          timeout=10).json()
+        typer.echo(f"store: {stats.get('store_counts')}")
+        typer.echo(f"rss: {stats.get('rss_bytes', 0) >> 20} MiB, "
+                   f"threads: {stats.get('num_threads')}")
+    except Exception:
+        pass
+    try:
+        models = httpx.get(f"{url}/v1/models", headers=H,
+                           timeout=10).json()
+        typer.echo("models: " + ", ".join(
+            m["id"] for m in models.get("data", [])[:20]))
+    except Exception:
+        pass
+
+
 @app.command("export-gguf")
 def export_gguf(preset: str = typer.Option("llama3-8b"),
                 ckpt: str = typer.Option("", help="safetensors dir "

@@ -1334,6 +1334,56 @@ def create_app(cfg: Optional[ServerConfig] = None,
                               "apps", "knowledge")},
         }
 
+    @app.get("/debug/threads")
+    async def debug_threads(user: AuthUser = Depends(admin_dep)):
+        # python stack dump (pprof goroutine-profile equivalent)
+        import sys
+        import traceback
+        import threading as _t
+        names = {t.ident: t.name for t in _t.enumerate()}
+        out = {}
+        for tid, frame in sys._current_frames().items():
+            out[f"{names.get(tid, '?')}-{tid}"] = \
+                traceback.format_stack(frame)[-6:]
+        return out
+
+    @app.post("/api/v1/admin/janitor")
+    async def run_janitor(request: Request,
+                          user: AuthUser = Depends(admin_dep)):
+        """Retention sweep (reference janitor/cleanup manager): prune
+        llm_calls/usage_metrics/step_info older than retention_days and
+        sessions idle past session_retention_days (0 = keep)."""
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        import time as _t
+        days = float(body.get("retention_days", 30))
+        sdays = float(body.get("session_retention_days", 0))
+        cutoff_ms = int((_t.time() - days * 86400) * 1000)
+        cutoff_s = _t.time() - days * 86400
+        pruned = {"llm_calls": 0, "usage_metrics": 0, "step_info": 0,
+                  "sessions": 0, "interactions": 0}
+        for table, ts_field, cut in (("llm_calls", "created", cutoff_ms),
+                                     ("usage_metrics", "ts", cutoff_s),
+                                     ("step_info", "created", cutoff_ms)):
+            for doc in store.list(table, limit=100000, desc=False):
+                v = doc.get(ts_field, 0)
+                if v and v < cut:
+                    store.delete(table, doc["id"])
+                    pruned[table] += 1
+        if sdays > 0:
+            scut = (_t.time() - sdays * 86400) * 1000
+            for doc in store.list("sessions", limit=100000, desc=False):
+                if doc.get("updated", doc.get("created", 0)) < scut:
+                    for it in store.list("interactions",
+                                         parent=doc["id"]):
+                        store.delete("interactions", it["id"])
+                        pruned["interactions"] += 1
+                    store.delete("sessions", doc["id"])
+                    pruned["sessions"] += 1
+        return pruned
+
     @app.get("/")
     async def index():
         from fastapi.responses import HTMLResponse

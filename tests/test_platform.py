@@ -246,3 +246,33 @@ def test_openapi_spec_served(stack):
     paths = r.json()["paths"]
     assert "/v1/chat/completions" in paths
     assert len(paths) > 80
+
+
+def test_janitor_prunes_old_rows(stack):
+    app, client, _, key, store = stack
+    import time as _t
+    H_admin = {"Authorization": "Bearer admin-key"}
+    old_ms = int((_t.time() - 90 * 86400) * 1000)
+    store.put("llm_calls", "old1", {"id": "old1", "created": old_ms})
+    store.put("llm_calls", "new1", {"id": "new1",
+                                    "created": int(_t.time() * 1000)})
+    store.put("usage_metrics", "um1", {"id": "um1",
+                                       "ts": _t.time() - 90 * 86400})
+    r = client.post("/api/v1/admin/janitor",
+                    json={"retention_days": 30}, headers=H_admin)
+    assert r.status_code == 200
+    pruned = r.json()
+    assert pruned["llm_calls"] == 1 and pruned["usage_metrics"] == 1
+    assert store.get("llm_calls", "old1") is None
+    assert store.get("llm_calls", "new1") is not None
+    # non-admin denied
+    r = client.post("/api/v1/admin/janitor", json={}, headers=H(key))
+    assert r.status_code in (401, 403)
+
+
+def test_debug_threads(stack):
+    _, client, _, _, _ = stack
+    r = client.get("/debug/threads",
+                   headers={"Authorization": "Bearer admin-key"})
+    assert r.status_code == 200
+    assert any("MainThread" in k for k in r.json())
