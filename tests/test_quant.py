@@ -69,3 +69,19 @@ def test_estimate_halves_fp8_weights():
     assert fp8 < bf16
     # projection weights are ~13.4 GiB of the 8B model; halving saves ~6.7
     assert (bf16 - fp8) > 5 << 30
+
+
+def test_eval_quant_harness_runs(tmp_path):
+    """The accuracy harness produces the expected report shape."""
+    import json
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, "scripts/eval_quant.py", "--prompts", "3",
+         "--steps", "6", "--model", "tiny-gqa"],
+        capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stderr
+    rep = json.loads(r.stdout)
+    assert set(rep["variants"]) == {"bf16", "fp8_weights", "fp8_kv",
+                                    "fp8_both"}
+    assert rep["variants"]["fp8_weights"]["mean_kl"] < 0.05
