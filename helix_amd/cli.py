@@ -387,6 +387,50 @@ def status(url: str = typer.Option("http://localhost:8080"),
         pass
 
 
+@app.command()
+def doctor():
+    """Environment check for an MI355X node: ROCm, GPU arch, extension,
+    RCCL prerequisites."""
+    import shutil
+    ok = True
+
+    def check(name, cond, hint=""):
+        nonlocal ok
+        mark = "ok " if cond else "FAIL"
+        typer.echo(f"[{mark}] {name}" + (f" — {hint}" if (hint and not
+                                                          cond) else ""))
+        ok = ok and bool(cond)
+
+    import torch
+    check("python/torch", True)
+    typer.echo(f"       torch {torch.__version__}")
+    check("hipcc on PATH", shutil.which("hipcc") is not None,
+          "install ROCm / add /opt/rocm/bin to PATH")
+    import helix_amd.ops as ops
+    check("helix_amd._C extension built", ops.have_native(),
+          "PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace")
+    has_gpu = torch.cuda.is_available()
+    # informational: a control-plane-only node is valid without a GPU
+    typer.echo(f"[{'ok ' if has_gpu else 'n/a'}] GPU visible" +
+               ("" if has_gpu else " — control-plane-only mode"))
+    if has_gpu:
+        props = torch.cuda.get_device_properties(0)
+        arch = getattr(props, "gcnArchName", "?")
+        typer.echo(f"       {props.name} ({arch}), "
+                   f"{props.total_memory >> 30} GiB")
+        check("gfx950 (MI355X)", "gfx950" in str(arch),
+              "kernels are compiled for gfx950 only")
+        import os as _os
+        check("HSA_ENABLE_IPC_MODE_LEGACY=0",
+              _os.environ.get("HSA_ENABLE_IPC_MODE_LEGACY") == "0",
+              "export HSA_ENABLE_IPC_MODE_LEGACY=0 for multi-process "
+              "RCCL (dmabuf IPC)")
+        import torch.distributed as dist
+        check("torch.distributed nccl (RCCL)",
+              dist.is_nccl_available())
+    raise typer.Exit(0 if ok else 1)
+
+
 @app.command("export-gguf")
 def export_gguf(preset: str = typer.Option("llama3-8b"),
                 ckpt: str = typer.Option("", help="safetensors dir "

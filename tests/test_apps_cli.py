@@ -76,3 +76,12 @@ def test_cli_version():
     res = CliRunner().invoke(app, ["version"])
     assert res.exit_code == 0
     assert "helix_amd" in res.output
+
+
+def test_cli_doctor_runs():
+    import subprocess
+    import sys
+    r = subprocess.run([sys.executable, "-m", "helix_amd.cli", "doctor"],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "extension built" in r.stdout
