@@ -110,7 +110,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
     evaluations = EvaluationService(store, controller, pubsub)
     app.state.evaluations = evaluations
     from helix_amd.server.mcp_gateway import MCPGateway
-    mcp = MCPGateway(store, agent_runner, code_intel)
+    mcp = MCPGateway(store, agent_runner, code_intel, org_rt=org_rt)
     app.state.mcp = mcp
 
     app.state.cfg = cfg

@@ -13,11 +13,70 @@ log = logging.getLogger("helix_amd.mcp")
 PROTOCOL_VERSION = "2024-11-05"
 
 
+class _OrgTool:
+    """Lightweight skill adapter exposing an org-runtime operation as an
+    MCP tool (reference: "MCP surface for org primitives")."""
+
+    def __init__(self, name, description, parameters, fn):
+        self.name = name
+        self.description = description
+        self.parameters = parameters
+        self._fn = fn
+
+    async def execute(self, args, ctx):
+        return await self._fn(args, ctx)
+
+
 class MCPGateway:
-    def __init__(self, store, agent_runner, code_intel=None):
+    def __init__(self, store, agent_runner, code_intel=None, org_rt=None):
         self.store = store
         self.agent_runner = agent_runner
         self.code_intel = code_intel
+        self.org_rt = org_rt
+
+    def _org_tools(self, owner: str):
+        if self.org_rt is None:
+            return []
+        rt = self.org_rt
+
+        async def list_streams(args, ctx):
+            out = []
+            for m in self.store.list("memberships", limit=10000):
+                if m.get("user_id") != owner:
+                    continue
+                for st in rt.list_streams(m.get("org_id", "")):
+                    out.append({"id": st["id"], "name": st["name"],
+                                "org_id": st["org_id"]})
+            return json.dumps(out)
+
+        async def read_stream(args, ctx):
+            msgs = rt.stream_messages(args.get("stream_id", ""), limit=50)
+            return json.dumps([{ "sender": m["sender"], "text": m["text"],
+                                 "replies": m.get("replies", [])}
+                               for m in msgs])
+
+        async def post_message(args, ctx):
+            msg = await rt.post_message(args.get("stream_id", ""), owner,
+                                        args.get("text", ""))
+            return json.dumps({"id": msg["id"],
+                               "replies": msg.get("replies", [])})
+
+        sid = {"type": "object", "properties": {
+            "stream_id": {"type": "string"}}, "required": ["stream_id"]}
+        post = {"type": "object", "properties": {
+            "stream_id": {"type": "string"}, "text": {"type": "string"}},
+            "required": ["stream_id", "text"]}
+        return [
+            _OrgTool("org_list_streams",
+                     "List org streams you are a member of",
+                     {"type": "object", "properties": {}}, list_streams),
+            _OrgTool("org_read_stream",
+                     "Read recent messages (and bot replies) of a stream",
+                     sid, read_stream),
+            _OrgTool("org_post_message",
+                     "Post a message to an org stream; subscribed bots "
+                     "reply", post, post_message),
+        ]
 
     async def _tools_for_app(self, app_id: str, owner: str):
         from helix_amd.server.types import App
@@ -27,9 +86,11 @@ class MCPGateway:
         app = App.model_validate(doc)
         assistant = app.config.assistants[0] if app.config.assistants \
             else None
-        if assistant is None:
-            return []
-        return self.agent_runner.build_skills(assistant, owner, app_id)
+        skills = []
+        if assistant is not None:
+            skills = list(self.agent_runner.build_skills(assistant, owner,
+                                                         app_id))
+        return skills + self._org_tools(owner)
 
     async def handle(self, app_id: str, owner: str, req: dict) -> Optional[dict]:
         """Handle one JSON-RPC request; returns a response dict (or None

@@ -103,3 +103,42 @@ def test_stream_message_fans_out_to_subscribed_bots(stack):
     hist = client.get(f"/api/v1/streams/{strm['id']}/messages",
                       headers=H(key)).json()
     assert [m["text"] for m in hist] == ["What is our refund policy?"]
+
+
+def test_org_primitives_via_mcp(stack):
+    """Org streams exposed as MCP tools: list, post (bot replies), read."""
+    client, key, store = stack
+    oid = _mk_org(client, key)
+    pos = client.post(f"/api/v1/organizations/{oid}/positions",
+                      json={"name": "support"}, headers=H(key)).json()
+    bot = client.post(f"/api/v1/organizations/{oid}/bots",
+                      json={"name": "helper", "position_id": pos["id"]},
+                      headers=H(key)).json()
+    strm = client.post(f"/api/v1/organizations/{oid}/streams",
+                       json={"name": "general"}, headers=H(key)).json()
+    client.post(f"/api/v1/bots/{bot['id']}/subscribe",
+                json={"stream_id": strm["id"]}, headers=H(key))
+    app_id = client.post("/api/v1/apps", json={"config": {
+        "name": "mcp-app", "assistants": [{"name": "a"}]}},
+        headers=H(key)).json()["id"]
+
+    def rpc(method, params=None, rid=1):
+        return client.post(f"/api/v1/mcp/{app_id}", json={
+            "jsonrpc": "2.0", "id": rid, "method": method,
+            "params": params or {}}, headers=H(key)).json()
+
+    tools = [t["name"] for t in rpc("tools/list")["result"]["tools"]]
+    assert {"org_list_streams", "org_read_stream",
+            "org_post_message"} <= set(tools)
+    import json as _json
+    r = rpc("tools/call", {"name": "org_list_streams", "arguments": {}})
+    streams = _json.loads(r["result"]["content"][0]["text"])
+    assert streams and streams[0]["name"] == "general"
+    r = rpc("tools/call", {"name": "org_post_message", "arguments": {
+        "stream_id": strm["id"], "text": "hello from mcp"}})
+    posted = _json.loads(r["result"]["content"][0]["text"])
+    assert posted["replies"][0]["text"] == "mock response"
+    r = rpc("tools/call", {"name": "org_read_stream", "arguments": {
+        "stream_id": strm["id"]}})
+    hist = _json.loads(r["result"]["content"][0]["text"])
+    assert hist[0]["text"] == "hello from mcp"
