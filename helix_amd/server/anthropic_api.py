@@ -19,11 +19,39 @@ def anthropic_to_openai(req: dict) -> dict:
         messages.append({"role": "system", "content": system})
     for m in req.get("messages", []):
         content = m.get("content")
+        role = m.get("role", "user")
         if isinstance(content, list):
-            content = " ".join(b.get("text", "") for b in content
-                               if isinstance(b, dict) and
-                               b.get("type") == "text")
-        messages.append({"role": m.get("role", "user"), "content": content})
+            texts = []
+            tool_calls = []
+            tool_results = []
+            for b in content:
+                if not isinstance(b, dict):
+                    continue
+                if b.get("type") == "text":
+                    texts.append(b.get("text", ""))
+                elif b.get("type") == "tool_use":
+                    tool_calls.append({
+                        "id": b.get("id", ""), "type": "function",
+                        "function": {
+                            "name": b.get("name", ""),
+                            "arguments": json.dumps(b.get("input", {}))}})
+                elif b.get("type") == "tool_result":
+                    rc = b.get("content", "")
+                    if isinstance(rc, list):
+                        rc = " ".join(x.get("text", "") for x in rc
+                                      if isinstance(x, dict))
+                    tool_results.append({
+                        "role": "tool",
+                        "tool_call_id": b.get("tool_use_id", ""),
+                        "content": rc})
+            msg = {"role": role, "content": " ".join(texts)}
+            if tool_calls:
+                msg["tool_calls"] = tool_calls
+            if texts or tool_calls:
+                messages.append(msg)
+            messages.extend(tool_results)
+        else:
+            messages.append({"role": role, "content": content})
     out = {
         "model": req.get("model", ""),
         "messages": messages,
@@ -35,21 +63,54 @@ def anthropic_to_openai(req: dict) -> dict:
             out[k] = req[k]
     if req.get("stop_sequences"):
         out["stop"] = req["stop_sequences"]
+    # tool definitions (Anthropic input_schema -> OpenAI parameters)
+    if req.get("tools"):
+        out["tools"] = [{
+            "type": "function",
+            "function": {"name": t.get("name", ""),
+                         "description": t.get("description", ""),
+                         "parameters": t.get("input_schema", {})}}
+            for t in req["tools"]]
+    tc = req.get("tool_choice")
+    if isinstance(tc, dict):
+        if tc.get("type") == "tool":
+            out["tool_choice"] = {"type": "function", "function":
+                                  {"name": tc.get("name", "")}}
+        elif tc.get("type") in ("any", "auto"):
+            out["tool_choice"] = ("required" if tc["type"] == "any"
+                                  else "auto")
     return out
 
 
 def openai_to_anthropic(resp: dict) -> dict:
     choice = resp.get("choices", [{}])[0]
-    text = choice.get("message", {}).get("content", "") or ""
+    msg = choice.get("message", {})
+    text = msg.get("content", "") or ""
     usage = resp.get("usage", {})
-    stop_reason = {"stop": "end_turn", "length": "max_tokens"}.get(
+    stop_reason = {"stop": "end_turn", "length": "max_tokens",
+                   "tool_calls": "tool_use"}.get(
         choice.get("finish_reason", "stop"), "end_turn")
+    content = []
+    if text:
+        content.append({"type": "text", "text": text})
+    for tc in msg.get("tool_calls") or []:
+        fn = tc.get("function", {})
+        try:
+            args = json.loads(fn.get("arguments") or "{}")
+        except Exception:
+            args = {"_raw": fn.get("arguments", "")}
+        content.append({"type": "tool_use",
+                        "id": tc.get("id", f"toolu_{uuid.uuid4().hex[:16]}"),
+                        "name": fn.get("name", ""), "input": args})
+        stop_reason = "tool_use"
+    if not content:
+        content = [{"type": "text", "text": ""}]
     return {
         "id": f"msg_{uuid.uuid4().hex[:24]}",
         "type": "message",
         "role": "assistant",
         "model": resp.get("model", ""),
-        "content": [{"type": "text", "text": text}],
+        "content": content,
         "stop_reason": stop_reason,
         "stop_sequence": None,
         "usage": {"input_tokens": usage.get("prompt_tokens", 0),

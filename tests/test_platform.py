@@ -276,3 +276,47 @@ def test_debug_threads(stack):
                    headers={"Authorization": "Bearer admin-key"})
     assert r.status_code == 200
     assert any("MainThread" in k for k in r.json())
+
+
+def test_anthropic_tool_use_translation():
+    """/v1/messages tool-use round-trip translation (reference
+    api/pkg/anthropic proxies tool blocks)."""
+    from helix_amd.server.anthropic_api import (anthropic_to_openai,
+                                                openai_to_anthropic)
+    areq = {
+        "model": "m", "max_tokens": 64,
+        "tools": [{"name": "get_weather",
+                   "description": "Weather lookup",
+                   "input_schema": {"type": "object", "properties": {
+                       "city": {"type": "string"}}}}],
+        "tool_choice": {"type": "any"},
+        "messages": [
+            {"role": "user", "content": "Weather in Paris?"},
+            {"role": "assistant", "content": [
+                {"type": "tool_use", "id": "toolu_1",
+                 "name": "get_weather", "input": {"city": "Paris"}}]},
+            {"role": "user", "content": [
+                {"type": "tool_result", "tool_use_id": "toolu_1",
+                 "content": "18C, sunny"}]},
+        ],
+    }
+    oreq = anthropic_to_openai(areq)
+    assert oreq["tools"][0]["function"]["name"] == "get_weather"
+    assert oreq["tool_choice"] == "required"
+    roles = [m["role"] for m in oreq["messages"]]
+    assert roles == ["user", "assistant", "tool"]
+    assert oreq["messages"][1]["tool_calls"][0]["function"]["name"] == \
+        "get_weather"
+    assert oreq["messages"][2]["tool_call_id"] == "toolu_1"
+
+    oresp = {"model": "m", "choices": [{"finish_reason": "tool_calls",
+             "message": {"content": "",
+                         "tool_calls": [{"id": "call_9", "type": "function",
+                                         "function": {"name": "get_weather",
+                                                      "arguments":
+                                                      '{"city": "Paris"}'}}]}}],
+             "usage": {"prompt_tokens": 10, "completion_tokens": 5}}
+    aresp = openai_to_anthropic(oresp)
+    assert aresp["stop_reason"] == "tool_use"
+    blk = aresp["content"][0]
+    assert blk["type"] == "tool_use" and blk["input"] == {"city": "Paris"}
