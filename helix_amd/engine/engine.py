@@ -46,6 +46,7 @@ class Sequence:
     arrival_time: float = field(default_factory=time.monotonic)
     first_token_time: Optional[float] = None
     finish_reason: Optional[str] = None
+    logprobs: List[dict] = field(default_factory=list)  # when requested
     # streaming callback: fn(seq, new_token_id, finished)
     on_token: Optional[Callable] = None
 
@@ -457,7 +458,27 @@ class LLMEngine:
         temps = torch.from_numpy(temps_np).to(logits.device)
         seeds_t = torch.from_numpy(seeds_np).to(logits.device)
         toks = ops.sample_tokens(logits.contiguous(), temps, seeds_t)
-        return toks.cpu().tolist()
+        out = toks.cpu().tolist()
+        # top-k logprobs for sequences that requested them (one extra
+        # log_softmax + topk over just those rows)
+        lp_rows = [i for i, s_ in enumerate(batch) if s_.params.logprobs]
+        if lp_rows:
+            k = max(batch[i].params.logprobs for i in lp_rows)
+            sub = logits[lp_rows].float()
+            lsm = torch.log_softmax(sub, dim=-1)
+            topv, topi = lsm.topk(min(k, lsm.shape[-1]), dim=-1)
+            topv = topv.cpu().tolist()
+            topi = topi.cpu().tolist()
+            for j, i in enumerate(lp_rows):
+                tok = out[i]
+                tok_lp = float(lsm[j, tok])
+                kk = batch[i].params.logprobs
+                batch[i].logprobs.append({
+                    "token": tok, "logprob": tok_lp,
+                    "top_logprobs": [
+                        {"token": t, "logprob": v}
+                        for t, v in zip(topi[j][:kk], topv[j][:kk])]})
+        return out
 
     def _process_logits(self, batch, logits: torch.Tensor) -> torch.Tensor:
         for i, seq in enumerate(batch):

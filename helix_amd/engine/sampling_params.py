@@ -19,6 +19,7 @@ class SamplingParams:
     stop_token_ids: List[int] = field(default_factory=list)
     ignore_eos: bool = False
     seed: Optional[int] = None
+    logprobs: Optional[int] = None     # top-k logprobs per emitted token
 
     @property
     def needs_logit_processing(self) -> bool:

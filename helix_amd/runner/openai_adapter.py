@@ -48,6 +48,8 @@ def _params_from_request(req: dict, max_model_len: int) -> SamplingParams:
         presence_penalty=float(req.get("presence_penalty") or 0.0),
         frequency_penalty=float(req.get("frequency_penalty") or 0.0),
         seed=req.get("seed"),
+        logprobs=(int(req.get("top_logprobs") or 1)
+                  if req.get("logprobs") else None),
     ), ([stop] if isinstance(stop, str) else list(stop or []))
 
 
@@ -59,17 +61,22 @@ class TokenStream:
         self.q: asyncio.Queue = asyncio.Queue()
 
     def on_token(self, seq, token_id: int, finished: bool):
+        lp = None
+        lps = getattr(seq, "logprobs", None)
+        if lps:
+            lp = lps[-1]       # engine appended this token's entry
         try:
             self.loop.call_soon_threadsafe(
-                self.q.put_nowait, (token_id, finished, seq.finish_reason))
+                self.q.put_nowait,
+                (token_id, finished, seq.finish_reason, lp))
         except RuntimeError:
             # client's event loop is gone (disconnect) — tokens go nowhere
             pass
 
     async def __aiter__(self):
         while True:
-            tok, fin, reason = await self.q.get()
-            yield tok, fin, reason
+            tok, fin, reason, lp = await self.q.get()
+            yield tok, fin, reason, lp
             if fin:
                 return
 
@@ -103,14 +110,17 @@ async def chat_completion(service: RunnerService, req: dict,
     choices = []
     total_completion = 0
     for i in range(n):
-        text, finish_reason, ntok = await _generate_one(
+        text, finish_reason, ntok, lp_content = await _generate_one(
             inst, f"{rid}-{i}", prompt_ids, params, stop_strs, tok, loop)
         total_completion += ntok
-        choices.append({
+        choice = {
             "index": i,
             "message": {"role": "assistant", "content": text},
             "finish_reason": finish_reason,
-        })
+        }
+        if params.logprobs:
+            choice["logprobs"] = {"content": lp_content}
+        choices.append(choice)
     return {
         "id": rid,
         "object": "chat.completion",
@@ -133,9 +143,12 @@ async def _generate_one(inst, seq_id, prompt_ids, params, stop_strs, tok,
     text = ""
     finish_reason = "stop"
     ntok = 0
-    async for token_id, fin, reason in ts.__aiter__():
+    lp_content = []
+    async for token_id, fin, reason, lp in ts.__aiter__():
         ntok += 1
         text += detok.push(token_id)
+        if lp is not None:
+            lp_content.append(_lp_entry(lp, tok))
         if fin:
             finish_reason = reason or "stop"
             break
@@ -145,7 +158,18 @@ async def _generate_one(inst, seq_id, prompt_ids, params, stop_strs, tok,
             text = text[:hit]
             finish_reason = "stop"
             break
-    return text, finish_reason, ntok
+    return text, finish_reason, ntok, lp_content
+
+
+def _lp_entry(lp: dict, tok) -> dict:
+    """Engine logprob record -> OpenAI chat logprobs content entry."""
+    return {
+        "token": tok.decode([lp["token"]]),
+        "logprob": lp["logprob"],
+        "top_logprobs": [{"token": tok.decode([t["token"]]),
+                          "logprob": t["logprob"]}
+                         for t in lp.get("top_logprobs", [])],
+    }
 
 
 def _find_stop(text: str, stop_strs: List[str]) -> Optional[int]:
@@ -183,7 +207,7 @@ async def _stream(service, inst, model, rid, created, prompt_ids, params,
         yield chunk({"role": "assistant", "content": ""})
         emitted = 0
         finish_reason = "stop"
-        async for token_id, fin, reason in ts.__aiter__():
+        async for token_id, fin, reason, _lp in ts.__aiter__():
             ntok += 1
             delta = detok.push(token_id)
             hit = _find_stop(detok.emitted, stop_strs)

@@ -401,3 +401,35 @@ def test_row_pool_stress_chunked_preempt():
         assert len(eng.seqs[f"s{i}"].output_ids) == 24, i
     assert eng.num_free_blocks() == 48
     assert len(eng._free_rows) == 6
+
+
+def test_logprobs_requested():
+    """SamplingParams.logprobs: per-token top-k logprobs are recorded
+    and the sampled token's logprob is consistent with its rank."""
+    import math
+    eng = LLMEngine(EngineConfig(model="tiny", max_model_len=128,
+                                 max_num_seqs=4, kv_cache_blocks=64,
+                                 eos_token_id=-1, seed=0),
+                    device="cpu")
+    sp = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True,
+                        logprobs=3)
+    out = eng.generate([[1, 2, 3]], sp)[0]
+    seq = eng.seqs[list(eng.seqs)[0]]
+    assert len(seq.logprobs) == 5
+    for step, tok in zip(seq.logprobs, out):
+        assert step["token"] == tok
+        assert len(step["top_logprobs"]) == 3
+        # greedy: the sampled token IS the top-1
+        assert step["top_logprobs"][0]["token"] == tok
+        assert math.isclose(step["top_logprobs"][0]["logprob"],
+                            step["logprob"], rel_tol=1e-5)
+        assert step["logprob"] <= 0.0
+    # no logprobs unless requested
+    eng2 = LLMEngine(EngineConfig(model="tiny", max_model_len=128,
+                                  max_num_seqs=4, kv_cache_blocks=64,
+                                  eos_token_id=-1),
+                     device="cpu")
+    eng2.generate([[1, 2, 3]], SamplingParams(temperature=0.0,
+                                              max_tokens=3,
+                                              ignore_eos=True))
+    assert all(not s.logprobs for s in eng2.seqs.values())

@@ -198,3 +198,30 @@ def test_register_spec_and_load(tmp_path):
         "name": "tiny-fp8", "preset": "tiny"}, headers=H)
     assert r.status_code == 409        # loaded: must unload first
     svc.shutdown()
+
+
+def test_openai_logprobs(tmp_path):
+    """OpenAI logprobs field on the native runner surface."""
+    import asyncio
+    from helix_amd.runner.openai_adapter import chat_completion
+    from helix_amd.runner.service import RunnerService
+    svc = RunnerService(device="cpu")
+    try:
+        resp = asyncio.run(chat_completion(svc, {
+            "model": "tiny", "messages": [{"role": "user",
+                                           "content": "hi"}],
+            "max_tokens": 4, "temperature": 0, "logprobs": True,
+            "top_logprobs": 2}))
+        lp = resp["choices"][0]["logprobs"]["content"]
+        assert len(lp) == 4
+        assert len(lp[0]["top_logprobs"]) == 2
+        assert lp[0]["logprob"] <= 0.0
+        assert isinstance(lp[0]["token"], str)
+        # absent unless requested
+        resp2 = asyncio.run(chat_completion(svc, {
+            "model": "tiny", "messages": [{"role": "user",
+                                           "content": "hi"}],
+            "max_tokens": 2, "temperature": 0}))
+        assert "logprobs" not in resp2["choices"][0]
+    finally:
+        svc.shutdown()
