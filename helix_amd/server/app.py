@@ -276,6 +276,36 @@ def create_app(cfg: Optional[ServerConfig] = None,
         body.setdefault("model", model)
         return await _chat(request, user, body)
 
+    @app.post("/v1/completions")
+    async def legacy_completions(request: Request,
+                                 user: AuthUser = Depends(auth_dep)):
+        """Legacy text-completions: translated onto the chat surface
+        (prompt -> user message; text_completion response shape)."""
+        body = await request.json()
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):
+            prompt = prompt[0] if prompt else ""
+        chat_body = {k: v for k, v in body.items()
+                     if k not in ("prompt", "echo", "suffix")}
+        chat_body["messages"] = [{"role": "user", "content": prompt}]
+        if body.get("stream"):
+            raise HTTPException(400, "streaming not supported on the "
+                                     "legacy completions surface; use "
+                                     "/v1/chat/completions")
+        resp = await _chat(request, user, chat_body)
+        if isinstance(resp, dict):
+            choices = [{
+                "index": c.get("index", i),
+                "text": c.get("message", {}).get("content", ""),
+                "finish_reason": c.get("finish_reason"),
+                "logprobs": c.get("logprobs"),
+            } for i, c in enumerate(resp.get("choices", []))]
+            return {"id": resp.get("id"), "object": "text_completion",
+                    "created": resp.get("created"),
+                    "model": resp.get("model"), "choices": choices,
+                    "usage": resp.get("usage")}
+        return resp
+
     @app.post("/v1/embeddings")
     async def embeddings(request: Request,
                          user: AuthUser = Depends(auth_dep)):

@@ -414,3 +414,13 @@ def test_wedged_interaction_auto_errors(stack):
     assert n == 1
     assert store.get("interactions", stale.id)["state"] == "error"
     assert store.get("interactions", fresh.id)["state"] == "waiting"
+
+
+def test_legacy_completions_surface(stack):
+    _, client, _, key, _ = stack
+    r = client.post("/v1/completions", json={
+        "model": "mock-model", "prompt": "Say hi"}, headers=H(key))
+    assert r.status_code == 200
+    body = r.json()
+    assert body["object"] == "text_completion"
+    assert body["choices"][0]["text"] == "mock response"
