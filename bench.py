@@ -36,6 +36,12 @@ def main():
                     choices=["dp", "tp"],
                     help="dp: one engine per GPU (weak scaling); "
                          "tp: one tensor-parallel engine over all GPUs")
+    ap.add_argument("--quant", type=str, default=None,
+                    choices=[None, "fp8"],
+                    help="fp8: e4m3 W8A8 projections (opt-in)")
+    ap.add_argument("--kv-dtype", type=str, default="bf16",
+                    choices=["bf16", "fp8"],
+                    help="fp8: e4m3 KV cache (opt-in)")
     args = ap.parse_args()
 
     import torch.distributed as dist
@@ -67,6 +73,8 @@ def main():
         max_prefill_tokens=16384,
         eos_token_id=-1,
         seed=rank,
+        quantization=args.quant,
+        kv_cache_dtype=args.kv_dtype,
     )
     if tp:
         # SPMD tensor parallelism: every rank runs the identical engine
@@ -130,7 +138,7 @@ def main():
             "higher_is_better": True,
             "scaling": "strong" if tp else "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": ("fp8-w8a8" if args.quant == "fp8" else "bf16"),
             "data": "synthetic",
             "config": {
                 "model": args.model,
@@ -139,6 +147,7 @@ def main():
                 "parallelism": (f"tp{world}" if tp else f"dp{world}"),
                 "ttft_p50_ms": round(ttft_p50, 2),
                 "prefill_s": round(prefill_s, 3),
+                "kv_cache_dtype": args.kv_dtype,
             },
         }))
     if distributed:
