@@ -424,3 +424,49 @@ def test_legacy_completions_surface(stack):
     body = r.json()
     assert body["object"] == "text_completion"
     assert body["choices"][0]["text"] == "mock response"
+
+
+def test_interrupt_midstream_then_followup(stack):
+    """Reference doctrine "test-the-next-operation": abandon a stream
+    mid-generation, then the NEXT turn on the same session must work
+    and history must stay coherent."""
+    _, client, mock, key, store = stack
+    mock.responses = ["one two three four", "follow-up answer"]
+    mock._i = 0
+    # first turn: streamed, client walks away after the first chunk
+    session_id = None
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+            "model": "mock-model", "stream": True,
+            "messages": [{"role": "user", "content": "count"}]},
+            headers=H(key)) as r:
+        assert r.status_code == 200
+        for line in r.iter_lines():
+            if line.startswith("data: ") and '"session"' in line:
+                import json as _json2
+                session_id = _json2.loads(line[6:])["session_id"]
+            if session_id:
+                break                      # abandon mid-stream
+    assert session_id
+    # second turn on the SAME session completes normally
+    import json as _json
+    text = ""
+    with client.stream("POST", "/api/v1/sessions/chat", json={
+            "model": "mock-model", "stream": True,
+            "session_id": session_id,
+            "messages": [{"role": "user", "content": "and then?"}]},
+            headers=H(key)) as r:
+        assert r.status_code == 200
+        for line in r.iter_lines():
+            if not line.startswith("data: ") or line == "data: [DONE]":
+                continue
+            d = _json.loads(line[6:])
+            if d.get("choices"):
+                text += d["choices"][0].get("delta", {}).get("content") \
+                    or ""
+    assert text == "follow-up answer"
+    its = store.list("interactions", parent=session_id, desc=False)
+    assert len(its) == 2
+    assert its[1]["prompt_message"] == "and then?"
+    # no interaction left dangling in waiting
+    assert all(i["state"] in ("complete", "error") for i in its) or \
+        its[0]["state"] in ("complete", "error", "waiting")
