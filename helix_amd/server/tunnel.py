@@ -47,13 +47,19 @@ class TunnelRegistry:
             raise ConnectionError(f"runner {runner_id} has no tunnel")
         rid = uuid.uuid4().hex
         self._pending[rid] = asyncio.Queue()
-        await q.put({"id": rid, "path": path, "body": body})
+        try:
+            await q.put({"id": rid, "path": path, "body": body})
+        except BaseException:
+            self._pending.pop(rid, None)
+            raise
         return rid
 
     async def events(self, rid: str, timeout: float = 300.0
                      ) -> AsyncIterator[dict]:
         """Yield reply events for a request until 'end'/'error'."""
         q = self._pending.get(rid)
+        if q is None:                      # already reaped / unknown id
+            return
         try:
             while True:
                 ev = await asyncio.wait_for(q.get(), timeout)
@@ -119,6 +125,8 @@ class TunnelClient:
                 from helix_amd.server.providers import ProviderError
                 raise ProviderError(ev.get("message", "tunnel error"),
                                     ev.get("status", 502))
+        from helix_amd.server.providers import ProviderError
+        raise ProviderError("tunnel closed without response", 502)
 
 
 # ---------------------------------------------------------------------------
