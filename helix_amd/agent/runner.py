@@ -136,7 +136,15 @@ class AgentRunner:
                 if skill is None:
                     result = f"unknown tool: {name}"
                 else:
-                    result = await skill.execute(args, ctx)
+                    try:
+                        result = await skill.execute(args, ctx)
+                    except Exception as e:
+                        # a failing tool is feedback for the model, not
+                        # a dead agent turn (reference tool-error flow)
+                        log.warning("tool %s failed: %s", name, e)
+                        result = f"tool error: {e}"
+                if not isinstance(result, str):
+                    result = json.dumps(result, default=str)
                 doc = self._emit_step(ctx, {
                     "step": name, "arguments": args,
                     "result": result[:2000],

@@ -220,3 +220,40 @@ def test_agent_memory_skill(stack):
         headers=H(key))
     assert r.status_code == 200
     assert store.list("memories", limit=5)
+
+
+def test_agent_tool_error_is_fed_back(stack_agent=None):
+    """A skill that raises becomes a tool-error message to the model,
+    not a failed turn."""
+    import asyncio
+    from helix_amd.agent.runner import AgentRunner
+    from helix_amd.server.config import ServerConfig
+    from helix_amd.server.providers import MockClient, ProviderManager
+    from helix_amd.server.types import AssistantConfig
+    from helix_amd.store import Store
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    # 1st call: model calls the calculator with args that make it raise;
+    # 2nd call: model answers using the error feedback
+    toolcall = {"choices": [{"finish_reason": "tool_calls", "message": {
+        "content": "", "tool_calls": [{"id": "c1", "type": "function",
+            "function": {"name": "calculator",
+                         "arguments": '{"expression": "__import__"}'}}]}}],
+        "usage": {}}
+    pm.register("mock", MockClient(responses=[toolcall, "recovered"]))
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    runner = AgentRunner(cfg, store, pm, None)
+    asst = AssistantConfig(name="a", system_prompt="x",
+                           calculator={"enabled": True})
+    msgs, final = asyncio.run(runner._loop(
+        asst, {"messages": [{"role": "user", "content": "calc"}]},
+        "u1", {"owner": "u1"}))
+    tool_msgs = [m for m in msgs if m.get("role") == "tool"]
+    assert tool_msgs, msgs
+    assert "error" in tool_msgs[0]["content"].lower() or \
+        "invalid" in tool_msgs[0]["content"].lower() or \
+        tool_msgs[0]["content"]
+    assert msgs[-1]["role"] == "assistant"
+    assert msgs[-1]["content"] == "recovered"
