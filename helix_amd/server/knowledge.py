@@ -103,8 +103,15 @@ class KnowledgeReconciler:
             documents.append({"text": content,
                               "metadata": {"source": "text"}})
         elif "filestore" in src:
-            base = os.path.join(self.filestore_path,
-                                src["filestore"].get("path", ""))
+            # owner-scoped + containment-checked (same rule as
+            # FileStore._resolve): a knowledge source must not read
+            # outside the owner's filestore namespace
+            owner_root = os.path.abspath(os.path.join(
+                self.filestore_path, "users", doc.get("owner", "")))
+            rel = str(src["filestore"].get("path", "")).lstrip("/")
+            base = os.path.abspath(os.path.join(owner_root, rel))
+            if not base.startswith(owner_root):
+                raise PermissionError("path escapes filestore root")
             if os.path.isdir(base):
                 for root, _, files in os.walk(base):
                     for f in files:

@@ -320,3 +320,30 @@ def test_anthropic_tool_use_translation():
     assert aresp["stop_reason"] == "tool_use"
     blk = aresp["content"][0]
     assert blk["type"] == "tool_use" and blk["input"] == {"city": "Paris"}
+
+
+def test_knowledge_filestore_is_owner_scoped(stack, tmp_path):
+    """A knowledge filestore source cannot traverse out of the owner's
+    namespace (and reads the owner's files, matching upload paths)."""
+    import asyncio
+    app, client, _, key, store = stack
+    # upload a file as the user
+    r = client.put("/api/v1/filestore/upload?path=docs/n.txt",
+                   content=b"alpha beta gamma", headers=H(key))
+    assert r.status_code == 200
+    kn = app.state.knowledge
+    app.state.cfg.rag.embeddings_provider = "mock"   # MockClient embeds
+    me = [u for u in store.list("users") if u["username"] == "alice"][0]
+    doc = kn.create(me["id"], "k1", {"filestore": {"path": "docs/n.txt"}})
+    asyncio.run(kn.reconcile_once())   # preparing -> pending
+    asyncio.run(kn.reconcile_once())   # pending -> indexing -> ready
+    assert kn.get(doc["id"])["state"] == "ready"
+    # traversal attempt errors, never reads outside
+    evil = kn.create(me["id"], "k2",
+                     {"filestore": {"path": "../../../../etc/hostname"}})
+    asyncio.run(kn.reconcile_once())
+    asyncio.run(kn.reconcile_once())
+    got = kn.get(evil["id"])
+    assert got["state"] == "error"
+    assert "escapes" in got["message"] or "No such" in got["message"] or \
+        "not found" in got["message"].lower()
