@@ -217,3 +217,29 @@ def test_spec_task_implementation_agent(tmp_path):
                          doc["branch"]) == "def login():\n    return True\n"
     log = git.log(rid, doc["branch"], 3)
     assert log[0]["subject"].startswith("add login module")
+
+
+def test_git_ref_and_path_injection_rejected(stack):
+    app, client, key, store = stack
+    pid = client.post("/api/v1/projects", json={"name": "sec"},
+                      headers=H(key)).json()["id"]
+    rid = store.get("projects", pid)["repo_id"]
+    git = app.state.git
+    git.commit_files(rid, {"a.txt": "hello"}, "init")
+    # option-looking refs are rejected, not passed to git
+    assert git.log(rid, "--output=/tmp/pwn") == []
+    assert git.ls_tree(rid, "--output=/tmp/pwn2") == []
+    import pytest as _pt
+    with _pt.raises(ValueError):
+        git.read_file(rid, "a.txt", ref="--help")
+    with _pt.raises(ValueError):
+        git.read_file(rid, "--flag", ref="main")
+    import os as _os
+    assert not _os.path.exists("/tmp/pwn")
+    # commit manifests cannot escape the worktree
+    with _pt.raises(ValueError):
+        git.commit_files(rid, {"../../evil.txt": "x"}, "bad")
+    with _pt.raises(ValueError):
+        git.commit_files(rid, {"ok.txt": "x"}, "bad", branch="--force")
+    # normal flows still work
+    assert git.read_file(rid, "a.txt", ref="main") == "hello"
