@@ -15,9 +15,11 @@ class FileStore:
         os.makedirs(self.root, exist_ok=True)
 
     def _resolve(self, owner: str, path: str) -> str:
-        base = os.path.join(self.root, "users", owner)
+        base = os.path.abspath(os.path.join(self.root, "users", owner))
         full = os.path.abspath(os.path.join(base, path.lstrip("/")))
-        if not full.startswith(base):
+        # note the os.sep suffix: plain startswith would accept a
+        # sibling dir whose name has this owner's as a prefix
+        if full != base and not full.startswith(base + os.sep):
             raise PermissionError("path escapes filestore root")
         return full
 

@@ -110,7 +110,8 @@ class KnowledgeReconciler:
                 self.filestore_path, "users", doc.get("owner", "")))
             rel = str(src["filestore"].get("path", "")).lstrip("/")
             base = os.path.abspath(os.path.join(owner_root, rel))
-            if not base.startswith(owner_root):
+            if base != owner_root and \
+                    not base.startswith(owner_root + os.sep):
                 raise PermissionError("path escapes filestore root")
             if os.path.isdir(base):
                 for root, _, files in os.walk(base):
