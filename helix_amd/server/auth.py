@@ -97,7 +97,7 @@ class Authenticator:
     def resolve(self, token: str) -> Optional[AuthUser]:
         if not token:
             return None
-        if token == self.admin_api_key:
+        if hmac.compare_digest(token, self.admin_api_key):
             return AuthUser(id="admin", username="admin", admin=True)
         if token.count(".") == 2:                      # JWT shape
             claims = jwt_decode(token, self.jwt_secret)
@@ -115,7 +115,8 @@ class Authenticator:
                         admin=(user or {}).get("admin", False))
 
     def is_runner(self, token: str) -> bool:
-        return bool(token) and token == self.runner_token
+        return bool(token) and hmac.compare_digest(token,
+                                                   self.runner_token)
 
 
 def bearer_token(request: Request) -> str:
