@@ -29,6 +29,29 @@ class AgentRunner:
         self.rag = rag
 
     # ------------------------------------------------------------------
+    def _resolve_secrets(self, owner: str, mapping: dict) -> dict:
+        """Interpolate ${secrets.NAME} in tool header/query values
+        (reference tools secret injection), decrypting at-rest values."""
+        import re as _re
+        from helix_amd.server.crypto import decrypt_str, secrets_key
+        key = secrets_key(self.cfg.web.admin_api_key)
+
+        def sub(val: str) -> str:
+            def repl(m):
+                doc = self.store.get("secrets",
+                                     f"{owner}:{m.group(1)}")
+                if doc is None:
+                    return m.group(0)
+                try:
+                    return decrypt_str(doc.get("value", ""), key)
+                except ValueError:
+                    return m.group(0)
+            return _re.sub(r"\$\{secrets\.([A-Za-z0-9_\-]+)\}", repl,
+                           val)
+
+        return {k: sub(v) if isinstance(v, str) else v
+                for k, v in (mapping or {}).items()}
+
     def build_skills(self, assistant: AssistantConfig, owner: str,
                      app_id: str = "") -> List[Skill]:
         skills: List[Skill] = []
@@ -40,7 +63,10 @@ class AgentRunner:
                 [k.name for k in assistant.knowledge], owner))
         for api in assistant.apis:
             try:
-                skills.append(APISkill(api))
+                sk = APISkill(api)
+                sk.headers = self._resolve_secrets(owner, sk.headers)
+                sk.query = self._resolve_secrets(owner, sk.query)
+                skills.append(sk)
             except Exception as e:
                 log.warning("failed to build API skill %s: %s", api.name, e)
         if assistant.web_search.get("enabled",

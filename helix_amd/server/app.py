@@ -1173,8 +1173,11 @@ def create_app(cfg: Optional[ServerConfig] = None,
         if not name:
             raise HTTPException(400, "name required")
         sid = f"{user.id}:{name}"
+        from helix_amd.server.crypto import encrypt_str, secrets_key
         store.put("secrets", sid, {"id": sid, "name": name,
-                                   "value": body.get("value", "")},
+                                   "value": encrypt_str(
+                                       str(body.get("value", "")),
+                                       secrets_key(cfg.web.admin_api_key))},
                   owner=user.id)
         return {"name": name, "ok": True}
 

@@ -1,0 +1,73 @@
+"""Secret encryption at rest + ${secrets.NAME} tool interpolation."""
+import pytest
+from fastapi.testclient import TestClient
+
+from helix_amd.server.app import create_app
+from helix_amd.server.config import ServerConfig
+from helix_amd.server.crypto import decrypt_str, encrypt_str
+from helix_amd.server.providers import MockClient, ProviderManager
+from helix_amd.store import Store
+
+
+def test_crypto_roundtrip_and_tamper():
+    ct = encrypt_str("tok-123", "k1")
+    assert ct.startswith("enc1:")
+    assert "tok-123" not in ct
+    assert decrypt_str(ct, "k1") == "tok-123"
+    with pytest.raises(ValueError):
+        decrypt_str(ct, "wrong-key")
+    # tamper
+    bad = ct[:-6] + ("AAAAAA" if not ct.endswith("AAAAAA") else "BBBBBB")
+    with pytest.raises(Exception):
+        decrypt_str(bad, "k1")
+    # legacy plaintext passes through
+    assert decrypt_str("plain-old", "k1") == "plain-old"
+    # unique nonces: same plaintext encrypts differently
+    assert encrypt_str("x", "k") != encrypt_str("x", "k")
+
+
+def test_secret_encrypted_at_rest_and_interpolated(tmp_path):
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    cfg.filestore.path = str(tmp_path / "fs")
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    pm.register("mock", MockClient())
+    app = create_app(cfg, store=store, providers=pm)
+    client = TestClient(app)
+    key = client.post("/api/v1/users", json={"username": "sue"},
+                      headers={"Authorization": "Bearer admin-key"}
+                      ).json()["api_key"]
+    H = {"Authorization": f"Bearer {key}"}
+    r = client.post("/api/v1/secrets",
+                    json={"name": "API_TOKEN", "value": "sk-verysecret"},
+                    headers=H)
+    assert r.status_code == 200
+    me = [u for u in store.list("users") if u["username"] == "sue"][0]
+    row = store.get("secrets", f"{me['id']}:API_TOKEN")
+    assert row["value"].startswith("enc1:")
+    assert "sk-verysecret" not in row["value"]
+
+    # interpolation into an API skill's headers
+    from helix_amd.server.types import AssistantConfig, ToolAPIConfig
+    runner = app.state  # agent runner lives in create_app scope; rebuild
+    from helix_amd.agent.runner import AgentRunner
+    ar = AgentRunner(cfg, store, pm, None)
+    asst = AssistantConfig(name="a", apis=[ToolAPIConfig(
+        name="svc", url="http://127.0.0.1:9",
+        schema_="openapi: 3.0.0\ninfo: {title: t, version: '1'}\npaths: {}",
+        headers={"Authorization": "Bearer ${secrets.API_TOKEN}",
+                 "X-Plain": "keep"})])
+    skills = ar.build_skills(asst, me["id"])
+    api = [s for s in skills if s.name.startswith("api_")][0]
+    assert api.headers["Authorization"] == "Bearer sk-verysecret"
+    assert api.headers["X-Plain"] == "keep"
+    # unknown secret placeholders stay literal (no crash)
+    asst2 = AssistantConfig(name="b", apis=[ToolAPIConfig(
+        name="svc2", url="http://127.0.0.1:9",
+        schema_="openapi: 3.0.0\ninfo: {title: t, version: '1'}\npaths: {}",
+        headers={"A": "${secrets.MISSING}"})])
+    api2 = [s for s in ar.build_skills(asst2, me["id"])
+            if s.name.startswith("api_")][0]
+    assert api2.headers["A"] == "${secrets.MISSING}"
