@@ -19,10 +19,21 @@ class PubSub:
         for pattern, queues in list(self._subs.items()):
             if fnmatch.fnmatch(topic, pattern):
                 for q in list(queues):
-                    await q.put((topic, msg))
+                    # slow-consumer policy (NATS-like): drop the OLDEST
+                    # message rather than grow without bound or block
+                    # every publisher behind one dead SSE client
+                    while True:
+                        try:
+                            q.put_nowait((topic, msg))
+                            break
+                        except asyncio.QueueFull:
+                            try:
+                                q.get_nowait()
+                            except asyncio.QueueEmpty:
+                                break
 
     async def subscribe(self, pattern: str) -> "Subscription":
-        q: asyncio.Queue = asyncio.Queue()
+        q: asyncio.Queue = asyncio.Queue(maxsize=1024)
         async with self._lock:
             self._subs[pattern].append(q)
         return Subscription(self, pattern, q)

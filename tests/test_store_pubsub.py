@@ -86,3 +86,22 @@ def test_pubsub_request_reply():
         await task
         assert out["echo"] == 42
     asyncio.run(run())
+
+
+def test_pubsub_slow_consumer_drops_oldest():
+    import asyncio
+    from helix_amd.server.pubsub import PubSub
+
+    async def run():
+        ps = PubSub()
+        sub = await ps.subscribe("t.*")
+        for i in range(1500):            # > maxsize
+            await ps.publish("t.x", i)
+        # newest survive; oldest dropped
+        topic, first = await sub.get(timeout=1)
+        assert first == 1500 - 1024
+        for _ in range(1023):
+            _, last = await sub.get(timeout=1)
+        assert last == 1499
+        await sub.close()
+    asyncio.run(run())
