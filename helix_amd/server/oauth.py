@@ -8,6 +8,7 @@ from typing import Dict, List, Optional
 
 import httpx
 
+from helix_amd.server.crypto import decrypt_str, encrypt_str, secrets_key
 from helix_amd.server.types import new_id
 
 # Built-in provider endpoint templates (reference oauth providers incl.
@@ -32,9 +33,12 @@ BUILTIN_PROVIDERS = {
 
 
 class OAuthManager:
-    def __init__(self, store, http_client: Optional[httpx.AsyncClient] = None):
+    def __init__(self, store, http_client: Optional[httpx.AsyncClient] = None,
+                 enc_key: str = ""):
         self.store = store
         self._http = http_client
+        # tokens are encrypted at rest (same scheme as secrets)
+        self._key = enc_key or secrets_key("oauth")
 
     def _client(self) -> httpx.AsyncClient:
         if self._http is None:
@@ -86,15 +90,27 @@ class OAuthManager:
     def save_token(self, owner: str, provider: str, tok: dict) -> dict:
         doc = {
             "id": f"{owner}:{provider}", "provider": provider,
-            "access_token": tok.get("access_token", ""),
-            "refresh_token": tok.get("refresh_token", ""),
+            "access_token": encrypt_str(tok.get("access_token", ""),
+                                        self._key),
+            "refresh_token": encrypt_str(tok.get("refresh_token", ""),
+                                         self._key),
             "expires_at": time.time() + float(tok.get("expires_in", 3600)),
         }
         self.store.put("oauth_tokens", doc["id"], doc, owner=owner)
-        return doc
+        return {**doc, "access_token": tok.get("access_token", ""),
+                "refresh_token": tok.get("refresh_token", "")}
 
     def token_for(self, owner: str, provider: str) -> Optional[dict]:
-        return self.store.get("oauth_tokens", f"{owner}:{provider}")
+        doc = self.store.get("oauth_tokens", f"{owner}:{provider}")
+        if doc is None:
+            return None
+        out = dict(doc)
+        for f in ("access_token", "refresh_token"):
+            try:
+                out[f] = decrypt_str(doc.get(f, ""), self._key)
+            except ValueError:
+                out[f] = ""
+        return out
 
     async def get_valid_token(self, owner: str, provider: str
                               ) -> Optional[str]:

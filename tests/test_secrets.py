@@ -71,3 +71,21 @@ def test_secret_encrypted_at_rest_and_interpolated(tmp_path):
     api2 = [s for s in ar.build_skills(asst2, me["id"])
             if s.name.startswith("api_")][0]
     assert api2.headers["A"] == "${secrets.MISSING}"
+
+
+def test_oauth_tokens_encrypted_at_rest():
+    import asyncio
+    from helix_amd.server.oauth import OAuthManager
+    store = Store(":memory:")
+    om = OAuthManager(store)
+    om.save_token("u1", "github", {"access_token": "gho_abc123",
+                                   "refresh_token": "ghr_xyz",
+                                   "expires_in": 3600})
+    raw = store.get("oauth_tokens", "u1:github")
+    assert raw["access_token"].startswith("enc1:")
+    assert "gho_abc123" not in raw["access_token"]
+    tok = om.token_for("u1", "github")
+    assert tok["access_token"] == "gho_abc123"
+    assert tok["refresh_token"] == "ghr_xyz"
+    got = asyncio.run(om.get_valid_token("u1", "github"))
+    assert got == "gho_abc123"
