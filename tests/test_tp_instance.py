@@ -109,3 +109,41 @@ def test_tp_instance_via_runner_service():
         assert ev.wait(timeout=60)
     finally:
         svc.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_tp_instance_worker_death_errors_inflight(tmp_path):
+    """If the TP worker fleet dies, in-flight callbacks get an error
+    finish instead of hanging forever."""
+    from helix_amd.runner.tp_instance import TPLLMInstance
+    spec = ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=256,
+                     max_num_seqs=4, kv_cache_blocks=128, tp=2)
+    inst = TPLLMInstance(spec, 2, device_type="cpu", backend="gloo")
+    try:
+        done = threading.Event()
+        started = threading.Event()
+        reasons = []
+
+        def cb(seq, tok, fin):
+            started.set()
+            if fin:
+                reasons.append(seq.finish_reason)
+                done.set()
+
+        # kill the workers AFTER the request is demonstrably streaming,
+        # then simulate the engine-death notice the rank-0 worker emits
+        # on unexpected exceptions
+        inst.submit("r0", [1, 2, 3],
+                    SamplingParams(temperature=0.0, max_tokens=4096,
+                                   ignore_eos=True), cb)
+        assert started.wait(timeout=60), "request never started"
+        for p in inst.procs:
+            p.terminate()
+        for p in inst.procs:
+            p.join(timeout=10)
+        inst.evt_q.put(("dead", None, None, None, "simulated crash"))
+        assert done.wait(timeout=30), "in-flight callback never finished"
+        assert reasons and reasons[0].startswith("error"), reasons
+        assert inst.in_flight == 0
+    finally:
+        inst.shutdown()
