@@ -5,9 +5,19 @@ invariant the GPU RCCL path relies on).
 """
 import os
 
+import socket
+
 import pytest
 import torch
 import torch.multiprocessing as mp
+
+
+def _free_port() -> str:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return str(p)
 
 from helix_amd.models.llama import PRESETS, LlamaForCausalLM, PrefillMeta
 from helix_amd.parallel import shard_llama_state_dict
@@ -27,9 +37,9 @@ def _full_forward(cfg_name, ids, positions, cu, max_len):
 
 
 def _tp_worker(rank, world, cfg_name, ids, positions, cu, max_len, sd,
-               out_path):
+               out_path, port):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
-                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29612",
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=port,
                       LOCAL_RANK=str(rank))
     from helix_amd import parallel
     parallel.init_tp(world, backend="gloo")
@@ -64,9 +74,10 @@ def test_tp2_matches_full(cfg_name):
     with tempfile.TemporaryDirectory() as td:
         out_path = os.path.join(td, "logits.pt")
         ctx = mp.get_context("spawn")
+        port = _free_port()
         procs = [ctx.Process(target=_tp_worker,
                              args=(r, 2, cfg_name, ids, positions, cu,
-                                   max(lens), sd, out_path))
+                                   max(lens), sd, out_path, port))
                  for r in range(2)]
         for p in procs:
             p.start()
@@ -87,9 +98,9 @@ def test_shard_shapes():
         m2.load_state_dict(shard, strict=True)
 
 
-def _tp_engine_worker(rank, world, sd, prompts, out_path):
+def _tp_engine_worker(rank, world, sd, prompts, out_path, port):
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
-                      MASTER_ADDR="127.0.0.1", MASTER_PORT="29613",
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=port,
                       LOCAL_RANK=str(rank))
     from helix_amd import parallel
     from helix_amd.engine.engine import EngineConfig, LLMEngine
@@ -133,8 +144,9 @@ def test_tp2_engine_generate_matches_single():
     with tempfile.TemporaryDirectory() as td:
         out_path = os.path.join(td, "out.pt")
         ctx = mp.get_context("spawn")
+        port = _free_port()
         procs = [ctx.Process(target=_tp_engine_worker,
-                             args=(r, 2, sd, prompts, out_path))
+                             args=(r, 2, sd, prompts, out_path, port))
                  for r in range(2)]
         for p in procs:
             p.start()
