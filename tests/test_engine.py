@@ -433,3 +433,22 @@ def test_logprobs_requested():
                                               max_tokens=3,
                                               ignore_eos=True))
     assert all(not s.logprobs for s in eng2.seqs.values())
+
+
+def test_chunked_prefill_with_sliding_window():
+    """Chunked prefill composed with sliding-window attention must match
+    the unchunked result (tiny-sw: window 24)."""
+    import torch
+    torch.manual_seed(2)
+    prompt = [((i * 29) % 500) + 1 for i in range(120)]
+    sp = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+
+    def run(budget):
+        eng = LLMEngine(EngineConfig(model="tiny-sw", max_model_len=256,
+                                     max_num_seqs=4, kv_cache_blocks=128,
+                                     max_prefill_tokens=budget,
+                                     eos_token_id=-1, seed=6),
+                        device="cpu")
+        return eng.generate([prompt], sp)[0]
+
+    assert run(32) == run(8192)
