@@ -280,6 +280,38 @@ def spectask_plan(task_id: str, url: str = typer.Option("", "--url")):
 model_app = typer.Typer(help="Model catalog & local models")
 app.add_typer(model_app, name="model")
 
+app_app = typer.Typer(help="Manage apps/agents")
+app.add_typer(app_app, name="app")
+
+
+@app_app.command("list")
+def app_list(url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    for a in httpx.get(f"{api}/api/v1/apps", headers=headers).json():
+        cfg = (a.get("config") or {}).get("helix") or a.get("config") or {}
+        n = len(cfg.get("assistants") or [])
+        typer.echo(f"{a['id']:<30} {cfg.get('name', ''):<24} "
+                   f"{n} assistant(s)")
+
+
+@app_app.command("get")
+def app_get(app_id: str, url: str = typer.Option("", "--url")):
+    import json as _json
+
+    import httpx
+    api, headers = _api(url)
+    r = httpx.get(f"{api}/api/v1/apps/{app_id}", headers=headers)
+    typer.echo(_json.dumps(r.json(), indent=2))
+
+
+@app_app.command("delete")
+def app_delete(app_id: str, url: str = typer.Option("", "--url")):
+    import httpx
+    api, headers = _api(url)
+    r = httpx.delete(f"{api}/api/v1/apps/{app_id}", headers=headers)
+    typer.echo("deleted" if r.status_code == 200 else f"error: {r.text}")
+
 
 @model_app.command("list")
 def model_list(url: str = typer.Option("", "--url")):
