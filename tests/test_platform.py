@@ -358,3 +358,12 @@ def test_filestore_prefix_sibling_rejected(tmp_path):
     import pytest as _pt
     with _pt.raises((PermissionError, FileNotFoundError)):
         fs.read("u1", "../u1evil/secret.txt")
+
+
+def test_webui_served(stack):
+    _, client, _, _, _ = stack
+    r = client.get("/")
+    assert r.status_code == 200
+    assert "<title>helix_amd</title>" in r.text
+    for pane in ("Chat", "Apps", "Knowledge", "Runners", "Usage"):
+        assert pane in r.text
