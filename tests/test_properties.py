@@ -1,0 +1,92 @@
+"""Property-based tests (hypothesis) for core invariants: allocator
+refcounts, chunker coverage, crypto roundtrip, tokenizer identity,
+GGUF metadata roundtrip."""
+import string
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from helix_amd.engine.kv_cache import BlockAllocator
+from helix_amd.rag.chunker import chunk_text
+from helix_amd.server.crypto import decrypt_str, encrypt_str
+from helix_amd.utils.tokenizer import ByteTokenizer
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.sampled_from(["alloc", "free", "share"]),
+                min_size=1, max_size=120),
+       st.integers(min_value=2, max_value=24))
+def test_block_allocator_invariants(ops, nblocks):
+    """Random alloc/free/share sequences: no double-allocation, counts
+    conserved, freeing everything restores capacity."""
+    a = BlockAllocator(nblocks)
+    live = []          # blocks we hold (with multiplicity = refcount)
+    for op in ops:
+        if op == "alloc" and a.can_allocate(1):
+            (b,) = a.allocate(1)
+            live.append(b)
+        elif op == "free" and live:
+            b = live.pop()
+            a.free([b])
+        elif op == "share" and live:
+            b = live[0]
+            a.share(b)
+            live.append(b)
+        # invariant: a block's refcount equals our multiplicity
+        for b in set(live):
+            assert a.ref.get(b, 0) == live.count(b)
+        assert a.free_count + len(set(live)) == nblocks
+    a.free(list(live))
+    assert a.free_count == nblocks
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.text(alphabet=string.printable, min_size=0, max_size=4000),
+       st.integers(min_value=32, max_value=512))
+def test_chunker_coverage_and_bounds(text, size):
+    chunks = chunk_text(text, chunk_size=size, overlap=min(16, size // 4))
+    joined = "\n".join(c["text"] for c in chunks)
+    # every non-whitespace word of the input appears in some chunk
+    for w in text.split():
+        assert w in joined or any(w in c["text"] for c in chunks) or \
+            len(w) > size  # long words are window-split
+    # chunks never wildly exceed the budget (word packing + overlap)
+    for c in chunks:
+        assert len(c["text"]) <= 2 * size + 64
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.text(min_size=0, max_size=500),
+       st.text(min_size=1, max_size=40))
+def test_crypto_roundtrip_any_unicode(plaintext, key):
+    assert decrypt_str(encrypt_str(plaintext, key), key) == plaintext
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.text(min_size=0, max_size=500))
+def test_byte_tokenizer_identity(text):
+    t = ByteTokenizer()
+    assert t.decode(t.encode(text)) == text
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.dictionaries(
+    st.text(alphabet=string.ascii_letters + ".", min_size=1, max_size=24),
+    st.one_of(st.integers(min_value=0, max_value=2**31 - 1),
+              st.text(max_size=40), st.booleans(),
+              st.floats(min_value=-1e6, max_value=1e6,
+                        allow_nan=False, width=32)),
+    max_size=8))
+def test_gguf_metadata_roundtrip(tmp_path_factory, meta):
+    import torch
+
+    from helix_amd.engine import gguf
+    path = str(tmp_path_factory.mktemp("gguf") / "m.gguf")
+    gguf.write_gguf(path, meta, {"w": torch.zeros(2, 2)})
+    g = gguf.GGUFFile(path)
+    for k, v in meta.items():
+        got = g.metadata[k]
+        if isinstance(v, float):
+            assert abs(got - v) <= max(1e-3, abs(v) * 1e-5)
+        else:
+            assert got == v
