@@ -90,3 +90,43 @@ def test_gguf_metadata_roundtrip(tmp_path_factory, meta):
             assert abs(got - v) <= max(1e-3, abs(v) * 1e-5)
         else:
             assert got == v
+
+
+@settings(max_examples=15, deadline=None)
+@given(st.lists(
+    st.one_of(
+        st.tuples(st.just("add"),
+                  st.integers(min_value=1, max_value=90),   # prompt len
+                  st.integers(min_value=1, max_value=12)),  # max_tokens
+        st.tuples(st.just("cancel"), st.integers(min_value=0, max_value=20),
+                  st.just(0)),
+        st.tuples(st.just("step"), st.just(0), st.just(0)),
+    ), min_size=1, max_size=40))
+def test_engine_fuzz_conserves_blocks(ops):
+    """Random add/cancel/step interleavings: the engine never crashes,
+    and when drained every block and row slot returns."""
+    from helix_amd.engine.engine import EngineConfig, LLMEngine
+    from helix_amd.engine.sampling_params import SamplingParams
+    eng = LLMEngine(EngineConfig(model="tiny", max_model_len=128,
+                                 max_num_seqs=4, kv_cache_blocks=32,
+                                 max_prefill_tokens=48, eos_token_id=-1,
+                                 seed=1),
+                    device="cpu")
+    n = 0
+    for op, a, b in ops:
+        if op == "add":
+            n += 1
+            eng.add_request(f"f{n}", [(i % 400) + 1 for i in range(a)],
+                            SamplingParams(temperature=0.0, max_tokens=b,
+                                           ignore_eos=True))
+        elif op == "cancel":
+            eng.cancel(f"f{a}")          # may or may not exist
+        else:
+            eng.step()
+    guard = 0
+    while eng.has_work:
+        eng.step()
+        guard += 1
+        assert guard < 2000
+    assert eng.num_free_blocks() == 32
+    assert len(eng._free_rows) == 4
