@@ -150,9 +150,12 @@ def paged_attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
 
 def sample_tokens(logits: torch.Tensor, temperatures: torch.Tensor,
                   seeds: torch.Tensor) -> torch.Tensor:
-    """Greedy for T<=0. For T>0 the HIP kernel uses Gumbel-max with its own
-    RNG; the reference uses torch sampling (distribution-equal, not
-    bit-equal — GPU tests check greedy exactly and stochastic statistically).
+    """Greedy for T<=0; Gumbel-max for T>0 driven by the per-row seeds.
+
+    Seed-determinism matters even on CPU: SPMD-TP ranks each run this
+    sampler and must pick identical tokens (the HIP kernel derives its
+    Gumbel noise from the same seeds; distribution-equal, not bit-equal
+    — GPU tests check greedy exactly and stochastic statistically).
     """
     out = torch.empty(logits.shape[0], dtype=torch.int64,
                       device=logits.device)
@@ -162,8 +165,11 @@ def sample_tokens(logits: torch.Tensor, temperatures: torch.Tensor,
         if t <= 0:
             out[i] = int(row.argmax())
         else:
-            probs = torch.softmax(row / t, dim=-1)
-            out[i] = int(torch.multinomial(probs, 1))
+            g = torch.Generator()
+            g.manual_seed(int(seeds[i]) & 0x7FFFFFFFFFFFFFFF)
+            u = torch.rand(row.shape[0], generator=g)
+            gumbel = -torch.log(-torch.log(u.clamp(min=1e-20)))
+            out[i] = int((row / t + gumbel).argmax())
     return out
 
 
