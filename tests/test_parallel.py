@@ -155,3 +155,65 @@ def test_tp2_engine_generate_matches_single():
         assert all(p.exitcode == 0 for p in procs)
         got = torch.load(out_path)
     assert got == want
+
+
+def _tp_temp_worker(rank, world, sd, prompts, out_path, port):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=port,
+                      LOCAL_RANK=str(rank))
+    from helix_amd import parallel
+    from helix_amd.engine.engine import EngineConfig, LLMEngine
+    from helix_amd.engine.sampling_params import SamplingParams
+    parallel.init_tp(world, backend="gloo")
+    cfg = PRESETS["tiny-gqa"]
+    model = LlamaForCausalLM(cfg, tp_size=world, tp_rank=rank).float()
+    model.load_state_dict(shard_llama_state_dict(sd, cfg, world, rank),
+                          strict=True)
+    eng = LLMEngine(EngineConfig(model="tiny-gqa", max_model_len=256,
+                                 max_num_seqs=4, kv_cache_blocks=128,
+                                 eos_token_id=-1),
+                    device="cpu", model=model, tp_size=world, tp_rank=rank)
+    out = eng.generate(prompts, SamplingParams(temperature=0.8,
+                                               max_tokens=8, seed=77,
+                                               ignore_eos=True))
+    # EVERY rank writes: both must have sampled identical tokens
+    torch.save(out, f"{out_path}.{rank}")
+    torch.distributed.destroy_process_group()
+
+
+def test_tp2_temperature_sampling_lockstep():
+    """At temperature>0, both TP ranks must sample the same tokens (the
+    seed-driven Gumbel sampler is the lockstep mechanism) and match the
+    single-process engine."""
+    import tempfile
+
+    from helix_amd.engine.engine import EngineConfig, LLMEngine
+    from helix_amd.engine.sampling_params import SamplingParams
+    cfg = PRESETS["tiny-gqa"]
+    torch.manual_seed(0)
+    full = LlamaForCausalLM(cfg).float()
+    full.init_random(0)
+    sd = full.state_dict()
+    prompts = [[1, 2, 3, 4, 5], [9, 8, 7]]
+    eng = LLMEngine(EngineConfig(model="tiny-gqa", max_model_len=256,
+                                 max_num_seqs=4, kv_cache_blocks=128,
+                                 eos_token_id=-1),
+                    device="cpu", model=full)
+    want = eng.generate(prompts, SamplingParams(temperature=0.8,
+                                                max_tokens=8, seed=77,
+                                                ignore_eos=True))
+    with tempfile.TemporaryDirectory() as td:
+        out_path = os.path.join(td, "out.pt")
+        ctx = mp.get_context("spawn")
+        port = _free_port()
+        procs = [ctx.Process(target=_tp_temp_worker,
+                             args=(r, 2, sd, prompts, out_path, port))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=300)
+        assert all(p.exitcode == 0 for p in procs)
+        r0 = torch.load(f"{out_path}.0")
+        r1 = torch.load(f"{out_path}.1")
+    assert r0 == r1 == want
