@@ -5,6 +5,7 @@ sessions, apps/agents, runner plane, admin.
 from __future__ import annotations
 
 import asyncio
+import os
 import json
 import logging
 from typing import Optional
@@ -1379,6 +1380,24 @@ def create_app(cfg: Optional[ServerConfig] = None,
             out[f"{names.get(tid, '?')}-{tid}"] = \
                 traceback.format_stack(frame)[-6:]
         return out
+
+    @app.post("/api/v1/admin/backup")
+    async def store_backup(request: Request,
+                           user: AuthUser = Depends(admin_dep)):
+        """Online store backup to a server-side path (reference
+        Postgres dump role)."""
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        import time as _t
+        path = body.get("path") or os.path.join(
+            cfg.filestore.path, "backups",
+            _t.strftime("helix-%Y%m%d-%H%M%S.db"))
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        store.backup(path)
+        return {"ok": True, "path": path,
+                "bytes": os.path.getsize(path)}
 
     @app.post("/api/v1/admin/janitor")
     async def run_janitor(request: Request,

@@ -62,6 +62,17 @@ class Store:
                 (id, owner, parent, now, now, blob))
             self._db.commit()
 
+    def backup(self, path: str):
+        """Consistent online backup (sqlite3 backup API) — the
+        operational role of the reference's Postgres dump/restore."""
+        import sqlite3 as _sq
+        with self._lock:
+            dst = _sq.connect(path)
+            try:
+                self._db.backup(dst)
+            finally:
+                dst.close()
+
     def get(self, table: str, id: str) -> Optional[Dict[str, Any]]:
         with self._lock:
             row = self._db.execute(

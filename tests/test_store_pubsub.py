@@ -105,3 +105,32 @@ def test_pubsub_slow_consumer_drops_oldest():
         assert last == 1499
         await sub.close()
     asyncio.run(run())
+
+
+def test_store_backup_restore(tmp_path):
+    from helix_amd.store import Store
+    s = Store(":memory:")
+    s.put("apps", "a1", {"id": "a1", "x": 1}, owner="u")
+    s.put("sessions", "s1", {"id": "s1"}, owner="u")
+    path = str(tmp_path / "bak.db")
+    s.backup(path)
+    restored = Store(path)
+    assert restored.get("apps", "a1") == {"id": "a1", "x": 1}
+    assert restored.get("sessions", "s1") == {"id": "s1"}
+
+
+def test_admin_backup_endpoint(tmp_path):
+    from fastapi.testclient import TestClient
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import ServerConfig
+    from helix_amd.store import Store
+    cfg = ServerConfig()
+    cfg.filestore.path = str(tmp_path / "fs")
+    app = create_app(cfg, store=Store(":memory:"))
+    client = TestClient(app)
+    r = client.post("/api/v1/admin/backup", json={},
+                    headers={"Authorization": "Bearer admin-key"})
+    assert r.status_code == 200
+    import os as _os
+    assert _os.path.exists(r.json()["path"])
+    assert Store(r.json()["path"]).count("users") >= 0
