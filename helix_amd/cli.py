@@ -341,6 +341,19 @@ def model_unload(model: str, url: str = typer.Option("", "--url")):
 
 
 @app.command()
+def backup(url: str = typer.Option("", "--url"),
+           path: str = typer.Option("", "--path",
+                                    help="server-side destination")):
+    """Online control-plane store backup (admin)."""
+    import httpx
+    api, headers = _api(url)
+    body = {"path": path} if path else {}
+    r = httpx.post(f"{api}/api/v1/admin/backup", json=body,
+                   headers=headers, timeout=120)
+    typer.echo(r.json())
+
+
+@app.command()
 def version():
     from helix_amd import __version__
     typer.echo(f"helix_amd {__version__}")
