@@ -122,10 +122,24 @@ class CUDAGraphRunner:
         entry["graph"].replay()
         return entry["logits"][:B]
 
-    def warmup(self, len_buckets=(MIN_LEN_BUCKET, 2 * MIN_LEN_BUCKET)):
-        """Eagerly capture all batch buckets so serving never pays
-        multi-second capture latency mid-request (cold-start cost moves
-        to model load)."""
+    def all_len_buckets(self):
+        """Every seqlen bucket up to max_model_len (512, 1024, ...)."""
+        out = []
+        b = MIN_LEN_BUCKET
+        while True:
+            out.append(min(b, self.max_model_len))
+            if b >= self.max_model_len:
+                break
+            b *= 2
+        return out
+
+    def warmup(self, len_buckets=None):
+        """Eagerly capture ALL (batch, seqlen) buckets so serving never
+        pays multi-second capture latency mid-request — including long
+        contexts (cold-start cost moves to model load; bound it by
+        lowering the spec's max_model_len if load time matters more)."""
+        if len_buckets is None:
+            len_buckets = self.all_len_buckets()
         self.in_ids.zero_()
         self.in_pos.zero_()
         self.in_slots.fill_(-1)
