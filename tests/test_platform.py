@@ -367,3 +367,26 @@ def test_webui_served(stack):
     assert "<title>helix_amd</title>" in r.text
     for pane in ("Chat", "Apps", "Knowledge", "Runners", "Usage"):
         assert pane in r.text
+
+
+def test_janitor_session_retention(stack):
+    _, client, _, _, store = stack
+    import time as _t
+    H_admin = {"Authorization": "Bearer admin-key"}
+    old_ms = int((_t.time() - 90 * 86400) * 1000)
+    store.put("sessions", "old_s", {"id": "old_s", "updated": old_ms},
+              owner="u")
+    store.put("interactions", "old_i", {"id": "old_i",
+                                        "session_id": "old_s"},
+              owner="u", parent="old_s")
+    store.put("sessions", "new_s", {"id": "new_s",
+                                    "updated": int(_t.time() * 1000)},
+              owner="u")
+    r = client.post("/api/v1/admin/janitor",
+                    json={"retention_days": 30,
+                          "session_retention_days": 30}, headers=H_admin)
+    pruned = r.json()
+    assert pruned["sessions"] == 1 and pruned["interactions"] == 1
+    assert store.get("sessions", "old_s") is None
+    assert store.get("interactions", "old_i") is None
+    assert store.get("sessions", "new_s") is not None
