@@ -51,7 +51,7 @@ def test_tp_instance_matches_single_process(tmp_path):
     spec = ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=256,
                      max_num_seqs=4, kv_cache_blocks=128, tp=2)
     inst = TPLLMInstance(spec, 2, device_type="cpu", backend="gloo",
-                         sd_path=sd_path)
+                         sd_path=sd_path, start_timeout=120)
     try:
         got0 = _collect_stream(inst, "s0", prompts[0], sp)
         got1 = _collect_stream(inst, "s1", prompts[1], sp)
@@ -118,7 +118,8 @@ def test_tp_instance_worker_death_errors_inflight(tmp_path):
     from helix_amd.runner.tp_instance import TPLLMInstance
     spec = ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=256,
                      max_num_seqs=4, kv_cache_blocks=128, tp=2)
-    inst = TPLLMInstance(spec, 2, device_type="cpu", backend="gloo")
+    inst = TPLLMInstance(spec, 2, device_type="cpu", backend="gloo",
+                         start_timeout=120)
     try:
         done = threading.Event()
         started = threading.Event()
