@@ -130,3 +130,34 @@ def test_e2e_models_endpoint(e2e):
                   headers={"Authorization": f"Bearer {key}"})
     ids = [m["id"] for m in r.json()["data"]]
     assert "tiny" in ids
+
+
+def test_e2e_cli_test_verb(e2e, tmp_path):
+    """`helix-amd test -f helix.yaml` runs assistant tests against the
+    live stack with the LLM judge (reference `helix test`)."""
+    import os
+    import subprocess
+    import sys
+    base, key, _ = e2e
+    yaml_path = tmp_path / "app.yaml"
+    yaml_path.write_text("""
+name: cli-test-app
+assistants:
+  - name: default
+    model: tiny
+    tests:
+      - name: replies
+        steps:
+          - prompt: "say anything"
+            expected_output: "any reply at all is accepted by the judge"
+""")
+    env = dict(os.environ, HELIX_URL=base, HELIX_API_KEY=key)
+    r = subprocess.run(
+        [sys.executable, "-m", "helix_amd.cli", "test", "-f",
+         str(yaml_path), "--url", base],
+        capture_output=True, text=True, timeout=180, env=env)
+    # tiny random-init model judges arbitrarily: PASS or FAIL both prove
+    # the loop ran end-to-end; crash/timeout would be a real failure
+    assert r.returncode in (0, 1), r.stderr
+    assert "replies" in r.stdout
+    assert ("PASS" in r.stdout) or ("FAIL" in r.stdout)
