@@ -124,6 +124,7 @@ class LLMInstance:
         self.intake: queue.Queue = queue.Queue()
         self.wake = threading.Event()
         self.stop = False
+        self.draining = False
         self.last_used = time.time()
         self.thread = threading.Thread(target=self._loop, daemon=True,
                                        name=f"engine-{spec.name}")
@@ -178,6 +179,15 @@ class LLMInstance:
 
     def submit(self, seq_id: str, prompt_ids: List[int],
                params: SamplingParams, on_token) -> None:
+        if self.draining:
+            if on_token is not None:
+                class _F:
+                    pass
+                f = _F()
+                f.seq_id = seq_id
+                f.finish_reason = "error: draining for shutdown"
+                on_token(f, 0, True)
+            return
         self.last_used = time.time()
         self.intake.put(("submit", (seq_id, prompt_ids, params, on_token)))
         self.wake.set()
@@ -185,6 +195,18 @@ class LLMInstance:
     def cancel(self, seq_id: str):
         self.intake.put(("cancel", seq_id))
         self.wake.set()
+
+    def drain(self, timeout: float = 30.0) -> bool:
+        """Stop accepting new work and wait for in-flight sequences to
+        finish (rolling-restart support). Returns True if fully drained
+        before the deadline."""
+        self.draining = True
+        deadline = time.time() + timeout
+        while time.time() < deadline:
+            if self.in_flight == 0:
+                return True
+            time.sleep(0.05)
+        return self.in_flight == 0
 
     def shutdown(self):
         self.stop = True
@@ -337,6 +359,16 @@ class RunnerService:
                 "last_used": inst.last_used,
             })
         return out
+
+    def drain(self, timeout: float = 30.0) -> bool:
+        """Drain every instance (new submissions rejected; in-flight
+        finishes) within the shared deadline."""
+        deadline = time.time() + timeout
+        ok = True
+        for inst in list(self.instances.values()):
+            if hasattr(inst, "drain"):
+                ok &= inst.drain(max(0.1, deadline - time.time()))
+        return ok
 
     def shutdown(self):
         with self._lock:

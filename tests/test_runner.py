@@ -225,3 +225,39 @@ def test_openai_logprobs(tmp_path):
         assert "logprobs" not in resp2["choices"][0]
     finally:
         svc.shutdown()
+
+
+def test_graceful_drain():
+    """drain(): in-flight requests finish; new submissions are rejected
+    with an error finish."""
+    import threading
+    from helix_amd.engine.sampling_params import SamplingParams
+    from helix_amd.runner.service import RunnerService
+    svc = RunnerService(device="cpu")
+    try:
+        inst = svc.ensure_loaded("tiny")
+        done = threading.Event()
+        toks = []
+
+        def cb(seq, tok, fin):
+            toks.append(tok)
+            if fin:
+                done.set()
+        inst.submit("d0", [1, 2, 3],
+                    SamplingParams(temperature=0.0, max_tokens=20,
+                                   ignore_eos=True), cb)
+        assert svc.drain(timeout=60) is True
+        assert done.is_set()
+        assert len(toks) == 20           # in-flight ran to completion
+        # post-drain submissions are rejected immediately
+        rejected = threading.Event()
+        reasons = []
+
+        def cb2(seq, tok, fin):
+            reasons.append(seq.finish_reason)
+            rejected.set()
+        inst.submit("d1", [4, 5], SamplingParams(max_tokens=4), cb2)
+        assert rejected.wait(timeout=5)
+        assert reasons[0].startswith("error")
+    finally:
+        svc.shutdown()
