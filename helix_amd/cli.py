@@ -86,7 +86,14 @@ def runner(api_url: str = typer.Option("http://localhost:8080"),
         asyncio.run(assignment_loop(
             api_url, cfg.runner_plane.runner_token, runner_id, svc))
     threading.Thread(target=poll_assignment, daemon=True).start()
-    uvicorn.run(api, host="0.0.0.0", port=port, log_level="info")
+    try:
+        uvicorn.run(api, host="0.0.0.0", port=port, log_level="info")
+    finally:
+        # graceful drain on SIGTERM/SIGINT (rolling restarts): reject
+        # new work, let in-flight sequences finish, then free the GPU
+        typer.echo("draining in-flight requests...")
+        svc.drain(timeout=30)
+        svc.shutdown()
 
 
 def _local_ip() -> str:
