@@ -46,8 +46,11 @@ class CalculatorSkill(Skill):
                                                          (int, float)):
             return node.value
         if isinstance(node, ast.BinOp) and type(node.op) in self._OPS:
-            return self._OPS[type(node.op)](self._eval(node.left),
-                                            self._eval(node.right))
+            left = self._eval(node.left)
+            right = self._eval(node.right)
+            if isinstance(node.op, ast.Pow) and abs(right) > 1024:
+                raise ValueError("exponent too large")
+            return self._OPS[type(node.op)](left, right)
         if isinstance(node, ast.UnaryOp) and type(node.op) in self._OPS:
             return self._OPS[type(node.op)](self._eval(node.operand))
         raise ValueError(f"unsupported expression element: {node!r}")
