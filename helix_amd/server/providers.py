@@ -346,13 +346,17 @@ class LoggingClient(Client):
                     call.prompt_tokens = usage.get("prompt_tokens", 0)
                     call.completion_tokens = usage.get("completion_tokens", 0)
                 yield chunk
-            call.duration_ms = int((time.monotonic() - t0) * 1000)
-            self._log(call)
-        except Exception as e:
-            call.error = str(e)
-            call.duration_ms = int((time.monotonic() - t0) * 1000)
-            self._log(call)
+        except BaseException as e:
+            # BaseException so a client disconnect (GeneratorExit) is
+            # also recorded; the original exception propagates unchanged
+            if isinstance(e, GeneratorExit):
+                call.error = "client disconnected"
+            else:
+                call.error = str(e)
             raise
+        finally:
+            call.duration_ms = int((time.monotonic() - t0) * 1000)
+            self._log(call)
 
     async def embeddings(self, req: dict) -> dict:
         return await self.inner.embeddings(req)
