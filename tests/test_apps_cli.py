@@ -82,6 +82,6 @@ def test_cli_doctor_runs():
     import subprocess
     import sys
     r = subprocess.run([sys.executable, "-m", "helix_amd.cli", "doctor"],
-                       capture_output=True, text=True, timeout=120)
+                       capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "extension built" in r.stdout

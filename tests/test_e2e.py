@@ -155,7 +155,7 @@ assistants:
     r = subprocess.run(
         [sys.executable, "-m", "helix_amd.cli", "test", "-f",
          str(yaml_path), "--url", base],
-        capture_output=True, text=True, timeout=180, env=env)
+        capture_output=True, text=True, timeout=300, env=env)
     # tiny random-init model judges arbitrarily: PASS or FAIL both prove
     # the loop ran end-to-end; crash/timeout would be a real failure
     assert r.returncode in (0, 1), r.stderr

@@ -135,7 +135,7 @@ def test_weights_loader_dispatches_gguf(tmp_path):
     r = subprocess.run(
         [sys.executable, "-m", "helix_amd.cli", "export-gguf",
          "--preset", "tiny", "-o", out, "--seed", "3"],
-        capture_output=True, text=True, timeout=120)
+        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr
     from helix_amd.engine.weights import load_llama_weights
     torch.manual_seed(0)

@@ -79,7 +79,7 @@ def test_eval_quant_harness_runs(tmp_path):
     r = subprocess.run(
         [sys.executable, "scripts/eval_quant.py", "--prompts", "3",
          "--steps", "6", "--model", "tiny-gqa"],
-        capture_output=True, text=True, timeout=240)
+        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr
     rep = json.loads(r.stdout)
     assert set(rep["variants"]) == {"bf16", "fp8_weights", "fp8_kv",
