@@ -91,6 +91,31 @@ def create_runner_app(service: RunnerService,
     async def healthz():
         return {"ok": True}
 
+    @app.get("/metrics")
+    async def prometheus():
+        # minimal runner-side Prometheus text (control plane has the
+        # full registry; this covers per-node scraping)
+        lines = [
+            "# TYPE helix_runner_models_loaded gauge",
+            f"helix_runner_models_loaded {len(service.loaded_models())}",
+        ]
+        for m in service.status():
+            mid = m["model_id"]
+            lines.append(f'helix_runner_in_flight{{model="{mid}"}} '
+                         f'{m["in_flight"]}')
+            lines.append(f'helix_runner_model_bytes{{model="{mid}"}} '
+                         f'{m["memory_bytes"]}')
+        try:
+            import torch
+            if torch.cuda.is_available():
+                free, total = torch.cuda.mem_get_info()
+                lines.append(f"helix_runner_hbm_free_bytes {free}")
+                lines.append(f"helix_runner_hbm_total_bytes {total}")
+        except Exception:
+            pass
+        from fastapi.responses import PlainTextResponse
+        return PlainTextResponse("\n".join(lines) + "\n")
+
     @app.get("/api/v1/status")
     async def status():
         return {

@@ -261,3 +261,19 @@ def test_graceful_drain():
         assert reasons[0].startswith("error")
     finally:
         svc.shutdown()
+
+
+def test_runner_metrics_endpoint():
+    from fastapi.testclient import TestClient
+    from helix_amd.runner.http import create_runner_app
+    from helix_amd.runner.service import RunnerService
+    svc = RunnerService(device="cpu")
+    try:
+        svc.ensure_loaded("tiny")
+        client = TestClient(create_runner_app(svc, "r-test"))
+        r = client.get("/metrics")
+        assert r.status_code == 200
+        assert "helix_runner_models_loaded 1" in r.text
+        assert 'helix_runner_in_flight{model="tiny"}' in r.text
+    finally:
+        svc.shutdown()
