@@ -19,6 +19,28 @@ def _free_port() -> str:
     s.close()
     return str(p)
 
+
+def _spawn2(target, mkargs, timeout=300, attempts=2):
+    """Run a 2-rank spawn job; retry once on nonzero exits (rare gloo
+    rendezvous flakes under load). mkargs(rank, port) -> args tuple."""
+    ctx = mp.get_context("spawn")
+    for attempt in range(attempts):
+        port = _free_port()
+        procs = [ctx.Process(target=target, args=mkargs(r, port))
+                 for r in range(2)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=timeout)
+        codes = [p.exitcode for p in procs]
+        for p in procs:
+            if p.is_alive():
+                p.terminate()
+        if all(c == 0 for c in codes):
+            return
+        if attempt == attempts - 1:
+            raise AssertionError(f"spawn ranks failed: {codes}")
+
 from helix_amd.models.llama import PRESETS, LlamaForCausalLM, PrefillMeta
 from helix_amd.parallel import shard_llama_state_dict
 
@@ -73,17 +95,9 @@ def test_tp2_matches_full(cfg_name):
     import tempfile
     with tempfile.TemporaryDirectory() as td:
         out_path = os.path.join(td, "logits.pt")
-        ctx = mp.get_context("spawn")
-        port = _free_port()
-        procs = [ctx.Process(target=_tp_worker,
-                             args=(r, 2, cfg_name, ids, positions, cu,
-                                   max(lens), sd, out_path, port))
-                 for r in range(2)]
-        for p in procs:
-            p.start()
-        for p in procs:
-            p.join(timeout=180)
-        assert all(p.exitcode == 0 for p in procs)
+        _spawn2(_tp_worker,
+                lambda r, port: (r, 2, cfg_name, ids, positions, cu,
+                                 max(lens), sd, out_path, port))
         tp_logits = torch.load(out_path)
     torch.testing.assert_close(tp_logits, full_logits, atol=2e-3, rtol=2e-3)
 
@@ -143,16 +157,8 @@ def test_tp2_engine_generate_matches_single():
                                                 ignore_eos=True))
     with tempfile.TemporaryDirectory() as td:
         out_path = os.path.join(td, "out.pt")
-        ctx = mp.get_context("spawn")
-        port = _free_port()
-        procs = [ctx.Process(target=_tp_engine_worker,
-                             args=(r, 2, sd, prompts, out_path, port))
-                 for r in range(2)]
-        for p in procs:
-            p.start()
-        for p in procs:
-            p.join(timeout=300)
-        assert all(p.exitcode == 0 for p in procs)
+        _spawn2(_tp_engine_worker,
+                lambda r, port: (r, 2, sd, prompts, out_path, port))
         got = torch.load(out_path)
     assert got == want
 
@@ -204,16 +210,8 @@ def test_tp2_temperature_sampling_lockstep():
                                                 ignore_eos=True))
     with tempfile.TemporaryDirectory() as td:
         out_path = os.path.join(td, "out.pt")
-        ctx = mp.get_context("spawn")
-        port = _free_port()
-        procs = [ctx.Process(target=_tp_temp_worker,
-                             args=(r, 2, sd, prompts, out_path, port))
-                 for r in range(2)]
-        for p in procs:
-            p.start()
-        for p in procs:
-            p.join(timeout=300)
-        assert all(p.exitcode == 0 for p in procs)
+        _spawn2(_tp_temp_worker,
+                lambda r, port: (r, 2, sd, prompts, out_path, port))
         r0 = torch.load(f"{out_path}.0")
         r1 = torch.load(f"{out_path}.1")
     assert r0 == r1 == want
