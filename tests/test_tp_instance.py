@@ -15,6 +15,16 @@ from helix_amd.models.llama import LlamaForCausalLM, PRESETS
 from helix_amd.runner.service import ModelSpec
 
 
+def _mk_inst(spec, **kw):
+    """Create a TPLLMInstance with one retry (rare gloo rendezvous
+    flakes under load)."""
+    from helix_amd.runner.tp_instance import TPLLMInstance
+    try:
+        return TPLLMInstance(spec, spec.tp, **kw)
+    except RuntimeError:
+        return TPLLMInstance(spec, spec.tp, **kw)
+
+
 def _collect_stream(inst, seq_id, prompt, params):
     done = threading.Event()
     toks = []
@@ -32,7 +42,6 @@ def _collect_stream(inst, seq_id, prompt, params):
 
 @pytest.mark.timeout(300)
 def test_tp_instance_matches_single_process(tmp_path):
-    from helix_amd.runner.tp_instance import TPLLMInstance
     cfg = PRESETS["tiny-gqa"]
     torch.manual_seed(0)
     full = LlamaForCausalLM(cfg).float()
@@ -50,8 +59,8 @@ def test_tp_instance_matches_single_process(tmp_path):
 
     spec = ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=256,
                      max_num_seqs=4, kv_cache_blocks=128, tp=2)
-    inst = TPLLMInstance(spec, 2, device_type="cpu", backend="gloo",
-                         sd_path=sd_path, start_timeout=120)
+    inst = _mk_inst(spec, device_type="cpu", backend="gloo",
+                    sd_path=sd_path, start_timeout=120)
     try:
         got0 = _collect_stream(inst, "s0", prompts[0], sp)
         got1 = _collect_stream(inst, "s1", prompts[1], sp)
@@ -115,11 +124,10 @@ def test_tp_instance_via_runner_service():
 def test_tp_instance_worker_death_errors_inflight(tmp_path):
     """If the TP worker fleet dies, in-flight callbacks get an error
     finish instead of hanging forever."""
-    from helix_amd.runner.tp_instance import TPLLMInstance
     spec = ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=256,
                      max_num_seqs=4, kv_cache_blocks=128, tp=2)
-    inst = TPLLMInstance(spec, 2, device_type="cpu", backend="gloo",
-                         start_timeout=120)
+    inst = _mk_inst(spec, device_type="cpu", backend="gloo",
+                    start_timeout=120)
     try:
         done = threading.Event()
         started = threading.Event()
