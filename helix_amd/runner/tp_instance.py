@@ -163,6 +163,8 @@ class TPLLMInstance:
             max_model_len=spec.max_model_len,
             max_num_seqs=spec.max_num_seqs,
             kv_cache_blocks=spec.kv_cache_blocks,
+            quantization=getattr(spec, "quantization", None),
+            kv_cache_dtype=getattr(spec, "kv_cache_dtype", "bf16"),
         )
         if getattr(spec, "eos_token_id", None) is not None:
             engine_kwargs["eos_token_id"] = spec.eos_token_id
