@@ -230,15 +230,19 @@ class TPLLMInstance:
                     except Exception:
                         log.exception("on_token callback failed")
             elif kind == "dead":
-                log.error("TP engine died: %s", evt[4])
-                with self._lock:
-                    cbs, self._cbs = self._cbs, {}
-                    self._inflight = 0
-                for seq_id, cb in cbs.items():
-                    try:
-                        cb(_SeqProxy(seq_id, f"error: {evt[4]}"), 0, True)
-                    except Exception:
-                        pass
+                self._on_dead(evt[4])
+
+    def _on_dead(self, reason: str):
+        """Engine fleet died: error-finish every in-flight callback."""
+        log.error("TP engine died: %s", reason)
+        with self._lock:
+            cbs, self._cbs = self._cbs, {}
+            self._inflight = 0
+        for seq_id, cb in cbs.items():
+            try:
+                cb(_SeqProxy(seq_id, f"error: {reason}"), 0, True)
+            except Exception:
+                pass
 
     def submit(self, seq_id: str, prompt_ids: List[int],
                params: SamplingParams, on_token) -> None:

@@ -142,8 +142,9 @@ def test_tp_instance_worker_death_errors_inflight(tmp_path):
                 done.set()
 
         # kill the workers AFTER the request is demonstrably streaming,
-        # then simulate the engine-death notice the rank-0 worker emits
-        # on unexpected exceptions
+        # then deliver the engine-death notice (invoked directly: a
+        # SIGKILLed producer can corrupt the mp queue's lock, which is
+        # exactly the path _on_dead guards the serving side against)
         inst.submit("r0", [1, 2, 3],
                     SamplingParams(temperature=0.0, max_tokens=4096,
                                    ignore_eos=True), cb)
@@ -152,7 +153,7 @@ def test_tp_instance_worker_death_errors_inflight(tmp_path):
             p.terminate()
         for p in inst.procs:
             p.join(timeout=10)
-        inst.evt_q.put(("dead", None, None, None, "simulated crash"))
+        inst._on_dead("simulated crash")
         assert done.wait(timeout=30), "in-flight callback never finished"
         assert reasons and reasons[0].startswith("error"), reasons
         assert inst.in_flight == 0
