@@ -258,14 +258,22 @@ class TPLLMInstance:
 
     def shutdown(self):
         self.stop = True
-        try:
-            self.cmd_q.put(("stop",))
-        except Exception:
-            pass
-        for p in self.procs:
-            p.join(timeout=30)
+        # a killed worker can die holding the queue's internal lock, so
+        # only enqueue the stop command while the whole fleet is alive
+        # (otherwise terminate directly — nothing would consume it)
+        if all(p.is_alive() for p in self.procs):
+            try:
+                self.cmd_q.put(("stop",))
+            except Exception:
+                pass
+            for p in self.procs:
+                p.join(timeout=30)
         for p in self.procs:
             if p.is_alive():
                 p.terminate()
+        for p in self.procs:
+            p.join(timeout=10)
+        self.cmd_q.cancel_join_thread()
+        self.evt_q.cancel_join_thread()
         if hasattr(self, "_pump"):
             self._pump.join(timeout=5)
