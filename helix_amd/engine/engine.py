@@ -9,6 +9,7 @@ running sequences one decode token over the paged KV cache.
 from __future__ import annotations
 
 import time
+import zlib
 from dataclasses import dataclass, field
 from enum import Enum
 from typing import Callable, Dict, List, Optional
@@ -451,8 +452,11 @@ class LLMEngine:
         seeds_np = np.empty(B, dtype=np.int64)
         for i, s in enumerate(batch):
             temps_np[i] = s.params.temperature
+            # Fallback seed must be deterministic across processes: SPMD TP
+            # ranks each run this code and must draw identical Gumbel noise.
+            # hash() is per-process randomized (PYTHONHASHSEED) — use crc32.
             base = s.params.seed if s.params.seed is not None else (
-                hash(s.seq_id) & 0x7FFFFFFF)
+                zlib.crc32(s.seq_id.encode()) & 0x7FFFFFFF)
             seeds_np[i] = ((base + len(s.output_ids) * _SEED_MIX)
                            & 0x7FFFFFFFFFFFFFFF)
         temps = torch.from_numpy(temps_np).to(logits.device)

@@ -836,6 +836,12 @@ def create_app(cfg: Optional[ServerConfig] = None,
     @app.post("/api/v1/teams/{tid}/members")
     async def add_team_member(tid: str, request: Request,
                               user: AuthUser = Depends(auth_dep)):
+        team = store.get("teams", tid)
+        if not team:
+            raise HTTPException(404, "team not found")
+        if rbac.member_role(team["org_id"], user.id) not in ("owner", "admin") \
+                and not user.admin:
+            raise HTTPException(403, "org admin only")
         body = await request.json()
         return rbac.add_team_member(tid, body["user_id"])
 
@@ -1217,6 +1223,17 @@ def create_app(cfg: Optional[ServerConfig] = None,
     # ------------------------------------------------------------------
     # Projects / spec-tasks / git / code-intel (reference services)
     # ------------------------------------------------------------------
+    def _owned(table: str, rid: str, user: AuthUser) -> dict:
+        """Load an id-addressed doc and enforce owner-or-admin (the same
+        guard sessions/knowledge/triggers use; reference authz.go
+        authorizeUserToResource). 404 if missing, 403 if not yours."""
+        doc = store.get(table, rid)
+        if not doc:
+            raise HTTPException(404, "not found")
+        if doc.get("owner") != user.id and not user.admin:
+            raise HTTPException(403, "forbidden")
+        return doc
+
     @app.post("/api/v1/projects")
     async def create_project(request: Request,
                              user: AuthUser = Depends(auth_dep)):
@@ -1230,17 +1247,20 @@ def create_app(cfg: Optional[ServerConfig] = None,
     @app.post("/api/v1/projects/{pid}/tasks")
     async def create_task(pid: str, request: Request,
                           user: AuthUser = Depends(auth_dep)):
+        _owned("projects", pid, user)
         body = await request.json()
         return spec_tasks.create_task(user.id, pid, body.get("title", ""),
                                       body.get("description", ""))
 
     @app.get("/api/v1/projects/{pid}/tasks")
     async def list_tasks(pid: str, user: AuthUser = Depends(auth_dep)):
+        _owned("projects", pid, user)
         return spec_tasks.list_tasks(pid)
 
     @app.post("/api/v1/spec-tasks/{tid}/transition")
     async def transition_task(tid: str, request: Request,
                               user: AuthUser = Depends(auth_dep)):
+        _owned("spec_tasks", tid, user)
         body = await request.json()
         try:
             return spec_tasks.transition(tid, body.get("state", ""))
@@ -1249,6 +1269,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
 
     @app.post("/api/v1/spec-tasks/{tid}/plan")
     async def plan_task(tid: str, user: AuthUser = Depends(auth_dep)):
+        _owned("spec_tasks", tid, user)
         try:
             return await spec_tasks.plan(tid)
         except KeyError:
@@ -1257,6 +1278,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
     @app.post("/api/v1/spec-tasks/{tid}/implement")
     async def implement_task(tid: str,
                              user: AuthUser = Depends(auth_dep)):
+        _owned("spec_tasks", tid, user)
         try:
             return await spec_tasks.implement(tid)
         except KeyError:
@@ -1267,6 +1289,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
     @app.post("/api/v1/spec-tasks/{tid}/comments")
     async def comment_task(tid: str, request: Request,
                            user: AuthUser = Depends(auth_dep)):
+        _owned("spec_tasks", tid, user)
         body = await request.json()
         return spec_tasks.add_comment(tid, user.id, body.get("text", ""))
 
@@ -1301,6 +1324,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
 
     @app.post("/api/v1/evaluation-suites/{sid}/runs")
     async def run_suite(sid: str, user: AuthUser = Depends(auth_dep)):
+        _owned("evaluation_runs", f"suite:{sid}", user)
         try:
             return await evaluations.run_suite(sid)
         except KeyError:
@@ -1311,6 +1335,8 @@ def create_app(cfg: Optional[ServerConfig] = None,
         run = evaluations.get_run(rid)
         if run is None:
             raise HTTPException(404, "run not found")
+        if run.get("owner") != user.id and not user.admin:
+            raise HTTPException(403, "forbidden")
         return run
 
     @app.get("/api/v1/git/repos")
@@ -1320,21 +1346,25 @@ def create_app(cfg: Optional[ServerConfig] = None,
     @app.get("/api/v1/git/repos/{rid}/log")
     async def repo_log(rid: str, user: AuthUser = Depends(auth_dep),
                        ref: str = "HEAD"):
+        _owned("git_repositories", rid, user)
         return git_svc.log(rid, ref)
 
     @app.get("/api/v1/git/repos/{rid}/files")
     async def repo_files(rid: str, user: AuthUser = Depends(auth_dep),
                          ref: str = "HEAD"):
+        _owned("git_repositories", rid, user)
         return git_svc.ls_tree(rid, ref)
 
     @app.post("/api/v1/git/repos/{rid}/index")
     async def index_repo(rid: str, user: AuthUser = Depends(auth_dep)):
+        _owned("git_repositories", rid, user)
         n = await code_intel.index_repo(rid)
         return {"chunks": n}
 
     @app.post("/api/v1/git/repos/{rid}/search")
     async def search_repo(rid: str, request: Request,
                           user: AuthUser = Depends(auth_dep)):
+        _owned("git_repositories", rid, user)
         body = await request.json()
         return await code_intel.query(rid, body.get("query", ""),
                                       body.get("k", 6))

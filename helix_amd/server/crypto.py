@@ -10,6 +10,7 @@ from __future__ import annotations
 import base64
 import hashlib
 import hmac
+import logging
 import os
 import secrets as _secrets
 
@@ -43,7 +44,17 @@ def decrypt_str(blob: str, key: str) -> str:
     """Inverse of encrypt_str; raises ValueError on tamper/bad key.
     Plaintext values (pre-encryption rows) pass through unchanged."""
     if not blob.startswith("enc1:"):
-        return blob                      # legacy plaintext row
+        # Legacy plaintext row (pre-encryption). Accepting it silently
+        # would let a store-level writer strip encryption (downgrade), so
+        # log loudly; HELIX_STRICT_SECRETS=1 rejects outright (set it once
+        # all rows are migrated via re-encryption on next write).
+        if os.environ.get("HELIX_STRICT_SECRETS") == "1":
+            raise ValueError("plaintext secret row rejected "
+                             "(HELIX_STRICT_SECRETS=1)")
+        logging.getLogger("helix_amd.crypto").warning(
+            "decrypt_str: legacy plaintext secret row encountered; "
+            "it will be re-encrypted on next write")
+        return blob
     raw = base64.b64decode(blob[5:])
     nonce, ct, tag = raw[:16], raw[16:-16], raw[-16:]
     mk = _derive(key, b"mac")
