@@ -41,6 +41,11 @@ void gemm_fp8(torch::Tensor out, torch::Tensor x, torch::Tensor w,
               torch::Tensor x_scale, torch::Tensor w_scale,
               c10::optional<torch::Tensor> bias, int64_t act);
 void mfma_probe(torch::Tensor d, torch::Tensor a, torch::Tensor b);
+torch::Tensor ar_create(int64_t world, int64_t rank, int64_t capacity);
+void ar_open(std::vector<torch::Tensor> handles);
+int64_t ar_capacity();
+void ar_allreduce(torch::Tensor inp, torch::Tensor out);
+void ar_destroy();
 void mfma_probe_fp8(torch::Tensor d, torch::Tensor a, torch::Tensor b,
                     int64_t scale_a, int64_t scale_b);
 
@@ -66,6 +71,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fp8", &gemm_fp8,
         "MX-fp8 e4m3 MFMA GEMM with epilogue per-row/col dequant");
   m.def("mfma_probe", &mfma_probe, "16x16x32 MFMA layout probe");
+  m.def("ar_create", &ar_create,
+        "Allocate one-shot allreduce mailbox; returns IPC handle bytes");
+  m.def("ar_open", &ar_open, "Map peer mailboxes from IPC handles");
+  m.def("ar_capacity", &ar_capacity, "Opened AR data capacity in bytes");
+  m.def("ar_allreduce", &ar_allreduce,
+        "One-shot bf16 allreduce over xGMI peer mailboxes");
+  m.def("ar_destroy", &ar_destroy, "Tear down the allreduce context");
   m.def("mfma_probe_fp8", &mfma_probe_fp8,
         "16x16x128 MX-fp8 MFMA layout probe (unity e8m0 scales)");
 }

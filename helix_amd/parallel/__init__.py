@@ -56,10 +56,82 @@ def tp_size() -> int:
     return _TP_SIZE
 
 
+_CUSTOM_AR = False        # one-shot xGMI allreduce initialized
+_CUSTOM_AR_CAP = 0        # its data capacity (bytes)
+
+
+def init_custom_allreduce(capacity_bytes: int = 32 * 1024 * 1024) -> bool:
+    """Set up the one-shot xGMI allreduce (ops/hip/allreduce.hip) for the
+    TP group: allocate this rank's mailbox, exchange hipIpc handles over
+    the existing process group, map the peers. Decode-sized messages
+    (<= capacity) then bypass RCCL's per-link-bound ring entirely —
+    pulled point-to-point over each GPU pair's own xGMI link — and the
+    kernel is hipGraph-capturable (per-block device counters, no resets).
+
+    Returns True when active. Safe no-op on CPU / TP=1 / missing IPC.
+    """
+    global _CUSTOM_AR, _CUSTOM_AR_CAP
+    if _TP_SIZE <= 1 or not torch.cuda.is_available():
+        return False
+    if _CUSTOM_AR:
+        return True
+    from helix_amd import ops
+    if not ops.have_native():
+        return False
+    try:
+        handle = ops._native().ar_create(_TP_SIZE, _TP_RANK, capacity_bytes)
+        gathered: list = [None] * _TP_SIZE
+        dist.all_gather_object(gathered, handle.numpy().tobytes(),
+                               group=_TP_GROUP)
+        handles = [torch.frombuffer(bytearray(b), dtype=torch.uint8)
+                   for b in gathered]
+        ops._native().ar_open(handles)
+        _CUSTOM_AR, _CUSTOM_AR_CAP = True, capacity_bytes
+        return True
+    except Exception:
+        import logging
+        logging.getLogger("helix_amd.parallel").exception(
+            "custom allreduce init failed; falling back to RCCL")
+        try:
+            ops._native().ar_destroy()
+        except Exception:
+            pass
+        return False
+
+
+def destroy_custom_allreduce():
+    global _CUSTOM_AR, _CUSTOM_AR_CAP
+    if _CUSTOM_AR:
+        from helix_amd import ops
+        ops._native().ar_destroy()
+        _CUSTOM_AR, _CUSTOM_AR_CAP = False, 0
+
+
 def tp_all_reduce(x: torch.Tensor) -> torch.Tensor:
-    """Sum partial activations across the TP group (row-parallel output)."""
-    if _TP_SIZE > 1:
-        dist.all_reduce(x, group=_TP_GROUP)
+    """Sum partial activations across the TP group (row-parallel output).
+
+    Dispatch order: one-shot xGMI kernel (decode-sized bf16 messages) →
+    RCCL ring (large prefill messages) → CPU round-trip (gloo PG with a
+    CUDA tensor: tests running two ranks on one GPU, where RCCL cannot).
+    """
+    if _TP_SIZE <= 1:
+        return x
+    if _CUSTOM_AR and x.dtype == torch.bfloat16 and x.is_cuda \
+            and x.numel() * 2 <= _CUSTOM_AR_CAP:
+        from helix_amd import ops
+        xc = x if x.is_contiguous() else x.contiguous()
+        ops._native().ar_allreduce(xc, xc)
+        if xc is not x:
+            x.copy_(xc)
+        return x
+    if x.is_cuda and dist.get_backend(_TP_GROUP) == "gloo":
+        # gloo cannot reduce CUDA bf16 tensors; round-trip through host
+        # fp32 (slow, correctness path for single-GPU multi-process tests)
+        h = x.detach().float().cpu()
+        dist.all_reduce(h, group=_TP_GROUP)
+        x.copy_(h.to(x.dtype))
+        return x
+    dist.all_reduce(x, group=_TP_GROUP)
     return x
 
 

@@ -4,10 +4,18 @@ one TP-sharded model from N spawned per-GPU worker processes.
 The reference delegates multi-GPU serving to vLLM's internal executor
 (SURVEY.md §2.8 "Model containers"); here it is a native runner
 component: one process per GPU (RCCL over xGMI; gloo on CPU for tests),
-SPMD engine loop on every rank, commands fanned out by rank 0 via
-``broadcast_object_list`` so all ranks execute the identical
-add_request/cancel/step sequence, and token events flowing back to the
-serving process over a multiprocessing queue.
+SPMD engine loop on every rank.
+
+Command plane (round-2 redesign): the serving process fans every
+command out to ALL ranks over per-rank multiprocessing queues, stamped
+with a monotonically increasing sequence number. Per engine iteration
+the ranks agree on "apply commands up to watermark W" via ONE int64
+tensor broadcast (RCCL over xGMI on GPU, gloo on CPU) issued async and
+overlapped with the decode step — no pickled-object collective on the
+hot path (the round-1 design broadcast pickled command lists every
+step, a host-side sync per decode iteration). Decode all-reduces go
+through the one-shot xGMI kernel (ops/hip/allreduce.hip) when
+available, which also makes hipGraph capture safe under TP.
 
 The public surface (``submit`` / ``cancel`` / ``in_flight`` /
 ``shutdown``) matches ``runner.service.LLMInstance`` so the OpenAI
@@ -15,6 +23,7 @@ adapter and RunnerService treat TP and single-GPU instances uniformly.
 """
 from __future__ import annotations
 
+import collections
 import logging
 import os
 import queue as _queue
@@ -50,8 +59,9 @@ class _SeqProxy:
 
 def _tp_worker(rank: int, world: int, preset: str, engine_kwargs: dict,
                master_port: int, backend: str, device_type: str,
-               sd_path: Optional[str], seed: int, cmd_q, evt_q):
-    """One TP rank: SPMD engine loop driven by rank-0 command broadcasts."""
+               sd_path: Optional[str], seed: int, cmd_q, evt_q,
+               device_index: Optional[int] = None):
+    """One TP rank: SPMD engine loop, watermark-synchronized commands."""
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
                       LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
                       MASTER_PORT=str(master_port))
@@ -62,12 +72,22 @@ def _tp_worker(rank: int, world: int, preset: str, engine_kwargs: dict,
 
     try:
         if device_type == "cuda":
-            torch.cuda.set_device(rank)
-            device = f"cuda:{rank}"
+            di = rank if device_index is None else device_index
+            torch.cuda.set_device(di)
+            device = f"cuda:{di}"
         else:
             device = "cpu"
         parallel.init_tp(world, backend=backend)
         cfg = PRESETS[preset]
+        # One-shot xGMI allreduce sized for the decode messages
+        # (B x hidden bf16); prefill messages exceed it and fall back to
+        # RCCL rings automatically (both correct, SURVEY §2.6 plan).
+        custom_ar = False
+        if device_type == "cuda" and world > 1:
+            cap = max(8 << 20,
+                      engine_kwargs.get("max_num_seqs", 64) *
+                      cfg.hidden_size * 2)
+            custom_ar = parallel.init_custom_allreduce(cap)
         model = None
         if sd_path is not None:
             # shard a known full checkpoint (tests / real weights)
@@ -85,33 +105,29 @@ def _tp_worker(rank: int, world: int, preset: str, engine_kwargs: dict,
                 strict=True)
             model = model.to(device)
         ecfg = EngineConfig(model=preset, seed=seed, **engine_kwargs)
+        # hipGraph capture under TP is safe only when decode all-reduces
+        # run on the capturable one-shot kernel (or TP==1, no collective).
+        if world > 1 and not custom_ar:
+            ecfg.enforce_eager = True
         eng = LLMEngine(ecfg, device=device, model=model,
                         tp_size=world, tp_rank=rank)
 
         def cb(seq, tok, fin):
             evt_q.put(("tok", seq.seq_id, tok, fin, seq.finish_reason))
 
-        if rank == 0:
-            evt_q.put(("ready", None, None, None, None))
-        while True:
-            if rank == 0:
-                cmds = []
-                timeout = 0.0 if eng.has_work else 0.05
-                try:
-                    cmds.append(cmd_q.get(timeout=timeout))
-                except _queue.Empty:
-                    pass
-                while True:
-                    try:
-                        cmds.append(cmd_q.get_nowait())
-                    except _queue.Empty:
-                        break
-                obj = [cmds]
-            else:
-                obj = [None]
-            dist.broadcast_object_list(obj, src=0)
-            stop = False
-            for c in obj[0]:
+        # -- watermark plumbing -------------------------------------------
+        wm_device = device if backend == "nccl" else "cpu"
+        wm = torch.zeros(1, dtype=torch.int64, device=wm_device)
+        applied = 0
+        pending = collections.deque()   # (seqno, cmd) popped ahead of apply
+        stop = False
+
+        def apply_cmd(sn: int, c: tuple):
+            """Apply one command; exceptions finish just that request
+            (all ranks take the same branch — commands and engine state
+            are identical) instead of killing the fleet."""
+            nonlocal stop
+            try:
                 if c[0] == "stop":
                     stop = True
                 elif c[0] == "submit":
@@ -123,15 +139,47 @@ def _tp_worker(rank: int, world: int, preset: str, engine_kwargs: dict,
                         eng.seqs[c[1]].finish_reason is None
                     eng.cancel(c[1])
                     if rank == 0 and existed:
-                        # the engine does not emit an event on cancel
-                        # (single-process adapters return immediately);
-                        # across processes the serving side needs closure
-                        # to release its callback + inflight slot
+                        # engine emits no event on cancel; serving side
+                        # needs closure to release its inflight slot
                         evt_q.put(("tok", c[1], 0, True, "cancelled"))
-            if stop:
-                break
+            except Exception as e:
+                log.exception("command %r failed on rank %d", c[0], rank)
+                if rank == 0 and c[0] == "submit":
+                    evt_q.put(("tok", c[1], 0, True, f"error: {e}"))
+
+        dist.barrier()
+        if rank == 0:
+            evt_q.put(("ready", None, None, None, None))
+        while not stop:
+            # 1. rank 0 picks this round's watermark from its own queue
+            if rank == 0:
+                if not eng.has_work and not pending:
+                    try:
+                        pending.append(cmd_q.get(timeout=0.05))
+                    except _queue.Empty:
+                        pass
+                while True:
+                    try:
+                        pending.append(cmd_q.get_nowait())
+                    except _queue.Empty:
+                        break
+                wm.fill_(pending[-1][0] if pending else applied)
+            # 2. broadcast the watermark async; overlap with the step
+            work = dist.broadcast(wm, src=0, async_op=True)
             if eng.has_work:
                 eng.step()
+            work.wait()
+            target = int(wm.item())
+            # 3. apply commands (applied, target] in seqno order — they
+            # are guaranteed to arrive on this rank's own queue
+            while applied < target:
+                if pending:
+                    sn, c = pending.popleft()
+                else:
+                    sn, c = cmd_q.get(timeout=300.0)
+                apply_cmd(sn, c)
+                applied = sn
+        parallel.destroy_custom_allreduce()
         dist.destroy_process_group()
     except Exception as e:  # surface worker death to the serving process
         log.exception("tp worker %d died", rank)
@@ -149,7 +197,8 @@ class TPLLMInstance:
 
     def __init__(self, spec, tp_size: int, device_type: Optional[str] = None,
                  backend: Optional[str] = None, sd_path: Optional[str] = None,
-                 seed: int = 0, start_timeout: float = 600.0):
+                 seed: int = 0, start_timeout: float = 600.0,
+                 device_indices: Optional[List[int]] = None):
         self.spec = spec
         self.tp_size = tp_size
         if device_type is None:
@@ -157,8 +206,12 @@ class TPLLMInstance:
         if backend is None:
             backend = "nccl" if device_type == "cuda" else "gloo"
         ctx = torch.multiprocessing.get_context("spawn")
-        self.cmd_q = ctx.Queue()
+        # one command queue per rank: commands fan out host-side, so the
+        # GPU-side per-step collective is a single int64 watermark
+        self.cmd_qs = [ctx.Queue() for _ in range(tp_size)]
         self.evt_q = ctx.Queue()
+        self._seqno = 0
+        self._send_lock = threading.Lock()
         engine_kwargs = dict(
             max_model_len=spec.max_model_len,
             max_num_seqs=spec.max_num_seqs,
@@ -173,13 +226,14 @@ class TPLLMInstance:
             ctx.Process(target=_tp_worker,
                         args=(r, tp_size, spec.preset, engine_kwargs, port,
                               backend, device_type, sd_path, seed,
-                              self.cmd_q, self.evt_q),
+                              self.cmd_qs[r], self.evt_q,
+                              device_indices[r] if device_indices else None),
                         daemon=True)
             for r in range(tp_size)
         ]
         for p in self.procs:
             p.start()
-        # wait for rank0's engine to come up (weights + graphs)
+        # wait for the fleet's engines to come up (weights + graphs)
         deadline = time.time() + start_timeout
         while True:
             try:
@@ -205,10 +259,38 @@ class TPLLMInstance:
         self._pump = threading.Thread(target=self._pump_events, daemon=True,
                                       name=f"tp-pump-{spec.name}")
         self._pump.start()
+        # liveness monitor: a dead rank would hang the others' watermark
+        # broadcast (fleet-wide failure detection)
+        self._monitor = threading.Thread(target=self._monitor_procs,
+                                         daemon=True,
+                                         name=f"tp-mon-{spec.name}")
+        self._monitor.start()
 
     @property
     def in_flight(self) -> int:
         return self._inflight
+
+    def _send(self, cmd: tuple):
+        """Stamp a command and fan it out to every rank's queue."""
+        with self._send_lock:
+            self._seqno += 1
+            sn = self._seqno
+            for q in self.cmd_qs:
+                q.put((sn, cmd))
+
+    def _monitor_procs(self):
+        while not self.stop:
+            time.sleep(0.5)
+            if self.stop:
+                return
+            dead = [p for p in self.procs
+                    if not p.is_alive() and p.exitcode not in (None, 0)]
+            if dead:
+                self._on_dead(f"rank exited with code {dead[0].exitcode}")
+                for p in self.procs:
+                    if p.is_alive():
+                        p.terminate()
+                return
 
     def _pump_events(self):
         while not self.stop:
@@ -251,19 +333,19 @@ class TPLLMInstance:
             if on_token is not None:
                 self._cbs[seq_id] = on_token
             self._inflight += 1
-        self.cmd_q.put(("submit", seq_id, list(prompt_ids), params))
+        self._send(("submit", seq_id, list(prompt_ids), params))
 
     def cancel(self, seq_id: str):
-        self.cmd_q.put(("cancel", seq_id))
+        self._send(("cancel", seq_id))
 
     def shutdown(self):
         self.stop = True
-        # a killed worker can die holding the queue's internal lock, so
+        # a killed worker can die holding a queue's internal lock, so
         # only enqueue the stop command while the whole fleet is alive
         # (otherwise terminate directly — nothing would consume it)
         if all(p.is_alive() for p in self.procs):
             try:
-                self.cmd_q.put(("stop",))
+                self._send(("stop",))
             except Exception:
                 pass
             for p in self.procs:
@@ -273,7 +355,8 @@ class TPLLMInstance:
                 p.terminate()
         for p in self.procs:
             p.join(timeout=10)
-        self.cmd_q.cancel_join_thread()
+        for q in self.cmd_qs:
+            q.cancel_join_thread()
         self.evt_q.cancel_join_thread()
         if hasattr(self, "_pump"):
             self._pump.join(timeout=5)

@@ -182,3 +182,44 @@ def test_tp_instance_forwards_serving_options():
     from helix_amd.models.quant import FP8Linear
     assert isinstance(eng.model.layers[0].attn.qkv_proj, FP8Linear)
     assert eng.kv.caches[0][0].dtype == torch.uint8
+
+
+@pytest.mark.timeout(300)
+def test_tp_lockstep_stochastic_no_explicit_seed(tmp_path):
+    """temperature>0 WITHOUT an explicit seed must stay in lockstep
+    across SPMD ranks: the engine's fallback seed is derived from the
+    seq_id (crc32), not per-process hash() randomization. Equivalence
+    vs a single-process engine proves both ranks sampled identically."""
+    cfg = PRESETS["tiny-gqa"]
+    torch.manual_seed(0)
+    full = LlamaForCausalLM(cfg).float()
+    full.init_random(0)
+    sd_path = str(tmp_path / "full_sd.pt")
+    torch.save(full.state_dict(), sd_path)
+
+    prompts = [[1, 2, 3, 4, 5], [9, 8, 7, 6]]
+    sp = SamplingParams(temperature=0.8, max_tokens=8, ignore_eos=True)
+    assert sp.seed is None
+    eng = LLMEngine(EngineConfig(model="tiny-gqa", max_model_len=256,
+                                 max_num_seqs=4, kv_cache_blocks=128,
+                                 eos_token_id=-1),
+                    device="cpu", model=full)
+    # same seq ids as the TP run below: the crc32(seq_id) fallback seed
+    # must make them sample identically
+    eng.add_request("req-0", prompts[0], sp)
+    eng.add_request("req-1", prompts[1], sp)
+    while eng.has_work:
+        eng.step()
+    want = [eng.seqs["req-0"].output_ids, eng.seqs["req-1"].output_ids]
+
+    spec = ModelSpec("tiny-gqa", "llm", "tiny-gqa", max_model_len=256,
+                     max_num_seqs=4, kv_cache_blocks=128, tp=2)
+    inst = _mk_inst(spec, device_type="cpu", backend="gloo",
+                    sd_path=sd_path, start_timeout=120)
+    try:
+        got0 = _collect_stream(inst, "req-0", prompts[0], sp)
+        got1 = _collect_stream(inst, "req-1", prompts[1], sp)
+        assert got0 == want[0]
+        assert got1 == want[1]
+    finally:
+        inst.shutdown()
