@@ -13,6 +13,8 @@ import socket
 import sys
 import time
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 
@@ -28,6 +30,8 @@ SIZES = [2048, 8192, 65536, 262144, 1048576, 2097152, 4194304]  # elements
 
 
 def worker(rank, world, port, q):
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
     os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
                       MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
     import torch.distributed as dist
@@ -64,12 +68,20 @@ def worker(rank, world, port, q):
         q.put(rows)
 
 
+def worker_safe(rank, world, port, q):
+    try:
+        worker(rank, world, port, q)
+    except Exception as e:
+        q.put(f"rank {rank} error: {e}")
+        raise
+
+
 def main():
     ctx = torch.multiprocessing.get_context("spawn")
     q = ctx.Queue()
     port = _free_port()
-    procs = [ctx.Process(target=worker, args=(r, 2, port, q), daemon=True)
-             for r in range(2)]
+    procs = [ctx.Process(target=worker_safe, args=(r, 2, port, q),
+                         daemon=True) for r in range(2)]
     for p in procs:
         p.start()
     rows = q.get(timeout=600)
