@@ -37,7 +37,7 @@ def run_coresident(args) -> None:
     serve mixed traffic). Reported value = total LLM tokens/s."""
     from helix_amd.engine.engine import EngineConfig, LLMEngine
     from helix_amd.engine.sampling_params import SamplingParams
-    from helix_amd.runner.service import MODEL_SPECS, EmbeddingInstance
+    from helix_amd.runner.service import DEFAULT_SPECS, EmbeddingInstance
 
     device = "cuda:0"
     b1, b2 = args.batch, max(32, args.batch // 4)
@@ -52,7 +52,7 @@ def run_coresident(args) -> None:
                            kv_cache_blocks=blocks(bsz),
                            kv_cache_dtype=args.kv_dtype)
         engines.append(LLMEngine(cfg, device=device))
-    bge = EmbeddingInstance(MODEL_SPECS["bge-base"], device)
+    bge = EmbeddingInstance(DEFAULT_SPECS["bge-base"], device)
     emb_texts = [f"document {i} about GPUs, HBM and xGMI topology" * 4
                  for i in range(16)]
     # Pre-tokenize once; the timed loop issues the raw encoder forward on
