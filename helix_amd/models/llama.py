@@ -111,6 +111,15 @@ class DecodeMeta:
     is_prefill: bool = False
 
 
+class HLinear(nn.Linear):
+    """nn.Linear routed through ops.linear: decode-shaped (skinny-M)
+    GEMMs run the owned split-K MFMA kernel, prefill-shaped ones stay on
+    hipBLASLt (SURVEY §2.8 'Q/K/V + O projections, MLP GEMMs')."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.linear(x, self.weight, self.bias)
+
+
 class Attention(nn.Module):
     def __init__(self, cfg: LlamaConfig, tp_size: int = 1, tp_rank: int = 0):
         super().__init__()
@@ -123,8 +132,8 @@ class Attention(nn.Module):
         self.tp_size = tp_size
         self.window = cfg.sliding_window or 0
         q, kv, h = self.nh * self.hd, self.nkv * self.hd, cfg.hidden_size
-        self.qkv_proj = nn.Linear(h, q + 2 * kv, bias=cfg.attention_bias)
-        self.o_proj = nn.Linear(q, h, bias=False)
+        self.qkv_proj = HLinear(h, q + 2 * kv, bias=cfg.attention_bias)
+        self.o_proj = HLinear(q, h, bias=False)
 
     def forward(self, x: torch.Tensor, cos_sin: torch.Tensor, kv_cache,
                 meta) -> torch.Tensor:
@@ -177,8 +186,8 @@ class MLP(nn.Module):
         assert cfg.intermediate_size % tp_size == 0
         self.tp_size = tp_size
         i = cfg.intermediate_size // tp_size
-        self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * i, bias=False)
-        self.down_proj = nn.Linear(i, cfg.hidden_size, bias=False)
+        self.gate_up_proj = HLinear(cfg.hidden_size, 2 * i, bias=False)
+        self.down_proj = HLinear(i, cfg.hidden_size, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         out = self.down_proj(ops.silu_and_mul(self.gate_up_proj(x)))
@@ -224,8 +233,8 @@ class LlamaForCausalLM(nn.Module):
         if cfg.tie_embeddings:
             self.lm_head = None
         else:
-            self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,
-                                     bias=False)
+            self.lm_head = HLinear(cfg.hidden_size, cfg.vocab_size,
+                                   bias=False)
         cs = make_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_base)
         self.register_buffer("cos_sin", cs, persistent=False)
 

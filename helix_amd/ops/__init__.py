@@ -7,6 +7,8 @@ logic is testable without a GPU.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 
 from . import reference as ref
@@ -265,6 +267,57 @@ def gemm_bf16(x: torch.Tensor, w: torch.Tensor, bias=None,
         _native().gemm_bf16(out, x, w, bias, act)
         return out
     return ref.gemm_bf16(x, w, bias, act)
+
+
+_SKINNY_MAX_M = 1024
+_SKINNY_SCRATCH: dict = {}        # device -> fp32 workspace tensor
+_SKINNY_SWIZZLE = os.environ.get("HELIX_SKINNY_SWIZZLE", "1") == "1"
+_SKINNY_ENABLED = os.environ.get("HELIX_SKINNY_GEMM", "1") == "1"
+
+
+def _skinny_split(M: int, N: int, K: int) -> int:
+    """Split-K factor: enough workgroups to put ~3 on each of the 256
+    CUs (m97-structure occupancy), subject to K dividing into 64s."""
+    bm = 128 if M > 64 else 64
+    tiles = ((M + bm - 1) // bm) * ((N + 127) // 128)
+    s = 1
+    while s < 8 and tiles * s < 768 and K % (s * 2 * 64) == 0:
+        s *= 2
+    return s
+
+
+def gemm_skinny_bf16(x: torch.Tensor, w: torch.Tensor, bias=None,
+                     act: int = 0) -> torch.Tensor:
+    """Decode-shape GEMM (M <= 1024): split-K skinny tiles, fused
+    bias/act epilogue. Caller guarantees bf16 CUDA contiguous inputs."""
+    M, K = x.shape
+    N = w.shape[0]
+    out = torch.empty(M, N, dtype=x.dtype, device=x.device)
+    s = _skinny_split(M, N, K)
+    scratch = None
+    if s > 1:
+        need = s * M * N
+        key = x.device.index
+        cur = _SKINNY_SCRATCH.get(key)
+        if cur is None or cur.numel() < need:
+            cur = torch.empty(need, dtype=torch.float32, device=x.device)
+            _SKINNY_SCRATCH[key] = cur
+        scratch = cur
+    _native().gemm_skinny_bf16(out, x, w, bias, scratch, s, act,
+                               1 if _SKINNY_SWIZZLE else 0)
+    return out
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, bias=None) -> torch.Tensor:
+    """Projection dispatch: skinny-M decode GEMMs on the owned kernel,
+    everything else (prefill-sized M, CPU, odd dims) via torch/hipBLASLt."""
+    if (_SKINNY_ENABLED and x.is_cuda and x.dtype == torch.bfloat16
+            and x.dim() == 2
+            and w.dtype == torch.bfloat16 and 0 < x.shape[0] <= _SKINNY_MAX_M
+            and x.shape[1] % 64 == 0 and w.shape[0] % 16 == 0
+            and x.is_contiguous() and w.is_contiguous()):
+        return gemm_skinny_bf16(x, w, bias)
+    return torch.nn.functional.linear(x, w, bias)
 
 
 def mfma_probe(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:

@@ -37,6 +37,10 @@ void sample_tokens(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperatures, torch::Tensor seeds);
 void gemm_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias, int64_t act);
+void gemm_skinny_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+                      c10::optional<torch::Tensor> bias,
+                      c10::optional<torch::Tensor> scratch, int64_t split_k,
+                      int64_t act, int64_t swizzle);
 void gemm_fp8(torch::Tensor out, torch::Tensor x, torch::Tensor w,
               torch::Tensor x_scale, torch::Tensor w_scale,
               c10::optional<torch::Tensor> bias, int64_t act);
@@ -68,6 +72,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Varlen causal flash prefill attention (MFMA)");
   m.def("sample_tokens", &sample_tokens, "Greedy/Gumbel token sampling");
   m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM: x @ w^T (+bias, act)");
+  m.def("gemm_skinny_bf16", &gemm_skinny_bf16,
+        "Skinny-M decode GEMM: split-K + XCD-chunked tile swizzle");
   m.def("gemm_fp8", &gemm_fp8,
         "MX-fp8 e4m3 MFMA GEMM with epilogue per-row/col dequant");
   m.def("mfma_probe", &mfma_probe, "16x16x32 MFMA layout probe");

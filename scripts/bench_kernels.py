@@ -69,6 +69,26 @@ def bench_prefill():
               f"{flops/t/1e12:7.1f} TF (dense peak 2500)")
 
 
+
+def bench_skinny():
+    print("== gemm_skinny_bf16 (split-K, XCD swizzle) vs hipBLASLt ==")
+    import helix_amd.ops as _o
+    for (M, N, K) in [(512, 6144, 4096), (512, 4096, 4096),
+                      (512, 28672, 4096), (512, 4096, 14336),
+                      (512, 128256, 4096),
+                      (256, 4096, 4096), (64, 6144, 4096),
+                      (128, 28672, 4096)]:
+        x = torch.randn(M, K, dtype=torch.bfloat16, device=DEV) / 8
+        w = torch.randn(N, K, dtype=torch.bfloat16, device=DEV) / 8
+        s_k = _o._skinny_split(M, N, K)
+        t1 = timeit(lambda: ops.gemm_skinny_bf16(x, w), iters=40)
+        t2 = timeit(lambda: torch.nn.functional.linear(x, w), iters=40)
+        fl = 2 * M * N * K
+        print(f"M{M:5d} N{N:6d} K{K:5d} splitK={s_k}: "
+              f"ours {t1*1e6:8.1f} us ({fl/t1/1e12:7.1f} TF) | "
+              f"hipblaslt {t2*1e6:8.1f} us ({fl/t2/1e12:7.1f} TF)")
+
+
 def bench_gemm():
     print("== gemm_bf16 (ours) vs torch/hipBLASLt ==")
     for (M, N, K) in [(512, 3072, 768), (8192, 768, 768), (4096, 4096, 4096),
@@ -136,7 +156,7 @@ if __name__ == "__main__":
     which = sys.argv[1] if len(sys.argv) > 1 else "all"
     torch.manual_seed(0)
     fns = {"decode": bench_decode, "prefill": bench_prefill,
-           "gemm": bench_gemm, "fp8": bench_fp8, "norm": bench_norm,
+           "gemm": bench_gemm, "skinny": bench_skinny, "fp8": bench_fp8, "norm": bench_norm,
            "sample": bench_sample}
     if which == "all":
         for f in fns.values():

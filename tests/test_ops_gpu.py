@@ -451,3 +451,37 @@ def test_fp8_kv_fused_rope_write_gpu():
     bound = 0.135 * torch.maximum(ka.abs(), kb.abs()) + 0.02
     bad = ((ka - kb).abs() > bound).sum().item()
     assert bad == 0, f"{bad} K elements off by more than one e4m3 ulp"
+
+
+@pytest.mark.parametrize("m,n,k", [
+    (1, 128, 64),                 # tiny decode batch, BM=64 path
+    (64, 4096, 4096),             # o_proj shape, small batch
+    (512, 4096, 4096),            # o_proj @ B=512 (split-K path)
+    (512, 6144, 4096),            # qkv_proj
+    (512, 4096, 14336),           # down_proj (deep split-K)
+    (200, 1024, 512),             # odd M tail
+])
+def test_gemm_skinny_bf16(m, n, k):
+    torch.manual_seed(21)
+    x = torch.randn(m, k, dtype=torch.bfloat16, device=DEV) / (k ** 0.25)
+    w = torch.randn(n, k, dtype=torch.bfloat16, device=DEV) / (k ** 0.25)
+    bias = torch.randn(n, dtype=torch.bfloat16, device=DEV)
+    got = ops.gemm_skinny_bf16(x, w, bias)
+    want = ref.gemm_bf16(x.cpu(), w.cpu(), bias.cpu())
+    assert_close_bf16(got, want, atol=5e-2, rtol=5e-2)
+    # no-bias path
+    got2 = ops.gemm_skinny_bf16(x, w, None)
+    want2 = ref.gemm_bf16(x.cpu(), w.cpu(), None)
+    assert_close_bf16(got2, want2, atol=5e-2, rtol=5e-2)
+
+
+def test_linear_dispatch_matches_torch():
+    """ops.linear (the HLinear route) must agree with F.linear on the
+    decode shapes it owns."""
+    torch.manual_seed(22)
+    for m, n, k in [(17, 512, 256), (512, 1024, 512)]:
+        x = torch.randn(m, k, dtype=torch.bfloat16, device=DEV) / (k ** .25)
+        w = torch.randn(n, k, dtype=torch.bfloat16, device=DEV) / (k ** .25)
+        got = ops.linear(x, w, None)
+        want = torch.nn.functional.linear(x.cpu().float(), w.cpu().float())
+        assert_close_bf16(got, want.to(torch.bfloat16), atol=5e-2, rtol=5e-2)
