@@ -272,7 +272,7 @@ def gemm_bf16(x: torch.Tensor, w: torch.Tensor, bias=None,
 _SKINNY_MAX_M = 1024
 _SKINNY_SCRATCH: dict = {}        # device -> fp32 workspace tensor
 _SKINNY_SWIZZLE = os.environ.get("HELIX_SKINNY_SWIZZLE", "1") == "1"
-_SKINNY_ENABLED = os.environ.get("HELIX_SKINNY_GEMM", "1") == "1"
+_SKINNY_ENABLED = os.environ.get("HELIX_SKINNY_GEMM", "0") == "1"
 
 
 def _skinny_split(M: int, N: int, K: int) -> int:
