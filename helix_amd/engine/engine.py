@@ -48,6 +48,7 @@ class Sequence:
     first_token_time: Optional[float] = None
     finish_reason: Optional[str] = None
     logprobs: List[dict] = field(default_factory=list)  # when requested
+    pen_init: bool = False       # GPU penalty tables primed for this row
     # streaming callback: fn(seq, new_token_id, finished)
     on_token: Optional[Callable] = None
 
@@ -113,6 +114,10 @@ class LLMEngine:
                 self.model_cfg.head_dim, cfg.max_model_len, self.device)
         else:
             self.decode_ws = None
+        # GPU penalty tables (lazy): [max_num_seqs, V] output counts +
+        # prompt-seen bitmap, read by the fused sampling kernel
+        self._pen_counts: Optional[torch.Tensor] = None
+        self._pen_seen: Optional[torch.Tensor] = None
         # Persistent host-side batch state: per-seq block-table rows and
         # scratch arrays. Rebuilding these as Python-list -> torch.tensor
         # per step cost ~4 ms at B=512 (512 small tensor constructions);
@@ -291,6 +296,7 @@ class LLMEngine:
         if seq.row >= 0:
             self._free_rows.append(seq.row)
             seq.row = -1
+        seq.pen_init = False
 
     def _run_prefill(self, batch: List[Sequence]):
         bs = self.cfg.block_size
@@ -441,13 +447,50 @@ class LLMEngine:
         return list(zip(batch, tokens))
 
     # ------------------------------------------------------------------
+    def _ensure_pen_row(self, seq: Sequence):
+        """Lazy per-row penalty tables on GPU: seen-prompt bitmap +
+        output-token counts, maintained incrementally so the sampling
+        kernel applies penalties without any host-side set()/Counter."""
+        if self._pen_counts is None:
+            V = self.model_cfg.vocab_size
+            R = self.cfg.max_num_seqs
+            self._pen_counts = torch.zeros(R, V, dtype=torch.int32,
+                                           device=self.device)
+            self._pen_seen = torch.zeros(R, V, dtype=torch.uint8,
+                                         device=self.device)
+        if not seq.pen_init and seq.row >= 0:
+            row = seq.row
+            self._pen_counts[row].zero_()
+            self._pen_seen[row].zero_()
+            pid = torch.tensor(seq.prompt_ids, dtype=torch.int64,
+                               device=self.device)
+            self._pen_seen[row][pid] = 1
+            if seq.output_ids:   # preemption recompute: rebuild counts
+                oid = torch.tensor(seq.output_ids, dtype=torch.int64,
+                                   device=self.device)
+                self._pen_counts[row].index_put_(
+                    (oid,), torch.ones(len(seq.output_ids),
+                                       dtype=torch.int32,
+                                       device=self.device),
+                    accumulate=True)
+            seq.pen_init = True
+
     def _sample(self, batch: List[Sequence], logits: torch.Tensor) -> List[int]:
         # wait: decode path's new input token is appended by _append_token;
         # here logits are [B, V].
-        needs_proc = any(s.params.needs_logit_processing for s in batch)
-        if needs_proc:
-            logits = self._process_logits(batch, logits.float())
         B = len(batch)
+        lp_rows = [i for i, s_ in enumerate(batch) if s_.params.logprobs]
+        needs_proc = any(s.params.needs_logit_processing for s in batch)
+        gpu_fast = logits.is_cuda and needs_proc
+        if needs_proc and not gpu_fast:
+            logits = self._process_logits(batch, logits.float())
+        elif gpu_fast and lp_rows:
+            # logprob rows report over the FILTERED distribution, so they
+            # keep the reference host path; their kernel params are
+            # neutralized below (no double filtering/penalties)
+            sub = logits[lp_rows].float()
+            self._process_logits([batch[i] for i in lp_rows], sub)
+            logits[lp_rows] = sub.to(logits.dtype)
         temps_np = np.empty(B, dtype=np.float32)
         seeds_np = np.empty(B, dtype=np.int64)
         for i, s in enumerate(batch):
@@ -461,7 +504,50 @@ class LLMEngine:
                            & 0x7FFFFFFFFFFFFFFF)
         temps = torch.from_numpy(temps_np).to(logits.device)
         seeds_t = torch.from_numpy(seeds_np).to(logits.device)
-        toks = ops.sample_tokens(logits.contiguous(), temps, seeds_t)
+        if gpu_fast:
+            lp_set = set(lp_rows)
+            topp_np = np.ones(B, dtype=np.float32)
+            topk_np = np.zeros(B, dtype=np.int32)
+            rep_np = np.ones(B, dtype=np.float32)
+            pres_np = np.zeros(B, dtype=np.float32)
+            freq_np = np.zeros(B, dtype=np.float32)
+            map_np = np.full(B, -1, dtype=np.int32)
+            pen_idx: List[int] = []
+            for i, s in enumerate(batch):
+                if i in lp_set:
+                    continue            # host-processed already
+                p = s.params
+                topp_np[i] = p.top_p
+                topk_np[i] = p.top_k
+                if (p.repetition_penalty != 1.0 or p.presence_penalty != 0.0
+                        or p.frequency_penalty != 0.0):
+                    rep_np[i] = p.repetition_penalty
+                    pres_np[i] = p.presence_penalty
+                    freq_np[i] = p.frequency_penalty
+                    self._ensure_pen_row(s)
+                    map_np[i] = s.row
+                    pen_idx.append(i)
+            dev = logits.device
+            toks = ops.sample_tokens_ext(
+                logits.contiguous(), temps, seeds_t,
+                torch.from_numpy(topp_np).to(dev),
+                torch.from_numpy(topk_np).to(dev),
+                torch.from_numpy(rep_np).to(dev),
+                torch.from_numpy(pres_np).to(dev),
+                torch.from_numpy(freq_np).to(dev),
+                self._pen_counts, self._pen_seen,
+                torch.from_numpy(map_np).to(dev))
+            if pen_idx:
+                # incremental count update with the just-sampled tokens
+                rows_t = torch.tensor([batch[i].row for i in pen_idx],
+                                      dtype=torch.int64, device=dev)
+                sel = toks[pen_idx]
+                self._pen_counts.index_put_(
+                    (rows_t, sel), torch.ones(len(pen_idx),
+                                              dtype=torch.int32, device=dev),
+                    accumulate=True)
+        else:
+            toks = ops.sample_tokens(logits.contiguous(), temps, seeds_t)
         out = toks.cpu().tolist()
         # top-k logprobs for sequences that requested them (one extra
         # log_softmax + topk over just those rows)

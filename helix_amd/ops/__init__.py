@@ -228,6 +228,22 @@ def sample_tokens(logits: torch.Tensor, temperatures: torch.Tensor,
     return ref.sample_tokens(logits, temperatures, seeds)
 
 
+def sample_tokens_ext(logits: torch.Tensor, temperatures: torch.Tensor,
+                      seeds: torch.Tensor, top_p: torch.Tensor,
+                      top_k: torch.Tensor, rep_pen: torch.Tensor,
+                      pres_pen: torch.Tensor, freq_pen: torch.Tensor,
+                      counts=None, seen=None, row_map=None) -> torch.Tensor:
+    """Fused top-k/top-p/penalty Gumbel sampling (GPU only): one block
+    per row, histogram-located threshold, no vocab sort. Penalties read
+    per-row token-count tables indexed by row_map (engine row slots)."""
+    out = torch.empty(logits.shape[0], dtype=torch.int64,
+                      device=logits.device)
+    _native().sample_tokens_ext(out, logits, temperatures, seeds, top_p,
+                                top_k, rep_pen, pres_pen, freq_pen,
+                                counts, seen, row_map)
+    return out
+
+
 def quantize_fp8(t: torch.Tensor, dim: int = -1):
     """Per-row (dim=-1 reduces over the last axis) e4m3 quantization.
     Returns (bytes uint8 view, float32 scales)."""

@@ -35,6 +35,14 @@ void attn_prefill(torch::Tensor out, torch::Tensor q, torch::Tensor k,
                   double scale, bool causal, int64_t window);
 void sample_tokens(torch::Tensor out, torch::Tensor logits,
                    torch::Tensor temperatures, torch::Tensor seeds);
+void sample_tokens_ext(torch::Tensor out, torch::Tensor logits,
+                       torch::Tensor temperatures, torch::Tensor seeds,
+                       torch::Tensor top_p, torch::Tensor top_k,
+                       torch::Tensor rep_pen, torch::Tensor pres_pen,
+                       torch::Tensor freq_pen,
+                       c10::optional<torch::Tensor> counts,
+                       c10::optional<torch::Tensor> seen,
+                       c10::optional<torch::Tensor> row_map);
 void gemm_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,
                c10::optional<torch::Tensor> bias, int64_t act);
 void gemm_skinny_bf16(torch::Tensor out, torch::Tensor x, torch::Tensor w,
@@ -71,6 +79,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill", &attn_prefill,
         "Varlen causal flash prefill attention (MFMA)");
   m.def("sample_tokens", &sample_tokens, "Greedy/Gumbel token sampling");
+  m.def("sample_tokens_ext", &sample_tokens_ext,
+        "Fused top-k/top-p/penalty Gumbel sampling (histogram threshold)");
   m.def("gemm_bf16", &gemm_bf16, "MFMA bf16 GEMM: x @ w^T (+bias, act)");
   m.def("gemm_skinny_bf16", &gemm_skinny_bf16,
         "Skinny-M decode GEMM: split-K + XCD-chunked tile swizzle");

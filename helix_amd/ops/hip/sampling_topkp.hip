@@ -1,0 +1,266 @@
+// Fused top-k / top-p / penalty sampling (CDNA4, no vocab sort).
+//
+// SURVEY.md §2.8 "vocab-GEMM + fused softmax/top-p sampling kernel":
+// request fields temperature/top_p/top_k/penalties flow from
+// AssistantConfig (reference types.go:1636-1661). Round-1 did these in
+// per-sequence host Python (engine._process_logits) — full-vocab sorts
+// per row per step; this kernel replaces that with one block per row:
+//
+//   pass A: online (max, sumexp) over the penalty-adjusted logits
+//   pass B: LDS histogram of exp-mass + counts binned by logit value;
+//           prefix-scan locates the top-p / top-k threshold bin
+//   pass C: Gumbel-argmax restricted to values above the threshold
+//
+// Threshold granularity is one histogram bin (RANGE/BINS = 16/2048 =
+// 0.0078 logit units): the kept set equals torch's exact sorted-cumsum
+// set except that the *boundary* bin is kept whole — a superset no
+// larger than one bin's worth of mass. Penalties (repetition, presence,
+// frequency) are applied inline from per-row token-count tables the
+// engine maintains incrementally on the GPU (no host set() per step).
+#include "common.h"
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+using namespace helix;
+
+namespace {
+
+constexpr int BINS = 2048;
+constexpr float RANGE = 16.0f;   // logit span below max covered by bins
+
+template <typename T>
+__device__ __forceinline__ float raw_val(const T* p, int i);
+template <>
+__device__ __forceinline__ float raw_val<uint16_t>(const uint16_t* p, int i) {
+  return bf16_to_f32(p[i]);
+}
+template <>
+__device__ __forceinline__ float raw_val<float>(const float* p, int i) {
+  return p[i];
+}
+
+struct PenaltyCtx {
+  const int* cnt;        // output-token counts for this row (or null)
+  const uint8_t* seen;   // prompt-token bitmap for this row (or null)
+  float rep, pres, freq;
+  bool active;
+};
+
+__device__ __forceinline__ float adjust(float v, int i, const PenaltyCtx& p) {
+  if (!p.active) return v;
+  const int c = p.cnt[i];
+  const bool s = p.seen[i] || c > 0;
+  if (s && p.rep != 1.f) v = (v > 0.f) ? v / p.rep : v * p.rep;
+  if (s) v -= p.pres;
+  v -= p.freq * (float)c;
+  return v;
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void sample_topkp_kernel(
+    int64_t* __restrict__ out, const T* __restrict__ logits,
+    const float* __restrict__ temperatures,
+    const uint64_t* __restrict__ seeds, const float* __restrict__ top_p,
+    const int* __restrict__ top_k, const float* __restrict__ rep_pen,
+    const float* __restrict__ pres_pen, const float* __restrict__ freq_pen,
+    const int* __restrict__ counts, const uint8_t* __restrict__ seen,
+    const int* __restrict__ row_map, int V) {
+  const int row = blockIdx.x;
+  const T* lrow = logits + (int64_t)row * V;
+  const float temp = temperatures[row];
+  const uint64_t seed = seeds[row];
+  const bool greedy = temp <= 0.f;
+  const float inv_t = greedy ? 1.f : 1.f / temp;
+  const float p_p = top_p[row];
+  const int k_k = top_k[row];
+
+  PenaltyCtx pen{nullptr, nullptr, 1.f, 0.f, 0.f, false};
+  const int slot = row_map ? row_map[row] : -1;
+  if (slot >= 0 && counts != nullptr) {
+    pen.rep = rep_pen[row];
+    pen.pres = pres_pen[row];
+    pen.freq = freq_pen[row];
+    pen.active = pen.rep != 1.f || pen.pres != 0.f || pen.freq != 0.f;
+    if (pen.active) {
+      pen.cnt = counts + (int64_t)slot * V;
+      pen.seen = seen + (int64_t)slot * V;
+    }
+  }
+  const bool filter = !greedy && (p_p < 1.f || k_k > 0);
+
+  __shared__ float h_mass[BINS];
+  __shared__ int h_cnt[BINS];
+  __shared__ float s_m, s_z, s_thresh;
+  __shared__ float red_f[4];
+  __shared__ int red_i[4];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+
+  float thresh = -INFINITY;
+  if (filter) {
+    // ---- pass A: online max + sumexp of adjusted (unscaled) logits ----
+    float m = -INFINITY, z = 0.f;
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      const float v = adjust(raw_val(lrow, i), i, pen);
+      if (v > m) {
+        z = z * __expf(m - v) + 1.f;
+        m = v;
+      } else {
+        z += __expf(v - m);
+      }
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      const float om = __shfl_xor(m, off, WAVE);
+      const float oz = __shfl_xor(z, off, WAVE);
+      const float nm = fmaxf(m, om);
+      // -inf lanes (no elements) contribute 0, not NaN (inf - inf)
+      const float sa = (m == -INFINITY) ? 0.f : __expf(m - nm);
+      const float sb = (om == -INFINITY) ? 0.f : __expf(om - nm);
+      z = z * sa + oz * sb;
+      m = nm;
+    }
+    if (lane == 0) { red_f[wid] = m; red_i[wid] = 0; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float mm = red_f[0];
+      for (int w = 1; w < 4; ++w) mm = fmaxf(mm, red_f[w]);
+      s_m = mm;
+    }
+    __syncthreads();
+    const float M = s_m;
+    // re-scale partial z to global max and sum
+    float zg = (lane == 0 && m != -INFINITY) ? z * __expf(m - M) : 0.f;
+    __syncthreads();
+    if (lane == 0) red_f[wid] = zg;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float zz = 0.f;
+      for (int w = 0; w < 4; ++w) zz += red_f[w];
+      s_z = zz;
+    }
+    // ---- pass B: histogram ----
+    for (int b = threadIdx.x; b < BINS; b += blockDim.x) {
+      h_mass[b] = 0.f;
+      h_cnt[b] = 0;
+    }
+    __syncthreads();
+    const float scale_b = (float)BINS / RANGE;
+    for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      const float v = adjust(raw_val(lrow, i), i, pen);
+      int b = (int)((M - v) * scale_b);
+      b = max(0, min(BINS - 1, b));
+      atomicAdd(&h_mass[b], __expf(v - M));
+      atomicAdd(&h_cnt[b], 1);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      const float target = p_p * s_z;
+      float cum = 0.f;
+      int cnt = 0;
+      int b_p = BINS - 1, b_k = BINS - 1;
+      bool got_p = (p_p >= 1.f), got_k = (k_k <= 0);
+      for (int b = 0; b < BINS; ++b) {
+        cum += h_mass[b];
+        cnt += h_cnt[b];
+        if (!got_p && cum >= target) { b_p = b; got_p = true; }
+        if (!got_k && cnt >= k_k) { b_k = b; got_k = true; }
+        if (got_p && got_k) break;
+      }
+      int b = BINS - 1;
+      if (p_p < 1.f) b = min(b, b_p);
+      if (k_k > 0) b = min(b, b_k);
+      // keep everything strictly above the boundary bin's lower edge
+      s_thresh = (b >= BINS - 1) ? -INFINITY : M - (float)(b + 1) / scale_b;
+    }
+    __syncthreads();
+    thresh = s_thresh;
+  }
+
+  // ---- pass C: Gumbel-argmax over the kept set ----
+  float best = -INFINITY;
+  int best_i = 0;
+  for (int i = threadIdx.x; i < V; i += blockDim.x) {
+    float v = adjust(raw_val(lrow, i), i, pen);
+    if (filter && !(v > thresh)) continue;
+    v *= inv_t;
+    if (!greedy) {
+      const float u = u64_to_uniform(splitmix64(seed ^ (uint64_t)i));
+      v += -__logf(-__logf(u));
+    }
+    if (v > best || (v == best && i < best_i)) {
+      best = v;
+      best_i = i;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ov = __shfl_xor(best, off, WAVE);
+    const int oi = __shfl_xor(best_i, off, WAVE);
+    if (ov > best || (ov == best && oi < best_i)) {
+      best = ov;
+      best_i = oi;
+    }
+  }
+  __syncthreads();
+  if (lane == 0) { red_f[wid] = best; red_i[wid] = best_i; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    for (int w = 1; w < 4; ++w)
+      if (red_f[w] > best || (red_f[w] == best && red_i[w] < best_i)) {
+        best = red_f[w];
+        best_i = red_i[w];
+      }
+    out[row] = best_i;
+  }
+}
+
+}  // namespace
+
+void sample_tokens_ext(torch::Tensor out, torch::Tensor logits,
+                       torch::Tensor temperatures, torch::Tensor seeds,
+                       torch::Tensor top_p, torch::Tensor top_k,
+                       torch::Tensor rep_pen, torch::Tensor pres_pen,
+                       torch::Tensor freq_pen,
+                       c10::optional<torch::Tensor> counts,
+                       c10::optional<torch::Tensor> seen,
+                       c10::optional<torch::Tensor> row_map) {
+  const int B = logits.size(0);
+  const int V = logits.size(1);
+  TORCH_CHECK(out.scalar_type() == torch::kInt64);
+  TORCH_CHECK(top_p.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(top_k.scalar_type() == torch::kInt32);
+  const int* cnt_ptr = nullptr;
+  const uint8_t* seen_ptr = nullptr;
+  const int* map_ptr = nullptr;
+  if (counts.has_value()) {
+    TORCH_CHECK(counts->scalar_type() == torch::kInt32);
+    TORCH_CHECK(seen.has_value() && row_map.has_value());
+    cnt_ptr = counts->data_ptr<int>();
+    seen_ptr = (const uint8_t*)seen->data_ptr();
+    map_ptr = row_map->data_ptr<int>();
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  if (logits.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((sample_topkp_kernel<uint16_t>), dim3(B), dim3(256),
+                       0, stream, out.data_ptr<int64_t>(),
+                       (const uint16_t*)logits.data_ptr(),
+                       temperatures.data_ptr<float>(),
+                       (const uint64_t*)seeds.data_ptr(),
+                       top_p.data_ptr<float>(), top_k.data_ptr<int>(),
+                       rep_pen.data_ptr<float>(), pres_pen.data_ptr<float>(),
+                       freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr,
+                       map_ptr, V);
+  } else {
+    TORCH_CHECK(logits.scalar_type() == torch::kFloat32);
+    hipLaunchKernelGGL((sample_topkp_kernel<float>), dim3(B), dim3(256),
+                       0, stream, out.data_ptr<int64_t>(),
+                       logits.data_ptr<float>(),
+                       temperatures.data_ptr<float>(),
+                       (const uint64_t*)seeds.data_ptr(),
+                       top_p.data_ptr<float>(), top_k.data_ptr<int>(),
+                       rep_pen.data_ptr<float>(), pres_pen.data_ptr<float>(),
+                       freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr,
+                       map_ptr, V);
+  }
+}
