@@ -39,8 +39,12 @@ constexpr int PART_QUANT = 128;
 // kv_cache_dtype="fp8"): halves the KV traffic this kernel is
 // latency/BW-bound on; dequant is one v_cvt per element in the math
 // loops. MFMA_A and KV8 are mutually exclusive instantiations.
-template <int DHEAD, int G, bool MFMA_A, bool KV8>
-__global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_kernel(
+// VPS: phase-C V rows staged RAW (u16x8) per batch — the round-2
+// deep-pipelining lever (PMC wait:busy 14:1, latency-bound): more V
+// bytes in flight per wave before any conversion VALU touches them.
+// OCC: launch_bounds waves/SIMD floor override (0 = round-1 default).
+template <int DHEAD, int G, bool MFMA_A, bool KV8, int VPS = 4, int OCC = 0>
+__global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
     float* __restrict__ tmp_ml,          // [B, Hq, maxP, 2]
@@ -269,22 +273,48 @@ __global__ __launch_bounds__(NTHREADS, (G <= 2 ? 4 : 3)) void paged_attn_decode_
         }
       };
       int ps = 0;
-      for (; ps + 4 <= npass; ps += 4) {
-        int64_t voff[4];
+      if constexpr (!KV8) {
+        // raw-staged batches: VPS x 16B loads issue back-to-back (pure,
+        // no dependent converts between them); conversion + FMA consume
+        // them in order while later loads are still in flight
+        for (; ps + VPS <= npass; ps += VPS) {
+          u16x8 vraw[VPS];
 #pragma unroll
-        for (int u = 0; u < 4; ++u)
-          voff[u] = vrow_off(base + (ps + u) * C_PAR + cpar);
-        float v4[4][8];
+          for (int u = 0; u < VPS; ++u)
+            vraw[u] = *reinterpret_cast<const u16x8*>(
+                v_cache + vrow_off(base + (ps + u) * C_PAR + cpar));
 #pragma unroll
-        for (int u = 0; u < 4; ++u) load_v(voff[u], v4[u]);
+          for (int u = 0; u < VPS; ++u) {
+            const int tok_i = (ps + u) * C_PAR + cpar;
+            float v[8];
 #pragma unroll
-        for (int u = 0; u < 4; ++u) {
-          const int tok_i = (ps + u) * C_PAR + cpar;
+            for (int i = 0; i < 8; ++i) v[i] = bf16_to_f32(vraw[u][i]);
 #pragma unroll
-          for (int g = 0; g < G; ++g) {
-            const float pv = s_lds[g][tok_i];
+            for (int g = 0; g < G; ++g) {
+              const float pv = s_lds[g][tok_i];
 #pragma unroll
-            for (int i = 0; i < 8; ++i) acc[g][i] += pv * v4[u][i];
+              for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
+            }
+          }
+        }
+      } else {
+        for (; ps + 4 <= npass; ps += 4) {
+          int64_t voff[4];
+#pragma unroll
+          for (int u = 0; u < 4; ++u)
+            voff[u] = vrow_off(base + (ps + u) * C_PAR + cpar);
+          float v4[4][8];
+#pragma unroll
+          for (int u = 0; u < 4; ++u) load_v(voff[u], v4[u]);
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            const int tok_i = (ps + u) * C_PAR + cpar;
+#pragma unroll
+            for (int g = 0; g < G; ++g) {
+              const float pv = s_lds[g][tok_i];
+#pragma unroll
+              for (int i = 0; i < 8; ++i) acc[g][i] += pv * v4[u][i];
+            }
           }
         }
       }
@@ -396,6 +426,22 @@ static bool use_mfma_a() {
   return v == 1;
 }
 
+static int env_int(const char* name, int dflt) {
+  const char* e = getenv(name);
+  return e ? atoi(e) : dflt;
+}
+// A/B knobs for the round-2 decode pipeline (defaults = measured best)
+static int decode_vps() {
+  static int v = -1;
+  if (v < 0) v = env_int("HELIX_DECODE_VPS", 8);
+  return v;
+}
+static int decode_occ() {
+  static int v = -1;
+  if (v < 0) v = env_int("HELIX_DECODE_OCC", 0);
+  return v;
+}
+
 template <int DHEAD, int G>
 void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
                    const uint16_t* q, const uint16_t* kc, const uint16_t* vc,
@@ -403,22 +449,23 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
                    int Hkv, int block_size, int max_blocks, int eff_part,
                    int nparts, int max_parts, int window, bool kv8,
                    hipStream_t stream) {
+#define LAUNCH_PD(MF, K8, VPS, OCC)                                          \
+  hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, MF, K8, VPS, OCC>),\
+                     dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,  \
+                     tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,      \
+                     block_size, max_blocks, eff_part, max_parts, window)
   if (kv8) {
-    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, false, true>),
-                       dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
-                       tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
-                       block_size, max_blocks, eff_part, max_parts, window);
+    LAUNCH_PD(false, true, 4, 0);
   } else if (use_mfma_a()) {
-    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, true, false>),
-                       dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
-                       tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
-                       block_size, max_blocks, eff_part, max_parts, window);
+    LAUNCH_PD(true, false, 4, 0);
   } else {
-    hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, false, false>),
-                       dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,
-                       tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,
-                       block_size, max_blocks, eff_part, max_parts, window);
+    const int vps = decode_vps(), occ = decode_occ();
+    if (vps >= 8 && occ >= 4) LAUNCH_PD(false, false, 8, 4);
+    else if (vps >= 8)        LAUNCH_PD(false, false, 8, 0);
+    else if (occ >= 4)        LAUNCH_PD(false, false, 4, 4);
+    else                      LAUNCH_PD(false, false, 4, 0);
   }
+#undef LAUNCH_PD
 }
 
 }  // namespace
