@@ -433,7 +433,7 @@ static int env_int(const char* name, int dflt) {
 // A/B knobs for the round-2 decode pipeline (defaults = measured best)
 static int decode_vps() {
   static int v = -1;
-  if (v < 0) v = env_int("HELIX_DECODE_VPS", 8);
+  if (v < 0) v = env_int("HELIX_DECODE_VPS", 4);
   return v;
 }
 static int decode_occ() {
