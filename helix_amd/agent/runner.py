@@ -92,7 +92,7 @@ class AgentRunner:
                "interaction_id": ctx.get("interaction_id", ""),
                "created": time.time(), **step}
         self.store.put("step_info", sid, doc, owner=ctx.get("owner", ""),
-                       parent=ctx.get("session_id", ""))
+                       parent=ctx.get("session_id", ""), buffered=True)
         return doc
 
     async def _publish_step(self, ctx: dict, doc: dict):

@@ -295,7 +295,8 @@ class LoggingClient(Client):
     def _log(self, call: LLMCall):
         if self.store is not None:
             self.store.put("llm_calls", call.id, call.model_dump(),
-                           owner=call.owner, parent=call.session_id)
+                           owner=call.owner, parent=call.session_id,
+                           buffered=True)
         if self.usage_logger is not None:
             self.usage_logger(call)
 

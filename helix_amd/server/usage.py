@@ -36,7 +36,8 @@ class UsageService:
             "prompt_tokens": call.prompt_tokens,
             "completion_tokens": call.completion_tokens,
             "duration_ms": call.duration_ms, "cost_usd": cost, "day": day,
-            "ts": time.time()}, owner=call.owner, parent=call.session_id)
+            "ts": time.time()}, owner=call.owner, parent=call.session_id,
+            buffered=True)
         rid = f"{call.owner}:{call.provider}:{call.model}:{day}"
         roll = self.store.get("usage_rollups", rid) or {
             "id": rid, "owner": call.owner, "provider": call.provider,
