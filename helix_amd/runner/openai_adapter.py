@@ -13,6 +13,8 @@ from typing import AsyncIterator, Dict, List, Optional
 
 from helix_amd.engine.sampling_params import SamplingParams
 from helix_amd.runner.service import LLMInstance, RunnerService
+from helix_amd.utils.chat_templates import (parse_tool_calls,
+                                             template_for_model)
 from helix_amd.utils.tokenizer import get_tokenizer
 
 
@@ -90,8 +92,11 @@ async def chat_completion(service: RunnerService, req: dict,
     inst = await loop.run_in_executor(None, service.ensure_loaded, model)
     assert isinstance(inst, LLMInstance), f"{model} is not an LLM"
     tok = get_tokenizer(model)
+    tools = req.get("tools")
     if "messages" in req and req["messages"] is not None:
-        prompt_ids = tok.apply_chat_template(req["messages"])
+        prompt_ids = tok.apply_chat_template(
+            req["messages"], template=template_for_model(model),
+            tools=tools)
     else:
         prompt_ids = tok.encode(str(req.get("prompt", "")), add_bos=True)
     max_ctx = inst.spec.max_model_len
@@ -104,7 +109,7 @@ async def chat_completion(service: RunnerService, req: dict,
 
     if req.get("stream"):
         return _stream(service, inst, model, rid, created, prompt_ids,
-                       params, stop_strs, tok)
+                       params, stop_strs, tok, tools)
 
     n = int(req.get("n") or 1)
     choices = []
@@ -113,9 +118,17 @@ async def chat_completion(service: RunnerService, req: dict,
         text, finish_reason, ntok, lp_content = await _generate_one(
             inst, f"{rid}-{i}", prompt_ids, params, stop_strs, tok, loop)
         total_completion += ntok
+        message = {"role": "assistant", "content": text}
+        if tools:
+            content, calls = parse_tool_calls(text)
+            if calls:
+                message = {"role": "assistant",
+                           "content": content or None,
+                           "tool_calls": calls}
+                finish_reason = "tool_calls"
         choice = {
             "index": i,
-            "message": {"role": "assistant", "content": text},
+            "message": message,
             "finish_reason": finish_reason,
         }
         if params.logprobs:
@@ -182,7 +195,7 @@ def _find_stop(text: str, stop_strs: List[str]) -> Optional[int]:
 
 
 async def _stream(service, inst, model, rid, created, prompt_ids, params,
-                  stop_strs, tok) -> AsyncIterator[dict]:
+                  stop_strs, tok, tools=None) -> AsyncIterator[dict]:
     loop = asyncio.get_event_loop()
     ts = TokenStream(loop)
     inst.submit(rid, prompt_ids, params, ts.on_token)
@@ -207,6 +220,7 @@ async def _stream(service, inst, model, rid, created, prompt_ids, params,
         yield chunk({"role": "assistant", "content": ""})
         emitted = 0
         finish_reason = "stop"
+        muted = False   # inside a <tool_call> block: buffer, don't emit
         async for token_id, fin, reason, _lp in ts.__aiter__():
             ntok += 1
             delta = detok.push(token_id)
@@ -214,18 +228,33 @@ async def _stream(service, inst, model, rid, created, prompt_ids, params,
             if hit is not None:
                 inst.cancel(rid)
                 keep = max(0, hit - emitted)
-                if keep:
+                if keep and not muted:
                     yield chunk({"content": delta[:keep]})
                 finish_reason = "stop"
                 finished = True
                 break
-            if delta:
+            if tools and not muted and "<tool_call>" in detok.emitted:
+                # stop emitting content once a tool call begins; the
+                # parsed calls are delivered in the final delta
+                muted = True
+                pre = detok.emitted.index("<tool_call>")
+                keep = max(0, pre - emitted)
+                if keep:
+                    yield chunk({"content": delta[:keep]})
+                emitted = len(detok.emitted)
+            if delta and not muted:
                 emitted += len(delta)
                 yield chunk({"content": delta})
             if fin:
                 finish_reason = reason or "stop"
                 finished = True
                 break
+        if tools:
+            _, calls = parse_tool_calls(detok.emitted)
+            if calls:
+                finish_reason = "tool_calls"
+                yield chunk({"tool_calls": [
+                    {"index": i, **c} for i, c in enumerate(calls)]})
         yield chunk({}, finish=finish_reason, usage={
             "prompt_tokens": len(prompt_ids),
             "completion_tokens": ntok,

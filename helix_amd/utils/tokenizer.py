@@ -42,8 +42,16 @@ class ByteTokenizer:
         return bs.decode("utf-8", errors="replace")
 
     def apply_chat_template(self, messages: List[dict],
-                            add_generation_prompt: bool = True) -> List[int]:
+                            add_generation_prompt: bool = True,
+                            template: str = "", tools=None) -> List[int]:
+        from helix_amd.utils.chat_templates import _tools_system_suffix
         ids = [BOS]
+        suffix = _tools_system_suffix(tools)
+        saw_system = any(m.get("role") == "system" for m in messages)
+        if suffix and not saw_system:
+            ids.append(ROLE_SYSTEM)
+            ids.extend(self.encode(suffix.strip()))
+            ids.append(EOS)
         for m in messages:
             role = m.get("role", "user")
             content = m.get("content") or ""
@@ -51,6 +59,9 @@ class ByteTokenizer:
                 # OpenAI content-parts form
                 content = " ".join(p.get("text", "") for p in content
                                    if isinstance(p, dict))
+            if role == "system" and suffix:
+                content += suffix
+                suffix = ""
             ids.append(_ROLE_TOKENS.get(role, ROLE_USER))
             ids.extend(self.encode(content))
             ids.append(EOS)
@@ -82,17 +93,16 @@ class HFTokenizer:
     def decode(self, ids):
         return self._tok.decode(list(ids), skip_special_tokens=True)
 
-    def apply_chat_template(self, messages, add_generation_prompt=True):
-        parts = []
-        for m in messages:
-            content = m.get("content") or ""
-            if not isinstance(content, str):
-                content = " ".join(p.get("text", "") for p in content
-                                   if isinstance(p, dict))
-            parts.append(f"<|{m.get('role', 'user')}|>\n{content}")
-        if add_generation_prompt:
-            parts.append("<|assistant|>\n")
-        return self.encode("\n".join(parts), add_bos=True)
+    def apply_chat_template(self, messages, add_generation_prompt=True,
+                            template: str = "", tools=None):
+        """Render with the model family's real prompt format (llama3
+        header tokens / mistral [INST] / qwen ChatML) and encode; the
+        special strings are added tokens in real tokenizer.json files."""
+        from helix_amd.utils.chat_templates import TEMPLATES
+        render = TEMPLATES.get(template or "llama3", TEMPLATES["llama3"])
+        text = render(messages, add_generation_prompt, tools=tools)
+        # templates carry their own BOS markers (<|begin_of_text|>/<s>)
+        return self.encode(text, add_bos=False)
 
 
 def get_tokenizer(model: str = ""):
