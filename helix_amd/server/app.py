@@ -638,9 +638,13 @@ def create_app(cfg: Optional[ServerConfig] = None,
     async def create_knowledge(request: Request,
                                user: AuthUser = Depends(auth_dep)):
         body = await request.json()
-        doc = knowledge.create(user.id, body.get("name", ""),
-                               body.get("source", {}),
-                               body.get("app_id", ""))
+        try:
+            doc = knowledge.create(user.id, body.get("name", ""),
+                                   body.get("source", {}),
+                                   body.get("app_id", ""),
+                                   body.get("refresh_schedule", ""))
+        except ValueError as e:
+            raise HTTPException(400, str(e))
         return doc
 
     @app.get("/api/v1/knowledge")

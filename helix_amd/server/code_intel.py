@@ -9,30 +9,13 @@ import logging
 import os
 from typing import List, Optional
 
+from helix_amd.rag.chunker import chunk_code
+
 log = logging.getLogger("helix_amd.code_intel")
 
 CODE_EXT = {".py", ".go", ".rs", ".c", ".cc", ".cpp", ".h", ".hpp", ".hip",
             ".cu", ".js", ".ts", ".tsx", ".java", ".rb", ".sh", ".md",
             ".yaml", ".yml", ".toml", ".json"}
-
-
-def chunk_code(text: str, path: str, max_lines: int = 60,
-               overlap: int = 10) -> List[dict]:
-    """Line-window chunking with path metadata (code-aware enough for
-    retrieval; semantic parsing can come later)."""
-    lines = text.splitlines()
-    if not lines:
-        return []
-    chunks = []
-    step = max(1, max_lines - overlap)
-    for i in range(0, len(lines), step):
-        seg = lines[i:i + max_lines]
-        chunks.append({
-            "text": f"// {path}:{i + 1}\n" + "\n".join(seg),
-            "metadata": {"path": path, "start_line": i + 1}})
-        if i + max_lines >= len(lines):
-            break
-    return chunks
 
 
 class CodeIntelService:
