@@ -11,22 +11,40 @@ import logging
 import time
 from typing import Any, AsyncIterator, Dict, List, Optional
 
-from helix_amd.agent.skills import (APISkill, CalculatorSkill,
-                                    KnowledgeSkill, MemorySkill, Skill,
-                                    WebSearchSkill)
+from helix_amd.agent.skills import (APISkill, BrowserSkill, CalculatorSkill,
+                                    EmailSkill, KnowledgeSkill, MemorySkill,
+                                    Skill, WebSearchSkill, build_mcp_skills)
 from helix_amd.server import pubsub as ps
 from helix_amd.server.types import AssistantConfig, new_id
 
 log = logging.getLogger("helix_amd.agent")
 
+_json_re = __import__("re").compile(r"\{.*\}", __import__("re").S)
+
 
 class AgentRunner:
-    def __init__(self, cfg, store, providers, pubsub, rag=None):
+    def __init__(self, cfg, store, providers, pubsub, rag=None,
+                 notifications=None):
         self.cfg = cfg
         self.store = store
         self.providers = providers
         self.pubsub = pubsub
         self.rag = rag
+        self.notifications = notifications
+
+    def _small_llm(self, assistant: AssistantConfig, owner: str):
+        """Async messages->str callable on the small generation model
+        slot (used by the browser skill's processOutput pass,
+        browser_skill.go:333)."""
+        provider, model = self._model_for(assistant,
+                                          "small_generation_model")
+        client = self.providers.get_client(provider, owner)
+
+        async def call(messages):
+            resp = await client.chat({"model": model,
+                                      "messages": messages})
+            return resp["choices"][0]["message"].get("content", "")
+        return call
 
     # ------------------------------------------------------------------
     def _resolve_secrets(self, owner: str, mapping: dict) -> dict:
@@ -75,7 +93,84 @@ class AgentRunner:
             skills.append(WebSearchSkill(os.environ.get("SEARXNG_URL", "")))
         if assistant.memory.get("enabled", bool(assistant.memory)):
             skills.append(MemorySkill(self.store, owner, app_id))
+        if assistant.browser.get("enabled", bool(assistant.browser)):
+            skills.append(BrowserSkill(assistant.browser,
+                                       llm=self._small_llm(assistant,
+                                                           owner)))
+        if assistant.email.get("enabled", bool(assistant.email)):
+            user = self.store.get("users", owner) or {}
+            skills.append(EmailSkill(self.notifications,
+                                     user.get("email", "")))
         return skills
+
+    async def resolve_skills(self, assistant: AssistantConfig, owner: str,
+                             app_id: str = "") -> List[Skill]:
+        """build_skills + remote MCP tool discovery (async because each
+        MCP server is queried with tools/list, mcp_skill.go:18)."""
+        skills = self.build_skills(assistant, owner, app_id)
+        for mcp_cfg in assistant.mcps:
+            try:
+                skills.extend(await build_mcp_skills(mcp_cfg))
+            except Exception as e:
+                log.warning("MCP server %s unavailable: %s",
+                            mcp_cfg.get("url", "?"), e)
+        return skills
+
+    # ------------------------------------------------------------------
+    async def is_actionable(self, assistant: AssistantConfig,
+                            messages: List[dict], owner: str,
+                            ctx: dict) -> dict:
+        """Classify whether the latest user input needs a tool call
+        (reference api/pkg/tools/informative_or_actionable.go: JSON
+        {needs_tool: yes|no, api, justification}, retried, over the
+        tools' name+description list and truncated history)."""
+        skills = self.build_skills(assistant, owner,
+                                   ctx.get("app_id", ""))
+        if not skills:
+            return {"needs_tool": "no", "api": "",
+                    "justification": "no tools available"}
+        tool_lines = "\n".join(f"- {s.name}: {s.description}"
+                               for s in skills)
+        system = (
+            "You decide whether the user's last input requires calling "
+            "one of the available tools (actionable) or can be answered "
+            "directly (informative).\nAvailable tools:\n" + tool_lines +
+            "\nRespond with ONLY a JSON object: "
+            '{"needs_tool": "yes"|"no", "api": "<tool name or empty>", '
+            '"justification": "<why>"}')
+        history = [m for m in messages if m.get("role") in
+                   ("user", "assistant")][-6:]
+        provider, model = self._model_for(assistant,
+                                          "small_reasoning_model")
+        client = self.providers.get_client(provider, owner)
+        from helix_amd.server.providers import LoggingClient
+        client = LoggingClient(client, self.store)
+        req_messages = [{"role": "system", "content": system}] + history
+        if req_messages[-1]["role"] == "user":
+            req_messages[-1] = dict(
+                req_messages[-1],
+                content=(req_messages[-1].get("content") or "") +
+                "\nReturn the corresponding json for the last user input")
+        for attempt in range(3):
+            resp = await client.chat({
+                "model": model, "messages": req_messages,
+                "temperature": 0.0,
+                "_ctx": {**ctx, "step": "is_actionable"}})
+            text = resp["choices"][0]["message"].get("content", "")
+            m = _json_re.search(text)
+            if m:
+                try:
+                    out = json.loads(m.group(0))
+                    if str(out.get("needs_tool", "")).lower() in \
+                            ("yes", "no"):
+                        out["needs_tool"] = out["needs_tool"].lower()
+                        return out
+                except json.JSONDecodeError:
+                    pass
+            log.warning("is_actionable parse failure (attempt %d): %r",
+                        attempt, text[:200])
+        return {"needs_tool": "no", "api": "",
+                "justification": "classification failed; informative"}
 
     def _model_for(self, assistant: AssistantConfig, slot: str) -> tuple:
         """4-slot model selection (reference llm_client.go:14-19)."""
@@ -108,7 +203,7 @@ class AgentRunner:
         """Run the agentic loop; returns (messages, final_req) where
         final_req is what produces the user-facing answer."""
         app_id = ctx.get("app_id", "")
-        skills = self.build_skills(assistant, owner, app_id)
+        skills = await self.resolve_skills(assistant, owner, app_id)
         tools = [s.to_tool() for s in skills]
         by_name = {s.name: s for s in skills}
 

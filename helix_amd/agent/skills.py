@@ -223,3 +223,190 @@ class MemorySkill(Skill):
         if not mems:
             return "no memories stored"
         return "\n".join(f"- {m['content']}" for m in mems)
+
+
+# ---------------------------------------------------------------------------
+class BrowserSkill(Skill):
+    """Fetch a URL, run readability extraction, and (optionally) distill
+    the content with the small generation model (reference
+    api/pkg/agent/skill/browser_skill.go:89-179: url+prompt parameters,
+    NoBrowser HTTP fetch path, readability parse, processOutput LLM
+    pass, per-URL cache)."""
+
+    name = "browser"
+    description = ("Open a URL in a browser, extract the readable "
+                   "content, and answer a prompt about it.")
+    parameters = {"type": "object", "properties": {
+        "url": {"type": "string", "description": "The URL to visit"},
+        "prompt": {"type": "string",
+                   "description": "The prompt to run on the fetched "
+                                  "content"}},
+        "required": ["url", "prompt"]}
+
+    def __init__(self, config: dict | None = None, llm=None,
+                 http_client=None):
+        import httpx
+        self.config = config or {}
+        self.llm = llm                  # async callable(messages) -> str
+        self._http = http_client or httpx.AsyncClient(
+            timeout=30, follow_redirects=True)
+        self._cache: Dict[str, str] = {}
+
+    async def execute(self, args: dict, ctx: dict) -> str:
+        from helix_amd.server.extract import extract_html
+        url = args.get("url", "")
+        prompt = args.get("prompt", "")
+        if not url:
+            return "browser error: url is required"
+        if not url.startswith(("http://", "https://")):
+            return "browser error: only http(s) URLs are supported"
+        text = self._cache.get(url) if self.config.get("cache", True) \
+            else None
+        if text is None:
+            try:
+                r = await self._http.get(url)
+            except Exception as e:
+                return f"browser error: {e}"
+            if r.status_code >= 400:
+                return f"browser error: HTTP {r.status_code} for {url}"
+            page = extract_html(r.text)
+            title = f"# {page['title']}\n\n" if page["title"] else ""
+            text = (title + page["text"])[:16000]
+            if self.config.get("cache", True):
+                self._cache[url] = text
+        if self.config.get("process_output", True) and \
+                self.llm is not None and prompt:
+            try:
+                return await self.llm([
+                    {"role": "system",
+                     "content": "You extract the information requested "
+                                "from fetched web content. Answer from "
+                                "the content only."},
+                    {"role": "user",
+                     "content": f"Request: {prompt}\n\nContent:\n{text}"}])
+            except Exception as e:
+                log.warning("browser process_output failed: %s", e)
+        return text
+
+
+# ---------------------------------------------------------------------------
+class EmailSkill(Skill):
+    """Send an email to the requesting user (reference
+    email_sending_skill.go:88-160: subject+message parameters, delivery
+    to meta.UserEmail only — an agent cannot email arbitrary
+    addresses)."""
+
+    name = "send_email"
+    description = "Send an email to the current user."
+    parameters = {"type": "object", "properties": {
+        "subject": {"type": "string"},
+        "message": {"type": "string"}},
+        "required": ["subject", "message"]}
+
+    def __init__(self, notifications, user_email: str):
+        self.notifications = notifications
+        self.user_email = user_email
+
+    async def execute(self, args: dict, ctx: dict) -> str:
+        subject = args.get("subject", "")
+        message = args.get("message", "")
+        if not subject or not message:
+            return "email error: subject and message are required"
+        if not self.user_email:
+            return "email error: the current user has no email address"
+        email = getattr(self.notifications, "email", None)
+        if email is None or not getattr(email, "host", ""):
+            return "email error: no email provider configured"
+        ok = email.send(self.user_email, subject, message)
+        return "Email sent" if ok else "email error: delivery failed"
+
+
+# ---------------------------------------------------------------------------
+class MCPClient:
+    """Minimal MCP client over streamable-HTTP JSON-RPC (reference
+    api/pkg/agent/skill/mcp/mcp_client.go): initialize, tools/list,
+    tools/call. Compatible with helix_amd's own MCP gateway."""
+
+    def __init__(self, url: str, headers: dict | None = None,
+                 http_client=None):
+        import httpx
+        self.url = url
+        self.headers = headers or {}
+        self._http = http_client or httpx.AsyncClient(timeout=30)
+        self._id = 0
+        self._initialized = False
+
+    async def _rpc(self, method: str, params: dict | None = None):
+        self._id += 1
+        r = await self._http.post(self.url, headers=self.headers, json={
+            "jsonrpc": "2.0", "id": self._id, "method": method,
+            "params": params or {}})
+        body = r.json()
+        if "error" in body:
+            raise RuntimeError(f"MCP {method}: {body['error'].get('message')}")
+        return body.get("result", {})
+
+    async def _ensure_init(self):
+        if not self._initialized:
+            await self._rpc("initialize", {
+                "protocolVersion": "2024-11-05",
+                "clientInfo": {"name": "helix_amd", "version": "1.0"},
+                "capabilities": {}})
+            self._initialized = True
+
+    async def list_tools(self) -> List[dict]:
+        await self._ensure_init()
+        return (await self._rpc("tools/list")).get("tools", [])
+
+    async def call_tool(self, name: str, arguments: dict) -> str:
+        await self._ensure_init()
+        result = await self._rpc("tools/call",
+                                 {"name": name, "arguments": arguments})
+        parts = []
+        for c in result.get("content", []):
+            if c.get("type") == "text":
+                parts.append(c.get("text", ""))
+            else:
+                parts.append(json.dumps(c))
+        out = "\n".join(parts)
+        if result.get("isError"):
+            return f"tool error: {out}"
+        return out
+
+
+def _sanitize_tool_name(name: str) -> str:
+    import re as _re
+    return _re.sub(r"[^A-Za-z0-9_\-]", "_", name)[:64]
+
+
+class MCPToolSkill(Skill):
+    """One remote MCP tool surfaced as an agent skill (reference
+    mcp_skill.go:18-60: per-tool skills named mcp_<name>, inputSchema
+    converted to OpenAI tool parameters)."""
+
+    def __init__(self, client: MCPClient, tool: dict):
+        self.client = client
+        self.tool_name = tool.get("name", "")
+        self.name = f"mcp_{_sanitize_tool_name(self.tool_name)}"
+        self.description = tool.get("description", "") or \
+            f"MCP tool {self.tool_name}"
+        schema = tool.get("inputSchema") or {}
+        self.parameters = {
+            "type": "object",
+            "properties": schema.get("properties", {}) or {},
+            **({"required": schema["required"]}
+               if schema.get("required") else {})}
+
+    async def execute(self, args: dict, ctx: dict) -> str:
+        try:
+            return await self.client.call_tool(self.tool_name, args)
+        except Exception as e:
+            return f"mcp error: {e}"
+
+
+async def build_mcp_skills(mcp_cfg: dict, http_client=None) -> List[Skill]:
+    """Connect to one configured MCP server and surface each remote
+    tool (config shape: {url, headers?} — reference ToolMCPClientConfig)."""
+    client = MCPClient(mcp_cfg.get("url", ""),
+                       mcp_cfg.get("headers") or {}, http_client)
+    return [MCPToolSkill(client, t) for t in await client.list_tools()]

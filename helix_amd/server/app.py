@@ -72,13 +72,16 @@ def create_app(cfg: Optional[ServerConfig] = None,
 
     from helix_amd.agent.runner import AgentRunner
     from helix_amd.rag.service import RAGService
+    from helix_amd.server.notifications import NotificationService
+    notifications = NotificationService(store=store)
     from helix_amd.server.filestore import FileStore
     from helix_amd.server.knowledge import KnowledgeReconciler
     from helix_amd.server.models_catalog import ModelCatalog
     from helix_amd.server.triggers import TriggerManager
     from helix_amd.server.usage import QuotaExceededError, UsageService
     rag = RAGService(cfg, store, providers)
-    agent_runner = AgentRunner(cfg, store, providers, pubsub, rag=rag)
+    agent_runner = AgentRunner(cfg, store, providers, pubsub, rag=rag,
+                               notifications=notifications)
     knowledge = KnowledgeReconciler(cfg, store, rag)
     catalog = ModelCatalog(store)
     from helix_amd.server.metrics import Metrics
@@ -89,7 +92,6 @@ def create_app(cfg: Optional[ServerConfig] = None,
 
     from helix_amd.server.code_intel import CodeIntelService
     from helix_amd.server.git_service import GitService
-    from helix_amd.server.notifications import NotificationService
     from helix_amd.server.oauth import OAuthManager
     from helix_amd.server.rbac import RBACService
     from helix_amd.server.runner_profiles import (ProfileService,
@@ -106,7 +108,6 @@ def create_app(cfg: Optional[ServerConfig] = None,
     git_svc = GitService(store, cfg.filestore.path)
     spec_tasks = SpecTaskService(store, controller, git_svc)
     code_intel = CodeIntelService(rag, git_svc)
-    notifications = NotificationService(store=store)
     from helix_amd.server.evaluations import EvaluationService
     evaluations = EvaluationService(store, controller, pubsub)
     app.state.evaluations = evaluations

@@ -149,6 +149,15 @@ class Controller:
             model = self.cfg.inference.default_model
         return provider, model
 
+    @staticmethod
+    def _has_tools(assistant) -> bool:
+        """Non-agent assistants with tools go through the actionable
+        classification chain (reference ChainStrategy.IsActionable,
+        tools/informative_or_actionable.go) before tool execution."""
+        return bool(assistant.apis or assistant.mcps or
+                    assistant.browser or assistant.email or
+                    assistant.zapier)
+
     async def chat_completion(self, req: dict, owner: str,
                               app_id: str = "", assistant_id: str = "",
                               ctx: Optional[dict] = None) -> dict:
@@ -159,6 +168,14 @@ class Controller:
                 raise ProviderError("agent runtime unavailable", 500)
             return await self.agent_runner.run_blocking(
                 assistant, req, owner, ctx or {})
+        if assistant is not None and self.agent_runner is not None and \
+                self._has_tools(assistant):
+            cls = await self.agent_runner.is_actionable(
+                assistant, req.get("messages", []), owner,
+                {**(ctx or {}), "app_id": app.id if app else ""})
+            if cls.get("needs_tool") == "yes":
+                return await self.agent_runner.run_blocking(
+                    assistant, req, owner, ctx or {})
         req = self._apply_assistant(req, assistant)
         if assistant is not None:
             req["messages"] = await self.enrich_with_knowledge(
@@ -182,6 +199,16 @@ class Controller:
                     assistant, req, owner, ctx or {}):
                 yield chunk
             return
+        if assistant is not None and self.agent_runner is not None and \
+                self._has_tools(assistant):
+            cls = await self.agent_runner.is_actionable(
+                assistant, req.get("messages", []), owner,
+                {**(ctx or {}), "app_id": app.id if app else ""})
+            if cls.get("needs_tool") == "yes":
+                async for chunk in self.agent_runner.run_stream(
+                        assistant, req, owner, ctx or {}):
+                    yield chunk
+                return
         req = self._apply_assistant(req, assistant)
         if assistant is not None:
             req["messages"] = await self.enrich_with_knowledge(
