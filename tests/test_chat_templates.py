@@ -131,8 +131,7 @@ def test_adapter_nonstream_tool_calls(monkeypatch):
            "messages": [{"role": "user", "content": "2+2?"}],
            "tools": [{"type": "function",
                       "function": {"name": "calc", "parameters": {}}}]}
-    resp = asyncio.get_event_loop().run_until_complete(
-        oa.chat_completion(FakeSvc(), req))
+    resp = asyncio.run(oa.chat_completion(FakeSvc(), req))
     ch = resp["choices"][0]
     assert ch["finish_reason"] == "tool_calls"
     assert ch["message"]["tool_calls"][0]["function"]["name"] == "calc"
