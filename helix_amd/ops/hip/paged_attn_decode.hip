@@ -43,7 +43,15 @@ constexpr int PART_QUANT = 128;
 // deep-pipelining lever (PMC wait:busy 14:1, latency-bound): more V
 // bytes in flight per wave before any conversion VALU touches them.
 // OCC: launch_bounds waves/SIMD floor override (0 = round-1 default).
-template <int DHEAD, int G, bool MFMA_A, bool KV8, int VPS = 4, int OCC = 0>
+// KPRE: inter-chunk K prefetch. The r2 ISA audit (profiles/
+// r02_decode_isa.md) showed the compiler sinking phase A's 16-deep raw
+// K staging into the consume loop — steady state s_waitcnt vmcnt(1),
+// i.e. only ~2 VMEM loads in flight per wave, matching the 14:1
+// wait:busy PMC. KPRE hoists the next chunk's 16 b128 K loads to just
+// after the current chunk's scores are consumed, so they land during
+// phases B+C (softmax + V) instead of head-of-line blocking phase A.
+template <int DHEAD, int G, bool MFMA_A, bool KV8, int VPS = 4, int OCC = 0,
+          bool KPRE = false>
 __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void paged_attn_decode_kernel(
     uint16_t* __restrict__ out,          // [B, Hq, D] (used when nparts==1)
     float* __restrict__ tmp_out,         // [B, Hq, maxP, D]
@@ -130,6 +138,22 @@ __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void 
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
 
+  // KPRE: raw K rows for the chunk about to be scored, loaded one full
+  // chunk ahead (16 b128 loads live across phases B+C of the previous
+  // chunk — the latency they need to hide).
+  constexpr bool USE_KPRE = KPRE && !MFMA_A && !KV8;
+  u16x8 kpre[USE_KPRE ? DHEAD / 8 : 1];
+  auto issue_kpre = [&](int cbase) {
+    const int tok = min(cbase + (int)threadIdx.x, p_end - 1);
+    const uint16_t* krow =
+        k_cache + (((int64_t)btable[tok / block_size] * Hkv + hkv) *
+                       (int64_t)block_size + tok % block_size) * DHEAD;
+#pragma unroll
+    for (int j = 0; j < (USE_KPRE ? DHEAD / 8 : 1); ++j)
+      kpre[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+  };
+  if constexpr (USE_KPRE) issue_kpre(p_start);
+
   for (int base = p_start; base < p_end; base += CHUNK) {
     const int chunk_n = min(CHUNK, p_end - base);
 
@@ -173,6 +197,29 @@ __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void 
           }
         }
       }
+    } else if constexpr (USE_KPRE) {
+      if ((int)threadIdx.x < chunk_n) {
+        float s[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) s[g] = 0.f;
+#pragma unroll
+        for (int j = 0; j < DHEAD / 8; ++j) {
+          float kv[8];
+#pragma unroll
+          for (int i = 0; i < 8; ++i) kv[i] = bf16_to_f32(kpre[j][i]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+#pragma unroll
+            for (int i = 0; i < 8; ++i)
+              s[g] += q_lds[g][j * 8 + i] * kv[i];
+          }
+        }
+#pragma unroll
+        for (int g = 0; g < G; ++g) s_lds[g][threadIdx.x] = s[g];
+      }
+      // scores consumed: issue the NEXT chunk's K rows now so they
+      // fly during softmax + V accumulation
+      if (base + CHUNK < p_end) issue_kpre(base + CHUNK);
     } else if ((int)threadIdx.x < chunk_n) {
       const int tok = base + threadIdx.x;
       const int64_t blk = btable[tok / block_size];
@@ -441,6 +488,11 @@ static int decode_occ() {
   if (v < 0) v = env_int("HELIX_DECODE_OCC", 0);
   return v;
 }
+static int decode_kpre() {
+  static int v = -1;
+  if (v < 0) v = env_int("HELIX_DECODE_KPRE", 1);
+  return v;
+}
 
 template <int DHEAD, int G>
 void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
@@ -449,21 +501,27 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
                    int Hkv, int block_size, int max_blocks, int eff_part,
                    int nparts, int max_parts, int window, bool kv8,
                    hipStream_t stream) {
-#define LAUNCH_PD(MF, K8, VPS, OCC)                                          \
-  hipLaunchKernelGGL((paged_attn_decode_kernel<DHEAD, G, MF, K8, VPS, OCC>),\
-                     dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,  \
-                     tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,      \
-                     block_size, max_blocks, eff_part, max_parts, window)
+#define LAUNCH_PD(MF, K8, VPS, OCC, KP)                                      \
+  hipLaunchKernelGGL(                                                        \
+      (paged_attn_decode_kernel<DHEAD, G, MF, K8, VPS, OCC, KP>),            \
+      dim3(B, Hkv, nparts), dim3(NTHREADS), 0, stream, out,                  \
+      tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,                      \
+      block_size, max_blocks, eff_part, max_parts, window)
   if (kv8) {
-    LAUNCH_PD(false, true, 4, 0);
+    LAUNCH_PD(false, true, 4, 0, false);
   } else if (use_mfma_a()) {
-    LAUNCH_PD(true, false, 4, 0);
+    LAUNCH_PD(true, false, 4, 0, false);
   } else {
     const int vps = decode_vps(), occ = decode_occ();
-    if (vps >= 8 && occ >= 4) LAUNCH_PD(false, false, 8, 4);
-    else if (vps >= 8)        LAUNCH_PD(false, false, 8, 0);
-    else if (occ >= 4)        LAUNCH_PD(false, false, 4, 4);
-    else                      LAUNCH_PD(false, false, 4, 0);
+    if (decode_kpre()) {
+      if (occ == 2)      LAUNCH_PD(false, false, 4, 2, true);
+      else               LAUNCH_PD(false, false, 4, 0, true);
+    }
+    else if (occ == 2)        LAUNCH_PD(false, false, 4, 2, false);
+    else if (vps >= 8 && occ >= 4) LAUNCH_PD(false, false, 8, 4, false);
+    else if (vps >= 8)        LAUNCH_PD(false, false, 8, 0, false);
+    else if (occ >= 4)        LAUNCH_PD(false, false, 4, 4, false);
+    else                      LAUNCH_PD(false, false, 4, 0, false);
   }
 #undef LAUNCH_PD
 }

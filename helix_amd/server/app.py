@@ -489,6 +489,68 @@ def create_app(cfg: Optional[ServerConfig] = None,
         return {"access_token": auth.issue_jwt(user, ttl),
                 "token_type": "Bearer", "expires_in": ttl}
 
+    # -- OIDC login mode (reference api/pkg/auth/oidc.go +
+    #    session_manager.go: discovery, state+nonce, code exchange,
+    #    ID-token verification, user mapping, session token) ----------
+    app.state.oidc_client = None
+    if cfg.oidc.enabled and cfg.oidc.issuer:
+        from helix_amd.server.oidc import OIDCClient
+        app.state.oidc_client = OIDCClient(
+            cfg.oidc.issuer, cfg.oidc.client_id, cfg.oidc.client_secret,
+            cfg.oidc.redirect_url,
+            allowed_domains=cfg.oidc.allowed_domains)
+
+    @app.get("/api/v1/auth/oidc/login")
+    async def oidc_login():
+        client = app.state.oidc_client
+        if client is None:
+            raise HTTPException(404, "OIDC is not configured")
+        import time as _t
+        state, nonce = new_id("oidcst"), new_id("nonce")
+        store.put("oidc_states", state,
+                  {"id": state, "nonce": nonce, "ts": _t.time()})
+        return {"url": await client.get_auth_url(state, nonce),
+                "state": state}
+
+    @app.get("/api/v1/auth/oidc/callback")
+    async def oidc_callback(code: str, state: str):
+        from helix_amd.server.oidc import OIDCError
+        client = app.state.oidc_client
+        if client is None:
+            raise HTTPException(404, "OIDC is not configured")
+        import time as _t
+        st = store.get("oidc_states", state)
+        if st is None or _t.time() - st.get("ts", 0) > 600:
+            raise HTTPException(400, "invalid or expired state")
+        store.delete("oidc_states", state)
+        try:
+            tok = await client.exchange(code)
+            claims = await client.verify_id_token(
+                tok.get("id_token", ""), st.get("nonce", ""))
+        except OIDCError as e:
+            raise HTTPException(401, str(e))
+        sub = claims.get("sub", "")
+        email = claims.get("email", "")
+        username = claims.get("preferred_username") or email or sub
+        import hashlib as _h
+        uid = f"user_{_h.sha1(('oidc:' + sub).encode()).hexdigest()[:16]}"
+        doc = store.get("users", uid)
+        if doc is None:
+            doc = {"id": uid, "username": username, "admin": False,
+                   "email": email, "oidc_sub": sub,
+                   "name": claims.get("name", "")}
+            store.put("users", uid, doc, owner=uid)
+        elif email and doc.get("email") != email:
+            doc["email"] = email
+            store.put("users", uid, doc, owner=uid)
+        jwt = auth.issue_jwt(AuthUser(id=uid, username=username,
+                                      admin=doc.get("admin", False)),
+                             ttl_s=86400)
+        return {"access_token": jwt, "token_type": "Bearer",
+                "expires_in": 86400,
+                "user": {"id": uid, "username": username,
+                         "email": email}}
+
     @app.websocket("/api/v1/ws/user")
     async def ws_user(ws: WebSocket):
         token = ws.query_params.get("access_token", "")

@@ -101,6 +101,25 @@ class WebServerConfig:
 
 
 @dataclass
+class OIDCConfig:
+    """OIDC login mode (reference api/pkg/auth/oidc.go OIDCConfig:
+    KEYCLOAK/OIDC issuer + client credentials + allowed email
+    domains)."""
+    enabled: bool = field(
+        default_factory=lambda: _env("OIDC_ENABLED", "") in
+        ("1", "true", "yes"))
+    issuer: str = field(default_factory=lambda: _env("OIDC_ISSUER", ""))
+    client_id: str = field(
+        default_factory=lambda: _env("OIDC_CLIENT_ID", "helix"))
+    client_secret: str = field(
+        default_factory=lambda: _env("OIDC_CLIENT_SECRET", ""))
+    redirect_url: str = field(
+        default_factory=lambda: _env("OIDC_REDIRECT_URL", ""))
+    allowed_domains: str = field(
+        default_factory=lambda: _env("OIDC_ALLOWED_DOMAINS", ""))
+
+
+@dataclass
 class StoreConfig:
     path: str = field(
         default_factory=lambda: _env("HELIX_STORE_PATH", "helix.db"))
@@ -120,6 +139,7 @@ class ServerConfig:
     rag: RAGConfig = field(default_factory=RAGConfig)
     web: WebServerConfig = field(default_factory=WebServerConfig)
     store: StoreConfig = field(default_factory=StoreConfig)
+    oidc: OIDCConfig = field(default_factory=OIDCConfig)
     filestore: FileStoreConfig = field(default_factory=FileStoreConfig)
     # agent loop cap (reference agent.go:26 maxIterations default 10)
     agent_max_iterations: int = field(
