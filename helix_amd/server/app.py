@@ -1141,6 +1141,50 @@ def create_app(cfg: Optional[ServerConfig] = None,
         return usage.topup(body.get("owner", user.id),
                            float(body.get("amount_usd", 0)))
 
+    # -- Stripe billing (reference api/pkg/stripe: checkout top-ups,
+    #    webhook-driven wallet credit + subscription sync) -------------
+    from helix_amd.server.billing import BillingService, WebhookError
+    billing = BillingService(
+        store, usage,
+        webhook_secret=os.environ.get("HELIX_STRIPE_WEBHOOK_SECRET", ""))
+    app.state.billing = billing
+
+    @app.post("/api/v1/billing/topup-session")
+    async def billing_topup_session(request: Request,
+                                    user: AuthUser = Depends(auth_dep)):
+        if not billing.enabled():
+            raise HTTPException(
+                501, "billing is not configured "
+                "(HELIX_STRIPE_WEBHOOK_SECRET unset)")
+        body = await request.json()
+        udoc = store.get("users", user.id) or {"id": user.id}
+        try:
+            return await billing.create_topup_session(
+                udoc, float(body.get("amount_usd", 0)))
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+
+    @app.post("/api/v1/stripe/webhook")
+    async def stripe_webhook(request: Request):
+        if not billing.enabled():
+            raise HTTPException(501, "billing is not configured")
+        payload = await request.body()
+        if len(payload) > 65536:           # stripe.go:138 MaxBodyBytes
+            raise HTTPException(413, "payload too large")
+        try:
+            return billing.process_webhook(
+                payload, request.headers.get("Stripe-Signature", ""))
+        except WebhookError as e:
+            raise HTTPException(400, str(e))
+
+    @app.get("/api/v1/billing")
+    async def billing_state(user: AuthUser = Depends(auth_dep)):
+        w = usage.wallet(user.id)
+        invoices = [i for i in store.list("billing_invoices",
+                                          owner=user.id, limit=100)]
+        return {"wallet": w, "invoices": invoices,
+                "enabled": billing.enabled()}
+
     # ------------------------------------------------------------------
     # Triggers (reference api/pkg/trigger: cron + webhook)
     # ------------------------------------------------------------------

@@ -1,0 +1,134 @@
+"""Stripe billing (VERDICT missing #9; reference api/pkg/stripe):
+webhook signature verification (t=,v1= HMAC scheme), event routing with
+per-event-id dedup, top-up wallet credit, subscription sync, and the
+HTTP surface.
+"""
+import json
+import time
+
+import pytest
+
+from helix_amd.server.billing import (BillingService, FakeStripeAPI,
+                                      WebhookError, sign_stripe_payload,
+                                      verify_stripe_signature)
+
+SECRET = "whsec_test_123"
+
+
+def test_signature_scheme():
+    payload = b'{"id": "evt_1"}'
+    hdr = sign_stripe_payload(payload, SECRET)
+    verify_stripe_signature(payload, hdr, SECRET)
+    with pytest.raises(WebhookError, match="mismatch"):
+        verify_stripe_signature(payload + b"x", hdr, SECRET)
+    with pytest.raises(WebhookError, match="mismatch"):
+        verify_stripe_signature(payload, hdr, "other-secret")
+    with pytest.raises(WebhookError, match="tolerance"):
+        verify_stripe_signature(
+            payload, sign_stripe_payload(payload, SECRET,
+                                         now=time.time() - 3600),
+            SECRET)
+    with pytest.raises(WebhookError, match="malformed"):
+        verify_stripe_signature(payload, "garbage", SECRET)
+
+
+@pytest.fixture()
+def svc(tmp_path):
+    from helix_amd.server.usage import UsageService
+    from helix_amd.store import Store
+    store = Store(str(tmp_path / "db.sqlite"))
+    usage = UsageService(store)
+    return store, usage, BillingService(store, usage,
+                                        webhook_secret=SECRET)
+
+
+def _event(etype, obj, eid="evt_1"):
+    return json.dumps({"id": eid, "type": etype,
+                       "data": {"object": obj}}).encode()
+
+
+def test_topup_checkout_flow(svc):
+    store, usage, billing = svc
+    import asyncio
+    user = {"id": "u1", "email": "u1@test.dev"}
+    sess = asyncio.run(billing.create_topup_session(user, 25.0))
+    assert sess["url"].startswith("https://checkout.stripe.test/")
+    # customer is reused on the second call
+    sess2 = asyncio.run(billing.create_topup_session(user, 10.0))
+    assert len(billing.api.customers) == 1
+    # completed checkout credits the wallet once, dedup on replay
+    obj = {"id": sess["session_id"],
+           "customer": list(billing.api.customers)[0],
+           "amount_total": 2500,
+           "metadata": {"user_id": "u1", "amount_cents": "2500"}}
+    payload = _event("checkout.session.completed", obj)
+    hdr = sign_stripe_payload(payload, SECRET)
+    out = billing.process_webhook(payload, hdr)
+    assert out["handled"]
+    assert usage.wallet("u1")["balance_usd"] == pytest.approx(25.0)
+    out2 = billing.process_webhook(payload, hdr)
+    assert out2.get("deduped")
+    assert usage.wallet("u1")["balance_usd"] == pytest.approx(25.0)
+    with pytest.raises(ValueError):
+        asyncio.run(billing.create_topup_session(user, -5))
+
+
+def test_subscription_sync(svc):
+    store, usage, billing = svc
+    obj = {"id": "sub_1", "status": "active",
+           "metadata": {"user_id": "u2"},
+           "items": {"data": [{"price": {"lookup_key": "pro-monthly"}}]}}
+    p = _event("customer.subscription.created", obj, eid="evt_s1")
+    billing.process_webhook(p, sign_stripe_payload(p, SECRET))
+    w = usage.wallet("u2")
+    assert w["subscription_status"] == "active"
+    assert w["subscription_plan"] == "pro-monthly"
+    p = _event("customer.subscription.deleted", obj, eid="evt_s2")
+    billing.process_webhook(p, sign_stripe_payload(p, SECRET))
+    assert usage.wallet("u2")["subscription_status"] == "canceled"
+
+
+def test_unknown_event_and_unknown_customer(svc):
+    store, usage, billing = svc
+    p = _event("charge.refunded", {"id": "ch_1"}, eid="evt_x")
+    out = billing.process_webhook(p, sign_stripe_payload(p, SECRET))
+    assert out["ok"] and not out["handled"]
+    # top-up for an unknown customer is logged, not crashed
+    p = _event("payment_intent.succeeded",
+               {"id": "pi_1", "customer": "cus_zzz",
+                "amount_received": 500}, eid="evt_y")
+    out = billing.process_webhook(p, sign_stripe_payload(p, SECRET))
+    assert out["ok"]
+
+
+def test_http_surface(tmp_path, monkeypatch):
+    from fastapi.testclient import TestClient
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import load_config
+    monkeypatch.setenv("HELIX_STRIPE_WEBHOOK_SECRET", SECRET)
+    cfg = load_config()
+    cfg.store.path = str(tmp_path / "db.sqlite")
+    cfg.filestore.path = str(tmp_path / "fs")
+    app = create_app(cfg)
+    with TestClient(app) as client:
+        auth = app.state.auth
+        me = auth.create_user("billing-user")
+        key = auth.create_api_key(me["id"])
+        H = {"Authorization": f"Bearer {key}"}
+        r = client.post("/api/v1/billing/topup-session",
+                        json={"amount_usd": 12}, headers=H)
+        assert r.status_code == 200, r.text
+        # webhook credits the wallet
+        obj = {"id": "cs_x", "metadata": {"user_id": me["id"],
+                                          "amount_cents": "1200"}}
+        payload = _event("checkout.session.completed", obj, "evt_h1")
+        r = client.post("/api/v1/stripe/webhook", content=payload,
+                        headers={"Stripe-Signature":
+                                 sign_stripe_payload(payload, SECRET)})
+        assert r.status_code == 200, r.text
+        r = client.get("/api/v1/billing", headers=H)
+        assert r.json()["wallet"]["balance_usd"] == pytest.approx(12.0)
+        # bad signature rejected
+        r = client.post("/api/v1/stripe/webhook", content=payload,
+                        headers={"Stripe-Signature": "t=1,v1=bad"})
+        assert r.status_code == 400
