@@ -113,14 +113,16 @@ class Store:
     def flush(self, table: Optional[str] = None):
         """Commit pending buffered writes (all tables or one) in a single
         transaction. Last-write-wins per id, like the reference's
-        websocket accumulator."""
+        websocket accumulator. Entries stay in the buffer until AFTER
+        the commit — a concurrent get() must hit either the buffer or
+        the committed row, never the gap between them (WAL readers do
+        not see uncommitted writes)."""
         with self._pending_lock:
             if table is None:
                 items = list(self._pending.items())
-                self._pending.clear()
             else:
-                keys = [k for k in self._pending if k[0] == table]
-                items = [(k, self._pending.pop(k)) for k in keys]
+                items = [(k, v) for k, v in self._pending.items()
+                         if k[0] == table]
         if not items:
             return
         with self._lock:
@@ -134,6 +136,11 @@ class Store:
                           updated=excluded.updated, doc=excluded.doc""",
                     (id_, owner, parent, ts, ts, blob))
             self._db.commit()
+        with self._pending_lock:
+            for k, v in items:
+                # drop only if not overwritten since the snapshot
+                if self._pending.get(k) is v:
+                    del self._pending[k]
 
     # ------------------------------------------------------------------
     def put(self, table: str, id: str, doc: Dict[str, Any],
