@@ -518,7 +518,8 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
       // vs 26,556 baseline at B=512 (profiles/r02_decode_kpre.md);
       // HELIX_DECODE_OCC=3 pins the prefetch variant at the old floor
       // for A/B.
-      if (occ == 0 || occ == 2) LAUNCH_PD(false, false, 4, 2, true);
+      if (vps >= 8)             LAUNCH_PD(false, false, 8, 2, true);
+      else if (occ == 0 || occ == 2) LAUNCH_PD(false, false, 4, 2, true);
       else                      LAUNCH_PD(false, false, 4, 0, true);
     }
     else if (occ == 2)        LAUNCH_PD(false, false, 4, 2, false);

@@ -106,7 +106,13 @@ def create_app(cfg: Optional[ServerConfig] = None,
     oauth = OAuthManager(store)
     profiles = ProfileService(store)
     git_svc = GitService(store, cfg.filestore.path)
-    spec_tasks = SpecTaskService(store, controller, git_svc)
+    from helix_amd.server.sandbox import SandboxError, SandboxManager
+    sandboxes = SandboxManager(
+        store, os.path.join(cfg.filestore.path, "sandboxes"),
+        golden_dir=os.environ.get("HELIX_SANDBOX_GOLDEN", ""))
+    app.state.sandboxes = sandboxes
+    spec_tasks = SpecTaskService(store, controller, git_svc,
+                                 sandboxes=sandboxes)
     code_intel = CodeIntelService(rag, git_svc)
     from helix_amd.server.evaluations import EvaluationService
     evaluations = EvaluationService(store, controller, pubsub)
@@ -1479,6 +1485,128 @@ def create_app(cfg: Optional[ServerConfig] = None,
         body = await request.json()
         return await code_intel.query(rid, body.get("query", ""),
                                       body.get("k", 6))
+
+    # ------------------------------------------------------------------
+    # Sandboxes (reference api/pkg/hydra dev containers + api/pkg/sandbox:
+    # create/list/get/delete, exec, file IO, PTY terminal)
+    # ------------------------------------------------------------------
+    def _owned_sandbox(sid: str, user: AuthUser) -> dict:
+        doc = sandboxes.get(sid)
+        if doc is None:
+            raise HTTPException(404, "sandbox not found")
+        if doc.get("owner") != user.id and not user.admin:
+            raise HTTPException(403, "not your sandbox")
+        return doc
+
+    @app.post("/api/v1/sandboxes")
+    async def create_sandbox(request: Request,
+                             user: AuthUser = Depends(auth_dep)):
+        body = await request.json()
+        return sandboxes.create(user.id, body.get("name", ""),
+                                body.get("session_id", ""))
+
+    @app.get("/api/v1/sandboxes")
+    async def list_sandboxes(user: AuthUser = Depends(auth_dep)):
+        return sandboxes.list(user.id)
+
+    @app.get("/api/v1/sandboxes/{sid}")
+    async def get_sandbox(sid: str, user: AuthUser = Depends(auth_dep)):
+        return _owned_sandbox(sid, user)
+
+    @app.delete("/api/v1/sandboxes/{sid}")
+    async def delete_sandbox(sid: str,
+                             user: AuthUser = Depends(auth_dep)):
+        _owned_sandbox(sid, user)
+        return {"ok": sandboxes.delete(sid)}
+
+    @app.post("/api/v1/sandboxes/{sid}/exec")
+    async def exec_in_sandbox(sid: str, request: Request,
+                              user: AuthUser = Depends(auth_dep)):
+        _owned_sandbox(sid, user)
+        body = await request.json()
+        cmd = body.get("command", "")
+        if not cmd:
+            raise HTTPException(400, "command required")
+        try:
+            return await asyncio.to_thread(
+                sandboxes.exec, sid, cmd,
+                min(float(body.get("timeout_s", 60)), 300),
+                body.get("cwd", ""), body.get("env"))
+        except SandboxError as e:
+            raise HTTPException(400, str(e))
+
+    @app.get("/api/v1/sandboxes/{sid}/files")
+    async def sandbox_files(sid: str, path: str = "",
+                            user: AuthUser = Depends(auth_dep)):
+        _owned_sandbox(sid, user)
+        try:
+            return sandboxes.list_files(sid, path)
+        except (SandboxError, OSError) as e:
+            raise HTTPException(400, str(e))
+
+    @app.get("/api/v1/sandboxes/{sid}/file")
+    async def sandbox_read_file(sid: str, path: str,
+                                user: AuthUser = Depends(auth_dep)):
+        _owned_sandbox(sid, user)
+        try:
+            data = sandboxes.read_file(sid, path)
+        except (SandboxError, OSError) as e:
+            raise HTTPException(400, str(e))
+        return {"path": path,
+                "content": data.decode("utf-8", errors="replace")}
+
+    @app.put("/api/v1/sandboxes/{sid}/file")
+    async def sandbox_write_file(sid: str, request: Request,
+                                 user: AuthUser = Depends(auth_dep)):
+        _owned_sandbox(sid, user)
+        body = await request.json()
+        try:
+            sandboxes.write_file(sid, body.get("path", ""),
+                                 body.get("content", "").encode())
+        except (SandboxError, OSError) as e:
+            raise HTTPException(400, str(e))
+        return {"ok": True}
+
+    @app.websocket("/api/v1/sandboxes/{sid}/terminal")
+    async def sandbox_terminal(ws: WebSocket, sid: str):
+        user = auth.resolve(ws.query_params.get("access_token", ""))
+        if user is None:
+            await ws.close(code=4401)
+            return
+        doc = sandboxes.get(sid)
+        if doc is None or (doc.get("owner") != user.id and
+                           not user.admin):
+            await ws.close(code=4403)
+            return
+        await ws.accept()
+        pid, master = sandboxes.open_terminal(sid)
+        loop = asyncio.get_event_loop()
+
+        async def pump_out():
+            try:
+                while True:
+                    data = await loop.run_in_executor(
+                        None, lambda: os.read(master, 4096))
+                    if not data:
+                        break
+                    await ws.send_bytes(data)
+            except (OSError, RuntimeError):
+                pass
+
+        out_task = asyncio.ensure_future(pump_out())
+        try:
+            while True:
+                data = await ws.receive_bytes()
+                os.write(master, data)
+        except Exception:
+            pass
+        finally:
+            out_task.cancel()
+            try:
+                os.kill(pid, 9)
+                os.close(master)
+            except OSError:
+                pass
 
     @app.get("/api/v1/config")
     async def get_config(user: AuthUser = Depends(auth_dep)):

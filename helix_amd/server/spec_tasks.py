@@ -27,7 +27,8 @@ _NEXT = {
 
 
 class SpecTaskService:
-    def __init__(self, store, controller=None, git=None):
+    def __init__(self, store, controller=None, git=None, sandboxes=None):
+        self.sandboxes = sandboxes
         self.store = store
         self.controller = controller
         self.git = git
@@ -125,12 +126,16 @@ class SpecTaskService:
             return self.transition(tid, "failed")
 
     # -- implementation agent (reference spec_task_orchestrator impl
-    # phase; runs in-process against the project's bare repo instead of
-    # a sandboxed dev container — no per-session dockerd on this stack) --
+    # phase: implementation runs in sandboxes; here the process-level
+    # SandboxManager replaces the per-session dockerd dev container) --
     async def implement(self, tid: str, model: str = "") -> dict:
         """Drive the approved spec to a commit on a task branch: the LLM
-        emits a JSON file-manifest which is committed to
-        `helix/task-{id}`; the task moves in_progress -> pr."""
+        emits a JSON file-manifest which is materialized into a sandbox
+        workspace alongside the repo tree; the manifest's optional
+        `verify` command runs inside the sandbox (rlimited, scrubbed
+        env) and its outcome is recorded on the task before the files
+        are committed to `helix/task-{id}`; the task moves
+        in_progress -> pr."""
         import json as _json
         doc = self.transition(tid, "in_progress")
         if self.controller is None or self.git is None:
@@ -172,13 +177,55 @@ class SpecTaskService:
                 # the model produced no manifest: record the raw output
                 files = {f"tasks/{tid}/output.md": text}
             msg = manifest.get("message") or f"task: {doc['title']}"
+            verify = None
+            if self.sandboxes is not None:
+                verify = self._sandbox_verify(doc, rid, tree, files,
+                                              manifest.get("verify", ""))
             branch = f"helix/task-{tid[-8:]}"
             self.git.commit_files(rid, files, msg, branch=branch)
             doc = self.get_task(tid)
             doc["branch"] = branch
+            if verify is not None:
+                doc["verify"] = verify
             self.store.put("spec_tasks", tid, doc, owner=doc["owner"],
                            parent=doc["project_id"])
             return self.transition(tid, "pr")
         except Exception:
             log.exception("implementation failed for %s", tid)
             return self.transition(tid, "failed")
+
+    def _sandbox_verify(self, doc, rid, tree, files, verify_cmd):
+        """Materialize repo + manifest into a fresh sandbox and run the
+        verify command there (reference: implementation agents execute
+        in hydra sandboxes, spec_task_orchestrator.go)."""
+        sbx = self.sandboxes.create(doc["owner"],
+                                    name=f"task-{doc['id'][-8:]}",
+                                    session_id=doc["id"])
+        try:
+            for path in tree[:500]:
+                try:
+                    content = self.git.read_file(rid, path)
+                except Exception:
+                    continue
+                self.sandboxes.write_file(sbx["id"], path,
+                                          content.encode())
+            for path, content in files.items():
+                self.sandboxes.write_file(
+                    sbx["id"], path,
+                    content.encode() if isinstance(content, str)
+                    else content)
+            if not verify_cmd:
+                return {"ran": False}
+            result = self.sandboxes.exec(sbx["id"], verify_cmd,
+                                         timeout_s=120)
+            return {"ran": True, "command": verify_cmd,
+                    "exit_code": result["exit_code"],
+                    "timed_out": result["timed_out"],
+                    "output": (result["stdout"] +
+                               result["stderr"])[-2000:]}
+        except Exception as e:
+            log.warning("sandbox verify failed for %s: %s",
+                        doc["id"], e)
+            return {"ran": False, "error": str(e)}
+        finally:
+            self.sandboxes.delete(sbx["id"])
