@@ -514,13 +514,20 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
   } else {
     const int vps = decode_vps(), occ = decode_occ();
     if (decode_kpre()) {
-      // measured default: K-prefetch + waves/EU floor 2 = 27,995 tok/s
-      // vs 26,556 baseline at B=512 (profiles/r02_decode_kpre.md);
-      // HELIX_DECODE_OCC=3 pins the prefetch variant at the old floor
-      // for A/B.
-      if (vps >= 8)             LAUNCH_PD(false, false, 8, 2, true);
+      // measured defaults (profiles/r02_decode_kpre.md): K-prefetch +
+      // waves/EU floor 2 everywhere (+5.4% e2e at B=512). When the
+      // grid underfills the chip (< 2048 workgroups on 256 CUs x 4
+      // SIMDs), per-wave pipelining has to replace occupancy-based
+      // latency hiding: the 8-deep V staging wins those shapes
+      // (4.25-4.29 vs 3.6-3.7 TB/s at B<=64 / L>=2048) but loses the
+      // full-grid B=512 headline (27.5k vs 28.1k tok/s), so it is
+      // grid-gated, not global. HELIX_DECODE_VPS / _OCC override.
+      const bool vps_forced = getenv("HELIX_DECODE_VPS") != nullptr;
+      const bool small_grid = B * Hkv * nparts < 2048;
+      const bool deep = vps_forced ? (vps >= 8) : small_grid;
+      if (deep)                      LAUNCH_PD(false, false, 8, 2, true);
       else if (occ == 0 || occ == 2) LAUNCH_PD(false, false, 4, 2, true);
-      else                      LAUNCH_PD(false, false, 4, 0, true);
+      else                           LAUNCH_PD(false, false, 4, 0, true);
     }
     else if (occ == 2)        LAUNCH_PD(false, false, 4, 2, false);
     else if (vps >= 8 && occ >= 4) LAUNCH_PD(false, false, 8, 4, false);
