@@ -575,6 +575,81 @@ def create_app(cfg: Optional[ServerConfig] = None,
         finally:
             await sub.close()
 
+    # -- cross-process message bus (reference embedded NATS WS listener
+    #    + JetStream, nats.go:119-163,608-699): external processes SUB/
+    #    PUB ephemeral topics and FETCH/ACK durable streams over one WS --
+    from helix_amd.server.pubsub import StreamBus
+    bus = StreamBus(store, pubsub)
+    app.state.bus = bus
+
+    @app.websocket("/api/v1/ws/bus")
+    async def ws_bus(ws: WebSocket):
+        token = ws.query_params.get("access_token", "")
+        user = auth.resolve(token)
+        is_runner = token and token == cfg.runner_plane.runner_token
+        if user is None and not is_runner:
+            await ws.close(code=4401)
+            return
+        await ws.accept()
+        subs = {}
+        forwarders = []
+
+        async def forward(pattern, sub):
+            try:
+                while True:
+                    topic, msg = await sub.get()
+                    await ws.send_json({"op": "msg", "pattern": pattern,
+                                        "topic": topic, "payload": msg})
+            except Exception:
+                pass
+
+        try:
+            while True:
+                frame = await ws.receive_json()
+                op = frame.get("op")
+                if op == "sub":
+                    pattern = frame.get("pattern", "")
+                    if pattern and pattern not in subs:
+                        sub = await pubsub.subscribe(pattern)
+                        subs[pattern] = sub
+                        forwarders.append(
+                            asyncio.ensure_future(forward(pattern, sub)))
+                elif op == "unsub":
+                    sub = subs.pop(frame.get("pattern", ""), None)
+                    if sub:
+                        await sub.close()
+                elif op == "pub":
+                    await pubsub.publish(frame.get("topic", ""),
+                                         frame.get("payload"))
+                elif op == "stream_pub":
+                    seq = await bus.publish_notify(
+                        frame.get("stream", "default"),
+                        frame.get("subject", ""), frame.get("payload"))
+                    await ws.send_json({"op": "pub_ack", "seq": seq})
+                elif op == "fetch":
+                    msgs = bus.fetch(frame.get("stream", "default"),
+                                     frame.get("durable", "d"),
+                                     int(frame.get("batch", 10)),
+                                     frame.get("subject_filter", "*"))
+                    await ws.send_json({"op": "batch",
+                                        "messages": msgs})
+                elif op == "ack":
+                    bus.ack(frame.get("stream", "default"),
+                            frame.get("durable", "d"),
+                            int(frame.get("seq", 0)))
+                elif op == "ping":
+                    await ws.send_json({"op": "pong"})
+        except Exception:
+            pass
+        finally:
+            for f in forwarders:
+                f.cancel()
+            for sub in subs.values():
+                try:
+                    await sub.close()
+                except Exception:
+                    pass
+
     # ------------------------------------------------------------------
     # Apps / agents (registered under BOTH aliases, agent_routes.go:12-50)
     # ------------------------------------------------------------------
