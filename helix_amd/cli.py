@@ -550,5 +550,202 @@ def export_gguf(preset: str = typer.Option("llama3-8b"),
                f"{est['weights'] >> 20} MiB")
 
 
+# ---------------------------------------------------------------------------
+# Round-2 subcommand families (reference root.go:45-72 has ~24: org,
+# project, spectask, mcp, evals, member, team ... — these map onto the
+# HelixClient library, client.py).
+
+def _client():
+    from helix_amd.client import HelixClient
+    return HelixClient()
+
+
+org_app = typer.Typer(help="Organizations and teams")
+app.add_typer(org_app, name="org")
+
+
+@org_app.command("list")
+def org_list():
+    for o in _client().list_organizations():
+        typer.echo(f"{o['id']}  {o.get('name', '')}")
+
+
+@org_app.command("create")
+def org_create(name: str):
+    o = _client().create_organization(name)
+    typer.echo(o["id"])
+
+
+@org_app.command("add-member")
+def org_add_member(org_id: str, user_id: str,
+                   role: str = typer.Option("member")):
+    _client().add_org_member(org_id, user_id, role)
+    typer.echo("added")
+
+
+@org_app.command("teams")
+def org_teams(org_id: str):
+    for t in _client().list_teams(org_id):
+        typer.echo(f"{t['id']}  {t.get('name', '')}")
+
+
+@org_app.command("create-team")
+def org_create_team(org_id: str, name: str):
+    typer.echo(_client().create_team(org_id, name)["id"])
+
+
+project_app = typer.Typer(help="Projects and spec-driven tasks")
+app.add_typer(project_app, name="project")
+
+
+@project_app.command("list")
+def project_list():
+    for p in _client().list_projects():
+        typer.echo(f"{p['id']}  {p.get('name', '')}")
+
+
+@project_app.command("create")
+def project_create(name: str):
+    typer.echo(_client().create_project(name)["id"])
+
+
+@project_app.command("tasks")
+def project_tasks(project_id: str):
+    for t in _client().list_tasks(project_id):
+        typer.echo(f"{t['id']}  [{t.get('state', '?')}] "
+                   f"{t.get('title', '')}")
+
+
+spectask_app = typer.Typer(help="Spec-task kanban operations")
+app.add_typer(spectask_app, name="spectask")
+
+
+@spectask_app.command("create")
+def spectask_create(project_id: str, title: str,
+                    description: str = typer.Option("")):
+    typer.echo(_client().create_task(project_id, title,
+                                     description)["id"])
+
+
+@spectask_app.command("transition")
+def spectask_transition(task_id: str, state: str):
+    t = _client().transition_task(task_id, state)
+    typer.echo(t.get("state", ""))
+
+
+@spectask_app.command("plan")
+def spectask_plan(task_id: str):
+    t = _client().plan_task(task_id)
+    typer.echo(t.get("state", ""))
+
+
+@spectask_app.command("implement")
+def spectask_implement(task_id: str):
+    t = _client().implement_task(task_id)
+    typer.echo(f"{t.get('state', '')} branch={t.get('branch', '')}")
+
+
+mcp_app = typer.Typer(help="MCP gateway operations")
+app.add_typer(mcp_app, name="mcp")
+
+
+@mcp_app.command("tools")
+def mcp_tools(app_id: str):
+    """List the MCP tools an app exposes (tools/list JSON-RPC)."""
+    c = _client()
+    out = c.request("POST", f"/api/v1/mcp/{app_id}",
+                    body={"jsonrpc": "2.0", "id": 1,
+                          "method": "tools/list", "params": {}})
+    for t in (out.get("result", {}) or {}).get("tools", []):
+        typer.echo(f"{t['name']}: {t.get('description', '')[:80]}")
+
+
+@mcp_app.command("call")
+def mcp_call(app_id: str, tool: str,
+             args: str = typer.Option("{}", help="JSON arguments")):
+    c = _client()
+    out = c.request("POST", f"/api/v1/mcp/{app_id}",
+                    body={"jsonrpc": "2.0", "id": 1,
+                          "method": "tools/call",
+                          "params": {"name": tool,
+                                     "arguments": json.loads(args)}})
+    typer.echo(json.dumps(out.get("result", out), indent=2))
+
+
+evals_app = typer.Typer(help="Evaluation suites")
+app.add_typer(evals_app, name="evals")
+
+
+@evals_app.command("create")
+def evals_create(app_id: str, name: str,
+                 cases_file: str = typer.Option(..., "-f")):
+    with open(cases_file) as fh:
+        cases = json.load(fh)
+    s = _client().create_evaluation_suite(app_id, name, cases)
+    typer.echo(s["id"])
+
+
+@evals_app.command("run")
+def evals_run(suite_id: str):
+    r = _client().run_evaluation_suite(suite_id)
+    typer.echo(json.dumps(r, indent=2))
+
+
+@evals_app.command("show")
+def evals_show(run_id: str):
+    typer.echo(json.dumps(_client().get_evaluation_run(run_id),
+                          indent=2))
+
+
+sandbox_app = typer.Typer(help="Sandboxed workspaces")
+app.add_typer(sandbox_app, name="sandbox")
+
+
+@sandbox_app.command("create")
+def sandbox_create(name: str = typer.Option("")):
+    typer.echo(_client().create_sandbox(name)["id"])
+
+
+@sandbox_app.command("list")
+def sandbox_list():
+    for s in _client().list_sandboxes():
+        typer.echo(f"{s['id']}  {s.get('name', '')}  "
+                   f"{s.get('state', '')}")
+
+
+@sandbox_app.command("exec")
+def sandbox_exec(sandbox_id: str, command: str,
+                 timeout_s: float = typer.Option(60.0)):
+    r = _client().sandbox_exec(sandbox_id, command, timeout_s)
+    if r["stdout"]:
+        typer.echo(r["stdout"], nl=False)
+    if r["stderr"]:
+        typer.echo(r["stderr"], nl=False, err=True)
+    raise typer.Exit(code=0 if r["exit_code"] == 0 else 1)
+
+
+@sandbox_app.command("rm")
+def sandbox_rm(sandbox_id: str):
+    _client().delete_sandbox(sandbox_id)
+    typer.echo("deleted")
+
+
+billing_app = typer.Typer(help="Billing and usage")
+app.add_typer(billing_app, name="billing")
+
+
+@billing_app.command("show")
+def billing_show():
+    typer.echo(json.dumps(_client().billing(), indent=2))
+
+
+@billing_app.command("topup")
+def billing_topup(amount_usd: float):
+    c = _client()
+    out = c.request("POST", "/api/v1/billing/topup-session",
+                    body={"amount_usd": amount_usd})
+    typer.echo(out.get("url", ""))
+
+
 if __name__ == "__main__":
     app()

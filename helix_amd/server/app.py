@@ -981,6 +981,13 @@ def create_app(cfg: Optional[ServerConfig] = None,
         body = await request.json()
         return rbac.create_team(oid, body.get("name", ""))
 
+    @app.get("/api/v1/organizations/{oid}/teams")
+    async def list_org_teams(oid: str,
+                             user: AuthUser = Depends(auth_dep)):
+        if rbac.member_role(oid, user.id) is None and not user.admin:
+            raise HTTPException(403, "not a member")
+        return rbac.list_teams(oid)
+
     @app.post("/api/v1/teams/{tid}/members")
     async def add_team_member(tid: str, request: Request,
                               user: AuthUser = Depends(auth_dep)):
