@@ -19,12 +19,29 @@ def chunk_text(text: str, chunk_size: int = 512, overlap: int = 64,
     for p in paras:
         if len(p) <= chunk_size:
             pieces.append(p)
-        else:
-            step = max(1, chunk_size - overlap)
-            for i in range(0, len(p), step):
-                pieces.append(p[i:i + chunk_size])
-                if i + chunk_size >= len(p):
-                    break
+            continue
+        # long paragraph: pack whole words into windows, overlap on a
+        # word boundary — a blind p[i:i+size] slice can split a word
+        # longer than the overlap across two chunks, losing it from
+        # retrieval (caught by the chunker coverage property test)
+        cur = ""
+        for w in p.split():
+            while len(w) > chunk_size:          # single oversized token
+                if cur:
+                    pieces.append(cur)
+                    cur = ""
+                pieces.append(w[:chunk_size])
+                w = w[max(1, chunk_size - overlap):]
+            if cur and len(cur) + 1 + len(w) > chunk_size:
+                pieces.append(cur)
+                tail = cur[-overlap:] if overlap else ""
+                sp = tail.find(" ")
+                tail = tail[sp + 1:] if sp >= 0 else ""
+                cur = (tail + " " + w) if tail else w
+            else:
+                cur = (cur + " " + w) if cur else w
+        if cur:
+            pieces.append(cur)
     chunks: List[dict] = []
     cur = ""
     for piece in pieces:
