@@ -157,3 +157,89 @@ async def stream_anthropic_events(chunks: AsyncIterator[dict],
         "delta": {"stop_reason": finish, "stop_sequence": None},
         "usage": {"output_tokens": out_tokens}})
     yield ev("message_stop", {"type": "message_stop"})
+
+
+# ---------------------------------------------------------------------------
+# Native passthrough (reference api/pkg/anthropic/anthropic_proxy.go:
+# reverse proxy that forwards the RAW /v1/messages body to the upstream
+# Anthropic-compatible endpoint — preserving cache_control blocks and
+# thus prompt caching — swaps the caller's helix token for the provider
+# key (anthropic_proxy.go:158,212), and retries the thinking.type
+# adaptive<->enabled mismatch (anthropic_proxy.go:108-111)).
+
+class AnthropicPassthrough:
+    def __init__(self, base_url: str, api_key: str, http_client=None,
+                 version: str = "2023-06-01"):
+        self.base_url = base_url.rstrip("/")
+        self.api_key = api_key
+        self.version = version
+        self._http = http_client
+
+    def _client(self):
+        if self._http is None:
+            import httpx
+            self._http = httpx.AsyncClient(timeout=600)
+        return self._http
+
+    def _headers(self, incoming: dict | None = None) -> dict:
+        h = {"content-type": "application/json",
+             "x-api-key": self.api_key,
+             "anthropic-version":
+                 (incoming or {}).get("anthropic-version", self.version)}
+        beta = (incoming or {}).get("anthropic-beta")
+        if beta:
+            h["anthropic-beta"] = beta
+        return h
+
+    @staticmethod
+    def _flip_thinking(body: dict) -> dict | None:
+        th = body.get("thinking")
+        if not isinstance(th, dict) or "type" not in th:
+            return None
+        flipped = dict(body)
+        t = th.get("type")
+        alt = {"adaptive": "enabled", "enabled": "adaptive"}.get(t)
+        if alt is None:
+            return None
+        flipped["thinking"] = dict(th, type=alt)
+        return flipped
+
+    async def forward(self, body: dict, incoming_headers: dict
+                      | None = None):
+        """Non-streaming: returns (status_code, response_json). Retries
+        once with the alternate thinking.type on a 400 naming it."""
+        url = self.base_url + "/v1/messages"
+        r = await self._client().post(url, json=body,
+                                      headers=self._headers(
+                                          incoming_headers))
+        if r.status_code == 400:
+            try:
+                err = r.json()
+            except Exception:
+                err = {}
+            if "thinking" in json.dumps(err):
+                flipped = self._flip_thinking(body)
+                if flipped is not None:
+                    r = await self._client().post(
+                        url, json=flipped,
+                        headers=self._headers(incoming_headers))
+        try:
+            return r.status_code, r.json()
+        except Exception:
+            return r.status_code, {"type": "error", "error": {
+                "type": "api_error", "message": r.text[:500]}}
+
+    async def forward_stream(self, body: dict,
+                             incoming_headers: dict | None = None):
+        """Streaming: yields raw SSE bytes verbatim from upstream."""
+        url = self.base_url + "/v1/messages"
+        client = self._client()
+        async with client.stream("POST", url, json=body,
+                                 headers=self._headers(
+                                     incoming_headers)) as r:
+            if r.status_code >= 400:
+                raw = await r.aread()
+                yield (b"event: error\ndata: " + raw[:2000] + b"\n\n")
+                return
+            async for chunk in r.aiter_bytes():
+                yield chunk

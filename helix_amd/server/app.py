@@ -1149,10 +1149,35 @@ def create_app(cfg: Optional[ServerConfig] = None,
     @app.post("/v1/messages")
     async def anthropic_messages(request: Request,
                                  user: AuthUser = Depends(auth_dep)):
-        from helix_amd.server.anthropic_api import (anthropic_to_openai,
+        from helix_amd.server.anthropic_api import (AnthropicPassthrough,
+                                                    anthropic_to_openai,
                                                     openai_to_anthropic,
                                                     stream_anthropic_events)
         areq = await request.json()
+        # native passthrough when an Anthropic endpoint is configured:
+        # raw body forwarded => prompt caching (cache_control) survives
+        base = os.environ.get("ANTHROPIC_BASE_URL", "")
+        akey = os.environ.get("ANTHROPIC_API_KEY", "")
+        if not base:
+            for ep in store.list("provider_endpoints", limit=1000):
+                if ep.get("provider") == "anthropic" and \
+                        ep.get("base_url"):
+                    base = ep["base_url"]
+                    akey = ep.get("api_key", "")
+                    break
+        if base:
+            pt = getattr(app.state, "anthropic_passthrough", None)
+            if pt is None or pt.base_url != base.rstrip("/"):
+                pt = AnthropicPassthrough(base, akey)
+                app.state.anthropic_passthrough = pt
+            in_h = {k.lower(): v for k, v in request.headers.items()
+                    if k.lower().startswith("anthropic-")}
+            if areq.get("stream"):
+                return StreamingResponse(
+                    pt.forward_stream(areq, in_h),
+                    media_type="text/event-stream")
+            status, body = await pt.forward(areq, in_h)
+            return JSONResponse(body, status_code=status)
         oreq = anthropic_to_openai(areq)
         ctx = {"owner": user.id}
         if oreq.get("stream"):
