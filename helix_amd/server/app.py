@@ -939,6 +939,19 @@ def create_app(cfg: Optional[ServerConfig] = None,
             raise HTTPException(409, why)
         return {"ok": True}
 
+    @app.get("/api/v1/runners/{runner_id}/compatible-profiles")
+    async def compatible_profiles(runner_id: str,
+                                  user: AuthUser = Depends(admin_dep)):
+        """Profiles whose GPU requirements this runner satisfies
+        (reference gpucloud scenario 2 `compatibility_filter`)."""
+        from helix_amd.server.runner_profiles import filter_compatible
+        state = next((r for r in router.runners()
+                      if r.runner_id == runner_id), None)
+        if state is None:
+            raise HTTPException(404, "runner not found")
+        return [p.model_dump()
+                for p in filter_compatible(profiles.list(), state.gpus)]
+
     @app.get("/api/v1/runner/{runner_id}/assignment")
     async def get_assignment(runner_id: str, _=Depends(runner_dep)):
         # runner polls its assigned profile (reference compose-manager
