@@ -13,7 +13,9 @@ from typing import Any, AsyncIterator, Dict, List, Optional
 
 from helix_amd.agent.skills import (APISkill, BrowserSkill, CalculatorSkill,
                                     EmailSkill, KnowledgeSkill, MemorySkill,
-                                    Skill, WebSearchSkill, build_mcp_skills)
+                                    ProjectSkill, RepositorySkill,
+                                    SandboxSkill, Skill, WebSearchSkill,
+                                    build_mcp_skills)
 from helix_amd.server import pubsub as ps
 from helix_amd.server.types import AssistantConfig, new_id
 
@@ -31,6 +33,11 @@ class AgentRunner:
         self.pubsub = pubsub
         self.rag = rag
         self.notifications = notifications
+        # late-bound platform services (set by create_app after the
+        # full dependency graph exists)
+        self.git = None
+        self.spec_tasks = None
+        self.sandboxes = None
 
     def _small_llm(self, assistant: AssistantConfig, owner: str):
         """Async messages->str callable on the small generation model
@@ -101,6 +108,17 @@ class AgentRunner:
             user = self.store.get("users", owner) or {}
             skills.append(EmailSkill(self.notifications,
                                      user.get("email", "")))
+        if self.git is not None and \
+                assistant.repository.get("enabled",
+                                         bool(assistant.repository)):
+            skills.append(RepositorySkill(self.git, owner))
+        if self.spec_tasks is not None and \
+                assistant.project.get("enabled", bool(assistant.project)):
+            skills.append(ProjectSkill(self.spec_tasks, owner))
+        if self.sandboxes is not None and \
+                assistant.sandbox.get("enabled", bool(assistant.sandbox)):
+            skills.append(SandboxSkill(self.sandboxes, owner,
+                                       app_id))
         return skills
 
     async def resolve_skills(self, assistant: AssistantConfig, owner: str,

@@ -410,3 +410,186 @@ async def build_mcp_skills(mcp_cfg: dict, http_client=None) -> List[Skill]:
     client = MCPClient(mcp_cfg.get("url", ""),
                        mcp_cfg.get("headers") or {}, http_client)
     return [MCPToolSkill(client, t) for t in await client.list_tools()]
+
+
+# ---------------------------------------------------------------------------
+class RepositorySkill(Skill):
+    """Repository tools over the in-platform git service (reference
+    api/pkg/agent/skill/repository: list_files, find_files, get_file,
+    grep — helix_repository_skill.go). One skill, action-dispatched,
+    scoped to the owner's repositories."""
+
+    name = "repository"
+    description = ("Explore the user's git repositories: "
+                   "action=list_repos|list_files|find_files|get_file|"
+                   "grep.")
+    parameters = {"type": "object", "properties": {
+        "action": {"type": "string",
+                   "enum": ["list_repos", "list_files", "find_files",
+                            "get_file", "grep"]},
+        "repo_id": {"type": "string"},
+        "path": {"type": "string",
+                 "description": "file path (get_file)"},
+        "pattern": {"type": "string",
+                    "description": "glob (find_files) or regex (grep)"},
+        "ref": {"type": "string", "description": "git ref, default HEAD"}},
+        "required": ["action"]}
+
+    def __init__(self, git, owner: str):
+        self.git = git
+        self.owner = owner
+
+    def _repo(self, repo_id: str):
+        doc = self.git.get(repo_id)
+        if doc is None or doc.get("owner") != self.owner:
+            raise ValueError(f"repository not found: {repo_id}")
+        return doc
+
+    async def execute(self, args: dict, ctx: dict) -> str:
+        import fnmatch
+        import re as _re
+        action = args.get("action", "")
+        ref = args.get("ref") or "HEAD"
+        if action == "list_repos":
+            repos = self.git.list(self.owner)
+            return "\n".join(f"{r['id']}  {r.get('name', '')}"
+                             for r in repos) or "no repositories"
+        rid = args.get("repo_id", "")
+        self._repo(rid)
+        if action == "list_files":
+            return "\n".join(self.git.ls_tree(rid, ref)[:500])
+        if action == "find_files":
+            pat = args.get("pattern", "*")
+            hits = [p for p in self.git.ls_tree(rid, ref)
+                    if fnmatch.fnmatch(p, pat)]
+            return "\n".join(hits[:200]) or "no matches"
+        if action == "get_file":
+            return self.git.read_file(rid, args.get("path", ""),
+                                      ref)[:16000]
+        if action == "grep":
+            rx = _re.compile(args.get("pattern", ""))
+            out = []
+            for p in self.git.ls_tree(rid, ref)[:500]:
+                try:
+                    content = self.git.read_file(rid, p, ref)
+                except Exception:
+                    continue
+                for i, line in enumerate(content.splitlines(), 1):
+                    if rx.search(line):
+                        out.append(f"{p}:{i}: {line.strip()[:200]}")
+                        if len(out) >= 100:
+                            return "\n".join(out)
+            return "\n".join(out) or "no matches"
+        return f"unknown action: {action}"
+
+
+class ProjectSkill(Skill):
+    """Spec-task project management (reference
+    api/pkg/agent/skill/project: create/get/list/update/start spec
+    tasks — the optimus PM agent's tool family)."""
+
+    name = "project"
+    description = ("Manage spec-driven tasks on the user's projects: "
+                   "action=list_projects|list_tasks|get_task|"
+                   "create_task|update_task|start_task.")
+    parameters = {"type": "object", "properties": {
+        "action": {"type": "string",
+                   "enum": ["list_projects", "list_tasks", "get_task",
+                            "create_task", "update_task", "start_task"]},
+        "project_id": {"type": "string"},
+        "task_id": {"type": "string"},
+        "title": {"type": "string"},
+        "description": {"type": "string"},
+        "state": {"type": "string",
+                  "description": "target state (update_task)"}},
+        "required": ["action"]}
+
+    def __init__(self, spec_tasks, owner: str):
+        self.svc = spec_tasks
+        self.owner = owner
+
+    def _task(self, task_id: str):
+        doc = self.svc.get_task(task_id)
+        if doc is None or doc.get("owner") != self.owner:
+            raise ValueError(f"task not found: {task_id}")
+        return doc
+
+    async def execute(self, args: dict, ctx: dict) -> str:
+        action = args.get("action", "")
+        if action == "list_projects":
+            ps = self.svc.list_projects(self.owner)
+            return "\n".join(f"{p['id']}  {p.get('name', '')}"
+                             for p in ps) or "no projects"
+        if action == "list_tasks":
+            ts = self.svc.list_tasks(args.get("project_id", ""))
+            return "\n".join(
+                f"{t['id']}  [{t.get('state')}] {t.get('title', '')}"
+                for t in ts if t.get("owner") == self.owner) or "no tasks"
+        if action == "get_task":
+            t = self._task(args.get("task_id", ""))
+            return json.dumps({k: t.get(k) for k in
+                               ("id", "title", "state", "description",
+                                "spec", "branch", "verify")},
+                              default=str)
+        if action == "create_task":
+            t = self.svc.create_task(self.owner,
+                                     args.get("project_id", ""),
+                                     args.get("title", ""),
+                                     args.get("description", ""))
+            return f"created {t['id']} in backlog"
+        if action == "update_task":
+            self._task(args.get("task_id", ""))
+            t = self.svc.transition(args.get("task_id", ""),
+                                    args.get("state", ""))
+            return f"{t['id']} -> {t['state']}"
+        if action == "start_task":
+            self._task(args.get("task_id", ""))
+            t = await self.svc.plan(args.get("task_id", ""))
+            return f"{t['id']} planned -> {t['state']}"
+        return f"unknown action: {action}"
+
+
+class SandboxSkill(Skill):
+    """Run shell commands in the session's sandbox workspace
+    (reference: hydra dev-container exec surfaced to agents)."""
+
+    name = "run_command"
+    description = ("Run a shell command in your sandboxed workspace "
+                   "and get stdout/stderr back.")
+    parameters = {"type": "object", "properties": {
+        "command": {"type": "string"},
+        "timeout_s": {"type": "number"}},
+        "required": ["command"]}
+
+    def __init__(self, sandboxes, owner: str, session_id: str = ""):
+        self.sandboxes = sandboxes
+        self.owner = owner
+        self.session_id = session_id
+        self._sandbox_id: Optional[str] = None
+
+    def _ensure(self) -> str:
+        if self._sandbox_id:
+            doc = self.sandboxes.get(self._sandbox_id)
+            if doc is not None:
+                return self._sandbox_id
+        for doc in self.sandboxes.list(self.owner):
+            if doc.get("session_id") == self.session_id:
+                self._sandbox_id = doc["id"]
+                return doc["id"]
+        doc = self.sandboxes.create(self.owner, name="agent-workspace",
+                                    session_id=self.session_id)
+        self._sandbox_id = doc["id"]
+        return doc["id"]
+
+    async def execute(self, args: dict, ctx: dict) -> str:
+        import asyncio as _aio
+        sid = self._ensure()
+        timeout = min(float(args.get("timeout_s") or 60), 300)
+        r = await _aio.to_thread(self.sandboxes.exec, sid,
+                                 args.get("command", ""), timeout)
+        out = r["stdout"]
+        if r["stderr"]:
+            out += ("\n[stderr]\n" + r["stderr"])
+        if r["timed_out"]:
+            out += "\n[timed out]"
+        return f"exit={r['exit_code']}\n{out}"[:8000]

@@ -113,6 +113,9 @@ def create_app(cfg: Optional[ServerConfig] = None,
     app.state.sandboxes = sandboxes
     spec_tasks = SpecTaskService(store, controller, git_svc,
                                  sandboxes=sandboxes)
+    agent_runner.git = git_svc
+    agent_runner.spec_tasks = spec_tasks
+    agent_runner.sandboxes = sandboxes
     code_intel = CodeIntelService(rag, git_svc)
     from helix_amd.server.evaluations import EvaluationService
     evaluations = EvaluationService(store, controller, pubsub)

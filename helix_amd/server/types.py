@@ -88,6 +88,9 @@ class AssistantConfig(BaseModel):
     calculator: Dict[str, Any] = {}
     email: Dict[str, Any] = {}
     memory: Dict[str, Any] = {}
+    repository: Dict[str, Any] = {}
+    project: Dict[str, Any] = {}
+    sandbox: Dict[str, Any] = {}
     tests: List[AssistantTest] = []
     is_default: bool = False
 
