@@ -1172,8 +1172,8 @@ def create_app(cfg: Optional[ServerConfig] = None,
         areq = await request.json()
         # native passthrough when an Anthropic endpoint is configured:
         # raw body forwarded => prompt caching (cache_control) survives
-        base = os.environ.get("ANTHROPIC_BASE_URL", "")
-        akey = os.environ.get("ANTHROPIC_API_KEY", "")
+        base = os.environ.get("HELIX_ANTHROPIC_BASE_URL", "")
+        akey = os.environ.get("HELIX_ANTHROPIC_API_KEY", "")
         if not base:
             for ep in store.list("provider_endpoints", limit=1000):
                 if ep.get("provider") == "anthropic" and \

@@ -116,8 +116,14 @@ def s4_inference(http, model):
         "messages": [{"role": "user", "content": "Say hello."}],
         "max_tokens": 16}, timeout=180)
     r.raise_for_status()
-    text = r.json()["choices"][0]["message"]["content"]
-    assert isinstance(text, str) and len(text) > 0
+    out = r.json()
+    # random-init weights at full vocab can sample ids the byte-level
+    # tokenizer decodes to '' — assert the serving contract (structure
+    # + generated tokens), not linguistic output
+    msg = out["choices"][0]["message"]
+    assert msg["role"] == "assistant" and isinstance(
+        msg.get("content", ""), str)
+    assert out.get("usage", {}).get("completion_tokens", 0) > 0, out
     r = http.post(API + "/v1/embeddings", headers=ADMIN, json={
         "model": "bge-base", "input": ["hello world"]}, timeout=180)
     r.raise_for_status()

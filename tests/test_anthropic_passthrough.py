@@ -90,8 +90,8 @@ def test_route_uses_passthrough_when_configured(tmp_path, monkeypatch):
     from fastapi.testclient import TestClient
     from helix_amd.server.app import create_app
     from helix_amd.server.config import load_config
-    monkeypatch.setenv("ANTHROPIC_BASE_URL", "https://api.anthropic.test")
-    monkeypatch.setenv("ANTHROPIC_API_KEY", "sk-prov")
+    monkeypatch.setenv("HELIX_ANTHROPIC_BASE_URL", "https://api.anthropic.test")
+    monkeypatch.setenv("HELIX_ANTHROPIC_API_KEY", "sk-prov")
     cfg = load_config()
     cfg.store.path = str(tmp_path / "db.sqlite")
     cfg.filestore.path = str(tmp_path / "fs")
@@ -119,7 +119,7 @@ def test_route_translates_without_endpoint(tmp_path, monkeypatch):
     from fastapi.testclient import TestClient
     from helix_amd.server.app import create_app
     from helix_amd.server.config import load_config
-    monkeypatch.delenv("ANTHROPIC_BASE_URL", raising=False)
+    monkeypatch.delenv("HELIX_ANTHROPIC_BASE_URL", raising=False)
     from helix_amd.server.providers import MockClient, ProviderManager
     from helix_amd.store import Store
     cfg = load_config()
