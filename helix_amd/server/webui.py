@@ -45,6 +45,9 @@ INDEX_HTML = """<!doctype html>
   <button data-p="apps">Apps</button>
   <button data-p="knw">Knowledge</button>
   <button data-p="run">Runners</button>
+  <button data-p="prj">Projects</button>
+  <button data-p="org">Orgs</button>
+  <button data-p="sbx">Sandboxes</button>
   <button data-p="use">Usage</button>
  </div>
  <button onclick="newSession()">+ new session</button>
@@ -81,6 +84,35 @@ INDEX_HTML = """<!doctype html>
   <div id="knwlist"></div>
  </div></div>
  <div class="pane" id="p-run"><div class="lst" id="runlist"></div></div>
+ <div class="pane" id="p-prj"><div class="lst">
+  <div class="card"><b>Projects</b>
+   <input id="pname" placeholder="new project name">
+   <button onclick="newProject()">create</button>
+   <select id="psel" onchange="loadBoard()"></select>
+   <input id="tname" placeholder="new task title">
+   <button onclick="newTask()">add task</button>
+  </div>
+  <div id="board" style="display:flex;gap:8px;align-items:flex-start">
+  </div>
+ </div></div>
+ <div class="pane" id="p-org"><div class="lst">
+  <div class="card"><b>Organizations</b>
+   <input id="oname" placeholder="new org name">
+   <button onclick="newOrg()">create</button>
+  </div>
+  <div id="orglist"></div>
+ </div></div>
+ <div class="pane" id="p-sbx"><div class="lst">
+  <div class="card"><b>Sandboxes</b>
+   <button onclick="newSandbox()">new sandbox</button>
+   <select id="ssel"></select>
+   <input id="scmd" size="40" placeholder="command"
+    onkeydown="if(event.key==='Enter')runCmd()">
+   <button onclick="runCmd()">run</button>
+   <pre id="sout" style="background:#111;color:#9f9;padding:8px;
+     min-height:80px;white-space:pre-wrap"></pre>
+  </div>
+ </div></div>
  <div class="pane" id="p-use"><div class="lst" id="uselist"></div></div>
 </div>
 <script>
@@ -96,7 +128,8 @@ for(const b of document.querySelectorAll('#tabs button')){
     b.classList.add('on');
     $('p-'+b.dataset.p).classList.add('on');
     ({apps:loadApps, knw:loadKnowledge, run:loadRunners,
-      use:loadUsage})[b.dataset.p]?.();
+      use:loadUsage, prj:loadProjects, org:loadOrgs,
+      sbx:loadSandboxes})[b.dataset.p]?.();
   };
 }
 async function loadModels(){
@@ -217,6 +250,128 @@ async function loadUsage(){
   const u=await r.json();
   el.innerHTML='<div class="card"><pre>'+
     JSON.stringify(u,null,2).slice(0,4000)+'</pre></div>';
+}
+// ---- projects / kanban pane (reference frontend kanban board) ----
+const STATES=['backlog','planning','spec_review','in_progress','pr',
+              'merged','failed'];
+async function loadProjects(){
+  const r=await fetch('/api/v1/projects',{headers:H()});if(!r.ok)return;
+  const sel=$('psel');const cur=sel.value;sel.innerHTML='';
+  for(const p of await r.json()){
+    const o=document.createElement('option');
+    o.value=p.id;o.textContent=p.name||p.id;sel.appendChild(o);}
+  if(cur)sel.value=cur;
+  loadBoard();
+}
+async function newProject(){
+  await fetch('/api/v1/projects',{method:'POST',headers:H(),
+    body:JSON.stringify({name:$('pname').value})});
+  $('pname').value='';loadProjects();
+}
+async function newTask(){
+  const pid=$('psel').value;if(!pid)return;
+  await fetch('/api/v1/projects/'+pid+'/tasks',{method:'POST',
+    headers:H(),body:JSON.stringify({title:$('tname').value})});
+  $('tname').value='';loadBoard();
+}
+async function moveTask(tid,st){
+  await fetch('/api/v1/spec-tasks/'+tid+'/transition',{method:'POST',
+    headers:H(),body:JSON.stringify({state:st})});
+  loadBoard();
+}
+async function planTask(tid){
+  await fetch('/api/v1/spec-tasks/'+tid+'/plan',{method:'POST',
+    headers:H()});loadBoard();
+}
+async function implementTask(tid){
+  await fetch('/api/v1/spec-tasks/'+tid+'/implement',{method:'POST',
+    headers:H()});loadBoard();
+}
+async function loadBoard(){
+  const pid=$('psel').value;const el=$('board');el.innerHTML='';
+  if(!pid)return;
+  const r=await fetch('/api/v1/projects/'+pid+'/tasks',{headers:H()});
+  if(!r.ok)return;
+  const tasks=await r.json();
+  for(const st of STATES){
+    const col=document.createElement('div');
+    col.style.cssText='flex:1;background:#eee;border-radius:8px;'+
+      'padding:6px;min-height:120px';
+    col.innerHTML='<b style="font-size:12px">'+st+'</b>';
+    for(const t of tasks.filter(x=>x.state===st)){
+      const c=document.createElement('div');c.className='card';
+      c.style.padding='6px';c.style.fontSize='12px';
+      let btns='';
+      if(st==='backlog')btns='<button onclick="planTask(\''+t.id+
+        '\')">plan</button>';
+      if(st==='spec_review')btns='<button onclick="implementTask(\''+
+        t.id+'\')">implement</button>';
+      c.innerHTML='<b>'+t.title+'</b><br>'+btns+
+        ' <select onchange="moveTask(\''+t.id+
+        '\',this.value)"><option>move...</option>'+
+        STATES.map(x=>'<option>'+x+'</option>').join('')+'</select>';
+      col.appendChild(c);}
+    el.appendChild(col);}
+}
+// ---- orgs pane (reference org admin) ----
+async function newOrg(){
+  await fetch('/api/v1/organizations',{method:'POST',headers:H(),
+    body:JSON.stringify({name:$('oname').value})});
+  $('oname').value='';loadOrgs();
+}
+async function loadOrgs(){
+  const r=await fetch('/api/v1/organizations',{headers:H()});
+  if(!r.ok)return;
+  const el=$('orglist');el.innerHTML='';
+  for(const o of await r.json()){
+    const d=document.createElement('div');d.className='card';
+    d.innerHTML='<b>'+(o.name||o.id)+'</b> <small>'+o.id+'</small>'+
+      '<div id="teams-'+o.id+'"></div>'+
+      '<input id="tn-'+o.id+'" placeholder="team name" size="12">'+
+      '<button onclick="newTeam(\''+o.id+'\')">add team</button> '+
+      '<input id="mu-'+o.id+'" placeholder="user id" size="16">'+
+      '<button onclick="addMember(\''+o.id+'\')">add member</button>';
+    el.appendChild(d);
+    fetch('/api/v1/organizations/'+o.id+'/teams',{headers:H()})
+      .then(r=>r.ok?r.json():[]).then(ts=>{
+        $('teams-'+o.id).textContent='teams: '+
+          (ts.map(t=>t.name).join(', ')||'none');});
+  }
+}
+async function newTeam(oid){
+  await fetch('/api/v1/organizations/'+oid+'/teams',{method:'POST',
+    headers:H(),body:JSON.stringify({name:$('tn-'+oid).value})});
+  loadOrgs();
+}
+async function addMember(oid){
+  await fetch('/api/v1/organizations/'+oid+'/members',{method:'POST',
+    headers:H(),body:JSON.stringify({user_id:$('mu-'+oid).value})});
+  loadOrgs();
+}
+// ---- sandboxes pane (reference dev-container exec) ----
+async function loadSandboxes(){
+  const r=await fetch('/api/v1/sandboxes',{headers:H()});if(!r.ok)return;
+  const sel=$('ssel');const cur=sel.value;sel.innerHTML='';
+  for(const s of await r.json()){
+    const o=document.createElement('option');
+    o.value=s.id;o.textContent=s.name||s.id;sel.appendChild(o);}
+  if(cur)sel.value=cur;
+}
+async function newSandbox(){
+  await fetch('/api/v1/sandboxes',{method:'POST',headers:H(),
+    body:JSON.stringify({name:'ui-'+Date.now()})});
+  loadSandboxes();
+}
+async function runCmd(){
+  const sid=$('ssel').value;if(!sid)return;
+  $('sout').textContent+='$ '+$('scmd').value+'\n';
+  const r=await fetch('/api/v1/sandboxes/'+sid+'/exec',{method:'POST',
+    headers:H(),body:JSON.stringify({command:$('scmd').value})});
+  if(r.ok){const o=await r.json();
+    $('sout').textContent+=o.stdout+(o.stderr||'')+
+      (o.exit_code?('[exit '+o.exit_code+']\n'):'');}
+  else $('sout').textContent+='[error '+r.status+']\n';
+  $('scmd').value='';
 }
 loadModels();loadSessions();
 </script></body></html>"""
