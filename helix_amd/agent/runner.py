@@ -15,7 +15,7 @@ from helix_amd.agent.skills import (APISkill, BrowserSkill, CalculatorSkill,
                                     EmailSkill, KnowledgeSkill, MemorySkill,
                                     ProjectSkill, RepositorySkill,
                                     SandboxSkill, Skill, WebSearchSkill,
-                                    build_mcp_skills)
+                                    ZapierSkill, build_mcp_skills)
 from helix_amd.server import pubsub as ps
 from helix_amd.server.types import AssistantConfig, new_id
 
@@ -108,6 +108,9 @@ class AgentRunner:
             user = self.store.get("users", owner) or {}
             skills.append(EmailSkill(self.notifications,
                                      user.get("email", "")))
+        for zc in assistant.zapier:
+            cfg2 = self._resolve_secrets(owner, dict(zc))
+            skills.append(ZapierSkill(cfg2))
         if self.git is not None and \
                 assistant.repository.get("enabled",
                                          bool(assistant.repository)):

@@ -593,3 +593,59 @@ class SandboxSkill(Skill):
         if r["timed_out"]:
             out += "\n[timed out]"
         return f"exit={r['exit_code']}\n{out}"[:8000]
+
+
+# ---------------------------------------------------------------------------
+class ZapierSkill(Skill):
+    """Zapier NLA tool (reference api/pkg/tools/zapier.go: exposed
+    actions listed and executed with natural-language instructions via
+    the user's NLA API key)."""
+
+    name = "zapier"
+    description = ("Run the user's Zapier actions (email, sheets, CRM "
+                   "...): action=list shows them, action=execute runs "
+                   "one with natural-language instructions.")
+    parameters = {"type": "object", "properties": {
+        "action": {"type": "string", "enum": ["list", "execute"]},
+        "action_id": {"type": "string"},
+        "instructions": {"type": "string"}},
+        "required": ["action"]}
+
+    BASE = "https://nla.zapier.com/api/v1"
+
+    def __init__(self, config: dict, http_client=None):
+        import httpx
+        self.api_key = (config or {}).get("api_key", "")
+        self._http = http_client or httpx.AsyncClient(timeout=60)
+
+    def _headers(self):
+        return {"X-API-Key": self.api_key}
+
+    async def execute(self, args: dict, ctx: dict) -> str:
+        if not self.api_key:
+            return "zapier error: no API key configured"
+        try:
+            if args.get("action") == "list":
+                r = await self._http.get(self.BASE + "/exposed/",
+                                         headers=self._headers())
+                if r.status_code != 200:
+                    return f"zapier error: HTTP {r.status_code}"
+                rows = r.json().get("results", [])
+                return "\n".join(f"{x.get('id')}: {x.get('description')}"
+                                 for x in rows) or "no exposed actions"
+            aid = args.get("action_id", "")
+            if not aid:
+                return "zapier error: action_id required for execute"
+            r = await self._http.post(
+                self.BASE + f"/exposed/{aid}/execute/",
+                headers=self._headers(),
+                json={"instructions": args.get("instructions", "")})
+            if r.status_code != 200:
+                return f"zapier error: HTTP {r.status_code}"
+            out = r.json()
+            if out.get("status") == "error":
+                return f"zapier error: {out.get('error', 'unknown')}"
+            return json.dumps(out.get("result",
+                                      out))[:4000]
+        except Exception as e:
+            return f"zapier error: {e}"

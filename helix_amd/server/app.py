@@ -1351,6 +1351,50 @@ def create_app(cfg: Optional[ServerConfig] = None,
             payload = {}
         return await triggers.fire(doc, payload)
 
+    @app.post("/api/v1/discord/interactions/{tid}")
+    async def discord_interactions(tid: str, request: Request):
+        """Discord interactions inbound (Ed25519-signed; reference
+        api/pkg/trigger/discord): PING->PONG + slash commands."""
+        doc = store.get("triggers", tid)
+        if doc is None or doc.get("kind") != "discord":
+            raise HTTPException(404, "discord trigger not found")
+        body = await request.body()
+        try:
+            return await triggers.handle_discord_event(
+                doc, body,
+                request.headers.get("X-Signature-Timestamp", ""),
+                request.headers.get("X-Signature-Ed25519", ""))
+        except PermissionError:
+            raise HTTPException(401, "bad discord signature")
+
+    @app.post("/api/v1/azure-devops/webhook/{tid}")
+    async def azure_devops_webhook(tid: str, request: Request):
+        """Azure DevOps service-hook inbound (basic-auth checked)."""
+        doc = store.get("triggers", tid)
+        if doc is None or doc.get("kind") != "azure_devops":
+            raise HTTPException(404, "azure devops trigger not found")
+        body = await request.body()
+        try:
+            return await triggers.handle_azure_devops_event(
+                doc, body, request.headers.get("Authorization", ""))
+        except PermissionError:
+            raise HTTPException(401, "bad azure devops auth")
+
+    @app.post("/api/v1/crisp/webhook/{tid}")
+    async def crisp_webhook(tid: str, request: Request):
+        """Crisp webhook inbound (HMAC-signed)."""
+        doc = store.get("triggers", tid)
+        if doc is None or doc.get("kind") != "crisp":
+            raise HTTPException(404, "crisp trigger not found")
+        body = await request.body()
+        try:
+            return await triggers.handle_crisp_event(
+                doc, body,
+                request.headers.get("X-Crisp-Request-Timestamp", ""),
+                request.headers.get("X-Crisp-Signature", ""))
+        except PermissionError:
+            raise HTTPException(401, "bad crisp signature")
+
     @app.post("/api/v1/slack/events/{tid}")
     async def slack_events(tid: str, request: Request):
         """Slack Events API inbound (reference api/pkg/trigger slack):

@@ -207,3 +207,112 @@ class SlackTeamsMixin:
 # attach to TriggerManager (kept as a separate block for readability)
 TriggerManager.handle_slack_event = SlackTeamsMixin.handle_slack_event
 TriggerManager.handle_teams_event = SlackTeamsMixin.handle_teams_event
+
+
+# -- inbound: Discord / Azure DevOps / Crisp (reference api/pkg/trigger
+#    discord, azure_devops, crisp — the round-1 gaps)
+
+def verify_discord_signature(public_key_hex: str, timestamp: str,
+                             body: bytes, signature_hex: str) -> bool:
+    """Discord interactions auth: Ed25519 over timestamp+body with the
+    application public key (X-Signature-Ed25519 / X-Signature-Timestamp)."""
+    from helix_amd.server.ed25519 import verify
+    try:
+        pk = bytes.fromhex(public_key_hex)
+        sig = bytes.fromhex(signature_hex or "")
+    except ValueError:
+        return False
+    return verify(timestamp.encode() + body, sig, pk)
+
+
+def verify_crisp_signature(secret: str, timestamp: str, body: bytes,
+                           signature: str) -> bool:
+    """Crisp webhook signing: HMAC-SHA256 of "[{ts};{body}]" hex
+    (X-Crisp-Signature / X-Crisp-Request-Timestamp)."""
+    base = b"[" + timestamp.encode() + b";" + body + b"]"
+    want = hmac.new(secret.encode(), base, hashlib.sha256).hexdigest()
+    return hmac.compare_digest(want, signature or "")
+
+
+class ChatPlatformMixin:
+    async def handle_discord_event(self, doc: dict, body: bytes,
+                                   timestamp: str,
+                                   signature: str) -> dict:
+        pk = doc.get("config", {}).get("public_key", "")
+        if not pk or not verify_discord_signature(pk, timestamp, body,
+                                                  signature):
+            raise PermissionError("bad discord signature")
+        try:
+            payload = json.loads(body or b"{}")
+        except Exception:
+            payload = {}
+        # type 1 = PING -> PONG (Discord's endpoint validation)
+        if payload.get("type") == 1:
+            return {"type": 1}
+        # type 2 = APPLICATION_COMMAND: run the agent
+        if payload.get("type") == 2:
+            data = payload.get("data", {})
+            opts = {o.get("name"): o.get("value")
+                    for o in data.get("options", [])}
+            text = opts.get("prompt") or opts.get("message") or \
+                data.get("name", "")
+            res = await self.fire(doc, {
+                "source": "discord",
+                "user": (payload.get("member", {}).get("user", {})
+                         or payload.get("user", {})).get("id"),
+                "channel": payload.get("channel_id"), "text": text})
+            return {"type": 4, "data": {
+                "content": f"Started session {res['session_id']}"}}
+        return {"type": 4, "data": {"content": "unsupported"}}
+
+    async def handle_azure_devops_event(self, doc: dict, body: bytes,
+                                        auth_header: str) -> dict:
+        """Service-hook webhook with basic auth (reference
+        trigger/azure_devops: PR/work-item events drive sessions)."""
+        want = doc.get("config", {}).get("basic_auth", "")
+        if want:
+            got = ""
+            if auth_header.startswith("Basic "):
+                try:
+                    got = base64.b64decode(auth_header[6:]).decode()
+                except Exception:
+                    got = ""
+            if not hmac.compare_digest(want, got):
+                raise PermissionError("bad azure devops auth")
+        try:
+            payload = json.loads(body or b"{}")
+        except Exception:
+            payload = {}
+        event = payload.get("eventType", "")
+        msg = (payload.get("message", {}) or {}).get("text", "")
+        res = await self.fire(doc, {
+            "source": "azure_devops", "event": event,
+            "text": msg or f"Azure DevOps event {event}",
+            "resource": payload.get("resource", {})})
+        return {"ok": True, **res}
+
+    async def handle_crisp_event(self, doc: dict, body: bytes,
+                                 timestamp: str, signature: str) -> dict:
+        secret = doc.get("config", {}).get("signing_secret", "")
+        if secret and not verify_crisp_signature(secret, timestamp,
+                                                 body, signature):
+            raise PermissionError("bad crisp signature")
+        try:
+            payload = json.loads(body or b"{}")
+        except Exception:
+            payload = {}
+        if payload.get("event") == "message:send":
+            data = payload.get("data", {})
+            if data.get("from") == "user":     # loop prevention
+                res = await self.fire(doc, {
+                    "source": "crisp",
+                    "session": data.get("session_id"),
+                    "text": data.get("content", "")})
+                return {"ok": True, **res}
+        return {"ok": True, "ignored": True}
+
+
+TriggerManager.handle_discord_event = ChatPlatformMixin.handle_discord_event
+TriggerManager.handle_azure_devops_event = \
+    ChatPlatformMixin.handle_azure_devops_event
+TriggerManager.handle_crisp_event = ChatPlatformMixin.handle_crisp_event
