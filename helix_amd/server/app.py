@@ -208,6 +208,27 @@ def create_app(cfg: Optional[ServerConfig] = None,
             if t:
                 t.cancel()
 
+    # error reporting (the reference wires Sentry; here unhandled
+    # exceptions land in the error_events table with a fingerprint for
+    # grouping, listed via /api/v1/admin/errors)
+    def record_error(where: str, exc: BaseException):
+        import hashlib as _h
+        import time as _t
+        import traceback
+        tb = traceback.format_exc()[-4000:]
+        fp = _h.sha1(f"{type(exc).__name__}:{where}".encode()
+                     ).hexdigest()[:16]
+        doc = store.get("error_events", fp) or {
+            "id": fp, "type": type(exc).__name__, "where": where,
+            "first_seen": _t.time(), "count": 0}
+        doc["count"] += 1
+        doc["last_seen"] = _t.time()
+        doc["message"] = str(exc)[:500]
+        doc["traceback"] = tb
+        store.put("error_events", fp, doc, buffered=True)
+
+    app.state.record_error = record_error
+
     @app.middleware("http")
     async def _prom_mw(request: Request, call_next):
         import time as _t
@@ -215,8 +236,9 @@ def create_app(cfg: Optional[ServerConfig] = None,
         try:
             resp = await call_next(request)
             status = resp.status_code
-        except Exception:
+        except Exception as exc:
             status = 500
+            record_error(request.url.path, exc)
             raise
         finally:
             route = request.scope.get("route")
@@ -1489,6 +1511,11 @@ def create_app(cfg: Optional[ServerConfig] = None,
     async def create_user(request: Request,
                           user: AuthUser = Depends(admin_dep)):
         body = await request.json()
+        from helix_amd.server.license import LicenseError
+        try:
+            app.state.licenses.check_seat()
+        except LicenseError as e:
+            raise HTTPException(402, str(e))
         u = auth.create_user(body["username"], bool(body.get("admin")))
         key = auth.create_api_key(u["id"])
         return {**u, "api_key": key}
@@ -1835,6 +1862,36 @@ def create_app(cfg: Optional[ServerConfig] = None,
         return {"ok": True, "path": path,
                 "bytes": os.path.getsize(path)}
 
+    # -- license (reference api/pkg/license: signed envelope, expiry,
+    #    revocation, seat limits; development mode without one) -------
+    from helix_amd.server.license import LicenseError, LicenseManager
+    _pub_hex = os.environ.get("HELIX_LICENSE_PUBKEY", "")
+    licenses = LicenseManager(
+        store, bytes.fromhex(_pub_hex) if _pub_hex else b"\x00" * 32)
+    app.state.licenses = licenses
+
+    @app.get("/api/v1/license")
+    async def license_status(user: AuthUser = Depends(admin_dep)):
+        licenses.revalidate()
+        return licenses.status()
+
+    @app.post("/api/v1/license")
+    async def install_license(request: Request,
+                              user: AuthUser = Depends(admin_dep)):
+        body = await request.json()
+        try:
+            lic = licenses.install(body.get("envelope", ""))
+        except LicenseError as e:
+            raise HTTPException(400, str(e))
+        return lic.to_dict()
+
+    @app.get("/api/v1/admin/errors")
+    async def list_errors(user: AuthUser = Depends(admin_dep)):
+        store.flush("error_events")
+        rows = store.list("error_events", limit=200)
+        rows.sort(key=lambda r: -r.get("last_seen", 0))
+        return rows
+
     @app.post("/api/v1/admin/janitor")
     async def run_janitor(request: Request,
                           user: AuthUser = Depends(admin_dep)):
@@ -1870,6 +1927,18 @@ def create_app(cfg: Optional[ServerConfig] = None,
                         pruned["interactions"] += 1
                     store.delete("sessions", doc["id"])
                     pruned["sessions"] += 1
+        # idle-sandbox GC (reference hydra handleGCReconcile)
+        sbx_hours = float(body.get("sandbox_idle_hours", 24))
+        pruned["sandboxes"] = 0
+        if sbx_hours > 0:
+            cut = _t.time() - sbx_hours * 3600
+            for doc in store.list("sandboxes", limit=100000):
+                last = max(doc.get("last_exec", 0),
+                           doc.get("created", 0))
+                if last < cut:
+                    sandboxes.delete(doc["id"])
+                    pruned["sandboxes"] += 1
+        licenses.revalidate()
         return pruned
 
     @app.get("/")

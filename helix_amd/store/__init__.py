@@ -35,7 +35,7 @@ _TABLES = [
     "triggers", "organizations", "teams", "memberships", "access_grants",
     "runner_profiles", "projects", "spec_tasks", "git_repositories",
     "evaluation_runs", "system_settings", "wallets", "transactions",
-    "rag_chunks", "rag_alias", "oidc_states", "billing_customers", "billing_events", "billing_invoices", "sandboxes", "bus_messages", "bus_consumers", "usage_rollups", "runner_assignments",
+    "rag_chunks", "rag_alias", "oidc_states", "billing_customers", "billing_events", "billing_invoices", "sandboxes", "bus_messages", "bus_consumers", "error_events", "usage_rollups", "runner_assignments",
     "org_positions", "org_bots", "org_streams", "org_messages",
 ]
 
