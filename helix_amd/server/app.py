@@ -1088,6 +1088,54 @@ def create_app(cfg: Optional[ServerConfig] = None,
         except ValueError as e:
             raise HTTPException(400, str(e))
 
+    @app.put("/api/v1/bots/{bid}/parents")
+    async def set_bot_parents(bid: str, request: Request,
+                              user: AuthUser = Depends(auth_dep)):
+        """Reporting lines (reference org_reporting_lines: bot -> its
+        managers, cycle-guarded DAG)."""
+        b = await request.json()
+        bot = org_rt.get_bot(bid)
+        if bot is None:
+            raise HTTPException(404, "bot not found")
+        _org_member(bot["org_id"], user)
+        try:
+            return org_rt.set_reporting_lines(
+                bid, list(b.get("parent_ids", [])))
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+
+    @app.delete("/api/v1/bots/{bid}")
+    async def delete_bot(bid: str, user: AuthUser = Depends(auth_dep)):
+        bot = org_rt.get_bot(bid)
+        if bot is None:
+            raise HTTPException(404, "bot not found")
+        _org_member(bot["org_id"], user)
+        return {"ok": org_rt.delete_bot_cascade(bid)}
+
+    @app.get("/api/v1/organizations/{oid}/chart")
+    async def org_chart(oid: str, user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        return org_rt.chart(oid)
+
+    @app.post("/api/v1/bots/{bid}/escalate")
+    async def bot_escalate(bid: str, request: Request,
+                           user: AuthUser = Depends(auth_dep)):
+        b = await request.json()
+        bot = org_rt.get_bot(bid)
+        if bot is None:
+            raise HTTPException(404, "bot not found")
+        _org_member(bot["org_id"], user)
+        try:
+            return await org_rt.escalate(bid, b.get("text", ""))
+        except ValueError as e:
+            raise HTTPException(400, str(e))
+
+    @app.get("/api/v1/organizations/{oid}/audit")
+    async def org_audit(oid: str, user: AuthUser = Depends(auth_dep)):
+        _org_member(oid, user)
+        store.flush("org_audit")
+        return org_rt.audit_log(oid)
+
     @app.post("/api/v1/organizations/{oid}/streams")
     async def create_stream(oid: str, request: Request,
                             user: AuthUser = Depends(auth_dep)):

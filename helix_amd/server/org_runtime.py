@@ -129,6 +129,19 @@ class OrgRuntime:
 
     async def _bot_turn(self, bot: dict, stream: dict, text: str,
                         sender: str) -> str:
+        t0 = time.time()
+        try:
+            reply = await self._bot_turn_inner(bot, stream, text, sender)
+            self._audit(bot, stream.get("name", ""), sender,
+                        int((time.time() - t0) * 1000), True)
+            return reply
+        except Exception as e:
+            self._audit(bot, stream.get("name", ""), sender,
+                        int((time.time() - t0) * 1000), False, str(e))
+            raise
+
+    async def _bot_turn_inner(self, bot: dict, stream: dict, text: str,
+                              sender: str) -> str:
         pos = self.store.get("org_positions", bot["position_id"]) or {}
         prompt = (f"[stream {stream['name']}] message from {sender}: "
                   f"{text}")
@@ -149,3 +162,115 @@ class OrgRuntime:
         # store refetch: run_session_turn persists the interaction
         doc = self.store.get("interactions", interaction.id)
         return (doc or {}).get("response_message", "")
+
+
+# ---------------------------------------------------------------------------
+# Org-graph mechanics (reference QA.md mental model: Bots form a
+# cycle-guarded DAG of reporting lines; activations are audited;
+# messages can be escalated up the line).
+
+class OrgGraphMixin:
+    def set_reporting_lines(self, bot_id: str,
+                            parent_ids: List[str]) -> dict:
+        """Replace a bot's managers (org_reporting_lines role). The
+        graph must stay a DAG — adding a line that closes a cycle is
+        rejected (reference: cycle-guarded chart)."""
+        bot = self.store.get("org_bots", bot_id)
+        if bot is None:
+            raise ValueError("bot not found")
+        org = bot["org_id"]
+        for pid in parent_ids:
+            p = self.store.get("org_bots", pid)
+            if p is None or p["org_id"] != org:
+                raise ValueError(f"parent bot not in org: {pid}")
+            if pid == bot_id:
+                raise ValueError("a bot cannot report to itself")
+        # cycle guard: walk up from each proposed parent; reaching
+        # bot_id again means the new lines close a loop
+        def ancestors(start: str, seen: set):
+            if start in seen:
+                return
+            seen.add(start)
+            doc = self.store.get("org_bots", start) or {}
+            for pp in doc.get("parent_ids", []):
+                ancestors(pp, seen)
+        for pid in parent_ids:
+            seen: set = set()
+            ancestors(pid, seen)
+            if bot_id in seen:
+                raise ValueError(
+                    f"reporting line {pid} -> {bot_id} closes a cycle")
+        bot["parent_ids"] = list(dict.fromkeys(parent_ids))
+        self.store.put("org_bots", bot_id, bot,
+                       owner=bot.get("owner"), parent=org)
+        return bot
+
+    def chart(self, org_id: str) -> dict:
+        """Chart-tab data: bots as nodes, reporting lines as edges."""
+        bots = self.list_bots(org_id)
+        ids = {b["id"] for b in bots}
+        edges = []
+        for b in bots:
+            for pid in b.get("parent_ids", []):
+                if pid in ids:
+                    edges.append({"manager": pid, "report": b["id"]})
+        return {"nodes": [{"id": b["id"], "name": b["name"],
+                           "state": b.get("state", "idle"),
+                           "turns": b.get("turns", 0)} for b in bots],
+                "edges": edges}
+
+    def delete_bot_cascade(self, bot_id: str) -> bool:
+        """Bot deletion drops every reporting line referencing it
+        (reference ON DELETE CASCADE)."""
+        bot = self.store.get("org_bots", bot_id)
+        if bot is None:
+            return False
+        for other in self.list_bots(bot["org_id"]):
+            if bot_id in other.get("parent_ids", []):
+                other["parent_ids"] = [p for p in other["parent_ids"]
+                                       if p != bot_id]
+                self.store.put("org_bots", other["id"], other,
+                               owner=other.get("owner"),
+                               parent=other["org_id"])
+        return self.store.delete("org_bots", bot_id)
+
+    async def escalate(self, bot_id: str, text: str) -> List[dict]:
+        """Send a message up the reporting line: each manager gets an
+        activation and the replies come back (reference escalation
+        path in the worker loop)."""
+        bot = self.store.get("org_bots", bot_id)
+        if bot is None:
+            raise ValueError("bot not found")
+        out = []
+        for pid in bot.get("parent_ids", []):
+            mgr = self.store.get("org_bots", pid)
+            if mgr is None:
+                continue
+            pseudo_stream = {"org_id": bot["org_id"],
+                             "name": f"escalation from {bot['name']}"}
+            try:
+                reply = await self._bot_turn(mgr, pseudo_stream, text,
+                                             bot["name"])
+            except Exception as e:
+                reply = f"error: {e}"
+            out.append({"manager": mgr["name"], "reply": reply})
+        return out
+
+    def audit_log(self, org_id: str, limit: int = 100) -> List[dict]:
+        rows = self.store.list("org_audit", parent=org_id, limit=limit)
+        return sorted(rows, key=lambda r: -r.get("ts", 0))
+
+    def _audit(self, bot: dict, trigger: str, sender: str,
+               duration_ms: int, ok: bool, detail: str = ""):
+        aid = new_id("orgaud")
+        self.store.put("org_audit", aid, {
+            "id": aid, "org_id": bot["org_id"], "bot_id": bot["id"],
+            "bot_name": bot.get("name", ""), "trigger": trigger,
+            "sender": sender, "duration_ms": duration_ms, "ok": ok,
+            "detail": detail[:500], "ts": time.time()},
+            parent=bot["org_id"], buffered=True)
+
+
+for _name in ("set_reporting_lines", "chart", "delete_bot_cascade",
+              "escalate", "audit_log", "_audit"):
+    setattr(OrgRuntime, _name, getattr(OrgGraphMixin, _name))

@@ -36,7 +36,7 @@ _TABLES = [
     "runner_profiles", "projects", "spec_tasks", "git_repositories",
     "evaluation_runs", "system_settings", "wallets", "transactions",
     "rag_chunks", "rag_alias", "oidc_states", "billing_customers", "billing_events", "billing_invoices", "sandboxes", "bus_messages", "bus_consumers", "error_events", "usage_rollups", "runner_assignments",
-    "org_positions", "org_bots", "org_streams", "org_messages",
+    "org_positions", "org_bots", "org_streams", "org_messages", "org_audit",
 ]
 
 _FLUSH_INTERVAL = 0.2   # reference accumulator cadence (200 ms)

@@ -142,3 +142,67 @@ def test_org_primitives_via_mcp(stack):
         "stream_id": strm["id"]}})
     hist = _json.loads(r["result"]["content"][0]["text"])
     assert hist[0]["text"] == "hello from mcp"
+
+
+def _mk_bot(client, key, oid, name):
+    pos = client.post(f"/api/v1/organizations/{oid}/positions",
+                      json={"name": f"pos-{name}"},
+                      headers=H(key)).json()
+    return client.post(f"/api/v1/organizations/{oid}/bots",
+                       json={"name": name, "position_id": pos["id"]},
+                       headers=H(key)).json()
+
+
+def test_reporting_lines_dag_and_chart(stack):
+    """Cycle-guarded reporting-line DAG + chart view (reference
+    org QA.md mental model: org_reporting_lines, ReactFlow chart)."""
+    client, key, _ = stack
+    oid = _mk_org(client, key)
+    root = _mk_bot(client, key, oid, "b-root")
+    eng = _mk_bot(client, key, oid, "b-eng")
+    dev = _mk_bot(client, key, oid, "b-dev")
+    # dev -> eng -> root
+    r = client.put(f"/api/v1/bots/{eng['id']}/parents",
+                   json={"parent_ids": [root["id"]]}, headers=H(key))
+    assert r.status_code == 200
+    client.put(f"/api/v1/bots/{dev['id']}/parents",
+               json={"parent_ids": [eng["id"]]}, headers=H(key))
+    # closing the cycle root -> dev is rejected
+    r = client.put(f"/api/v1/bots/{root['id']}/parents",
+                   json={"parent_ids": [dev["id"]]}, headers=H(key))
+    assert r.status_code == 400 and "cycle" in r.json()["detail"]
+    # self-report rejected
+    r = client.put(f"/api/v1/bots/{root['id']}/parents",
+                   json={"parent_ids": [root["id"]]}, headers=H(key))
+    assert r.status_code == 400
+    chart = client.get(f"/api/v1/organizations/{oid}/chart",
+                       headers=H(key)).json()
+    assert len(chart["nodes"]) == 3
+    assert {"manager": root["id"], "report": eng["id"]} in chart["edges"]
+    assert {"manager": eng["id"], "report": dev["id"]} in chart["edges"]
+    # deleting eng cascades its lines (dev loses its manager)
+    client.delete(f"/api/v1/bots/{eng['id']}", headers=H(key))
+    chart = client.get(f"/api/v1/organizations/{oid}/chart",
+                       headers=H(key)).json()
+    assert len(chart["nodes"]) == 2 and chart["edges"] == []
+
+
+def test_escalation_and_audit(stack):
+    client, key, store = stack
+    oid = _mk_org(client, key)
+    mgr = _mk_bot(client, key, oid, "b-mgr")
+    worker = _mk_bot(client, key, oid, "b-worker")
+    client.put(f"/api/v1/bots/{worker['id']}/parents",
+               json={"parent_ids": [mgr["id"]]}, headers=H(key))
+    r = client.post(f"/api/v1/bots/{worker['id']}/escalate",
+                    json={"text": "prod is down"}, headers=H(key))
+    assert r.status_code == 200, r.text
+    out = r.json()
+    assert out and out[0]["manager"] == "b-mgr"
+    assert out[0]["reply"]              # mock model replied
+    # activation landed in the audit log
+    audit = client.get(f"/api/v1/organizations/{oid}/audit",
+                       headers=H(key)).json()
+    assert audit and audit[0]["bot_name"] == "b-mgr"
+    assert audit[0]["ok"] is True
+    assert audit[0]["duration_ms"] >= 0
