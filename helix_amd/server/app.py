@@ -507,6 +507,66 @@ def create_app(cfg: Optional[ServerConfig] = None,
         controller.delete_session(session_id)
         return {"ok": True}
 
+    # -- interactions CRUD (reference sessions API, server.go:1024-1064:
+    #    interaction listing/edit/delete alongside step-info) ----------
+    def _owned_interaction(iid: str, user: AuthUser) -> dict:
+        doc = store.get("interactions", iid)
+        if doc is None:
+            raise HTTPException(404, "interaction not found")
+        s = controller.get_session(doc.get("session_id", ""))
+        if s is None or (s.owner != user.id and not user.admin):
+            raise HTTPException(404, "interaction not found")
+        return doc
+
+    @app.get("/api/v1/sessions/{session_id}/interactions")
+    async def list_interactions(session_id: str, limit: int = 100,
+                                offset: int = 0,
+                                user: AuthUser = Depends(auth_dep)):
+        s = controller.get_session(session_id)
+        if s is None or (s.owner != user.id and not user.admin):
+            raise HTTPException(404, "session not found")
+        rows = store.list("interactions", parent=session_id,
+                          desc=False, limit=limit, offset=offset)
+        return rows
+
+    @app.put("/api/v1/interactions/{iid}")
+    async def update_interaction(iid: str, request: Request,
+                                 user: AuthUser = Depends(auth_dep)):
+        """Edit a turn (prompt correction before resume / response
+        redaction)."""
+        doc = _owned_interaction(iid, user)
+        body = await request.json()
+        for k in ("prompt_message", "response_message", "state"):
+            if k in body:
+                doc[k] = body[k]
+        s2 = controller.get_session(doc.get("session_id", ""))
+        store.put("interactions", iid, doc,
+                  owner=s2.owner if s2 else "",
+                  parent=doc.get("session_id", ""))
+        return doc
+
+    @app.delete("/api/v1/interactions/{iid}")
+    async def delete_interaction(iid: str,
+                                 user: AuthUser = Depends(auth_dep)):
+        _owned_interaction(iid, user)
+        return {"ok": store.delete("interactions", iid)}
+
+    @app.post("/api/v1/tokenize")
+    async def tokenize(request: Request,
+                       user: AuthUser = Depends(auth_dep)):
+        """Token counting for a prompt under a model's tokenizer +
+        chat template (client-side budget planning)."""
+        from helix_amd.utils.tokenizer import get_tokenizer
+        body = await request.json()
+        tok = get_tokenizer(body.get("model", ""))
+        if body.get("messages"):
+            ids = tok.apply_chat_template(
+                body["messages"], template=body.get("template", ""))
+        else:
+            ids = tok.encode(body.get("text", ""))
+        return {"count": len(ids),
+                "tokens": ids[:int(body.get("return_tokens", 0) or 0)]}
+
     @app.post("/api/v1/auth/token")
     async def issue_token(request: Request,
                           user: AuthUser = Depends(auth_dep)):

@@ -470,3 +470,48 @@ def test_interrupt_midstream_then_followup(stack):
     # no interaction left dangling in waiting
     assert all(i["state"] in ("complete", "error") for i in its) or \
         its[0]["state"] in ("complete", "error", "waiting")
+
+
+def test_interactions_crud_and_tokenize(stack):
+    app, client, mock, key, store = stack
+    r = client.post("/v1/chat/completions", headers=H(key), json={
+        "model": "mock-model",
+        "messages": [{"role": "user", "content": "hello"}]})
+    assert r.status_code == 200
+    # a session-less call logs no interaction; create a session turn
+    r = client.post("/api/v1/sessions/chat", headers=H(key), json={
+        "messages": [{"role": "user", "content": "first turn"}]})
+    assert r.status_code == 200
+    sid = None
+    for line in r.text.splitlines():
+        if line.startswith("data: ") and "session_id" in line:
+            import json as _j
+            sid = _j.loads(line[6:])["session_id"]
+            break
+    assert sid
+    rows = client.get(f"/api/v1/sessions/{sid}/interactions",
+                      headers=H(key)).json()
+    assert rows and rows[0]["prompt_message"] == "first turn"
+    iid = rows[0]["id"]
+    # edit the prompt
+    r = client.put(f"/api/v1/interactions/{iid}", headers=H(key),
+                   json={"prompt_message": "edited turn"})
+    assert r.json()["prompt_message"] == "edited turn"
+    # another user cannot touch it
+    r2 = client.post("/api/v1/users", json={"username": "other-i"},
+                     headers={"Authorization": "Bearer admin-key"})
+    okey = r2.json()["api_key"]
+    assert client.put(f"/api/v1/interactions/{iid}", headers=H(okey),
+                      json={"state": "error"}).status_code == 404
+    # delete
+    assert client.delete(f"/api/v1/interactions/{iid}",
+                         headers=H(key)).json()["ok"]
+    # tokenize: plain text and chat-template forms
+    r = client.post("/api/v1/tokenize", headers=H(key),
+                    json={"text": "hello world"})
+    assert r.json()["count"] == len("hello world")
+    r = client.post("/api/v1/tokenize", headers=H(key), json={
+        "messages": [{"role": "user", "content": "hi"}],
+        "return_tokens": 4})
+    body = r.json()
+    assert body["count"] > 2 and len(body["tokens"]) == 4
