@@ -38,6 +38,7 @@ class AgentRunner:
         self.git = None
         self.spec_tasks = None
         self.sandboxes = None
+        self.external_agents = None
 
     def _small_llm(self, assistant: AssistantConfig, owner: str):
         """Async messages->str callable on the small generation model
@@ -304,8 +305,29 @@ class AgentRunner:
         return messages, {"model": model, "provider": provider}
 
     # ------------------------------------------------------------------
+    def _external_key(self, ctx: dict) -> str:
+        return ctx.get("session_id") or ctx.get("app_id") or "default"
+
+    def _external_payload(self, assistant, req, ctx) -> dict:
+        return {"messages": req.get("messages", []),
+                "model": assistant.model or req.get("model", ""),
+                "session_id": ctx.get("session_id", ""),
+                "interaction_id": ctx.get("interaction_id", "")}
+
     async def run_blocking(self, assistant: AssistantConfig, req: dict,
                            owner: str, ctx: dict) -> dict:
+        if assistant.agent_type == "zed_external" and \
+                self.external_agents is not None:
+            # turns execute on the externally connected agent
+            # (reference controller_external_agent.go RunExternalAgent)
+            from helix_amd.server.external_agent import ExternalAgentError
+            try:
+                text = await self.external_agents.run_turn_blocking(
+                    self._external_key(ctx),
+                    self._external_payload(assistant, req, ctx))
+            except ExternalAgentError as e:
+                text = f"[external agent error] {e}"
+            return _completion_dict(req.get("model", ""), text)
         messages, final = await self._loop(assistant, req, owner, ctx)
         if final is None and messages and messages[-1]["role"] == "assistant":
             content = messages[-1].get("content", "")
@@ -324,6 +346,33 @@ class AgentRunner:
 
     async def run_stream(self, assistant: AssistantConfig, req: dict,
                          owner: str, ctx: dict) -> AsyncIterator[dict]:
+        if assistant.agent_type == "zed_external" and \
+                self.external_agents is not None:
+            from helix_amd.server.external_agent import ExternalAgentError
+            base_id = new_id("chatcmpl")
+            created = int(time.time())
+            model = req.get("model", "")
+            try:
+                async for chunk in self.external_agents.run_turn(
+                        self._external_key(ctx),
+                        self._external_payload(assistant, req, ctx)):
+                    yield {"id": base_id,
+                           "object": "chat.completion.chunk",
+                           "created": created, "model": model,
+                           "choices": [{"index": 0, "delta": {
+                               "content": chunk},
+                               "finish_reason": None}]}
+            except ExternalAgentError as e:
+                yield {"id": base_id, "object": "chat.completion.chunk",
+                       "created": created, "model": model,
+                       "choices": [{"index": 0, "delta": {
+                           "content": f"[external agent error] {e}"},
+                           "finish_reason": None}]}
+            yield {"id": base_id, "object": "chat.completion.chunk",
+                   "created": created, "model": model,
+                   "choices": [{"index": 0, "delta": {},
+                                "finish_reason": "stop"}]}
+            return
         messages, final = await self._loop(assistant, req, owner, ctx)
         if final is None and messages and messages[-1]["role"] == "assistant":
             content = messages[-1].get("content", "")

@@ -116,6 +116,10 @@ def create_app(cfg: Optional[ServerConfig] = None,
     agent_runner.git = git_svc
     agent_runner.spec_tasks = spec_tasks
     agent_runner.sandboxes = sandboxes
+    from helix_amd.server.external_agent import ExternalAgentRegistry
+    external_agents = ExternalAgentRegistry()
+    app.state.external_agents = external_agents
+    agent_runner.external_agents = external_agents
     code_intel = CodeIntelService(rag, git_svc)
     from helix_amd.server.evaluations import EvaluationService
     evaluations = EvaluationService(store, controller, pubsub)
@@ -666,6 +670,43 @@ def create_app(cfg: Optional[ServerConfig] = None,
     from helix_amd.server.pubsub import StreamBus
     bus = StreamBus(store, pubsub)
     app.state.bus = bus
+
+    @app.websocket("/api/v1/external-agents/ws")
+    async def external_agent_ws(ws: WebSocket):
+        """External agent uplink (reference external-agent executor):
+        the agent connects with its agent_id (session or app scope),
+        receives chat commands, streams delta/done/error frames back."""
+        token = ws.query_params.get("access_token", "")
+        agent_id = ws.query_params.get("agent_id", "")
+        user = auth.resolve(token)
+        is_runner = token and token == cfg.runner_plane.runner_token
+        if (user is None and not is_runner) or not agent_id:
+            await ws.close(code=4401)
+            return
+        await ws.accept()
+        outbox = external_agents.attach(agent_id)
+
+        async def pump_out():
+            try:
+                while True:
+                    frame = await outbox.get()
+                    await ws.send_json(frame)
+            except Exception:
+                pass
+
+        out_task = asyncio.ensure_future(pump_out())
+        try:
+            while True:
+                frame = await ws.receive_json()
+                if frame.get("type") == "ping":
+                    await ws.send_json({"type": "pong"})
+                    continue
+                external_agents.deliver(frame)
+        except Exception:
+            pass
+        finally:
+            out_task.cancel()
+            external_agents.detach(agent_id)
 
     @app.websocket("/api/v1/ws/bus")
     async def ws_bus(ws: WebSocket):
