@@ -215,6 +215,274 @@ __global__ __launch_bounds__(256) void sample_topkp_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Partitioned variant (round 2). The one-block-per-row kernel above
+// underfills the chip at serving batch sizes (B=512 -> 512 workgroups
+// on 256 CUs x 4 SIMDs) and serializes three full-vocab passes per
+// block: measured 843 us/step at B=512, V=128k (profiles). Splitting
+// each row across VP=8 partition blocks (grid B x 8 = 4096 wgs) runs
+// the same three passes chip-filled:
+//   K1 partial (max, sumexp) per (row, partition) + zero the row's
+//      global histogram slice
+//   K2 combine partials (redundant, 8 floats) + LDS histogram of the
+//      partition, sparse-flushed to the row's global histogram
+//   K3 combine + serial 2048-bin threshold scan (redundant) + partial
+//      Gumbel-argmax over the partition
+//   K4 combine the 8 partition winners (lowest index wins ties)
+// Non-filtered rows (greedy / no top-k/p) skip K2's histogram and
+// K3's scan and just do the partitioned argmax.
+
+constexpr int VP = 8;              // vocab partitions per row
+
+template <typename T>
+__global__ __launch_bounds__(256) void topkp_partial_kernel(
+    const T* __restrict__ logits, const float* __restrict__ temperatures,
+    const float* __restrict__ top_p, const int* __restrict__ top_k,
+    const float* __restrict__ rep_pen, const float* __restrict__ pres_pen,
+    const float* __restrict__ freq_pen, const int* __restrict__ counts,
+    const uint8_t* __restrict__ seen, const int* __restrict__ row_map,
+    float* __restrict__ part_mz,       // [B, VP, 2]
+    float* __restrict__ hist_mass,     // [B, BINS]
+    int* __restrict__ hist_cnt,        // [B, BINS]
+    int V) {
+  const int row = blockIdx.x, part = blockIdx.y;
+  // zero this block's slice of the row's histogram (K2 depends on it)
+  const int bins_per = BINS / VP;
+  for (int b = part * bins_per + threadIdx.x;
+       b < (part + 1) * bins_per; b += blockDim.x) {
+    hist_mass[(int64_t)row * BINS + b] = 0.f;
+    hist_cnt[(int64_t)row * BINS + b] = 0;
+  }
+  const float temp = temperatures[row];
+  const bool filter = temp > 0.f && (top_p[row] < 1.f || top_k[row] > 0);
+  if (!filter) return;
+  PenaltyCtx pen{nullptr, nullptr, 1.f, 0.f, 0.f, false};
+  const int slot = row_map ? row_map[row] : -1;
+  if (slot >= 0 && counts != nullptr) {
+    pen.rep = rep_pen[row]; pen.pres = pres_pen[row];
+    pen.freq = freq_pen[row];
+    pen.active = pen.rep != 1.f || pen.pres != 0.f || pen.freq != 0.f;
+    if (pen.active) {
+      pen.cnt = counts + (int64_t)slot * V;
+      pen.seen = seen + (int64_t)slot * V;
+    }
+  }
+  const T* lrow = logits + (int64_t)row * V;
+  const int chunk = (V + VP - 1) / VP;
+  const int lo = part * chunk, hi = min(V, lo + chunk);
+  float m = -INFINITY, z = 0.f;
+  for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    const float v = adjust(raw_val(lrow, i), i, pen);
+    if (v > m) { z = z * __expf(m - v) + 1.f; m = v; }
+    else z += __expf(v - m);
+  }
+  __shared__ float red_m[4], red_z[4];
+  const int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float om = __shfl_xor(m, off, WAVE);
+    const float oz = __shfl_xor(z, off, WAVE);
+    const float nm = fmaxf(m, om);
+    // -inf lanes (no elements) contribute 0, not NaN (inf - inf)
+    const float sa = (m == -INFINITY) ? 0.f : __expf(m - nm);
+    const float sb = (om == -INFINITY) ? 0.f : __expf(om - nm);
+    z = z * sa + oz * sb;
+    m = nm;
+  }
+  if (lane == 0) { red_m[wid] = m; red_z[wid] = z; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+#pragma unroll
+    for (int w = 1; w < 4; ++w) {
+      const float om = red_m[w], oz = red_z[w];
+      const float nm = fmaxf(m, om);
+      const float sa = (m == -INFINITY) ? 0.f : __expf(m - nm);
+      const float sb = (om == -INFINITY) ? 0.f : __expf(om - nm);
+      z = z * sa + oz * sb;
+      m = nm;
+    }
+    part_mz[((int64_t)row * VP + part) * 2] = m;
+    part_mz[((int64_t)row * VP + part) * 2 + 1] = z;
+  }
+}
+
+__device__ __forceinline__ void combine_mz(const float* part_mz, int row,
+                                           float& M, float& Z) {
+  M = -INFINITY; Z = 0.f;
+#pragma unroll
+  for (int p = 0; p < VP; ++p) {
+    const float m = part_mz[((int64_t)row * VP + p) * 2];
+    const float z = part_mz[((int64_t)row * VP + p) * 2 + 1];
+    const float nm = fmaxf(M, m);
+    const float sa = (M == -INFINITY) ? 0.f : __expf(M - nm);
+    const float sb = (m == -INFINITY) ? 0.f : __expf(m - nm);
+    Z = Z * sa + z * sb;
+    M = nm;
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void topkp_hist_kernel(
+    const T* __restrict__ logits, const float* __restrict__ temperatures,
+    const float* __restrict__ top_p, const int* __restrict__ top_k,
+    const float* __restrict__ rep_pen, const float* __restrict__ pres_pen,
+    const float* __restrict__ freq_pen, const int* __restrict__ counts,
+    const uint8_t* __restrict__ seen, const int* __restrict__ row_map,
+    const float* __restrict__ part_mz, float* __restrict__ hist_mass,
+    int* __restrict__ hist_cnt, int V) {
+  const int row = blockIdx.x, part = blockIdx.y;
+  const float temp = temperatures[row];
+  const bool filter = temp > 0.f && (top_p[row] < 1.f || top_k[row] > 0);
+  if (!filter) return;
+  PenaltyCtx pen{nullptr, nullptr, 1.f, 0.f, 0.f, false};
+  const int slot = row_map ? row_map[row] : -1;
+  if (slot >= 0 && counts != nullptr) {
+    pen.rep = rep_pen[row]; pen.pres = pres_pen[row];
+    pen.freq = freq_pen[row];
+    pen.active = pen.rep != 1.f || pen.pres != 0.f || pen.freq != 0.f;
+    if (pen.active) {
+      pen.cnt = counts + (int64_t)slot * V;
+      pen.seen = seen + (int64_t)slot * V;
+    }
+  }
+  float M, Z;
+  combine_mz(part_mz, row, M, Z);
+  __shared__ float h_mass[BINS];
+  __shared__ int h_cnt[BINS];
+  for (int b = threadIdx.x; b < BINS; b += blockDim.x) {
+    h_mass[b] = 0.f; h_cnt[b] = 0;
+  }
+  __syncthreads();
+  const T* lrow = logits + (int64_t)row * V;
+  const int chunk = (V + VP - 1) / VP;
+  const int lo = part * chunk, hi = min(V, lo + chunk);
+  const float scale_b = (float)BINS / RANGE;
+  for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    const float v = adjust(raw_val(lrow, i), i, pen);
+    int b = (int)((M - v) * scale_b);
+    b = max(0, min(BINS - 1, b));
+    atomicAdd(&h_mass[b], __expf(v - M));
+    atomicAdd(&h_cnt[b], 1);
+  }
+  __syncthreads();
+  // sparse flush: logits concentrate in few bins
+  for (int b = threadIdx.x; b < BINS; b += blockDim.x) {
+    if (h_cnt[b]) {
+      atomicAdd(&hist_mass[(int64_t)row * BINS + b], h_mass[b]);
+      atomicAdd(&hist_cnt[(int64_t)row * BINS + b], h_cnt[b]);
+    }
+  }
+}
+
+template <typename T>
+__global__ __launch_bounds__(256) void topkp_argmax_kernel(
+    const T* __restrict__ logits, const float* __restrict__ temperatures,
+    const uint64_t* __restrict__ seeds, const float* __restrict__ top_p,
+    const int* __restrict__ top_k, const float* __restrict__ rep_pen,
+    const float* __restrict__ pres_pen, const float* __restrict__ freq_pen,
+    const int* __restrict__ counts, const uint8_t* __restrict__ seen,
+    const int* __restrict__ row_map, const float* __restrict__ part_mz,
+    const float* __restrict__ hist_mass, const int* __restrict__ hist_cnt,
+    float* __restrict__ win, int* __restrict__ win_i, int V) {
+  const int row = blockIdx.x, part = blockIdx.y;
+  const float temp = temperatures[row];
+  const uint64_t seed = seeds[row];
+  const bool greedy = temp <= 0.f;
+  const float inv_t = greedy ? 1.f : 1.f / temp;
+  const float p_p = top_p[row];
+  const int k_k = top_k[row];
+  const bool filter = !greedy && (p_p < 1.f || k_k > 0);
+  PenaltyCtx pen{nullptr, nullptr, 1.f, 0.f, 0.f, false};
+  const int slot = row_map ? row_map[row] : -1;
+  if (slot >= 0 && counts != nullptr) {
+    pen.rep = rep_pen[row]; pen.pres = pres_pen[row];
+    pen.freq = freq_pen[row];
+    pen.active = pen.rep != 1.f || pen.pres != 0.f || pen.freq != 0.f;
+    if (pen.active) {
+      pen.cnt = counts + (int64_t)slot * V;
+      pen.seen = seen + (int64_t)slot * V;
+    }
+  }
+  float thresh = -INFINITY;
+  if (filter) {
+    float M, Z;
+    combine_mz(part_mz, row, M, Z);
+    const float scale_b = (float)BINS / RANGE;
+    const float target = p_p * Z;
+    float cum = 0.f;
+    int cnt = 0;
+    int b_p = BINS - 1, b_k = BINS - 1;
+    bool got_p = (p_p >= 1.f), got_k = (k_k <= 0);
+    for (int b = 0; b < BINS; ++b) {
+      cum += hist_mass[(int64_t)row * BINS + b];
+      cnt += hist_cnt[(int64_t)row * BINS + b];
+      if (!got_p && cum >= target) { b_p = b; got_p = true; }
+      if (!got_k && cnt >= k_k) { b_k = b; got_k = true; }
+      if (got_p && got_k) break;
+    }
+    int b = BINS - 1;
+    if (p_p < 1.f) b = min(b, b_p);
+    if (k_k > 0) b = min(b, b_k);
+    thresh = (b >= BINS - 1) ? -INFINITY : M - (float)(b + 1) / scale_b;
+  }
+  const T* lrow = logits + (int64_t)row * V;
+  const int chunk = (V + VP - 1) / VP;
+  const int lo = part * chunk, hi = min(V, lo + chunk);
+  float best = -INFINITY;
+  int best_i = lo;
+  for (int i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    float v = adjust(raw_val(lrow, i), i, pen);
+    if (filter && !(v > thresh)) continue;
+    v *= inv_t;
+    if (!greedy) {
+      const float u = u64_to_uniform(splitmix64(seed ^ (uint64_t)i));
+      v += -__logf(-__logf(u));
+    }
+    if (v > best || (v == best && i < best_i)) { best = v; best_i = i; }
+  }
+  __shared__ float red_f[4];
+  __shared__ int red_i[4];
+  const int lane = threadIdx.x & (WAVE - 1), wid = threadIdx.x / WAVE;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ov = __shfl_xor(best, off, WAVE);
+    const int oi = __shfl_xor(best_i, off, WAVE);
+    if (ov > best || (ov == best && oi < best_i)) { best = ov; best_i = oi; }
+  }
+  if (lane == 0) { red_f[wid] = best; red_i[wid] = best_i; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+#pragma unroll
+    for (int w = 1; w < 4; ++w)
+      if (red_f[w] > best || (red_f[w] == best && red_i[w] < best_i)) {
+        best = red_f[w]; best_i = red_i[w];
+      }
+    win[(int64_t)row * VP + part] = best;
+    win_i[(int64_t)row * VP + part] = best_i;
+  }
+}
+
+__global__ __launch_bounds__(64) void topkp_final_kernel(
+    int64_t* __restrict__ out, const float* __restrict__ win,
+    const int* __restrict__ win_i) {
+  const int row = blockIdx.x;
+  if (threadIdx.x != 0) return;
+  float best = win[(int64_t)row * VP];
+  int best_i = win_i[(int64_t)row * VP];
+#pragma unroll
+  for (int p = 1; p < VP; ++p) {
+    const float v = win[(int64_t)row * VP + p];
+    const int i = win_i[(int64_t)row * VP + p];
+    if (v > best || (v == best && i < best_i)) { best = v; best_i = i; }
+  }
+  out[row] = best_i;
+}
+
+static int sampler_mono() {
+  const char* e = getenv("HELIX_SAMPLER_MONO");
+  return e ? atoi(e) : 0;
+}
+
 }  // namespace
 
 void sample_tokens_ext(torch::Tensor out, torch::Tensor logits,
@@ -241,26 +509,52 @@ void sample_tokens_ext(torch::Tensor out, torch::Tensor logits,
     map_ptr = row_map->data_ptr<int>();
   }
   auto stream = at::hip::getCurrentHIPStream();
+  if (sampler_mono()) {
+    // round-1 one-block-per-row kernel (A/B fallback)
+    if (logits.scalar_type() == torch::kBFloat16) {
+      hipLaunchKernelGGL((sample_topkp_kernel<uint16_t>), dim3(B), dim3(256),
+                         0, stream, out.data_ptr<int64_t>(),
+                         (const uint16_t*)logits.data_ptr(),
+                         temperatures.data_ptr<float>(),
+                         (const uint64_t*)seeds.data_ptr(),
+                         top_p.data_ptr<float>(), top_k.data_ptr<int>(),
+                         rep_pen.data_ptr<float>(), pres_pen.data_ptr<float>(),
+                         freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr,
+                         map_ptr, V);
+    } else {
+      TORCH_CHECK(logits.scalar_type() == torch::kFloat32);
+      hipLaunchKernelGGL((sample_topkp_kernel<float>), dim3(B), dim3(256),
+                         0, stream, out.data_ptr<int64_t>(),
+                         logits.data_ptr<float>(),
+                         temperatures.data_ptr<float>(),
+                         (const uint64_t*)seeds.data_ptr(),
+                         top_p.data_ptr<float>(), top_k.data_ptr<int>(),
+                         rep_pen.data_ptr<float>(), pres_pen.data_ptr<float>(),
+                         freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr,
+                         map_ptr, V);
+    }
+    return;
+  }
+  // partitioned path: chip-filled grid (B x VP); scratch comes from the
+  // torch caching allocator (stable sizes -> no real allocations after
+  // the first step)
+  auto opts_f = torch::TensorOptions().dtype(torch::kFloat32)
+                    .device(logits.device());
+  auto opts_i = torch::TensorOptions().dtype(torch::kInt32)
+                    .device(logits.device());
+  auto part_mz = torch::empty({B, VP, 2}, opts_f);
+  auto hist_mass = torch::empty({B, BINS}, opts_f);
+  auto hist_cnt = torch::empty({B, BINS}, opts_i);
+  auto win = torch::empty({B, VP}, opts_f);
+  auto win_i = torch::empty({B, VP}, opts_i);
+#define SAMP_ARGS(T)                                                              (const T*)logits.data_ptr(), temperatures.data_ptr<float>(),                top_p.data_ptr<float>(), top_k.data_ptr<int>(),                             rep_pen.data_ptr<float>(), pres_pen.data_ptr<float>(),                      freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr, map_ptr
+#define LAUNCH_SAMP(T)                                                        do {                                                                          hipLaunchKernelGGL((topkp_partial_kernel<T>), dim3(B, VP), dim3(256),                          0, stream, SAMP_ARGS(T),                                                    part_mz.data_ptr<float>(),                                                  hist_mass.data_ptr<float>(),                                                hist_cnt.data_ptr<int>(), V);                            hipLaunchKernelGGL((topkp_hist_kernel<T>), dim3(B, VP), dim3(256),                             0, stream, SAMP_ARGS(T),                                                    part_mz.data_ptr<float>(),                                                  hist_mass.data_ptr<float>(),                                                hist_cnt.data_ptr<int>(), V);                            hipLaunchKernelGGL((topkp_argmax_kernel<T>), dim3(B, VP), dim3(256),                           0, stream, (const T*)logits.data_ptr(),                                     temperatures.data_ptr<float>(),                                             (const uint64_t*)seeds.data_ptr(),                                          top_p.data_ptr<float>(), top_k.data_ptr<int>(),                             rep_pen.data_ptr<float>(),                                                  pres_pen.data_ptr<float>(),                                                 freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr,                              map_ptr, part_mz.data_ptr<float>(),                                         hist_mass.data_ptr<float>(),                                                hist_cnt.data_ptr<int>(), win.data_ptr<float>(),                            win_i.data_ptr<int>(), V);                               hipLaunchKernelGGL(topkp_final_kernel, dim3(B), dim3(64), 0, stream,                           out.data_ptr<int64_t>(), win.data_ptr<float>(),                             win_i.data_ptr<int>());                                } while (0)
   if (logits.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL((sample_topkp_kernel<uint16_t>), dim3(B), dim3(256),
-                       0, stream, out.data_ptr<int64_t>(),
-                       (const uint16_t*)logits.data_ptr(),
-                       temperatures.data_ptr<float>(),
-                       (const uint64_t*)seeds.data_ptr(),
-                       top_p.data_ptr<float>(), top_k.data_ptr<int>(),
-                       rep_pen.data_ptr<float>(), pres_pen.data_ptr<float>(),
-                       freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr,
-                       map_ptr, V);
+    LAUNCH_SAMP(uint16_t);
   } else {
     TORCH_CHECK(logits.scalar_type() == torch::kFloat32);
-    hipLaunchKernelGGL((sample_topkp_kernel<float>), dim3(B), dim3(256),
-                       0, stream, out.data_ptr<int64_t>(),
-                       logits.data_ptr<float>(),
-                       temperatures.data_ptr<float>(),
-                       (const uint64_t*)seeds.data_ptr(),
-                       top_p.data_ptr<float>(), top_k.data_ptr<int>(),
-                       rep_pen.data_ptr<float>(), pres_pen.data_ptr<float>(),
-                       freq_pen.data_ptr<float>(), cnt_ptr, seen_ptr,
-                       map_ptr, V);
+    LAUNCH_SAMP(float);
   }
+#undef LAUNCH_SAMP
+#undef SAMP_ARGS
 }
