@@ -229,61 +229,6 @@ def session_list(url: str = typer.Option("", "--url")):
         typer.echo(f"{s['id']}  {s['name'][:40]}")
 
 
-project_app = typer.Typer(help="Projects & spec-tasks")
-app.add_typer(project_app, name="project")
-
-
-@project_app.command("create")
-def project_create(name: str, url: str = typer.Option("", "--url")):
-    import httpx
-    api, headers = _api(url)
-    r = httpx.post(f"{api}/api/v1/projects", json={"name": name},
-                   headers=headers)
-    typer.echo(r.json())
-
-
-@project_app.command("list")
-def project_list(url: str = typer.Option("", "--url")):
-    import httpx
-    api, headers = _api(url)
-    for p in httpx.get(f"{api}/api/v1/projects", headers=headers).json():
-        typer.echo(f"{p['id']}  {p['name']}")
-
-
-spectask_app = typer.Typer(help="Spec-task kanban")
-app.add_typer(spectask_app, name="spectask")
-
-
-@spectask_app.command("create")
-def spectask_create(project_id: str, title: str,
-                    description: str = typer.Option("", "--description"),
-                    url: str = typer.Option("", "--url")):
-    import httpx
-    api, headers = _api(url)
-    r = httpx.post(f"{api}/api/v1/projects/{project_id}/tasks",
-                   json={"title": title, "description": description},
-                   headers=headers)
-    typer.echo(r.json())
-
-
-@spectask_app.command("list")
-def spectask_list(project_id: str, url: str = typer.Option("", "--url")):
-    import httpx
-    api, headers = _api(url)
-    for t in httpx.get(f"{api}/api/v1/projects/{project_id}/tasks",
-                       headers=headers).json():
-        typer.echo(f"{t['id']}  [{t['state']:<12}] {t['title']}")
-
-
-@spectask_app.command("plan")
-def spectask_plan(task_id: str, url: str = typer.Option("", "--url")):
-    import httpx
-    api, headers = _api(url)
-    r = httpx.post(f"{api}/api/v1/spec-tasks/{task_id}/plan",
-                   headers=headers, timeout=300)
-    typer.echo(r.json())
-
-
 model_app = typer.Typer(help="Model catalog & local models")
 app.add_typer(model_app, name="model")
 
@@ -728,6 +673,71 @@ def sandbox_exec(sandbox_id: str, command: str,
 def sandbox_rm(sandbox_id: str):
     _client().delete_sandbox(sandbox_id)
     typer.echo("deleted")
+
+
+fs_app = typer.Typer(help="Filestore operations")
+app.add_typer(fs_app, name="fs")
+
+
+@fs_app.command("list")
+def fs_list(path: str = typer.Argument("")):
+    for f in _client().filestore_list(path):
+        kind = "d" if f.get("dir") else "f"
+        typer.echo(f"{kind}  {f.get('size', 0):>10}  {f['name']}")
+
+
+@fs_app.command("upload")
+def fs_upload(local: str, remote: str = typer.Argument("")):
+    with open(local, "rb") as fh:
+        _client().filestore_upload(remote or local.split("/")[-1],
+                                   fh.read())
+    typer.echo("uploaded")
+
+
+@fs_app.command("rm")
+def fs_rm(path: str):
+    _client().filestore_delete(path)
+    typer.echo("deleted")
+
+
+user_app = typer.Typer(help="User administration")
+app.add_typer(user_app, name="user")
+
+
+@user_app.command("create")
+def user_create(username: str, admin: bool = typer.Option(False)):
+    c = _client()
+    out = c.request("POST", "/api/v1/users",
+                    body={"username": username, "admin": admin})
+    typer.echo(f"{out['id']}  api_key={out.get('api_key', '')}")
+
+
+@user_app.command("list")
+def user_list():
+    for u in _client().request("GET", "/api/v1/users"):
+        typer.echo(f"{u['id']}  {u.get('username', '')}"
+                   f"{'  [admin]' if u.get('admin') else ''}")
+
+
+provider_app = typer.Typer(help="Provider endpoints")
+app.add_typer(provider_app, name="provider")
+
+
+@provider_app.command("list")
+def provider_list():
+    out = _client().request("GET", "/api/v1/provider-endpoints")
+    for pvd in out:
+        typer.echo(f"{pvd.get('provider', '')}  "
+                   f"{pvd.get('base_url', '')}")
+
+
+@provider_app.command("add")
+def provider_add(provider: str, base_url: str,
+                 api_key: str = typer.Option("")):
+    _client().request("POST", "/api/v1/provider-endpoints",
+                      body={"provider": provider, "base_url": base_url,
+                            "api_key": api_key})
+    typer.echo("added")
 
 
 billing_app = typer.Typer(help="Billing and usage")

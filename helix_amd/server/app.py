@@ -1656,6 +1656,13 @@ def create_app(cfg: Optional[ServerConfig] = None,
     # ------------------------------------------------------------------
     # Users / keys / admin
     # ------------------------------------------------------------------
+    @app.get("/api/v1/users")
+    async def list_users(user: AuthUser = Depends(admin_dep)):
+        """Admin user listing (reference client.go:275 ListUsers)."""
+        return [{k: u.get(k) for k in
+                 ("id", "username", "admin", "email")}
+                for u in store.list("users", limit=10000)]
+
     @app.post("/api/v1/users")
     async def create_user(request: Request,
                           user: AuthUser = Depends(admin_dep)):
