@@ -38,6 +38,7 @@ class LlamaConfig:
     tie_embeddings: bool = False
     sliding_window: int = 0        # 0 = full attention (Mistral v0.1: 4096)
     attention_bias: bool = False   # Qwen2-style QKV bias
+    rope_scaling: dict = field(default_factory=dict)  # llama3.1 scheme
 
     @property
     def q_size(self) -> int:
@@ -50,6 +51,13 @@ class LlamaConfig:
 
 PRESETS = {
     "llama3-8b": LlamaConfig(),
+    # Llama-3.1: same weights shape as 3.0 but 128k context via the
+    # llama3 rope-scaling scheme (HF config.json rope_scaling)
+    "llama3.1-8b": LlamaConfig(
+        name="llama3.1-8b", max_position=32768, rope_base=500000.0,
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 8192}),
     "llama3-70b": LlamaConfig(
         name="llama3-70b", hidden_size=8192, intermediate_size=28672,
         num_layers=80, num_heads=64, num_kv_heads=8),
@@ -235,7 +243,8 @@ class LlamaForCausalLM(nn.Module):
         else:
             self.lm_head = HLinear(cfg.hidden_size, cfg.vocab_size,
                                    bias=False)
-        cs = make_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_base)
+        cs = make_cos_sin_cache(cfg.head_dim, cfg.max_position, cfg.rope_base,
+                                rope_scaling=cfg.rope_scaling or None)
         self.register_buffer("cos_sin", cs, persistent=False)
 
     def _apply(self, fn, recurse=True):

@@ -50,11 +50,38 @@ def rotary_embedding(positions: torch.Tensor, q: torch.Tensor,
 
 
 def make_cos_sin_cache(head_dim: int, max_pos: int, base: float = 10000.0,
-                       device="cpu") -> torch.Tensor:
-    """[max_pos, D] fp32 rows = [cos(0..D/2), sin(0..D/2)]."""
+                       device="cpu",
+                       rope_scaling: dict | None = None) -> torch.Tensor:
+    """[max_pos, D] fp32 rows = [cos(0..D/2), sin(0..D/2)].
+
+    rope_scaling supports the llama3.1 scheme (HF config.json
+    rope_scaling: rope_type "llama3" with factor / low_freq_factor /
+    high_freq_factor / original_max_position_embeddings): wavelengths
+    longer than original/low_freq are divided by `factor`, shorter than
+    original/high_freq are untouched, with a smooth ramp between —
+    the long-context extension Llama-3.1 ships with."""
+    import math
     half = head_dim // 2
     inv_freq = 1.0 / (base ** (torch.arange(half, dtype=torch.float32,
                                             device=device) / half))
+    if rope_scaling and rope_scaling.get("rope_type", rope_scaling.get(
+            "type", "")) == "llama3":
+        factor = float(rope_scaling.get("factor", 8.0))
+        lo = float(rope_scaling.get("low_freq_factor", 1.0))
+        hi = float(rope_scaling.get("high_freq_factor", 4.0))
+        orig = float(rope_scaling.get(
+            "original_max_position_embeddings", 8192))
+        wavelen = 2 * math.pi / inv_freq
+        low_wl = orig / lo
+        high_wl = orig / hi
+        scaled = inv_freq / factor
+        # smooth interpolation in the medium-frequency band
+        smooth = (orig / wavelen - lo) / (hi - lo)
+        smooth = smooth.clamp(0.0, 1.0)
+        mid = (1 - smooth) * scaled + smooth * inv_freq
+        inv_freq = torch.where(wavelen > low_wl, scaled,
+                               torch.where(wavelen < high_wl,
+                                           inv_freq, mid))
     t = torch.arange(max_pos, dtype=torch.float32, device=device)
     freqs = torch.outer(t, inv_freq)  # [max_pos, half]
     return torch.cat([freqs.cos(), freqs.sin()], dim=-1)
