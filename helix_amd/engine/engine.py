@@ -49,6 +49,7 @@ class Sequence:
     finish_reason: Optional[str] = None
     logprobs: List[dict] = field(default_factory=list)  # when requested
     pen_init: bool = False       # GPU penalty tables primed for this row
+    json_mask: object = None     # JSONByteMask when params.json_mode
     # streaming callback: fn(seq, new_token_id, finished)
     on_token: Optional[Callable] = None
 
@@ -493,10 +494,52 @@ class LLMEngine:
                     accumulate=True)
             seq.pen_init = True
 
+    def _apply_json_masks(self, batch: List[Sequence],
+                          logits: torch.Tensor):
+        """Grammar-constrained decoding (response_format json_object):
+        mask each JSON row's logits to the automaton's allowed byte
+        tokens; a complete document admits only EOS."""
+        from helix_amd.engine.json_mode import JSONByteMask
+        from helix_amd.utils.tokenizer import N_SPECIAL
+        for i, s in enumerate(batch):
+            if not s.params.json_mode:
+                continue
+            if s.json_mask is None:
+                s.json_mask = JSONByteMask()
+            m = s.json_mask
+            if m.complete:
+                allowed = [self.cfg.eos_token_id]                     if self.cfg.eos_token_id >= 0 else []
+            else:
+                V = logits.shape[1]
+                allowed = [b + N_SPECIAL for b in m.allowed_bytes()
+                           if b + N_SPECIAL < V]
+            row = logits[i]
+            if not allowed:
+                continue
+            idx = torch.tensor(allowed, dtype=torch.int64,
+                               device=row.device)
+            keep = row[idx].clone()
+            row.fill_(float("-inf"))
+            row[idx] = keep
+
+    def _advance_json_masks(self, batch: List[Sequence],
+                            out: List[int]):
+        from helix_amd.utils.tokenizer import N_SPECIAL
+        for i, s in enumerate(batch):
+            if s.params.json_mode and s.json_mask is not None:
+                b = out[i] - N_SPECIAL
+                if 0 <= b < 256:
+                    s.json_mask.push_byte(b)
+
     def _sample(self, batch: List[Sequence], logits: torch.Tensor) -> List[int]:
         # wait: decode path's new input token is appended by _append_token;
         # here logits are [B, V].
         B = len(batch)
+        if any(s.params.json_mode for s in batch):
+            # logits may be an inference-mode tensor (graph/forward
+            # output): clone before the in-place grammar mask
+            logits = logits.clone()
+            self._apply_json_masks(batch, logits)
         lp_rows = [i for i, s_ in enumerate(batch) if s_.params.logprobs]
         needs_proc = any(s.params.needs_logit_processing for s in batch)
         gpu_fast = logits.is_cuda and needs_proc
@@ -611,6 +654,8 @@ class LLMEngine:
             seeds_t = torch.from_numpy(hs[:B].copy())
             toks = ops.sample_tokens(logits.contiguous(), temps, seeds_t)
         out = toks.cpu().tolist()
+        if any(s.params.json_mode for s in batch):
+            self._advance_json_masks(batch, out)
         # top-k logprobs for sequences that requested them (one extra
         # log_softmax + topk over just those rows)
         lp_rows = [i for i, s_ in enumerate(batch) if s_.params.logprobs]

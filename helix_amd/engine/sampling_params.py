@@ -20,6 +20,7 @@ class SamplingParams:
     ignore_eos: bool = False
     seed: Optional[int] = None
     logprobs: Optional[int] = None     # top-k logprobs per emitted token
+    json_mode: bool = False            # grammar-constrain output to JSON
 
     @property
     def needs_logit_processing(self) -> bool:

@@ -52,6 +52,8 @@ def _params_from_request(req: dict, max_model_len: int) -> SamplingParams:
         seed=req.get("seed"),
         logprobs=(int(req.get("top_logprobs") or 1)
                   if req.get("logprobs") else None),
+        json_mode=(req.get("response_format") or {}).get(
+            "type") == "json_object",
     ), ([stop] if isinstance(stop, str) else list(stop or []))
 
 
