@@ -535,10 +535,25 @@ class LLMEngine:
         # wait: decode path's new input token is appended by _append_token;
         # here logits are [B, V].
         B = len(batch)
-        if any(s.params.json_mode for s in batch):
+        needs_bias = any(s.params.logit_bias for s in batch)
+        needs_json = any(s.params.json_mode for s in batch)
+        if needs_bias or needs_json:
             # logits may be an inference-mode tensor (graph/forward
-            # output): clone before the in-place grammar mask
+            # output): clone before in-place bias / grammar masking
             logits = logits.clone()
+        if needs_bias:
+            for i, s in enumerate(batch):
+                lb = s.params.logit_bias
+                if not lb:
+                    continue
+                idx = torch.tensor([int(k) for k in lb],
+                                   dtype=torch.int64,
+                                   device=logits.device)
+                vals = torch.tensor([float(v) for v in lb.values()],
+                                    dtype=logits.dtype,
+                                    device=logits.device)
+                logits[i].index_add_(0, idx, vals)
+        if needs_json:
             self._apply_json_masks(batch, logits)
         lp_rows = [i for i, s_ in enumerate(batch) if s_.params.logprobs]
         needs_proc = any(s.params.needs_logit_processing for s in batch)

@@ -21,6 +21,7 @@ class SamplingParams:
     seed: Optional[int] = None
     logprobs: Optional[int] = None     # top-k logprobs per emitted token
     json_mode: bool = False            # grammar-constrain output to JSON
+    logit_bias: Optional[dict] = None  # token id -> additive bias
 
     @property
     def needs_logit_processing(self) -> bool:

@@ -54,6 +54,8 @@ def _params_from_request(req: dict, max_model_len: int) -> SamplingParams:
                   if req.get("logprobs") else None),
         json_mode=(req.get("response_format") or {}).get(
             "type") == "json_object",
+        logit_bias={int(k): float(v) for k, v in
+                    (req.get("logit_bias") or {}).items()} or None,
     ), ([stop] if isinstance(stop, str) else list(stop or []))
 
 

@@ -107,3 +107,24 @@ def test_adapter_maps_response_format():
     assert p.json_mode is True
     p, _ = _params_from_request({"max_tokens": 9}, 2048)
     assert p.json_mode is False
+
+
+def test_logit_bias():
+    """OpenAI logit_bias: additive per-token bias before sampling
+    (a large positive bias makes the token deterministic)."""
+    import torch
+
+    from helix_amd.engine.engine import EngineConfig, LLMEngine
+    from helix_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny", max_num_seqs=2,
+                                 max_model_len=64, kv_cache_blocks=32,
+                                 eos_token_id=-1), device="cpu")
+    sp = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True,
+                        logit_bias={77: 1000.0})
+    out = eng.generate([[1, 2, 3]], sp)[0]
+    assert out == [77, 77, 77, 77]
+    # adapter mapping (string keys, OpenAI wire form)
+    from helix_amd.runner.openai_adapter import _params_from_request
+    p, _ = _params_from_request({"logit_bias": {"42": -5}}, 128)
+    assert p.logit_bias == {42: -5.0}
