@@ -683,6 +683,17 @@ def create_app(cfg: Optional[ServerConfig] = None,
         if (user is None and not is_runner) or not agent_id:
             await ws.close(code=4401)
             return
+        if not is_runner and not (user and user.admin):
+            # the agent key is a session/app id: only its owner may
+            # serve turns for it (otherwise any authenticated user
+            # could hijack another tenant's zed_external sessions)
+            sess = controller.get_session(agent_id)
+            app_doc = store.get("apps", agent_id)
+            owns = (sess is not None and sess.owner == user.id) or                 (app_doc is not None and
+                 app_doc.get("owner") == user.id)
+            if not owns:
+                await ws.close(code=4403)
+                return
         await ws.accept()
         outbox = external_agents.attach(agent_id)
 

@@ -176,9 +176,21 @@ def test_ws_route_auth(tmp_path):
         except Exception:
             ok = True
         assert ok
-        # valid connect registers + ping/pong works
+        # a random agent_id not owned by the caller is rejected
+        try:
+            with client.websocket_connect(
+                    f"/api/v1/external-agents/ws?agent_id=sess-7"
+                    f"&access_token={key}") as ws:
+                ws.receive_json()
+            hijack_blocked = False
+        except Exception:
+            hijack_blocked = True
+        assert hijack_blocked
+        # owning a session grants the uplink; ping/pong works
+        sess = app.state.controller.create_session(
+            auth.resolve(key).id, name="mine")
         with client.websocket_connect(
-                f"/api/v1/external-agents/ws?agent_id=sess-7"
+                f"/api/v1/external-agents/ws?agent_id={sess.id}"
                 f"&access_token={key}") as ws:
             ws.send_json({"type": "ping"})
             assert ws.receive_json() == {"type": "pong"}
