@@ -799,6 +799,31 @@ def create_app(cfg: Optional[ServerConfig] = None,
             raise HTTPException(403, "forbidden")
         return a
 
+    def _reconcile_app_triggers(a: App, owner: str):
+        """helix.yaml `triggers:` -> TriggerManager rows (reference
+        apply registers cron/slack/webhook triggers from the app
+        config; stale rows for the app are replaced)."""
+        for t in store.list("triggers", limit=10000):
+            if t.get("app_id") == a.id and t.get("from_app_config"):
+                store.delete("triggers", t["id"])
+        made = []
+        for spec in (a.config.triggers or []):
+            kind = spec.get("kind") or spec.get("type") or ""
+            config = dict(spec.get("config") or
+                          {k: v for k, v in spec.items()
+                           if k not in ("kind", "type")})
+            if not kind:
+                continue
+            try:
+                doc = triggers.create(owner, a.id, kind, config)
+                doc["from_app_config"] = True
+                store.put("triggers", doc["id"], doc, owner=owner,
+                          parent=a.id)
+                made.append(doc["id"])
+            except ValueError as e:
+                raise HTTPException(400, f"bad trigger {kind}: {e}")
+        return made
+
     for prefix in ("/api/v1/apps", "/api/v1/agents"):
         def _bind(prefix=prefix):
             @app.post(prefix, name=f"create_app_{prefix}")
@@ -811,6 +836,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
                         global_=bool(body.get("global", False)))
                 store.put("apps", a.id, a.model_dump(by_alias=True),
                           owner=user.id)
+                _reconcile_app_triggers(a, user.id)
                 return a.model_dump(by_alias=True)
 
             @app.get(prefix, name=f"list_apps_{prefix}")
@@ -834,6 +860,7 @@ def create_app(cfg: Optional[ServerConfig] = None,
                 a.updated = now_ms()
                 store.put("apps", a.id, a.model_dump(by_alias=True),
                           owner=a.owner)
+                _reconcile_app_triggers(a, a.owner)
                 return a.model_dump(by_alias=True)
 
             @app.delete(prefix + "/{app_id}", name=f"delete_app_{prefix}")
@@ -841,6 +868,10 @@ def create_app(cfg: Optional[ServerConfig] = None,
                                    user: AuthUser = Depends(auth_dep)):
                 _get_app(app_id, user)
                 store.delete("apps", app_id)
+                for t in store.list("triggers", limit=10000):
+                    if t.get("app_id") == app_id and \
+                            t.get("from_app_config"):
+                        store.delete("triggers", t["id"])
                 return {"ok": True}
         _bind()
 

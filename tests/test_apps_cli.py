@@ -85,3 +85,45 @@ def test_cli_doctor_runs():
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "extension built" in r.stdout
+
+
+def test_helix_yaml_triggers_register(tmp_path):
+    """helix.yaml `triggers:` are reconciled into the trigger manager
+    on app create/update/delete (reference apply behavior)."""
+    from fastapi.testclient import TestClient
+
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import load_config
+    cfg = load_config()
+    cfg.store.path = str(tmp_path / "db.sqlite")
+    cfg.filestore.path = str(tmp_path / "fs")
+    app = create_app(cfg)
+    with TestClient(app) as client:
+        auth = app.state.auth
+        key = auth.create_api_key(auth.create_user("u")["id"])
+        H = {"Authorization": f"Bearer {key}"}
+        r = client.post("/api/v1/apps", headers=H, json={"config": {
+            "name": "cronapp", "assistants": [{"name": "a"}],
+            "triggers": [{"kind": "cron",
+                          "config": {"schedule": "0 9 * * *",
+                                     "prompt": "daily digest"}}]}})
+        assert r.status_code == 200, r.text
+        aid = r.json()["id"]
+        trs = client.get("/api/v1/triggers", headers=H).json()
+        assert [t["kind"] for t in trs] == ["cron"]
+        assert trs[0]["app_id"] == aid
+        # update swaps the trigger set atomically
+        client.put(f"/api/v1/apps/{aid}", headers=H, json={"config": {
+            "name": "cronapp", "assistants": [{"name": "a"}],
+            "triggers": [{"kind": "webhook", "config": {}}]}})
+        trs = client.get("/api/v1/triggers", headers=H).json()
+        assert [t["kind"] for t in trs] == ["webhook"]
+        # bad cron rejected with 400
+        r = client.put(f"/api/v1/apps/{aid}", headers=H, json={"config": {
+            "name": "cronapp", "assistants": [{"name": "a"}],
+            "triggers": [{"kind": "cron",
+                          "config": {"schedule": "junk"}}]}})
+        assert r.status_code == 400
+        # delete clears app-config triggers
+        client.delete(f"/api/v1/apps/{aid}", headers=H)
+        assert client.get("/api/v1/triggers", headers=H).json() == []
