@@ -2125,6 +2125,23 @@ def create_app(cfg: Optional[ServerConfig] = None,
                         pruned["interactions"] += 1
                     store.delete("sessions", doc["id"])
                     pruned["sessions"] += 1
+        # bounded retention for bus streams and error events
+        keep_bus = int(body.get("bus_keep_last", 10000))
+        streams = {r["id"].split(":", 1)[0]
+                   for r in store.list("bus_messages", limit=100000)}
+        pruned["bus_messages"] = 0
+        for stream in streams:
+            before = len(store.list("bus_messages", parent=stream,
+                                    limit=1000000))
+            bus.purge(stream, keep_last=keep_bus)
+            after = len(store.list("bus_messages", parent=stream,
+                                   limit=1000000))
+            pruned["bus_messages"] += before - after
+        pruned["error_events"] = 0
+        for doc in store.list("error_events", limit=100000):
+            if doc.get("last_seen", 0) < cutoff_s:
+                store.delete("error_events", doc["id"])
+                pruned["error_events"] += 1
         # idle-sandbox GC (reference hydra handleGCReconcile)
         sbx_hours = float(body.get("sandbox_idle_hours", 24))
         pruned["sandboxes"] = 0
