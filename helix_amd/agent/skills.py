@@ -102,20 +102,18 @@ class APISkill(Skill):
     query params injected."""
 
     def __init__(self, tool_cfg, http_client=None):
-        import httpx
         self.cfg = tool_cfg
         self.base_url = tool_cfg.url.rstrip("/")
         self.headers = dict(tool_cfg.headers or {})
         self.query = dict(tool_cfg.query or {})
-        self._http = http_client or httpx.AsyncClient(timeout=30)
+        self._http = http_client   # lazy: skills are built per request
+                                   # (incl. is_actionable) — eager
+                                   # clients leak sockets under load
+
         self.operations = self._parse_spec(tool_cfg.schema_)
         self.name = f"api_{tool_cfg.name}".replace(" ", "_").lower()
         self.description = (tool_cfg.description or
                             f"Call the {tool_cfg.name} API")
-        ops = {oid: {"type": "object", "properties": {
-            "parameters": {"type": "object"},
-            "body": {"type": "object"}}}
-            for oid in self.operations}
         self.parameters = {"type": "object", "properties": {
             "operation_id": {"type": "string",
                              "enum": list(self.operations.keys())},
@@ -123,6 +121,12 @@ class APISkill(Skill):
                            "description": "path/query parameters"},
             "body": {"type": "object", "description": "JSON request body"},
         }, "required": ["operation_id"]}
+
+    def _client(self):
+        if self._http is None:
+            import httpx
+            self._http = httpx.AsyncClient(timeout=30)
+        return self._http
 
     @staticmethod
     def _parse_spec(spec_text: str) -> Dict[str, dict]:
@@ -157,7 +161,7 @@ class APISkill(Skill):
                 path = path.replace("{" + k + "}", str(v))
                 params.pop(k)
         try:
-            r = await self._http.request(
+            r = await self._client().request(
                 o["method"], self.base_url + path,
                 params={**self.query, **params},
                 headers=self.headers,
@@ -176,15 +180,20 @@ class WebSearchSkill(Skill):
         "query": {"type": "string"}}, "required": ["query"]}
 
     def __init__(self, searxng_url: str = "", http_client=None):
-        import httpx
         self.url = searxng_url
-        self._http = http_client or httpx.AsyncClient(timeout=20)
+        self._http = http_client
+
+    def _client(self):
+        if self._http is None:
+            import httpx
+            self._http = httpx.AsyncClient(timeout=20)
+        return self._http
 
     async def execute(self, args: dict, ctx: dict) -> str:
         if not self.url:
             return "web search is not configured (no SEARXNG_URL)"
         try:
-            r = await self._http.get(self.url + "/search", params={
+            r = await self._client().get(self.url + "/search", params={
                 "q": args.get("query", ""), "format": "json"})
             results = r.json().get("results", [])[:5]
             return "\n".join(f"- {x.get('title')}: {x.get('content', '')} "
@@ -245,12 +254,17 @@ class BrowserSkill(Skill):
 
     def __init__(self, config: dict | None = None, llm=None,
                  http_client=None):
-        import httpx
         self.config = config or {}
         self.llm = llm                  # async callable(messages) -> str
-        self._http = http_client or httpx.AsyncClient(
-            timeout=30, follow_redirects=True)
+        self._http = http_client
         self._cache: Dict[str, str] = {}
+
+    def _client(self):
+        if self._http is None:
+            import httpx
+            self._http = httpx.AsyncClient(timeout=30,
+                                           follow_redirects=True)
+        return self._http
 
     async def execute(self, args: dict, ctx: dict) -> str:
         from helix_amd.server.extract import extract_html
@@ -264,7 +278,7 @@ class BrowserSkill(Skill):
             else None
         if text is None:
             try:
-                r = await self._http.get(url)
+                r = await self._client().get(url)
             except Exception as e:
                 return f"browser error: {e}"
             if r.status_code >= 400:
@@ -614,9 +628,14 @@ class ZapierSkill(Skill):
     BASE = "https://nla.zapier.com/api/v1"
 
     def __init__(self, config: dict, http_client=None):
-        import httpx
         self.api_key = (config or {}).get("api_key", "")
-        self._http = http_client or httpx.AsyncClient(timeout=60)
+        self._http = http_client
+
+    def _client(self):
+        if self._http is None:
+            import httpx
+            self._http = httpx.AsyncClient(timeout=60)
+        return self._http
 
     def _headers(self):
         return {"X-API-Key": self.api_key}
@@ -626,8 +645,8 @@ class ZapierSkill(Skill):
             return "zapier error: no API key configured"
         try:
             if args.get("action") == "list":
-                r = await self._http.get(self.BASE + "/exposed/",
-                                         headers=self._headers())
+                r = await self._client().get(self.BASE + "/exposed/",
+                                             headers=self._headers())
                 if r.status_code != 200:
                     return f"zapier error: HTTP {r.status_code}"
                 rows = r.json().get("results", [])
@@ -636,7 +655,7 @@ class ZapierSkill(Skill):
             aid = args.get("action_id", "")
             if not aid:
                 return "zapier error: action_id required for execute"
-            r = await self._http.post(
+            r = await self._client().post(
                 self.BASE + f"/exposed/{aid}/execute/",
                 headers=self._headers(),
                 json={"instructions": args.get("instructions", "")})
