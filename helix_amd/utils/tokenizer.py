@@ -74,14 +74,60 @@ _DEFAULT = ByteTokenizer()
 _CACHE = {}
 
 
+def _jinja_raise(msg):
+    raise ValueError(msg)
+
+
+def _token_str(tok):
+    """HF tokenizer_config token fields are strings or AddedToken
+    dicts."""
+    if isinstance(tok, dict):
+        return tok.get("content", "")
+    return tok or ""
+
+
 class HFTokenizer:
     """Real tokenizer files (tokenizer.json) when a deployment provides
     them via HELIX_TOKENIZER_DIR/<model>/tokenizer.json — same interface
     as ByteTokenizer."""
 
     def __init__(self, path: str):
+        import json as _json
+        import os as _os
+
         from tokenizers import Tokenizer
         self._tok = Tokenizer.from_file(path)
+        # the model's own jinja chat template, when the deployment
+        # ships tokenizer_config.json next to tokenizer.json (HF
+        # layout) — takes precedence over the built-in family
+        # templates
+        self.chat_template = None
+        cfg_path = _os.path.join(_os.path.dirname(path),
+                                 "tokenizer_config.json")
+        if _os.path.exists(cfg_path):
+            try:
+                with open(cfg_path) as fh:
+                    tc = _json.load(fh)
+                tpl = tc.get("chat_template")
+                if isinstance(tpl, list):      # HF multi-template form
+                    tpl = next((t.get("template") for t in tpl
+                                if t.get("name") == "default"),
+                               tpl[0].get("template")
+                               if tpl else None)
+                if tpl:
+                    import jinja2
+                    env = jinja2.Environment(
+                        trim_blocks=True, lstrip_blocks=True,
+                        undefined=jinja2.ChainableUndefined)
+                    env.globals["raise_exception"] = _jinja_raise
+                    env.filters["tojson"] = _json.dumps
+                    self.chat_template = env.from_string(tpl)
+                self._special = {
+                    "bos_token": _token_str(tc.get("bos_token")),
+                    "eos_token": _token_str(tc.get("eos_token")),
+                }
+            except Exception:
+                self.chat_template = None
         self.vocab_size = self._tok.get_vocab_size()
         self.bos_token_id = self._tok.token_to_id("<|begin_of_text|>") or             self._tok.token_to_id("<s>") or 1
         self.eos_token_id = self._tok.token_to_id("<|end_of_text|>") or             self._tok.token_to_id("</s>") or 2
@@ -98,6 +144,14 @@ class HFTokenizer:
         """Render with the model family's real prompt format (llama3
         header tokens / mistral [INST] / qwen ChatML) and encode; the
         special strings are added tokens in real tokenizer.json files."""
+        if self.chat_template is not None:
+            # the model's own jinja template (exact HF semantics for
+            # the fields real templates use)
+            text = self.chat_template.render(
+                messages=messages, tools=tools,
+                add_generation_prompt=add_generation_prompt,
+                **getattr(self, "_special", {}))
+            return self.encode(text, add_bos=False)
         from helix_amd.utils.chat_templates import TEMPLATES
         render = TEMPLATES.get(template or "llama3", TEMPLATES["llama3"])
         text = render(messages, add_generation_prompt, tools=tools)

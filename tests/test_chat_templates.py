@@ -135,3 +135,43 @@ def test_adapter_nonstream_tool_calls(monkeypatch):
     ch = resp["choices"][0]
     assert ch["finish_reason"] == "tool_calls"
     assert ch["message"]["tool_calls"][0]["function"]["name"] == "calc"
+
+
+def test_hf_jinja_chat_template(tmp_path, monkeypatch):
+    """A model shipping its own chat_template (HF tokenizer_config
+    jinja) takes precedence over the built-in family templates."""
+    import json
+
+    from helix_amd.utils.tokenizer import get_tokenizer
+
+    mdir = tmp_path / "jinja-model"
+    mdir.mkdir()
+    # minimal real tokenizer.json (WordLevel with a tiny vocab)
+    vocab = {c: i for i, c in enumerate(
+        ["<s>", "</s>", "[UNK]", "user", "assistant", ":", "hi",
+         "there", "SYS", "\n", " "])}
+    (mdir / "tokenizer.json").write_text(json.dumps({
+        "version": "1.0",
+        "truncation": None, "padding": None,
+        "added_tokens": [], "normalizer": None,
+        "pre_tokenizer": {"type": "Whitespace"},
+        "post_processor": None, "decoder": None,
+        "model": {"type": "WordLevel", "vocab": vocab,
+                  "unk_token": "[UNK]"}}))
+    (mdir / "tokenizer_config.json").write_text(json.dumps({
+        "bos_token": "<s>", "eos_token": "</s>",
+        "chat_template":
+            "{{ bos_token }}{% for m in messages %}"
+            "{{ m.role }} : {{ m.content }}\n{% endfor %}"
+            "{% if add_generation_prompt %}assistant :{% endif %}"}))
+    monkeypatch.setenv("HELIX_TOKENIZER_DIR", str(tmp_path))
+    tok = get_tokenizer("jinja-model")
+    assert tok.chat_template is not None
+    ids = tok.apply_chat_template(
+        [{"role": "user", "content": "hi there"}],
+        add_generation_prompt=True)
+    text_tokens = [k for k, v in sorted(vocab.items(),
+                                        key=lambda kv: kv[1])]
+    rendered = " ".join(text_tokens[i] for i in ids)
+    assert "user" in rendered and "hi" in rendered
+    assert rendered.rstrip().endswith("assistant :")
