@@ -2000,6 +2000,69 @@ def create_app(cfg: Optional[ServerConfig] = None,
             except OSError:
                 pass
 
+    # -- git smart-HTTP (reference git_http_server.go): real clone /
+    #    push against platform repos via `git http-backend`, auth'd
+    #    with the same bearer keys (git clients send them via
+    #    http.extraHeader, or basic auth with the key as password) ----
+    def _git_http_user(request: Request) -> Optional[AuthUser]:
+        hdr = request.headers.get("Authorization", "")
+        if hdr.startswith("Basic "):
+            import base64 as _b64
+            try:
+                raw = _b64.b64decode(hdr[6:]).decode()
+                _, _, pwd = raw.partition(":")
+                return auth.resolve(pwd)
+            except Exception:
+                return None
+        if hdr.startswith("Bearer "):
+            return auth.resolve(hdr[7:])
+        return None
+
+    def _git_http_repo(rid: str, request: Request) -> dict:
+        user = _git_http_user(request)
+        if user is None:
+            raise HTTPException(
+                401, "authentication required",
+                headers={"WWW-Authenticate": 'Basic realm="helix-git"'})
+        doc = git_svc.get(rid)
+        if doc is None or (doc.get("owner") != user.id and
+                           not user.admin):
+            raise HTTPException(404, "repository not found")
+        return doc
+
+    @app.get("/api/v1/git/repos/{rid}.git/info/refs")
+    async def git_info_refs(rid: str, request: Request,
+                            service: str = ""):
+        from helix_amd.server.git_service import run_http_backend
+        doc = _git_http_repo(rid, request)
+        if service not in ("git-upload-pack", "git-receive-pack"):
+            raise HTTPException(400, "dumb HTTP protocol not served")
+        status, headers, payload = await asyncio.to_thread(
+            run_http_backend, doc["path"], "GET",
+            f"/{rid}.git/info/refs", f"service={service}", "", b"")
+        from fastapi.responses import Response
+        return Response(content=payload, status_code=status,
+                        media_type=headers.get(
+                            "Content-Type",
+                            f"application/x-{service}-advertisement"))
+
+    @app.post("/api/v1/git/repos/{rid}.git/{service}")
+    async def git_service_rpc(rid: str, service: str, request: Request):
+        from helix_amd.server.git_service import run_http_backend
+        doc = _git_http_repo(rid, request)
+        if service not in ("git-upload-pack", "git-receive-pack"):
+            raise HTTPException(404, "unknown git service")
+        body = await request.body()
+        status, headers, payload = await asyncio.to_thread(
+            run_http_backend, doc["path"], "POST",
+            f"/{rid}.git/{service}", "",
+            request.headers.get("Content-Type", ""), body)
+        from fastapi.responses import Response
+        return Response(content=payload, status_code=status,
+                        media_type=headers.get(
+                            "Content-Type",
+                            f"application/x-{service}-result"))
+
     @app.get("/api/v1/config")
     async def get_config(user: AuthUser = Depends(auth_dep)):
         return {"version": "helix_amd-0.1.0",

@@ -130,3 +130,48 @@ class GitService:
                  check=False)
             _git(wt, "push", "-q", "origin", branch)
             return _git(wt, "rev-parse", "HEAD").strip()
+
+
+# ---------------------------------------------------------------------------
+# Smart-HTTP transport (reference api/pkg/server/git_http_server.go):
+# real `git clone` / `git push` against platform repos, bridged to
+# `git http-backend` (the stock CGI) with auth handled by the API.
+
+def run_http_backend(repo_path: str, method: str, path_info: str,
+                     query: str, content_type: str,
+                     body: bytes) -> tuple:
+    """Invoke git's CGI once; returns (status, headers, payload)."""
+    env = {
+        "GIT_PROJECT_ROOT": os.path.dirname(repo_path),
+        "GIT_HTTP_EXPORT_ALL": "1",
+        "PATH_INFO": path_info,
+        "REQUEST_METHOD": method,
+        "QUERY_STRING": query or "",
+        "CONTENT_TYPE": content_type or "",
+        "CONTENT_LENGTH": str(len(body)),
+        "GATEWAY_INTERFACE": "CGI/1.1",
+        "REMOTE_ADDR": "127.0.0.1",
+        "REMOTE_USER": "helix",
+        "PATH": os.environ.get("PATH", "/usr/bin:/bin"),
+    }
+    proc = subprocess.run(
+        ["git", "http-backend"], input=body, env=env,
+        capture_output=True, timeout=60)
+    raw = proc.stdout
+    sep = raw.find(b"\r\n\r\n")
+    if sep < 0:
+        sep = raw.find(b"\n\n")
+        head, payload = raw[:sep], raw[sep + 2:]
+    else:
+        head, payload = raw[:sep], raw[sep + 4:]
+    status = 200
+    headers = {}
+    for line in head.decode(errors="replace").splitlines():
+        if ":" not in line:
+            continue
+        k, v = line.split(":", 1)
+        if k.strip().lower() == "status":
+            status = int(v.strip().split()[0])
+        else:
+            headers[k.strip()] = v.strip()
+    return status, headers, payload
