@@ -1795,6 +1795,17 @@ def create_app(cfg: Optional[ServerConfig] = None,
         except (ValueError, RuntimeError) as e:
             raise HTTPException(400, str(e))
 
+    @app.post("/api/v1/spec-tasks/{tid}/merge")
+    async def merge_task(tid: str, user: AuthUser = Depends(auth_dep)):
+        doc = spec_tasks.get_task(tid)
+        if doc is None or (doc.get("owner") != user.id and
+                           not user.admin):
+            raise HTTPException(404, "task not found")
+        try:
+            return await asyncio.to_thread(spec_tasks.merge, tid)
+        except ValueError as e:
+            raise HTTPException(409, str(e))
+
     @app.post("/api/v1/spec-tasks/{tid}/comments")
     async def comment_task(tid: str, request: Request,
                            user: AuthUser = Depends(auth_dep)):

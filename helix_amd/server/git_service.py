@@ -131,6 +131,36 @@ class GitService:
             _git(wt, "push", "-q", "origin", branch)
             return _git(wt, "rev-parse", "HEAD").strip()
 
+    def merge_branch(self, rid: str, branch: str,
+                     into: str = "main",
+                     message: str = "") -> dict:
+        """Merge `branch` into `into` (the spec-task pr -> merged
+        transition; reference forge-PR merge role). Returns
+        {merged, commit} or raises ValueError on conflicts."""
+        import tempfile
+        bare = self._path(rid)
+        _safe_ref(branch)
+        _safe_ref(into)
+        with tempfile.TemporaryDirectory() as td:
+            wt = os.path.join(td, "wt")
+            subprocess.run(["git", "clone", "-q", bare, wt], check=True,
+                           capture_output=True)
+            _git(wt, "checkout", "-B", into, f"origin/{into}",
+                 check=False)
+            out = subprocess.run(
+                ["git", "-c", "user.email=agent@helix",
+                 "-c", "user.name=helix-agent", "merge", "--no-ff",
+                 "-m", message or f"merge {branch} into {into}",
+                 f"origin/{branch}"],
+                cwd=wt, capture_output=True, text=True)
+            if out.returncode != 0:
+                raise ValueError(
+                    f"merge conflict: {out.stdout[-300:]}"
+                    f"{out.stderr[-300:]}")
+            _git(wt, "push", "-q", "origin", into)
+            return {"merged": True,
+                    "commit": _git(wt, "rev-parse", "HEAD").strip()}
+
 
 # ---------------------------------------------------------------------------
 # Smart-HTTP transport (reference api/pkg/server/git_http_server.go):

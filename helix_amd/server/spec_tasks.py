@@ -229,3 +229,23 @@ class SpecTaskService:
             return {"ran": False, "error": str(e)}
         finally:
             self.sandboxes.delete(sbx["id"])
+
+    def merge(self, tid: str) -> dict:
+        """pr -> merged with a REAL git merge of the task branch
+        (reference: PR merge completes the spec-task flow)."""
+        doc = self.get_task(tid)
+        if doc is None:
+            raise KeyError(tid)
+        if doc.get("state") != "pr":
+            raise ValueError(f"task is {doc.get('state')}, not pr")
+        branch = doc.get("branch", "")
+        project = self.store.get("projects", doc["project_id"])
+        rid = (project or {}).get("repo_id")
+        if self.git is not None and rid and branch:
+            result = self.git.merge_branch(
+                rid, branch, message=f"task: {doc['title']} ({tid})")
+            doc = self.get_task(tid)
+            doc["merge_commit"] = result["commit"]
+            self.store.put("spec_tasks", tid, doc, owner=doc["owner"],
+                           parent=doc["project_id"])
+        return self.transition(tid, "merged")

@@ -187,3 +187,57 @@ def test_spec_task_implement_verifies_in_sandbox(tmp_path):
         assert "OK" in doc["verify"]["output"]
         # sandbox was cleaned up
         assert app.state.sandboxes.list(me["id"]) == []
+
+
+def test_spec_task_merge_completes_flow(tmp_path):
+    """pr -> merged performs a real git merge of the task branch into
+    main (the forge-PR role, platform-native)."""
+    import asyncio
+    import json as _json
+
+    from fastapi.testclient import TestClient
+
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import load_config
+    cfg = load_config()
+    cfg.store.path = str(tmp_path / "db.sqlite")
+    cfg.filestore.path = str(tmp_path / "fs")
+    app = create_app(cfg)
+    with TestClient(app) as client:
+        auth = app.state.auth
+        me = auth.create_user("dev2")
+        key = auth.create_api_key(me["id"])
+        H = {"Authorization": f"Bearer {key}"}
+        tasks = app.state.spec_tasks
+        proj = tasks.create_project(me["id"], "p")
+        rid = app.state.store.get("projects",
+                                  proj["id"]).get("repo_id")
+        app.state.git.commit_files(rid, {"base.txt": "base"}, "init")
+        task = tasks.create_task(me["id"], proj["id"], "feature", "d")
+        doc = tasks.get_task(task["id"])
+        doc["spec"] = "do it"
+        app.state.store.put("spec_tasks", task["id"], doc,
+                            owner=me["id"], parent=proj["id"])
+        manifest = {"message": "add feature",
+                    "files": {"feature.txt": "done"}}
+
+        class FakeController:
+            cfg = app.state.cfg
+
+            async def chat_completion(self, *a, **kw):
+                return {"choices": [{"message": {
+                    "content": _json.dumps(manifest)}}]}
+
+        tasks.controller = FakeController()
+        tasks.transition(task["id"], "planning")
+        tasks.transition(task["id"], "spec_review")
+        asyncio.run(tasks.implement(task["id"]))
+        assert tasks.get_task(task["id"])["state"] == "pr"
+        r = client.post(f"/api/v1/spec-tasks/{task['id']}/merge",
+                        headers=H)
+        assert r.status_code == 200, r.text
+        got = tasks.get_task(task["id"])
+        assert got["state"] == "merged" and got["merge_commit"]
+        # main now contains both files
+        tree = app.state.git.ls_tree(rid)
+        assert "feature.txt" in tree and "base.txt" in tree
