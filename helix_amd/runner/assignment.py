@@ -24,6 +24,9 @@ def apply_profile(service: RunnerService, profile: dict) -> dict:
             max_model_len=int(m.get("max_model_len", 8192)),
             max_num_seqs=int(m.get("max_num_seqs", 64)),
             kv_cache_blocks=m.get("kv_cache_blocks"),
+            tp=int(m.get("tp", 1)),
+            quantization=m.get("quantization"),
+            kv_cache_dtype=m.get("kv_cache_dtype", "bf16"),
         )
         wanted[spec.name] = spec
         service.specs[spec.name] = spec
