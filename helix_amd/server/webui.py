@@ -287,6 +287,10 @@ async function implementTask(tid){
   await fetch('/api/v1/spec-tasks/'+tid+'/implement',{method:'POST',
     headers:H()});loadBoard();
 }
+async function mergeTask(tid){
+  await fetch('/api/v1/spec-tasks/'+tid+'/merge',{method:'POST',
+    headers:H()});loadBoard();
+}
 async function loadBoard(){
   const pid=$('psel').value;const el=$('board');el.innerHTML='';
   if(!pid)return;
@@ -306,6 +310,8 @@ async function loadBoard(){
         '\')">plan</button>';
       if(st==='spec_review')btns='<button onclick="implementTask(\''+
         t.id+'\')">implement</button>';
+      if(st==='pr')btns='<button onclick="mergeTask(\''+t.id+
+        '\')">merge</button>';
       c.innerHTML='<b>'+t.title+'</b><br>'+btns+
         ' <select onchange="moveTask(\''+t.id+
         '\',this.value)"><option>move...</option>'+
