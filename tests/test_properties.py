@@ -130,3 +130,35 @@ def test_engine_fuzz_conserves_blocks(ops):
         assert guard < 2000
     assert eng.num_free_blocks() == 32
     assert len(eng._free_rows) == 4
+
+
+json_values = st.recursive(
+    st.one_of(st.none(), st.booleans(),
+              st.integers(min_value=-10**9, max_value=10**9),
+              st.floats(allow_nan=False, allow_infinity=False,
+                        width=32),
+              st.text(max_size=20)),
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(st.text(max_size=8), children, max_size=4)),
+    max_leaves=12)
+
+
+@settings(max_examples=120, deadline=None)
+@given(st.dictionaries(st.text(max_size=8), json_values, max_size=4),
+       st.sampled_from([None, (",", ":"), (", ", ": ")]),
+       st.sampled_from([None, 1, 2]))
+def test_json_mask_accepts_every_dumps(doc, seps, indent):
+    """Property: whatever json.dumps emits, the byte automaton accepts
+    byte-for-byte and reports completion (the grammar mask can never
+    corner a model out of a document json.dumps could write)."""
+    import json as _json
+
+    from helix_amd.engine.json_mode import JSONByteMask
+    raw = _json.dumps(doc, separators=seps, indent=indent).encode()
+    m = JSONByteMask(strict_object=True)
+    for b in raw:
+        assert b in m.allowed_bytes(), \
+            f"byte {bytes([b])!r} refused in {raw[:80]!r}"
+        assert m.push_byte(b)
+    assert m.complete
