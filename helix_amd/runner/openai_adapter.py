@@ -285,6 +285,10 @@ async def embeddings(service: RunnerService, req: dict) -> dict:
         "model": model,
         "data": [{"object": "embedding", "index": i, "embedding": v}
                  for i, v in enumerate(vecs)],
-        "usage": {"prompt_tokens": sum(len(t) for t in inputs),
-                  "total_tokens": sum(len(t) for t in inputs)},
+        "usage": {"prompt_tokens": sum(
+                      len(t) if isinstance(t, str) else 1
+                      for t in inputs),
+                  "total_tokens": sum(
+                      len(t) if isinstance(t, str) else 1
+                      for t in inputs)},
     }

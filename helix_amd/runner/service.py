@@ -51,6 +51,8 @@ DEFAULT_SPECS = {
     "qwen2-7b": ModelSpec("qwen2-7b", "llm", "qwen2-7b",
                           max_model_len=32768),
     "bge-base": ModelSpec("bge-base", "embedding", "bge-base"),
+    "siglip-base": ModelSpec("siglip-base", "vision", "siglip-base"),
+    "tiny-vit": ModelSpec("tiny-vit", "vision", "tiny-vit"),
     "bge-large": ModelSpec("bge-large", "embedding", "bge-large"),
     # tiny models for CPU tests
     "tiny": ModelSpec("tiny", "llm", "tiny", max_model_len=256,
@@ -65,6 +67,14 @@ def estimate_model_bytes(spec: ModelSpec, block_size: int = 16) -> int:
     """Admission estimate: weights + KV budget + workspace headroom.
     Descendant of the reference's GGUF estimator (api/pkg/memory/estimate.go)
     re-based on bf16 dense checkpoints."""
+    if spec.kind == "vision":
+        from helix_amd.models.vit import VIT_PRESETS
+        vcfg = VIT_PRESETS[spec.preset]
+        n = vcfg.patch_dim * vcfg.hidden_size + \
+            vcfg.num_patches * vcfg.hidden_size
+        per_layer = 4 * vcfg.hidden_size ** 2 + \
+            2 * vcfg.hidden_size * vcfg.intermediate_size
+        return int((n + vcfg.num_layers * per_layer) * 2 * 1.3) + (512 << 20)
     if spec.kind == "embedding":
         cfg = BERT_PRESETS[spec.preset]
         n = cfg.vocab_size * cfg.hidden_size + cfg.max_position * cfg.hidden_size
@@ -254,6 +264,50 @@ class EmbeddingInstance:
             torch.cuda.empty_cache()
 
 
+class VisionEmbeddingInstance:
+    """SigLIP2-role image embedder (reference kodit vision path):
+    base64/bytes images -> normalized vectors in the retrieval space."""
+
+    def __init__(self, spec: ModelSpec, device: str):
+        from helix_amd.models.vit import VIT_PRESETS, ViTEmbeddingModel
+        self.spec = spec
+        self.device = torch.device(device)
+        cfg = VIT_PRESETS[spec.preset]
+        dtype = torch.bfloat16 if self.device.type == "cuda"             else torch.float32
+        self.model = ViTEmbeddingModel(cfg).to(dtype).to(self.device)
+        self.model.init_random(0)
+        self.lock = threading.Lock()
+        self.last_used = time.time()
+
+    @property
+    def in_flight(self) -> int:
+        return 0
+
+    def embed(self, inputs: List) -> List[List[float]]:
+        """inputs: base64 strings, data URLs, or raw bytes."""
+        import base64 as _b64
+        self.last_used = time.time()
+        blobs = []
+        for item in inputs:
+            if isinstance(item, dict):
+                item = item.get("image", "")
+            if isinstance(item, bytes):
+                blobs.append(item)
+                continue
+            s = str(item)
+            if s.startswith("data:"):
+                s = s.split(",", 1)[-1]
+            blobs.append(_b64.b64decode(s))
+        with self.lock:
+            out = self.model.embed_images(blobs)
+        return out.cpu().tolist()
+
+    def shutdown(self):
+        del self.model
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
+
 class RunnerService:
     """Owns model instances on one GPU (or CPU for tests); packs/evicts
     under the HBM budget."""
@@ -322,6 +376,8 @@ class RunnerService:
             t0 = time.time()
             if spec.kind == "embedding":
                 inst = EmbeddingInstance(spec, self.device)
+            elif spec.kind == "vision":
+                inst = VisionEmbeddingInstance(spec, self.device)
             elif spec.tp > 1:
                 from helix_amd.runner.tp_instance import TPLLMInstance
                 inst = TPLLMInstance(spec, spec.tp)

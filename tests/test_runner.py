@@ -279,3 +279,37 @@ def test_runner_metrics_endpoint():
         assert 'helix_runner_in_flight{model="tiny"}' in r.text
     finally:
         svc.shutdown()
+
+
+def test_vision_embeddings_instance():
+    """SigLIP2-role image embedder (reference kodit vision path):
+    base64/data-URL images embed to stable normalized vectors through
+    the runner's OpenAI embeddings surface."""
+    import asyncio
+    import base64
+    import io
+
+    import numpy as np
+    from PIL import Image
+
+    from helix_amd.runner.openai_adapter import embeddings
+    from helix_amd.runner.service import RunnerService
+
+    svc = RunnerService(device="cpu")
+
+    def png(color):
+        img = Image.new("RGB", (48, 48), color)
+        buf = io.BytesIO()
+        img.save(buf, "PNG")
+        return base64.b64encode(buf.getvalue()).decode()
+
+    red, blue = png((255, 0, 0)), png((0, 0, 255))
+    out = asyncio.run(embeddings(svc, {
+        "model": "tiny-vit",
+        "input": [{"image": red}, {"image": blue},
+                  "data:image/png;base64," + red]}))
+    vecs = [np.array(d["embedding"]) for d in out["data"]]
+    assert len(vecs) == 3 and len(vecs[0]) == 128
+    assert abs(vecs[0] @ vecs[2] - 1.0) < 1e-3   # same image, same vec
+    assert vecs[0] @ vecs[1] < 0.999             # different images
+    assert abs(np.linalg.norm(vecs[0]) - 1.0) < 1e-3
