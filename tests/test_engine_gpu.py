@@ -135,3 +135,30 @@ def test_fp8_kv_engine_gpu():
     for g, w in zip(eager, want):
         assert len(g) == 12
         assert sum(a == b for a, b in zip(g, w)) >= 6
+
+
+@pytest.mark.gpu
+def test_vision_embeddings_gpu_matches_cpu():
+    """tiny-vit on the HIP kernel stack matches the CPU fp32 reference
+    path within bf16 tolerance."""
+    import base64
+    import io
+
+    import numpy as np
+    from PIL import Image
+
+    from helix_amd.runner.service import RunnerService
+
+    def png(color):
+        img = Image.new("RGB", (48, 48), color)
+        buf = io.BytesIO()
+        img.save(buf, "PNG")
+        return base64.b64encode(buf.getvalue()).decode()
+
+    imgs = [png((255, 0, 0)), png((10, 200, 30))]
+    gpu = RunnerService(device="cuda:0").ensure_loaded("tiny-vit")
+    cpu = RunnerService(device="cpu").ensure_loaded("tiny-vit")
+    vg = [np.array(v) for v in gpu.embed(imgs)]
+    vc = [np.array(v) for v in cpu.embed(imgs)]
+    for g, c in zip(vg, vc):
+        assert g @ c > 0.98, f"gpu/cpu cosine {g @ c}"
