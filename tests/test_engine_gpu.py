@@ -158,6 +158,10 @@ def test_vision_embeddings_gpu_matches_cpu():
     imgs = [png((255, 0, 0)), png((10, 200, 30))]
     gpu = RunnerService(device="cuda:0").ensure_loaded("tiny-vit")
     cpu = RunnerService(device="cpu").ensure_loaded("tiny-vit")
+    # same weights on both devices (CPU and GPU RNG streams differ)
+    gpu.model.load_state_dict(
+        {k: v.to(torch.bfloat16)
+         for k, v in cpu.model.state_dict().items()})
     vg = [np.array(v) for v in gpu.embed(imgs)]
     vc = [np.array(v) for v in cpu.embed(imgs)]
     for g, c in zip(vg, vc):
