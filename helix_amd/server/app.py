@@ -727,6 +727,22 @@ def create_app(cfg: Optional[ServerConfig] = None,
         if user is None and not is_runner:
             await ws.close(code=4401)
             return
+        privileged = is_runner or (user is not None and user.admin)
+
+        def _topic_allowed(topic: str) -> bool:
+            # non-privileged connections are scoped to their own
+            # namespaces: session events and user topics they own
+            # (otherwise any key could read every tenant's streams)
+            if privileged:
+                return True
+            uid = user.id
+            return topic.startswith((f"session.{uid}.", f"user.{uid}."))
+
+        def _stream_allowed(stream: str) -> bool:
+            if privileged:
+                return True
+            return stream.startswith(f"user-{user.id}-")
+
         await ws.accept()
         subs = {}
         forwarders = []
@@ -746,6 +762,11 @@ def create_app(cfg: Optional[ServerConfig] = None,
                 op = frame.get("op")
                 if op == "sub":
                     pattern = frame.get("pattern", "")
+                    if pattern and not _topic_allowed(pattern):
+                        await ws.send_json({"op": "error",
+                                            "message": "pattern not "
+                                                       "allowed"})
+                        continue
                     if pattern and pattern not in subs:
                         sub = await pubsub.subscribe(pattern)
                         subs[pattern] = sub
@@ -756,14 +777,32 @@ def create_app(cfg: Optional[ServerConfig] = None,
                     if sub:
                         await sub.close()
                 elif op == "pub":
-                    await pubsub.publish(frame.get("topic", ""),
-                                         frame.get("payload"))
+                    topic = frame.get("topic", "")
+                    if not _topic_allowed(topic):
+                        await ws.send_json({"op": "error",
+                                            "message": "topic not "
+                                                       "allowed"})
+                        continue
+                    await pubsub.publish(topic, frame.get("payload"))
                 elif op == "stream_pub":
+                    stream = frame.get("stream", "default")
+                    if not _stream_allowed(stream):
+                        await ws.send_json({"op": "error",
+                                            "message": "stream not "
+                                                       "allowed"})
+                        continue
                     seq = await bus.publish_notify(
-                        frame.get("stream", "default"),
-                        frame.get("subject", ""), frame.get("payload"))
+                        stream, frame.get("subject", ""),
+                        frame.get("payload"))
                     await ws.send_json({"op": "pub_ack", "seq": seq})
                 elif op == "fetch":
+                    if not _stream_allowed(frame.get("stream",
+                                                     "default")):
+                        await ws.send_json({"op": "batch",
+                                            "messages": [],
+                                            "error": "stream not "
+                                                     "allowed"})
+                        continue
                     msgs = bus.fetch(frame.get("stream", "default"),
                                      frame.get("durable", "d"),
                                      int(frame.get("batch", 10)),
@@ -771,6 +810,9 @@ def create_app(cfg: Optional[ServerConfig] = None,
                     await ws.send_json({"op": "batch",
                                         "messages": msgs})
                 elif op == "ack":
+                    if not _stream_allowed(frame.get("stream",
+                                                     "default")):
+                        continue
                     bus.ack(frame.get("stream", "default"),
                             frame.get("durable", "d"),
                             int(frame.get("seq", 0)))

@@ -101,7 +101,7 @@ def test_ws_bridge_pubsub_and_streams(tmp_path):
     app = create_app(cfg)
     with TestClient(app) as client:
         auth = app.state.auth
-        me = auth.create_user("bus-user")
+        me = auth.create_user("bus-user", admin=True)
         key = auth.create_api_key(me["id"])
         url = f"/api/v1/ws/bus?access_token={key}"
         with client.websocket_connect(url) as ws1, \
@@ -130,6 +130,20 @@ def test_ws_bridge_pubsub_and_streams(tmp_path):
             ws1.send_json({"op": "fetch", "stream": "jobs",
                            "durable": "ci-worker", "batch": 5})
             assert ws1.receive_json()["messages"] == []
+        # non-admin users are scoped to their own namespaces
+        plain = auth.create_api_key(auth.create_user("plain")["id"])
+        with client.websocket_connect(
+                f"/api/v1/ws/bus?access_token={plain}") as ws:
+            ws.send_json({"op": "sub", "pattern": "session.*"})
+            assert ws.receive_json()["op"] == "error"
+            ws.send_json({"op": "stream_pub", "stream": "jobs",
+                          "subject": "x", "payload": {}})
+            assert ws.receive_json()["op"] == "error"
+            uid = auth.resolve(plain).id
+            ws.send_json({"op": "stream_pub",
+                          "stream": f"user-{uid}-tasks",
+                          "subject": "x", "payload": {"n": 1}})
+            assert ws.receive_json()["op"] == "pub_ack"
         # unauthenticated connections are rejected
         try:
             with client.websocket_connect("/api/v1/ws/bus") as ws:
