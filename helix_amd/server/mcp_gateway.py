@@ -84,6 +84,13 @@ class MCPGateway:
         if doc is None:
             raise KeyError(app_id)
         app = App.model_validate(doc)
+        caller = self.store.get("users", owner) or {}
+        if app.owner != owner and not app.global_ and \
+                not caller.get("admin"):
+            # skills resolve with the CALLER's secrets/knowledge, but a
+            # private app's config (prompts, API endpoints) is still
+            # the owner's — same guard as the HTTP app routes
+            raise KeyError(app_id)
         assistant = app.config.assistants[0] if app.config.assistants \
             else None
         skills = []

@@ -125,3 +125,26 @@ def test_crawler_follows_same_host_links(tmp_path, monkeypatch):
     docs = asyncio.run(kn._crawl({
         "urls": ["https://docs.test/start"], "max_pages": 1}))
     assert len(docs) == 1
+
+
+def test_mcp_gateway_guards_private_apps(server):
+    app, tc, key = server
+    r = tc.post("/api/v1/apps", headers={
+        "Authorization": f"Bearer {key}"}, json={
+        "config": {"name": "private-app", "assistants": [
+            {"name": "a", "calculator": {"enabled": True}}]}})
+    app_id = r.json()["id"]
+    other = app.state.auth.create_api_key(
+        app.state.auth.create_user("intruder")["id"])
+    r = tc.post(f"/api/v1/mcp/{app_id}",
+                headers={"Authorization": f"Bearer {other}"},
+                json={"jsonrpc": "2.0", "id": 1,
+                      "method": "tools/list", "params": {}})
+    body = r.json()
+    assert "error" in body, body
+    # the owner still lists tools fine
+    r = tc.post(f"/api/v1/mcp/{app_id}",
+                headers={"Authorization": f"Bearer {key}"},
+                json={"jsonrpc": "2.0", "id": 2,
+                      "method": "tools/list", "params": {}})
+    assert "result" in r.json()
