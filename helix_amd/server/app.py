@@ -2122,6 +2122,47 @@ def create_app(cfg: Optional[ServerConfig] = None,
                 "default_model": cfg.inference.default_model,
                 "default_provider": cfg.inference.default_provider}
 
+    @app.get("/api/v1/search")
+    async def global_search(q: str = "",
+                            user: AuthUser = Depends(auth_dep)):
+        """Owner-scoped search across sessions, apps, knowledge and
+        spec tasks (the reference frontend's global search bar)."""
+        needle = q.lower().strip()
+        if not needle:
+            return {"sessions": [], "apps": [], "knowledge": [],
+                    "tasks": []}
+
+        def match(doc, *fields):
+            return any(needle in str(doc.get(f, "")).lower()
+                       for f in fields)
+
+        out = {
+            "sessions": [
+                {"id": d["id"], "name": d.get("name", "")}
+                for d in store.list("sessions", owner=user.id,
+                                    limit=2000)
+                if match(d, "name")][:20],
+            "apps": [
+                {"id": d["id"],
+                 "name": (d.get("config", {}) or {}).get("name", "")}
+                for d in store.list("apps", owner=user.id, limit=2000)
+                if needle in str((d.get("config", {}) or {}).get(
+                    "name", "")).lower()][:20],
+            "knowledge": [
+                {"id": d["id"], "name": d.get("name", ""),
+                 "state": d.get("state", "")}
+                for d in store.list("knowledge", owner=user.id,
+                                    limit=2000)
+                if match(d, "name")][:20],
+            "tasks": [
+                {"id": d["id"], "title": d.get("title", ""),
+                 "state": d.get("state", "")}
+                for d in store.list("spec_tasks", owner=user.id,
+                                    limit=2000)
+                if match(d, "title", "description")][:20],
+        }
+        return out
+
     @app.get("/healthz")
     async def healthz():
         return {"ok": True}
