@@ -57,6 +57,9 @@ INDEX_HTML = """<!doctype html>
  <div id="cfg">
   key <input id="key" size="14" value="admin-key">
   model <select id="model"></select>
+  <input id="srch" size="18" placeholder="search..."
+   onkeydown="if(event.key==='Enter')doSearch()">
+  <span id="srchout" style="font-size:12px"></span>
   <span id="status"></span>
  </div>
  <div class="pane on" id="p-chat">
@@ -378,6 +381,18 @@ async function runCmd(){
       (o.exit_code?('[exit '+o.exit_code+']\n'):'');}
   else $('sout').textContent+='[error '+r.status+']\n';
   $('scmd').value='';
+}
+async function doSearch(){
+  const q=$('srch').value.trim();if(!q)return;
+  const r=await fetch('/api/v1/search?q='+encodeURIComponent(q),
+    {headers:H()});
+  if(!r.ok){$('srchout').textContent='error';return;}
+  const o=await r.json();
+  const n=o.sessions.length+o.apps.length+o.knowledge.length+
+    o.tasks.length;
+  $('srchout').textContent=n+' hits: '+
+    o.sessions.slice(0,3).map(s=>s.name).join(', ');
+  if(o.sessions.length)openSession(o.sessions[0].id);
 }
 loadModels();loadSessions();
 </script></body></html>"""
