@@ -154,27 +154,6 @@ __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void 
   };
   if constexpr (USE_KPRE) issue_kpre(p_start);
 
-  // V-prefetch (KPRE variants): the next chunk's FIRST VPS-batch of V
-  // rows — V addresses depend only on token indices, not scores, so
-  // these loads fly during the next chunk's phases A+B and phase C
-  // starts with data already in registers.
-  u16x8 vpre[USE_KPRE ? VPS : 1];
-  int vpre_base = -1;                    // chunk base the staging is for
-  auto issue_vpre = [&](int cbase) {
-    const int cn = min(CHUNK, p_end - cbase);
-#pragma unroll
-    for (int u = 0; u < (USE_KPRE ? VPS : 1); ++u) {
-      const int tok_i = u * C_PAR + cpar;
-      const int tok = cbase + min(tok_i, cn - 1);
-      const int64_t blk = btable[tok / block_size];
-      vpre[u] = *reinterpret_cast<const u16x8*>(
-          v_cache + (((blk * Hkv + hkv) * (int64_t)block_size +
-                      tok % block_size)) * DHEAD + d8);
-    }
-    vpre_base = cbase;
-  };
-  if constexpr (USE_KPRE) issue_vpre(p_start);
-
   for (int base = p_start; base < p_end; base += CHUNK) {
     const int chunk_n = min(CHUNK, p_end - base);
 
@@ -341,27 +320,6 @@ __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void 
         }
       };
       int ps = 0;
-      if constexpr (USE_KPRE) {
-        // batch 0 was prefetched during the previous chunk's epilogue
-        // (or the preamble): consume it straight from registers
-        if (vpre_base == base) {
-#pragma unroll
-          for (int u = 0; u < VPS; ++u) {
-            if (u >= npass) break;
-            const int tok_i = u * C_PAR + cpar;
-            float v[8];
-#pragma unroll
-            for (int i = 0; i < 8; ++i) v[i] = bf16_to_f32(vpre[u][i]);
-#pragma unroll
-            for (int g = 0; g < G; ++g) {
-              const float pv = s_lds[g][tok_i];
-#pragma unroll
-              for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
-            }
-          }
-          ps = min(VPS, npass);
-        }
-      }
       if constexpr (!KV8) {
         // raw-staged batches: VPS x 16B loads issue back-to-back (pure,
         // no dependent converts between them); conversion + FMA consume
@@ -417,11 +375,6 @@ __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void 
 #pragma unroll
           for (int i = 0; i < 8; ++i) acc[g][i] += pv * v[i];
         }
-      }
-      if constexpr (USE_KPRE) {
-        // stage the NEXT chunk's first V batch now: the loads overlap
-        // its phase A math + phase B softmax
-        if (base + CHUNK < p_end) issue_vpre(base + CHUNK);
       }
     }
     __syncthreads();
@@ -609,7 +562,14 @@ void paged_attn_decode(torch::Tensor out, torch::Tensor q,
 
   const int max_len = (int)max_len_hint;
   const int max_useful = cdiv(max_len, PART_QUANT);
-  int nparts = std::max(1, 1024 / std::max(1, B * Hkv));
+  // split-K sizing target: enough workgroups to fill 256 CUs. 1024 was
+  // round-1's tuning; HELIX_DECODE_TARGET_WGS re-tunes it for the
+  // round-2 pipeline on underfilled (small-B, long-L) shapes.
+  static int target_wgs = [] {
+    const char* e = getenv("HELIX_DECODE_TARGET_WGS");
+    return e ? atoi(e) : 1024;
+  }();
+  int nparts = std::max(1, target_wgs / std::max(1, B * Hkv));
   nparts = std::min({nparts, max_useful, max_parts});
   int eff_part = cdiv(cdiv(max_len, nparts), PART_QUANT) * PART_QUANT;
   nparts = cdiv(max_len, eff_part);
