@@ -284,7 +284,8 @@ def create_app(cfg: Optional[ServerConfig] = None,
     # ------------------------------------------------------------------
     async def _chat(request: Request, user: AuthUser, body: dict):
         app_id = body.pop("app_id", "") or \
-            request.headers.get("X-Helix-App-Id", "")
+            request.headers.get("X-Helix-App-Id", "") or \
+            user.app_id
         assistant_id = body.pop("assistant_id", "")
         ctx = {"owner": user.id}
         if body.get("stream"):
@@ -916,6 +917,33 @@ def create_app(cfg: Optional[ServerConfig] = None,
                         store.delete("triggers", t["id"])
                 return {"ok": True}
         _bind()
+
+    @app.post("/api/v1/apps/{app_id}/keys")
+    async def create_app_key(app_id: str, request: Request,
+                             user: AuthUser = Depends(auth_dep)):
+        """App-scoped API key (reference GetAppAPIKeys, client/app.go:
+        44): requests made with it default to this app."""
+        a = _get_app(app_id, user)
+        if a.owner != user.id and not user.admin:
+            raise HTTPException(403, "only the owner can mint app keys")
+        try:
+            body = await request.json()
+        except Exception:
+            body = {}
+        key = auth.create_api_key(a.owner,
+                                  body.get("name", f"app-{app_id[:8]}"),
+                                  app_id=app_id)
+        return {"key": key, "app_id": app_id}
+
+    @app.get("/api/v1/apps/{app_id}/keys")
+    async def list_app_keys(app_id: str,
+                            user: AuthUser = Depends(auth_dep)):
+        a = _get_app(app_id, user)
+        if a.owner != user.id and not user.admin:
+            raise HTTPException(403, "forbidden")
+        return [{"id": k["id"][:12] + "...", "name": k.get("name", "")}
+                for k in store.list("api_keys", owner=a.owner)
+                if k.get("app_id") == app_id]
 
     # ------------------------------------------------------------------
     # Local-model admin (reference local-models/load|unload,

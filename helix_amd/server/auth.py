@@ -58,6 +58,8 @@ class AuthUser:
     id: str
     username: str = ""
     admin: bool = False
+    app_id: str = ""      # set for app-scoped API keys (client/app.go
+                          # GetAppAPIKeys): requests default to this app
 
 
 class Authenticator:
@@ -73,10 +75,12 @@ class Authenticator:
         self.store.put("users", uid, doc, owner=uid)
         return doc
 
-    def create_api_key(self, owner: str, name: str = "default") -> str:
+    def create_api_key(self, owner: str, name: str = "default",
+                       app_id: str = "") -> str:
         key = f"hl-{secrets.token_hex(24)}"
         self.store.put("api_keys", key, {"id": key, "owner": owner,
-                                         "name": name}, owner=owner)
+                                         "name": name,
+                                         "app_id": app_id}, owner=owner)
         return key
 
     # -- JWT (self-issued HS256) --------------------------------------
@@ -112,7 +116,8 @@ class Authenticator:
         user = self.store.get("users", doc["owner"])
         return AuthUser(id=doc["owner"],
                         username=(user or {}).get("username", ""),
-                        admin=(user or {}).get("admin", False))
+                        admin=(user or {}).get("admin", False),
+                        app_id=doc.get("app_id", ""))
 
     def is_runner(self, token: str) -> bool:
         return bool(token) and hmac.compare_digest(token,

@@ -127,3 +127,52 @@ def test_helix_yaml_triggers_register(tmp_path):
         # delete clears app-config triggers
         client.delete(f"/api/v1/apps/{aid}", headers=H)
         assert client.get("/api/v1/triggers", headers=H).json() == []
+
+
+def test_app_scoped_api_keys(tmp_path):
+    """App API keys (reference client/app.go:44 GetAppAPIKeys):
+    requests made with an app key default to that app's assistant."""
+    import json as _json
+
+    from fastapi.testclient import TestClient
+
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import ServerConfig
+    from helix_amd.server.providers import MockClient, ProviderManager
+    from helix_amd.store import Store
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    cfg.filestore.path = str(tmp_path / "fs")
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    pm.register("mock", MockClient())
+    app = create_app(cfg, store=store, providers=pm)
+    with TestClient(app) as client:
+        auth = app.state.auth
+        me = auth.create_user("owner-ak")
+        key = auth.create_api_key(me["id"])
+        H = {"Authorization": f"Bearer {key}"}
+        r = client.post("/api/v1/apps", headers=H, json={"config": {
+            "name": "sys-app", "assistants": [
+                {"name": "a", "system_prompt": "ALWAYS SAY BANANA"}]}})
+        aid = r.json()["id"]
+        akey = client.post(f"/api/v1/apps/{aid}/keys", headers=H,
+                           json={}).json()["key"]
+        r = client.post("/v1/chat/completions",
+                        headers={"Authorization": f"Bearer {akey}"},
+                        json={"model": "mock-model",
+                              "messages": [{"role": "user",
+                                            "content": "hi"}]})
+        assert r.status_code == 200, r.text
+        calls = store.list("llm_calls", limit=5)
+        assert any("BANANA" in _json.dumps(c) for c in calls)
+        # listing shows the key (truncated), other users cannot mint
+        assert len(client.get(f"/api/v1/apps/{aid}/keys",
+                              headers=H).json()) == 1
+        intruder = auth.create_api_key(
+            auth.create_user("intruder-ak")["id"])
+        r = client.post(f"/api/v1/apps/{aid}/keys",
+                        headers={"Authorization": f"Bearer {intruder}"},
+                        json={})
+        assert r.status_code in (403, 404)
