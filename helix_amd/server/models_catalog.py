@@ -15,15 +15,24 @@ STATIC_MODEL_INFO: Dict[str, dict] = {
     "llama3-70b": {"context_length": 8192, "prompt_price_per_m": 0.6,
                    "completion_price_per_m": 0.8, "family": "llama",
                    "runtime": "helix_amd", "kind": "chat"},
+    "llama3.1-8b": {"context_length": 32768, "prompt_price_per_m": 0.05,
+                    "completion_price_per_m": 0.10, "family": "llama",
+                    "runtime": "helix_amd", "kind": "chat"},
     "mistral-7b": {"context_length": 8192, "prompt_price_per_m": 0.05,
                    "completion_price_per_m": 0.10, "family": "mistral",
                    "runtime": "helix_amd", "kind": "chat"},
+    "qwen2-7b": {"context_length": 32768, "prompt_price_per_m": 0.05,
+                 "completion_price_per_m": 0.10, "family": "qwen",
+                 "runtime": "helix_amd", "kind": "chat"},
     "bge-base": {"context_length": 512, "prompt_price_per_m": 0.005,
                  "completion_price_per_m": 0.0, "family": "bge",
                  "runtime": "helix_amd", "kind": "embedding"},
     "bge-large": {"context_length": 512, "prompt_price_per_m": 0.01,
                   "completion_price_per_m": 0.0, "family": "bge",
                   "runtime": "helix_amd", "kind": "embedding"},
+    "siglip-base": {"context_length": 0, "prompt_price_per_m": 0.01,
+                    "completion_price_per_m": 0.0, "family": "siglip",
+                    "runtime": "helix_amd", "kind": "vision-embedding"},
     "gpt-4o": {"context_length": 128000, "prompt_price_per_m": 2.5,
                "completion_price_per_m": 10.0, "family": "openai",
                "runtime": "external", "kind": "chat"},
