@@ -54,13 +54,16 @@ class RAGService:
                                          self.cfg.rag.chunk_overlap, md))
         return chunks
 
-    async def index(self, knowledge_id: str, documents: List[dict]) -> int:
+    async def index(self, knowledge_id: str, documents: List[dict],
+                    progress=None) -> int:
         """documents: [{text, metadata?}] -> chunk, embed, store, swap."""
         return await self.index_chunks(knowledge_id,
-                                       self._chunks_of(documents))
+                                       self._chunks_of(documents),
+                                       progress=progress)
 
     async def index_chunks(self, knowledge_id: str,
-                           chunks: List[dict]) -> int:
+                           chunks: List[dict],
+                           progress=None) -> int:
         alias = self.store.get("rag_alias", knowledge_id) or \
             {"id": knowledge_id, "ns": knowledge_id, "version": 0}
         old_ns = alias["ns"]
@@ -71,6 +74,8 @@ class RAGService:
             batch = chunks[i:i + B]
             vecs = await self._embed([c["text"] for c in batch])
             self.vs.add(ns, batch, vecs)
+            if progress is not None:
+                progress(min(i + B, len(chunks)), len(chunks))
         # atomic swap: queries resolve through the alias row
         self.store.put("rag_alias", knowledge_id,
                        {"id": knowledge_id, "ns": ns, "version": ver})

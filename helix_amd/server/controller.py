@@ -309,6 +309,15 @@ class Controller:
                 yield chunk
             interaction.response_message = text
             interaction.state = InteractionState.COMPLETE
+            # auto-title after the first completed turn (reference:
+            # the controller names sessions with the small model)
+            if session.name in ("", "New Session") or \
+                    session.name == (interaction.prompt_message
+                                     or "")[:40]:
+                try:
+                    await self._autotitle(session, interaction, text)
+                except Exception:
+                    log.debug("session auto-title failed", exc_info=True)
         except Exception as e:
             interaction.state = InteractionState.ERROR
             interaction.error = str(e)
@@ -330,3 +339,29 @@ class Controller:
                     ps.session_queue(session.owner, session.id),
                     {"type": "done", "interaction_id": interaction.id,
                      "message": text, "state": interaction.state.value})
+
+    async def _autotitle(self, session: Session,
+                         interaction: Interaction, answer: str):
+        """Name the session from its first exchange using the default
+        provider (small-model role); falls back silently."""
+        provider = self.cfg.inference.default_provider
+        client = self.providers.get_client(provider, session.owner)
+        resp = await client.chat({
+            "model": session.model_name or
+            self.cfg.inference.default_model,
+            "max_tokens": 16,
+            "messages": [
+                {"role": "system",
+                 "content": "Reply with a 3-6 word title for this "
+                            "conversation. No quotes."},
+                {"role": "user",
+                 "content": f"Q: {interaction.prompt_message[:400]}\n"
+                            f"A: {answer[:400]}"}]})
+        title = resp["choices"][0]["message"].get("content", "").strip()
+        title = title.strip('"' + "'").splitlines()[0][:60]
+        if title:
+            session.name = title
+            session.updated = now_ms()
+            self.store.put("sessions", session.id, session.model_dump(),
+                           owner=session.owner,
+                           parent=session.parent_app)

@@ -244,8 +244,15 @@ class KnowledgeReconciler:
         else:
             raise ValueError(f"unsupported knowledge source: {list(src)}")
         # versioned swap happens inside the RAG layer: the previous
-        # index keeps serving queries until the new one is complete
-        count = await self.rag.index(doc["id"], documents)
+        # index keeps serving queries until the new one is complete;
+        # batch progress lands on the row (reference progress_percent)
+        def _progress(done: int, total: int):
+            doc["progress_percent"] = int(100 * done / max(1, total))
+            self._save(doc)
+
+        count = await self.rag.index(doc["id"], documents,
+                                     progress=_progress)
+        doc["progress_percent"] = 100
         if skipped:
             doc["_skipped"] = True
             doc["_skip_note"] = f"skipped {len(skipped)}: " + \

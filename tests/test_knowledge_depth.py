@@ -217,7 +217,7 @@ def test_async_jobs_do_not_block_other_rows(platform):
         real_index = rag.index
         gate = asyncio.Event()
 
-        async def gated_index(kid, docs):
+        async def gated_index(kid, docs, progress=None):
             if any("SLOW" in d.get("text", "") for d in docs):
                 await gate.wait()
             return await real_index(kid, docs)
@@ -264,3 +264,24 @@ def test_indexing_recovery_after_restart(platform):
     asyncio.run(kn.reconcile_once())   # re-queue
     asyncio.run(kn.reconcile_once())
     assert kn.get(doc["id"])["state"] == "ready"
+
+
+def test_indexing_progress_lands_on_row(platform):
+    cfg, store, rag, kn = platform
+    big = "\n\n".join(f"paragraph number {i} " + "x" * 400
+                      for i in range(40))
+    doc = kn.create("u1", "big", {"text": big})
+    seen = []
+    orig_save = kn._save
+
+    def spy(d):
+        if "progress_percent" in d:
+            seen.append(d["progress_percent"])
+        orig_save(d)
+    kn._save = spy
+    asyncio.run(kn.reconcile_once())
+    asyncio.run(kn.reconcile_once())
+    got = kn.get(doc["id"])
+    assert got["state"] == "ready"
+    assert got["progress_percent"] == 100
+    assert seen and seen[0] <= 100 and sorted(seen) == seen
