@@ -94,7 +94,6 @@ def runner_responses_queue(owner: str, request_id: str) -> str:
 # fetched messages are leased for ack_wait_s, unacked leases expire and
 # the message is redelivered.
 
-import json as _json
 import time as _time
 
 
