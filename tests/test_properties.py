@@ -162,3 +162,55 @@ def test_json_mask_accepts_every_dumps(doc, seps, indent):
             f"byte {bytes([b])!r} refused in {raw[:80]!r}"
         assert m.push_byte(b)
     assert m.complete
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.sampled_from(["pub", "fetch", "ack", "expire"]),
+                min_size=5, max_size=40),
+       st.integers(min_value=1, max_value=4))
+def test_stream_bus_at_least_once(ops_seq, batch):
+    """Property: any interleaving of publish/fetch/ack/lease-expiry
+    delivers every published message at least once, never delivers a
+    message while it is leased, and acked messages never reappear."""
+    from helix_amd.server.pubsub import StreamBus
+    from helix_amd.store import Store
+
+    st_ = Store(":memory:")
+    bus = StreamBus(st_, ack_wait_s=1e9)     # expiry only when forced
+    published, acked, leased = set(), set(), {}
+    n = 0
+    for op in ops_seq:
+        if op == "pub":
+            n += 1
+            seq = bus.publish("s", "subj", {"n": n})
+            published.add(seq)
+        elif op == "fetch":
+            got = bus.fetch("s", "w", batch=batch)
+            for m in got:
+                assert m["seq"] not in leased, "delivered while leased"
+                assert m["seq"] not in acked, "delivered after ack"
+                leased[m["seq"]] = True
+        elif op == "ack" and leased:
+            seq = sorted(leased)[0]
+            del leased[seq]
+            bus.ack("s", "w", seq)
+            acked.add(seq)
+        elif op == "expire" and leased:
+            # force every lease to expire
+            doc = st_.get("bus_consumers", "s:w")
+            if doc:
+                doc["inflight"] = {k: 0.0 for k in doc["inflight"]}
+                st_.put("bus_consumers", "s:w", doc)
+            leased.clear()
+    # drain: everything published must eventually deliver
+    doc = st_.get("bus_consumers", "s:w")
+    if doc:
+        doc["inflight"] = {k: 0.0 for k in doc["inflight"]}
+        st_.put("bus_consumers", "s:w", doc)
+    seen = set(acked)
+    for _ in range(len(published) + 1):
+        for m in bus.fetch("s", "w", batch=50):
+            seen.add(m["seq"])
+            bus.ack("s", "w", m["seq"])
+    assert seen >= published, f"lost {published - seen}"
+    st_.close()
