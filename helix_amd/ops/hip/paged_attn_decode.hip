@@ -142,17 +142,27 @@ __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void 
   // chunk ahead (16 b128 loads live across phases B+C of the previous
   // chunk — the latency they need to hide).
   constexpr bool USE_KPRE = KPRE && !MFMA_A && !KV8;
+  constexpr bool USE_KPRE8 = KPRE && !MFMA_A && KV8;
   u16x8 kpre[USE_KPRE ? DHEAD / 8 : 1];
+  u8x16 kpre8[USE_KPRE8 ? DHEAD / 16 : 1];
   auto issue_kpre = [&](int cbase) {
     const int tok = min(cbase + (int)threadIdx.x, p_end - 1);
-    const uint16_t* krow =
-        k_cache + (((int64_t)btable[tok / block_size] * Hkv + hkv) *
-                       (int64_t)block_size + tok % block_size) * DHEAD;
+    const int64_t roff =
+        (((int64_t)btable[tok / block_size] * Hkv + hkv) *
+             (int64_t)block_size + tok % block_size) * DHEAD;
+    if constexpr (USE_KPRE8) {
+      const uint8_t* krow = (const uint8_t*)k_cache + roff;
 #pragma unroll
-    for (int j = 0; j < (USE_KPRE ? DHEAD / 8 : 1); ++j)
-      kpre[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+      for (int j = 0; j < DHEAD / 16; ++j)
+        kpre8[j] = *reinterpret_cast<const u8x16*>(krow + j * 16);
+    } else {
+      const uint16_t* krow = k_cache + roff;
+#pragma unroll
+      for (int j = 0; j < (USE_KPRE ? DHEAD / 8 : 1); ++j)
+        kpre[j] = *reinterpret_cast<const u16x8*>(krow + j * 8);
+    }
   };
-  if constexpr (USE_KPRE) issue_kpre(p_start);
+  if constexpr (USE_KPRE || USE_KPRE8) issue_kpre(p_start);
 
   for (int base = p_start; base < p_end; base += CHUNK) {
     const int chunk_n = min(CHUNK, p_end - base);
@@ -219,6 +229,27 @@ __global__ __launch_bounds__(NTHREADS, (OCC > 0 ? OCC : (G <= 2 ? 4 : 3))) void 
       }
       // scores consumed: issue the NEXT chunk's K rows now so they
       // fly during softmax + V accumulation
+      if (base + CHUNK < p_end) issue_kpre(base + CHUNK);
+    } else if constexpr (USE_KPRE8) {
+      if ((int)threadIdx.x < chunk_n) {
+        float s[G];
+#pragma unroll
+        for (int g = 0; g < G; ++g) s[g] = 0.f;
+#pragma unroll
+        for (int j = 0; j < DHEAD / 16; ++j) {
+          float kv[16];
+#pragma unroll
+          for (int i = 0; i < 16; ++i) kv[i] = fp8_to_f32(kpre8[j][i]);
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+#pragma unroll
+            for (int i = 0; i < 16; ++i)
+              s[g] += q_lds[g][j * 16 + i] * kv[i];
+          }
+        }
+#pragma unroll
+        for (int g = 0; g < G; ++g) s_lds[g][threadIdx.x] = s[g];
+      }
       if (base + CHUNK < p_end) issue_kpre(base + CHUNK);
     } else if ((int)threadIdx.x < chunk_n) {
       const int tok = base + threadIdx.x;
@@ -508,7 +539,8 @@ void launch_decode(uint16_t* out, float* tmp_out, float* tmp_ml,
       tmp_out, tmp_ml, q, kc, vc, bt, lens, scale, Hkv,                      \
       block_size, max_blocks, eff_part, max_parts, window)
   if (kv8) {
-    LAUNCH_PD(false, true, 4, 0, false);
+    if (decode_kpre()) LAUNCH_PD(false, true, 4, 2, true);
+    else               LAUNCH_PD(false, true, 4, 0, false);
   } else if (use_mfma_a()) {
     LAUNCH_PD(true, false, 4, 0, false);
   } else {
