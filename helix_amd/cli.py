@@ -187,6 +187,30 @@ def knowledge_list(url: str = typer.Option("", "--url")):
                    f"v{k.get('version', 0)} chunks={k.get('chunks', 0)}")
 
 
+@knowledge_app.command("create")
+def knowledge_create(name: str,
+                     text: str = typer.Option("", help="inline text"),
+                     path: str = typer.Option("", help="filestore path"),
+                     s3: str = typer.Option("", help="bucket/prefix"),
+                     url: str = typer.Option("", "--web-url"),
+                     refresh_schedule: str = typer.Option("")):
+    """Create a knowledge source (text / filestore / s3 / web)."""
+    if text:
+        source = {"text": text}
+    elif path:
+        source = {"filestore": {"path": path}}
+    elif s3:
+        bucket, _, prefix = s3.partition("/")
+        source = {"s3": {"bucket": bucket, "prefix": prefix}}
+    elif url:
+        source = {"web": {"urls": [url]}}
+    else:
+        typer.echo("one of --text/--path/--s3/--web-url required")
+        raise typer.Exit(2)
+    doc = _client().create_knowledge(name, source, refresh_schedule)
+    typer.echo(f"{doc['id']}  {doc['state']}")
+
+
 @knowledge_app.command("refresh")
 def knowledge_refresh(kid: str, url: str = typer.Option("", "--url")):
     import httpx

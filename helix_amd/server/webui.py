@@ -220,14 +220,21 @@ async function addKnowledge(){
       source:{text:$('ktext').value}})});
   loadKnowledge();
 }
+async function refreshKnw(kid){
+  await fetch('/api/v1/knowledge/'+kid+'/refresh',{method:'POST',
+    headers:H()});loadKnowledge();
+}
 async function loadKnowledge(){
   const r=await fetch('/api/v1/knowledge',{headers:H()});if(!r.ok)return;
   const el=$('knwlist');el.innerHTML='';
   for(const k of await r.json()){
     const d=document.createElement('div');d.className='card';
+    const prog=(k.state==='indexing'&&k.progress_percent!=null)
+      ?(' '+k.progress_percent+'%'):'';
     d.innerHTML='<b>'+k.name+'</b> <span class="pill'+
-      (k.state==='error'?' err':'')+'">'+k.state+'</span>'+
-      (k.version?' v'+k.version:'');
+      (k.state==='error'?' err':'')+'">'+k.state+prog+'</span>'+
+      (k.version?' v'+k.version:'')+
+      ' <button onclick="refreshKnw(\''+k.id+'\')">refresh</button>';
     el.appendChild(d);}
 }
 // ---- runners pane ----
