@@ -559,13 +559,20 @@ class LLMEngine:
         needs_proc = any(s.params.needs_logit_processing for s in batch)
         gpu_fast = logits.is_cuda and needs_proc
         if needs_proc and not gpu_fast:
-            logits = self._process_logits(batch, logits.float())
+            # float() is a no-op on fp32 CPU engines, and the model
+            # forward runs under inference_mode: clone so the in-place
+            # top-k/top-p/penalty edits are legal
+            logits = self._process_logits(
+                batch, logits.float().clone()
+                if logits.is_inference() else logits.float())
         elif gpu_fast and lp_rows:
             # logprob rows report over the FILTERED distribution, so they
             # keep the reference host path; their kernel params are
             # neutralized below (no double filtering/penalties)
-            sub = logits[lp_rows].float()
+            sub = logits[lp_rows].float().clone()
             self._process_logits([batch[i] for i in lp_rows], sub)
+            if logits.is_inference():
+                logits = logits.clone()
             logits[lp_rows] = sub.to(logits.dtype)
         on_gpu = logits.is_cuda
         if on_gpu:
@@ -754,7 +761,12 @@ class LLMEngine:
     def generate(self, prompts: List[List[int]],
                  params: SamplingParams) -> List[List[int]]:
         """Synchronous batch generate (used by tests and bench)."""
-        ids = [f"gen-{id(prompts)}-{i}" for i in range(len(prompts))]
+        # ids must be identical across SPMD TP ranks: the seedless
+        # sampling fallback seeds from crc32(seq_id), so an id built
+        # from a memory address (id(prompts)) silently diverges ranks
+        self._gen_calls = getattr(self, "_gen_calls", 0) + 1
+        ids = [f"gen-{self._gen_calls}-{i}"
+               for i in range(len(prompts))]
         for sid, p in zip(ids, prompts):
             self.add_request(sid, p, params)
         while self.has_work:

@@ -215,3 +215,66 @@ def test_tp2_temperature_sampling_lockstep():
         r0 = torch.load(f"{out_path}.0")
         r1 = torch.load(f"{out_path}.1")
     assert r0 == r1 == want
+
+
+def _tp_feature_worker(rank, world, sd, prompts, out_path, port):
+    """Rank worker exercising the round-2 sampling features under
+    SPMD: seedless temperature sampling over MULTIPLE steps (the
+    cached staging must advance seeds identically on every rank),
+    JSON-mode masking, and logit_bias."""
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      MASTER_ADDR="127.0.0.1", MASTER_PORT=port,
+                      LOCAL_RANK=str(rank))
+    from helix_amd import parallel
+    from helix_amd.engine.engine import EngineConfig, LLMEngine
+    from helix_amd.engine.sampling_params import SamplingParams
+    parallel.init_tp(world, backend="gloo")
+    cfg = PRESETS["tiny-gqa"]
+    model = LlamaForCausalLM(cfg, tp_size=world, tp_rank=rank).float()
+    model.load_state_dict(shard_llama_state_dict(sd, cfg, world, rank),
+                          strict=True)
+    eng = LLMEngine(EngineConfig(model="tiny-gqa", max_model_len=256,
+                                 max_num_seqs=4, kv_cache_blocks=128,
+                                 eos_token_id=-1),
+                    device="cpu", model=model, tp_size=world,
+                    tp_rank=rank)
+    outs = []
+    # seedless sampling (deterministic crc32 fallback + cached
+    # staging advancing outlens per step)
+    outs.append(eng.generate(
+        prompts, SamplingParams(temperature=0.9, top_p=0.8,
+                                max_tokens=10, ignore_eos=True)))
+    # json grammar masking
+    outs.append(eng.generate(
+        [prompts[0]], SamplingParams(temperature=1.0, seed=None,
+                                     max_tokens=10, ignore_eos=True,
+                                     json_mode=True)))
+    # logit bias pins the output
+    outs.append(eng.generate(
+        [prompts[1]], SamplingParams(temperature=0.7, max_tokens=4,
+                                     ignore_eos=True,
+                                     logit_bias={33: 500.0})))
+    if out_path:
+        torch.save(outs, out_path + f".r{rank}")
+    torch.distributed.destroy_process_group()
+
+
+def test_tp2_round2_sampling_features_lockstep():
+    """Both ranks must emit IDENTICAL tokens for seedless sampling,
+    JSON mode, and logit_bias — any divergence silently corrupts TP
+    state (ADVICE r1 high-severity class of bug)."""
+    import tempfile
+    cfg = PRESETS["tiny-gqa"]
+    torch.manual_seed(0)
+    full = LlamaForCausalLM(cfg).float()
+    full.init_random(0)
+    sd = full.state_dict()
+    prompts = [[1, 2, 3, 4, 5], [9, 8, 7]]
+    with tempfile.TemporaryDirectory() as td:
+        out_path = os.path.join(td, "out.pt")
+        _spawn2(_tp_feature_worker,
+                lambda r, port: (r, 2, sd, prompts, out_path, port))
+        r0 = torch.load(out_path + ".r0")
+        r1 = torch.load(out_path + ".r1")
+    assert r0 == r1, "rank outputs diverged"
+    assert r0[2][0] == [33, 33, 33, 33]    # bias honored under TP
