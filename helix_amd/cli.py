@@ -253,6 +253,28 @@ def session_list(url: str = typer.Option("", "--url")):
         typer.echo(f"{s['id']}  {s['name'][:40]}")
 
 
+@session_app.command("export")
+def session_export(session_id: str,
+                   out: str = typer.Option("", "-o")):
+    """Export a session with its interactions as JSON (docs say this
+    exists; backup/portability role)."""
+    c = _client()
+    doc = c.get_session(session_id)
+    blob = json.dumps(doc, indent=2, default=str)
+    if out:
+        with open(out, "w") as fh:
+            fh.write(blob)
+        typer.echo(f"wrote {out}")
+    else:
+        typer.echo(blob)
+
+
+@session_app.command("delete")
+def session_delete(session_id: str):
+    _client().delete_session(session_id)
+    typer.echo("deleted")
+
+
 model_app = typer.Typer(help="Model catalog & local models")
 app.add_typer(model_app, name="model")
 
