@@ -44,7 +44,7 @@ constexpr int PART_QUANT = 128;
 // bytes in flight per wave before any conversion VALU touches them.
 // OCC: launch_bounds waves/SIMD floor override (0 = round-1 default).
 // KPRE: inter-chunk K prefetch. The r2 ISA audit (profiles/
-// r02_decode_isa.md) showed the compiler sinking phase A's 16-deep raw
+// r02_decode_kpre.md) showed the compiler sinking phase A's 16-deep raw
 // K staging into the consume loop — steady state s_waitcnt vmcnt(1),
 // i.e. only ~2 VMEM loads in flight per wave, matching the 14:1
 // wait:busy PMC. KPRE hoists the next chunk's 16 b128 K loads to just
