@@ -186,6 +186,17 @@ class BillingService:
             log.warning("top-up event with unknown customer: %s",
                         obj.get("customer"))
             return
+        # one PAYMENT can arrive as both checkout.session.completed and
+        # payment_intent.succeeded (distinct event ids): credit once
+        # per payment, keyed by the payment_intent (or the session id)
+        pay_id = obj.get("payment_intent") or obj.get("id", "")
+        if pay_id:
+            guard = f"pay:{pay_id}"
+            if self.store.get("billing_events", guard):
+                return
+            self.store.put("billing_events", guard,
+                           {"id": guard, "type": "payment-credit",
+                            "ts": time.time()})
         md = obj.get("metadata") or {}
         cents = int(md.get("amount_cents") or
                     obj.get("amount_total") or
@@ -193,7 +204,7 @@ class BillingService:
         if cents <= 0:
             return
         self.usage.topup(uid, cents / 100.0,
-                         ref=f"stripe:{obj.get('id', '')}")
+                         ref=f"stripe:{pay_id}")
 
     def _handle_subscription(self, etype: str, obj: dict):
         uid = self._user_for_customer(obj)

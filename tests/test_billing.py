@@ -132,3 +132,20 @@ def test_http_surface(tmp_path, monkeypatch):
         r = client.post("/api/v1/stripe/webhook", content=payload,
                         headers={"Stripe-Signature": "t=1,v1=bad"})
         assert r.status_code == 400
+
+
+def test_checkout_and_payment_intent_credit_once(svc):
+    """Stripe sends BOTH checkout.session.completed and
+    payment_intent.succeeded for one payment (distinct event ids):
+    the wallet must be credited exactly once."""
+    store, usage, billing = svc
+    cs = {"id": "cs_dup", "payment_intent": "pi_dup",
+          "metadata": {"user_id": "u9", "amount_cents": "1000"}}
+    pi = {"id": "pi_dup",
+          "metadata": {"user_id": "u9", "amount_cents": "1000"},
+          "amount_received": 1000}
+    p1 = _event("checkout.session.completed", cs, eid="evt_c1")
+    p2 = _event("payment_intent.succeeded", pi, eid="evt_p1")
+    billing.process_webhook(p1, sign_stripe_payload(p1, SECRET))
+    billing.process_webhook(p2, sign_stripe_payload(p2, SECRET))
+    assert usage.wallet("u9")["balance_usd"] == pytest.approx(10.0)
