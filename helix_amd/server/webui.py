@@ -121,6 +121,9 @@ INDEX_HTML = """<!doctype html>
 <script>
 let sessionId = null;
 const $ = id => document.getElementById(id);
+const esc = x => String(x==null?'':x).replace(/[&<>"']/g,
+  ch => ({'&':'&amp;','<':'&lt;','>':'&gt;','"':'&quot;',
+          "'":'&#39;'}[ch]));
 const H = () => ({'Authorization':'Bearer '+$('key').value,
                   'Content-Type':'application/json'});
 for(const b of document.querySelectorAll('#tabs button')){
@@ -208,7 +211,7 @@ async function loadApps(){
   for(const a of await r.json()){
     const d=document.createElement('div');d.className='card';
     const cfg=a.config?.helix||a.config||{};
-    d.innerHTML='<b>'+(cfg.name||a.id)+'</b> <span class="pill">'+
+    d.innerHTML='<b>'+esc(cfg.name||a.id)+'</b> <span class="pill">'+
       (cfg.assistants?.length||0)+' assistants</span><br><small>'+a.id+
       '</small>';
     el.appendChild(d);}
@@ -231,7 +234,7 @@ async function loadKnowledge(){
     const d=document.createElement('div');d.className='card';
     const prog=(k.state==='indexing'&&k.progress_percent!=null)
       ?(' '+k.progress_percent+'%'):'';
-    d.innerHTML='<b>'+k.name+'</b> <span class="pill'+
+    d.innerHTML='<b>'+esc(k.name)+'</b> <span class="pill'+
       (k.state==='error'?' err':'')+'">'+k.state+prog+'</span>'+
       (k.version?' v'+k.version:'')+
       ' <button onclick="refreshKnw(\''+k.id+'\')">refresh</button>';
@@ -247,7 +250,7 @@ async function loadRunners(){
   for(const x of rs){
     const d=document.createElement('div');d.className='card';
     const models=(x.models||[]).map(m=>m.model_id||m).join(', ');
-    d.innerHTML='<b>'+x.id+'</b> <span class="pill">'+
+    d.innerHTML='<b>'+esc(x.id)+'</b> <span class="pill">'+
       (x.status||'ready')+'</span><br>GPU: '+(x.gpu||'?')+
       '<br>models: '+models;
     el.appendChild(d);}
@@ -322,7 +325,7 @@ async function loadBoard(){
         t.id+'\')">implement</button>';
       if(st==='pr')btns='<button onclick="mergeTask(\''+t.id+
         '\')">merge</button>';
-      c.innerHTML='<b>'+t.title+'</b><br>'+btns+
+      c.innerHTML='<b>'+esc(t.title)+'</b><br>'+btns+
         ' <select onchange="moveTask(\''+t.id+
         '\',this.value)"><option>move...</option>'+
         STATES.map(x=>'<option>'+x+'</option>').join('')+'</select>';
@@ -341,7 +344,7 @@ async function loadOrgs(){
   const el=$('orglist');el.innerHTML='';
   for(const o of await r.json()){
     const d=document.createElement('div');d.className='card';
-    d.innerHTML='<b>'+(o.name||o.id)+'</b> <small>'+o.id+'</small>'+
+    d.innerHTML='<b>'+esc(o.name||o.id)+'</b> <small>'+o.id+'</small>'+
       '<div id="teams-'+o.id+'"></div>'+
       '<input id="tn-'+o.id+'" placeholder="team name" size="12">'+
       '<button onclick="newTeam(\''+o.id+'\')">add team</button> '+
