@@ -285,3 +285,27 @@ def test_indexing_progress_lands_on_row(platform):
     assert got["state"] == "ready"
     assert got["progress_percent"] == 100
     assert seen and seen[0] <= 100 and sorted(seen) == seen
+
+
+def test_extract_html_malformed_inputs():
+    """Readability must never raise on hostile/malformed markup."""
+    cases = [
+        "",
+        "just plain text, no tags",
+        "<div><p>unclosed everywhere",
+        "<html><body>" + "<div>" * 2000 + "deep" + "</div>" * 10,
+        "<p>" + "x" * 200000 + "</p>",
+        "<script>while(1){}</script><p>content that stays here ok</p>",
+        "<a href='/x'>" * 500,
+        "\x00\x01<binaryish>\xff content",
+        "<p>ünïcødé ✓ \U0001F600 content long enough to keep</p>",
+    ]
+    for html in cases:
+        out = extract_html(html)
+        assert isinstance(out["text"], str)
+        assert isinstance(out["links"], list)
+    out = extract_html(cases[5])
+    assert "while(1)" not in out["text"]
+    assert "content that stays" in out["text"]
+    out = extract_html(cases[8])
+    assert "ünïcødé" in out["text"]
