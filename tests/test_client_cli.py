@@ -127,3 +127,19 @@ def test_client_chat_and_stream(tmp_path):
                        for ch in chunks if ch.get("choices"))
         assert text
         assert chunks[-1]["choices"][0].get("finish_reason") or True
+
+
+def test_cli_help_paths_render():
+    """Every registered family's --help renders without import or
+    signature errors (catches broken typer wiring)."""
+    from typer.testing import CliRunner
+
+    from helix_amd.cli import app as cli_app
+    runner = CliRunner()
+    out = runner.invoke(cli_app, ["--help"])
+    assert out.exit_code == 0, out.output
+    for fam in ("org", "project", "spectask", "mcp", "evals",
+                "sandbox", "billing", "fs", "user", "provider",
+                "knowledge", "secret", "session", "model", "app"):
+        r = runner.invoke(cli_app, [fam, "--help"])
+        assert r.exit_code == 0, f"{fam}: {r.output}"
