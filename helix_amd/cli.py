@@ -786,6 +786,32 @@ def provider_add(provider: str, base_url: str,
     typer.echo("added")
 
 
+image_app = typer.Typer(help="Image generation (diffusion models)")
+app.add_typer(image_app, name="image")
+
+
+@image_app.command("generate")
+def image_generate(prompt: str,
+                   model: str = typer.Option("flux-lite"),
+                   out: str = typer.Option("out.png", "--out", "-o"),
+                   n: int = typer.Option(1),
+                   size: str = typer.Option(""),
+                   steps: int = typer.Option(0),
+                   seed: int = typer.Option(-1)):
+    """Generate image(s) and write PNG files."""
+    import base64
+    resp = _client().images_generate(
+        prompt, model=model, n=n, size=size, steps=steps,
+        seed=None if seed < 0 else seed)
+    for i, item in enumerate(resp.get("data", [])):
+        path = out if n == 1 else \
+            out.replace(".png", f"-{i}.png") if out.endswith(".png") \
+            else f"{out}-{i}.png"
+        with open(path, "wb") as f:
+            f.write(base64.b64decode(item["b64_json"]))
+        typer.echo(path)
+
+
 billing_app = typer.Typer(help="Billing and usage")
 app.add_typer(billing_app, name="billing")
 

@@ -113,6 +113,21 @@ class HelixClient:
                         return
                     yield json.loads(data)
 
+    def images_generate(self, prompt: str, model: str = "",
+                        n: int = 1, size: str = "", steps: int = 0,
+                        seed: Optional[int] = None) -> dict:
+        """POST /v1/images/generations — returns b64_json PNGs."""
+        body: dict = {"prompt": prompt, "n": n}
+        if model:
+            body["model"] = model
+        if size:
+            body["size"] = size
+        if steps:
+            body["steps"] = steps
+        if seed is not None:
+            body["seed"] = seed
+        return self._post("/v1/images/generations", body)
+
     def list_sessions(self) -> List[dict]:
         return self._get("/api/v1/sessions")
 

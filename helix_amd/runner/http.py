@@ -17,7 +17,8 @@ from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, StreamingResponse
 
 from helix_amd.runner import gpudetect
-from helix_amd.runner.openai_adapter import chat_completion, embeddings
+from helix_amd.runner.openai_adapter import (chat_completion, embeddings,
+                                             images_generations)
 from helix_amd.runner.service import (ModelNotFoundError, NoCapacityError,
                                       RunnerService)
 
@@ -78,6 +79,10 @@ def create_runner_app(service: RunnerService,
     @app.post("/v1/embeddings")
     async def embed(request: Request):
         return await embeddings(service, await request.json())
+
+    @app.post("/v1/images/generations")
+    async def images(request: Request):
+        return await images_generations(service, await request.json())
 
     @app.get("/v1/models")
     async def models():

@@ -12,7 +12,8 @@ import uuid
 from typing import AsyncIterator, Dict, List, Optional
 
 from helix_amd.engine.sampling_params import SamplingParams
-from helix_amd.runner.service import LLMInstance, RunnerService
+from helix_amd.runner.service import (LLMInstance, ModelNotFoundError,
+                                      RunnerService)
 from helix_amd.utils.chat_templates import (parse_tool_calls,
                                              template_for_model)
 from helix_amd.utils.tokenizer import get_tokenizer
@@ -270,6 +271,32 @@ async def _stream(service, inst, model, rid, created, prompt_ids, params,
         # helix_openai_server.go:285-293 conn hard-close semantics).
         if not finished:
             inst.cancel(rid)
+
+
+async def images_generations(service: RunnerService, req: dict) -> dict:
+    """OpenAI images surface (reference delegates to a diffusers
+    container via `/v1/images/generations`, inferenceproxy/proxy.go:113).
+    Always returns b64_json (no public URL storage on a runner)."""
+    import base64
+    model = req.get("model", "flux-lite")
+    loop = asyncio.get_event_loop()
+    inst = await loop.run_in_executor(None, service.ensure_loaded, model)
+    if not hasattr(inst, "generate"):
+        raise ModelNotFoundError(f"{model} is not an image model")
+    n = min(int(req.get("n", 1) or 1), 8)
+    steps = min(int(req.get("steps", 8) or 8), 64)
+    seed = req.get("seed")
+    pngs = await loop.run_in_executor(
+        None, lambda: inst.generate(
+            str(req.get("prompt", "")), n=n, steps=steps,
+            seed=int(seed) if seed is not None else None,
+            size=req.get("size")))
+    return {
+        "created": int(time.time()),
+        "model": model,
+        "data": [{"b64_json": base64.b64encode(p).decode(),
+                  "revised_prompt": None} for p in pngs],
+    }
 
 
 async def embeddings(service: RunnerService, req: dict) -> dict:

@@ -54,6 +54,8 @@ DEFAULT_SPECS = {
     "siglip-base": ModelSpec("siglip-base", "vision", "siglip-base"),
     "tiny-vit": ModelSpec("tiny-vit", "vision", "tiny-vit"),
     "bge-large": ModelSpec("bge-large", "embedding", "bge-large"),
+    "flux-lite": ModelSpec("flux-lite", "image", "flux-lite"),
+    "tiny-dit": ModelSpec("tiny-dit", "image", "tiny-dit"),
     # tiny models for CPU tests
     "tiny": ModelSpec("tiny", "llm", "tiny", max_model_len=256,
                       kv_cache_blocks=256),
@@ -67,6 +69,14 @@ def estimate_model_bytes(spec: ModelSpec, block_size: int = 16) -> int:
     """Admission estimate: weights + KV budget + workspace headroom.
     Descendant of the reference's GGUF estimator (api/pkg/memory/estimate.go)
     re-based on bf16 dense checkpoints."""
+    if spec.kind == "image":
+        from helix_amd.models.dit import DIT_PRESETS
+        dcfg = DIT_PRESETS[spec.preset]
+        h = dcfg.hidden
+        per_layer = 18 * h * h          # qkv+o+mlp(8h^2)+adaLN mod(6h^2)
+        n = dcfg.vocab_size * h + dcfg.seq_len * h + \
+            dcfg.depth * per_layer + 40 * dcfg.vae_ch ** 2 * 9
+        return int(n * 2 * 1.3) + (1 << 30)
     if spec.kind == "vision":
         from helix_amd.models.vit import VIT_PRESETS
         vcfg = VIT_PRESETS[spec.preset]
@@ -308,6 +318,67 @@ class VisionEmbeddingInstance:
             torch.cuda.empty_cache()
 
 
+class ImageGenInstance:
+    """Diffusion image generator (the diffusers-container role of the
+    reference, SURVEY §2.8 last row): prompt -> PNG bytes via the
+    rectified-flow DiT pipeline (models/dit.py)."""
+
+    def __init__(self, spec: ModelSpec, device: str):
+        from helix_amd.models.dit import DIT_PRESETS, DiffusionImageModel
+        self.spec = spec
+        self.device = torch.device(device)
+        self.cfg = DIT_PRESETS[spec.preset]
+        dtype = torch.bfloat16 if self.device.type == "cuda" \
+            else torch.float32
+        self.model = DiffusionImageModel(self.cfg).to(dtype).to(
+            self.device)
+        self.model.init_random(0)
+        self.tokenizer = get_tokenizer(spec.name)
+        self.lock = threading.Lock()
+        self.last_used = time.time()
+
+    @property
+    def in_flight(self) -> int:
+        return 0
+
+    def generate(self, prompt: str, n: int = 1, steps: int = 8,
+                 seed: Optional[int] = None,
+                 size: Optional[str] = None) -> List[bytes]:
+        """-> PNG blobs. `size` ("WxH") resizes the preset's native
+        output; seed defaults to a fresh draw per request."""
+        import io as _io
+
+        from PIL import Image
+        self.last_used = time.time()
+        if seed is None:
+            seed = int.from_bytes(__import__("os").urandom(4), "little")
+        ids = self.tokenizer.encode(prompt)
+        with self.lock:
+            # distinct images per sample: consecutive seeds
+            imgs = [self.model.generate([ids], steps=steps,
+                                        seed=seed + i)[0]
+                    for i in range(max(1, int(n)))]
+        out = []
+        for img in imgs:
+            pil = Image.fromarray(
+                img.permute(1, 2, 0).cpu().numpy(), mode="RGB")
+            if size:
+                try:
+                    w, h = (int(v) for v in size.lower().split("x"))
+                    pil = pil.resize((w, h), Image.BILINEAR)
+                except ValueError:
+                    pass
+            buf = _io.BytesIO()
+            pil.save(buf, format="PNG")
+            out.append(buf.getvalue())
+        return out
+
+    def shutdown(self):
+        del self.model
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
+
 class RunnerService:
     """Owns model instances on one GPU (or CPU for tests); packs/evicts
     under the HBM budget."""
@@ -378,6 +449,8 @@ class RunnerService:
                 inst = EmbeddingInstance(spec, self.device)
             elif spec.kind == "vision":
                 inst = VisionEmbeddingInstance(spec, self.device)
+            elif spec.kind == "image":
+                inst = ImageGenInstance(spec, self.device)
             elif spec.tp > 1:
                 from helix_amd.runner.tp_instance import TPLLMInstance
                 inst = TPLLMInstance(spec, spec.tp)

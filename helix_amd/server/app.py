@@ -353,6 +353,19 @@ def create_app(cfg: Optional[ServerConfig] = None,
         client = providers.get_client(provider, user.id)
         return await client.embeddings({**body, "model": model})
 
+    @app.post("/v1/images/generations")
+    async def images_generations(request: Request,
+                                 user: AuthUser = Depends(auth_dep)):
+        """OpenAI images surface (SURVEY §2.8 last row: the reference
+        proxies this to a diffusers container; here it dispatches to a
+        runner's native rectified-flow DiT engine)."""
+        body = await request.json()
+        provider, model = providers.resolve(
+            body.get("model", ""), cfg.inference.default_provider,
+            user.id)
+        client = providers.get_client(provider, user.id)
+        return await client.images({**body, "model": model})
+
     @app.get("/v1/models")
     async def models(request: Request, user: AuthUser = Depends(auth_dep)):
         # anthropic-version header => Anthropic-style model list

@@ -38,6 +38,11 @@ class Client:
     async def embeddings(self, req: dict) -> dict:
         raise NotImplementedError
 
+    async def images(self, req: dict) -> dict:
+        raise ProviderError(
+            f"provider {self.provider or '?'} does not support image "
+            "generation", 501)
+
     async def list_models(self) -> List[str]:
         return []
 
@@ -102,6 +107,26 @@ class MockClient(Client):
                          "embedding": [random.random() for _ in range(8)]})
         return {"object": "list", "data": data, "model": req.get("model")}
 
+    async def images(self, req: dict) -> dict:
+        import base64
+        import struct
+        import zlib
+        self.calls.append(req)
+        # minimal valid 1x1 red PNG built by hand
+        def chunk(tag, payload):
+            return (struct.pack(">I", len(payload)) + tag + payload +
+                    struct.pack(">I", zlib.crc32(tag + payload)))
+        raw = zlib.compress(b"\x00\xff\x00\x00")
+        png = (b"\x89PNG\r\n\x1a\n" +
+               chunk(b"IHDR", struct.pack(">IIBBBBB", 1, 1, 8, 2,
+                                          0, 0, 0)) +
+               chunk(b"IDAT", raw) + chunk(b"IEND", b""))
+        n = max(1, int(req.get("n", 1) or 1))
+        return {"created": int(time.time()),
+                "model": req.get("model", "mock-image"),
+                "data": [{"b64_json": base64.b64encode(png).decode(),
+                          "revised_prompt": None}] * n}
+
     async def list_models(self) -> List[str]:
         return ["mock-model"]
 
@@ -147,6 +172,13 @@ class OpenAIHTTPClient(Client):
             raise ProviderError(r.text, r.status_code)
         return r.json()
 
+    async def images(self, req: dict) -> dict:
+        r = await self._http.post(f"{self.base_url}/images/generations",
+                                  json=req)
+        if r.status_code != 200:
+            raise ProviderError(r.text, r.status_code)
+        return r.json()
+
     async def list_models(self) -> List[str]:
         try:
             r = await self._http.get(f"{self.base_url}/models")
@@ -179,6 +211,10 @@ class LocalRunnerClient(Client):
     async def embeddings(self, req: dict) -> dict:
         from helix_amd.runner.openai_adapter import embeddings
         return await embeddings(self.service, req)
+
+    async def images(self, req: dict) -> dict:
+        from helix_amd.runner.openai_adapter import images_generations
+        return await images_generations(self.service, req)
 
     async def list_models(self) -> List[str]:
         return list(self.service.specs.keys())
@@ -246,6 +282,16 @@ class RouterClient(Client):
             raise ProviderError(r.text, r.status_code)
         return r.json()
 
+    async def images(self, req: dict) -> dict:
+        addr = self._pick(req.get("model", ""))
+        if not isinstance(addr, str):
+            return await addr.images(req)
+        r = await self._http.post(f"{addr}/v1/images/generations",
+                                  json=req)
+        if r.status_code != 200:
+            raise ProviderError(r.text, r.status_code)
+        return r.json()
+
     async def list_models(self) -> List[str]:
         return self.router.available_models()
 
@@ -276,6 +322,9 @@ class RetryableClient(Client):
 
     async def embeddings(self, req: dict) -> dict:
         return await self.inner.embeddings(req)
+
+    async def images(self, req: dict) -> dict:
+        return await self.inner.images(req)
 
     async def list_models(self) -> List[str]:
         return await self.inner.list_models()
@@ -361,6 +410,9 @@ class LoggingClient(Client):
 
     async def embeddings(self, req: dict) -> dict:
         return await self.inner.embeddings(req)
+
+    async def images(self, req: dict) -> dict:
+        return await self.inner.images(req)
 
     async def list_models(self) -> List[str]:
         return await self.inner.list_models()

@@ -115,9 +115,8 @@ class TunnelClient:
             elif ev["type"] in ("end", "response"):
                 return
 
-    async def embeddings(self, req: dict) -> dict:
-        rid = await self.registry.send_request(self.runner_id,
-                                               "/v1/embeddings", req)
+    async def _unary(self, path: str, req: dict) -> dict:
+        rid = await self.registry.send_request(self.runner_id, path, req)
         async for ev in self.registry.events(rid, self.timeout):
             if ev["type"] == "response":
                 return ev["data"]
@@ -127,6 +126,12 @@ class TunnelClient:
                                     ev.get("status", 502))
         from helix_amd.server.providers import ProviderError
         raise ProviderError("tunnel closed without response", 502)
+
+    async def embeddings(self, req: dict) -> dict:
+        return await self._unary("/v1/embeddings", req)
+
+    async def images(self, req: dict) -> dict:
+        return await self._unary("/v1/images/generations", req)
 
 
 # ---------------------------------------------------------------------------
@@ -168,7 +173,9 @@ async def tunnel_loop(api_url: str, runner_token: str, runner_id: str,
 
 
 async def _serve_one(http, api_url, headers, runner_id, service, msg):
-    from helix_amd.runner.openai_adapter import chat_completion, embeddings
+    from helix_amd.runner.openai_adapter import (chat_completion,
+                                                 embeddings,
+                                                 images_generations)
     rid = msg["id"]
     reply_url = f"{api_url}/api/v1/runner/tunnel/{runner_id}/reply"
 
@@ -180,6 +187,9 @@ async def _serve_one(http, api_url, headers, runner_id, service, msg):
         body = msg.get("body") or {}
         if msg.get("path") == "/v1/embeddings":
             out = await embeddings(service, body)
+            await reply({"type": "response", "data": out})
+        elif msg.get("path") == "/v1/images/generations":
+            out = await images_generations(service, body)
             await reply({"type": "response", "data": out})
         elif body.get("stream"):
             it = await chat_completion(service, body, request_id=rid)

@@ -124,6 +124,31 @@ def test_e2e_embeddings_via_router(e2e):
     assert len(r.json()["data"][0]["embedding"]) == 128
 
 
+def test_e2e_images_via_router(e2e):
+    base, key, svc = e2e
+    svc.ensure_loaded("tiny-dit")
+    r0 = httpx.get(f"{base}/api/v1/admin/runners",
+                   headers={"Authorization": "Bearer admin-key"}).json()
+    hb = {"runner_id": "runner-e2e", "address": r0[0]["address"],
+          "gpus": [], "models": [
+              {"model_id": "tiny", "state": "ready"},
+              {"model_id": "tiny-dit", "state": "ready"}]}
+    httpx.post(f"{base}/api/v1/runner/heartbeat", json=hb,
+               headers={"Authorization": "Bearer runner-token"})
+    r = httpx.post(f"{base}/v1/images/generations", json={
+        "model": "tiny-dit", "prompt": "a blue circle", "steps": 3,
+        "seed": 5}, timeout=120,
+        headers={"Authorization": f"Bearer {key}"})
+    assert r.status_code == 200, r.text
+    import base64
+    import io
+
+    from PIL import Image
+    img = Image.open(io.BytesIO(
+        base64.b64decode(r.json()["data"][0]["b64_json"])))
+    assert img.size == (32, 32)
+
+
 def test_e2e_models_endpoint(e2e):
     base, key, _ = e2e
     r = httpx.get(f"{base}/v1/models",

@@ -318,10 +318,16 @@ def test_switch_agent_and_step_info(stack):
                       headers=H(key)).status_code == 200
 
 
-def test_images_501_and_anthropic_model_list(stack):
+def test_images_route_and_anthropic_model_list(stack):
+    """/v1/images/generations is a real surface now (R2: native DiT
+    engine, tests/test_imagegen.py); via the mock provider it returns
+    b64 PNG data."""
     _, client, _, key, _ = stack
-    assert client.post("/v1/images/generations",
-                       headers=H(key)).status_code == 501
+    r = client.post("/v1/images/generations",
+                    json={"prompt": "a cat", "n": 2}, headers=H(key))
+    assert r.status_code == 200, r.text
+    assert len(r.json()["data"]) == 2
+    assert r.json()["data"][0]["b64_json"]
     r = client.get("/v1/models", headers={**H(key),
                                           "anthropic-version": "2023-06-01"})
     assert r.json()["has_more"] is False
