@@ -171,7 +171,11 @@ class DiTModel(nn.Module):
         super().__init__()
         self.cfg = cfg
         h = cfg.hidden
-        self.in_w = nn.Parameter(torch.empty(h, cfg.patch_dim))
+        # patch rows are zero-padded to a K multiple of 64 — the native
+        # MFMA GEMM's K-granularity (gemm_bf16.hip); padded weight
+        # columns multiply zeros and are inert
+        self.in_k = ((cfg.patch_dim + 63) // 64) * 64
+        self.in_w = nn.Parameter(torch.empty(h, self.in_k))
         self.in_b = nn.Parameter(torch.zeros(h))
         self.txt_emb = nn.Embedding(cfg.vocab_size, h)
         self.pos_emb = nn.Embedding(cfg.seq_len, h)
@@ -211,8 +215,10 @@ class DiTModel(nn.Module):
         B = lat.shape[0]
         P, Tt, L = cfg.num_patches, cfg.text_len, cfg.seq_len
         dtype = self.in_w.dtype
-        x_img = ops.gemm_bf16(self._patchify(lat).to(dtype),
-                              self.in_w, self.in_b)
+        rows = self._patchify(lat).to(dtype)
+        if self.in_k != cfg.patch_dim:
+            rows = F.pad(rows, (0, self.in_k - cfg.patch_dim))
+        x_img = ops.gemm_bf16(rows.contiguous(), self.in_w, self.in_b)
         tx = self.txt_emb(text_ids.clamp(0, cfg.vocab_size - 1))
         x = torch.cat([tx, x_img.view(B, P, -1)], dim=1).view(B * L, -1)
         pos = torch.arange(L, device=lat.device).repeat(B)
