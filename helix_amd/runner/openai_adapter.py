@@ -279,6 +279,10 @@ async def images_generations(service: RunnerService, req: dict) -> dict:
     Always returns b64_json (no public URL storage on a runner)."""
     import base64
     model = req.get("model", "flux-lite")
+    spec = service.specs.get(model)
+    if spec is not None and spec.kind != "image":
+        # refuse before admission — don't load an LLM just to 404
+        raise ModelNotFoundError(f"{model} is not an image model")
     loop = asyncio.get_event_loop()
     inst = await loop.run_in_executor(None, service.ensure_loaded, model)
     if not hasattr(inst, "generate"):
