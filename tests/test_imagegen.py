@@ -130,6 +130,27 @@ def test_provider_client_plumbing():
     assert img.size == (1, 1)
 
 
+def test_flux_lite_preset_shapes():
+    """The production preset (298M params, 288 tokens) builds and one
+    velocity forward has the right shape — catches preset-only shape
+    bugs the tiny preset can't see."""
+    cfg = DIT_PRESETS["flux-lite"]
+    pipe = DiffusionImageModel(cfg).init_random(0)
+    m = pipe.dit
+    n_params = sum(p.numel() for p in m.parameters())
+    assert 200e6 < n_params < 500e6
+    lat = torch.randn(1, cfg.latent_ch, cfg.latent_size,
+                      cfg.latent_size)
+    v = m(lat, torch.tensor([0.5]),
+          torch.zeros(1, cfg.text_len, dtype=torch.int64))
+    assert v.shape == (1, cfg.latent_ch, cfg.latent_size,
+                       cfg.latent_size)
+    assert torch.isfinite(v).all()
+    # the native GEMM needs K%64: every projection K in this preset
+    assert m.in_k % 64 == 0 and cfg.hidden % 64 == 0
+    assert cfg.time_dim % 64 == 0
+
+
 @pytest.mark.gpu
 def test_dit_generate_gpu():
     """tiny-dit on the GPU: bf16 weights, same CPU-drawn noise prior;
