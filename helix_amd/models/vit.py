@@ -8,8 +8,11 @@ attention + layer_norm + fused-GELU GEMMs) — no new kernels needed,
 which is the point: the encoder infra serves text AND vision.
 
 Weights are random-init offline (synthetic parity like every model
-here); `load_hf_vit` maps standard HF ViT/SigLIP safetensors names for
-deployments that ship a checkpoint.
+here). A faithful HF SigLIP import is deliberately NOT shipped: SigLIP
+is pre-LN while these blocks are post-LN (bge layout), so a name-map
+alone would silently change semantics; deployments with a checkpoint
+should read it via engine/weights.py `iter_safetensors` into a pre-LN
+variant first.
 """
 from __future__ import annotations
 
