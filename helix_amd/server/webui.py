@@ -48,6 +48,7 @@ INDEX_HTML = """<!doctype html>
   <button data-p="prj">Projects</button>
   <button data-p="org">Orgs</button>
   <button data-p="sbx">Sandboxes</button>
+  <button data-p="img">Images</button>
   <button data-p="use">Usage</button>
  </div>
  <button onclick="newSession()">+ new session</button>
@@ -115,6 +116,17 @@ INDEX_HTML = """<!doctype html>
    <pre id="sout" style="background:#111;color:#9f9;padding:8px;
      min-height:80px;white-space:pre-wrap"></pre>
   </div>
+ </div></div>
+ <div class="pane" id="p-img"><div class="lst">
+  <div class="card"><b>Image generation</b><br>
+   <input id="iprompt" size="40" placeholder="prompt"
+    onkeydown="if(event.key==='Enter')genImage()">
+   model <input id="imodel" size="10" value="flux-lite">
+   n <input id="inum" size="2" value="1">
+   <button onclick="genImage()">generate</button>
+   <span id="imsg"></span>
+  </div>
+  <div id="imgout" style="display:flex;gap:8px;flex-wrap:wrap"></div>
  </div></div>
  <div class="pane" id="p-use"><div class="lst" id="uselist"></div></div>
 </div>
@@ -391,6 +403,21 @@ async function runCmd(){
       (o.exit_code?('[exit '+o.exit_code+']\n'):'');}
   else $('sout').textContent+='[error '+r.status+']\n';
   $('scmd').value='';
+}
+async function genImage(){
+  $('imsg').textContent='generating...';
+  const r = await fetch('/v1/images/generations',{method:'POST',
+    headers:H(),body:JSON.stringify({prompt:$('iprompt').value,
+      model:$('imodel').value,n:parseInt($('inum').value)||1})});
+  if(!r.ok){$('imsg').textContent='error '+r.status;return;}
+  $('imsg').textContent='';
+  const out = $('imgout'); out.innerHTML='';
+  for(const d of (await r.json()).data){
+    const im=document.createElement('img');
+    im.src='data:image/png;base64,'+d.b64_json;
+    im.style.cssText='max-width:256px;border:1px solid #ddd;'+
+      'border-radius:8px';
+    out.appendChild(im);}
 }
 async function doSearch(){
   const q=$('srch').value.trim();if(!q)return;
