@@ -33,6 +33,10 @@ STATIC_MODEL_INFO: Dict[str, dict] = {
     "siglip-base": {"context_length": 0, "prompt_price_per_m": 0.01,
                     "completion_price_per_m": 0.0, "family": "siglip",
                     "runtime": "helix_amd", "kind": "vision-embedding"},
+    "flux-lite": {"context_length": 32, "prompt_price_per_m": 0.0,
+                  "completion_price_per_m": 0.0, "family": "flux",
+                  "runtime": "helix_amd", "kind": "image",
+                  "price_per_image": 0.002},
     "gpt-4o": {"context_length": 128000, "prompt_price_per_m": 2.5,
                "completion_price_per_m": 10.0, "family": "openai",
                "runtime": "external", "kind": "chat"},
