@@ -287,8 +287,8 @@ async def images_generations(service: RunnerService, req: dict) -> dict:
     inst = await loop.run_in_executor(None, service.ensure_loaded, model)
     if not hasattr(inst, "generate"):
         raise ModelNotFoundError(f"{model} is not an image model")
-    n = min(int(req.get("n", 1) or 1), 8)
-    steps = min(int(req.get("steps", 8) or 8), 64)
+    n = max(1, min(int(req.get("n", 1) or 1), 8))
+    steps = max(1, min(int(req.get("steps", 8) or 8), 64))
     seed = req.get("seed")
     pngs = await loop.run_in_executor(
         None, lambda: inst.generate(
