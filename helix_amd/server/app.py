@@ -378,13 +378,6 @@ def create_app(cfg: Optional[ServerConfig] = None,
         data = await providers.aggregate_models(user.id)
         return {"object": "list", "data": data}
 
-    @app.post("/v1/images/generations")
-    async def images(user: AuthUser = Depends(auth_dep)):
-        # kept returning 501 until a diffusion engine exists
-        # (SURVEY.md §2.8 "Image generation ... out of MVP")
-        raise HTTPException(501, "image generation is not implemented in "
-                                 "the MI355X runner yet")
-
     # ------------------------------------------------------------------
     # Sessions API (reference server.go:1024-1064)
     # ------------------------------------------------------------------
