@@ -100,3 +100,28 @@ def test_tunnel_embeddings(tunnel_stack):
         headers={"Authorization": f"Bearer {key}"})
     assert r.status_code == 200, r.text
     assert len(r.json()["data"][0]["embedding"]) == 128
+
+
+def test_tunnel_images(tunnel_stack):
+    """images/generations over the reverse tunnel (server/tunnel.py
+    `_unary` + `_serve_one` image branch)."""
+    base, key, app = tunnel_stack
+    app.state.runner_service = None  # force router -> tunnel path
+    httpx.post(f"{base}/api/v1/runner/heartbeat", json={
+        "runner_id": "nat-runner", "address": "tunnel:nat-runner",
+        "gpus": [], "models": [{"model_id": "tiny", "state": "ready"},
+                               {"model_id": "tiny-dit",
+                                "state": "ready"}]},
+        headers={"Authorization": "Bearer runner-token"})
+    r = httpx.post(f"{base}/v1/images/generations", json={
+        "model": "tiny-dit", "prompt": "tunnel image", "steps": 3,
+        "seed": 2}, timeout=120,
+        headers={"Authorization": f"Bearer {key}"})
+    assert r.status_code == 200, r.text
+    import base64
+    import io
+
+    from PIL import Image
+    img = Image.open(io.BytesIO(
+        base64.b64decode(r.json()["data"][0]["b64_json"])))
+    assert img.size == (32, 32)
