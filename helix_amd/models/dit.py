@@ -109,10 +109,15 @@ class TinyVAE(nn.Module):
             nn.Conv2d(c, 3, 3, padding=1))
 
     def encode(self, images: torch.Tensor) -> torch.Tensor:
-        return self.enc(images)
+        # NHWC: MIOpen's fast bf16 conv path — NCHW bf16 falls back to
+        # a naive kernel that dominated the whole pipeline (rocprof:
+        # 354 of 450 ms; profiles/r02_imagegen.md)
+        return self.enc(images.contiguous(
+            memory_format=torch.channels_last))
 
     def decode(self, latents: torch.Tensor) -> torch.Tensor:
-        return self.dec(latents)
+        return self.dec(latents.contiguous(
+            memory_format=torch.channels_last)).contiguous()
 
 
 class DiTBlock(nn.Module):
@@ -276,7 +281,7 @@ class DiffusionImageModel(nn.Module):
         super().__init__()
         self.cfg = cfg
         self.dit = DiTModel(cfg)
-        self.vae = TinyVAE(cfg)
+        self.vae = TinyVAE(cfg).to(memory_format=torch.channels_last)
 
     @torch.inference_mode()
     def generate(self, prompts_ids: List[List[int]], steps: int = 8,
