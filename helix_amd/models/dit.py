@@ -109,9 +109,10 @@ class TinyVAE(nn.Module):
             nn.Conv2d(c, 3, 3, padding=1))
 
     def encode(self, images: torch.Tensor) -> torch.Tensor:
-        # NHWC: MIOpen's fast bf16 conv path — NCHW bf16 falls back to
-        # a naive kernel that dominated the whole pipeline (rocprof:
-        # 354 of 450 ms; profiles/r02_imagegen.md)
+        # NHWC keeps MIOpen on its igemm bf16 conv path from the first
+        # call (NCHW bf16 runs a naive kernel until MIOpen's find
+        # completes; e2e steady state measured identical — see
+        # profiles/r02_imagegen.md postscript)
         return self.enc(images.contiguous(
             memory_format=torch.channels_last))
 
